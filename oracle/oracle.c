@@ -1,0 +1,546 @@
+/* oracle.c — CPU restatement of the StarRocks BE hot-path algorithms.
+ *
+ * TEST INFRASTRUCTURE ONLY (see oracle.h header comment and DESIGN.md §2).
+ *
+ * Each section cites the reference file:line (relative to /root/reference)
+ * whose algorithm it restates. No reference source is copied; the algorithms
+ * are re-derived from their published behaviour and pinned by the reference's
+ * own known-answer tests (ported in tests/test_oracle_golden.py).
+ *
+ * Build: gcc -O3 -mavx2 -fopenmp -shared -fPIC oracle.c -o liboracle.so
+ */
+#include "oracle.h"
+#include <stdlib.h>
+#include <string.h>
+#ifdef _OPENMP
+#include <omp.h>
+#endif
+
+/* ====================================================================== */
+/* Hashes                                                                  */
+/* ====================================================================== */
+
+/* CRC32-C (Castagnoli), byte-at-a-time table, matching x86 _mm_crc32_u8/u32
+ * semantics used by crc_hash_32 (be/src/base/hash/hash.h:96-132): words are
+ * consumed 4 bytes LSB-first (== _mm_crc32_u32 of the LE load), then the
+ * byte tail. Reflected polynomial 0x82F63B78. */
+static uint32_t crc32c_table[256];
+static int crc32c_init_done = 0;
+static void crc32c_init(void) {
+    if (crc32c_init_done) return;
+    for (uint32_t i = 0; i < 256; i++) {
+        uint32_t c = i;
+        for (int k = 0; k < 8; k++) c = (c >> 1) ^ (0x82F63B78u & (uint32_t) - (int)(c & 1));
+        crc32c_table[i] = c;
+    }
+    crc32c_init_done = 1;
+}
+static inline uint32_t crc32c_u8(uint32_t crc, uint8_t b) {
+    return (crc >> 8) ^ crc32c_table[(crc ^ b) & 0xFF];
+}
+
+/* phmap_mix<4> (be/src/base/hash/hash.h:26-34): l = a*0xcc9e2d51; l ^ l>>32 */
+static inline uint64_t phmap_mix4(uint64_t a) {
+    uint64_t l = a * 0xcc9e2d51ull;
+    return l ^ (l >> 32);
+}
+
+uint32_t orc_crc_hash_32(const void* data, int32_t bytes, uint32_t seed) {
+    crc32c_init();
+    const uint8_t* p = (const uint8_t*)data;
+    uint32_t h = seed;
+    int32_t words = bytes / 4;
+    int32_t tail = bytes % 4;
+    while (words--) {
+        h = crc32c_u8(h, p[0]); h = crc32c_u8(h, p[1]);
+        h = crc32c_u8(h, p[2]); h = crc32c_u8(h, p[3]);
+        p += 4;
+    }
+    while (tail--) { h = crc32c_u8(h, *p++); }
+    return (uint32_t)phmap_mix4(h);
+}
+
+/* HashUtil::fnv_hash (be/src/base/hash/hash_util.hpp:133-143) */
+uint32_t orc_fnv_hash(const void* data, int32_t bytes, uint32_t seed) {
+    const uint8_t* p = (const uint8_t*)data;
+    uint32_t h = seed;
+    while (bytes--) { h = (*p ^ h) * 16777619u; ++p; }
+    return h;
+}
+
+/* HashUtil::xorshift32 (be/src/base/hash/hash_util.hpp:240-246) */
+uint32_t orc_xorshift32(uint32_t x) {
+    x ^= x << 13; x ^= x >> 17; x ^= x << 5;
+    return x;
+}
+
+/* JoinKeyHash<T,4> (be/src/exec/join/join_hash_map_helper.h:34-44) */
+uint32_t orc_join_hash_u32(uint32_t v, uint32_t num_log_buckets) {
+    v ^= v >> (32 - num_log_buckets);
+    return (v * 2654435761u) >> (32 - num_log_buckets);
+}
+
+/* JoinKeyHash<T,8> (join_hash_map_helper.h:46-55) */
+uint32_t orc_join_hash_u64(uint64_t v, uint32_t num_log_buckets) {
+    v ^= v >> (64 - num_log_buckets);
+    return (uint32_t)((v * 11400714819323198485ull) >> (64 - num_log_buckets));
+}
+
+/* JoinKeyHash<Slice> (join_hash_map_helper.h:57-64), CRC_SEED 0x811C9DC5 */
+uint32_t orc_join_hash_slice(const void* p, int32_t n, uint32_t num_buckets) {
+    return orc_crc_hash_32(p, n, 0x811C9DC5u) & (num_buckets - 1);
+}
+
+/* JoinHashMapHelper::calc_bucket_size (join_hash_map_helper.h:68-78):
+ * NormalizeCapacity(n + (n-1)/4) + 1 with phmap NormalizeCapacity(n) =
+ * ~size_t{} >> LeadingZeros(n) (be/src/base/phmap/phmap.h:485-487). */
+uint32_t orc_calc_bucket_size(uint32_t size) {
+    uint64_t expect = (uint64_t)size + (size - 1) / 4;
+    if (expect >= (1ull << 31)) return 1u << 31;
+    uint64_t norm = expect ? (~0ull >> __builtin_clzll(expect)) : 1;
+    return (uint32_t)(norm + 1);
+}
+
+/* ====================================================================== */
+/* Deterministic synthetic data — splitmix64 finalizer, counter-based.     */
+/* Shared definition with the HIP generator (csrc/gpue_kernels.hip) and    */
+/* the numpy reimplementation in tests — all three must stay identical.    */
+/* ====================================================================== */
+
+static inline uint64_t sm64(uint64_t x) {
+    x += 0x9E3779B97F4A7C15ull;
+    x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+    x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+    return x ^ (x >> 31);
+}
+
+uint64_t orc_gen_u64(uint64_t seed, uint64_t tag, uint64_t i) {
+    return sm64(seed + tag * 0x9E3779B97F4A7C15ull + i);
+}
+
+void orc_gen_u32_mod(uint64_t seed, uint64_t tag, uint64_t start, uint64_t n,
+                     uint32_t mod, uint32_t add, uint32_t* out) {
+#pragma omp parallel for schedule(static)
+    for (uint64_t i = 0; i < n; i++) {
+        uint64_t v = orc_gen_u64(seed, tag, start + i);
+        out[i] = (mod ? (uint32_t)(v % mod) : (uint32_t)v) + add;
+    }
+}
+
+void orc_gen_i64(uint64_t seed, uint64_t tag, uint64_t start, uint64_t n, int64_t* out) {
+#pragma omp parallel for schedule(static)
+    for (uint64_t i = 0; i < n; i++) out[i] = (int64_t)orc_gen_u64(seed, tag, start + i);
+}
+
+/* Gregorian calendar from 1992-01-01; d_datekey = y*10000 + m*100 + d.
+ * SSB date dim: 2556 days (test/common/sql/ssb/create.sql date table). */
+static const int MDAYS[12] = {31, 28, 31, 30, 31, 30, 31, 31, 30, 31, 30, 31};
+static inline int is_leap(int y) { return (y % 4 == 0 && y % 100 != 0) || y % 400 == 0; }
+void orc_gen_dates(int32_t n_days, int32_t* datekey, int32_t* dyear) {
+    int y = 1992, m = 1, d = 1;
+    for (int32_t i = 0; i < n_days; i++) {
+        datekey[i] = y * 10000 + m * 100 + d;
+        if (dyear) dyear[i] = y;
+        int md = MDAYS[m - 1] + (m == 2 && is_leap(y));
+        if (++d > md) { d = 1; if (++m > 12) { m = 1; y++; } }
+    }
+}
+
+/* ====================================================================== */
+/* Join hash map methods                                                   */
+/* ====================================================================== */
+
+/* BucketChainedJoinHashMap::construct_hash_table
+ * (join_hash_map_method.hpp:37-85, non-null path): two passes — hash all
+ * keys into next[], then scatter next[i]=first[b]; first[b]=i. Rows 1-based. */
+void orc_bucket_chained_build_u32(const uint32_t* keys, uint32_t row_count,
+                                  uint32_t* first, uint32_t* next,
+                                  uint32_t bucket_size, uint32_t log_bucket_size) {
+    (void)bucket_size;
+    const uint32_t num_rows = row_count + 1;
+    for (uint32_t i = 1; i < num_rows; i++)
+        next[i] = orc_join_hash_u32(keys[i], log_bucket_size);
+    for (uint32_t i = 1; i < num_rows; i++) {
+        uint32_t b = next[i];
+        next[i] = first[b];
+        first[b] = i;
+    }
+}
+
+/* BucketChainedJoinHashMap::lookup_init (join_hash_map_method.hpp:88-120) */
+void orc_bucket_chained_lookup_u32(const uint32_t* probe_keys, uint32_t probe_rows,
+                                   const uint32_t* first, uint32_t bucket_size,
+                                   uint32_t log_bucket_size, uint32_t* heads) {
+    (void)bucket_size;
+    for (uint32_t i = 0; i < probe_rows; i++)
+        heads[i] = first[orc_join_hash_u32(probe_keys[i], log_bucket_size)];
+}
+
+/* TLinearChainedJoinHashMap (join_hash_map_method.h:118-150,
+ * join_hash_map_method.hpp:125-368): FP_BITS=8; hash in space
+ * bucket_size<<8 / log+8; first[b] = (hash<<24 fp) | 24-bit row index;
+ * linear probing with triangular increment; chains via next[]. */
+#define LC_FP_BITS 8u
+#define LC_DATA_MASK 0x00FFFFFFu
+static inline uint32_t lc_fp(uint32_t hash) { return hash << (32 - LC_FP_BITS); }
+static inline uint32_t lc_bucket(uint32_t hash) { return hash >> LC_FP_BITS; }
+
+void orc_linear_chained_build_u32(const uint32_t* keys, uint32_t row_count,
+                                  uint32_t* first, uint32_t* next,
+                                  uint32_t bucket_size, uint32_t log_bucket_size) {
+    const uint32_t num_rows = row_count + 1;
+    const uint32_t mask = bucket_size - 1;
+    for (uint32_t i = 1; i < num_rows; i++)
+        next[i] = orc_join_hash_u32(keys[i], log_bucket_size + LC_FP_BITS);
+    for (uint32_t i = 1; i < num_rows; i++) {
+        const uint32_t hash = next[i];
+        const uint32_t fp = lc_fp(hash);
+        uint32_t b = lc_bucket(hash);
+        uint32_t probe_times = 1;
+        for (;;) {
+            if (first[b] == 0) {
+                next[i] = 0;
+                first[b] = fp | i;
+                break;
+            }
+            if (fp == (first[b] & ~LC_DATA_MASK) && keys[i] == keys[first[b] & LC_DATA_MASK]) {
+                next[i] = first[b] & LC_DATA_MASK;
+                first[b] = fp | i;
+                break;
+            }
+            b = (b + probe_times) & mask;
+            probe_times++;
+        }
+    }
+}
+
+void orc_linear_chained_lookup_u32(const uint32_t* build_keys, const uint32_t* probe_keys,
+                                   uint32_t probe_rows, const uint32_t* first,
+                                   uint32_t bucket_size, uint32_t log_bucket_size,
+                                   uint32_t* heads) {
+    const uint32_t mask = bucket_size - 1;
+    for (uint32_t i = 0; i < probe_rows; i++) {
+        const uint32_t hash = orc_join_hash_u32(probe_keys[i], log_bucket_size + LC_FP_BITS);
+        const uint32_t fp = lc_fp(hash);
+        uint32_t b = lc_bucket(hash);
+        uint32_t probe_times = 1;
+        for (;;) {
+            if (first[b] == 0) { heads[i] = 0; break; }
+            const uint32_t cur = first[b];
+            if (fp == (cur & ~LC_DATA_MASK) && probe_keys[i] == build_keys[cur & LC_DATA_MASK]) {
+                heads[i] = cur & LC_DATA_MASK;
+                break;
+            }
+            b = (b + probe_times) & mask;
+            probe_times++;
+        }
+    }
+}
+
+/* RangeDirectMappingJoinHashMap (join_hash_map_method.hpp:625-707):
+ * first[key - min] = i (chained via next), probe checks [min,max]. */
+void orc_range_direct_build_i32(const int32_t* keys, uint32_t row_count,
+                                int64_t min_value, uint32_t* first, uint32_t* next) {
+    const uint32_t num_rows = row_count + 1;
+    for (uint32_t i = 1; i < num_rows; i++) {
+        const uint64_t b = (uint64_t)((int64_t)keys[i] - min_value);
+        next[i] = first[b];
+        first[b] = i;
+    }
+}
+
+void orc_range_direct_lookup_i32(const int32_t* probe_keys, uint64_t probe_rows,
+                                 int64_t min_value, int64_t max_value,
+                                 const uint32_t* first, uint32_t* heads) {
+    for (uint64_t i = 0; i < probe_rows; i++) {
+        if (probe_keys[i] >= min_value && probe_keys[i] <= max_value)
+            heads[i] = first[probe_keys[i] - min_value];
+        else
+            heads[i] = 0;
+    }
+}
+
+/* _probe_from_ht (join_hash_map.hpp:717-795): per probe row walk the chain,
+ * emit (probe_idx, build_idx) per key-equal build row. Emit-all variant —
+ * the chunk_size-resumable cursor is an iteration detail of the CPU's
+ * bounded output chunks; the emitted multiset is the result. */
+uint64_t orc_probe_emit_u32(const uint32_t* build_keys, const uint32_t* next,
+                            const uint32_t* probe_keys, const uint32_t* heads,
+                            uint32_t probe_rows, int collision_free,
+                            uint32_t* out_probe_idx, uint32_t* out_build_idx) {
+    uint64_t m = 0;
+    for (uint32_t i = 0; i < probe_rows; i++) {
+        uint32_t b = heads[i];
+        if (b == 0) continue;
+        if (collision_free) {
+            if (build_keys[b] == probe_keys[i]) {
+                out_probe_idx[m] = i; out_build_idx[m] = b; m++;
+            }
+            continue;
+        }
+        do {
+            if (build_keys[b] == probe_keys[i]) {
+                out_probe_idx[m] = i; out_build_idx[m] = b; m++;
+            }
+            b = next[b];
+        } while (b != 0);
+    }
+    return m;
+}
+
+/* ====================================================================== */
+/* Predicate filter                                                        */
+/* ====================================================================== */
+
+/* eval_conjuncts + Column::filter_range stream compaction
+ * (be/src/exprs/chunk_predicate_evaluator.cpp:31-80,
+ *  be/src/base/simd/filter.h:26-38): stable, order-preserving. */
+uint64_t orc_filter_i64_lt(const int64_t* in, uint64_t n, int64_t theta, int64_t* out) {
+    uint64_t k = 0;
+    for (uint64_t i = 0; i < n; i++)
+        if (in[i] < theta) out[k++] = in[i];
+    return k;
+}
+
+uint64_t orc_filter_i64_lt_mt(const int64_t* in, uint64_t n, int64_t theta, int64_t* out) {
+#ifdef _OPENMP
+    int nt = omp_get_max_threads();
+#else
+    int nt = 1;
+#endif
+    uint64_t* counts = (uint64_t*)calloc(nt + 1, sizeof(uint64_t));
+    uint64_t chunk = (n + nt - 1) / nt;
+#pragma omp parallel num_threads(nt)
+    {
+#ifdef _OPENMP
+        int t = omp_get_thread_num();
+#else
+        int t = 0;
+#endif
+        uint64_t lo = (uint64_t)t * chunk, hi = lo + chunk;
+        if (hi > n) hi = n;
+        uint64_t c = 0;
+        for (uint64_t i = lo; i < hi; i++) c += (in[i] < theta);
+        counts[t + 1] = c;
+#pragma omp barrier
+#pragma omp single
+        { for (int j = 1; j <= nt; j++) counts[j] += counts[j - 1]; }
+        uint64_t k = counts[t];
+        for (uint64_t i = lo; i < hi; i++)
+            if (in[i] < theta) out[k++] = in[i];
+    }
+    uint64_t total = counts[nt];
+    free(counts);
+    return total;
+}
+
+/* ====================================================================== */
+/* Exchange partition                                                      */
+/* ====================================================================== */
+
+/* ExchangeSinkOperator hash path (exchange_sink_operator.cpp:602-627,
+ * default _exchange_hash_function_version=0 ⇒ fnv,
+ * InternalService.thrift:378) + Shuffler HASH_PARTITIONED non-compat
+ * ReduceOp (shuffler.h:71-86, hash_util.hpp:251-262). */
+void orc_partition_channel_u32(const uint32_t* keys, uint64_t n, uint32_t num_channels,
+                               uint32_t* channel_ids) {
+#pragma omp parallel for schedule(static)
+    for (uint64_t i = 0; i < n; i++) {
+        uint32_t h = orc_fnv_hash(&keys[i], 4, 0x811C9DC5u); /* FNV_SEED */
+        channel_ids[i] = (uint32_t)(((uint64_t)h * num_channels) >> 32); /* ReduceOp */
+    }
+}
+
+/* counting-sort row layout (exchange_sink_operator.cpp:629-660): forward
+ * count, prefix-sum, then reverse iteration emit so each channel's rows stay
+ * in ascending source order. */
+void orc_partition_counting_sort(const uint32_t* channel_ids, uint64_t n,
+                                 uint32_t num_channels, uint64_t* start_points,
+                                 uint32_t* row_indexes) {
+    memset(start_points, 0, (num_channels + 1) * sizeof(uint64_t));
+    for (uint64_t i = 0; i < n; i++) start_points[channel_ids[i]]++;
+    for (uint32_t c = 1; c <= num_channels; c++) start_points[c] += start_points[c - 1];
+    for (int64_t i = (int64_t)n - 1; i >= 0; i--) {
+        row_indexes[start_points[channel_ids[i]] - 1] = (uint32_t)i;
+        start_points[channel_ids[i]]--;
+    }
+    /* start_points now holds start offsets per channel, [num_channels]==begin
+     * of last channel; recompute ends by re-adding counts */
+    for (uint32_t c = num_channels; c > 0; c--) start_points[c] = start_points[c - 1];
+    start_points[0] = 0;
+    for (uint64_t i = 0; i < n; i++) start_points[channel_ids[i] + 1]++;
+    for (uint32_t c = 1; c <= num_channels; c++) start_points[c] += start_points[c - 1];
+}
+
+/* ====================================================================== */
+/* SSB synthetic columns + dim builds (shared schema with the GPU engine)  */
+/* ====================================================================== */
+
+enum { TAG_ORDERDATE = 1, TAG_EXTPRICE = 2, TAG_DISCOUNT = 3,
+       TAG_PARTKEY = 4, TAG_SUPPKEY = 5, TAG_REVENUE = 6,
+       TAG_PCAT = 7, TAG_PBRD = 8, TAG_SREG = 9 };
+
+#define N_DAYS 2556
+#define N_PARTS_SF100 1400000u
+#define N_SUPPS_SF100 200000u
+
+void orc_gen_lineorder_q1(uint64_t seed, uint64_t row_start, uint64_t n,
+                          int32_t* lo_orderdate, int32_t* lo_extendedprice,
+                          int32_t* lo_discount) {
+    int32_t datekey[N_DAYS];
+    orc_gen_dates(N_DAYS, datekey, NULL);
+#pragma omp parallel for schedule(static)
+    for (uint64_t i = 0; i < n; i++) {
+        uint64_t r = row_start + i;
+        lo_orderdate[i] = datekey[orc_gen_u64(seed, TAG_ORDERDATE, r) % N_DAYS];
+        lo_extendedprice[i] = (int32_t)(orc_gen_u64(seed, TAG_EXTPRICE, r) % 100000u) + 1;
+        lo_discount[i] = (int32_t)(orc_gen_u64(seed, TAG_DISCOUNT, r) % 11u);
+    }
+}
+
+void orc_gen_lineorder_q21(uint64_t seed, uint64_t row_start, uint64_t n,
+                           int32_t* lo_partkey, int32_t* lo_suppkey,
+                           int32_t* lo_orderdate, int32_t* lo_revenue) {
+    int32_t datekey[N_DAYS];
+    orc_gen_dates(N_DAYS, datekey, NULL);
+#pragma omp parallel for schedule(static)
+    for (uint64_t i = 0; i < n; i++) {
+        uint64_t r = row_start + i;
+        lo_partkey[i] = (int32_t)(orc_gen_u64(seed, TAG_PARTKEY, r) % N_PARTS_SF100) + 1;
+        lo_suppkey[i] = (int32_t)(orc_gen_u64(seed, TAG_SUPPKEY, r) % N_SUPPS_SF100) + 1;
+        lo_orderdate[i] = datekey[orc_gen_u64(seed, TAG_ORDERDATE, r) % N_DAYS];
+        lo_revenue[i] = (int32_t)(orc_gen_u64(seed, TAG_REVENUE, r) % 10000000u);
+    }
+}
+
+uint32_t orc_part_category(uint64_t seed, uint32_t partkey) {
+    return (uint32_t)(orc_gen_u64(seed, TAG_PCAT, partkey) % 25u);
+}
+uint32_t orc_part_brand(uint64_t seed, uint32_t partkey) {
+    return orc_part_category(seed, partkey) * 40u +
+           (uint32_t)(orc_gen_u64(seed, TAG_PBRD, partkey) % 40u);
+}
+uint32_t orc_supp_region(uint64_t seed, uint32_t suppkey) {
+    return (uint32_t)(orc_gen_u64(seed, TAG_SREG, suppkey) % 5u);
+}
+
+/* Direct-mapped payload dim arrays: the reference's
+ * RANGE_DIRECT_MAPPING (join_hash_table.cpp:263-321) with the probe-side
+ * filter + payload gather folded into the `first` value. */
+void orc_build_date_dim(int32_t n_days, int32_t year_filter, int32_t* min_key,
+                        int32_t* max_key, uint32_t** first_out, uint32_t* size_out) {
+    int32_t* datekey = (int32_t*)malloc(n_days * sizeof(int32_t));
+    int32_t* dyear = (int32_t*)malloc(n_days * sizeof(int32_t));
+    orc_gen_dates(n_days, datekey, dyear);
+    int32_t mn = datekey[0], mx = datekey[n_days - 1];
+    uint32_t size = (uint32_t)(mx - mn + 1);
+    uint32_t* first = (uint32_t*)calloc(size, sizeof(uint32_t));
+    for (int32_t i = 0; i < n_days; i++) {
+        if (year_filter < 0 || dyear[i] == year_filter)
+            first[datekey[i] - mn] = (uint32_t)(dyear[i] - 1992) + 1u;
+    }
+    *min_key = mn; *max_key = mx; *first_out = first; *size_out = size;
+    free(datekey); free(dyear);
+}
+
+void orc_build_part_dim(uint64_t seed, uint32_t n_parts, int32_t category, uint32_t* first) {
+#pragma omp parallel for schedule(static)
+    for (uint32_t p = 1; p <= n_parts; p++) {
+        uint32_t cat = orc_part_category(seed, p);
+        first[p - 1] = (cat == (uint32_t)category) ? orc_part_brand(seed, p) + 1u : 0u;
+    }
+}
+
+void orc_build_supp_dim(uint64_t seed, uint32_t n_supps, int32_t region, uint32_t* first) {
+#pragma omp parallel for schedule(static)
+    for (uint32_t s = 1; s <= n_supps; s++)
+        first[s - 1] = (orc_supp_region(seed, s) == (uint32_t)region) ? 1u : 0u;
+}
+
+/* ====================================================================== */
+/* Fused pipelines (timed CPU baselines)                                   */
+/* ====================================================================== */
+
+/* Config 2: date dim build (filter d_year==year) + probe + SUM, following
+ * JoinHashTable::probe fast path (join_hash_map.hpp:752-761) +
+ * AggregateFunction SUM update_batch (be/src/exprs/agg/sum.h:45-181).
+ * Generation is NOT timed by callers: columns are materialised first. */
+int64_t orc_q1_pipeline(uint64_t seed, uint64_t row_start, uint64_t n_rows,
+                        int32_t year, int threads, uint64_t* match_count) {
+    int32_t* od = (int32_t*)malloc(n_rows * 4);
+    int32_t* ep = (int32_t*)malloc(n_rows * 4);
+    int32_t* dc = (int32_t*)malloc(n_rows * 4);
+    orc_gen_lineorder_q1(seed, row_start, n_rows, od, ep, dc);
+
+    int32_t mn, mx; uint32_t *first, size;
+    orc_build_date_dim(N_DAYS, year, &mn, &mx, &first, &size);
+
+#ifdef _OPENMP
+    if (threads > 0) omp_set_num_threads(threads);
+#endif
+    int64_t sum = 0; uint64_t matches = 0;
+#pragma omp parallel for schedule(static) reduction(+:sum) reduction(+:matches)
+    for (uint64_t i = 0; i < n_rows; i++) {
+        int32_t k = od[i];
+        if (k >= mn && k <= mx && first[k - mn] != 0) {
+            sum += (int64_t)ep[i] * dc[i];
+            matches++;
+        }
+    }
+    if (match_count) *match_count = matches;
+    free(od); free(ep); free(dc); free(first);
+    return sum;
+}
+
+/* Config 3: SSB Q2.1-shaped 3-way star probe + GROUP BY (d_year,p_brand),
+ * group id = (d_year-1992)*1000 + p_brand; SUM(lo_revenue) per group.
+ * Mirrors chained probe + Aggregator::compute_batch_agg_states
+ * (be/src/exec/aggregator.cpp:937-959) with a dense group space. */
+void orc_q21_pipeline(uint64_t seed, uint64_t row_start, uint64_t n_rows,
+                      int32_t category, int32_t region, int threads,
+                      int64_t* group_sums) {
+    int32_t* pk = (int32_t*)malloc(n_rows * 4);
+    int32_t* sk = (int32_t*)malloc(n_rows * 4);
+    int32_t* od = (int32_t*)malloc(n_rows * 4);
+    int32_t* rv = (int32_t*)malloc(n_rows * 4);
+    orc_gen_lineorder_q21(seed, row_start, n_rows, pk, sk, od, rv);
+
+    int32_t mn, mx; uint32_t *dfirst, dsize;
+    orc_build_date_dim(N_DAYS, -1, &mn, &mx, &dfirst, &dsize);
+    uint32_t* pfirst = (uint32_t*)malloc(N_PARTS_SF100 * sizeof(uint32_t));
+    orc_build_part_dim(seed, N_PARTS_SF100, category, pfirst);
+    uint32_t* sfirst = (uint32_t*)malloc(N_SUPPS_SF100 * sizeof(uint32_t));
+    orc_build_supp_dim(seed, N_SUPPS_SF100, region, sfirst);
+
+#ifdef _OPENMP
+    if (threads > 0) omp_set_num_threads(threads);
+    int nt = omp_get_max_threads();
+#else
+    int nt = 1;
+#endif
+    const int NG = 7 * 1000;
+    int64_t* partials = (int64_t*)calloc((size_t)nt * NG, sizeof(int64_t));
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        int t = omp_get_thread_num();
+#else
+        int t = 0;
+#endif
+        int64_t* local = partials + (size_t)t * NG;
+#pragma omp for schedule(static)
+        for (uint64_t i = 0; i < n_rows; i++) {
+            uint32_t brand1 = pfirst[pk[i] - 1];
+            if (brand1 == 0) continue;
+            if (sfirst[sk[i] - 1] == 0) continue;
+            uint32_t year1 = dfirst[od[i] - mn]; /* od always in range */
+            local[(year1 - 1) * 1000 + (brand1 - 1)] += rv[i];
+        }
+    }
+    for (int t = 0; t < nt; t++)
+        for (int g = 0; g < NG; g++) group_sums[g] += partials[(size_t)t * NG + g];
+    free(partials); free(pk); free(sk); free(od); free(rv);
+    free(dfirst); free(pfirst); free(sfirst);
+}
+
+void orc_free(void* p) { free(p); }

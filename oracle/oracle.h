@@ -1,0 +1,137 @@
+/* oracle.h — CPU restatement of the StarRocks BE hot-path algorithms.
+ *
+ * TEST INFRASTRUCTURE ONLY. Per DESIGN.md §2, only tests/, __graft_entry__'s
+ * smoke() and bench.py's cpu_baseline leg may load this library. The product
+ * path (starrocks_amd/ + libgpue.so) must never call into it.
+ *
+ * Every function cites the reference source (path:line relative to
+ * /root/reference) whose algorithm it restates. Parity is pinned by the
+ * reference's own known-answer tests, ported in tests/test_oracle_golden.py.
+ */
+#ifndef GPUE_ORACLE_H
+#define GPUE_ORACLE_H
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- hashes (be/src/base/hash/hash.h:96-132, exec/join/join_hash_map_helper.h:23-77) ---- */
+uint32_t orc_crc_hash_32(const void* data, int32_t bytes, uint32_t seed);
+uint32_t orc_fnv_hash(const void* data, int32_t bytes, uint32_t seed);
+uint32_t orc_xorshift32(uint32_t x);
+uint32_t orc_join_hash_u32(uint32_t v, uint32_t num_log_buckets);
+uint32_t orc_join_hash_u64(uint64_t v, uint32_t num_log_buckets);
+uint32_t orc_join_hash_slice(const void* p, int32_t n, uint32_t num_buckets);
+uint32_t orc_calc_bucket_size(uint32_t size);
+
+/* ---- deterministic synthetic data (stateless splitmix64 finalizer) ---- */
+uint64_t orc_gen_u64(uint64_t seed, uint64_t tag, uint64_t i);
+void orc_gen_u32_mod(uint64_t seed, uint64_t tag, uint64_t start, uint64_t n,
+                     uint32_t mod, uint32_t add, uint32_t* out);
+void orc_gen_i64(uint64_t seed, uint64_t tag, uint64_t start, uint64_t n, int64_t* out);
+
+/* SSB date dimension: n_days entries from 1992-01-01, d_datekey=y*10000+m*100+d */
+void orc_gen_dates(int32_t n_days, int32_t* datekey, int32_t* dyear);
+
+/* ---- join hash maps ---- */
+/* BucketChained build: exec/join/join_hash_map_method.hpp:37-85. keys is
+ * 1-based (keys[0] = sentinel default row, join_hash_table.cpp:590-596);
+ * first[bucket_size] and next[row_count+1] are zero-initialised by callee. */
+void orc_bucket_chained_build_u32(const uint32_t* keys, uint32_t row_count,
+                                  uint32_t* first, uint32_t* next,
+                                  uint32_t bucket_size, uint32_t log_bucket_size);
+/* lookup_init: join_hash_map_method.hpp:88-120 — chain head per probe row */
+void orc_bucket_chained_lookup_u32(const uint32_t* probe_keys, uint32_t probe_rows,
+                                   const uint32_t* first, uint32_t bucket_size,
+                                   uint32_t log_bucket_size, uint32_t* heads);
+
+/* TLinearChained (fp-packed) build+lookup: join_hash_map_method.hpp:125-368,
+ * join_hash_map_method.h:118-150 (FP_BITS=8, fp|24-bit index packed in first) */
+void orc_linear_chained_build_u32(const uint32_t* keys, uint32_t row_count,
+                                  uint32_t* first, uint32_t* next,
+                                  uint32_t bucket_size, uint32_t log_bucket_size);
+void orc_linear_chained_lookup_u32(const uint32_t* build_keys, const uint32_t* probe_keys,
+                                   uint32_t probe_rows, const uint32_t* first,
+                                   uint32_t bucket_size, uint32_t log_bucket_size,
+                                   uint32_t* heads);
+
+/* RangeDirectMapping: join_hash_map_method.hpp:625-707 */
+void orc_range_direct_build_i32(const int32_t* keys, uint32_t row_count,
+                                int64_t min_value, uint32_t* first, uint32_t* next);
+void orc_range_direct_lookup_i32(const int32_t* probe_keys, uint64_t probe_rows,
+                                 int64_t min_value, int64_t max_value,
+                                 const uint32_t* first, uint32_t* heads);
+
+/* Probe chain-walk, emit ALL (probe_idx, build_idx) pairs:
+ * join_hash_map.hpp:717-795 (_probe_from_ht); the chunk_size-resumable cursor
+ * is an iteration detail — the emitted multiset is the result. Returns number
+ * of pairs. build_keys/next are 1-based; heads from a lookup_init above.
+ * collision_free: skip key compare (is_collision_free_and_unique fast path,
+ * join_hash_map.hpp:752-761). */
+uint64_t orc_probe_emit_u32(const uint32_t* build_keys, const uint32_t* next,
+                            const uint32_t* probe_keys, const uint32_t* heads,
+                            uint32_t probe_rows, int collision_free,
+                            uint32_t* out_probe_idx, uint32_t* out_build_idx);
+
+/* ---- predicate filter (base/simd/filter.h:26-38 + chunk_predicate_evaluator.cpp:31-80) ----
+ * stable stream compaction of int64 values where v < theta; returns count */
+uint64_t orc_filter_i64_lt(const int64_t* in, uint64_t n, int64_t theta, int64_t* out);
+/* OpenMP-parallel variant used as the timed CPU baseline */
+uint64_t orc_filter_i64_lt_mt(const int64_t* in, uint64_t n, int64_t theta, int64_t* out);
+
+/* ---- exchange partition (exchange_sink_operator.cpp:611-660, shuffler.h:71-102,
+ * hash_util.hpp:120-262) ----
+ * hashes: per-row FNV over the 4-byte key with running seed FNV_SEED, then
+ * channel = ReduceOp(hash, num_channels) = (hash * n) >> 32 (HASH_PARTITIONED). */
+void orc_partition_channel_u32(const uint32_t* keys, uint64_t n, uint32_t num_channels,
+                               uint32_t* channel_ids);
+/* counting-sort row layout: start_points[ch+1] sizes then reverse emit
+ * (exchange_sink_operator.cpp:629-660). row_indexes gets source row per slot. */
+void orc_partition_counting_sort(const uint32_t* channel_ids, uint64_t n,
+                                 uint32_t num_channels, uint64_t* start_points,
+                                 uint32_t* row_indexes);
+
+/* ---- fused pipelines (the CPU baseline legs; OpenMP) ---- */
+/* Config 2 (SSB SF10 Q1-like): lineorder(lo_orderdate,lo_extendedprice,
+ * lo_discount) ⋈ date filtered d_year==year, SUM(extendedprice*discount).
+ * Columns generated internally from (seed, n_rows, rank offset). Returns sum;
+ * *match_count gets matched row count. */
+int64_t orc_q1_pipeline(uint64_t seed, uint64_t row_start, uint64_t n_rows,
+                        int32_t year, int threads, uint64_t* match_count);
+/* Config 3 (SSB SF100 Q2.1): 3-way star probe + GROUP BY (d_year,p_brand).
+ * group_sums must hold 7*1000 int64 (index (year-1992)*1000+brand), zeroed. */
+void orc_q21_pipeline(uint64_t seed, uint64_t row_start, uint64_t n_rows,
+                      int32_t category, int32_t region, int threads,
+                      int64_t* group_sums);
+
+/* column generators for the pipelines, exposed so tests can cross-check the
+ * GPU generator bit-for-bit */
+void orc_gen_lineorder_q1(uint64_t seed, uint64_t row_start, uint64_t n,
+                          int32_t* lo_orderdate, int32_t* lo_extendedprice,
+                          int32_t* lo_discount);
+void orc_gen_lineorder_q21(uint64_t seed, uint64_t row_start, uint64_t n,
+                           int32_t* lo_partkey, int32_t* lo_suppkey,
+                           int32_t* lo_orderdate, int32_t* lo_revenue);
+/* dim payload arrays as the GPU builds them:
+ * date:  first[datekey-min] = (d_year-1992)+1 if in range else 0
+ * part:  first[p-1] = brand+1 if p_category==category else 0
+ * supp:  first[s-1] = 1 if s_region==region else 0 */
+void orc_build_date_dim(int32_t n_days, int32_t year_filter /* -1: no filter */,
+                        int32_t* min_key, int32_t* max_key, uint32_t** first_out,
+                        uint32_t* size_out);
+void orc_build_part_dim(uint64_t seed, uint32_t n_parts, int32_t category, uint32_t* first);
+void orc_build_supp_dim(uint64_t seed, uint32_t n_supps, int32_t region, uint32_t* first);
+
+/* p_category / p_brand / s_region generators (shared with GPU) */
+uint32_t orc_part_category(uint64_t seed, uint32_t partkey);
+uint32_t orc_part_brand(uint64_t seed, uint32_t partkey);
+uint32_t orc_supp_region(uint64_t seed, uint32_t suppkey);
+
+void orc_free(void* p);
+
+#ifdef __cplusplus
+}
+#endif
+#endif
