@@ -1,0 +1,160 @@
+/* gpue.h — C-ABI boundary of the MI355X-native execution engine for the
+ * StarRocks BE hot path.
+ *
+ * These entry points are what thin C++ Operator wrappers inside the reference
+ * BE would call from the pipeline driver loop
+ * (reference be/src/exec/runtime/pipeline_driver.cpp:391-500); the mapping of
+ * each entry to the reference interface it replaces is given per declaration.
+ * INTEGRATION.md shows the Operator-side binding a maintainer would add.
+ *
+ * Conventions: extern "C"; plain pointers + sizes; opaque handles; int status
+ * returns (0 == GPUE_OK, matching Status::ok() of reference
+ * be/src/base/status.h); no torch types. The implementation (HIP/C++,
+ * starrocks_amd/csrc/gpue.hip) FAILS at session creation when no AMD GPU is
+ * present — there is no CPU fallback in the product path.
+ */
+#ifndef GPUE_H
+#define GPUE_H
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+#define GPUE_OK 0
+#define GPUE_ERR_HIP 1        /* HIP runtime failure (details: gpue_last_error) */
+#define GPUE_ERR_ARG 2        /* bad argument */
+#define GPUE_ERR_NO_GPU 3     /* no AMD GPU visible */
+
+/* Human-readable detail of the last non-OK status on this thread. */
+const char* gpue_last_error(void);
+
+/* ---- session ----
+ * Engine lifetime per device; analog of FragmentExecutor::prepare/close
+ * (reference be/src/orchestration/fragment_executor.h:40-42). One session ==
+ * one HIP device + one HIP stream (one pipeline driver <-> one stream,
+ * SURVEY.md §2 pipeline row). */
+typedef struct gpue_session gpue_session;
+int gpue_session_create(int device_index, gpue_session** out);
+void gpue_session_destroy(gpue_session* s);
+int gpue_device_count(int* out);
+int gpue_sync(gpue_session* s);
+
+/* ---- device buffers ----
+ * Raw HBM allocations; the engine-side analog of Chunk column containers
+ * (reference be/src/column/fixed_length_column_base.h:283) held resident on
+ * device. Chunked Operator pushes append into these (gpue_dbuf_h2d with
+ * dst_off). */
+typedef struct gpue_dbuf gpue_dbuf;
+int gpue_dbuf_alloc(gpue_session* s, uint64_t bytes, gpue_dbuf** out);
+void gpue_dbuf_free(gpue_dbuf* b);
+int gpue_dbuf_h2d(gpue_dbuf* b, const void* src, uint64_t bytes, uint64_t dst_off);
+int gpue_dbuf_d2h(gpue_dbuf* b, void* dst, uint64_t bytes, uint64_t src_off);
+int gpue_dbuf_memset(gpue_dbuf* b, int value, uint64_t bytes);
+
+/* ---- synthetic chunk source ----
+ * Replaces the scan operator with a seeded on-device generator (SURVEY.md §2
+ * scan row: storage engine OUT OF SCOPE, synthetic source instead). The
+ * generator (splitmix64 finalizer on row index) is bit-identical to
+ * oracle/oracle.c orc_gen_* and tests' numpy restatement. */
+int gpue_gen_u32_mod(gpue_session* s, gpue_dbuf* out, uint64_t seed, uint64_t tag,
+                     uint64_t row_start, uint64_t n, uint32_t mod, uint32_t add);
+int gpue_gen_i64(gpue_session* s, gpue_dbuf* out, uint64_t seed, uint64_t tag,
+                 uint64_t row_start, uint64_t n);
+/* SSB lineorder column sets (schema: reference test/common/sql/ssb/create.sql) */
+int gpue_gen_lineorder_q1(gpue_session* s, uint64_t seed, uint64_t row_start, uint64_t n,
+                          gpue_dbuf* lo_orderdate, gpue_dbuf* lo_extendedprice,
+                          gpue_dbuf* lo_discount);
+int gpue_gen_lineorder_q21(gpue_session* s, uint64_t seed, uint64_t row_start, uint64_t n,
+                           gpue_dbuf* lo_partkey, gpue_dbuf* lo_suppkey,
+                           gpue_dbuf* lo_orderdate, gpue_dbuf* lo_revenue);
+
+/* ---- scan + predicate filter ----
+ * Replaces ChunkPredicateEvaluator::eval_conjuncts + Column::filter_range
+ * (reference be/src/exprs/chunk_predicate_evaluator.cpp:31-80,
+ * be/src/base/simd/filter.h:26-38). Ordered (stable) compaction of int64
+ * values v < theta; output order bit-identical to the reference loop. */
+int gpue_scan_filter_i64_lt(gpue_session* s, gpue_dbuf* in, uint64_t n, int64_t theta,
+                            gpue_dbuf* out, uint64_t* out_count);
+
+/* ---- hash-join build ----
+ * Replaces JoinHashTable::build with the RANGE_DIRECT_MAPPING method the
+ * selector takes for dense int keys (reference
+ * be/src/exec/join/join_hash_table.cpp:263-321,
+ * join_hash_map_method.hpp:625-707). Build rows are 1-based; row 0 is the
+ * chain-end sentinel (join_hash_table.cpp:590-596).
+ *
+ * Payload variant (the fused fast path): first[key-min] holds
+ * a caller-provided uint32 payload (0 = dim row filtered out / absent). */
+typedef struct gpue_join_table gpue_join_table;
+int gpue_join_build_payload_i32(gpue_session* s, gpue_dbuf* keys /*i32*/,
+                                gpue_dbuf* payloads /*u32*/, uint64_t n_rows,
+                                gpue_join_table** out);
+/* Row-index variant: first/next chain structure exactly as the reference
+ * builds it (chain order under duplicate keys is scatter-order, which on GPU
+ * is nondeterministic — the emitted match multiset is identical). */
+int gpue_join_build_range_direct_i32(gpue_session* s, gpue_dbuf* keys /*i32, 1-based row 0 sentinel*/,
+                                     uint64_t row_count, gpue_join_table** out);
+void gpue_join_table_destroy(gpue_join_table* t);
+int gpue_join_table_minmax(gpue_join_table* t, int64_t* min_out, int64_t* max_out);
+/* d2h the first[] array (tests) */
+int gpue_join_table_first_d2h(gpue_join_table* t, uint32_t* dst, uint64_t n_entries);
+
+/* ---- hash-join probe ----
+ * Replaces lookup_init + _probe_from_ht chain-walk emit
+ * (reference be/src/exec/join/join_hash_map_method.hpp:88-120,
+ * join_hash_map.hpp:717-795). Emits all (probe_idx, build_idx) match pairs;
+ * out buffers must be sized for the match count (call with out_* NULL to get
+ * the count first — the CPU's chunk_size-resumable cursor maps to this
+ * two-phase count/emit, SURVEY.md §7 hard part (c)). */
+int gpue_join_probe_emit_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
+                             uint64_t n_rows, gpue_dbuf* out_probe_idx,
+                             gpue_dbuf* out_build_idx, uint64_t* match_count);
+
+/* ---- fused probe + aggregate pipelines ----
+ * Replace the probe -> output gather -> Aggregator::update_batch chain
+ * (reference be/src/exec/join/join_hash_map.hpp:284-330,
+ * be/src/exec/aggregator.cpp:937-959, be/src/exprs/agg/sum.h:45-181) with one
+ * fused kernel per BASELINE config. Sums are int64, order-independent =>
+ * bit-exact vs the oracle. */
+/* Config 2 (SSB SF10 Q1-shaped): SUM(lo_extendedprice*lo_discount) over rows
+ * whose lo_orderdate matches the (pre-filtered) date dim payload table. */
+int gpue_q1_join_sum(gpue_session* s, gpue_join_table* dates, gpue_dbuf* lo_orderdate,
+                     gpue_dbuf* lo_extendedprice, gpue_dbuf* lo_discount, uint64_t n_rows,
+                     int64_t* sum_out, uint64_t* match_count_out);
+/* Async bench variant: zero + launch into a caller-owned 16-byte accumulator
+ * ({int64 sum, uint64 count}); no sync, no transient allocation. */
+int gpue_q1_join_sum_async(gpue_session* s, gpue_join_table* dates, gpue_dbuf* lo_orderdate,
+                           gpue_dbuf* lo_extendedprice, gpue_dbuf* lo_discount,
+                           uint64_t n_rows, gpue_dbuf* acc);
+/* Config 3 (SSB SF100 Q2.1-shaped): 3-way star probe + GROUP BY
+ * (d_year,p_brand) -> dense group id (year_idx*1000+brand), SUM(lo_revenue).
+ * group_sums_out must hold 7000 int64. */
+int gpue_q21_star_agg(gpue_session* s, gpue_join_table* parts, gpue_join_table* supps,
+                      gpue_join_table* dates, gpue_dbuf* lo_partkey, gpue_dbuf* lo_suppkey,
+                      gpue_dbuf* lo_orderdate, gpue_dbuf* lo_revenue, uint64_t n_rows,
+                      int64_t* group_sums_out);
+
+/* Async bench variant of the star aggregate (7000 × int64 device buffer). */
+int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_table* supps,
+                            gpue_join_table* dates, gpue_dbuf* lo_partkey, gpue_dbuf* lo_suppkey,
+                            gpue_dbuf* lo_orderdate, gpue_dbuf* lo_revenue, uint64_t n_rows,
+                            gpue_dbuf* group_sums);
+
+/* ---- exchange partition (shuffle groundwork for configs 4-5) ----
+ * Replaces the ExchangeSinkOperator partition stage
+ * (reference be/src/exec/pipeline/exchange/exchange_sink_operator.cpp:611-660,
+ * shuffler.h:71-86): per-row FNV hash -> ReduceOp channel -> counting-sort
+ * row layout, bit-identical to the reference's (stable: each channel's rows
+ * ascend by source row). start_points has num_channels+1 entries. */
+int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t num_channels,
+                       uint64_t* start_points_out, gpue_dbuf* row_indexes_out);
+
+/* ---- event timing on the session stream (bench roofline evidence) ---- */
+int gpue_timer_start(gpue_session* s);
+int gpue_timer_stop(gpue_session* s, float* ms_out);
+
+#ifdef __cplusplus
+}
+#endif
+#endif

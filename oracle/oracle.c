@@ -364,10 +364,9 @@ void orc_partition_counting_sort(const uint32_t* channel_ids, uint64_t n,
         row_indexes[start_points[channel_ids[i]] - 1] = (uint32_t)i;
         start_points[channel_ids[i]]--;
     }
-    /* start_points now holds start offsets per channel, [num_channels]==begin
-     * of last channel; recompute ends by re-adding counts */
-    for (uint32_t c = num_channels; c > 0; c--) start_points[c] = start_points[c - 1];
-    start_points[0] = 0;
+    /* the reverse emit decremented start_points in place; recompute the
+     * channel boundaries cleanly: start_points[c] = rows in channels < c */
+    memset(start_points, 0, (num_channels + 1) * sizeof(uint64_t));
     for (uint64_t i = 0; i < n; i++) start_points[channel_ids[i] + 1]++;
     for (uint32_t c = 1; c <= num_channels; c++) start_points[c] += start_points[c - 1];
 }
@@ -541,6 +540,63 @@ void orc_q21_pipeline(uint64_t seed, uint64_t row_start, uint64_t n_rows,
         for (int g = 0; g < NG; g++) group_sums[g] += partials[(size_t)t * NG + g];
     free(partials); free(pk); free(sk); free(od); free(rv);
     free(dfirst); free(pfirst); free(sfirst);
+}
+
+/* Compute-only pipeline legs (columns pre-generated by the caller) — these
+ * are what bench.py's cpu_baseline TIMES, so data generation stays outside
+ * the measured region on both CPU and GPU. */
+int64_t orc_q1_kernel(const int32_t* od, const int32_t* ep, const int32_t* dc,
+                      uint64_t n_rows, const uint32_t* dfirst, int64_t mn, int64_t mx,
+                      int threads, uint64_t* match_count) {
+#ifdef _OPENMP
+    if (threads > 0) omp_set_num_threads(threads);
+#endif
+    int64_t sum = 0;
+    uint64_t matches = 0;
+#pragma omp parallel for schedule(static) reduction(+:sum) reduction(+:matches)
+    for (uint64_t i = 0; i < n_rows; i++) {
+        int32_t k = od[i];
+        if (k >= mn && k <= mx && dfirst[k - mn] != 0) {
+            sum += (int64_t)ep[i] * dc[i];
+            matches++;
+        }
+    }
+    if (match_count) *match_count = matches;
+    return sum;
+}
+
+void orc_q21_kernel(const int32_t* pk, const int32_t* sk, const int32_t* od,
+                    const int32_t* rv, uint64_t n_rows, const uint32_t* pfirst,
+                    const uint32_t* sfirst, const uint32_t* dfirst, int64_t dmin,
+                    int threads, int64_t* group_sums) {
+#ifdef _OPENMP
+    if (threads > 0) omp_set_num_threads(threads);
+    int nt = omp_get_max_threads();
+#else
+    int nt = 1;
+#endif
+    const int NG = 7 * 1000;
+    int64_t* partials = (int64_t*)calloc((size_t)nt * NG, sizeof(int64_t));
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        int t = omp_get_thread_num();
+#else
+        int t = 0;
+#endif
+        int64_t* local = partials + (size_t)t * NG;
+#pragma omp for schedule(static)
+        for (uint64_t i = 0; i < n_rows; i++) {
+            uint32_t brand1 = pfirst[pk[i] - 1];
+            if (brand1 == 0) continue;
+            if (sfirst[sk[i] - 1] == 0) continue;
+            uint32_t year1 = dfirst[od[i] - dmin];
+            local[(year1 - 1) * 1000 + (brand1 - 1)] += rv[i];
+        }
+    }
+    for (int t = 0; t < nt; t++)
+        for (int g = 0; g < NG; g++) group_sums[g] += partials[(size_t)t * NG + g];
+    free(partials);
 }
 
 void orc_free(void* p) { free(p); }

@@ -129,6 +129,15 @@ uint32_t orc_part_category(uint64_t seed, uint32_t partkey);
 uint32_t orc_part_brand(uint64_t seed, uint32_t partkey);
 uint32_t orc_supp_region(uint64_t seed, uint32_t suppkey);
 
+/* compute-only legs timed by bench.py's cpu_baseline (columns pre-generated) */
+int64_t orc_q1_kernel(const int32_t* od, const int32_t* ep, const int32_t* dc,
+                      uint64_t n_rows, const uint32_t* dfirst, int64_t mn, int64_t mx,
+                      int threads, uint64_t* match_count);
+void orc_q21_kernel(const int32_t* pk, const int32_t* sk, const int32_t* od,
+                    const int32_t* rv, uint64_t n_rows, const uint32_t* pfirst,
+                    const uint32_t* sfirst, const uint32_t* dfirst, int64_t dmin,
+                    int threads, int64_t* group_sums);
+
 void orc_free(void* p);
 
 #ifdef __cplusplus

@@ -1,0 +1,234 @@
+"""ctypes wrapper over liboracle.so — TEST INFRASTRUCTURE ONLY.
+
+Per DESIGN.md §2 only tests/, __graft_entry__.smoke() and bench.py's
+cpu_baseline leg may import this module. The product path must never route
+through it.
+"""
+
+import ctypes
+import os
+
+import numpy as np
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+
+c_u64 = ctypes.c_uint64
+c_i64 = ctypes.c_int64
+c_u32 = ctypes.c_uint32
+c_i32 = ctypes.c_int32
+c_vp = ctypes.c_void_p
+
+_lib = None
+_ref = None
+
+
+def load():
+    global _lib
+    if _lib is None:
+        p = os.path.join(_DIR, "liboracle.so")
+        if not os.path.exists(p):
+            raise RuntimeError(f"{p} missing — run `make -C oracle`")
+        _lib = ctypes.CDLL(p)
+        _decl(_lib)
+    return _lib
+
+
+def load_ref():
+    """The reference-header shim (oracle/_ref/ref.so), or None if not built."""
+    global _ref
+    if _ref is None:
+        p = os.path.join(_DIR, "_ref", "ref.so")
+        if not os.path.exists(p):
+            return None
+        _ref = ctypes.CDLL(p)
+        for f in ("ref_crc_hash_32", "ref_fnv_hash", "ref_xorshift32"):
+            getattr(_ref, f).restype = c_u32
+    return _ref
+
+
+def _decl(lib):
+    u = c_u32
+    lib.orc_crc_hash_32.restype = u
+    lib.orc_crc_hash_32.argtypes = [c_vp, c_i32, u]
+    lib.orc_fnv_hash.restype = u
+    lib.orc_fnv_hash.argtypes = [c_vp, c_i32, u]
+    lib.orc_xorshift32.restype = u
+    lib.orc_xorshift32.argtypes = [u]
+    lib.orc_join_hash_u32.restype = u
+    lib.orc_join_hash_u32.argtypes = [u, u]
+    lib.orc_join_hash_u64.restype = u
+    lib.orc_join_hash_u64.argtypes = [c_u64, u]
+    lib.orc_join_hash_slice.restype = u
+    lib.orc_join_hash_slice.argtypes = [c_vp, c_i32, u]
+    lib.orc_calc_bucket_size.restype = u
+    lib.orc_calc_bucket_size.argtypes = [u]
+    lib.orc_gen_u64.restype = c_u64
+    lib.orc_gen_u64.argtypes = [c_u64, c_u64, c_u64]
+    lib.orc_gen_dates.argtypes = [c_i32, c_vp, c_vp]
+    lib.orc_bucket_chained_build_u32.argtypes = [c_vp, u, c_vp, c_vp, u, u]
+    lib.orc_bucket_chained_lookup_u32.argtypes = [c_vp, u, c_vp, u, u, c_vp]
+    lib.orc_linear_chained_build_u32.argtypes = [c_vp, u, c_vp, c_vp, u, u]
+    lib.orc_linear_chained_lookup_u32.argtypes = [c_vp, c_vp, u, c_vp, u, u, c_vp]
+    lib.orc_range_direct_build_i32.argtypes = [c_vp, u, c_i64, c_vp, c_vp]
+    lib.orc_range_direct_lookup_i32.argtypes = [c_vp, c_u64, c_i64, c_i64, c_vp, c_vp]
+    lib.orc_probe_emit_u32.restype = c_u64
+    lib.orc_probe_emit_u32.argtypes = [c_vp, c_vp, c_vp, c_vp, u, c_i32, c_vp, c_vp]
+    lib.orc_filter_i64_lt.restype = c_u64
+    lib.orc_filter_i64_lt.argtypes = [c_vp, c_u64, c_i64, c_vp]
+    lib.orc_filter_i64_lt_mt.restype = c_u64
+    lib.orc_filter_i64_lt_mt.argtypes = [c_vp, c_u64, c_i64, c_vp]
+    lib.orc_partition_channel_u32.argtypes = [c_vp, c_u64, u, c_vp]
+    lib.orc_partition_counting_sort.argtypes = [c_vp, c_u64, u, c_vp, c_vp]
+    lib.orc_q1_pipeline.restype = c_i64
+    lib.orc_q1_pipeline.argtypes = [c_u64, c_u64, c_u64, c_i32, c_i32, ctypes.POINTER(c_u64)]
+    lib.orc_q21_pipeline.argtypes = [c_u64, c_u64, c_u64, c_i32, c_i32, c_i32, c_vp]
+    lib.orc_gen_lineorder_q1.argtypes = [c_u64, c_u64, c_u64, c_vp, c_vp, c_vp]
+    lib.orc_gen_lineorder_q21.argtypes = [c_u64, c_u64, c_u64, c_vp, c_vp, c_vp, c_vp]
+    lib.orc_q1_kernel.restype = c_i64
+    lib.orc_q1_kernel.argtypes = [c_vp, c_vp, c_vp, c_u64, c_vp, c_i64, c_i64, c_i32,
+                                  ctypes.POINTER(c_u64)]
+    lib.orc_q21_kernel.argtypes = [c_vp, c_vp, c_vp, c_vp, c_u64, c_vp, c_vp, c_vp,
+                                   c_i64, c_i32, c_vp]
+
+
+def _p(a: np.ndarray):
+    return a.ctypes.data_as(c_vp)
+
+
+def crc_hash_32(data: bytes, seed: int) -> int:
+    return load().orc_crc_hash_32(data, len(data), seed)
+
+
+def fnv_hash(data: bytes, seed: int) -> int:
+    return load().orc_fnv_hash(data, len(data), seed)
+
+
+def join_hash_slice(data: bytes, num_buckets: int) -> int:
+    return load().orc_join_hash_slice(data, len(data), num_buckets)
+
+
+def q1_pipeline(seed, row_start, n_rows, year, threads=0):
+    cnt = c_u64()
+    s = load().orc_q1_pipeline(seed, row_start, n_rows, year, threads, ctypes.byref(cnt))
+    return s, cnt.value
+
+
+def q21_pipeline(seed, row_start, n_rows, category, region, threads=0) -> np.ndarray:
+    out = np.zeros(7000, dtype=np.int64)
+    load().orc_q21_pipeline(seed, row_start, n_rows, category, region, threads, _p(out))
+    return out
+
+
+def filter_i64_lt(inp: np.ndarray, theta: int, mt=False) -> np.ndarray:
+    out = np.empty_like(inp)
+    fn = load().orc_filter_i64_lt_mt if mt else load().orc_filter_i64_lt
+    k = fn(_p(inp), len(inp), theta, _p(out))
+    return out[:k]
+
+
+def gen_lineorder_q1(seed, row_start, n):
+    od = np.empty(n, np.int32)
+    ep = np.empty(n, np.int32)
+    dc = np.empty(n, np.int32)
+    load().orc_gen_lineorder_q1(seed, row_start, n, _p(od), _p(ep), _p(dc))
+    return od, ep, dc
+
+
+def gen_lineorder_q21(seed, row_start, n):
+    pk = np.empty(n, np.int32)
+    sk = np.empty(n, np.int32)
+    od = np.empty(n, np.int32)
+    rv = np.empty(n, np.int32)
+    load().orc_gen_lineorder_q21(seed, row_start, n, _p(pk), _p(sk), _p(od), _p(rv))
+    return pk, sk, od, rv
+
+
+def q1_kernel(od, ep, dc, dfirst, mn, mx, threads=0):
+    cnt = c_u64()
+    s = load().orc_q1_kernel(_p(od), _p(ep), _p(dc), len(od), _p(dfirst), mn, mx,
+                             threads, ctypes.byref(cnt))
+    return s, cnt.value
+
+
+def q21_kernel(pk, sk, od, rv, pfirst, sfirst, dfirst, dmin, threads=0):
+    out = np.zeros(7000, np.int64)
+    load().orc_q21_kernel(_p(pk), _p(sk), _p(od), _p(rv), len(pk), _p(pfirst),
+                          _p(sfirst), _p(dfirst), dmin, threads, _p(out))
+    return out
+
+
+def range_direct_build(keys_1based: np.ndarray, min_value: int, max_value: int):
+    """keys_1based[0] is the sentinel row. Returns (first, next)."""
+    row_count = len(keys_1based) - 1
+    interval = max_value - min_value + 1
+    first = np.zeros(interval, np.uint32)
+    nxt = np.zeros(row_count + 1, np.uint32)
+    load().orc_range_direct_build_i32(_p(keys_1based), row_count, min_value, _p(first), _p(nxt))
+    return first, nxt
+
+
+def range_direct_lookup(probe_keys: np.ndarray, min_value, max_value, first: np.ndarray):
+    heads = np.zeros(len(probe_keys), np.uint32)
+    load().orc_range_direct_lookup_i32(_p(probe_keys), len(probe_keys), min_value, max_value,
+                                       _p(first), _p(heads))
+    return heads
+
+
+def probe_emit(build_keys: np.ndarray, nxt: np.ndarray, probe_keys: np.ndarray,
+               heads: np.ndarray, collision_free=False):
+    cap = max(len(probe_keys) * 8, 1024)
+    op = np.empty(cap, np.uint32)
+    ob = np.empty(cap, np.uint32)
+    m = load().orc_probe_emit_u32(_p(build_keys), _p(nxt), _p(probe_keys), _p(heads),
+                                  len(probe_keys), 1 if collision_free else 0, _p(op), _p(ob))
+    return op[:m].copy(), ob[:m].copy()
+
+
+def bucket_chained_build(keys_1based: np.ndarray):
+    row_count = len(keys_1based) - 1
+    bucket_size = load().orc_calc_bucket_size(row_count + 1)
+    log = int(bucket_size - 1).bit_length()
+    first = np.zeros(bucket_size, np.uint32)
+    nxt = np.zeros(row_count + 1, np.uint32)
+    load().orc_bucket_chained_build_u32(_p(keys_1based), row_count, _p(first), _p(nxt),
+                                        bucket_size, log)
+    return first, nxt, bucket_size, log
+
+
+def bucket_chained_lookup(probe_keys: np.ndarray, first: np.ndarray, bucket_size, log):
+    heads = np.zeros(len(probe_keys), np.uint32)
+    load().orc_bucket_chained_lookup_u32(_p(probe_keys), len(probe_keys), _p(first),
+                                         bucket_size, log, _p(heads))
+    return heads
+
+
+def linear_chained_build(keys_1based: np.ndarray):
+    row_count = len(keys_1based) - 1
+    bucket_size = load().orc_calc_bucket_size(row_count + 1)
+    log = int(bucket_size - 1).bit_length()
+    first = np.zeros(bucket_size, np.uint32)
+    nxt = np.zeros(row_count + 1, np.uint32)
+    load().orc_linear_chained_build_u32(_p(keys_1based), row_count, _p(first), _p(nxt),
+                                        bucket_size, log)
+    return first, nxt, bucket_size, log
+
+
+def linear_chained_lookup(build_keys: np.ndarray, probe_keys: np.ndarray, first, bucket_size, log):
+    heads = np.zeros(len(probe_keys), np.uint32)
+    load().orc_linear_chained_lookup_u32(_p(build_keys), _p(probe_keys), len(probe_keys),
+                                         _p(first), bucket_size, log, _p(heads))
+    return heads
+
+
+def partition_channels(keys: np.ndarray, num_channels: int) -> np.ndarray:
+    out = np.zeros(len(keys), np.uint32)
+    load().orc_partition_channel_u32(_p(keys), len(keys), num_channels, _p(out))
+    return out
+
+
+def partition_counting_sort(channel_ids: np.ndarray, num_channels: int):
+    sp = np.zeros(num_channels + 1, np.uint64)
+    ri = np.zeros(len(channel_ids), np.uint32)
+    load().orc_partition_counting_sort(_p(channel_ids), len(channel_ids), num_channels,
+                                       _p(sp), _p(ri))
+    return sp, ri

@@ -1,0 +1,253 @@
+"""ctypes host bindings over the C-ABI (include/gpue.h).
+
+This is the host side of the drop-in boundary: the same calls a thin C++
+Operator wrapper inside the reference BE would make from the pipeline driver
+loop (INTEGRATION.md shows that wrapper). There is NO CPU fallback: a missing
+libgpue.so or absent GPU raises GpueError.
+"""
+
+import ctypes
+import os
+
+import numpy as np
+
+_PKG_DIR = os.path.dirname(os.path.abspath(__file__))
+
+
+def lib_path() -> str:
+    p = os.environ.get("GPUE_LIB")
+    if p:
+        return p
+    return os.path.join(_PKG_DIR, "libgpue.so")
+
+
+_lib = None
+
+
+def _load():
+    global _lib
+    if _lib is not None:
+        return _lib
+    p = lib_path()
+    if not os.path.exists(p):
+        raise GpueError(
+            f"HIP engine library not found at {p}; build it with "
+            "`python -c 'import __graft_entry__; __graft_entry__.build()'` — "
+            "the product path has no CPU fallback."
+        )
+    _lib = ctypes.CDLL(p)
+    _declare(_lib)
+    return _lib
+
+
+class GpueError(RuntimeError):
+    pass
+
+
+c_u64 = ctypes.c_uint64
+c_i64 = ctypes.c_int64
+c_u32 = ctypes.c_uint32
+c_i32 = ctypes.c_int32
+c_vp = ctypes.c_void_p
+
+
+def _declare(lib):
+    lib.gpue_last_error.restype = ctypes.c_char_p
+    sigs = {
+        "gpue_session_create": (c_i32, [c_i32, ctypes.POINTER(c_vp)]),
+        "gpue_session_destroy": (None, [c_vp]),
+        "gpue_device_count": (c_i32, [ctypes.POINTER(c_i32)]),
+        "gpue_sync": (c_i32, [c_vp]),
+        "gpue_dbuf_alloc": (c_i32, [c_vp, c_u64, ctypes.POINTER(c_vp)]),
+        "gpue_dbuf_free": (None, [c_vp]),
+        "gpue_dbuf_h2d": (c_i32, [c_vp, c_vp, c_u64, c_u64]),
+        "gpue_dbuf_d2h": (c_i32, [c_vp, c_vp, c_u64, c_u64]),
+        "gpue_dbuf_memset": (c_i32, [c_vp, c_i32, c_u64]),
+        "gpue_gen_u32_mod": (c_i32, [c_vp, c_vp, c_u64, c_u64, c_u64, c_u64, c_u32, c_u32]),
+        "gpue_gen_i64": (c_i32, [c_vp, c_vp, c_u64, c_u64, c_u64, c_u64]),
+        "gpue_gen_lineorder_q1": (c_i32, [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp, c_vp]),
+        "gpue_gen_lineorder_q21": (c_i32, [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp, c_vp, c_vp]),
+        "gpue_scan_filter_i64_lt": (c_i32, [c_vp, c_vp, c_u64, c_i64, c_vp, ctypes.POINTER(c_u64)]),
+        "gpue_join_build_payload_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
+        "gpue_join_build_range_direct_i32": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
+        "gpue_join_table_destroy": (None, [c_vp]),
+        "gpue_join_table_minmax": (c_i32, [c_vp, ctypes.POINTER(c_i64), ctypes.POINTER(c_i64)]),
+        "gpue_join_table_first_d2h": (c_i32, [c_vp, c_vp, c_u64]),
+        "gpue_join_probe_emit_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp, c_vp, ctypes.POINTER(c_u64)]),
+        "gpue_q1_join_sum": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_u64,
+                                     ctypes.POINTER(c_i64), ctypes.POINTER(c_u64)]),
+        "gpue_q21_star_agg": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
+        "gpue_q1_join_sum_async": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
+        "gpue_q21_star_agg_async": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
+        "gpue_partition_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
+        "gpue_timer_start": (c_i32, [c_vp]),
+        "gpue_timer_stop": (c_i32, [c_vp, ctypes.POINTER(ctypes.c_float)]),
+    }
+    for name, (res, args) in sigs.items():
+        fn = getattr(lib, name)
+        fn.restype = res
+        fn.argtypes = args
+
+
+def _ck(lib, status):
+    if status != 0:
+        raise GpueError(f"gpue status {status}: {lib.gpue_last_error().decode()}")
+
+
+class DBuf:
+    """Device HBM buffer (columnar container analog)."""
+
+    def __init__(self, eng: "Engine", nbytes: int):
+        self._lib = eng._lib
+        self.nbytes = nbytes
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_dbuf_alloc(eng._h, nbytes, ctypes.byref(h)))
+        self._h = h
+
+    def h2d(self, arr: np.ndarray, dst_off: int = 0):
+        arr = np.ascontiguousarray(arr)
+        _ck(self._lib, self._lib.gpue_dbuf_h2d(self._h, arr.ctypes.data_as(c_vp),
+                                               arr.nbytes, dst_off))
+        return self
+
+    def d2h(self, dtype, count: int, src_off: int = 0) -> np.ndarray:
+        out = np.empty(count, dtype=dtype)
+        _ck(self._lib, self._lib.gpue_dbuf_d2h(self._h, out.ctypes.data_as(c_vp),
+                                               out.nbytes, src_off))
+        return out
+
+    def free(self):
+        if self._h:
+            self._lib.gpue_dbuf_free(self._h)
+            self._h = None
+
+
+class JoinTable:
+    def __init__(self, eng: "Engine", handle):
+        self._lib = eng._lib
+        self._h = handle
+
+    @property
+    def minmax(self):
+        mn, mx = c_i64(), c_i64()
+        _ck(self._lib, self._lib.gpue_join_table_minmax(self._h, ctypes.byref(mn), ctypes.byref(mx)))
+        return mn.value, mx.value
+
+    def first_d2h(self, n_entries: int) -> np.ndarray:
+        out = np.empty(n_entries, dtype=np.uint32)
+        _ck(self._lib, self._lib.gpue_join_table_first_d2h(self._h, out.ctypes.data_as(c_vp), n_entries))
+        return out
+
+    def destroy(self):
+        if self._h:
+            self._lib.gpue_join_table_destroy(self._h)
+            self._h = None
+
+
+class Engine:
+    """One session == one device + one HIP stream (one pipeline driver)."""
+
+    def __init__(self, device: int = 0):
+        self._lib = _load()
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_session_create(device, ctypes.byref(h)))
+        self._h = h
+        self.device = device
+
+    @staticmethod
+    def device_count() -> int:
+        lib = _load()
+        n = c_i32()
+        lib.gpue_device_count(ctypes.byref(n))
+        return n.value
+
+    def close(self):
+        if self._h:
+            self._lib.gpue_session_destroy(self._h)
+            self._h = None
+
+    def sync(self):
+        _ck(self._lib, self._lib.gpue_sync(self._h))
+
+    def alloc(self, nbytes: int) -> DBuf:
+        return DBuf(self, nbytes)
+
+    # ---- synthetic chunk source (scan replacement) ----
+    def gen_u32_mod(self, out: DBuf, seed, tag, row_start, n, mod, add=0):
+        _ck(self._lib, self._lib.gpue_gen_u32_mod(self._h, out._h, seed, tag, row_start, n, mod, add))
+
+    def gen_i64(self, out: DBuf, seed, tag, row_start, n):
+        _ck(self._lib, self._lib.gpue_gen_i64(self._h, out._h, seed, tag, row_start, n))
+
+    def gen_lineorder_q1(self, seed, row_start, n, od: DBuf, ep: DBuf, dc: DBuf):
+        _ck(self._lib, self._lib.gpue_gen_lineorder_q1(self._h, seed, row_start, n, od._h, ep._h, dc._h))
+
+    def gen_lineorder_q21(self, seed, row_start, n, pk, sk, od, rv):
+        _ck(self._lib, self._lib.gpue_gen_lineorder_q21(self._h, seed, row_start, n,
+                                                        pk._h, sk._h, od._h, rv._h))
+
+    # ---- operators ----
+    def scan_filter_i64_lt(self, inp: DBuf, n, theta, out: DBuf) -> int:
+        cnt = c_u64()
+        _ck(self._lib, self._lib.gpue_scan_filter_i64_lt(self._h, inp._h, n, theta, out._h,
+                                                         ctypes.byref(cnt)))
+        return cnt.value
+
+    def join_build_payload(self, keys: DBuf, payloads: DBuf, n_rows) -> JoinTable:
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_join_build_payload_i32(self._h, keys._h, payloads._h,
+                                                             n_rows, ctypes.byref(h)))
+        return JoinTable(self, h)
+
+    def join_build_range_direct(self, keys: DBuf, row_count) -> JoinTable:
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_join_build_range_direct_i32(self._h, keys._h, row_count,
+                                                                  ctypes.byref(h)))
+        return JoinTable(self, h)
+
+    def join_probe_emit(self, table: JoinTable, probe_keys: DBuf, n_rows,
+                        out_probe: DBuf = None, out_build: DBuf = None) -> int:
+        cnt = c_u64()
+        op = out_probe._h if out_probe else None
+        ob = out_build._h if out_build else None
+        _ck(self._lib, self._lib.gpue_join_probe_emit_i32(self._h, table._h, probe_keys._h,
+                                                          n_rows, op, ob, ctypes.byref(cnt)))
+        return cnt.value
+
+    def q1_join_sum(self, dates: JoinTable, od: DBuf, ep: DBuf, dc: DBuf, n):
+        s, c = c_i64(), c_u64()
+        _ck(self._lib, self._lib.gpue_q1_join_sum(self._h, dates._h, od._h, ep._h, dc._h, n,
+                                                  ctypes.byref(s), ctypes.byref(c)))
+        return s.value, c.value
+
+    def q1_join_sum_async(self, dates: JoinTable, od: DBuf, ep: DBuf, dc: DBuf, n, acc: DBuf):
+        _ck(self._lib, self._lib.gpue_q1_join_sum_async(self._h, dates._h, od._h, ep._h,
+                                                        dc._h, n, acc._h))
+
+    def q21_star_agg_async(self, parts, supps, dates, pk, sk, od, rv, n, group_sums: DBuf):
+        _ck(self._lib, self._lib.gpue_q21_star_agg_async(self._h, parts._h, supps._h, dates._h,
+                                                         pk._h, sk._h, od._h, rv._h, n,
+                                                         group_sums._h))
+
+    def q21_star_agg(self, parts: JoinTable, supps: JoinTable, dates: JoinTable,
+                     pk: DBuf, sk: DBuf, od: DBuf, rv: DBuf, n) -> np.ndarray:
+        out = np.zeros(7000, dtype=np.int64)
+        _ck(self._lib, self._lib.gpue_q21_star_agg(self._h, parts._h, supps._h, dates._h,
+                                                   pk._h, sk._h, od._h, rv._h, n,
+                                                   out.ctypes.data_as(c_vp)))
+        return out
+
+    def partition(self, keys: DBuf, n, num_channels, row_indexes: DBuf) -> np.ndarray:
+        sp = np.zeros(num_channels + 1, dtype=np.uint64)
+        _ck(self._lib, self._lib.gpue_partition_i32(self._h, keys._h, n, num_channels,
+                                                    sp.ctypes.data_as(c_vp), row_indexes._h))
+        return sp
+
+    # ---- stream event timing (bench) ----
+    def timer_start(self):
+        _ck(self._lib, self._lib.gpue_timer_start(self._h))
+
+    def timer_stop(self) -> float:
+        ms = ctypes.c_float()
+        _ck(self._lib, self._lib.gpue_timer_stop(self._h, ctypes.byref(ms)))
+        return ms.value

@@ -1,0 +1,125 @@
+"""Deterministic synthetic SSB/TPC-H columns — numpy restatement.
+
+This is the SAME counter-based generator (splitmix64 finalizer) implemented in
+oracle/oracle.c (orc_gen_*) and starrocks_amd/csrc/gpue.hip (k_gen_*): all
+three produce bit-identical columns for a given (seed, tag, row index), so CPU
+oracle and GPU engine operate on identical data with no transfer.
+
+Schema follows the reference's SSB DDL (reference
+test/common/sql/ssb/create.sql) at the granularity BASELINE.json's configs
+need; SURVEY.md §8d fixes the column value ranges.
+"""
+
+import numpy as np
+
+_GOLD = np.uint64(0x9E3779B97F4A7C15)
+_M1 = np.uint64(0xBF58476D1CE4E5B9)
+_M2 = np.uint64(0x94D049BB133111EB)
+
+# tags shared with oracle/oracle.c and csrc/gpue.hip
+TAG_ORDERDATE = 1
+TAG_EXTPRICE = 2
+TAG_DISCOUNT = 3
+TAG_PARTKEY = 4
+TAG_SUPPKEY = 5
+TAG_REVENUE = 6
+TAG_PCAT = 7
+TAG_PBRD = 8
+TAG_SREG = 9
+
+N_DAYS = 2556
+N_PARTS_SF100 = 1_400_000
+N_SUPPS_SF100 = 200_000
+SF10_LINEORDER_ROWS = 59_986_052
+SF100_LINEORDER_ROWS = 600_000_000
+
+
+def sm64(x: np.ndarray) -> np.ndarray:
+    with np.errstate(over="ignore"):
+        x = x.astype(np.uint64) + _GOLD
+        x = (x ^ (x >> np.uint64(30))) * _M1
+        x = (x ^ (x >> np.uint64(27))) * _M2
+        return x ^ (x >> np.uint64(31))
+
+
+def gen_u64(seed: int, tag: int, idx: np.ndarray) -> np.ndarray:
+    with np.errstate(over="ignore"):
+        base = np.uint64(seed) + np.uint64(tag) * _GOLD
+        return sm64(base + idx.astype(np.uint64))
+
+
+def gen_dates(n_days: int = N_DAYS):
+    """datekey/d_year arrays from 1992-01-01 (identical to orc_gen_dates)."""
+    mdays = [31, 28, 31, 30, 31, 30, 31, 31, 30, 31, 30, 31]
+    datekey = np.empty(n_days, dtype=np.int32)
+    dyear = np.empty(n_days, dtype=np.int32)
+    y, m, d = 1992, 1, 1
+    for i in range(n_days):
+        datekey[i] = y * 10000 + m * 100 + d
+        dyear[i] = y
+        leap = (y % 4 == 0 and y % 100 != 0) or y % 400 == 0
+        md = mdays[m - 1] + (1 if (m == 2 and leap) else 0)
+        d += 1
+        if d > md:
+            d = 1
+            m += 1
+            if m > 12:
+                m, y = 1, y + 1
+    return datekey, dyear
+
+
+def gen_lineorder_q1(seed: int, row_start: int, n: int):
+    idx = np.arange(row_start, row_start + n, dtype=np.uint64)
+    datekey, _ = gen_dates()
+    od = datekey[(gen_u64(seed, TAG_ORDERDATE, idx) % np.uint64(N_DAYS)).astype(np.int64)]
+    ep = (gen_u64(seed, TAG_EXTPRICE, idx) % np.uint64(100000)).astype(np.int32) + 1
+    dc = (gen_u64(seed, TAG_DISCOUNT, idx) % np.uint64(11)).astype(np.int32)
+    return od.astype(np.int32), ep, dc
+
+
+def gen_lineorder_q21(seed: int, row_start: int, n: int):
+    idx = np.arange(row_start, row_start + n, dtype=np.uint64)
+    datekey, _ = gen_dates()
+    pk = (gen_u64(seed, TAG_PARTKEY, idx) % np.uint64(N_PARTS_SF100)).astype(np.int32) + 1
+    sk = (gen_u64(seed, TAG_SUPPKEY, idx) % np.uint64(N_SUPPS_SF100)).astype(np.int32) + 1
+    od = datekey[(gen_u64(seed, TAG_ORDERDATE, idx) % np.uint64(N_DAYS)).astype(np.int64)]
+    rv = (gen_u64(seed, TAG_REVENUE, idx) % np.uint64(10000000)).astype(np.int32)
+    return pk, sk, od.astype(np.int32), rv
+
+
+def part_category(seed: int, partkeys: np.ndarray) -> np.ndarray:
+    return (gen_u64(seed, TAG_PCAT, partkeys) % np.uint64(25)).astype(np.uint32)
+
+
+def part_brand(seed: int, partkeys: np.ndarray) -> np.ndarray:
+    cat = part_category(seed, partkeys)
+    return cat * 40 + (gen_u64(seed, TAG_PBRD, partkeys) % np.uint64(40)).astype(np.uint32)
+
+
+def supp_region(seed: int, suppkeys: np.ndarray) -> np.ndarray:
+    return (gen_u64(seed, TAG_SREG, suppkeys) % np.uint64(5)).astype(np.uint32)
+
+
+def build_date_dim_payload(year_filter: int | None):
+    """Direct-mapped date payload array: value (d_year-1992)+1
+    where the dim row passes the year filter, else 0. Returns (min_key,
+    max_key, first)."""
+    datekey, dyear = gen_dates()
+    mn, mx = int(datekey[0]), int(datekey[-1])
+    first = np.zeros(mx - mn + 1, dtype=np.uint32)
+    keep = np.ones(len(datekey), bool) if year_filter is None else (dyear == year_filter)
+    first[datekey[keep] - mn] = (dyear[keep] - 1992 + 1).astype(np.uint32)
+    return mn, mx, first
+
+
+def build_part_dim_payload(seed: int, n_parts: int, category: int) -> np.ndarray:
+    """first[p-1] = brand+1 if p_category==category else 0."""
+    pkeys = np.arange(1, n_parts + 1, dtype=np.uint64)
+    cat = part_category(seed, pkeys)
+    brand = part_brand(seed, pkeys)
+    return np.where(cat == category, brand + 1, 0).astype(np.uint32)
+
+
+def build_supp_dim_payload(seed: int, n_supps: int, region: int) -> np.ndarray:
+    skeys = np.arange(1, n_supps + 1, dtype=np.uint64)
+    return (supp_region(seed, skeys) == region).astype(np.uint32)

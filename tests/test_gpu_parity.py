@@ -1,0 +1,182 @@
+"""GPU↔oracle parity (the §8c gate). Every test here calls through the C-ABI
+(libgpue.so) and compares against the CPU oracle on identical seeded inputs:
+bit-exact for all integer work; join pair multisets compared sorted
+(match ORDER is the only nondeterminism — SURVEY.md §7 hard part (a)).
+
+Runs on a real MI355X via gpurun; marked gpu.
+"""
+
+import numpy as np
+import pytest
+
+from oracle import pyoracle as orc
+from starrocks_amd import gen
+
+pytestmark = pytest.mark.gpu
+
+SEED = 42
+
+
+def test_native_lib_is_intree(engine):
+    """The loaded engine must be the in-tree HIP library, not a fallback."""
+    from starrocks_amd.engine import lib_path
+    import os
+    assert os.path.exists(lib_path())
+    assert "starrocks_amd" in lib_path()
+
+
+def test_gen_u32_matches_oracle(engine):
+    n = 1_000_000
+    buf = engine.alloc(n * 4)
+    engine.gen_u32_mod(buf, SEED, 77, 123, n, 1000, 5)
+    got = buf.d2h(np.uint32, n)
+    idx = np.arange(123, 123 + n, dtype=np.uint64)
+    expect = (gen.gen_u64(SEED, 77, idx) % np.uint64(1000)).astype(np.uint32) + 5
+    assert np.array_equal(got, expect)
+    buf.free()
+
+
+def test_gen_lineorder_q1_matches_oracle(engine):
+    n = 2_000_000
+    od, ep, dc = engine.alloc(n * 4), engine.alloc(n * 4), engine.alloc(n * 4)
+    engine.gen_lineorder_q1(SEED, 500, n, od, ep, dc)
+    od_c, ep_c, dc_c = orc.gen_lineorder_q1(SEED, 500, n)
+    assert np.array_equal(od.d2h(np.int32, n), od_c)
+    assert np.array_equal(ep.d2h(np.int32, n), ep_c)
+    assert np.array_equal(dc.d2h(np.int32, n), dc_c)
+    for b in (od, ep, dc):
+        b.free()
+
+
+@pytest.mark.parametrize("n,sel", [(1_000_000, 0.1), (10_000_000, 0.5), (999_999, 0.01), (4096, 1.0), (1000, 0.0)])
+def test_filter_parity_bit_exact_ordered(engine, n, sel):
+    inp = engine.alloc(n * 8)
+    engine.gen_i64(inp, SEED, 3, 0, n)
+    data = inp.d2h(np.int64, n)
+    # theta at the requested selectivity over the uniform u64-as-i64 domain
+    theta = int(np.quantile(data.astype(np.float64), sel)) if 0 < sel < 1 else (
+        np.iinfo(np.int64).max if sel >= 1 else np.iinfo(np.int64).min)
+    out = engine.alloc(n * 8)
+    cnt = engine.scan_filter_i64_lt(inp, n, theta, out)
+    expect = orc.filter_i64_lt(data, theta)
+    assert cnt == len(expect)
+    got = out.d2h(np.int64, cnt) if cnt else np.empty(0, np.int64)
+    assert np.array_equal(got, expect)  # bit-exact INCLUDING order
+    inp.free()
+    out.free()
+
+
+def test_join_build_probe_emit_parity(engine):
+    """Range-direct build + probe emit vs oracle: multiset of (probe,build)."""
+    rng = np.random.default_rng(5)
+    build_keys = np.concatenate([[0], rng.integers(100, 5000, 20_000)]).astype(np.int32)
+    probe_keys = rng.integers(0, 6000, 300_000).astype(np.int32)
+
+    kb = engine.alloc(build_keys.nbytes)
+    kb.h2d(build_keys)
+    t = engine.join_build_range_direct(kb, len(build_keys) - 1)
+    mn, mx = t.minmax
+    assert mn == int(build_keys[1:].min()) and mx == int(build_keys[1:].max())
+
+    pb = engine.alloc(probe_keys.nbytes)
+    pb.h2d(probe_keys)
+    cnt = engine.join_probe_emit(t, pb, len(probe_keys))
+    op_buf = engine.alloc(max(cnt, 1) * 4)
+    ob_buf = engine.alloc(max(cnt, 1) * 4)
+    cnt2 = engine.join_probe_emit(t, pb, len(probe_keys), op_buf, ob_buf)
+    assert cnt2 == cnt
+    gop = op_buf.d2h(np.uint32, cnt)
+    gob = ob_buf.d2h(np.uint32, cnt)
+
+    first, nxt = orc.range_direct_build(build_keys, mn, mx)
+    heads = orc.range_direct_lookup(probe_keys, mn, mx, first)
+    eop, eob = orc.probe_emit(build_keys.view(np.uint32), nxt, probe_keys.view(np.uint32), heads)
+    assert len(eop) == cnt
+    got = np.sort(gop.astype(np.uint64) << np.uint64(32) | gob.astype(np.uint64))
+    exp = np.sort(eop.astype(np.uint64) << np.uint64(32) | eob.astype(np.uint64))
+    assert np.array_equal(got, exp)
+    for b in (kb, pb, op_buf, ob_buf):
+        b.free()
+    t.destroy()
+
+
+def _build_date_table(engine, year_filter):
+    datekey, dyear = gen.gen_dates()
+    mn, mx, first = gen.build_date_dim_payload(year_filter)
+    keys = engine.alloc(datekey.nbytes)
+    keys.h2d(datekey.astype(np.int32))
+    payload = np.where((dyear == year_filter) if year_filter else np.ones(len(dyear), bool),
+                       dyear - 1992 + 1, 0).astype(np.uint32)
+    pay = engine.alloc(payload.nbytes)
+    pay.h2d(payload)
+    t = engine.join_build_payload(keys, pay, len(datekey))
+    # the device payload table must equal the host-computed one
+    assert np.array_equal(t.first_d2h(mx - mn + 1), first)
+    keys.free()
+    pay.free()
+    return t
+
+
+@pytest.mark.parametrize("n_rows", [1_000_000, 10_000_000])
+def test_q1_join_sum_parity(engine, n_rows):
+    year = 1993
+    od, ep, dc = (engine.alloc(n_rows * 4) for _ in range(3))
+    engine.gen_lineorder_q1(SEED, 0, n_rows, od, ep, dc)
+    dates = _build_date_table(engine, year)
+    s, cnt = engine.q1_join_sum(dates, od, ep, dc, n_rows)
+    es, ecnt = orc.q1_pipeline(SEED, 0, n_rows, year)
+    assert (s, cnt) == (es, ecnt)  # bit-exact int64 sum
+    for b in (od, ep, dc):
+        b.free()
+    dates.destroy()
+
+
+def test_q21_star_agg_parity(engine):
+    n_rows, cat, reg = 5_000_000, 12, 2
+    pk, sk, od, rv = (engine.alloc(n_rows * 4) for _ in range(4))
+    engine.gen_lineorder_q21(SEED, 0, n_rows, pk, sk, od, rv)
+
+    pfirst = gen.build_part_dim_payload(SEED, gen.N_PARTS_SF100, cat)
+    sfirst = gen.build_supp_dim_payload(SEED, gen.N_SUPPS_SF100, reg)
+
+    pkeys = engine.alloc(gen.N_PARTS_SF100 * 4)
+    pkeys.h2d(np.arange(1, gen.N_PARTS_SF100 + 1, dtype=np.int32))
+    ppay = engine.alloc(pfirst.nbytes)
+    ppay.h2d(pfirst)
+    parts = engine.join_build_payload(pkeys, ppay, gen.N_PARTS_SF100)
+
+    skeys = engine.alloc(gen.N_SUPPS_SF100 * 4)
+    skeys.h2d(np.arange(1, gen.N_SUPPS_SF100 + 1, dtype=np.int32))
+    spay = engine.alloc(sfirst.nbytes)
+    spay.h2d(sfirst)
+    supps = engine.join_build_payload(skeys, spay, gen.N_SUPPS_SF100)
+
+    dates = _build_date_table(engine, None)
+
+    got = engine.q21_star_agg(parts, supps, dates, pk, sk, od, rv, n_rows)
+    expect = orc.q21_pipeline(SEED, 0, n_rows, cat, reg)
+    assert np.array_equal(got, expect)  # bit-exact per-group int64 sums
+    for b in (pk, sk, od, rv, pkeys, ppay, skeys, spay):
+        b.free()
+    for t in (parts, supps, dates):
+        t.destroy()
+
+
+def test_partition_parity(engine):
+    n, nch = 3_000_000, 8
+    keys = engine.alloc(n * 4)
+    engine.gen_u32_mod(keys, SEED, 9, 0, n, 0, 0)
+    host_keys = keys.d2h(np.uint32, n)
+    ri = engine.alloc(n * 4)
+    sp = engine.partition(keys, n, nch, ri)
+    ch = orc.partition_channels(host_keys, nch)
+    esp, _ = orc.partition_counting_sort(ch, nch)
+    assert np.array_equal(sp, esp)  # channel sizes bit-identical
+    got_ri = ri.d2h(np.uint32, n)
+    for c in range(nch):
+        rows = got_ri[int(sp[c]):int(sp[c + 1])]
+        # every row landed on its reference channel; per-channel row SET pinned
+        assert (ch[rows] == c).all()
+        assert len(np.unique(rows)) == len(rows)
+    keys.free()
+    ri.free()
