@@ -636,6 +636,10 @@ int gpue_join_probe_emit_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* pro
 // ---------------------------------------------------------------------------
 // fused probe + aggregate pipelines
 // ---------------------------------------------------------------------------
+// 16 B/lane vectorized (Guideline 13: 4×int32 per column per iteration) with
+// UNCONDITIONAL column loads — a load guarded by the probe branch serializes
+// behind per-element vmcnt waits (guide §5 item 4(c)); the probe gather is
+// made branchless by clamping the index into the (L2-resident) dim array.
 __global__ void k_q1_join_sum(const int32_t* __restrict__ od, const int32_t* __restrict__ ep,
                               const int32_t* __restrict__ dc, uint64_t n,
                               int64_t mn, int64_t mx, const uint32_t* __restrict__ dfirst,
@@ -643,10 +647,32 @@ __global__ void k_q1_join_sum(const int32_t* __restrict__ od, const int32_t* __r
                               unsigned long long* __restrict__ cnt_out) {
     int64_t sum = 0;
     uint64_t cnt = 0;
+    const uint64_t n4 = n / 4;
+    const uint32_t interval = (uint32_t)(mx - mn + 1);
+    const int4* __restrict__ od4 = (const int4*)od;
+    const int4* __restrict__ ep4 = (const int4*)ep;
+    const int4* __restrict__ dc4 = (const int4*)dc;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+        int4 k4 = od4[i];
+        int4 e4 = ep4[i];
+        int4 d4 = dc4[i];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            int32_t k = (&k4.x)[j];
+            uint32_t idx = (uint32_t)(k - (int32_t)mn);
+            uint32_t cidx = idx < interval ? idx : 0;
+            bool pass = (idx < interval) & (dfirst[cidx] != 0);
+            sum += pass ? (int64_t)(&e4.x)[j] * (&d4.x)[j] : 0;
+            cnt += pass;
+        }
+    }
+    // scalar tail (n % 4 rows), first few threads of the grid
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (uint64_t i = n4 * 4 + tid; i < n; i += stride) {
         int32_t k = od[i];
-        if (k >= mn && k <= mx && dfirst[k - mn] != 0) {
+        uint32_t idx = (uint32_t)(k - (int32_t)mn);
+        if (idx < interval && dfirst[idx] != 0) {
             sum += (int64_t)ep[i] * dc[i];
             cnt++;
         }
@@ -700,17 +726,47 @@ int gpue_q1_join_sum_async(gpue_session* s, gpue_join_table* dates, gpue_dbuf* o
 
 static constexpr int NG_Q21 = 7 * 1000; // (d_year-1992) in [0,7) × p_brand in [0,1000)
 
-__global__ void k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
-                               const int32_t* __restrict__ od, const int32_t* __restrict__ rv,
-                               uint64_t n, const uint32_t* __restrict__ pfirst,
-                               const uint32_t* __restrict__ sfirst,
-                               const uint32_t* __restrict__ dfirst, int64_t dmin,
-                               unsigned long long* __restrict__ group_sums) {
-    __shared__ unsigned long long g[NG_Q21]; // 56 KB LDS, 2 blocks/CU
+// 1024-thread blocks: the 56 KB LDS group array would cap a 256-thread block
+// at 2 blocks/CU = 8 waves/CU — far too few to hide the random pfirst/sfirst
+// gather latency. 1024 threads × 2 blocks/CU = 32 waves/CU (full) at 112 KB
+// LDS/CU (< 160 KB). Columns are read 16 B/lane (int4, Guideline 13) with
+// unconditional loads; only the dependent dim gathers stay branchy (they
+// carry the 1/25 × 1/5 selectivity, so skipping them saves real traffic).
+static constexpr int BLOCK_Q21 = 1024;
+
+__global__ __launch_bounds__(BLOCK_Q21) void
+k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
+               const int32_t* __restrict__ od, const int32_t* __restrict__ rv,
+               uint64_t n, const uint32_t* __restrict__ pfirst,
+               const uint32_t* __restrict__ sfirst,
+               const uint32_t* __restrict__ dfirst, int64_t dmin,
+               unsigned long long* __restrict__ group_sums) {
+    __shared__ unsigned long long g[NG_Q21]; // 56 KB LDS
     for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x) g[j] = 0;
     __syncthreads();
+    const uint64_t n4 = n / 4;
+    const int4* __restrict__ pk4 = (const int4*)pk;
+    const int4* __restrict__ sk4 = (const int4*)sk;
+    const int4* __restrict__ od4 = (const int4*)od;
+    const int4* __restrict__ rv4 = (const int4*)rv;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+        int4 p4 = pk4[i];
+        int4 s4 = sk4[i];
+        int4 o4 = od4[i];
+        int4 r4 = rv4[i];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint32_t brand1 = pfirst[(&p4.x)[j] - 1];
+            if (brand1 == 0) continue;
+            if (sfirst[(&s4.x)[j] - 1] == 0) continue;
+            uint32_t year1 = dfirst[(&o4.x)[j] - dmin];
+            atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)],
+                      (unsigned long long)(int64_t)(&r4.x)[j]);
+        }
+    }
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (uint64_t i = n4 * 4 + tid; i < n; i += stride) {
         uint32_t brand1 = pfirst[pk[i] - 1];
         if (brand1 == 0) continue;
         if (sfirst[sk[i] - 1] == 0) continue;
@@ -733,7 +789,7 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
     ARG_CHECK(group_sums->bytes >= NG_Q21 * sizeof(int64_t));
     ARG_CHECK(parts->min_key == 1 && supps->min_key == 1);
     HIP_CHECK(hipMemsetAsync(group_sums->ptr, 0, NG_Q21 * sizeof(int64_t), s->stream));
-    hipLaunchKernelGGL(k_q21_star_agg, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_q21_star_agg, dim3(512), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)od->ptr, (const int32_t*)rv->ptr, n,
                        parts->first, supps->first, dates->first, dates->min_key,
@@ -751,7 +807,7 @@ int gpue_q21_star_agg(gpue_session* s, gpue_join_table* parts, gpue_join_table* 
     unsigned long long* d_g = nullptr;
     HIP_CHECK(hipMalloc(&d_g, NG_Q21 * sizeof(unsigned long long)));
     HIP_CHECK(hipMemsetAsync(d_g, 0, NG_Q21 * sizeof(unsigned long long), s->stream));
-    hipLaunchKernelGGL(k_q21_star_agg, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_q21_star_agg, dim3(512), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)od->ptr, (const int32_t*)rv->ptr, n,
                        parts->first, supps->first, dates->first, dates->min_key, d_g);
