@@ -71,12 +71,18 @@ struct gpue_dbuf {
 };
 
 struct gpue_join_table {
-    gpue_session* s;
-    int64_t min_key, max_key;
-    uint32_t* first;    // device, (max-min+1) entries: payload or head row idx
-    uint32_t* next;     // device, (row_count+1) entries, or nullptr (payload variant)
-    uint64_t bucket_size;
-    uint64_t row_count;
+    gpue_session* s = nullptr;
+    int64_t min_key = 0, max_key = 0;
+    uint32_t* first = nullptr;   // device, (max-min+1) entries: payload or head row idx
+    uint32_t* next = nullptr;    // device, (row_count+1) entries, or nullptr (payload variant)
+    // Derived probe structures (payload variant only):
+    uint32_t* bitset = nullptr;  // 1 bit per key in [min,max] — the reference's
+                                 // RANGE_DIRECT_MAPPING_SET (join_hash_table.cpp:297-303)
+                                 // for semi-join probes; small enough to stage in LDS
+    uint16_t* first16 = nullptr; // 16-bit payload copy when all payloads < 65536:
+                                 // halves the random-gather footprint (L2 per XCD is 4 MiB)
+    uint64_t bucket_size = 0;
+    uint64_t row_count = 0;
 };
 
 int gpue_device_count(int* out) {
@@ -420,6 +426,34 @@ __global__ void k_build_payload(const int32_t* __restrict__ keys,
         first[keys[i] - min_key] = payloads[i];
 }
 
+// Derive the semi-join bitset (1 bit/key) and the 16-bit payload copy from
+// the u32 payload array — built once at build time, probed every launch.
+__global__ void k_derive_bitset(const uint32_t* __restrict__ first, uint64_t interval,
+                                uint32_t* __restrict__ bitset) {
+    uint64_t nwords = (interval + 31) / 32;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t w = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; w < nwords; w += stride) {
+        uint32_t bits = 0;
+        uint64_t base = w * 32;
+        #pragma unroll 4
+        for (int j = 0; j < 32; j++) {
+            uint64_t idx = base + j;
+            if (idx < interval && first[idx] != 0) bits |= (1u << j);
+        }
+        bitset[w] = bits;
+    }
+}
+
+__global__ void k_derive_u16(const uint32_t* __restrict__ first, uint64_t interval,
+                             uint16_t* __restrict__ first16, uint32_t* __restrict__ overflow) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < interval; i += stride) {
+        uint32_t v = first[i];
+        if (v >= 65536u) atomicOr(overflow, 1u);
+        first16[i] = (uint16_t)v;
+    }
+}
+
 __global__ void k_build_range_direct(const int32_t* __restrict__ keys, uint64_t row_count,
                                      int64_t min_key, uint32_t* __restrict__ first,
                                      uint32_t* __restrict__ next) {
@@ -456,13 +490,31 @@ int gpue_join_build_payload_i32(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* pay
     if (rc != GPUE_OK) return rc;
     uint64_t interval = (uint64_t)(mx - mn + 1);
     ARG_CHECK(interval < (1ull << 32)); // selector's RANGE_DIRECT gate (join_hash_table.cpp:287)
-    gpue_join_table* t = new gpue_join_table{s, mn, mx, nullptr, nullptr, interval, n_rows};
+    gpue_join_table* t = new gpue_join_table();
+    t->s = s; t->min_key = mn; t->max_key = mx;
+    t->bucket_size = interval; t->row_count = n_rows;
     HIP_CHECK(hipMalloc(&t->first, interval * sizeof(uint32_t)));
     HIP_CHECK(hipMemsetAsync(t->first, 0, interval * sizeof(uint32_t), s->stream));
     hipLaunchKernelGGL(k_build_payload, dim3(grid_for(n_rows)), dim3(BLOCK), 0, s->stream,
                        (const int32_t*)keys->ptr, (const uint32_t*)payloads->ptr, n_rows,
                        mn, t->first);
     HIP_CHECK(hipGetLastError());
+    // derive the probe-side structures
+    uint64_t nwords = (interval + 31) / 32;
+    HIP_CHECK(hipMalloc(&t->bitset, nwords * sizeof(uint32_t)));
+    hipLaunchKernelGGL(k_derive_bitset, dim3(grid_for(nwords)), dim3(BLOCK), 0, s->stream,
+                       t->first, interval, t->bitset);
+    uint32_t* d_ovf = nullptr;
+    HIP_CHECK(hipMalloc(&d_ovf, sizeof(uint32_t)));
+    HIP_CHECK(hipMemsetAsync(d_ovf, 0, sizeof(uint32_t), s->stream));
+    HIP_CHECK(hipMalloc(&t->first16, interval * sizeof(uint16_t)));
+    hipLaunchKernelGGL(k_derive_u16, dim3(grid_for(interval)), dim3(BLOCK), 0, s->stream,
+                       t->first, interval, t->first16, d_ovf);
+    uint32_t ovf = 0;
+    HIP_CHECK(hipMemcpyAsync(&ovf, d_ovf, sizeof(uint32_t), hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_ovf);
+    if (ovf) { (void)hipFree(t->first16); t->first16 = nullptr; }
     *out = t;
     return GPUE_OK;
 }
@@ -477,7 +529,9 @@ int gpue_join_build_range_direct_i32(gpue_session* s, gpue_dbuf* keys, uint64_t 
     if (rc != GPUE_OK) return rc;
     uint64_t interval = (uint64_t)(mx - mn + 1);
     ARG_CHECK(interval < (1ull << 32));
-    gpue_join_table* t = new gpue_join_table{s, mn, mx, nullptr, nullptr, interval, row_count};
+    gpue_join_table* t = new gpue_join_table();
+    t->s = s; t->min_key = mn; t->max_key = mx;
+    t->bucket_size = interval; t->row_count = row_count;
     HIP_CHECK(hipMalloc(&t->first, interval * sizeof(uint32_t)));
     HIP_CHECK(hipMalloc(&t->next, (row_count + 1) * sizeof(uint32_t)));
     HIP_CHECK(hipMemsetAsync(t->first, 0, interval * sizeof(uint32_t), s->stream));
@@ -493,6 +547,8 @@ void gpue_join_table_destroy(gpue_join_table* t) {
     if (!t) return;
     if (t->first) (void)hipFree(t->first);
     if (t->next) (void)hipFree(t->next);
+    if (t->bitset) (void)hipFree(t->bitset);
+    if (t->first16) (void)hipFree(t->first16);
     delete t;
 }
 
@@ -636,6 +692,63 @@ int gpue_join_probe_emit_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* pro
 // ---------------------------------------------------------------------------
 // fused probe + aggregate pipelines
 // ---------------------------------------------------------------------------
+// Semi-join probe from an LDS-resident bitset: the u32-payload gather version
+// below measured issue-bound at ~2.0 TB/s (PMC traffic == algorithmic bytes,
+// profiles/pmc_q1*, so not bandwidth) — 64 random lanes per wave serialize in
+// the vector-memory path. The date dim needs only membership (SSB Q1 is a
+// semi join — the reference's RANGE_DIRECT_MAPPING_SET, 1 bit/key,
+// join_hash_table.cpp:297-303), and its bitset (61130 bits = 7.6 KB) stages
+// in LDS once per block: probes become ds_read, off the TA path entirely.
+__global__ void k_q1_join_sum_bitset(const int32_t* __restrict__ od,
+                                     const int32_t* __restrict__ ep,
+                                     const int32_t* __restrict__ dc, uint64_t n,
+                                     int64_t mn, int64_t mx,
+                                     const uint32_t* __restrict__ gbits,
+                                     unsigned long long* __restrict__ sum_out,
+                                     unsigned long long* __restrict__ cnt_out) {
+    extern __shared__ uint32_t lbits[];
+    const uint32_t interval = (uint32_t)(mx - mn + 1);
+    const uint32_t nwords = (interval + 31) / 32;
+    for (uint32_t w = threadIdx.x; w < nwords; w += blockDim.x) lbits[w] = gbits[w];
+    __syncthreads();
+    int64_t sum = 0;
+    uint64_t cnt = 0;
+    const uint64_t n4 = n / 4;
+    const int4* __restrict__ od4 = (const int4*)od;
+    const int4* __restrict__ ep4 = (const int4*)ep;
+    const int4* __restrict__ dc4 = (const int4*)dc;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+        int4 k4 = od4[i];
+        int4 e4 = ep4[i];
+        int4 d4 = dc4[i];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint32_t idx = (uint32_t)((&k4.x)[j] - (int32_t)mn);
+            uint32_t cidx = idx < interval ? idx : 0;
+            bool pass = (idx < interval) & ((lbits[cidx >> 5] >> (cidx & 31)) & 1u);
+            sum += pass ? (int64_t)(&e4.x)[j] * (&d4.x)[j] : 0;
+            cnt += pass;
+        }
+    }
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (uint64_t i = n4 * 4 + tid; i < n; i += stride) {
+        uint32_t idx = (uint32_t)(od[i] - (int32_t)mn);
+        if (idx < interval && ((lbits[idx >> 5] >> (idx & 31)) & 1u)) {
+            sum += (int64_t)ep[i] * dc[i];
+            cnt++;
+        }
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        sum += __shfl_down((long long)sum, off, WAVE);
+        cnt += __shfl_down((unsigned long long)cnt, off, WAVE);
+    }
+    if ((threadIdx.x & (WAVE - 1)) == 0) {
+        atomicAdd(sum_out, (unsigned long long)sum);
+        atomicAdd(cnt_out, (unsigned long long)cnt);
+    }
+}
+
 // 16 B/lane vectorized (Guideline 13: 4×int32 per column per iteration) with
 // UNCONDITIONAL column loads — a load guarded by the probe branch serializes
 // behind per-element vmcnt waits (guide §5 item 4(c)); the probe gather is
@@ -687,6 +800,24 @@ __global__ void k_q1_join_sum(const int32_t* __restrict__ od, const int32_t* __r
     }
 }
 
+
+// choose the LDS-bitset probe when the dim bitset fits comfortably in LDS
+// (<= 32 KiB leaves >= 4 blocks/CU); fall back to the payload-gather kernel
+static void launch_q1(gpue_session* s, gpue_join_table* dates, const int32_t* od,
+                      const int32_t* ep, const int32_t* dc, uint64_t n,
+                      unsigned long long* sum_out, unsigned long long* cnt_out) {
+    uint64_t nwords = (dates->bucket_size + 31) / 32;
+    if (dates->bitset && nwords * 4 <= 32768) {
+        hipLaunchKernelGGL(k_q1_join_sum_bitset, dim3(grid_for(n)), dim3(BLOCK),
+                           nwords * 4, s->stream, od, ep, dc, n, dates->min_key,
+                           dates->max_key, dates->bitset, sum_out, cnt_out);
+    } else {
+        hipLaunchKernelGGL(k_q1_join_sum, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+                           od, ep, dc, n, dates->min_key, dates->max_key, dates->first,
+                           sum_out, cnt_out);
+    }
+}
+
 int gpue_q1_join_sum(gpue_session* s, gpue_join_table* dates, gpue_dbuf* od, gpue_dbuf* ep,
                      gpue_dbuf* dc, uint64_t n, int64_t* sum_out, uint64_t* match_count_out) {
     ARG_CHECK(s && dates && od && ep && dc && sum_out && match_count_out);
@@ -695,10 +826,8 @@ int gpue_q1_join_sum(gpue_session* s, gpue_join_table* dates, gpue_dbuf* od, gpu
     unsigned long long* d_acc = nullptr;
     HIP_CHECK(hipMalloc(&d_acc, 2 * sizeof(unsigned long long)));
     HIP_CHECK(hipMemcpyAsync(d_acc, h_acc, sizeof(h_acc), hipMemcpyHostToDevice, s->stream));
-    hipLaunchKernelGGL(k_q1_join_sum, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
-                       (const int32_t*)od->ptr, (const int32_t*)ep->ptr,
-                       (const int32_t*)dc->ptr, n, dates->min_key, dates->max_key,
-                       dates->first, d_acc, d_acc + 1);
+    launch_q1(s, dates, (const int32_t*)od->ptr, (const int32_t*)ep->ptr,
+              (const int32_t*)dc->ptr, n, d_acc, d_acc + 1);
     HIP_CHECK(hipMemcpyAsync(h_acc, d_acc, sizeof(h_acc), hipMemcpyDeviceToHost, s->stream));
     HIP_CHECK(hipStreamSynchronize(s->stream));
     (void)hipFree(d_acc);
@@ -715,11 +844,9 @@ int gpue_q1_join_sum_async(gpue_session* s, gpue_join_table* dates, gpue_dbuf* o
                            gpue_dbuf* ep, gpue_dbuf* dc, uint64_t n, gpue_dbuf* acc) {
     ARG_CHECK(s && dates && od && ep && dc && acc && acc->bytes >= 16);
     HIP_CHECK(hipMemsetAsync(acc->ptr, 0, 16, s->stream));
-    hipLaunchKernelGGL(k_q1_join_sum, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
-                       (const int32_t*)od->ptr, (const int32_t*)ep->ptr,
-                       (const int32_t*)dc->ptr, n, dates->min_key, dates->max_key,
-                       dates->first, (unsigned long long*)acc->ptr,
-                       (unsigned long long*)acc->ptr + 1);
+    launch_q1(s, dates, (const int32_t*)od->ptr, (const int32_t*)ep->ptr,
+              (const int32_t*)dc->ptr, n, (unsigned long long*)acc->ptr,
+              (unsigned long long*)acc->ptr + 1);
     HIP_CHECK(hipGetLastError());
     return GPUE_OK;
 }
@@ -737,9 +864,9 @@ static constexpr int BLOCK_Q21 = 1024;
 __global__ __launch_bounds__(BLOCK_Q21) void
 k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
                const int32_t* __restrict__ od, const int32_t* __restrict__ rv,
-               uint64_t n, const uint32_t* __restrict__ pfirst,
-               const uint32_t* __restrict__ sfirst,
-               const uint32_t* __restrict__ dfirst, int64_t dmin,
+               uint64_t n, const uint16_t* __restrict__ pfirst,
+               const uint16_t* __restrict__ sfirst,
+               const uint16_t* __restrict__ dfirst, int64_t dmin,
                unsigned long long* __restrict__ group_sums) {
     __shared__ unsigned long long g[NG_Q21]; // 56 KB LDS
     for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x) g[j] = 0;
@@ -788,11 +915,12 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
     ARG_CHECK(s && parts && supps && dates && pk && sk && od && rv && group_sums);
     ARG_CHECK(group_sums->bytes >= NG_Q21 * sizeof(int64_t));
     ARG_CHECK(parts->min_key == 1 && supps->min_key == 1);
+    ARG_CHECK(parts->first16 && supps->first16 && dates->first16);
     HIP_CHECK(hipMemsetAsync(group_sums->ptr, 0, NG_Q21 * sizeof(int64_t), s->stream));
     hipLaunchKernelGGL(k_q21_star_agg, dim3(512), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)od->ptr, (const int32_t*)rv->ptr, n,
-                       parts->first, supps->first, dates->first, dates->min_key,
+                       parts->first16, supps->first16, dates->first16, dates->min_key,
                        (unsigned long long*)group_sums->ptr);
     HIP_CHECK(hipGetLastError());
     return GPUE_OK;
@@ -804,13 +932,14 @@ int gpue_q21_star_agg(gpue_session* s, gpue_join_table* parts, gpue_join_table* 
     ARG_CHECK(s && parts && supps && dates && pk && sk && od && rv && group_sums_out);
     ARG_CHECK(pk->bytes >= n * 4 && sk->bytes >= n * 4 && od->bytes >= n * 4 && rv->bytes >= n * 4);
     ARG_CHECK(parts->min_key == 1 && supps->min_key == 1); // keys are 1..N
+    ARG_CHECK(parts->first16 && supps->first16 && dates->first16); // payloads < 65536
     unsigned long long* d_g = nullptr;
     HIP_CHECK(hipMalloc(&d_g, NG_Q21 * sizeof(unsigned long long)));
     HIP_CHECK(hipMemsetAsync(d_g, 0, NG_Q21 * sizeof(unsigned long long), s->stream));
     hipLaunchKernelGGL(k_q21_star_agg, dim3(512), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)od->ptr, (const int32_t*)rv->ptr, n,
-                       parts->first, supps->first, dates->first, dates->min_key, d_g);
+                       parts->first16, supps->first16, dates->first16, dates->min_key, d_g);
     HIP_CHECK(hipMemcpyAsync(group_sums_out, d_g, NG_Q21 * sizeof(int64_t),
                              hipMemcpyDeviceToHost, s->stream));
     HIP_CHECK(hipStreamSynchronize(s->stream));
