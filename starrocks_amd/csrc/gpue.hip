@@ -722,13 +722,23 @@ __global__ void k_q1_join_sum_bitset(const int32_t* __restrict__ od,
         int4 k4 = od4[i];
         int4 e4 = ep4[i];
         int4 d4 = dc4[i];
+        // phase 1: issue ALL probe ds_reads back-to-back (a ?: / if around a
+        // load makes hipcc emit per-element branches + vmcnt(0)/lgkmcnt(0)
+        // waits — guide §5 item 4(c); verified in the .s of the branchy form)
+        uint32_t idx[4], w[4];
         #pragma unroll
         for (int j = 0; j < 4; j++) {
-            uint32_t idx = (uint32_t)((&k4.x)[j] - (int32_t)mn);
-            uint32_t cidx = idx < interval ? idx : 0;
-            bool pass = (idx < interval) & ((lbits[cidx >> 5] >> (cidx & 31)) & 1u);
-            sum += pass ? (int64_t)(&e4.x)[j] * (&d4.x)[j] : 0;
-            cnt += pass;
+            idx[j] = (uint32_t)((&k4.x)[j] - (int32_t)mn);
+            uint32_t cidx = idx[j] < interval ? idx[j] : 0u;
+            w[j] = lbits[cidx >> 5] >> (cidx & 31);
+        }
+        // phase 2: branchless masked accumulate (bitwise mask, no select on
+        // the product so the multiply is unconditional VALU work)
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            int64_t pass = (int64_t)((idx[j] < interval) & (w[j] & 1u));
+            sum += ((int64_t)(&e4.x)[j] * (&d4.x)[j]) & -pass;
+            cnt += (uint64_t)pass;
         }
     }
     uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -882,13 +892,20 @@ k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
         int4 s4 = sk4[i];
         int4 o4 = od4[i];
         int4 r4 = rv4[i];
+        // phase 1: the part-table gathers are needed for EVERY row — issue all
+        // four back-to-back so they overlap (a branch around each would add a
+        // per-element wait, guide §5 item 4(c))
+        uint16_t brand1[4];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) brand1[j] = pfirst[(&p4.x)[j] - 1];
+        // phase 2: the supplier/date gathers carry the 1/25 × 1/5 selectivity —
+        // branching here skips real traffic for 96 % of rows
         #pragma unroll
         for (int j = 0; j < 4; j++) {
-            uint32_t brand1 = pfirst[(&p4.x)[j] - 1];
-            if (brand1 == 0) continue;
+            if (brand1[j] == 0) continue;
             if (sfirst[(&s4.x)[j] - 1] == 0) continue;
             uint32_t year1 = dfirst[(&o4.x)[j] - dmin];
-            atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)],
+            atomicAdd(&g[(year1 - 1) * 1000 + (brand1[j] - 1)],
                       (unsigned long long)(int64_t)(&r4.x)[j]);
         }
     }
@@ -1036,6 +1053,73 @@ int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t nc
     (void)hipFree(d_off);
     free(h_hist);
     free(h_off);
+    return GPUE_OK;
+}
+
+
+// ---------------------------------------------------------------------------
+// streaming microbenchmarks — establish the achievable ceilings the fused
+// kernels are judged against (read via gpue_ubench from tools/)
+// ---------------------------------------------------------------------------
+__global__ void k_ub_sum3(const int4* __restrict__ a, const int4* __restrict__ b,
+                          const int4* __restrict__ c, uint64_t n4,
+                          unsigned long long* __restrict__ out) {
+    int64_t sum = 0;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+        int4 x = a[i], y = b[i], z = c[i];
+        sum += (int64_t)x.x + x.y + x.z + x.w + y.x + y.y + y.z + y.w
+             + (int64_t)z.x + z.y + z.z + z.w;
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        sum += __shfl_down((long long)sum, off, WAVE);
+    if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(out, (unsigned long long)sum);
+}
+
+__global__ void k_ub_sum1(const int4* __restrict__ a, uint64_t n4,
+                          unsigned long long* __restrict__ out) {
+    int64_t sum = 0;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+        int4 x = a[i];
+        sum += (int64_t)x.x + x.y + x.z + x.w;
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        sum += __shfl_down((long long)sum, off, WAVE);
+    if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(out, (unsigned long long)sum);
+}
+
+extern "C" int gpue_ubench(gpue_session* s, int which, gpue_dbuf* a, gpue_dbuf* b,
+                           gpue_dbuf* c, uint64_t n_i32, int reps, float* ms_out);
+int gpue_ubench(gpue_session* s, int which, gpue_dbuf* a, gpue_dbuf* b, gpue_dbuf* c,
+                uint64_t n_i32, int reps, float* ms_out) {
+    ARG_CHECK(s && a && ms_out && reps > 0);
+    uint64_t n4 = n_i32 / 4;
+    unsigned long long* d_out = nullptr;
+    HIP_CHECK(hipMalloc(&d_out, sizeof(unsigned long long)));
+    HIP_CHECK(hipMemsetAsync(d_out, 0, sizeof(unsigned long long), s->stream));
+    // one untimed warm launch
+    for (int pass = 0; pass < 2; pass++) {
+        if (pass == 1) HIP_CHECK(hipEventRecord(s->ev_start, s->stream));
+        int iters = pass == 0 ? 1 : reps;
+        for (int r = 0; r < iters; r++) {
+            if (which == 0) {
+                hipLaunchKernelGGL(k_ub_sum1, dim3(grid_for(n4)), dim3(BLOCK), 0, s->stream,
+                                   (const int4*)a->ptr, n4, d_out);
+            } else {
+                ARG_CHECK(b && c);
+                hipLaunchKernelGGL(k_ub_sum3, dim3(grid_for(n4)), dim3(BLOCK), 0, s->stream,
+                                   (const int4*)a->ptr, (const int4*)b->ptr,
+                                   (const int4*)c->ptr, n4, d_out);
+            }
+        }
+    }
+    HIP_CHECK(hipEventRecord(s->ev_stop, s->stream));
+    HIP_CHECK(hipEventSynchronize(s->ev_stop));
+    float ms = 0;
+    HIP_CHECK(hipEventElapsedTime(&ms, s->ev_start, s->ev_stop));
+    (void)hipFree(d_out);
+    *ms_out = ms / reps;
     return GPUE_OK;
 }
 
