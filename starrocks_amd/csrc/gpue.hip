@@ -718,13 +718,13 @@ __global__ void k_q1_join_sum_bitset(const int32_t* __restrict__ od,
     const int4* __restrict__ ep4 = (const int4*)ep;
     const int4* __restrict__ dc4 = (const int4*)dc;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
-        int4 k4 = od4[i];
-        int4 e4 = ep4[i];
-        int4 d4 = dc4[i];
-        // phase 1: issue ALL probe ds_reads back-to-back (a ?: / if around a
-        // load makes hipcc emit per-element branches + vmcnt(0)/lgkmcnt(0)
-        // waits — guide §5 item 4(c); verified in the .s of the branchy form)
+    // Two independent quads in flight per iteration: without this the next
+    // iteration's loads only issue after the current LDS-probe/accumulate
+    // tail (loop-carried serialization) — the 3-stream ubench ceiling needs
+    // ≥6 outstanding loads per wave. Probe phase batches all 8 ds_reads; a
+    // ?:/if around a load would emit per-element branches + vmcnt(0) waits
+    // (guide §5 item 4(c); verified in the .s of the branchy form).
+    auto quad = [&](int4 k4, int4 e4, int4 d4) {
         uint32_t idx[4], w[4];
         #pragma unroll
         for (int j = 0; j < 4; j++) {
@@ -732,15 +732,22 @@ __global__ void k_q1_join_sum_bitset(const int32_t* __restrict__ od,
             uint32_t cidx = idx[j] < interval ? idx[j] : 0u;
             w[j] = lbits[cidx >> 5] >> (cidx & 31);
         }
-        // phase 2: branchless masked accumulate (bitwise mask, no select on
-        // the product so the multiply is unconditional VALU work)
         #pragma unroll
         for (int j = 0; j < 4; j++) {
             int64_t pass = (int64_t)((idx[j] < interval) & (w[j] & 1u));
             sum += ((int64_t)(&e4.x)[j] * (&d4.x)[j]) & -pass;
             cnt += (uint64_t)pass;
         }
+    };
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i + stride < n4; i += 2 * stride) {
+        int4 ka = od4[i], ea = ep4[i], da = dc4[i];
+        uint64_t i2 = i + stride;
+        int4 kb = od4[i2], eb = ep4[i2], db = dc4[i2];
+        quad(ka, ea, da);
+        quad(kb, eb, db);
     }
+    for (; i < n4; i += stride) quad(od4[i], ep4[i], dc4[i]);
     uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     for (uint64_t i = n4 * 4 + tid; i < n; i += stride) {
         uint32_t idx = (uint32_t)(od[i] - (int32_t)mn);
@@ -874,8 +881,9 @@ static constexpr int BLOCK_Q21 = 1024;
 __global__ __launch_bounds__(BLOCK_Q21) void
 k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
                const int32_t* __restrict__ od, const int32_t* __restrict__ rv,
-               uint64_t n, const uint16_t* __restrict__ pfirst,
-               const uint16_t* __restrict__ sfirst,
+               uint64_t n, const uint32_t* __restrict__ pbits,
+               const uint16_t* __restrict__ pfirst,
+               const uint32_t* __restrict__ sbits,
                const uint16_t* __restrict__ dfirst, int64_t dmin,
                unsigned long long* __restrict__ group_sums) {
     __shared__ unsigned long long g[NG_Q21]; // 56 KB LDS
@@ -892,28 +900,38 @@ k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
         int4 s4 = sk4[i];
         int4 o4 = od4[i];
         int4 r4 = rv4[i];
-        // phase 1: the part-table gathers are needed for EVERY row — issue all
-        // four back-to-back so they overlap (a branch around each would add a
-        // per-element wait, guide §5 item 4(c))
-        uint16_t brand1[4];
-        #pragma unroll
-        for (int j = 0; j < 4; j++) brand1[j] = pfirst[(&p4.x)[j] - 1];
-        // phase 2: the supplier/date gathers carry the 1/25 × 1/5 selectivity —
-        // branching here skips real traffic for 96 % of rows
+        // phase 1: runtime-filter probe — the part table's 1-bit membership
+        // set (175 KB, L2-resident; the reference pushes exactly this filter
+        // to the scan, runtime_filter.h:79 / SURVEY.md §8f) replaces a 2.8 MB
+        // payload gather for the 96 % of rows the category filter rejects.
+        // All four word-gathers issue back-to-back (guide §5 item 4(c)).
+        uint32_t pb[4];
         #pragma unroll
         for (int j = 0; j < 4; j++) {
-            if (brand1[j] == 0) continue;
-            if (sfirst[(&s4.x)[j] - 1] == 0) continue;
+            uint32_t idx = (uint32_t)((&p4.x)[j] - 1);
+            pb[j] = pbits[idx >> 5] >> (idx & 31);
+        }
+        // phase 2: survivors (≈4 %) gather the brand payload and probe the
+        // supplier bitset (25 KB) and date payload — branching here skips
+        // real traffic
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            if (!(pb[j] & 1u)) continue;
+            uint32_t brand1 = pfirst[(&p4.x)[j] - 1];
+            uint32_t sidx = (uint32_t)((&s4.x)[j] - 1);
+            if (!((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
             uint32_t year1 = dfirst[(&o4.x)[j] - dmin];
-            atomicAdd(&g[(year1 - 1) * 1000 + (brand1[j] - 1)],
+            atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)],
                       (unsigned long long)(int64_t)(&r4.x)[j]);
         }
     }
     uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     for (uint64_t i = n4 * 4 + tid; i < n; i += stride) {
-        uint32_t brand1 = pfirst[pk[i] - 1];
-        if (brand1 == 0) continue;
-        if (sfirst[sk[i] - 1] == 0) continue;
+        uint32_t pidx = (uint32_t)(pk[i] - 1);
+        if (!((pbits[pidx >> 5] >> (pidx & 31)) & 1u)) continue;
+        uint32_t brand1 = pfirst[pidx];
+        uint32_t sidx = (uint32_t)(sk[i] - 1);
+        if (!((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
         uint32_t year1 = dfirst[od[i] - dmin];
         atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)], (unsigned long long)(int64_t)rv[i]);
     }
@@ -937,8 +955,8 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
     hipLaunchKernelGGL(k_q21_star_agg, dim3(512), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)od->ptr, (const int32_t*)rv->ptr, n,
-                       parts->first16, supps->first16, dates->first16, dates->min_key,
-                       (unsigned long long*)group_sums->ptr);
+                       parts->bitset, parts->first16, supps->bitset, dates->first16,
+                       dates->min_key, (unsigned long long*)group_sums->ptr);
     HIP_CHECK(hipGetLastError());
     return GPUE_OK;
 }
@@ -956,7 +974,8 @@ int gpue_q21_star_agg(gpue_session* s, gpue_join_table* parts, gpue_join_table* 
     hipLaunchKernelGGL(k_q21_star_agg, dim3(512), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)od->ptr, (const int32_t*)rv->ptr, n,
-                       parts->first16, supps->first16, dates->first16, dates->min_key, d_g);
+                       parts->bitset, parts->first16, supps->bitset, dates->first16,
+                       dates->min_key, d_g);
     HIP_CHECK(hipMemcpyAsync(group_sums_out, d_g, NG_Q21 * sizeof(int64_t),
                              hipMemcpyDeviceToHost, s->stream));
     HIP_CHECK(hipStreamSynchronize(s->stream));
