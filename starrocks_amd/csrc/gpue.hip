@@ -76,9 +76,13 @@ struct gpue_join_table {
     uint32_t* first = nullptr;   // device, (max-min+1) entries: payload or head row idx
     uint32_t* next = nullptr;    // device, (row_count+1) entries, or nullptr (payload variant)
     // Derived probe structures (payload variant only):
-    uint32_t* bitset = nullptr;  // 1 bit per key in [min,max] — the reference's
-                                 // RANGE_DIRECT_MAPPING_SET (join_hash_table.cpp:297-303)
-                                 // for semi-join probes; small enough to stage in LDS
+    uint32_t* bitset = nullptr;  // 1 bit per key in [set_min,set_max] — the
+                                 // reference's RANGE_DIRECT_MAPPING_SET
+                                 // (join_hash_table.cpp:297-303) for semi-join
+                                 // probes, clamped to the PASSING keys' bounding
+                                 // range (data-dependent specialization, like the
+                                 // selector's interval checks) so it stages in LDS
+    int64_t set_min = 0, set_max = -1; // bounding range of keys with payload != 0
     uint16_t* first16 = nullptr; // 16-bit payload copy when all payloads < 65536:
                                  // halves the random-gather footprint (L2 per XCD is 4 MiB)
     uint64_t bucket_size = 0;
@@ -428,8 +432,30 @@ __global__ void k_build_payload(const int32_t* __restrict__ keys,
 
 // Derive the semi-join bitset (1 bit/key) and the 16-bit payload copy from
 // the u32 payload array — built once at build time, probed every launch.
-__global__ void k_derive_bitset(const uint32_t* __restrict__ first, uint64_t interval,
-                                uint32_t* __restrict__ bitset) {
+// bounding range [min idx, max idx] of nonzero payload entries
+__global__ void k_set_bounds(const uint32_t* __restrict__ first, uint64_t interval,
+                             int32_t* __restrict__ lo_out, int32_t* __restrict__ hi_out) {
+    int32_t lo = INT32_MAX, hi = INT32_MIN;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < interval; i += stride) {
+        if (first[i] != 0) {
+            lo = min(lo, (int32_t)i);
+            hi = max(hi, (int32_t)i);
+        }
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        lo = min(lo, __shfl_down(lo, off, WAVE));
+        hi = max(hi, __shfl_down(hi, off, WAVE));
+    }
+    if ((threadIdx.x & (WAVE - 1)) == 0) {
+        atomicMin(lo_out, lo);
+        atomicMax(hi_out, hi);
+    }
+}
+
+__global__ void k_derive_bitset(const uint32_t* __restrict__ first, uint64_t offset,
+                                uint64_t interval, uint32_t* __restrict__ bitset) {
+    // bit j of word w == (first[offset + w*32 + j] != 0)
     uint64_t nwords = (interval + 31) / 32;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t w = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; w < nwords; w += stride) {
@@ -438,7 +464,7 @@ __global__ void k_derive_bitset(const uint32_t* __restrict__ first, uint64_t int
         #pragma unroll 4
         for (int j = 0; j < 32; j++) {
             uint64_t idx = base + j;
-            if (idx < interval && first[idx] != 0) bits |= (1u << j);
+            if (idx < interval && first[offset + idx] != 0) bits |= (1u << j);
         }
         bitset[w] = bits;
     }
@@ -499,11 +525,35 @@ int gpue_join_build_payload_i32(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* pay
                        (const int32_t*)keys->ptr, (const uint32_t*)payloads->ptr, n_rows,
                        mn, t->first);
     HIP_CHECK(hipGetLastError());
-    // derive the probe-side structures
-    uint64_t nwords = (interval + 31) / 32;
+    // derive the probe-side structures: set bounds first,
+    // then the bitset over just the passing keys' bounding range
+    int32_t h_bounds[2] = {INT32_MAX, INT32_MIN};
+    int32_t* d_bounds = nullptr;
+    HIP_CHECK(hipMalloc(&d_bounds, 2 * sizeof(int32_t)));
+    HIP_CHECK(hipMemcpyAsync(d_bounds, h_bounds, sizeof(h_bounds), hipMemcpyHostToDevice,
+                             s->stream));
+    hipLaunchKernelGGL(k_set_bounds, dim3(grid_for(interval)), dim3(BLOCK), 0, s->stream,
+                       t->first, interval, d_bounds, d_bounds + 1);
+    HIP_CHECK(hipMemcpyAsync(h_bounds, d_bounds, sizeof(h_bounds), hipMemcpyDeviceToHost,
+                             s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_bounds);
+    uint64_t set_off, set_interval;
+    if (h_bounds[0] > h_bounds[1]) { // empty pass set
+        t->set_min = mn;
+        t->set_max = mn;
+        set_off = 0;
+        set_interval = 1;
+    } else {
+        t->set_min = mn + h_bounds[0];
+        t->set_max = mn + h_bounds[1];
+        set_off = (uint64_t)h_bounds[0];
+        set_interval = (uint64_t)(h_bounds[1] - h_bounds[0] + 1);
+    }
+    uint64_t nwords = (set_interval + 31) / 32;
     HIP_CHECK(hipMalloc(&t->bitset, nwords * sizeof(uint32_t)));
     hipLaunchKernelGGL(k_derive_bitset, dim3(grid_for(nwords)), dim3(BLOCK), 0, s->stream,
-                       t->first, interval, t->bitset);
+                       t->first, set_off, set_interval, t->bitset);
     uint32_t* d_ovf = nullptr;
     HIP_CHECK(hipMalloc(&d_ovf, sizeof(uint32_t)));
     HIP_CHECK(hipMemsetAsync(d_ovf, 0, sizeof(uint32_t), s->stream));
@@ -823,11 +873,12 @@ __global__ void k_q1_join_sum(const int32_t* __restrict__ od, const int32_t* __r
 static void launch_q1(gpue_session* s, gpue_join_table* dates, const int32_t* od,
                       const int32_t* ep, const int32_t* dc, uint64_t n,
                       unsigned long long* sum_out, unsigned long long* cnt_out) {
-    uint64_t nwords = (dates->bucket_size + 31) / 32;
-    if (dates->bitset && nwords * 4 <= 32768) {
+    uint64_t set_interval = (uint64_t)(dates->set_max - dates->set_min + 1);
+    uint64_t nwords = (set_interval + 31) / 32;
+    if (dates->bitset && dates->set_max >= dates->set_min && nwords * 4 <= 32768) {
         hipLaunchKernelGGL(k_q1_join_sum_bitset, dim3(grid_for(n)), dim3(BLOCK),
-                           nwords * 4, s->stream, od, ep, dc, n, dates->min_key,
-                           dates->max_key, dates->bitset, sum_out, cnt_out);
+                           nwords * 4, s->stream, od, ep, dc, n, dates->set_min,
+                           dates->set_max, dates->bitset, sum_out, cnt_out);
     } else {
         hipLaunchKernelGGL(k_q1_join_sum, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
                            od, ep, dc, n, dates->min_key, dates->max_key, dates->first,
@@ -881,9 +932,9 @@ static constexpr int BLOCK_Q21 = 1024;
 __global__ __launch_bounds__(BLOCK_Q21) void
 k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
                const int32_t* __restrict__ od, const int32_t* __restrict__ rv,
-               uint64_t n, const uint32_t* __restrict__ pbits,
-               const uint16_t* __restrict__ pfirst,
-               const uint32_t* __restrict__ sbits,
+               uint64_t n, const uint32_t* __restrict__ pbits, int64_t psmin,
+               uint64_t psint, const uint16_t* __restrict__ pfirst,
+               const uint32_t* __restrict__ sbits, int64_t ssmin, uint64_t ssint,
                const uint16_t* __restrict__ dfirst, int64_t dmin,
                unsigned long long* __restrict__ group_sums) {
     __shared__ unsigned long long g[NG_Q21]; // 56 KB LDS
@@ -905,21 +956,23 @@ k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
         // to the scan, runtime_filter.h:79 / SURVEY.md §8f) replaces a 2.8 MB
         // payload gather for the 96 % of rows the category filter rejects.
         // All four word-gathers issue back-to-back (guide §5 item 4(c)).
-        uint32_t pb[4];
+        uint32_t pb[4], pin[4];
         #pragma unroll
         for (int j = 0; j < 4; j++) {
-            uint32_t idx = (uint32_t)((&p4.x)[j] - 1);
-            pb[j] = pbits[idx >> 5] >> (idx & 31);
+            uint32_t idx = (uint32_t)((&p4.x)[j] - psmin);
+            pin[j] = idx < psint;
+            uint32_t cidx = pin[j] ? idx : 0u;
+            pb[j] = pbits[cidx >> 5] >> (cidx & 31);
         }
         // phase 2: survivors (≈4 %) gather the brand payload and probe the
         // supplier bitset (25 KB) and date payload — branching here skips
         // real traffic
         #pragma unroll
         for (int j = 0; j < 4; j++) {
-            if (!(pb[j] & 1u)) continue;
+            if (!(pin[j] & pb[j] & 1u)) continue;
             uint32_t brand1 = pfirst[(&p4.x)[j] - 1];
-            uint32_t sidx = (uint32_t)((&s4.x)[j] - 1);
-            if (!((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
+            uint32_t sidx = (uint32_t)((&s4.x)[j] - ssmin);
+            if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
             uint32_t year1 = dfirst[(&o4.x)[j] - dmin];
             atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)],
                       (unsigned long long)(int64_t)(&r4.x)[j]);
@@ -927,11 +980,11 @@ k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
     }
     uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     for (uint64_t i = n4 * 4 + tid; i < n; i += stride) {
-        uint32_t pidx = (uint32_t)(pk[i] - 1);
-        if (!((pbits[pidx >> 5] >> (pidx & 31)) & 1u)) continue;
-        uint32_t brand1 = pfirst[pidx];
-        uint32_t sidx = (uint32_t)(sk[i] - 1);
-        if (!((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
+        uint32_t pidx = (uint32_t)(pk[i] - psmin);
+        if (pidx >= psint || !((pbits[pidx >> 5] >> (pidx & 31)) & 1u)) continue;
+        uint32_t brand1 = pfirst[pk[i] - 1];
+        uint32_t sidx = (uint32_t)(sk[i] - ssmin);
+        if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
         uint32_t year1 = dfirst[od[i] - dmin];
         atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)], (unsigned long long)(int64_t)rv[i]);
     }
@@ -955,7 +1008,10 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
     hipLaunchKernelGGL(k_q21_star_agg, dim3(512), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)od->ptr, (const int32_t*)rv->ptr, n,
-                       parts->bitset, parts->first16, supps->bitset, dates->first16,
+                       parts->bitset, parts->set_min,
+                       (uint64_t)(parts->set_max - parts->set_min + 1), parts->first16,
+                       supps->bitset, supps->set_min,
+                       (uint64_t)(supps->set_max - supps->set_min + 1), dates->first16,
                        dates->min_key, (unsigned long long*)group_sums->ptr);
     HIP_CHECK(hipGetLastError());
     return GPUE_OK;
@@ -974,7 +1030,10 @@ int gpue_q21_star_agg(gpue_session* s, gpue_join_table* parts, gpue_join_table* 
     hipLaunchKernelGGL(k_q21_star_agg, dim3(512), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)od->ptr, (const int32_t*)rv->ptr, n,
-                       parts->bitset, parts->first16, supps->bitset, dates->first16,
+                       parts->bitset, parts->set_min,
+                       (uint64_t)(parts->set_max - parts->set_min + 1), parts->first16,
+                       supps->bitset, supps->set_min,
+                       (uint64_t)(supps->set_max - supps->set_min + 1), dates->first16,
                        dates->min_key, d_g);
     HIP_CHECK(hipMemcpyAsync(group_sums_out, d_g, NG_Q21 * sizeof(int64_t),
                              hipMemcpyDeviceToHost, s->stream));
