@@ -946,11 +946,10 @@ k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
     const int4* __restrict__ od4 = (const int4*)od;
     const int4* __restrict__ rv4 = (const int4*)rv;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
-        int4 p4 = pk4[i];
-        int4 s4 = sk4[i];
-        int4 o4 = od4[i];
-        int4 r4 = rv4[i];
+    // SQ counters (profiles/pmc_sq_q21): 64 % of wave cycles parked on memory
+    // waits at full occupancy — latency-bound. Two independent quads per
+    // iteration double the loads in flight per wave.
+    auto quad = [&](int4 p4, int4 s4, int4 o4, int4 r4) {
         // phase 1: runtime-filter probe — the part table's 1-bit membership
         // set (175 KB, L2-resident; the reference pushes exactly this filter
         // to the scan, runtime_filter.h:79 / SURVEY.md §8f) replaces a 2.8 MB
@@ -977,7 +976,16 @@ k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
             atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)],
                       (unsigned long long)(int64_t)(&r4.x)[j]);
         }
+    };
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i + stride < n4; i += 2 * stride) {
+        int4 pa = pk4[i], sa = sk4[i], oa = od4[i], ra = rv4[i];
+        uint64_t i2 = i + stride;
+        int4 pb_ = pk4[i2], sb = sk4[i2], ob = od4[i2], rb = rv4[i2];
+        quad(pa, sa, oa, ra);
+        quad(pb_, sb, ob, rb);
     }
+    for (; i < n4; i += stride) quad(pk4[i], sk4[i], od4[i], rv4[i]);
     uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     for (uint64_t i = n4 * 4 + tid; i < n; i += stride) {
         uint32_t pidx = (uint32_t)(pk[i] - psmin);
