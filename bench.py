@@ -1,18 +1,19 @@
 #!/usr/bin/env python3
 """bench.py — measures BASELINE.json's metric: SSB rows/sec through the
-join+agg pipeline on MI355X, N=1 workload = configs[1] (SSB SF10
+join+agg pipeline on MI355X. N=1 workload = configs[1] (SSB SF10
 lineorder⋈date + SUM — the single-GPU quoted config).
 
 Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for N>1
 launched via torch.distributed.run with one rank per GPU over RCCL. W untimed
-warmup steps, then EXACTLY K timed steps bracketed by barrier +
-synchronize on both sides; MAX elapsed over ranks; rank 0 prints ONE JSON
-line. A step = one pass of the fused join+aggregate over the rank's resident
-shard (inputs already in HBM) + the cross-rank partial-aggregate merge
- + the 16-byte result read.
+warmup steps, then EXACTLY K timed steps bracketed by barrier + synchronize on
+both sides; MAX elapsed over ranks; rank 0 prints ONE JSON line. A step = one
+pass of the fused join+aggregate over the rank's resident shard (inputs
+already in HBM) + the cross-rank partial-aggregate merge + the
+result read.
 
---workload q21 measures config 3 (SF100 Q2.1 star join) instead; the default
-(and what the driver records) is the q1 config-2 line.
+Workloads: q1 (default, config 2), q21 (config 3), q43 (config 4 — at N>1 it
+runs the MANDATED hash-partitioned mode: per-step partition kernel + gather +
+RCCL all-to-all of the probe columns on lo_custkey, then local probe+agg).
 """
 
 import argparse
@@ -28,11 +29,11 @@ import numpy as np
 
 SEED = 42
 SF10_ROWS = 59_986_052
-SF100_ROWS = 600_000_000     # config 3
+SF100_ROWS = 600_000_000     # configs 3-4
 Q1_YEAR = 1993
 Q21_CATEGORY, Q21_REGION = 12, 2
-Q1_BYTES_PER_ROW = 12        # 3 × int32 read (algorithmic, §8d)
-Q21_BYTES_PER_ROW = 16       # 4 × int32 read
+Q43_REGION, Q43_NATION, Q43_CATEGORY = 1, 7, 12
+BYTES_PER_ROW = {"q1": 12, "q21": 16, "q43": 24}  # algorithmic (§8d)
 HBM_PEAK_GBPS = 8000.0       # 8 TB/s spec peak (MI355X_MICROARCH.md)
 
 
@@ -41,65 +42,89 @@ def log(msg):
         print(msg, file=sys.stderr, flush=True)
 
 
-def build_dim_tables(eng, workload):
+def build_payload_table(eng, keys_i32, payload_u32):
+    k = eng.alloc(keys_i32.nbytes)
+    k.h2d(keys_i32)
+    p = eng.alloc(payload_u32.nbytes)
+    p.h2d(payload_u32)
+    t = eng.join_build_payload(k, p, len(keys_i32))
+    k.free()
+    p.free()
+    return t
+
+
+def build_dim_tables(eng, workload, rank=0, world=1):
     from starrocks_amd import gen
     tables = {}
     datekey, dyear = gen.gen_dates()
-    keys = eng.alloc(datekey.nbytes)
-    keys.h2d(datekey.astype(np.int32))
+    dkeys = datekey.astype(np.int32)
     if workload == "q1":
-        payload = np.where(dyear == Q1_YEAR, dyear - 1992 + 1, 0).astype(np.uint32)
-    else:
-        payload = (dyear - 1992 + 1).astype(np.uint32)
-    pay = eng.alloc(payload.nbytes)
-    pay.h2d(payload)
-    tables["dates"] = eng.join_build_payload(keys, pay, len(datekey))
-    keys.free(); pay.free()
+        dpay = np.where(dyear == Q1_YEAR, dyear - 1992 + 1, 0).astype(np.uint32)
+    elif workload == "q21":
+        dpay = (dyear - 1992 + 1).astype(np.uint32)
+    else:  # q43: 1997 -> 1, 1998 -> 2
+        dpay = np.where(dyear == 1997, 1, np.where(dyear == 1998, 2, 0)).astype(np.uint32)
+    tables["dates"] = build_payload_table(eng, dkeys, dpay)
     if workload == "q21":
-        pfirst = gen.build_part_dim_payload(SEED, gen.N_PARTS_SF100, Q21_CATEGORY)
-        k = eng.alloc(gen.N_PARTS_SF100 * 4)
-        k.h2d(np.arange(1, gen.N_PARTS_SF100 + 1, dtype=np.int32))
-        p = eng.alloc(pfirst.nbytes); p.h2d(pfirst)
-        tables["parts"] = eng.join_build_payload(k, p, gen.N_PARTS_SF100)
-        k.free(); p.free()
-        sfirst = gen.build_supp_dim_payload(SEED, gen.N_SUPPS_SF100, Q21_REGION)
-        k = eng.alloc(gen.N_SUPPS_SF100 * 4)
-        k.h2d(np.arange(1, gen.N_SUPPS_SF100 + 1, dtype=np.int32))
-        p = eng.alloc(sfirst.nbytes); p.h2d(sfirst)
-        tables["supps"] = eng.join_build_payload(k, p, gen.N_SUPPS_SF100)
-        k.free(); p.free()
+        tables["parts"] = build_payload_table(
+            eng, np.arange(1, gen.N_PARTS_SF100 + 1, dtype=np.int32),
+            gen.build_part_dim_payload(SEED, gen.N_PARTS_SF100, Q21_CATEGORY))
+        tables["supps"] = build_payload_table(
+            eng, np.arange(1, gen.N_SUPPS_SF100 + 1, dtype=np.int32),
+            gen.build_supp_dim_payload(SEED, gen.N_SUPPS_SF100, Q21_REGION))
+    elif workload == "q43":
+        cpay = gen.build_cust_dim_q43(SEED, gen.N_CUSTS_SF100, Q43_REGION)
+        if world > 1:
+            # hash-partitioned join on lo_custkey: this rank owns only the
+            # customers the partition function routes to it
+            ckeys = np.arange(1, gen.N_CUSTS_SF100 + 1, dtype=np.uint32)
+            owned = gen.partition_channels(ckeys, world) == rank
+            cpay = np.where(owned, cpay, 0).astype(np.uint32)
+        tables["custs"] = build_payload_table(
+            eng, np.arange(1, gen.N_CUSTS_SF100 + 1, dtype=np.int32), cpay)
+        tables["supps"] = build_payload_table(
+            eng, np.arange(1, gen.N_SUPPS_SF100 + 1, dtype=np.int32),
+            gen.build_supp_dim_q43(SEED, gen.N_SUPPS_SF100, Q43_NATION))
+        tables["parts"] = build_payload_table(
+            eng, np.arange(1, gen.N_PARTS_SF100 + 1, dtype=np.int32),
+            gen.build_part_dim_q43(SEED, gen.N_PARTS_SF100, Q43_CATEGORY))
     return tables
 
 
 def cpu_baseline(workload, rows_full):
     """Oracle (kind 'port') on this box's host cores; bounded sample
-    (~10-30 s of CPU work), compute-only region timed (generation excluded,
+    (~10 s of CPU work), compute-only region timed (generation excluded,
     matching the GPU timed region)."""
     from oracle import pyoracle as orc
     from starrocks_amd import gen
     cores = os.cpu_count()
-    sample = min(rows_full, 60_000_000)
     if workload == "q1":
+        sample = min(rows_full, 60_000_000)
         od, ep, dc = orc.gen_lineorder_q1(SEED, 0, sample)
         mn, mx, dfirst = gen.build_date_dim_payload(Q1_YEAR)
-        t0 = time.perf_counter()
-        passes = 0
-        while time.perf_counter() - t0 < 10.0:
-            orc.q1_kernel(od, ep, dc, dfirst, mn, mx)
-            passes += 1
-        dt = time.perf_counter() - t0
-    else:
+        run = lambda: orc.q1_kernel(od, ep, dc, dfirst, mn, mx)
+    elif workload == "q21":
         sample = min(rows_full, 30_000_000)
         pk, sk, od, rv = orc.gen_lineorder_q21(SEED, 0, sample)
         pfirst = gen.build_part_dim_payload(SEED, gen.N_PARTS_SF100, Q21_CATEGORY)
         sfirst = gen.build_supp_dim_payload(SEED, gen.N_SUPPS_SF100, Q21_REGION)
         mn, _, dfirst = gen.build_date_dim_payload(None)
-        t0 = time.perf_counter()
-        passes = 0
-        while time.perf_counter() - t0 < 10.0:
-            orc.q21_kernel(pk, sk, od, rv, pfirst, sfirst, dfirst, mn)
-            passes += 1
-        dt = time.perf_counter() - t0
+        run = lambda: orc.q21_kernel(pk, sk, od, rv, pfirst, sfirst, dfirst, mn)
+    else:
+        sample = min(rows_full, 30_000_000)
+        ck, sk, pk, od, rv, sc = orc.gen_lineorder_q43(SEED, 0, sample)
+        cfirst = gen.build_cust_dim_q43(SEED, gen.N_CUSTS_SF100, Q43_REGION)
+        sfirst = gen.build_supp_dim_q43(SEED, gen.N_SUPPS_SF100, Q43_NATION)
+        pfirst = gen.build_part_dim_q43(SEED, gen.N_PARTS_SF100, Q43_CATEGORY)
+        mn, _, dfirst = gen.build_date_dim_q43()
+        run = lambda: orc.q43_kernel(ck, sk, pk, od, rv, sc, cfirst, sfirst,
+                                     pfirst, dfirst, mn)
+    t0 = time.perf_counter()
+    passes = 0
+    while time.perf_counter() - t0 < 10.0:
+        run()
+        passes += 1
+    dt = time.perf_counter() - t0
     rate = passes * sample / dt
     return {"value": round(rate, 1), "unit": "rows/s", "cores": cores, "kind": "port",
             "sample": f"{passes} passes over {sample} rows ({dt:.1f}s, oracle -O3 -fopenmp, "
@@ -110,8 +135,7 @@ def read_pmc_traffic(workload):
     p = os.path.join(REPO, "profiles", f"pmc_{workload}.json")
     if os.path.exists(p):
         with open(p) as f:
-            d = json.load(f)
-        return d.get("traffic_bytes_per_launch")
+            return json.load(f).get("traffic_bytes_per_launch")
     return None
 
 
@@ -120,68 +144,133 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=50)
     ap.add_argument("--warmup", type=int, default=10)
-    ap.add_argument("--workload", choices=["q1", "q21"], default="q1")
+    ap.add_argument("--workload", choices=["q1", "q21", "q43"], default="q1")
     ap.add_argument("--rows", type=int, default=0, help="override rows per GPU")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", args.gpus))
     rank = int(os.environ.get("RANK", 0))
     local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    wl = args.workload
 
     from starrocks_amd.engine import Engine
 
     dist = None
+    torch = None
     if world > 1:
-        import torch
+        import torch as _torch
         import torch.distributed as tdist
+        torch = _torch
         torch.cuda.set_device(local_rank)
         tdist.init_process_group("nccl")
         dist = tdist
 
     eng = Engine(local_rank)
-    rows = args.rows or (SF10_ROWS if args.workload == "q1" else SF100_ROWS)
+    rows = args.rows or (SF10_ROWS if wl == "q1" else SF100_ROWS)
     row_start = rank * rows  # weak scaling: each rank owns its shard
 
-    log(f"[bench] workload={args.workload} rows/gpu={rows} world={world} "
+    log(f"[bench] workload={wl} rows/gpu={rows} world={world} "
         f"steps={args.steps} warmup={args.warmup}")
-
-    # ---- untimed setup: generate shard on device, build dim tables ----
     t_setup = time.perf_counter()
-    if args.workload == "q1":
+    tables = build_dim_tables(eng, wl, rank, world)
+    n_merge = {"q1": 2, "q21": 7000, "q43": 800}[wl]
+
+    # ---- untimed setup: generate shard on device, build step closure ----
+    if wl == "q1":
         cols = [eng.alloc(rows * 4) for _ in range(3)]
         eng.gen_lineorder_q1(SEED, row_start, rows, *cols)
-        tables = build_dim_tables(eng, "q1")
         acc = eng.alloc(16)
 
-        def step():
+        def kernel_only():
             eng.q1_join_sum_async(tables["dates"], cols[0], cols[1], cols[2], rows, acc)
-            vals = acc.d2h(np.int64, 2)  # result read (syncs stream)
-            return vals
-    else:
-        cols = [eng.alloc(rows * 4) for _ in range(4)]
-        eng.gen_lineorder_q21(SEED, row_start, rows, *cols)
-        tables = build_dim_tables(eng, "q21")
-        acc = eng.alloc(7000 * 8)
 
         def step():
+            kernel_only()
+            return acc.d2h(np.int64, 2)
+    elif wl == "q21":
+        cols = [eng.alloc(rows * 4) for _ in range(4)]
+        eng.gen_lineorder_q21(SEED, row_start, rows, *cols)
+        acc = eng.alloc(7000 * 8)
+
+        def kernel_only():
             eng.q21_star_agg_async(tables["parts"], tables["supps"], tables["dates"],
                                    cols[0], cols[1], cols[2], cols[3], rows, acc)
+
+        def step():
+            kernel_only()
             return acc.d2h(np.int64, 7000)
+    else:  # q43
+        acc = eng.alloc(800 * 8)
+        if world == 1:
+            cols = [eng.alloc(rows * 4) for _ in range(6)]
+            eng.gen_lineorder_q43(SEED, row_start, rows, *cols)
+
+            def kernel_only():
+                eng.q43_star_agg_async(tables["custs"], tables["supps"], tables["parts"],
+                                       tables["dates"], *cols, rows, acc)
+
+            def step():
+                kernel_only()
+                return acc.d2h(np.int64, 800)
+        else:
+            # hash-partitioned mode: columns live in torch cuda tensors so the
+            # RCCL all-to-all moves the same buffers the kernels fill
+            cols_t = [torch.empty(rows, dtype=torch.int32, device="cuda")
+                      for _ in range(6)]
+            cols = [eng.wrap_ptr(t.data_ptr(), rows * 4) for t in cols_t]
+            eng.gen_lineorder_q43(SEED, row_start, rows, *cols)
+            eng.sync()
+            send_t = [torch.empty(rows, dtype=torch.int32, device="cuda")
+                      for _ in range(6)]
+            send = [eng.wrap_ptr(t.data_ptr(), rows * 4) for t in send_t]
+            ridx = eng.alloc(rows * 4)
+
+            # size discovery (untimed): splits are static since data is static
+            sp = eng.partition(cols[0], rows, world, ridx)
+            in_splits = np.diff(sp).astype(np.int64)
+            in_t = torch.from_numpy(in_splits).cuda()
+            out_t = torch.empty(world, dtype=torch.int64, device="cuda")
+            dist.all_to_all_single(out_t, in_t)
+            out_splits = out_t.cpu().numpy()
+            n_recv = int(out_splits.sum())
+            recv_t = [torch.empty(max(n_recv, 1), dtype=torch.int32, device="cuda")
+                      for _ in range(6)]
+            recv = [eng.wrap_ptr(t.data_ptr(), max(n_recv, 1) * 4) for t in recv_t]
+            in_list = [int(x) for x in in_splits]
+            out_list = [int(x) for x in out_splits]
+
+            def kernel_only():
+                eng.q43_star_agg_async(tables["custs"], tables["supps"], tables["parts"],
+                                       tables["dates"], *recv, n_recv, acc)
+
+            def step():
+                # partition (crc/fnv->channel + counting sort) + gather: the
+                # exchange sink stage (exchange_sink_operator.cpp:611-660)
+                eng.partition(cols[0], rows, world, ridx)
+                for c, s_ in zip(cols, send):
+                    eng.gather_u32(c, ridx, rows, s_)
+                eng.sync()
+                # the brpc transmit_chunk leg -> RCCL all-to-all over xGMI
+                for st, rt in zip(send_t, recv_t):
+                    dist.all_to_all_single(rt[:n_recv], st, out_list, in_list)
+                torch.cuda.synchronize()
+                kernel_only()
+                return acc.d2h(np.int64, 800)
 
     eng.sync()
     log(f"[bench] setup {time.perf_counter()-t_setup:.1f}s")
 
     merge_buf = None
     if world > 1:
-        import torch
-        merge_buf = torch.zeros(2 if args.workload == "q1" else 7000,
-                                dtype=torch.int64, device="cuda")
+        merge_buf = torch.zeros(n_merge, dtype=torch.int64, device="cuda")
 
     def run_step():
         vals = step()
         if dist is not None:
+            # phase1 -> phase2 aggregate merge (agg_hash_variant.h merge_batch
+            # semantics) over RCCL
             merge_buf.copy_(torch.from_numpy(np.asarray(vals)))
-            dist.all_reduce(merge_buf)  # phase1→phase2 aggregate merge (RCCL)
+            dist.all_reduce(merge_buf)
             return merge_buf
         return vals
 
@@ -190,7 +279,6 @@ def main():
 
     # ---- timed region ----
     if dist is not None:
-        import torch
         dist.barrier()
         torch.cuda.synchronize()
     eng.sync()
@@ -199,13 +287,11 @@ def main():
         run_step()
     eng.sync()
     if dist is not None:
-        import torch
         torch.cuda.synchronize()
         dist.barrier()
     elapsed = time.perf_counter() - t0
 
     if dist is not None:
-        import torch
         t = torch.tensor([elapsed], device="cuda")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
@@ -214,26 +300,21 @@ def main():
     ms_per_step = elapsed / args.steps * 1000
 
     # ---- roofline evidence: HIP events around kernel-only launches ----
-    bytes_per_row = Q1_BYTES_PER_ROW if args.workload == "q1" else Q21_BYTES_PER_ROW
     R = 20
     eng.sync()
     eng.timer_start()
     for _ in range(R):
-        if args.workload == "q1":
-            eng.q1_join_sum_async(tables["dates"], cols[0], cols[1], cols[2], rows, acc)
-        else:
-            eng.q21_star_agg_async(tables["parts"], tables["supps"], tables["dates"],
-                                   cols[0], cols[1], cols[2], cols[3], rows, acc)
+        kernel_only()
     kernel_ms = eng.timer_stop() / R
-    algo_bytes = bytes_per_row * rows
+    kernel_rows = rows if not (wl == "q43" and world > 1) else max(n_recv, 1)
+    algo_bytes = BYTES_PER_ROW[wl] * kernel_rows
     achieved_gbps = algo_bytes / (kernel_ms / 1e3) / 1e9
     roofline = {"bound": "hbm", "achieved": round(achieved_gbps, 1), "peak": HBM_PEAK_GBPS,
                 "unit": "GB/s", "frac": round(achieved_gbps / HBM_PEAK_GBPS, 4),
-                "traffic": read_pmc_traffic(args.workload)}
+                "traffic": read_pmc_traffic(wl)}
 
-    result = None
     if rank == 0:
-        cb = cpu_baseline(args.workload, rows) if world == 1 else None
+        cb = cpu_baseline(wl, rows) if world == 1 else None
         result = {
             "metric": "ssb_join_agg_rows_per_sec",
             "value": round(value, 1),
@@ -248,22 +329,20 @@ def main():
             "dtype": "int64",
             "data": "synthetic",
             "config": {
-                "workload": ("ssb_sf10_q1_join_sum" if args.workload == "q1"
-                             else "ssb_sf100_q21_star_groupby"),
+                "workload": {"q1": "ssb_sf10_q1_join_sum",
+                             "q21": "ssb_sf100_q21_star_groupby",
+                             "q43": "ssb_sf100_q43_4way_star_groupby"}[wl],
                 "rows_per_gpu": rows,
                 "seed": SEED,
-                "parallelism": f"dp{world}-replicated-dims",
+                "parallelism": (f"dp{world}-hash-partitioned-alltoall"
+                                if wl == "q43" and world > 1
+                                else f"dp{world}-replicated-dims"),
             },
             "roofline": roofline,
             "cpu_baseline": cb,
         }
         print(json.dumps(result), flush=True)
 
-    for c in cols:
-        c.free()
-    acc.free()
-    for t in tables.values():
-        t.destroy()
     eng.close()
     if dist is not None:
         dist.destroy_process_group()

@@ -51,6 +51,12 @@ void gpue_dbuf_free(gpue_dbuf* b);
 int gpue_dbuf_h2d(gpue_dbuf* b, const void* src, uint64_t bytes, uint64_t dst_off);
 int gpue_dbuf_d2h(gpue_dbuf* b, void* dst, uint64_t bytes, uint64_t src_off);
 int gpue_dbuf_memset(gpue_dbuf* b, int value, uint64_t bytes);
+/* Wrap external device memory (e.g. a torch tensor's data_ptr) so kernels
+ * operate in place and torch.distributed (RCCL) moves the same buffers —
+ * the all-to-all leg of configs 4-5. Caller keeps ownership. */
+int gpue_dbuf_wrap(gpue_session* s, void* device_ptr, uint64_t bytes, gpue_dbuf** out);
+/* Expose the device pointer (torch interop / sub-buffer views). */
+int gpue_dbuf_ptr(gpue_dbuf* b, void** out);
 
 /* ---- synthetic chunk source ----
  * Replaces the scan operator with a seeded on-device generator (SURVEY.md §2
@@ -68,6 +74,10 @@ int gpue_gen_lineorder_q1(gpue_session* s, uint64_t seed, uint64_t row_start, ui
 int gpue_gen_lineorder_q21(gpue_session* s, uint64_t seed, uint64_t row_start, uint64_t n,
                            gpue_dbuf* lo_partkey, gpue_dbuf* lo_suppkey,
                            gpue_dbuf* lo_orderdate, gpue_dbuf* lo_revenue);
+int gpue_gen_lineorder_q43(gpue_session* s, uint64_t seed, uint64_t row_start, uint64_t n,
+                           gpue_dbuf* lo_custkey, gpue_dbuf* lo_suppkey,
+                           gpue_dbuf* lo_partkey, gpue_dbuf* lo_orderdate,
+                           gpue_dbuf* lo_revenue, gpue_dbuf* lo_supplycost);
 
 /* ---- scan + predicate filter ----
  * Replaces ChunkPredicateEvaluator::eval_conjuncts + Column::filter_range
@@ -141,6 +151,16 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
                             gpue_dbuf* lo_orderdate, gpue_dbuf* lo_revenue, uint64_t n_rows,
                             gpue_dbuf* group_sums);
 
+/* Config 4 (SSB Q4.3): 4-way star join + GROUP BY (d_year,s_city,p_brand) as
+ * compact filtered indexes -> 800 groups; SUM(lo_revenue - lo_supplycost).
+ * group_sums holds 800 x int64 on device. */
+int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs, gpue_join_table* supps,
+                            gpue_join_table* parts, gpue_join_table* dates,
+                            gpue_dbuf* lo_custkey, gpue_dbuf* lo_suppkey,
+                            gpue_dbuf* lo_partkey, gpue_dbuf* lo_orderdate,
+                            gpue_dbuf* lo_revenue, gpue_dbuf* lo_supplycost,
+                            uint64_t n_rows, gpue_dbuf* group_sums);
+
 /* ---- exchange partition (shuffle groundwork for configs 4-5) ----
  * Replaces the ExchangeSinkOperator partition stage
  * (reference be/src/exec/pipeline/exchange/exchange_sink_operator.cpp:611-660,
@@ -149,6 +169,12 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
  * ascend by source row). start_points has num_channels+1 entries. */
 int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t num_channels,
                        uint64_t* start_points_out, gpue_dbuf* row_indexes_out);
+
+/* Row gather by index — the exchange sink's add_rows_selective analog
+ * (exchange_sink_operator.cpp:670): materializes per-channel row slices at
+ * the counting-sorted indexes before the RCCL all-to-all. */
+int gpue_gather_u32(gpue_session* s, gpue_dbuf* in, gpue_dbuf* idx, uint64_t n,
+                    gpue_dbuf* out);
 
 /* ---- event timing on the session stream (bench roofline evidence) ---- */
 int gpue_timer_start(gpue_session* s);

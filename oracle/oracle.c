@@ -377,11 +377,14 @@ void orc_partition_counting_sort(const uint32_t* channel_ids, uint64_t n,
 
 enum { TAG_ORDERDATE = 1, TAG_EXTPRICE = 2, TAG_DISCOUNT = 3,
        TAG_PARTKEY = 4, TAG_SUPPKEY = 5, TAG_REVENUE = 6,
-       TAG_PCAT = 7, TAG_PBRD = 8, TAG_SREG = 9 };
+       TAG_PCAT = 7, TAG_PBRD = 8, TAG_SREG = 9,
+       TAG_CUSTKEY = 10, TAG_SUPPCOST = 11, TAG_CREG = 12,
+       TAG_SNAT = 13, TAG_SCITY = 14 };
 
 #define N_DAYS 2556
 #define N_PARTS_SF100 1400000u
 #define N_SUPPS_SF100 200000u
+#define N_CUSTS_SF100 3000000u
 
 void orc_gen_lineorder_q1(uint64_t seed, uint64_t row_start, uint64_t n,
                           int32_t* lo_orderdate, int32_t* lo_extendedprice,
@@ -410,6 +413,135 @@ void orc_gen_lineorder_q21(uint64_t seed, uint64_t row_start, uint64_t n,
         lo_orderdate[i] = datekey[orc_gen_u64(seed, TAG_ORDERDATE, r) % N_DAYS];
         lo_revenue[i] = (int32_t)(orc_gen_u64(seed, TAG_REVENUE, r) % 10000000u);
     }
+}
+
+/* SSB Q4.3 columns: 6 × int32 per lineorder row */
+void orc_gen_lineorder_q43(uint64_t seed, uint64_t row_start, uint64_t n,
+                           int32_t* lo_custkey, int32_t* lo_suppkey,
+                           int32_t* lo_partkey, int32_t* lo_orderdate,
+                           int32_t* lo_revenue, int32_t* lo_supplycost) {
+    int32_t datekey[N_DAYS];
+    orc_gen_dates(N_DAYS, datekey, NULL);
+#pragma omp parallel for schedule(static)
+    for (uint64_t i = 0; i < n; i++) {
+        uint64_t r = row_start + i;
+        lo_custkey[i] = (int32_t)(orc_gen_u64(seed, TAG_CUSTKEY, r) % N_CUSTS_SF100) + 1;
+        lo_suppkey[i] = (int32_t)(orc_gen_u64(seed, TAG_SUPPKEY, r) % N_SUPPS_SF100) + 1;
+        lo_partkey[i] = (int32_t)(orc_gen_u64(seed, TAG_PARTKEY, r) % N_PARTS_SF100) + 1;
+        lo_orderdate[i] = datekey[orc_gen_u64(seed, TAG_ORDERDATE, r) % N_DAYS];
+        lo_revenue[i] = (int32_t)(orc_gen_u64(seed, TAG_REVENUE, r) % 10000000u);
+        lo_supplycost[i] = (int32_t)(orc_gen_u64(seed, TAG_SUPPCOST, r) % 100000u) + 1;
+    }
+}
+
+uint32_t orc_cust_region(uint64_t seed, uint32_t custkey) {
+    return (uint32_t)(orc_gen_u64(seed, TAG_CREG, custkey) % 5u);
+}
+uint32_t orc_supp_nation(uint64_t seed, uint32_t suppkey) {
+    return (uint32_t)(orc_gen_u64(seed, TAG_SNAT, suppkey) % 25u);
+}
+uint32_t orc_supp_city_in_nation(uint64_t seed, uint32_t suppkey) {
+    return (uint32_t)(orc_gen_u64(seed, TAG_SCITY, suppkey) % 10u);
+}
+uint32_t orc_part_brand_in_category(uint64_t seed, uint32_t partkey) {
+    return (uint32_t)(orc_gen_u64(seed, TAG_PBRD, partkey) % 40u);
+}
+
+/* Q4.3 dim payload arrays (compact filtered indexes):
+ * cust:  1 if c_region==region else 0
+ * supp:  city_in_nation+1 (1..10) if s_nation==nation else 0
+ * part:  brand_in_category+1 (1..40) if p_category==category else 0
+ * date:  1 for 1997, 2 for 1998, else 0 */
+void orc_build_cust_dim_q43(uint64_t seed, uint32_t n_custs, int32_t region, uint32_t* first) {
+#pragma omp parallel for schedule(static)
+    for (uint32_t c = 1; c <= n_custs; c++)
+        first[c - 1] = (orc_cust_region(seed, c) == (uint32_t)region) ? 1u : 0u;
+}
+void orc_build_supp_dim_q43(uint64_t seed, uint32_t n_supps, int32_t nation, uint32_t* first) {
+#pragma omp parallel for schedule(static)
+    for (uint32_t s = 1; s <= n_supps; s++)
+        first[s - 1] = (orc_supp_nation(seed, s) == (uint32_t)nation)
+                               ? orc_supp_city_in_nation(seed, s) + 1u : 0u;
+}
+void orc_build_part_dim_q43(uint64_t seed, uint32_t n_parts, int32_t category, uint32_t* first) {
+#pragma omp parallel for schedule(static)
+    for (uint32_t p = 1; p <= n_parts; p++)
+        first[p - 1] = (orc_part_category(seed, p) == (uint32_t)category)
+                               ? orc_part_brand_in_category(seed, p) + 1u : 0u;
+}
+
+/* group id = (dpay-1)*400 + (spay-1)*40 + (ppay-1), 2*10*40 = 800 groups;
+ * SUM(lo_revenue - lo_supplycost) per group. */
+#define NG_Q43 800
+void orc_q43_kernel(const int32_t* ck, const int32_t* sk, const int32_t* pk,
+                    const int32_t* od, const int32_t* rv, const int32_t* sc,
+                    uint64_t n_rows, const uint32_t* cfirst, const uint32_t* sfirst,
+                    const uint32_t* pfirst, const uint32_t* dfirst, int64_t dmin,
+                    int threads, int64_t* group_sums) {
+#ifdef _OPENMP
+    if (threads > 0) omp_set_num_threads(threads);
+    int nt = omp_get_max_threads();
+#else
+    int nt = 1;
+#endif
+    int64_t* partials = (int64_t*)calloc((size_t)nt * NG_Q43, sizeof(int64_t));
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        int t = omp_get_thread_num();
+#else
+        int t = 0;
+#endif
+        int64_t* local = partials + (size_t)t * NG_Q43;
+#pragma omp for schedule(static)
+        for (uint64_t i = 0; i < n_rows; i++) {
+            uint32_t ppay = pfirst[pk[i] - 1];
+            if (ppay == 0) continue;
+            uint32_t spay = sfirst[sk[i] - 1];
+            if (spay == 0) continue;
+            if (cfirst[ck[i] - 1] == 0) continue;
+            uint32_t dpay = dfirst[od[i] - dmin];
+            if (dpay == 0) continue;
+            local[(dpay - 1) * 400 + (spay - 1) * 40 + (ppay - 1)] += (int64_t)rv[i] - sc[i];
+        }
+    }
+    for (int t = 0; t < nt; t++)
+        for (int g = 0; g < NG_Q43; g++) group_sums[g] += partials[(size_t)t * NG_Q43 + g];
+    free(partials);
+}
+
+void orc_q43_pipeline(uint64_t seed, uint64_t row_start, uint64_t n_rows,
+                      int32_t region, int32_t nation, int32_t category,
+                      int threads, int64_t* group_sums) {
+    int32_t* ck = (int32_t*)malloc(n_rows * 4);
+    int32_t* sk = (int32_t*)malloc(n_rows * 4);
+    int32_t* pk = (int32_t*)malloc(n_rows * 4);
+    int32_t* od = (int32_t*)malloc(n_rows * 4);
+    int32_t* rv = (int32_t*)malloc(n_rows * 4);
+    int32_t* sc = (int32_t*)malloc(n_rows * 4);
+    orc_gen_lineorder_q43(seed, row_start, n_rows, ck, sk, pk, od, rv, sc);
+
+    uint32_t* cfirst = (uint32_t*)malloc(N_CUSTS_SF100 * sizeof(uint32_t));
+    orc_build_cust_dim_q43(seed, N_CUSTS_SF100, region, cfirst);
+    uint32_t* sfirst = (uint32_t*)malloc(N_SUPPS_SF100 * sizeof(uint32_t));
+    orc_build_supp_dim_q43(seed, N_SUPPS_SF100, nation, sfirst);
+    uint32_t* pfirst = (uint32_t*)malloc(N_PARTS_SF100 * sizeof(uint32_t));
+    orc_build_part_dim_q43(seed, N_PARTS_SF100, category, pfirst);
+
+    int32_t* datekey = (int32_t*)malloc(N_DAYS * sizeof(int32_t));
+    int32_t* dyear = (int32_t*)malloc(N_DAYS * sizeof(int32_t));
+    orc_gen_dates(N_DAYS, datekey, dyear);
+    int32_t dmn = datekey[0], dmx = datekey[N_DAYS - 1];
+    uint32_t* dfirst = (uint32_t*)calloc((size_t)(dmx - dmn + 1), sizeof(uint32_t));
+    for (int32_t i = 0; i < N_DAYS; i++) {
+        if (dyear[i] == 1997) dfirst[datekey[i] - dmn] = 1;
+        else if (dyear[i] == 1998) dfirst[datekey[i] - dmn] = 2;
+    }
+
+    orc_q43_kernel(ck, sk, pk, od, rv, sc, n_rows, cfirst, sfirst, pfirst, dfirst,
+                   dmn, threads, group_sums);
+    free(ck); free(sk); free(pk); free(od); free(rv); free(sc);
+    free(cfirst); free(sfirst); free(pfirst); free(dfirst); free(datekey); free(dyear);
 }
 
 uint32_t orc_part_category(uint64_t seed, uint32_t partkey) {

@@ -129,6 +129,27 @@ uint32_t orc_part_category(uint64_t seed, uint32_t partkey);
 uint32_t orc_part_brand(uint64_t seed, uint32_t partkey);
 uint32_t orc_supp_region(uint64_t seed, uint32_t suppkey);
 
+/* ---- SSB Q4.3 (config 4): 4-way star join + 2-key GROUP BY ---- */
+void orc_gen_lineorder_q43(uint64_t seed, uint64_t row_start, uint64_t n,
+                           int32_t* lo_custkey, int32_t* lo_suppkey,
+                           int32_t* lo_partkey, int32_t* lo_orderdate,
+                           int32_t* lo_revenue, int32_t* lo_supplycost);
+uint32_t orc_cust_region(uint64_t seed, uint32_t custkey);
+uint32_t orc_supp_nation(uint64_t seed, uint32_t suppkey);
+uint32_t orc_supp_city_in_nation(uint64_t seed, uint32_t suppkey);
+uint32_t orc_part_brand_in_category(uint64_t seed, uint32_t partkey);
+void orc_build_cust_dim_q43(uint64_t seed, uint32_t n_custs, int32_t region, uint32_t* first);
+void orc_build_supp_dim_q43(uint64_t seed, uint32_t n_supps, int32_t nation, uint32_t* first);
+void orc_build_part_dim_q43(uint64_t seed, uint32_t n_parts, int32_t category, uint32_t* first);
+void orc_q43_kernel(const int32_t* ck, const int32_t* sk, const int32_t* pk,
+                    const int32_t* od, const int32_t* rv, const int32_t* sc,
+                    uint64_t n_rows, const uint32_t* cfirst, const uint32_t* sfirst,
+                    const uint32_t* pfirst, const uint32_t* dfirst, int64_t dmin,
+                    int threads, int64_t* group_sums /*800*/);
+void orc_q43_pipeline(uint64_t seed, uint64_t row_start, uint64_t n_rows,
+                      int32_t region, int32_t nation, int32_t category,
+                      int threads, int64_t* group_sums /*800*/);
+
 /* compute-only legs timed by bench.py's cpu_baseline (columns pre-generated) */
 int64_t orc_q1_kernel(const int32_t* od, const int32_t* ep, const int32_t* dc,
                       uint64_t n_rows, const uint32_t* dfirst, int64_t mn, int64_t mx,

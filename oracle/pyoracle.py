@@ -84,6 +84,12 @@ def _decl(lib):
     lib.orc_q21_pipeline.argtypes = [c_u64, c_u64, c_u64, c_i32, c_i32, c_i32, c_vp]
     lib.orc_gen_lineorder_q1.argtypes = [c_u64, c_u64, c_u64, c_vp, c_vp, c_vp]
     lib.orc_gen_lineorder_q21.argtypes = [c_u64, c_u64, c_u64, c_vp, c_vp, c_vp, c_vp]
+    lib.orc_gen_lineorder_q43.argtypes = [c_u64, c_u64, c_u64] + [c_vp] * 6
+    lib.orc_build_cust_dim_q43.argtypes = [c_u64, u, c_i32, c_vp]
+    lib.orc_build_supp_dim_q43.argtypes = [c_u64, u, c_i32, c_vp]
+    lib.orc_build_part_dim_q43.argtypes = [c_u64, u, c_i32, c_vp]
+    lib.orc_q43_kernel.argtypes = [c_vp] * 6 + [c_u64] + [c_vp] * 4 + [c_i64, c_i32, c_vp]
+    lib.orc_q43_pipeline.argtypes = [c_u64, c_u64, c_u64, c_i32, c_i32, c_i32, c_i32, c_vp]
     lib.orc_q1_kernel.restype = c_i64
     lib.orc_q1_kernel.argtypes = [c_vp, c_vp, c_vp, c_u64, c_vp, c_i64, c_i64, c_i32,
                                   ctypes.POINTER(c_u64)]
@@ -154,6 +160,27 @@ def q21_kernel(pk, sk, od, rv, pfirst, sfirst, dfirst, dmin, threads=0):
     out = np.zeros(7000, np.int64)
     load().orc_q21_kernel(_p(pk), _p(sk), _p(od), _p(rv), len(pk), _p(pfirst),
                           _p(sfirst), _p(dfirst), dmin, threads, _p(out))
+    return out
+
+
+def gen_lineorder_q43(seed, row_start, n):
+    cols = [np.empty(n, np.int32) for _ in range(6)]
+    load().orc_gen_lineorder_q43(seed, row_start, n, *[_p(c) for c in cols])
+    return cols  # ck, sk, pk, od, rv, sc
+
+
+def q43_pipeline(seed, row_start, n_rows, region=1, nation=7, category=12, threads=0):
+    out = np.zeros(800, np.int64)
+    load().orc_q43_pipeline(seed, row_start, n_rows, region, nation, category,
+                            threads, _p(out))
+    return out
+
+
+def q43_kernel(ck, sk, pk, od, rv, sc, cfirst, sfirst, pfirst, dfirst, dmin, threads=0):
+    out = np.zeros(800, np.int64)
+    load().orc_q43_kernel(_p(ck), _p(sk), _p(pk), _p(od), _p(rv), _p(sc), len(ck),
+                          _p(cfirst), _p(sfirst), _p(pfirst), _p(dfirst), dmin,
+                          threads, _p(out))
     return out
 
 

@@ -68,6 +68,7 @@ struct gpue_dbuf {
     gpue_session* s;
     void* ptr;
     uint64_t bytes;
+    bool owned = true; // false: wraps external device memory (e.g. a torch tensor)
 };
 
 struct gpue_join_table {
@@ -135,15 +136,33 @@ int gpue_dbuf_alloc(gpue_session* s, uint64_t bytes, gpue_dbuf** out) {
     HIP_CHECK(hipSetDevice(s->device));
     void* p = nullptr;
     HIP_CHECK(hipMalloc(&p, bytes));
-    gpue_dbuf* b = new gpue_dbuf{s, p, bytes};
+    gpue_dbuf* b = new gpue_dbuf{s, p, bytes, true};
     *out = b;
     return GPUE_OK;
 }
 
 void gpue_dbuf_free(gpue_dbuf* b) {
     if (!b) return;
-    (void)hipFree(b->ptr);
+    if (b->owned) (void)hipFree(b->ptr);
     delete b;
+}
+
+// Wrap external device memory (a torch tensor's data_ptr) as a gpue_dbuf so
+// kernels operate in place and torch.distributed (RCCL) moves the same
+// buffers — the all-to-all leg of configs 4-5.
+int gpue_dbuf_wrap(gpue_session* s, void* device_ptr, uint64_t bytes, gpue_dbuf** out) {
+    ARG_CHECK(s && device_ptr && out && bytes > 0);
+    gpue_dbuf* b = new gpue_dbuf{s, device_ptr, bytes, false};
+    *out = b;
+    return GPUE_OK;
+}
+
+// expose the device pointer (for torch interop / sub-buffer views)
+extern "C" int gpue_dbuf_ptr(gpue_dbuf* b, void** out);
+int gpue_dbuf_ptr(gpue_dbuf* b, void** out) {
+    ARG_CHECK(b && out);
+    *out = b->ptr;
+    return GPUE_OK;
 }
 
 int gpue_dbuf_h2d(gpue_dbuf* b, const void* src, uint64_t bytes, uint64_t dst_off) {
@@ -186,10 +205,13 @@ __host__ __device__ static inline uint64_t gen_u64(uint64_t seed, uint64_t tag, 
 // tags shared with oracle/oracle.c
 enum { TAG_ORDERDATE = 1, TAG_EXTPRICE = 2, TAG_DISCOUNT = 3,
        TAG_PARTKEY = 4, TAG_SUPPKEY = 5, TAG_REVENUE = 6,
-       TAG_PCAT = 7, TAG_PBRD = 8, TAG_SREG = 9 };
+       TAG_PCAT = 7, TAG_PBRD = 8, TAG_SREG = 9,
+       TAG_CUSTKEY = 10, TAG_SUPPCOST = 11, TAG_CREG = 12,
+       TAG_SNAT = 13, TAG_SCITY = 14 };
 static constexpr int N_DAYS = 2556;
 static constexpr uint32_t N_PARTS_SF100 = 1400000u;
 static constexpr uint32_t N_SUPPS_SF100 = 200000u;
+static constexpr uint32_t N_CUSTS_SF100 = 3000000u;
 
 __global__ void k_gen_u32_mod(uint32_t* out, uint64_t seed, uint64_t tag,
                               uint64_t row_start, uint64_t n, uint32_t mod, uint32_t add) {
@@ -229,6 +251,22 @@ __global__ void k_gen_lineorder_q21(int32_t* pk, int32_t* sk, int32_t* od, int32
         sk[i] = (int32_t)(gen_u64(seed, TAG_SUPPKEY, r) % N_SUPPS_SF100) + 1;
         od[i] = datekey[gen_u64(seed, TAG_ORDERDATE, r) % N_DAYS];
         rv[i] = (int32_t)(gen_u64(seed, TAG_REVENUE, r) % 10000000u);
+    }
+}
+
+__global__ void k_gen_lineorder_q43(int32_t* ck, int32_t* sk, int32_t* pk, int32_t* od,
+                                    int32_t* rv, int32_t* sc,
+                                    const int32_t* __restrict__ datekey,
+                                    uint64_t seed, uint64_t row_start, uint64_t n) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint64_t r = row_start + i;
+        ck[i] = (int32_t)(gen_u64(seed, TAG_CUSTKEY, r) % N_CUSTS_SF100) + 1;
+        sk[i] = (int32_t)(gen_u64(seed, TAG_SUPPKEY, r) % N_SUPPS_SF100) + 1;
+        pk[i] = (int32_t)(gen_u64(seed, TAG_PARTKEY, r) % N_PARTS_SF100) + 1;
+        od[i] = datekey[gen_u64(seed, TAG_ORDERDATE, r) % N_DAYS];
+        rv[i] = (int32_t)(gen_u64(seed, TAG_REVENUE, r) % 10000000u);
+        sc[i] = (int32_t)(gen_u64(seed, TAG_SUPPCOST, r) % 100000u) + 1;
     }
 }
 
@@ -279,6 +317,25 @@ int gpue_gen_lineorder_q1(gpue_session* s, uint64_t seed, uint64_t row_start, ui
     if (rc != GPUE_OK) return rc;
     hipLaunchKernelGGL(k_gen_lineorder_q1, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
                        (int32_t*)od->ptr, (int32_t*)ep->ptr, (int32_t*)dc->ptr,
+                       s->d_datekey, seed, row_start, n);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
+extern "C" int gpue_gen_lineorder_q43(gpue_session* s, uint64_t seed, uint64_t row_start,
+                                      uint64_t n, gpue_dbuf* ck, gpue_dbuf* sk, gpue_dbuf* pk,
+                                      gpue_dbuf* od, gpue_dbuf* rv, gpue_dbuf* sc);
+int gpue_gen_lineorder_q43(gpue_session* s, uint64_t seed, uint64_t row_start, uint64_t n,
+                           gpue_dbuf* ck, gpue_dbuf* sk, gpue_dbuf* pk, gpue_dbuf* od,
+                           gpue_dbuf* rv, gpue_dbuf* sc) {
+    ARG_CHECK(s && ck && sk && pk && od && rv && sc);
+    ARG_CHECK(ck->bytes >= n * 4 && sk->bytes >= n * 4 && pk->bytes >= n * 4 &&
+              od->bytes >= n * 4 && rv->bytes >= n * 4 && sc->bytes >= n * 4);
+    int rc = ensure_datekey(s);
+    if (rc != GPUE_OK) return rc;
+    hipLaunchKernelGGL(k_gen_lineorder_q43, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+                       (int32_t*)ck->ptr, (int32_t*)sk->ptr, (int32_t*)pk->ptr,
+                       (int32_t*)od->ptr, (int32_t*)rv->ptr, (int32_t*)sc->ptr,
                        s->d_datekey, seed, row_start, n);
     HIP_CHECK(hipGetLastError());
     return GPUE_OK;
@@ -1051,6 +1108,125 @@ int gpue_q21_star_agg(gpue_session* s, gpue_join_table* parts, gpue_join_table* 
 }
 
 // ---------------------------------------------------------------------------
+// Config 4 — SSB Q4.3: 4-way star join (customer ⋈ supplier ⋈ part ⋈ date)
+// + GROUP BY (d_year, s_city, p_brand) as compact filtered indexes:
+// gid = (dpay-1)*400 + (spay-1)*40 + (ppay-1), 2×10×40 = 800 groups
+// (SURVEY.md §8a cfg 4 ≈ 800 groups). SUM(lo_revenue − lo_supplycost).
+// Probe order by rejection rate × table cost: part bitset (1/25, 175 KB L2)
+// batched for every row; survivors probe supplier bitset (1/25, 25 KB),
+// customer bitset (1/5, 375 KB), date payload (2/7).
+// ---------------------------------------------------------------------------
+static constexpr int NG_Q43 = 800;
+
+__global__ __launch_bounds__(BLOCK_Q21) void
+k_q43_star_agg(const int32_t* __restrict__ ck, const int32_t* __restrict__ sk,
+               const int32_t* __restrict__ pk, const int32_t* __restrict__ od,
+               const int32_t* __restrict__ rv, const int32_t* __restrict__ sc,
+               uint64_t n,
+               const uint32_t* __restrict__ cbits, int64_t csmin, uint64_t csint,
+               const uint32_t* __restrict__ sbits, int64_t ssmin, uint64_t ssint,
+               const uint16_t* __restrict__ sfirst,
+               const uint32_t* __restrict__ pbits, int64_t psmin, uint64_t psint,
+               const uint16_t* __restrict__ pfirst,
+               const uint16_t* __restrict__ dfirst, int64_t dmin,
+               unsigned long long* __restrict__ group_sums) {
+    __shared__ unsigned long long g[NG_Q43];
+    for (int j = threadIdx.x; j < NG_Q43; j += blockDim.x) g[j] = 0;
+    __syncthreads();
+    const uint64_t n4 = n / 4;
+    const int4* __restrict__ ck4 = (const int4*)ck;
+    const int4* __restrict__ sk4 = (const int4*)sk;
+    const int4* __restrict__ pk4 = (const int4*)pk;
+    const int4* __restrict__ od4 = (const int4*)od;
+    const int4* __restrict__ rv4 = (const int4*)rv;
+    const int4* __restrict__ sc4 = (const int4*)sc;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    auto quad = [&](int4 c4, int4 s4, int4 p4, int4 o4, int4 r4, int4 x4) {
+        uint32_t pb[4], pin[4];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint32_t idx = (uint32_t)((&p4.x)[j] - psmin);
+            pin[j] = idx < psint;
+            uint32_t cidx = pin[j] ? idx : 0u;
+            pb[j] = pbits[cidx >> 5] >> (cidx & 31);
+        }
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            if (!(pin[j] & pb[j] & 1u)) continue;
+            uint32_t sidx = (uint32_t)((&s4.x)[j] - ssmin);
+            if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
+            uint32_t cidx = (uint32_t)((&c4.x)[j] - csmin);
+            if (cidx >= csint || !((cbits[cidx >> 5] >> (cidx & 31)) & 1u)) continue;
+            uint32_t dpay = dfirst[(&o4.x)[j] - dmin];
+            if (dpay == 0) continue;
+            uint32_t ppay = pfirst[(&p4.x)[j] - 1];
+            uint32_t spay = sfirst[(&s4.x)[j] - 1];
+            atomicAdd(&g[(dpay - 1) * 400 + (spay - 1) * 40 + (ppay - 1)],
+                      (unsigned long long)((int64_t)(&r4.x)[j] - (&x4.x)[j]));
+        }
+    };
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i + stride < n4; i += 2 * stride) {
+        int4 ca = ck4[i], sa = sk4[i], pa = pk4[i], oa = od4[i], ra = rv4[i], xa = sc4[i];
+        uint64_t i2 = i + stride;
+        int4 cb = ck4[i2], sb = sk4[i2], pb_ = pk4[i2], ob = od4[i2], rb = rv4[i2], xb = sc4[i2];
+        quad(ca, sa, pa, oa, ra, xa);
+        quad(cb, sb, pb_, ob, rb, xb);
+    }
+    for (; i < n4; i += stride) quad(ck4[i], sk4[i], pk4[i], od4[i], rv4[i], sc4[i]);
+    // scalar tail
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (uint64_t r = n4 * 4 + tid; r < n; r += stride) {
+        uint32_t pidx = (uint32_t)(pk[r] - psmin);
+        if (pidx >= psint || !((pbits[pidx >> 5] >> (pidx & 31)) & 1u)) continue;
+        uint32_t sidx = (uint32_t)(sk[r] - ssmin);
+        if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
+        uint32_t cidx = (uint32_t)(ck[r] - csmin);
+        if (cidx >= csint || !((cbits[cidx >> 5] >> (cidx & 31)) & 1u)) continue;
+        uint32_t dpay = dfirst[od[r] - dmin];
+        if (dpay == 0) continue;
+        uint32_t ppay = pfirst[pk[r] - 1];
+        uint32_t spay = sfirst[sk[r] - 1];
+        atomicAdd(&g[(dpay - 1) * 400 + (spay - 1) * 40 + (ppay - 1)],
+                  (unsigned long long)((int64_t)rv[r] - sc[r]));
+    }
+    __syncthreads();
+    for (int j = threadIdx.x; j < NG_Q43; j += blockDim.x)
+        if (g[j] != 0) atomicAdd(&group_sums[j], g[j]);
+}
+
+extern "C" int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs,
+                                       gpue_join_table* supps, gpue_join_table* parts,
+                                       gpue_join_table* dates, gpue_dbuf* ck, gpue_dbuf* sk,
+                                       gpue_dbuf* pk, gpue_dbuf* od, gpue_dbuf* rv,
+                                       gpue_dbuf* sc, uint64_t n, gpue_dbuf* group_sums);
+int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs, gpue_join_table* supps,
+                            gpue_join_table* parts, gpue_join_table* dates, gpue_dbuf* ck,
+                            gpue_dbuf* sk, gpue_dbuf* pk, gpue_dbuf* od, gpue_dbuf* rv,
+                            gpue_dbuf* sc, uint64_t n, gpue_dbuf* group_sums) {
+    ARG_CHECK(s && custs && supps && parts && dates && ck && sk && pk && od && rv && sc);
+    ARG_CHECK(group_sums && group_sums->bytes >= NG_Q43 * sizeof(int64_t));
+    ARG_CHECK(supps->first16 && parts->first16 && dates->first16);
+    ARG_CHECK(custs->bitset && supps->bitset && parts->bitset);
+    ARG_CHECK(custs->min_key == 1 && supps->min_key == 1 && parts->min_key == 1);
+    HIP_CHECK(hipMemsetAsync(group_sums->ptr, 0, NG_Q43 * sizeof(int64_t), s->stream));
+    hipLaunchKernelGGL(k_q43_star_agg, dim3(512), dim3(BLOCK_Q21), 0, s->stream,
+                       (const int32_t*)ck->ptr, (const int32_t*)sk->ptr,
+                       (const int32_t*)pk->ptr, (const int32_t*)od->ptr,
+                       (const int32_t*)rv->ptr, (const int32_t*)sc->ptr, n,
+                       custs->bitset, custs->set_min,
+                       (uint64_t)(custs->set_max - custs->set_min + 1),
+                       supps->bitset, supps->set_min,
+                       (uint64_t)(supps->set_max - supps->set_min + 1), supps->first16,
+                       parts->bitset, parts->set_min,
+                       (uint64_t)(parts->set_max - parts->set_min + 1), parts->first16,
+                       dates->first16, dates->min_key,
+                       (unsigned long long*)group_sums->ptr);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
+// ---------------------------------------------------------------------------
 // exchange partition (reference exchange_sink_operator.cpp:611-660 semantics:
 // FNV row hash -> ReduceOp channel -> counting-sort layout). Channel
 // ASSIGNMENT is bit-identical to the reference; within a channel, rows are
@@ -1206,6 +1382,32 @@ int gpue_ubench(gpue_session* s, int which, gpue_dbuf* a, gpue_dbuf* b, gpue_dbu
     HIP_CHECK(hipEventElapsedTime(&ms, s->ev_start, s->ev_stop));
     (void)hipFree(d_out);
     *ms_out = ms / reps;
+    return GPUE_OK;
+}
+
+// ---------------------------------------------------------------------------
+// row gather by index — the exchange sink's add_rows_selective analog
+// (reference exchange_sink_operator.cpp:670: per-channel row slices are
+// materialized by gathering source rows at the counting-sorted indexes)
+// ---------------------------------------------------------------------------
+__global__ void k_gather_u32(const uint32_t* __restrict__ in,
+                             const uint32_t* __restrict__ idx, uint64_t n,
+                             uint32_t* __restrict__ out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+        out[i] = in[idx[i]];
+}
+
+extern "C" int gpue_gather_u32(gpue_session* s, gpue_dbuf* in, gpue_dbuf* idx, uint64_t n,
+                               gpue_dbuf* out);
+int gpue_gather_u32(gpue_session* s, gpue_dbuf* in, gpue_dbuf* idx, uint64_t n,
+                    gpue_dbuf* out) {
+    ARG_CHECK(s && in && idx && out);
+    ARG_CHECK(idx->bytes >= n * 4 && out->bytes >= n * 4);
+    hipLaunchKernelGGL(k_gather_u32, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+                       (const uint32_t*)in->ptr, (const uint32_t*)idx->ptr, n,
+                       (uint32_t*)out->ptr);
+    HIP_CHECK(hipGetLastError());
     return GPUE_OK;
 }
 

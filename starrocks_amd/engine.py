@@ -67,6 +67,7 @@ def _declare(lib):
         "gpue_gen_i64": (c_i32, [c_vp, c_vp, c_u64, c_u64, c_u64, c_u64]),
         "gpue_gen_lineorder_q1": (c_i32, [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp, c_vp]),
         "gpue_gen_lineorder_q21": (c_i32, [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp, c_vp, c_vp]),
+        "gpue_gen_lineorder_q43": (c_i32, [c_vp, c_u64, c_u64, c_u64] + [c_vp] * 6),
         "gpue_scan_filter_i64_lt": (c_i32, [c_vp, c_vp, c_u64, c_i64, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_join_build_payload_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_build_range_direct_i32": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
@@ -78,8 +79,12 @@ def _declare(lib):
                                      ctypes.POINTER(c_i64), ctypes.POINTER(c_u64)]),
         "gpue_q21_star_agg": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_q1_join_sum_async": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
+        "gpue_dbuf_wrap": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
+        "gpue_dbuf_ptr": (c_i32, [c_vp, ctypes.POINTER(c_vp)]),
+        "gpue_q43_star_agg_async": (c_i32, [c_vp] * 11 + [c_u64, c_vp]),
         "gpue_q21_star_agg_async": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_partition_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
+        "gpue_gather_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_timer_start": (c_i32, [c_vp]),
         "gpue_timer_stop": (c_i32, [c_vp, ctypes.POINTER(ctypes.c_float)]),
     }
@@ -186,6 +191,11 @@ class Engine:
         _ck(self._lib, self._lib.gpue_gen_lineorder_q21(self._h, seed, row_start, n,
                                                         pk._h, sk._h, od._h, rv._h))
 
+    def gen_lineorder_q43(self, seed, row_start, n, ck, sk, pk, od, rv, sc):
+        _ck(self._lib, self._lib.gpue_gen_lineorder_q43(self._h, seed, row_start, n,
+                                                        ck._h, sk._h, pk._h, od._h,
+                                                        rv._h, sc._h))
+
     # ---- operators ----
     def scan_filter_i64_lt(self, inp: DBuf, n, theta, out: DBuf) -> int:
         cnt = c_u64()
@@ -220,6 +230,32 @@ class Engine:
                                                   ctypes.byref(s), ctypes.byref(c)))
         return s.value, c.value
 
+    def dbuf_ptr(self, b: DBuf) -> int:
+        p = c_vp()
+        _ck(self._lib, self._lib.gpue_dbuf_ptr(b._h, ctypes.byref(p)))
+        return p.value
+
+    def wrap_ptr_offset(self, b: DBuf, offset: int, nbytes: int) -> DBuf:
+        """Sub-buffer view (per-channel slice of a partitioned column)."""
+        return self.wrap_ptr(self.dbuf_ptr(b) + offset, nbytes)
+
+    def wrap_ptr(self, device_ptr: int, nbytes: int) -> DBuf:
+        """Wrap external device memory (e.g. torch_tensor.data_ptr())."""
+        b = DBuf.__new__(DBuf)
+        b._lib = self._lib
+        b.nbytes = nbytes
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_dbuf_wrap(self._h, c_vp(device_ptr), nbytes,
+                                                ctypes.byref(h)))
+        b._h = h
+        return b
+
+    def q43_star_agg_async(self, custs, supps, parts, dates, ck, sk, pk, od, rv, sc,
+                           n, group_sums: DBuf):
+        _ck(self._lib, self._lib.gpue_q43_star_agg_async(
+            self._h, custs._h, supps._h, parts._h, dates._h, ck._h, sk._h, pk._h,
+            od._h, rv._h, sc._h, n, group_sums._h))
+
     def q1_join_sum_async(self, dates: JoinTable, od: DBuf, ep: DBuf, dc: DBuf, n, acc: DBuf):
         _ck(self._lib, self._lib.gpue_q1_join_sum_async(self._h, dates._h, od._h, ep._h,
                                                         dc._h, n, acc._h))
@@ -236,6 +272,9 @@ class Engine:
                                                    pk._h, sk._h, od._h, rv._h, n,
                                                    out.ctypes.data_as(c_vp)))
         return out
+
+    def gather_u32(self, inp: DBuf, idx: DBuf, n, out: DBuf):
+        _ck(self._lib, self._lib.gpue_gather_u32(self._h, inp._h, idx._h, n, out._h))
 
     def partition(self, keys: DBuf, n, num_channels, row_indexes: DBuf) -> np.ndarray:
         sp = np.zeros(num_channels + 1, dtype=np.uint64)

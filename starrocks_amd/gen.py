@@ -26,10 +26,16 @@ TAG_REVENUE = 6
 TAG_PCAT = 7
 TAG_PBRD = 8
 TAG_SREG = 9
+TAG_CUSTKEY = 10
+TAG_SUPPCOST = 11
+TAG_CREG = 12
+TAG_SNAT = 13
+TAG_SCITY = 14
 
 N_DAYS = 2556
 N_PARTS_SF100 = 1_400_000
 N_SUPPS_SF100 = 200_000
+N_CUSTS_SF100 = 3_000_000
 SF10_LINEORDER_ROWS = 59_986_052
 SF100_LINEORDER_ROWS = 600_000_000
 
@@ -85,6 +91,66 @@ def gen_lineorder_q21(seed: int, row_start: int, n: int):
     od = datekey[(gen_u64(seed, TAG_ORDERDATE, idx) % np.uint64(N_DAYS)).astype(np.int64)]
     rv = (gen_u64(seed, TAG_REVENUE, idx) % np.uint64(10000000)).astype(np.int32)
     return pk, sk, od.astype(np.int32), rv
+
+
+def gen_lineorder_q43(seed: int, row_start: int, n: int):
+    idx = np.arange(row_start, row_start + n, dtype=np.uint64)
+    datekey, _ = gen_dates()
+    ck = (gen_u64(seed, TAG_CUSTKEY, idx) % np.uint64(N_CUSTS_SF100)).astype(np.int32) + 1
+    sk = (gen_u64(seed, TAG_SUPPKEY, idx) % np.uint64(N_SUPPS_SF100)).astype(np.int32) + 1
+    pk = (gen_u64(seed, TAG_PARTKEY, idx) % np.uint64(N_PARTS_SF100)).astype(np.int32) + 1
+    od = datekey[(gen_u64(seed, TAG_ORDERDATE, idx) % np.uint64(N_DAYS)).astype(np.int64)]
+    rv = (gen_u64(seed, TAG_REVENUE, idx) % np.uint64(10000000)).astype(np.int32)
+    sc = (gen_u64(seed, TAG_SUPPCOST, idx) % np.uint64(100000)).astype(np.int32) + 1
+    return ck, sk, pk, od.astype(np.int32), rv, sc
+
+
+def build_cust_dim_q43(seed: int, n_custs: int, region: int) -> np.ndarray:
+    keys = np.arange(1, n_custs + 1, dtype=np.uint64)
+    reg = (gen_u64(seed, TAG_CREG, keys) % np.uint64(5)).astype(np.uint32)
+    return (reg == region).astype(np.uint32)
+
+
+def build_supp_dim_q43(seed: int, n_supps: int, nation: int) -> np.ndarray:
+    keys = np.arange(1, n_supps + 1, dtype=np.uint64)
+    nat = (gen_u64(seed, TAG_SNAT, keys) % np.uint64(25)).astype(np.uint32)
+    city = (gen_u64(seed, TAG_SCITY, keys) % np.uint64(10)).astype(np.uint32)
+    return np.where(nat == nation, city + 1, 0).astype(np.uint32)
+
+
+def build_part_dim_q43(seed: int, n_parts: int, category: int) -> np.ndarray:
+    keys = np.arange(1, n_parts + 1, dtype=np.uint64)
+    cat = part_category(seed, keys)
+    brand_in_cat = (gen_u64(seed, TAG_PBRD, keys) % np.uint64(40)).astype(np.uint32)
+    return np.where(cat == category, brand_in_cat + 1, 0).astype(np.uint32)
+
+
+def build_date_dim_q43() -> tuple:
+    datekey, dyear = gen_dates()
+    mn, mx = int(datekey[0]), int(datekey[-1])
+    first = np.zeros(mx - mn + 1, dtype=np.uint32)
+    first[datekey[dyear == 1997] - mn] = 1
+    first[datekey[dyear == 1998] - mn] = 2
+    return mn, mx, first
+
+
+def fnv_u32(keys: np.ndarray) -> np.ndarray:
+    """HashUtil::fnv_hash over the 4 LE bytes of each uint32 key (reference
+    hash_util.hpp:133-143), vectorized; identical to oracle orc_fnv_hash and
+    the HIP fnv_u32."""
+    with np.errstate(over="ignore"):
+        h = np.full(keys.shape, 0x811C9DC5, np.uint32)
+        k = keys.astype(np.uint32)
+        for b in range(4):
+            byte = ((k >> np.uint32(8 * b)) & np.uint32(0xFF))
+            h = (byte ^ h) * np.uint32(16777619)
+        return h
+
+
+def partition_channels(keys: np.ndarray, num_channels: int) -> np.ndarray:
+    """ReduceOp channel assignment (shuffler.h:71-86): (fnv(key)*n)>>32."""
+    return ((fnv_u32(keys).astype(np.uint64) * np.uint64(num_channels))
+            >> np.uint64(32)).astype(np.uint32)
 
 
 def part_category(seed: int, partkeys: np.ndarray) -> np.ndarray:

@@ -162,6 +162,107 @@ def test_q21_star_agg_parity(engine):
         t.destroy()
 
 
+def _build_q43_tables(engine):
+    R, N, C = 1, 7, 12
+    tabs = {}
+    for name, keys, pay in [
+        ("custs", np.arange(1, gen.N_CUSTS_SF100 + 1, dtype=np.int32),
+         gen.build_cust_dim_q43(SEED, gen.N_CUSTS_SF100, R)),
+        ("supps", np.arange(1, gen.N_SUPPS_SF100 + 1, dtype=np.int32),
+         gen.build_supp_dim_q43(SEED, gen.N_SUPPS_SF100, N)),
+        ("parts", np.arange(1, gen.N_PARTS_SF100 + 1, dtype=np.int32),
+         gen.build_part_dim_q43(SEED, gen.N_PARTS_SF100, C)),
+    ]:
+        kb = engine.alloc(keys.nbytes)
+        kb.h2d(keys)
+        pb = engine.alloc(pay.nbytes)
+        pb.h2d(pay)
+        tabs[name] = engine.join_build_payload(kb, pb, len(keys))
+        kb.free()
+        pb.free()
+    datekey, dyear = gen.gen_dates()
+    dpay = np.where(dyear == 1997, 1, np.where(dyear == 1998, 2, 0)).astype(np.uint32)
+    kb = engine.alloc(datekey.nbytes)
+    kb.h2d(datekey.astype(np.int32))
+    pb = engine.alloc(dpay.nbytes)
+    pb.h2d(dpay)
+    tabs["dates"] = engine.join_build_payload(kb, pb, len(datekey))
+    kb.free()
+    pb.free()
+    return tabs
+
+
+def test_q43_star_agg_parity(engine):
+    from starrocks_amd.engine import c_u64, c_vp, _ck
+    import ctypes
+    n = 8_000_000
+    cols = [engine.alloc(n * 4) for _ in range(6)]
+    engine.gen_lineorder_q43(SEED, 0, n, *cols)
+    tabs = _build_q43_tables(engine)
+    acc = engine.alloc(800 * 8)
+    engine.q43_star_agg_async(tabs["custs"], tabs["supps"], tabs["parts"],
+                              tabs["dates"], *cols, n, acc)
+    got = acc.d2h(np.int64, 800)
+    from oracle import pyoracle as orc
+    expect = orc.q43_pipeline(SEED, 0, n, 1, 7, 12)
+    assert np.array_equal(got, expect)  # bit-exact int64 group sums
+    for b in cols:
+        b.free()
+    acc.free()
+    for t in tabs.values():
+        t.destroy()
+
+
+def test_q43_partition_gather_roundtrip(engine):
+    """Single-GPU simulation of the N>1 partitioned pipeline (bench.py q43):
+    partition rows by lo_custkey, gather all 6 columns into channel order,
+    run the agg per channel slice against the channel's customer shard, merge
+    — must equal the unpartitioned result bit-exactly. This is the same math
+    the RCCL all-to-all path executes across 8 GPUs."""
+    n, world = 4_000_000, 4
+    cols = [engine.alloc(n * 4) for _ in range(6)]
+    engine.gen_lineorder_q43(SEED, 0, n, *cols)
+    tabs = _build_q43_tables(engine)
+
+    ridx = engine.alloc(n * 4)
+    sp = engine.partition(cols[0], n, world, ridx)
+    send = [engine.alloc(n * 4) for _ in range(6)]
+    for c, s in zip(cols, send):
+        engine.gather_u32(c, ridx, n, s)
+
+    ckeys = np.arange(1, gen.N_CUSTS_SF100 + 1, dtype=np.uint32)
+    cust_owner = gen.partition_channels(ckeys, world)
+    cpay_full = gen.build_cust_dim_q43(SEED, gen.N_CUSTS_SF100, 1)
+    merged = np.zeros(800, np.int64)
+    acc = engine.alloc(800 * 8)
+    for r in range(world):
+        lo, hi = int(sp[r]), int(sp[r + 1])
+        if hi == lo:
+            continue
+        cpay_r = np.where(cust_owner == r, cpay_full, 0).astype(np.uint32)
+        kb = engine.alloc(ckeys.nbytes)
+        kb.h2d(ckeys.view(np.int32))
+        pb = engine.alloc(cpay_r.nbytes)
+        pb.h2d(cpay_r)
+        custs_r = engine.join_build_payload(kb, pb, len(ckeys))
+        kb.free()
+        pb.free()
+        # per-channel slice: wrap offsets into the gathered buffers
+        slice_bufs = [engine.wrap_ptr_offset(s, lo * 4, (hi - lo) * 4) for s in send]
+        engine.q43_star_agg_async(custs_r, tabs["supps"], tabs["parts"], tabs["dates"],
+                                  *slice_bufs, hi - lo, acc)
+        merged += acc.d2h(np.int64, 800)
+        custs_r.destroy()
+
+    from oracle import pyoracle as orc
+    expect = orc.q43_pipeline(SEED, 0, n, 1, 7, 12)
+    assert np.array_equal(merged, expect)
+    for b in cols + send + [ridx, acc]:
+        b.free()
+    for t in tabs.values():
+        t.destroy()
+
+
 def test_partition_parity(engine):
     n, nch = 3_000_000, 8
     keys = engine.alloc(n * 4)
