@@ -170,6 +170,18 @@ int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs, gpue_join_t
 int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t num_channels,
                        uint64_t* start_points_out, gpue_dbuf* row_indexes_out);
 
+/* ---- generic hash aggregate ----
+ * Replaces AggHashMapWithKey::compute_agg_states + update_batch +
+ * convert_hash_map_to_chunk (reference be/src/exec/agg_hash_map.h:112-290,
+ * aggregator.cpp:937-959,1742-1816) for HIGH-cardinality GROUP BY (group-key
+ * packed <= 8 B, as the SERIALIZED_FIXED_SIZE key constructors pack). SUM +
+ * COUNT states; emission order is table order (results are a set). Key
+ * sentinel ~0ull must not occur in keys. */
+int gpue_hash_agg_sum_u64(gpue_session* s, gpue_dbuf* keys /*u64*/, gpue_dbuf* vals /*i64*/,
+                          uint64_t n, uint64_t capacity_hint, gpue_dbuf* out_keys,
+                          gpue_dbuf* out_sums, gpue_dbuf* out_counts /*nullable*/,
+                          uint64_t max_out, uint64_t* n_groups);
+
 /* Row gather by index — the exchange sink's add_rows_selective analog
  * (exchange_sink_operator.cpp:670): materializes per-channel row slices at
  * the counting-sorted indexes before the RCCL all-to-all. */

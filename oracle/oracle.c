@@ -674,6 +674,60 @@ void orc_q21_pipeline(uint64_t seed, uint64_t row_start, uint64_t n_rows,
     free(dfirst); free(pfirst); free(sfirst);
 }
 
+/* ====================================================================== */
+/* Generic hash aggregate — restates AggHashMapWithKey::compute_agg_states
+ * + AggregateFunction SUM/COUNT update_batch (be/src/exec/agg_hash_map.h:
+ * 112-290, be/src/exec/aggregator.cpp:937-959, be/src/exprs/agg/sum.h:45):
+ * per row lazy-emplace the group key, accumulate into its packed state.
+ * The reference's phmap internals (H2 fingerprints, SSE groups) affect only
+ * iteration order, which is not part of the result (output compared as a
+ * key-sorted set); restated here as open-addressing linear probing.        */
+/* ====================================================================== */
+
+#define AGG_EMPTY 0xFFFFFFFFFFFFFFFFull
+
+static inline uint32_t agg_hash_u64(uint64_t v, uint32_t mask) {
+    /* multiplicative 64-bit (same family as JoinKeyHash<T,8>) */
+    return (uint32_t)((v * 11400714819323198485ull) >> 32) & mask;
+}
+
+/* returns number of groups, or UINT64_MAX if out capacity exceeded */
+uint64_t orc_hash_agg_sum_u64(const uint64_t* keys, const int64_t* vals, uint64_t n,
+                              uint64_t* out_keys, int64_t* out_sums, int64_t* out_counts,
+                              uint64_t max_out) {
+    uint64_t cap = 16;
+    while (cap < n * 2) cap <<= 1;
+    uint64_t* slots = (uint64_t*)malloc(cap * sizeof(uint64_t));
+    int64_t* sums = (int64_t*)calloc(cap, sizeof(int64_t));
+    int64_t* counts = (int64_t*)calloc(cap, sizeof(int64_t));
+    memset(slots, 0xFF, cap * sizeof(uint64_t));
+    uint32_t mask = (uint32_t)(cap - 1);
+    for (uint64_t i = 0; i < n; i++) {
+        uint64_t k = keys[i];
+        uint32_t s = agg_hash_u64(k, mask);
+        for (;;) {
+            if (slots[s] == AGG_EMPTY) { slots[s] = k; }
+            if (slots[s] == k) {
+                sums[s] += vals[i];
+                counts[s]++;
+                break;
+            }
+            s = (s + 1) & mask;
+        }
+    }
+    uint64_t g = 0;
+    for (uint64_t s = 0; s < cap; s++) {
+        if (slots[s] == AGG_EMPTY) continue;
+        if (g >= max_out) { g = UINT64_MAX; break; }
+        out_keys[g] = slots[s];
+        out_sums[g] = sums[s];
+        if (out_counts) out_counts[g] = counts[s];
+        g++;
+    }
+    free(slots); free(sums); free(counts);
+    return g;
+}
+
 /* Compute-only pipeline legs (columns pre-generated by the caller) — these
  * are what bench.py's cpu_baseline TIMES, so data generation stays outside
  * the measured region on both CPU and GPU. */

@@ -150,6 +150,12 @@ void orc_q43_pipeline(uint64_t seed, uint64_t row_start, uint64_t n_rows,
                       int32_t region, int32_t nation, int32_t category,
                       int threads, int64_t* group_sums /*800*/);
 
+/* generic hash aggregate (agg_hash_map.h:112-290 restatement); returns group
+ * count or UINT64_MAX if max_out exceeded */
+uint64_t orc_hash_agg_sum_u64(const uint64_t* keys, const int64_t* vals, uint64_t n,
+                              uint64_t* out_keys, int64_t* out_sums, int64_t* out_counts,
+                              uint64_t max_out);
+
 /* compute-only legs timed by bench.py's cpu_baseline (columns pre-generated) */
 int64_t orc_q1_kernel(const int32_t* od, const int32_t* ep, const int32_t* dc,
                       uint64_t n_rows, const uint32_t* dfirst, int64_t mn, int64_t mx,

@@ -90,6 +90,8 @@ def _decl(lib):
     lib.orc_build_part_dim_q43.argtypes = [c_u64, u, c_i32, c_vp]
     lib.orc_q43_kernel.argtypes = [c_vp] * 6 + [c_u64] + [c_vp] * 4 + [c_i64, c_i32, c_vp]
     lib.orc_q43_pipeline.argtypes = [c_u64, c_u64, c_u64, c_i32, c_i32, c_i32, c_i32, c_vp]
+    lib.orc_hash_agg_sum_u64.restype = c_u64
+    lib.orc_hash_agg_sum_u64.argtypes = [c_vp, c_vp, c_u64, c_vp, c_vp, c_vp, c_u64]
     lib.orc_q1_kernel.restype = c_i64
     lib.orc_q1_kernel.argtypes = [c_vp, c_vp, c_vp, c_u64, c_vp, c_i64, c_i64, c_i32,
                                   ctypes.POINTER(c_u64)]
@@ -182,6 +184,16 @@ def q43_kernel(ck, sk, pk, od, rv, sc, cfirst, sfirst, pfirst, dfirst, dmin, thr
                           _p(cfirst), _p(sfirst), _p(pfirst), _p(dfirst), dmin,
                           threads, _p(out))
     return out
+
+
+def hash_agg_sum(keys: np.ndarray, vals: np.ndarray):
+    cap = max(len(keys), 16)
+    ok = np.empty(cap, np.uint64)
+    os_ = np.empty(cap, np.int64)
+    oc = np.empty(cap, np.int64)
+    g = load().orc_hash_agg_sum_u64(_p(keys), _p(vals), len(keys), _p(ok), _p(os_), _p(oc), cap)
+    assert g != 2**64 - 1
+    return ok[:g].copy(), os_[:g].copy(), oc[:g].copy()
 
 
 def range_direct_build(keys_1based: np.ndarray, min_value: int, max_value: int):

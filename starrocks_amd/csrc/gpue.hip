@@ -1386,6 +1386,115 @@ int gpue_ubench(gpue_session* s, int which, gpue_dbuf* a, gpue_dbuf* b, gpue_dbu
 }
 
 // ---------------------------------------------------------------------------
+// Generic hash aggregate — the AggHashMapWithKey::compute_agg_states +
+// AggregateFunction::update_batch chain (reference be/src/exec/agg_hash_map.h
+// :112-290, aggregator.cpp:937-959, exprs/agg/sum.h:45) for HIGH-cardinality
+// GROUP BY, where the dense-group LDS shortcut of the SSB kernels does not
+// apply. Open-addressing linear probing in HBM; group slots claimed with a
+// 64-bit atomicCAS; SUM/COUNT accumulate with atomicAdd (order-independent
+// => bit-exact int64). The CPU's lazy_emplace + prefetch pipeline maps to
+// one CAS-claim + atomic update per row. Key sentinel: ~0ull.
+// ---------------------------------------------------------------------------
+static constexpr unsigned long long AGG_EMPTY = 0xFFFFFFFFFFFFFFFFull;
+
+__global__ void k_hash_agg_sum(const uint64_t* __restrict__ keys,
+                               const int64_t* __restrict__ vals, uint64_t n,
+                               unsigned long long* __restrict__ slots,
+                               unsigned long long* __restrict__ sums,
+                               unsigned long long* __restrict__ counts,
+                               uint64_t cap_mask) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        unsigned long long k = keys[i];
+        unsigned long long v = (unsigned long long)vals[i];
+        uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        for (;;) {
+            unsigned long long cur = slots[s];
+            if (cur == k) {
+                atomicAdd(&sums[s], v);
+                atomicAdd(&counts[s], 1ull);
+                break;
+            }
+            if (cur == AGG_EMPTY) {
+                unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
+                if (old == AGG_EMPTY || old == k) {
+                    atomicAdd(&sums[s], v);
+                    atomicAdd(&counts[s], 1ull);
+                    break;
+                }
+            }
+            s = (s + 1) & cap_mask;
+        }
+    }
+}
+
+// convert_hash_map_to_chunk analog (aggregator.cpp:1742-1816): iterate the
+// table, emit keys + finalized states (emission order is table order —
+// results are a set, compared key-sorted)
+__global__ void k_hash_agg_emit(const unsigned long long* __restrict__ slots,
+                                const unsigned long long* __restrict__ sums,
+                                const unsigned long long* __restrict__ counts,
+                                uint64_t cap, unsigned long long* __restrict__ cursor,
+                                uint64_t max_out, uint64_t* __restrict__ out_keys,
+                                int64_t* __restrict__ out_sums,
+                                int64_t* __restrict__ out_counts) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t s = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; s < cap; s += stride) {
+        if (slots[s] == AGG_EMPTY) continue;
+        unsigned long long pos = atomicAdd(cursor, 1ull);
+        if (pos >= max_out) continue; // caller checks n_groups > max_out
+        out_keys[pos] = slots[s];
+        out_sums[pos] = (int64_t)sums[s];
+        if (out_counts) out_counts[pos] = (int64_t)counts[s];
+    }
+}
+
+extern "C" int gpue_hash_agg_sum_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals,
+                                     uint64_t n, uint64_t capacity_hint, gpue_dbuf* out_keys,
+                                     gpue_dbuf* out_sums, gpue_dbuf* out_counts,
+                                     uint64_t max_out, uint64_t* n_groups);
+int gpue_hash_agg_sum_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uint64_t n,
+                          uint64_t capacity_hint, gpue_dbuf* out_keys, gpue_dbuf* out_sums,
+                          gpue_dbuf* out_counts, uint64_t max_out, uint64_t* n_groups) {
+    ARG_CHECK(s && keys && vals && out_keys && out_sums && n_groups);
+    ARG_CHECK(keys->bytes >= n * 8 && vals->bytes >= n * 8);
+    uint64_t cap = 16;
+    uint64_t want = capacity_hint ? capacity_hint : n * 2;
+    while (cap < want) cap <<= 1;
+    unsigned long long *d_slots = nullptr, *d_sums = nullptr, *d_counts = nullptr,
+                       *d_cursor = nullptr;
+    HIP_CHECK(hipMalloc(&d_slots, cap * 8));
+    HIP_CHECK(hipMalloc(&d_sums, cap * 8));
+    HIP_CHECK(hipMalloc(&d_counts, cap * 8));
+    HIP_CHECK(hipMalloc(&d_cursor, 8));
+    HIP_CHECK(hipMemsetAsync(d_slots, 0xFF, cap * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_sums, 0, cap * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_counts, 0, cap * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_cursor, 0, 8, s->stream));
+    hipLaunchKernelGGL(k_hash_agg_sum, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+                       (const uint64_t*)keys->ptr, (const int64_t*)vals->ptr, n,
+                       d_slots, d_sums, d_counts, cap - 1);
+    hipLaunchKernelGGL(k_hash_agg_emit, dim3(grid_for(cap)), dim3(BLOCK), 0, s->stream,
+                       d_slots, d_sums, d_counts, cap, d_cursor, max_out,
+                       (uint64_t*)out_keys->ptr, (int64_t*)out_sums->ptr,
+                       out_counts ? (int64_t*)out_counts->ptr : nullptr);
+    unsigned long long groups = 0;
+    HIP_CHECK(hipMemcpyAsync(&groups, d_cursor, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_slots);
+    (void)hipFree(d_sums);
+    (void)hipFree(d_counts);
+    (void)hipFree(d_cursor);
+    *n_groups = groups;
+    if (groups > max_out) {
+        snprintf(g_err, sizeof(g_err), "hash_agg: %llu groups exceed max_out %llu",
+                 groups, (unsigned long long)max_out);
+        return GPUE_ERR_ARG;
+    }
+    return GPUE_OK;
+}
+
+// ---------------------------------------------------------------------------
 // row gather by index — the exchange sink's add_rows_selective analog
 // (reference exchange_sink_operator.cpp:670: per-channel row slices are
 // materialized by gathering source rows at the counting-sorted indexes)

@@ -85,6 +85,8 @@ def _declare(lib):
         "gpue_q21_star_agg_async": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_partition_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
         "gpue_gather_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
+        "gpue_hash_agg_sum_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_u64, c_vp, c_vp, c_vp,
+                                          c_u64, ctypes.POINTER(c_u64)]),
         "gpue_timer_start": (c_i32, [c_vp]),
         "gpue_timer_stop": (c_i32, [c_vp, ctypes.POINTER(ctypes.c_float)]),
     }
@@ -272,6 +274,15 @@ class Engine:
                                                    pk._h, sk._h, od._h, rv._h, n,
                                                    out.ctypes.data_as(c_vp)))
         return out
+
+    def hash_agg_sum_u64(self, keys: DBuf, vals: DBuf, n, out_keys: DBuf, out_sums: DBuf,
+                         out_counts: DBuf = None, max_out=0, capacity_hint=0):
+        g = c_u64()
+        oc = out_counts._h if out_counts else None
+        _ck(self._lib, self._lib.gpue_hash_agg_sum_u64(
+            self._h, keys._h, vals._h, n, capacity_hint, out_keys._h, out_sums._h, oc,
+            max_out or out_keys.nbytes // 8, ctypes.byref(g)))
+        return g.value
 
     def gather_u32(self, inp: DBuf, idx: DBuf, n, out: DBuf):
         _ck(self._lib, self._lib.gpue_gather_u32(self._h, inp._h, idx._h, n, out._h))

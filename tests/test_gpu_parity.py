@@ -263,6 +263,38 @@ def test_q43_partition_gather_roundtrip(engine):
         t.destroy()
 
 
+@pytest.mark.parametrize("nkeys,card", [(2_000_000, 100), (2_000_000, 500_000), (1_000_000, 2**62)])
+def test_hash_agg_parity(engine, nkeys, card):
+    """Generic hash aggregate vs oracle: key-sorted (keys, sums, counts)
+    must match bit-exactly (int64, order-independent)."""
+    rng = np.random.default_rng(17)
+    keys = (rng.integers(0, card, nkeys).astype(np.uint64)
+            if card <= 2**32 else
+            rng.integers(0, 2**62, nkeys, dtype=np.int64).astype(np.uint64))
+    vals = rng.integers(-10**9, 10**9, nkeys).astype(np.int64)
+    kb = engine.alloc(keys.nbytes)
+    kb.h2d(keys)
+    vb = engine.alloc(vals.nbytes)
+    vb.h2d(vals)
+    max_out = min(nkeys, 1_200_000)
+    ok = engine.alloc(max_out * 8)
+    os_ = engine.alloc(max_out * 8)
+    oc = engine.alloc(max_out * 8)
+    g = engine.hash_agg_sum_u64(kb, vb, nkeys, ok, os_, oc, max_out)
+    gk = ok.d2h(np.uint64, g)
+    gs = os_.d2h(np.int64, g)
+    gc = oc.d2h(np.int64, g)
+    ek, es, ec = orc.hash_agg_sum(keys, vals)
+    assert g == len(ek)
+    order = np.argsort(gk)
+    eorder = np.argsort(ek)
+    assert np.array_equal(gk[order], ek[eorder])
+    assert np.array_equal(gs[order], es[eorder])
+    assert np.array_equal(gc[order], ec[eorder])
+    for b in (kb, vb, ok, os_, oc):
+        b.free()
+
+
 def test_partition_parity(engine):
     n, nch = 3_000_000, 8
     keys = engine.alloc(n * 4)
