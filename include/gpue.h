@@ -100,6 +100,11 @@ typedef struct gpue_join_table gpue_join_table;
 int gpue_join_build_payload_i32(gpue_session* s, gpue_dbuf* keys /*i32*/,
                                 gpue_dbuf* payloads /*u32*/, uint64_t n_rows,
                                 gpue_join_table** out);
+/* BUCKET_CHAINED variant — the selector's generic fallback for non-dense
+ * keys (join_hash_map_method.hpp:37-120): multiplicative-hash buckets +
+ * first/next chains; probe compares build keys along the chain. */
+int gpue_join_build_bucket_chained_u32(gpue_session* s, gpue_dbuf* keys /*u32, 1-based*/,
+                                       uint64_t row_count, gpue_join_table** out);
 /* Row-index variant: first/next chain structure exactly as the reference
  * builds it (chain order under duplicate keys is scatter-order, which on GPU
  * is nondeterministic — the emitted match multiset is identical). */

@@ -88,6 +88,12 @@ struct gpue_join_table {
                                  // halves the random-gather footprint (L2 per XCD is 4 MiB)
     uint64_t bucket_size = 0;
     uint64_t row_count = 0;
+    // method discriminator — the GPU analog of JoinHashMapSelector's choice
+    // (reference join_hash_table.cpp:164-344)
+    enum Kind { PAYLOAD = 0, RANGE_DIRECT = 1, BUCKET_CHAINED = 2 } kind = PAYLOAD;
+    uint32_t log_bucket_size = 0;
+    uint32_t* build_keys = nullptr; // chained methods keep the build keys for the
+                                    // probe-side equality check (1-based, row 0 sentinel)
 };
 
 int gpue_device_count(int* out) {
@@ -639,6 +645,7 @@ int gpue_join_build_range_direct_i32(gpue_session* s, gpue_dbuf* keys, uint64_t 
     gpue_join_table* t = new gpue_join_table();
     t->s = s; t->min_key = mn; t->max_key = mx;
     t->bucket_size = interval; t->row_count = row_count;
+    t->kind = gpue_join_table::RANGE_DIRECT;
     HIP_CHECK(hipMalloc(&t->first, interval * sizeof(uint32_t)));
     HIP_CHECK(hipMalloc(&t->next, (row_count + 1) * sizeof(uint32_t)));
     HIP_CHECK(hipMemsetAsync(t->first, 0, interval * sizeof(uint32_t), s->stream));
@@ -656,6 +663,7 @@ void gpue_join_table_destroy(gpue_join_table* t) {
     if (t->next) (void)hipFree(t->next);
     if (t->bitset) (void)hipFree(t->bitset);
     if (t->first16) (void)hipFree(t->first16);
+    if (t->build_keys) (void)hipFree(t->build_keys);
     delete t;
 }
 
@@ -670,6 +678,108 @@ int gpue_join_table_first_d2h(gpue_join_table* t, uint32_t* dst, uint64_t n_entr
     ARG_CHECK(t && dst && n_entries <= t->bucket_size);
     HIP_CHECK(hipMemcpy(dst, t->first, n_entries * sizeof(uint32_t), hipMemcpyDeviceToHost));
     return GPUE_OK;
+}
+
+// ---------------------------------------------------------------------------
+// BUCKET_CHAINED — the selector's generic fallback (reference
+// join_hash_map_method.hpp:37-120): bucket = JoinKeyHash (multiplicative,
+// join_hash_map_helper.h:36-44, bit-identical to the oracle), scatter
+// next[i] = atomicExch(&first[b], i). Chain ORDER under shared buckets is
+// scatter-order (the CPU's is reverse insertion order) — the emitted match
+// multiset is identical and that is the result (SURVEY.md §7 hard part (a)).
+// ---------------------------------------------------------------------------
+__device__ static inline uint32_t join_hash_u32(uint32_t v, uint32_t num_log_buckets) {
+    v ^= v >> (32 - num_log_buckets);
+    return (v * 2654435761u) >> (32 - num_log_buckets);
+}
+
+__global__ void k_build_bucket_chained(const uint32_t* __restrict__ keys, uint64_t row_count,
+                                       uint32_t log_bucket_size, uint32_t* __restrict__ first,
+                                       uint32_t* __restrict__ next) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i <= row_count;
+         i += stride)
+        next[i] = atomicExch(&first[join_hash_u32(keys[i], log_bucket_size)], (uint32_t)i);
+}
+
+// calc_bucket_size (join_hash_map_helper.h:68-78): NormalizeCapacity(n+(n-1)/4)+1
+static uint32_t calc_bucket_size(uint32_t size) {
+    uint64_t expect = (uint64_t)size + (size - 1) / 4;
+    if (expect >= (1ull << 31)) return 1u << 31;
+    uint64_t norm = expect ? (~0ull >> __builtin_clzll(expect)) : 1;
+    return (uint32_t)(norm + 1);
+}
+
+extern "C" int gpue_join_build_bucket_chained_u32(gpue_session* s, gpue_dbuf* keys,
+                                                  uint64_t row_count, gpue_join_table** out);
+int gpue_join_build_bucket_chained_u32(gpue_session* s, gpue_dbuf* keys, uint64_t row_count,
+                                       gpue_join_table** out) {
+    ARG_CHECK(s && keys && out && row_count > 0 && row_count + 1 < (1ull << 31));
+    ARG_CHECK(keys->bytes >= (row_count + 1) * 4);
+    gpue_join_table* t = new gpue_join_table();
+    t->s = s;
+    t->kind = gpue_join_table::BUCKET_CHAINED;
+    t->row_count = row_count;
+    t->bucket_size = calc_bucket_size((uint32_t)(row_count + 1));
+    t->log_bucket_size = (uint32_t)__builtin_ctzll(t->bucket_size);
+    HIP_CHECK(hipMalloc(&t->first, t->bucket_size * sizeof(uint32_t)));
+    HIP_CHECK(hipMalloc(&t->next, (row_count + 1) * sizeof(uint32_t)));
+    HIP_CHECK(hipMalloc(&t->build_keys, (row_count + 1) * sizeof(uint32_t)));
+    HIP_CHECK(hipMemsetAsync(t->first, 0, t->bucket_size * sizeof(uint32_t), s->stream));
+    HIP_CHECK(hipMemsetAsync(t->next, 0, (row_count + 1) * sizeof(uint32_t), s->stream));
+    HIP_CHECK(hipMemcpyAsync(t->build_keys, keys->ptr, (row_count + 1) * sizeof(uint32_t),
+                             hipMemcpyDeviceToDevice, s->stream));
+    hipLaunchKernelGGL(k_build_bucket_chained, dim3(grid_for(row_count)), dim3(BLOCK), 0,
+                       s->stream, t->build_keys, row_count, t->log_bucket_size, t->first,
+                       t->next);
+    HIP_CHECK(hipGetLastError());
+    *out = t;
+    return GPUE_OK;
+}
+
+// probe with key-equality compare along the chain (lookup_init +
+// _probe_from_ht with AreKeysInChainIdentical == false)
+__global__ void k_probe_count_bc(const uint32_t* __restrict__ probe_keys, uint64_t n,
+                                 uint32_t log_bucket_size, const uint32_t* __restrict__ first,
+                                 const uint32_t* __restrict__ next,
+                                 const uint32_t* __restrict__ build_keys,
+                                 uint32_t* __restrict__ row_counts) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint32_t k = probe_keys[i];
+        uint32_t b = first[join_hash_u32(k, log_bucket_size)];
+        uint32_t c = 0;
+        while (b != 0) {
+            c += (build_keys[b] == k);
+            b = next[b];
+        }
+        row_counts[i] = c;
+    }
+}
+
+__global__ void k_probe_emit_bc(const uint32_t* __restrict__ probe_keys, uint64_t n,
+                                uint32_t log_bucket_size, const uint32_t* __restrict__ first,
+                                const uint32_t* __restrict__ next,
+                                const uint32_t* __restrict__ build_keys,
+                                const uint32_t* __restrict__ row_counts,
+                                const uint64_t* __restrict__ row_offsets,
+                                uint32_t* __restrict__ out_probe,
+                                uint32_t* __restrict__ out_build) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        if (row_counts[i] == 0) continue;
+        uint32_t k = probe_keys[i];
+        uint64_t pos = row_offsets[i];
+        uint32_t b = first[join_hash_u32(k, log_bucket_size)];
+        while (b != 0) {
+            if (build_keys[b] == k) {
+                out_probe[pos] = (uint32_t)i;
+                out_build[pos] = b;
+                pos++;
+            }
+            b = next[b];
+        }
+    }
 }
 
 // ---------------------------------------------------------------------------
@@ -768,6 +878,11 @@ int gpue_join_probe_emit_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* pro
     uint64_t* d_offsets = nullptr;
     HIP_CHECK(hipMalloc(&d_counts, n_rows * sizeof(uint32_t)));
     HIP_CHECK(hipMalloc(&d_bsums, (nb + 1) * sizeof(uint64_t)));
+    if (t->kind == gpue_join_table::BUCKET_CHAINED) {
+        hipLaunchKernelGGL(k_probe_count_bc, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const uint32_t*)probe_keys->ptr, n_rows, t->log_bucket_size,
+                           t->first, t->next, t->build_keys, d_counts);
+    } else
     hipLaunchKernelGGL(k_probe_count_rd, dim3(nb), dim3(BLOCK), 0, s->stream,
                        (const int32_t*)probe_keys->ptr, n_rows, t->min_key, t->max_key,
                        t->first, t->next, d_counts);
@@ -784,6 +899,12 @@ int gpue_join_probe_emit_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* pro
         HIP_CHECK(hipMalloc(&d_offsets, n_rows * sizeof(uint64_t)));
         hipLaunchKernelGGL(k_scan_offsets, dim3(nb), dim3(BLOCK), 0, s->stream,
                            d_counts, n_rows, tile, d_bsums, d_offsets);
+        if (t->kind == gpue_join_table::BUCKET_CHAINED) {
+            hipLaunchKernelGGL(k_probe_emit_bc, dim3(nb), dim3(BLOCK), 0, s->stream,
+                               (const uint32_t*)probe_keys->ptr, n_rows, t->log_bucket_size,
+                               t->first, t->next, t->build_keys, d_counts, d_offsets,
+                               (uint32_t*)out_probe_idx->ptr, (uint32_t*)out_build_idx->ptr);
+        } else
         hipLaunchKernelGGL(k_probe_emit_rd, dim3(nb), dim3(BLOCK), 0, s->stream,
                            (const int32_t*)probe_keys->ptr, n_rows, t->min_key, t->first,
                            t->next, d_counts, d_offsets,
