@@ -33,7 +33,10 @@ SF100_ROWS = 600_000_000     # configs 3-4
 Q1_YEAR = 1993
 Q21_CATEGORY, Q21_REGION = 12, 2
 Q43_REGION, Q43_NATION, Q43_CATEGORY = 1, 7, 12
-BYTES_PER_ROW = {"q1": 12, "q21": 16, "q43": 24}  # algorithmic (§8d)
+BYTES_PER_ROW = {"q1": 12, "q21": 16, "q43": 24, "q3": 28}  # algorithmic (§8d)
+Q3_N_ORDERS, Q3_N_CUSTS = 450_000_000, 45_000_000  # SF300
+Q3_ROWS_PER_GPU = 225_000_000  # 8 GPUs × 225 M = the §8d 1.8 B-row dataset
+Q3_CUTOFF = 19950315
 HBM_PEAK_GBPS = 8000.0       # 8 TB/s spec peak (MI355X_MICROARCH.md)
 
 
@@ -131,6 +134,26 @@ def cpu_baseline(workload, rows_full):
                       f"OMP over {cores} cores, generation untimed)"}
 
 
+def cpu_baseline_q3(rows_full):
+    """Oracle Q3 probe+agg leg, single-threaded restatement (the hash agg is
+    sequential); bounded sample."""
+    from oracle import pyoracle as orc
+    import numpy as np
+    sample = min(rows_full, 5_000_000)
+    n_orders, n_custs = 2_000_000, 200_000  # scaled dims for the bounded sample
+    t0 = time.perf_counter()
+    passes = 0
+    while time.perf_counter() - t0 < 10.0:
+        orc.q3_pipeline(SEED, 0, sample, n_orders, n_custs)
+        passes += 1
+    dt = time.perf_counter() - t0
+    return {"value": round(passes * sample / dt, 1), "unit": "rows/s", "cores": 1,
+            "kind": "port",
+            "sample": f"{passes} passes over {sample} rows, dims scaled to "
+                      f"{n_orders}/{n_custs} ({dt:.1f}s, single-threaded oracle incl. "
+                      "generation+build — conservative for the CPU)"}
+
+
 def read_pmc_traffic(workload):
     p = os.path.join(REPO, "profiles", f"pmc_{workload}.json")
     if os.path.exists(p):
@@ -144,7 +167,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=50)
     ap.add_argument("--warmup", type=int, default=10)
-    ap.add_argument("--workload", choices=["q1", "q21", "q43"], default="q1")
+    ap.add_argument("--workload", choices=["q1", "q21", "q43", "q3"], default="q1")
     ap.add_argument("--rows", type=int, default=0, help="override rows per GPU")
     args = ap.parse_args()
 
@@ -165,15 +188,19 @@ def main():
         tdist.init_process_group("nccl")
         dist = tdist
 
+    if wl == "q3" and world > 1:
+        raise SystemExit("q3 multi-GPU (partitioned high-cardinality agg merge) lands in "
+                         "round 2 — run q3 at N=1 or use q43 for the partitioned mode")
     eng = Engine(local_rank)
-    rows = args.rows or (SF10_ROWS if wl == "q1" else SF100_ROWS)
+    rows = args.rows or {"q1": SF10_ROWS, "q21": SF100_ROWS, "q43": SF100_ROWS,
+                         "q3": Q3_ROWS_PER_GPU}[wl]
     row_start = rank * rows  # weak scaling: each rank owns its shard
 
     log(f"[bench] workload={wl} rows/gpu={rows} world={world} "
         f"steps={args.steps} warmup={args.warmup}")
     t_setup = time.perf_counter()
-    tables = build_dim_tables(eng, wl, rank, world)
-    n_merge = {"q1": 2, "q21": 7000, "q43": 800}[wl]
+    tables = build_dim_tables(eng, wl, rank, world) if wl != "q3" else None
+    n_merge = {"q1": 2, "q21": 7000, "q43": 800, "q3": 0}[wl]
 
     # ---- untimed setup: generate shard on device, build step closure ----
     if wl == "q1":
@@ -199,6 +226,36 @@ def main():
         def step():
             kernel_only()
             return acc.d2h(np.int64, 7000)
+    elif wl == "q3":
+        from oracle import pyoracle  # literal only (16-byte dictionary constant)
+        lit = pyoracle.mkt_literal(1)
+        mkt = eng.alloc(Q3_N_CUSTS * 16)
+        eng.gen_cust_mkt16(SEED, Q3_N_CUSTS, mkt)
+        cbits = eng.alloc((Q3_N_CUSTS + 31) // 32 * 4)
+        eng.bits_str16_eq(mkt, Q3_N_CUSTS, lit, cbits)
+        mkt.free()
+        oc, od_ = eng.alloc(Q3_N_ORDERS * 4), eng.alloc(Q3_N_ORDERS * 4)
+        eng.gen_orders_q3(SEED, Q3_N_ORDERS, Q3_N_CUSTS, oc, od_)
+        obits = eng.alloc((Q3_N_ORDERS + 31) // 32 * 4)
+        lk, ext, disc = (eng.alloc(rows * 8) for _ in range(3))
+        ship = eng.alloc(rows * 4)
+        eng.gen_lineitem_q3(SEED, row_start, rows, Q3_N_ORDERS, lk, ext, disc, ship)
+        max_out = 64_000_000
+        ok_b, os_b = eng.alloc(max_out * 8), eng.alloc(max_out * 8)
+
+        def kernel_only():
+            # fused lineitem filter + orders semi-probe + hash aggregate
+            # (includes the per-query agg-table allocate/clear + emit: the
+            # whole probe-side pass, matching the CPU's per-query hash map)
+            return eng.q3_probe_agg(lk, ext, disc, ship, rows, obits, Q3_CUTOFF,
+                                    ok_b, os_b, max_out, capacity_hint=128_000_000)
+
+        def step():
+            # orders build pass (the hash-join BUILD phase: scan orders,
+            # apply date + customer filters) + probe
+            eng.q3_order_bits(oc, od_, Q3_N_ORDERS, cbits, Q3_CUTOFF, obits)
+            g = kernel_only()
+            return np.array([g], np.int64)
     else:  # q43
         acc = eng.alloc(800 * 8)
         if world == 1:
@@ -314,7 +371,8 @@ def main():
                 "traffic": read_pmc_traffic(wl)}
 
     if rank == 0:
-        cb = cpu_baseline(wl, rows) if world == 1 else None
+        cb = cpu_baseline(wl, rows) if world == 1 and wl != "q3" else (
+            cpu_baseline_q3(rows) if world == 1 else None)
         result = {
             "metric": "ssb_join_agg_rows_per_sec",
             "value": round(value, 1),
@@ -331,7 +389,8 @@ def main():
             "config": {
                 "workload": {"q1": "ssb_sf10_q1_join_sum",
                              "q21": "ssb_sf100_q21_star_groupby",
-                             "q43": "ssb_sf100_q43_4way_star_groupby"}[wl],
+                             "q43": "ssb_sf100_q43_4way_star_groupby",
+                             "q3": "tpch_sf300_q3_join_highcard_agg"}[wl],
                 "rows_per_gpu": rows,
                 "seed": SEED,
                 "parallelism": (f"dp{world}-hash-partitioned-alltoall"

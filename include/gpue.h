@@ -175,6 +175,31 @@ int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs, gpue_join_t
 int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t num_channels,
                        uint64_t* start_points_out, gpue_dbuf* row_indexes_out);
 
+/* ---- config 5: TPC-H Q3-shaped ----
+ * lineitem ⋈ orders ⋈ customer; c_mktsegment = 16-byte dictionary string
+ * (SERIALIZED_FIXED_SIZE_LARGEINT packing, join_hash_table.cpp:185-192);
+ * revenue = extendedprice(cents) × (100−discount), scale-4 decimal int64;
+ * GROUP BY l_orderkey (high cardinality) via the hash aggregate below. */
+int gpue_gen_lineitem_q3(gpue_session* s, uint64_t seed, uint64_t row_start, uint64_t n,
+                         uint64_t n_orders, gpue_dbuf* l_orderkey /*i64*/,
+                         gpue_dbuf* l_extendedprice /*i64*/, gpue_dbuf* l_discount /*i64*/,
+                         gpue_dbuf* l_shipdate /*i32*/);
+int gpue_gen_orders_q3(gpue_session* s, uint64_t seed, uint64_t n_orders, uint32_t n_custs,
+                       gpue_dbuf* o_custkey, gpue_dbuf* o_orderdate);
+int gpue_gen_cust_mkt16(gpue_session* s, uint64_t seed, uint32_t n_custs, gpue_dbuf* out);
+/* 16-byte fixed-string equality -> membership bitset (two u64 compares) */
+int gpue_bits_str16_eq(gpue_session* s, gpue_dbuf* col16, uint64_t n, const void* lit16,
+                       gpue_dbuf* bits);
+/* orders pass bitset: o_orderdate < cutoff AND customer bit set */
+int gpue_q3_order_bits(gpue_session* s, gpue_dbuf* o_custkey, gpue_dbuf* o_orderdate,
+                       uint64_t n_orders, gpue_dbuf* cust_bits, int32_t cutoff,
+                       gpue_dbuf* order_bits);
+/* fused lineitem filter + orders semi-probe + hash aggregate */
+int gpue_q3_probe_agg(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
+                      gpue_dbuf* ship, uint64_t n, gpue_dbuf* order_bits, int32_t ship_cutoff,
+                      uint64_t capacity_hint, gpue_dbuf* out_keys, gpue_dbuf* out_sums,
+                      uint64_t max_out, uint64_t* n_groups);
+
 /* ---- generic hash aggregate ----
  * Replaces AggHashMapWithKey::compute_agg_states + update_batch +
  * convert_hash_map_to_chunk (reference be/src/exec/agg_hash_map.h:112-290,

@@ -675,6 +675,114 @@ void orc_q21_pipeline(uint64_t seed, uint64_t row_start, uint64_t n_rows,
 }
 
 /* ====================================================================== */
+/* Config 5 — TPC-H Q3-shaped (SURVEY.md §8d cfg 5): lineitem ⋈ orders ⋈
+ * customer; c_mktsegment is a 16-byte space-padded dictionary string
+ * (the selector's SERIALIZED_FIXED_SIZE_LARGEINT packing,
+ * join_hash_table.cpp:185-192); o_orderkey is dense 1..N_ORDERS (synthetic,
+ * so the orderkey column is implicit); GROUP BY l_orderkey is
+ * high-cardinality -> the generic hash aggregate. revenue units are
+ * scale-4 decimal as int64: extendedprice(cents) * (100 - discount).     */
+/* ====================================================================== */
+
+enum { TAG_LOKEY = 15, TAG_LEXT = 16, TAG_LDISC = 17, TAG_LSHIP = 18,
+       TAG_OCUST = 19, TAG_ODATE = 20, TAG_CMKT = 21 };
+#define N_ORDERS_SF300 450000000ull
+#define N_CUSTS_SF300 45000000u
+static const char* MKT_SEGMENTS[5] = {
+    "AUTOMOBILE      ", "BUILDING        ", "FURNITURE       ",
+    "MACHINERY       ", "HOUSEHOLD       "}; /* 16 B space-padded */
+
+void orc_gen_lineitem_q3(uint64_t seed, uint64_t row_start, uint64_t n,
+                         uint64_t n_orders, int64_t* l_orderkey,
+                         int64_t* l_extendedprice, int64_t* l_discount,
+                         int32_t* l_shipdate) {
+    int32_t datekey[N_DAYS];
+    orc_gen_dates(N_DAYS, datekey, NULL);
+#pragma omp parallel for schedule(static)
+    for (uint64_t i = 0; i < n; i++) {
+        uint64_t r = row_start + i;
+        l_orderkey[i] = (int64_t)(orc_gen_u64(seed, TAG_LOKEY, r) % n_orders) + 1;
+        l_extendedprice[i] = (int64_t)(orc_gen_u64(seed, TAG_LEXT, r) % 10000000u) + 1;
+        l_discount[i] = (int64_t)(orc_gen_u64(seed, TAG_LDISC, r) % 11u);
+        l_shipdate[i] = datekey[orc_gen_u64(seed, TAG_LSHIP, r) % N_DAYS];
+    }
+}
+
+void orc_gen_orders_q3(uint64_t seed, uint64_t n_orders, uint32_t n_custs,
+                       int32_t* o_custkey, int32_t* o_orderdate) {
+    int32_t datekey[N_DAYS];
+    orc_gen_dates(N_DAYS, datekey, NULL);
+#pragma omp parallel for schedule(static)
+    for (uint64_t o = 1; o <= n_orders; o++) {
+        o_custkey[o - 1] = (int32_t)(orc_gen_u64(seed, TAG_OCUST, o) % n_custs) + 1;
+        o_orderdate[o - 1] = datekey[orc_gen_u64(seed, TAG_ODATE, o) % N_DAYS];
+    }
+}
+
+/* c_mktsegment as 16-byte fixed strings (out: n*16 bytes) */
+void orc_gen_cust_mkt16(uint64_t seed, uint32_t n_custs, uint8_t* out) {
+#pragma omp parallel for schedule(static)
+    for (uint32_t c = 1; c <= n_custs; c++) {
+        uint32_t seg = (uint32_t)(orc_gen_u64(seed, TAG_CMKT, c) % 5u);
+        memcpy(out + (size_t)(c - 1) * 16, MKT_SEGMENTS[seg], 16);
+    }
+}
+
+const char* orc_mkt_segment_literal(int idx) { return MKT_SEGMENTS[idx]; }
+
+/* customer pass bitset: 16-byte string equality (SERIALIZED_FIXED_SIZE
+ * packing: two u64 compares) */
+void orc_q3_build_cust_bits(const uint8_t* mkt16, uint32_t n_custs,
+                            const char* lit16, uint8_t* bits) {
+    memset(bits, 0, (n_custs + 7) / 8);
+    uint64_t la, lb;
+    memcpy(&la, lit16, 8);
+    memcpy(&lb, lit16 + 8, 8);
+    for (uint32_t c = 0; c < n_custs; c++) {
+        uint64_t a, b;
+        memcpy(&a, mkt16 + (size_t)c * 16, 8);
+        memcpy(&b, mkt16 + (size_t)c * 16 + 8, 8);
+        if (a == la && b == lb) bits[c / 8] |= 1u << (c % 8);
+    }
+}
+
+/* orders pass bitset: o_orderdate < cutoff AND customer passes */
+void orc_q3_build_order_bits(const int32_t* o_custkey, const int32_t* o_orderdate,
+                             uint64_t n_orders, const uint8_t* cust_bits,
+                             int32_t date_cutoff, uint8_t* bits) {
+    memset(bits, 0, (n_orders + 7) / 8);
+    for (uint64_t o = 0; o < n_orders; o++) {
+        if (o_orderdate[o] < date_cutoff) {
+            uint32_t c = (uint32_t)o_custkey[o] - 1;
+            if (cust_bits[c / 8] & (1u << (c % 8))) bits[o / 8] |= 1u << (o % 8);
+        }
+    }
+}
+
+/* fused lineitem probe + hash agg: returns group count (keys=l_orderkey,
+ * sums=revenue scale-4) via orc_hash_agg_sum_u64 on the passing rows */
+uint64_t orc_q3_probe_agg(const int64_t* lk, const int64_t* ext, const int64_t* disc,
+                          const int32_t* ship, uint64_t n, const uint8_t* order_bits,
+                          int32_t ship_cutoff, uint64_t* out_keys, int64_t* out_sums,
+                          uint64_t max_out) {
+    uint64_t* keys = (uint64_t*)malloc(n * 8);
+    int64_t* vals = (int64_t*)malloc(n * 8);
+    uint64_t m = 0;
+    for (uint64_t i = 0; i < n; i++) {
+        if (ship[i] <= ship_cutoff) continue;
+        uint64_t o = (uint64_t)lk[i] - 1;
+        if (!(order_bits[o / 8] & (1u << (o % 8)))) continue;
+        keys[m] = (uint64_t)lk[i];
+        vals[m] = ext[i] * (100 - disc[i]);
+        m++;
+    }
+    uint64_t g = orc_hash_agg_sum_u64(keys, vals, m, out_keys, out_sums, NULL, max_out);
+    free(keys);
+    free(vals);
+    return g;
+}
+
+/* ====================================================================== */
 /* Generic hash aggregate — restates AggHashMapWithKey::compute_agg_states
  * + AggregateFunction SUM/COUNT update_batch (be/src/exec/agg_hash_map.h:
  * 112-290, be/src/exec/aggregator.cpp:937-959, be/src/exprs/agg/sum.h:45):

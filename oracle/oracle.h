@@ -150,6 +150,25 @@ void orc_q43_pipeline(uint64_t seed, uint64_t row_start, uint64_t n_rows,
                       int32_t region, int32_t nation, int32_t category,
                       int threads, int64_t* group_sums /*800*/);
 
+/* ---- TPC-H Q3 (config 5) ---- */
+void orc_gen_lineitem_q3(uint64_t seed, uint64_t row_start, uint64_t n,
+                         uint64_t n_orders, int64_t* l_orderkey,
+                         int64_t* l_extendedprice, int64_t* l_discount,
+                         int32_t* l_shipdate);
+void orc_gen_orders_q3(uint64_t seed, uint64_t n_orders, uint32_t n_custs,
+                       int32_t* o_custkey, int32_t* o_orderdate);
+void orc_gen_cust_mkt16(uint64_t seed, uint32_t n_custs, uint8_t* out);
+const char* orc_mkt_segment_literal(int idx);
+void orc_q3_build_cust_bits(const uint8_t* mkt16, uint32_t n_custs,
+                            const char* lit16, uint8_t* bits);
+void orc_q3_build_order_bits(const int32_t* o_custkey, const int32_t* o_orderdate,
+                             uint64_t n_orders, const uint8_t* cust_bits,
+                             int32_t date_cutoff, uint8_t* bits);
+uint64_t orc_q3_probe_agg(const int64_t* lk, const int64_t* ext, const int64_t* disc,
+                          const int32_t* ship, uint64_t n, const uint8_t* order_bits,
+                          int32_t ship_cutoff, uint64_t* out_keys, int64_t* out_sums,
+                          uint64_t max_out);
+
 /* generic hash aggregate (agg_hash_map.h:112-290 restatement); returns group
  * count or UINT64_MAX if max_out exceeded */
 uint64_t orc_hash_agg_sum_u64(const uint64_t* keys, const int64_t* vals, uint64_t n,

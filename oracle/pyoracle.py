@@ -90,6 +90,15 @@ def _decl(lib):
     lib.orc_build_part_dim_q43.argtypes = [c_u64, u, c_i32, c_vp]
     lib.orc_q43_kernel.argtypes = [c_vp] * 6 + [c_u64] + [c_vp] * 4 + [c_i64, c_i32, c_vp]
     lib.orc_q43_pipeline.argtypes = [c_u64, c_u64, c_u64, c_i32, c_i32, c_i32, c_i32, c_vp]
+    lib.orc_gen_lineitem_q3.argtypes = [c_u64, c_u64, c_u64, c_u64] + [c_vp] * 4
+    lib.orc_gen_orders_q3.argtypes = [c_u64, c_u64, u, c_vp, c_vp]
+    lib.orc_gen_cust_mkt16.argtypes = [c_u64, u, c_vp]
+    lib.orc_mkt_segment_literal.restype = ctypes.c_char_p
+    lib.orc_mkt_segment_literal.argtypes = [c_i32]
+    lib.orc_q3_build_cust_bits.argtypes = [c_vp, u, ctypes.c_char_p, c_vp]
+    lib.orc_q3_build_order_bits.argtypes = [c_vp, c_vp, c_u64, c_vp, c_i32, c_vp]
+    lib.orc_q3_probe_agg.restype = c_u64
+    lib.orc_q3_probe_agg.argtypes = [c_vp] * 4 + [c_u64, c_vp, c_i32, c_vp, c_vp, c_u64]
     lib.orc_hash_agg_sum_u64.restype = c_u64
     lib.orc_hash_agg_sum_u64.argtypes = [c_vp, c_vp, c_u64, c_vp, c_vp, c_vp, c_u64]
     lib.orc_q1_kernel.restype = c_i64
@@ -184,6 +193,52 @@ def q43_kernel(ck, sk, pk, od, rv, sc, cfirst, sfirst, pfirst, dfirst, dmin, thr
                           _p(cfirst), _p(sfirst), _p(pfirst), _p(dfirst), dmin,
                           threads, _p(out))
     return out
+
+
+def gen_lineitem_q3(seed, row_start, n, n_orders):
+    lk = np.empty(n, np.int64)
+    ext = np.empty(n, np.int64)
+    disc = np.empty(n, np.int64)
+    ship = np.empty(n, np.int32)
+    load().orc_gen_lineitem_q3(seed, row_start, n, n_orders, _p(lk), _p(ext), _p(disc), _p(ship))
+    return lk, ext, disc, ship
+
+
+def gen_orders_q3(seed, n_orders, n_custs):
+    oc = np.empty(n_orders, np.int32)
+    od = np.empty(n_orders, np.int32)
+    load().orc_gen_orders_q3(seed, n_orders, n_custs, _p(oc), _p(od))
+    return oc, od
+
+
+def gen_cust_mkt16(seed, n_custs):
+    out = np.empty(n_custs * 16, np.uint8)
+    load().orc_gen_cust_mkt16(seed, n_custs, _p(out))
+    return out
+
+
+def mkt_literal(idx=1) -> bytes:
+    return load().orc_mkt_segment_literal(idx)
+
+
+def q3_pipeline(seed, row_start, n, n_orders, n_custs, seg=1,
+                date_cutoff=19950315, ship_cutoff=19950315):
+    """Full Q3 oracle pipeline; returns (keys, sums) key-sorted."""
+    mkt = gen_cust_mkt16(seed, n_custs)
+    cbits = np.zeros((n_custs + 7) // 8, np.uint8)
+    load().orc_q3_build_cust_bits(_p(mkt), n_custs, mkt_literal(seg), _p(cbits))
+    oc, od = gen_orders_q3(seed, n_orders, n_custs)
+    obits = np.zeros((n_orders + 7) // 8, np.uint8)
+    load().orc_q3_build_order_bits(_p(oc), _p(od), n_orders, _p(cbits), date_cutoff, _p(obits))
+    lk, ext, disc, ship = gen_lineitem_q3(seed, row_start, n, n_orders)
+    cap = max(n, 16)
+    ok = np.empty(cap, np.uint64)
+    os_ = np.empty(cap, np.int64)
+    g = load().orc_q3_probe_agg(_p(lk), _p(ext), _p(disc), _p(ship), n, _p(obits),
+                                ship_cutoff, _p(ok), _p(os_), cap)
+    assert g != 2**64 - 1
+    order = np.argsort(ok[:g])
+    return ok[:g][order].copy(), os_[:g][order].copy()
 
 
 def hash_agg_sum(keys: np.ndarray, vals: np.ndarray):

@@ -218,6 +218,15 @@ static constexpr int N_DAYS = 2556;
 static constexpr uint32_t N_PARTS_SF100 = 1400000u;
 static constexpr uint32_t N_SUPPS_SF100 = 200000u;
 static constexpr uint32_t N_CUSTS_SF100 = 3000000u;
+enum { TAG_LOKEY = 15, TAG_LEXT = 16, TAG_LDISC = 17, TAG_LSHIP = 18,
+       TAG_OCUST = 19, TAG_ODATE = 20, TAG_CMKT = 21 };
+// 16-byte space-padded dictionary (identical to oracle MKT_SEGMENTS)
+__constant__ char MKT_SEGMENTS_DEV[5][16] = {
+    {'A','U','T','O','M','O','B','I','L','E',' ',' ',' ',' ',' ',' '},
+    {'B','U','I','L','D','I','N','G',' ',' ',' ',' ',' ',' ',' ',' '},
+    {'F','U','R','N','I','T','U','R','E',' ',' ',' ',' ',' ',' ',' '},
+    {'M','A','C','H','I','N','E','R','Y',' ',' ',' ',' ',' ',' ',' '},
+    {'H','O','U','S','E','H','O','L','D',' ',' ',' ',' ',' ',' ',' '}};
 
 __global__ void k_gen_u32_mod(uint32_t* out, uint64_t seed, uint64_t tag,
                               uint64_t row_start, uint64_t n, uint32_t mod, uint32_t add) {
@@ -1610,6 +1619,229 @@ int gpue_hash_agg_sum_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uin
     if (groups > max_out) {
         snprintf(g_err, sizeof(g_err), "hash_agg: %llu groups exceed max_out %llu",
                  groups, (unsigned long long)max_out);
+        return GPUE_ERR_ARG;
+    }
+    return GPUE_OK;
+}
+
+// ---------------------------------------------------------------------------
+// Config 5 — TPC-H Q3-shaped: lineitem ⋈ orders ⋈ customer with a 16-byte
+// dictionary-string filter (SERIALIZED_FIXED_SIZE_LARGEINT packing: two u64
+// compares — reference join_hash_table.cpp:185-192), decimal revenue as
+// scale-4 int64 (extendedprice cents × (100−discount)), and a
+// HIGH-cardinality GROUP BY l_orderkey through the hash aggregate below.
+// o_orderkey is dense 1..N (synthetic), so the orders "hash table" is the
+// RANGE_DIRECT bitset over the passing orders.
+// ---------------------------------------------------------------------------
+__global__ void k_gen_lineitem_q3(int64_t* lk, int64_t* ext, int64_t* disc, int32_t* ship,
+                                  const int32_t* __restrict__ datekey, uint64_t seed,
+                                  uint64_t row_start, uint64_t n, uint64_t n_orders) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint64_t r = row_start + i;
+        lk[i] = (int64_t)(gen_u64(seed, TAG_LOKEY, r) % n_orders) + 1;
+        ext[i] = (int64_t)(gen_u64(seed, TAG_LEXT, r) % 10000000u) + 1;
+        disc[i] = (int64_t)(gen_u64(seed, TAG_LDISC, r) % 11u);
+        ship[i] = datekey[gen_u64(seed, TAG_LSHIP, r) % N_DAYS];
+    }
+}
+
+__global__ void k_gen_orders_q3(int32_t* ocust, int32_t* odate,
+                                const int32_t* __restrict__ datekey, uint64_t seed,
+                                uint64_t n_orders, uint32_t n_custs) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t o = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; o <= n_orders;
+         o += stride) {
+        ocust[o - 1] = (int32_t)(gen_u64(seed, TAG_OCUST, o) % n_custs) + 1;
+        odate[o - 1] = datekey[gen_u64(seed, TAG_ODATE, o) % N_DAYS];
+    }
+}
+
+__global__ void k_gen_cust_mkt16(uint8_t* out, uint64_t seed, uint32_t n_custs) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t c = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; c <= n_custs;
+         c += stride) {
+        uint32_t seg = (uint32_t)(gen_u64(seed, TAG_CMKT, c) % 5u);
+        ulonglong2 v = *reinterpret_cast<const ulonglong2*>(MKT_SEGMENTS_DEV[seg]);
+        *reinterpret_cast<ulonglong2*>(out + (c - 1) * 16) = v;
+    }
+}
+
+// 16-byte fixed-string equality -> bitset (two u64 compares per row)
+__global__ void k_bits_str16_eq(const ulonglong2* __restrict__ col, uint64_t n,
+                                unsigned long long la, unsigned long long lb,
+                                uint32_t* __restrict__ bits) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        ulonglong2 v = col[i];
+        if (v.x == la && v.y == lb) atomicOr(&bits[i >> 5], 1u << (i & 31));
+    }
+}
+
+__global__ void k_q3_order_bits(const int32_t* __restrict__ ocust,
+                                const int32_t* __restrict__ odate, uint64_t n_orders,
+                                const uint32_t* __restrict__ cust_bits, int32_t cutoff,
+                                uint32_t* __restrict__ bits) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t o = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; o < n_orders;
+         o += stride) {
+        if (odate[o] < cutoff) {
+            uint32_t c = (uint32_t)ocust[o] - 1;
+            if ((cust_bits[c >> 5] >> (c & 31)) & 1u)
+                atomicOr(&bits[o >> 5], 1u << (o & 31));
+        }
+    }
+}
+
+// fused lineitem filter + orders semi-probe + hash-agg insert
+__global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
+                               const int64_t* __restrict__ ext,
+                               const int64_t* __restrict__ disc,
+                               const int32_t* __restrict__ ship, uint64_t n,
+                               const uint32_t* __restrict__ order_bits, int32_t ship_cutoff,
+                               unsigned long long* __restrict__ slots,
+                               unsigned long long* __restrict__ sums,
+                               unsigned long long* __restrict__ counts, uint64_t cap_mask) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        if (ship[i] <= ship_cutoff) continue;
+        unsigned long long k = (unsigned long long)lk[i];
+        uint64_t o = k - 1;
+        if (!((order_bits[o >> 5] >> (o & 31)) & 1u)) continue;
+        unsigned long long v = (unsigned long long)(ext[i] * (100 - disc[i]));
+        uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        for (;;) {
+            unsigned long long cur = slots[s];
+            if (cur == k) { atomicAdd(&sums[s], v); atomicAdd(&counts[s], 1ull); break; }
+            if (cur == AGG_EMPTY) {
+                unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
+                if (old == AGG_EMPTY || old == k) {
+                    atomicAdd(&sums[s], v);
+                    atomicAdd(&counts[s], 1ull);
+                    break;
+                }
+            }
+            s = (s + 1) & cap_mask;
+        }
+    }
+}
+
+extern "C" {
+int gpue_gen_lineitem_q3(gpue_session* s, uint64_t seed, uint64_t row_start, uint64_t n,
+                         uint64_t n_orders, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
+                         gpue_dbuf* ship);
+int gpue_gen_orders_q3(gpue_session* s, uint64_t seed, uint64_t n_orders, uint32_t n_custs,
+                       gpue_dbuf* ocust, gpue_dbuf* odate);
+int gpue_gen_cust_mkt16(gpue_session* s, uint64_t seed, uint32_t n_custs, gpue_dbuf* out);
+int gpue_bits_str16_eq(gpue_session* s, gpue_dbuf* col16, uint64_t n, const void* lit16,
+                       gpue_dbuf* bits);
+int gpue_q3_order_bits(gpue_session* s, gpue_dbuf* ocust, gpue_dbuf* odate, uint64_t n_orders,
+                       gpue_dbuf* cust_bits, int32_t cutoff, gpue_dbuf* order_bits);
+int gpue_q3_probe_agg(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
+                      gpue_dbuf* ship, uint64_t n, gpue_dbuf* order_bits, int32_t ship_cutoff,
+                      uint64_t capacity_hint, gpue_dbuf* out_keys, gpue_dbuf* out_sums,
+                      uint64_t max_out, uint64_t* n_groups);
+}
+
+int gpue_gen_lineitem_q3(gpue_session* s, uint64_t seed, uint64_t row_start, uint64_t n,
+                         uint64_t n_orders, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
+                         gpue_dbuf* ship) {
+    ARG_CHECK(s && lk && ext && disc && ship && n_orders > 0);
+    ARG_CHECK(lk->bytes >= n * 8 && ext->bytes >= n * 8 && disc->bytes >= n * 8 &&
+              ship->bytes >= n * 4);
+    int rc = ensure_datekey(s);
+    if (rc != GPUE_OK) return rc;
+    hipLaunchKernelGGL(k_gen_lineitem_q3, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+                       (int64_t*)lk->ptr, (int64_t*)ext->ptr, (int64_t*)disc->ptr,
+                       (int32_t*)ship->ptr, s->d_datekey, seed, row_start, n, n_orders);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
+int gpue_gen_orders_q3(gpue_session* s, uint64_t seed, uint64_t n_orders, uint32_t n_custs,
+                       gpue_dbuf* ocust, gpue_dbuf* odate) {
+    ARG_CHECK(s && ocust && odate && ocust->bytes >= n_orders * 4 &&
+              odate->bytes >= n_orders * 4);
+    int rc = ensure_datekey(s);
+    if (rc != GPUE_OK) return rc;
+    hipLaunchKernelGGL(k_gen_orders_q3, dim3(grid_for(n_orders)), dim3(BLOCK), 0, s->stream,
+                       (int32_t*)ocust->ptr, (int32_t*)odate->ptr, s->d_datekey, seed,
+                       n_orders, n_custs);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
+int gpue_gen_cust_mkt16(gpue_session* s, uint64_t seed, uint32_t n_custs, gpue_dbuf* out) {
+    ARG_CHECK(s && out && out->bytes >= (uint64_t)n_custs * 16);
+    hipLaunchKernelGGL(k_gen_cust_mkt16, dim3(grid_for(n_custs)), dim3(BLOCK), 0, s->stream,
+                       (uint8_t*)out->ptr, seed, n_custs);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
+int gpue_bits_str16_eq(gpue_session* s, gpue_dbuf* col16, uint64_t n, const void* lit16,
+                       gpue_dbuf* bits) {
+    ARG_CHECK(s && col16 && lit16 && bits);
+    ARG_CHECK(col16->bytes >= n * 16 && bits->bytes >= (n + 31) / 32 * 4);
+    unsigned long long la, lb;
+    memcpy(&la, lit16, 8);
+    memcpy(&lb, (const char*)lit16 + 8, 8);
+    HIP_CHECK(hipMemsetAsync(bits->ptr, 0, (n + 31) / 32 * 4, s->stream));
+    hipLaunchKernelGGL(k_bits_str16_eq, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+                       (const ulonglong2*)col16->ptr, n, la, lb, (uint32_t*)bits->ptr);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
+int gpue_q3_order_bits(gpue_session* s, gpue_dbuf* ocust, gpue_dbuf* odate, uint64_t n_orders,
+                       gpue_dbuf* cust_bits, int32_t cutoff, gpue_dbuf* order_bits) {
+    ARG_CHECK(s && ocust && odate && cust_bits && order_bits);
+    ARG_CHECK(order_bits->bytes >= (n_orders + 31) / 32 * 4);
+    HIP_CHECK(hipMemsetAsync(order_bits->ptr, 0, (n_orders + 31) / 32 * 4, s->stream));
+    hipLaunchKernelGGL(k_q3_order_bits, dim3(grid_for(n_orders)), dim3(BLOCK), 0, s->stream,
+                       (const int32_t*)ocust->ptr, (const int32_t*)odate->ptr, n_orders,
+                       (const uint32_t*)cust_bits->ptr, cutoff, (uint32_t*)order_bits->ptr);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
+int gpue_q3_probe_agg(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
+                      gpue_dbuf* ship, uint64_t n, gpue_dbuf* order_bits, int32_t ship_cutoff,
+                      uint64_t capacity_hint, gpue_dbuf* out_keys, gpue_dbuf* out_sums,
+                      uint64_t max_out, uint64_t* n_groups) {
+    ARG_CHECK(s && lk && ext && disc && ship && order_bits && out_keys && out_sums && n_groups);
+    uint64_t cap = 16;
+    uint64_t want = capacity_hint ? capacity_hint : n / 4;
+    while (cap < want) cap <<= 1;
+    unsigned long long *d_slots = nullptr, *d_sums = nullptr, *d_counts = nullptr,
+                       *d_cursor = nullptr;
+    HIP_CHECK(hipMalloc(&d_slots, cap * 8));
+    HIP_CHECK(hipMalloc(&d_sums, cap * 8));
+    HIP_CHECK(hipMalloc(&d_counts, cap * 8));
+    HIP_CHECK(hipMalloc(&d_cursor, 8));
+    HIP_CHECK(hipMemsetAsync(d_slots, 0xFF, cap * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_sums, 0, cap * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_counts, 0, cap * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_cursor, 0, 8, s->stream));
+    hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+                       (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
+                       (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
+                       (const uint32_t*)order_bits->ptr, ship_cutoff, d_slots, d_sums,
+                       d_counts, cap - 1);
+    hipLaunchKernelGGL(k_hash_agg_emit, dim3(grid_for(cap)), dim3(BLOCK), 0, s->stream,
+                       d_slots, d_sums, d_counts, cap, d_cursor, max_out,
+                       (uint64_t*)out_keys->ptr, (int64_t*)out_sums->ptr, nullptr);
+    unsigned long long groups = 0;
+    HIP_CHECK(hipMemcpyAsync(&groups, d_cursor, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_slots);
+    (void)hipFree(d_sums);
+    (void)hipFree(d_counts);
+    (void)hipFree(d_cursor);
+    *n_groups = groups;
+    if (groups > max_out) {
+        snprintf(g_err, sizeof(g_err), "q3: %llu groups exceed max_out %llu", groups,
+                 (unsigned long long)max_out);
         return GPUE_ERR_ARG;
     }
     return GPUE_OK;

@@ -86,6 +86,13 @@ def _declare(lib):
         "gpue_q21_star_agg_async": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_partition_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
         "gpue_gather_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
+        "gpue_gen_lineitem_q3": (c_i32, [c_vp, c_u64, c_u64, c_u64, c_u64] + [c_vp] * 4),
+        "gpue_gen_orders_q3": (c_i32, [c_vp, c_u64, c_u64, c_u32, c_vp, c_vp]),
+        "gpue_gen_cust_mkt16": (c_i32, [c_vp, c_u64, c_u32, c_vp]),
+        "gpue_bits_str16_eq": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_vp]),
+        "gpue_q3_order_bits": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp, c_i32, c_vp]),
+        "gpue_q3_probe_agg": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_vp, c_i32, c_u64,
+                                     c_vp, c_vp, c_u64, ctypes.POINTER(c_u64)]),
         "gpue_hash_agg_sum_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_u64, c_vp, c_vp, c_vp,
                                           c_u64, ctypes.POINTER(c_u64)]),
         "gpue_timer_start": (c_i32, [c_vp]),
@@ -281,6 +288,33 @@ class Engine:
                                                    pk._h, sk._h, od._h, rv._h, n,
                                                    out.ctypes.data_as(c_vp)))
         return out
+
+    def gen_lineitem_q3(self, seed, row_start, n, n_orders, lk, ext, disc, ship):
+        _ck(self._lib, self._lib.gpue_gen_lineitem_q3(self._h, seed, row_start, n, n_orders,
+                                                      lk._h, ext._h, disc._h, ship._h))
+
+    def gen_orders_q3(self, seed, n_orders, n_custs, ocust, odate):
+        _ck(self._lib, self._lib.gpue_gen_orders_q3(self._h, seed, n_orders, n_custs,
+                                                    ocust._h, odate._h))
+
+    def gen_cust_mkt16(self, seed, n_custs, out):
+        _ck(self._lib, self._lib.gpue_gen_cust_mkt16(self._h, seed, n_custs, out._h))
+
+    def bits_str16_eq(self, col16, n, lit16: bytes, bits):
+        assert len(lit16) == 16
+        _ck(self._lib, self._lib.gpue_bits_str16_eq(self._h, col16._h, n, lit16, bits._h))
+
+    def q3_order_bits(self, ocust, odate, n_orders, cust_bits, cutoff, order_bits):
+        _ck(self._lib, self._lib.gpue_q3_order_bits(self._h, ocust._h, odate._h, n_orders,
+                                                    cust_bits._h, cutoff, order_bits._h))
+
+    def q3_probe_agg(self, lk, ext, disc, ship, n, order_bits, ship_cutoff,
+                     out_keys, out_sums, max_out, capacity_hint=0):
+        g = c_u64()
+        _ck(self._lib, self._lib.gpue_q3_probe_agg(
+            self._h, lk._h, ext._h, disc._h, ship._h, n, order_bits._h, ship_cutoff,
+            capacity_hint, out_keys._h, out_sums._h, max_out, ctypes.byref(g)))
+        return g.value
 
     def hash_agg_sum_u64(self, keys: DBuf, vals: DBuf, n, out_keys: DBuf, out_sums: DBuf,
                          out_counts: DBuf = None, max_out=0, capacity_hint=0):

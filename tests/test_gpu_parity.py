@@ -327,6 +327,36 @@ def test_hash_agg_parity(engine, nkeys, card):
         b.free()
 
 
+def test_q3_pipeline_parity(engine):
+    """Config 5 (TPC-H Q3-shaped) on device vs oracle: string-dict filter,
+    orders semi-join bitset, high-cardinality GROUP BY — key-sorted (keys,
+    sums) bit-exact."""
+    n, n_orders, n_custs = 3_000_000, 200_000, 40_000
+    mkt = engine.alloc(n_custs * 16)
+    engine.gen_cust_mkt16(SEED, n_custs, mkt)
+    cbits = engine.alloc((n_custs + 31) // 32 * 4)
+    engine.bits_str16_eq(mkt, n_custs, orc.mkt_literal(1), cbits)
+    oc, od = engine.alloc(n_orders * 4), engine.alloc(n_orders * 4)
+    engine.gen_orders_q3(SEED, n_orders, n_custs, oc, od)
+    obits = engine.alloc((n_orders + 31) // 32 * 4)
+    engine.q3_order_bits(oc, od, n_orders, cbits, 19950315, obits)
+    lk, ext, disc = (engine.alloc(n * 8) for _ in range(3))
+    ship = engine.alloc(n * 4)
+    engine.gen_lineitem_q3(SEED, 0, n, n_orders, lk, ext, disc, ship)
+    max_out = n
+    ok_b, os_b = engine.alloc(max_out * 8), engine.alloc(max_out * 8)
+    g = engine.q3_probe_agg(lk, ext, disc, ship, n, obits, 19950315, ok_b, os_b, max_out)
+    gk = ok_b.d2h(np.uint64, g)
+    gs = os_b.d2h(np.int64, g)
+    order = np.argsort(gk)
+    ek, es = orc.q3_pipeline(SEED, 0, n, n_orders, n_custs)
+    assert g == len(ek)
+    assert np.array_equal(gk[order], ek)
+    assert np.array_equal(gs[order], es)
+    for b in (mkt, cbits, oc, od, obits, lk, ext, disc, ship, ok_b, os_b):
+        b.free()
+
+
 def test_partition_parity(engine):
     n, nch = 3_000_000, 8
     keys = engine.alloc(n * 4)
