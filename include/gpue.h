@@ -194,11 +194,21 @@ int gpue_bits_str16_eq(gpue_session* s, gpue_dbuf* col16, uint64_t n, const void
 int gpue_q3_order_bits(gpue_session* s, gpue_dbuf* o_custkey, gpue_dbuf* o_orderdate,
                        uint64_t n_orders, gpue_dbuf* cust_bits, int32_t cutoff,
                        gpue_dbuf* order_bits);
+/* persistent aggregate table (per-query hash map analog, reused across
+ * passes so repeated executions pay reset, not allocation) */
+typedef struct gpue_agg_table gpue_agg_table;
+int gpue_agg_table_create(gpue_session* s, uint64_t capacity, gpue_agg_table** out);
+void gpue_agg_table_destroy(gpue_agg_table* t);
 /* fused lineitem filter + orders semi-probe + hash aggregate */
 int gpue_q3_probe_agg(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
                       gpue_dbuf* ship, uint64_t n, gpue_dbuf* order_bits, int32_t ship_cutoff,
                       uint64_t capacity_hint, gpue_dbuf* out_keys, gpue_dbuf* out_sums,
                       uint64_t max_out, uint64_t* n_groups);
+/* variant over a persistent gpue_agg_table (reset + probe + emit) */
+int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
+                        gpue_dbuf* ship, uint64_t n, gpue_dbuf* order_bits,
+                        int32_t ship_cutoff, gpue_agg_table* at, gpue_dbuf* out_keys,
+                        gpue_dbuf* out_sums, uint64_t max_out, uint64_t* n_groups);
 
 /* ---- generic hash aggregate ----
  * Replaces AggHashMapWithKey::compute_agg_states + update_batch +

@@ -1527,6 +1527,50 @@ int gpue_ubench(gpue_session* s, int which, gpue_dbuf* a, gpue_dbuf* b, gpue_dbu
 // ---------------------------------------------------------------------------
 static constexpr unsigned long long AGG_EMPTY = 0xFFFFFFFFFFFFFFFFull;
 
+struct gpue_agg_table {
+    gpue_session* s;
+    uint64_t cap;
+    unsigned long long* slots;
+    unsigned long long* sums;
+    unsigned long long* counts;
+    unsigned long long* cursor;
+};
+
+extern "C" {
+int gpue_agg_table_create(gpue_session* s, uint64_t capacity, gpue_agg_table** out);
+void gpue_agg_table_destroy(gpue_agg_table* t);
+}
+
+int gpue_agg_table_create(gpue_session* s, uint64_t capacity, gpue_agg_table** out) {
+    ARG_CHECK(s && out && capacity >= 16);
+    uint64_t cap = 16;
+    while (cap < capacity) cap <<= 1;
+    gpue_agg_table* t = new gpue_agg_table{s, cap, nullptr, nullptr, nullptr, nullptr};
+    HIP_CHECK(hipMalloc(&t->slots, cap * 8));
+    HIP_CHECK(hipMalloc(&t->sums, cap * 8));
+    HIP_CHECK(hipMalloc(&t->counts, cap * 8));
+    HIP_CHECK(hipMalloc(&t->cursor, 8));
+    *out = t;
+    return GPUE_OK;
+}
+
+void gpue_agg_table_destroy(gpue_agg_table* t) {
+    if (!t) return;
+    (void)hipFree(t->slots);
+    (void)hipFree(t->sums);
+    (void)hipFree(t->counts);
+    (void)hipFree(t->cursor);
+    delete t;
+}
+
+static int agg_table_reset(gpue_agg_table* t) {
+    HIP_CHECK(hipMemsetAsync(t->slots, 0xFF, t->cap * 8, t->s->stream));
+    HIP_CHECK(hipMemsetAsync(t->sums, 0, t->cap * 8, t->s->stream));
+    HIP_CHECK(hipMemsetAsync(t->counts, 0, t->cap * 8, t->s->stream));
+    HIP_CHECK(hipMemsetAsync(t->cursor, 0, 8, t->s->stream));
+    return GPUE_OK;
+}
+
 __global__ void k_hash_agg_sum(const uint64_t* __restrict__ keys,
                                const int64_t* __restrict__ vals, uint64_t n,
                                unsigned long long* __restrict__ slots,
@@ -1802,6 +1846,39 @@ int gpue_q3_order_bits(gpue_session* s, gpue_dbuf* ocust, gpue_dbuf* odate, uint
                        (const int32_t*)ocust->ptr, (const int32_t*)odate->ptr, n_orders,
                        (const uint32_t*)cust_bits->ptr, cutoff, (uint32_t*)order_bits->ptr);
     HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
+extern "C" int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext,
+                                   gpue_dbuf* disc, gpue_dbuf* ship, uint64_t n,
+                                   gpue_dbuf* order_bits, int32_t ship_cutoff,
+                                   gpue_agg_table* at, gpue_dbuf* out_keys,
+                                   gpue_dbuf* out_sums, uint64_t max_out, uint64_t* n_groups);
+int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
+                        gpue_dbuf* ship, uint64_t n, gpue_dbuf* order_bits,
+                        int32_t ship_cutoff, gpue_agg_table* at, gpue_dbuf* out_keys,
+                        gpue_dbuf* out_sums, uint64_t max_out, uint64_t* n_groups) {
+    ARG_CHECK(s && lk && ext && disc && ship && order_bits && at && out_keys && out_sums &&
+              n_groups);
+    int rc = agg_table_reset(at);
+    if (rc != GPUE_OK) return rc;
+    hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+                       (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
+                       (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
+                       (const uint32_t*)order_bits->ptr, ship_cutoff, at->slots, at->sums,
+                       at->counts, at->cap - 1);
+    hipLaunchKernelGGL(k_hash_agg_emit, dim3(grid_for(at->cap)), dim3(BLOCK), 0, s->stream,
+                       at->slots, at->sums, at->counts, at->cap, at->cursor, max_out,
+                       (uint64_t*)out_keys->ptr, (int64_t*)out_sums->ptr, nullptr);
+    unsigned long long groups = 0;
+    HIP_CHECK(hipMemcpyAsync(&groups, at->cursor, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    *n_groups = groups;
+    if (groups > max_out) {
+        snprintf(g_err, sizeof(g_err), "q3: %llu groups exceed max_out %llu", groups,
+                 (unsigned long long)max_out);
+        return GPUE_ERR_ARG;
+    }
     return GPUE_OK;
 }
 

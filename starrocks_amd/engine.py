@@ -91,6 +91,10 @@ def _declare(lib):
         "gpue_gen_cust_mkt16": (c_i32, [c_vp, c_u64, c_u32, c_vp]),
         "gpue_bits_str16_eq": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_vp]),
         "gpue_q3_order_bits": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp, c_i32, c_vp]),
+        "gpue_agg_table_create": (c_i32, [c_vp, c_u64, ctypes.POINTER(c_vp)]),
+        "gpue_agg_table_destroy": (None, [c_vp]),
+        "gpue_q3_probe_agg_t": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_vp, c_i32, c_vp,
+                                       c_vp, c_vp, c_u64, ctypes.POINTER(c_u64)]),
         "gpue_q3_probe_agg": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_vp, c_i32, c_u64,
                                      c_vp, c_vp, c_u64, ctypes.POINTER(c_u64)]),
         "gpue_hash_agg_sum_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_u64, c_vp, c_vp, c_vp,
@@ -307,6 +311,22 @@ class Engine:
     def q3_order_bits(self, ocust, odate, n_orders, cust_bits, cutoff, order_bits):
         _ck(self._lib, self._lib.gpue_q3_order_bits(self._h, ocust._h, odate._h, n_orders,
                                                     cust_bits._h, cutoff, order_bits._h))
+
+    def agg_table_create(self, capacity):
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_agg_table_create(self._h, capacity, ctypes.byref(h)))
+        return h
+
+    def agg_table_destroy(self, h):
+        self._lib.gpue_agg_table_destroy(h)
+
+    def q3_probe_agg_t(self, lk, ext, disc, ship, n, order_bits, ship_cutoff, at,
+                       out_keys, out_sums, max_out):
+        g = c_u64()
+        _ck(self._lib, self._lib.gpue_q3_probe_agg_t(
+            self._h, lk._h, ext._h, disc._h, ship._h, n, order_bits._h, ship_cutoff,
+            at, out_keys._h, out_sums._h, max_out, ctypes.byref(g)))
+        return g.value
 
     def q3_probe_agg(self, lk, ext, disc, ship, n, order_bits, ship_cutoff,
                      out_keys, out_sums, max_out, capacity_hint=0):
