@@ -1563,10 +1563,10 @@ void gpue_agg_table_destroy(gpue_agg_table* t) {
     delete t;
 }
 
-static int agg_table_reset(gpue_agg_table* t) {
+static int agg_table_reset(gpue_agg_table* t, bool with_counts = true) {
     HIP_CHECK(hipMemsetAsync(t->slots, 0xFF, t->cap * 8, t->s->stream));
     HIP_CHECK(hipMemsetAsync(t->sums, 0, t->cap * 8, t->s->stream));
-    HIP_CHECK(hipMemsetAsync(t->counts, 0, t->cap * 8, t->s->stream));
+    if (with_counts) HIP_CHECK(hipMemsetAsync(t->counts, 0, t->cap * 8, t->s->stream));
     HIP_CHECK(hipMemsetAsync(t->cursor, 0, 8, t->s->stream));
     return GPUE_OK;
 }
@@ -1619,7 +1619,7 @@ __global__ void k_hash_agg_emit(const unsigned long long* __restrict__ slots,
         if (pos >= max_out) continue; // caller checks n_groups > max_out
         out_keys[pos] = slots[s];
         out_sums[pos] = (int64_t)sums[s];
-        if (out_counts) out_counts[pos] = (int64_t)counts[s];
+        if (out_counts != nullptr) out_counts[pos] = (int64_t)counts[s];
     }
 }
 
@@ -1722,18 +1722,39 @@ __global__ void k_bits_str16_eq(const ulonglong2* __restrict__ col, uint64_t n,
     }
 }
 
+// Each lane evaluates 4 consecutive orders (int4 loads); 8-lane groups pack
+// their 32 predicate bits into one word via shuffles — no atomics, each
+// 32-bit bitset word written exactly once.
 __global__ void k_q3_order_bits(const int32_t* __restrict__ ocust,
                                 const int32_t* __restrict__ odate, uint64_t n_orders,
                                 const uint32_t* __restrict__ cust_bits, int32_t cutoff,
                                 uint32_t* __restrict__ bits) {
+    uint64_t n4 = (n_orders + 3) / 4; // quads; caller sizes bits to whole words
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t o = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; o < n_orders;
-         o += stride) {
-        if (odate[o] < cutoff) {
-            uint32_t c = (uint32_t)ocust[o] - 1;
-            if ((cust_bits[c >> 5] >> (c & 31)) & 1u)
-                atomicOr(&bits[o >> 5], 1u << (o & 31));
+    int lane = threadIdx.x & (WAVE - 1);
+    // wave-uniform loop bound (q - lane == this wave's first quad) so every
+    // lane reaches the shuffles together
+    for (uint64_t q = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; q - lane < n4;
+         q += stride) {
+        uint64_t base = q * 4;
+        uint32_t b4 = 0;
+        if (q < n4) {
+            #pragma unroll
+            for (int j = 0; j < 4; j++) {
+                uint64_t o = base + j;
+                if (o < n_orders && odate[o] < cutoff) {
+                    uint32_t c = (uint32_t)ocust[o] - 1;
+                    b4 |= ((cust_bits[c >> 5] >> (c & 31)) & 1u) << j;
+                }
+            }
         }
+        uint32_t word = 0;
+        #pragma unroll
+        for (int k = 0; k < 8; k++)
+            word |= (uint32_t)__shfl((int)b4, (lane & ~7) + k, WAVE) << (4 * k);
+        // 8 quads = 32 orders = one bitset word, written exactly once by the
+        // group leader (group base quad is 8-aligned: blockDim % 64 == 0)
+        if (q < n4 && (lane & 7) == 0) bits[q / 8] = word;
     }
 }
 
@@ -1754,16 +1775,13 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
         if (!((order_bits[o >> 5] >> (o & 31)) & 1u)) continue;
         unsigned long long v = (unsigned long long)(ext[i] * (100 - disc[i]));
         uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        (void)counts; // COUNT is not in Q3's select list — one CAS + one add
         for (;;) {
             unsigned long long cur = slots[s];
-            if (cur == k) { atomicAdd(&sums[s], v); atomicAdd(&counts[s], 1ull); break; }
+            if (cur == k) { atomicAdd(&sums[s], v); break; }
             if (cur == AGG_EMPTY) {
                 unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
-                if (old == AGG_EMPTY || old == k) {
-                    atomicAdd(&sums[s], v);
-                    atomicAdd(&counts[s], 1ull);
-                    break;
-                }
+                if (old == AGG_EMPTY || old == k) { atomicAdd(&sums[s], v); break; }
             }
             s = (s + 1) & cap_mask;
         }
@@ -1860,7 +1878,7 @@ int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbu
                         gpue_dbuf* out_sums, uint64_t max_out, uint64_t* n_groups) {
     ARG_CHECK(s && lk && ext && disc && ship && order_bits && at && out_keys && out_sums &&
               n_groups);
-    int rc = agg_table_reset(at);
+    int rc = agg_table_reset(at, /*with_counts=*/false);
     if (rc != GPUE_OK) return rc;
     hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
                        (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
