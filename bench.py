@@ -203,10 +203,19 @@ def main():
     n_merge = {"q1": 2, "q21": 7000, "q43": 800, "q3": 0}[wl]
 
     # ---- untimed setup: generate shard on device, build step closure ----
+    # At N>1 the accumulator is torch-backed (wrap_ptr) so the phase1→phase2
+    # merge is ONE RCCL all-reduce on the kernel's own output buffer — no
+    # host round trip.
+    def make_acc(n_items):
+        if world > 1:
+            t = torch.zeros(n_items, dtype=torch.int64, device="cuda")
+            return eng.wrap_ptr(t.data_ptr(), n_items * 8), t
+        return eng.alloc(n_items * 8), None
+
     if wl == "q1":
         cols = [eng.alloc(rows * 4) for _ in range(3)]
         eng.gen_lineorder_q1(SEED, row_start, rows, *cols)
-        acc = eng.alloc(16)
+        acc, acc_t = make_acc(2)
 
         def kernel_only():
             eng.q1_join_sum_async(tables["dates"], cols[0], cols[1], cols[2], rows, acc)
@@ -217,7 +226,7 @@ def main():
     elif wl == "q21":
         cols = [eng.alloc(rows * 4) for _ in range(4)]
         eng.gen_lineorder_q21(SEED, row_start, rows, *cols)
-        acc = eng.alloc(7000 * 8)
+        acc, acc_t = make_acc(7000)
 
         def kernel_only():
             eng.q21_star_agg_async(tables["parts"], tables["supps"], tables["dates"],
@@ -259,7 +268,7 @@ def main():
             g = kernel_only()
             return np.array([g], np.int64)
     else:  # q43
-        acc = eng.alloc(800 * 8)
+        acc, acc_t = make_acc(800)
         if world == 1:
             cols = [eng.alloc(rows * 4) for _ in range(6)]
             eng.gen_lineorder_q43(SEED, row_start, rows, *cols)
@@ -319,19 +328,27 @@ def main():
     eng.sync()
     log(f"[bench] setup {time.perf_counter()-t_setup:.1f}s")
 
-    merge_buf = None
-    if world > 1:
-        merge_buf = torch.zeros(n_merge, dtype=torch.int64, device="cuda")
-
     def run_step():
-        vals = step()
         if dist is not None:
+            kernel_only() if wl != "q43" else step_partitioned_kernel()
+            eng.sync()
             # phase1 -> phase2 aggregate merge (agg_hash_variant.h merge_batch
-            # semantics) over RCCL
-            merge_buf.copy_(torch.from_numpy(np.asarray(vals)))
-            dist.all_reduce(merge_buf)
-            return merge_buf
-        return vals
+            # semantics): one RCCL all-reduce on the kernel's output buffer
+            dist.all_reduce(acc_t)
+            return acc_t.cpu().numpy()  # per-step result read
+        return step()
+
+    step_partitioned_kernel = None
+    if wl == "q43" and world > 1:
+        def step_partitioned_kernel():
+            eng.partition(cols[0], rows, world, ridx)
+            for c, s_ in zip(cols, send):
+                eng.gather_u32(c, ridx, rows, s_)
+            eng.sync()
+            for st, rt in zip(send_t, recv_t):
+                dist.all_to_all_single(rt[:n_recv], st, out_list, in_list)
+            torch.cuda.synchronize()
+            kernel_only()
 
     for _ in range(args.warmup):
         run_step()
