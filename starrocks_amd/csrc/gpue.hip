@@ -47,8 +47,14 @@ static constexpr int WAVE = 64;
 static constexpr uint64_t MAX_GRID = 2048; // Guideline 11: grid-stride past this
 
 static inline uint32_t grid_for(uint64_t n) {
+    static int override_ = -2; // GPUE_GRID env: experiment hook for block-count sweeps
+    if (override_ == -2) {
+        const char* e = getenv("GPUE_GRID");
+        override_ = e ? atoi(e) : -1;
+    }
+    uint64_t cap = override_ > 0 ? (uint64_t)override_ : MAX_GRID;
     uint64_t b = (n + BLOCK - 1) / BLOCK;
-    if (b > MAX_GRID) b = MAX_GRID;
+    if (b > cap) b = cap;
     if (b == 0) b = 1;
     return (uint32_t)b;
 }
@@ -1164,13 +1170,32 @@ k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
                       (unsigned long long)(int64_t)(&r4.x)[j]);
         }
     };
+    // 2-deep pipeline on the pk -> pbits dependency chain: the NEXT pair's
+    // pk quads are loaded before the current pair's probes, so the chain's
+    // two serial memory latencies overlap the current iteration's work
+    // (+8 VGPRs only — the other columns stay loaded at use).
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int4 pa = {}, pb_ = {};
+    bool have = i + stride < n4;
+    if (have) {
+        pa = pk4[i];
+        pb_ = pk4[i + stride];
+    }
     for (; i + stride < n4; i += 2 * stride) {
-        int4 pa = pk4[i], sa = sk4[i], oa = od4[i], ra = rv4[i];
+        uint64_t nx = i + 2 * stride;
+        int4 npa = {}, npb = {};
+        bool nhave = nx + stride < n4;
+        if (nhave) {
+            npa = pk4[nx];
+            npb = pk4[nx + stride];
+        }
         uint64_t i2 = i + stride;
-        int4 pb_ = pk4[i2], sb = sk4[i2], ob = od4[i2], rb = rv4[i2];
+        int4 sa = sk4[i], oa = od4[i], ra = rv4[i];
+        int4 sb = sk4[i2], ob = od4[i2], rb = rv4[i2];
         quad(pa, sa, oa, ra);
         quad(pb_, sb, ob, rb);
+        pa = npa;
+        pb_ = npb;
     }
     for (; i < n4; i += stride) quad(pk4[i], sk4[i], od4[i], rv4[i]);
     uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1789,12 +1814,25 @@ __global__ void k_q3_order_bits(const int32_t* __restrict__ ocust,
         uint64_t base = q * 4;
         uint32_t b4 = 0;
         if (q < n4) {
-            #pragma unroll
-            for (int j = 0; j < 4; j++) {
-                uint64_t o = base + j;
-                if (o < n_orders && odate[o] < cutoff) {
-                    uint32_t c = (uint32_t)ocust[o] - 1;
-                    b4 |= ((cust_bits[c >> 5] >> (c & 31)) & 1u) << j;
+            if (base + 4 <= n_orders) {
+                // 16 B/lane vector loads for the two streamed columns; the
+                // customer-bit gather is issued only for date-passing orders
+                int4 d4 = ((const int4*)odate)[q];
+                int4 c4 = ((const int4*)ocust)[q];
+                #pragma unroll
+                for (int j = 0; j < 4; j++) {
+                    if ((&d4.x)[j] < cutoff) {
+                        uint32_t c = (uint32_t)(&c4.x)[j] - 1;
+                        b4 |= ((cust_bits[c >> 5] >> (c & 31)) & 1u) << j;
+                    }
+                }
+            } else {
+                for (int j = 0; j < 4; j++) {
+                    uint64_t o = base + j;
+                    if (o < n_orders && odate[o] < cutoff) {
+                        uint32_t c = (uint32_t)ocust[o] - 1;
+                        b4 |= ((cust_bits[c >> 5] >> (c & 31)) & 1u) << j;
+                    }
                 }
             }
         }
