@@ -1170,32 +1170,15 @@ k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
                       (unsigned long long)(int64_t)(&r4.x)[j]);
         }
     };
-    // 2-deep pipeline on the pk -> pbits dependency chain: the NEXT pair's
-    // pk quads are loaded before the current pair's probes, so the chain's
-    // two serial memory latencies overlap the current iteration's work
-    // (+8 VGPRs only — the other columns stay loaded at use).
+    // (a 2-deep pk prefetch pipeline measured 9 % SLOWER here — the extra
+    // registers/branches cost more than the overlapped latency buys)
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    int4 pa = {}, pb_ = {};
-    bool have = i + stride < n4;
-    if (have) {
-        pa = pk4[i];
-        pb_ = pk4[i + stride];
-    }
     for (; i + stride < n4; i += 2 * stride) {
-        uint64_t nx = i + 2 * stride;
-        int4 npa = {}, npb = {};
-        bool nhave = nx + stride < n4;
-        if (nhave) {
-            npa = pk4[nx];
-            npb = pk4[nx + stride];
-        }
+        int4 pa = pk4[i], sa = sk4[i], oa = od4[i], ra = rv4[i];
         uint64_t i2 = i + stride;
-        int4 sa = sk4[i], oa = od4[i], ra = rv4[i];
-        int4 sb = sk4[i2], ob = od4[i2], rb = rv4[i2];
+        int4 pb_ = pk4[i2], sb = sk4[i2], ob = od4[i2], rb = rv4[i2];
         quad(pa, sa, oa, ra);
         quad(pb_, sb, ob, rb);
-        pa = npa;
-        pb_ = npb;
     }
     for (; i < n4; i += stride) quad(pk4[i], sk4[i], od4[i], rv4[i]);
     uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
