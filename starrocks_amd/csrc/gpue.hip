@@ -46,17 +46,35 @@ static constexpr int BLOCK = 256;
 static constexpr int WAVE = 64;
 static constexpr uint64_t MAX_GRID = 2048; // Guideline 11: grid-stride past this
 
-static inline uint32_t grid_for(uint64_t n) {
-    static int override_ = -2; // GPUE_GRID env: experiment hook for block-count sweeps
-    if (override_ == -2) {
-        const char* e = getenv("GPUE_GRID");
-        override_ = e ? atoi(e) : -1;
-    }
-    uint64_t cap = override_ > 0 ? (uint64_t)override_ : MAX_GRID;
+static uint32_t env_cap(const char* name, uint32_t dflt) {
+    const char* e = getenv(name);
+    int v = e ? atoi(e) : 0;
+    return v > 0 ? (uint32_t)v : dflt;
+}
+
+static inline uint32_t grid_capped(uint64_t n, uint64_t cap) {
     uint64_t b = (n + BLOCK - 1) / BLOCK;
     if (b > cap) b = cap;
     if (b == 0) b = 1;
     return (uint32_t)b;
+}
+
+// latency-bound kernels (chain walks, hash inserts): many blocks for wave
+// occupancy (GPUE_GRID env overrides for experiments)
+static inline uint32_t grid_for(uint64_t n) {
+    static uint32_t cap = 0;
+    if (!cap) cap = env_cap("GPUE_GRID", MAX_GRID);
+    return grid_capped(n, cap);
+}
+
+// streaming-bound kernels: measured FASTEST at ~1 block/CU (256 blocks) —
+// the q1 grid sweep (profiles/r01 q1_grid_sweep) runs 5205 GB/s @SF10 /
+// 6342 GB/s @600 M rows at 256 blocks vs 2819/4895 at the 2048-block cap:
+// fewer, longer-lived blocks stream long contiguous ranges per wave
+static inline uint32_t grid_stream(uint64_t n) {
+    static uint32_t cap = 0;
+    if (!cap) cap = env_cap("GPUE_GRID_STREAM", 256);
+    return grid_capped(n, cap);
 }
 
 // ---------------------------------------------------------------------------
@@ -316,7 +334,7 @@ static int ensure_datekey(gpue_session* s) {
 int gpue_gen_u32_mod(gpue_session* s, gpue_dbuf* out, uint64_t seed, uint64_t tag,
                      uint64_t row_start, uint64_t n, uint32_t mod, uint32_t add) {
     ARG_CHECK(s && out && out->bytes >= n * 4);
-    hipLaunchKernelGGL(k_gen_u32_mod, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_gen_u32_mod, dim3(grid_stream(n)), dim3(BLOCK), 0, s->stream,
                        (uint32_t*)out->ptr, seed, tag, row_start, n, mod, add);
     HIP_CHECK(hipGetLastError());
     return GPUE_OK;
@@ -325,7 +343,7 @@ int gpue_gen_u32_mod(gpue_session* s, gpue_dbuf* out, uint64_t seed, uint64_t ta
 int gpue_gen_i64(gpue_session* s, gpue_dbuf* out, uint64_t seed, uint64_t tag,
                  uint64_t row_start, uint64_t n) {
     ARG_CHECK(s && out && out->bytes >= n * 8);
-    hipLaunchKernelGGL(k_gen_i64, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_gen_i64, dim3(grid_stream(n)), dim3(BLOCK), 0, s->stream,
                        (int64_t*)out->ptr, seed, tag, row_start, n);
     HIP_CHECK(hipGetLastError());
     return GPUE_OK;
@@ -336,7 +354,7 @@ int gpue_gen_lineorder_q1(gpue_session* s, uint64_t seed, uint64_t row_start, ui
     ARG_CHECK(s && od && ep && dc && od->bytes >= n * 4 && ep->bytes >= n * 4 && dc->bytes >= n * 4);
     int rc = ensure_datekey(s);
     if (rc != GPUE_OK) return rc;
-    hipLaunchKernelGGL(k_gen_lineorder_q1, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_gen_lineorder_q1, dim3(grid_stream(n)), dim3(BLOCK), 0, s->stream,
                        (int32_t*)od->ptr, (int32_t*)ep->ptr, (int32_t*)dc->ptr,
                        s->d_datekey, seed, row_start, n);
     HIP_CHECK(hipGetLastError());
@@ -354,7 +372,7 @@ int gpue_gen_lineorder_q43(gpue_session* s, uint64_t seed, uint64_t row_start, u
               od->bytes >= n * 4 && rv->bytes >= n * 4 && sc->bytes >= n * 4);
     int rc = ensure_datekey(s);
     if (rc != GPUE_OK) return rc;
-    hipLaunchKernelGGL(k_gen_lineorder_q43, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_gen_lineorder_q43, dim3(grid_stream(n)), dim3(BLOCK), 0, s->stream,
                        (int32_t*)ck->ptr, (int32_t*)sk->ptr, (int32_t*)pk->ptr,
                        (int32_t*)od->ptr, (int32_t*)rv->ptr, (int32_t*)sc->ptr,
                        s->d_datekey, seed, row_start, n);
@@ -368,7 +386,7 @@ int gpue_gen_lineorder_q21(gpue_session* s, uint64_t seed, uint64_t row_start, u
               od->bytes >= n * 4 && rv->bytes >= n * 4);
     int rc = ensure_datekey(s);
     if (rc != GPUE_OK) return rc;
-    hipLaunchKernelGGL(k_gen_lineorder_q21, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_gen_lineorder_q21, dim3(grid_stream(n)), dim3(BLOCK), 0, s->stream,
                        (int32_t*)pk->ptr, (int32_t*)sk->ptr, (int32_t*)od->ptr,
                        (int32_t*)rv->ptr, s->d_datekey, seed, row_start, n);
     HIP_CHECK(hipGetLastError());
@@ -1069,11 +1087,11 @@ static void launch_q1(gpue_session* s, gpue_join_table* dates, const int32_t* od
     uint64_t set_interval = (uint64_t)(dates->set_max - dates->set_min + 1);
     uint64_t nwords = (set_interval + 31) / 32;
     if (dates->bitset && dates->set_max >= dates->set_min && nwords * 4 <= 32768) {
-        hipLaunchKernelGGL(k_q1_join_sum_bitset, dim3(grid_for(n)), dim3(BLOCK),
+        hipLaunchKernelGGL(k_q1_join_sum_bitset, dim3(grid_stream(n)), dim3(BLOCK),
                            nwords * 4, s->stream, od, ep, dc, n, dates->set_min,
                            dates->set_max, dates->bitset, sum_out, cnt_out);
     } else {
-        hipLaunchKernelGGL(k_q1_join_sum, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+        hipLaunchKernelGGL(k_q1_join_sum, dim3(grid_stream(n)), dim3(BLOCK), 0, s->stream,
                            od, ep, dc, n, dates->min_key, dates->max_key, dates->first,
                            sum_out, cnt_out);
     }
@@ -1208,7 +1226,7 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
     ARG_CHECK(parts->min_key == 1 && supps->min_key == 1);
     ARG_CHECK(parts->first16 && supps->first16 && dates->first16);
     HIP_CHECK(hipMemsetAsync(group_sums->ptr, 0, NG_Q21 * sizeof(int64_t), s->stream));
-    hipLaunchKernelGGL(k_q21_star_agg, dim3(512), dim3(BLOCK_Q21), 0, s->stream,
+    hipLaunchKernelGGL(k_q21_star_agg, dim3(env_cap("GPUE_GRID_WIDE", 512)), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)od->ptr, (const int32_t*)rv->ptr, n,
                        parts->bitset, parts->set_min,
@@ -1230,7 +1248,7 @@ int gpue_q21_star_agg(gpue_session* s, gpue_join_table* parts, gpue_join_table* 
     unsigned long long* d_g = nullptr;
     HIP_CHECK(hipMalloc(&d_g, NG_Q21 * sizeof(unsigned long long)));
     HIP_CHECK(hipMemsetAsync(d_g, 0, NG_Q21 * sizeof(unsigned long long), s->stream));
-    hipLaunchKernelGGL(k_q21_star_agg, dim3(512), dim3(BLOCK_Q21), 0, s->stream,
+    hipLaunchKernelGGL(k_q21_star_agg, dim3(env_cap("GPUE_GRID_WIDE", 512)), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)od->ptr, (const int32_t*)rv->ptr, n,
                        parts->bitset, parts->set_min,
@@ -1348,7 +1366,7 @@ int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs, gpue_join_t
     ARG_CHECK(custs->bitset && supps->bitset && parts->bitset);
     ARG_CHECK(custs->min_key == 1 && supps->min_key == 1 && parts->min_key == 1);
     HIP_CHECK(hipMemsetAsync(group_sums->ptr, 0, NG_Q43 * sizeof(int64_t), s->stream));
-    hipLaunchKernelGGL(k_q43_star_agg, dim3(512), dim3(BLOCK_Q21), 0, s->stream,
+    hipLaunchKernelGGL(k_q43_star_agg, dim3(env_cap("GPUE_GRID_WIDE", 512)), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)ck->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)pk->ptr, (const int32_t*)od->ptr,
                        (const int32_t*)rv->ptr, (const int32_t*)sc->ptr, n,
@@ -1504,11 +1522,11 @@ int gpue_ubench(gpue_session* s, int which, gpue_dbuf* a, gpue_dbuf* b, gpue_dbu
         int iters = pass == 0 ? 1 : reps;
         for (int r = 0; r < iters; r++) {
             if (which == 0) {
-                hipLaunchKernelGGL(k_ub_sum1, dim3(grid_for(n4)), dim3(BLOCK), 0, s->stream,
+                hipLaunchKernelGGL(k_ub_sum1, dim3(grid_stream(n4)), dim3(BLOCK), 0, s->stream,
                                    (const int4*)a->ptr, n4, d_out);
             } else {
                 ARG_CHECK(b && c);
-                hipLaunchKernelGGL(k_ub_sum3, dim3(grid_for(n4)), dim3(BLOCK), 0, s->stream,
+                hipLaunchKernelGGL(k_ub_sum3, dim3(grid_stream(n4)), dim3(BLOCK), 0, s->stream,
                                    (const int4*)a->ptr, (const int4*)b->ptr,
                                    (const int4*)c->ptr, n4, d_out);
             }
@@ -1884,7 +1902,7 @@ int gpue_gen_lineitem_q3(gpue_session* s, uint64_t seed, uint64_t row_start, uin
               ship->bytes >= n * 4);
     int rc = ensure_datekey(s);
     if (rc != GPUE_OK) return rc;
-    hipLaunchKernelGGL(k_gen_lineitem_q3, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_gen_lineitem_q3, dim3(grid_stream(n)), dim3(BLOCK), 0, s->stream,
                        (int64_t*)lk->ptr, (int64_t*)ext->ptr, (int64_t*)disc->ptr,
                        (int32_t*)ship->ptr, s->d_datekey, seed, row_start, n, n_orders);
     HIP_CHECK(hipGetLastError());
@@ -1897,7 +1915,7 @@ int gpue_gen_orders_q3(gpue_session* s, uint64_t seed, uint64_t n_orders, uint32
               odate->bytes >= n_orders * 4);
     int rc = ensure_datekey(s);
     if (rc != GPUE_OK) return rc;
-    hipLaunchKernelGGL(k_gen_orders_q3, dim3(grid_for(n_orders)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_gen_orders_q3, dim3(grid_stream(n_orders)), dim3(BLOCK), 0, s->stream,
                        (int32_t*)ocust->ptr, (int32_t*)odate->ptr, s->d_datekey, seed,
                        n_orders, n_custs);
     HIP_CHECK(hipGetLastError());
@@ -1906,7 +1924,7 @@ int gpue_gen_orders_q3(gpue_session* s, uint64_t seed, uint64_t n_orders, uint32
 
 int gpue_gen_cust_mkt16(gpue_session* s, uint64_t seed, uint32_t n_custs, gpue_dbuf* out) {
     ARG_CHECK(s && out && out->bytes >= (uint64_t)n_custs * 16);
-    hipLaunchKernelGGL(k_gen_cust_mkt16, dim3(grid_for(n_custs)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_gen_cust_mkt16, dim3(grid_stream(n_custs)), dim3(BLOCK), 0, s->stream,
                        (uint8_t*)out->ptr, seed, n_custs);
     HIP_CHECK(hipGetLastError());
     return GPUE_OK;
@@ -1920,7 +1938,7 @@ int gpue_bits_str16_eq(gpue_session* s, gpue_dbuf* col16, uint64_t n, const void
     memcpy(&la, lit16, 8);
     memcpy(&lb, (const char*)lit16 + 8, 8);
     HIP_CHECK(hipMemsetAsync(bits->ptr, 0, (n + 31) / 32 * 4, s->stream));
-    hipLaunchKernelGGL(k_bits_str16_eq, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_bits_str16_eq, dim3(grid_stream(n)), dim3(BLOCK), 0, s->stream,
                        (const ulonglong2*)col16->ptr, n, la, lb, (uint32_t*)bits->ptr);
     HIP_CHECK(hipGetLastError());
     return GPUE_OK;
@@ -1931,7 +1949,7 @@ int gpue_q3_order_bits(gpue_session* s, gpue_dbuf* ocust, gpue_dbuf* odate, uint
     ARG_CHECK(s && ocust && odate && cust_bits && order_bits);
     ARG_CHECK(order_bits->bytes >= (n_orders + 31) / 32 * 4);
     HIP_CHECK(hipMemsetAsync(order_bits->ptr, 0, (n_orders + 31) / 32 * 4, s->stream));
-    hipLaunchKernelGGL(k_q3_order_bits, dim3(grid_for(n_orders)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_q3_order_bits, dim3(grid_stream(n_orders)), dim3(BLOCK), 0, s->stream,
                        (const int32_t*)ocust->ptr, (const int32_t*)odate->ptr, n_orders,
                        (const uint32_t*)cust_bits->ptr, cutoff, (uint32_t*)order_bits->ptr);
     HIP_CHECK(hipGetLastError());
@@ -1951,7 +1969,7 @@ int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbu
               n_groups);
     int rc = agg_table_reset(at, /*with_counts=*/false);
     if (rc != GPUE_OK) return rc;
-    hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))), dim3(BLOCK), 0, s->stream,
                        (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
                        (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
                        (const uint32_t*)order_bits->ptr, ship_cutoff, at->slots, at->sums,
@@ -1989,7 +2007,7 @@ int gpue_q3_probe_agg(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf*
     HIP_CHECK(hipMemsetAsync(d_sums, 0, cap * 8, s->stream));
     HIP_CHECK(hipMemsetAsync(d_counts, 0, cap * 8, s->stream));
     HIP_CHECK(hipMemsetAsync(d_cursor, 0, 8, s->stream));
-    hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))), dim3(BLOCK), 0, s->stream,
                        (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
                        (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
                        (const uint32_t*)order_bits->ptr, ship_cutoff, d_slots, d_sums,
