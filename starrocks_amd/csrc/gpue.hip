@@ -1949,7 +1949,7 @@ int gpue_q3_order_bits(gpue_session* s, gpue_dbuf* ocust, gpue_dbuf* odate, uint
     ARG_CHECK(s && ocust && odate && cust_bits && order_bits);
     ARG_CHECK(order_bits->bytes >= (n_orders + 31) / 32 * 4);
     HIP_CHECK(hipMemsetAsync(order_bits->ptr, 0, (n_orders + 31) / 32 * 4, s->stream));
-    hipLaunchKernelGGL(k_q3_order_bits, dim3(grid_stream(n_orders)), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(k_q3_order_bits, dim3(grid_for(n_orders)), dim3(BLOCK), 0, s->stream,
                        (const int32_t*)ocust->ptr, (const int32_t*)odate->ptr, n_orders,
                        (const uint32_t*)cust_bits->ptr, cutoff, (uint32_t*)order_bits->ptr);
     HIP_CHECK(hipGetLastError());
