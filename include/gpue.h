@@ -105,6 +105,11 @@ int gpue_join_build_payload_i32(gpue_session* s, gpue_dbuf* keys /*i32*/,
  * first/next chains; probe compares build keys along the chain. */
 int gpue_join_build_bucket_chained_u32(gpue_session* s, gpue_dbuf* keys /*u32, 1-based*/,
                                        uint64_t row_count, gpue_join_table** out);
+/* LINEAR_CHAINED variant — the selector's preferred method under its
+ * 16M-bucket cap (join_hash_map_method.h:118-150): 8-bit fingerprint packed
+ * in first[], linear probing, same-key chains via next[]. */
+int gpue_join_build_linear_chained_u32(gpue_session* s, gpue_dbuf* keys /*u32, 1-based*/,
+                                       uint64_t row_count, gpue_join_table** out);
 /* Row-index variant: first/next chain structure exactly as the reference
  * builds it (chain order under duplicate keys is scatter-order, which on GPU
  * is nondeterministic — the emitted match multiset is identical). */

@@ -114,7 +114,8 @@ struct gpue_join_table {
     uint64_t row_count = 0;
     // method discriminator — the GPU analog of JoinHashMapSelector's choice
     // (reference join_hash_table.cpp:164-344)
-    enum Kind { PAYLOAD = 0, RANGE_DIRECT = 1, BUCKET_CHAINED = 2 } kind = PAYLOAD;
+    enum Kind { PAYLOAD = 0, RANGE_DIRECT = 1, BUCKET_CHAINED = 2,
+                LINEAR_CHAINED = 3 } kind = PAYLOAD;
     uint32_t log_bucket_size = 0;
     uint32_t* build_keys = nullptr; // chained methods keep the build keys for the
                                     // probe-side equality check (1-based, row 0 sentinel)
@@ -770,6 +771,148 @@ int gpue_join_build_bucket_chained_u32(gpue_session* s, gpue_dbuf* keys, uint64_
     return GPUE_OK;
 }
 
+// ---------------------------------------------------------------------------
+// LINEAR_CHAINED — the selector's preferred method under the 16 M-bucket cap
+// (reference join_hash_map_method.h:118-150, .hpp:125-368): hash in space
+// bucket<<8, 8-bit fingerprint packed in the top byte of first[], linear
+// probing with triangular increment, same-key rows chained via next[].
+// Parallel build claims slots with CAS; concurrent colliding inserts may
+// land keys at different probe offsets than the CPU's sequential order, and
+// chain order is scatter-order — probes follow the same (fp,key) search so
+// the match multiset is identical (SURVEY.md §7 hard part (b)).
+// ---------------------------------------------------------------------------
+static constexpr uint32_t LC_FP_BITS = 8u;
+static constexpr uint32_t LC_DATA_MASK = 0x00FFFFFFu;
+
+__global__ void k_build_linear_chained(const uint32_t* __restrict__ keys, uint64_t row_count,
+                                       uint32_t log_bucket_size, uint32_t bucket_mask,
+                                       uint32_t* __restrict__ first,
+                                       uint32_t* __restrict__ next) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i <= row_count;
+         i += stride) {
+        uint32_t k = keys[i];
+        uint32_t hash = join_hash_u32(k, log_bucket_size + LC_FP_BITS);
+        uint32_t fp = hash << (32 - LC_FP_BITS);
+        uint32_t b = hash >> LC_FP_BITS;
+        uint32_t probe_times = 1;
+        for (;;) {
+            uint32_t cur = first[b];
+            if (cur == 0) {
+                next[i] = 0;
+                uint32_t old = atomicCAS(&first[b], 0u, fp | (uint32_t)i);
+                if (old == 0) break;
+                cur = old; // someone claimed it — fall through to compare
+            }
+            if ((cur & ~LC_DATA_MASK) == fp && keys[cur & LC_DATA_MASK] == k) {
+                // same key: push-front onto its chain
+                uint32_t old = cur;
+                for (;;) {
+                    next[i] = old & LC_DATA_MASK;
+                    uint32_t seen = atomicCAS(&first[b], old, fp | (uint32_t)i);
+                    if (seen == old) break;
+                    old = seen; // head moved (same key/fp by construction)
+                }
+                break;
+            }
+            b = (b + probe_times) & bucket_mask;
+            probe_times++;
+        }
+    }
+}
+
+extern "C" int gpue_join_build_linear_chained_u32(gpue_session* s, gpue_dbuf* keys,
+                                                  uint64_t row_count, gpue_join_table** out);
+int gpue_join_build_linear_chained_u32(gpue_session* s, gpue_dbuf* keys, uint64_t row_count,
+                                       gpue_join_table** out) {
+    ARG_CHECK(s && keys && out && row_count > 0 && row_count < LC_DATA_MASK);
+    ARG_CHECK(keys->bytes >= (row_count + 1) * 4);
+    gpue_join_table* t = new gpue_join_table();
+    t->s = s;
+    t->kind = gpue_join_table::LINEAR_CHAINED;
+    t->row_count = row_count;
+    t->bucket_size = calc_bucket_size((uint32_t)(row_count + 1));
+    ARG_CHECK(t->bucket_size <= LC_DATA_MASK); // the 16M cap (join_hash_map_method.h:136)
+    t->log_bucket_size = (uint32_t)__builtin_ctzll(t->bucket_size);
+    HIP_CHECK(hipMalloc(&t->first, t->bucket_size * sizeof(uint32_t)));
+    HIP_CHECK(hipMalloc(&t->next, (row_count + 1) * sizeof(uint32_t)));
+    HIP_CHECK(hipMalloc(&t->build_keys, (row_count + 1) * sizeof(uint32_t)));
+    HIP_CHECK(hipMemsetAsync(t->first, 0, t->bucket_size * sizeof(uint32_t), s->stream));
+    HIP_CHECK(hipMemsetAsync(t->next, 0, (row_count + 1) * sizeof(uint32_t), s->stream));
+    HIP_CHECK(hipMemcpyAsync(t->build_keys, keys->ptr, (row_count + 1) * sizeof(uint32_t),
+                             hipMemcpyDeviceToDevice, s->stream));
+    hipLaunchKernelGGL(k_build_linear_chained, dim3(grid_for(row_count)), dim3(BLOCK), 0,
+                       s->stream, t->build_keys, row_count, t->log_bucket_size,
+                       (uint32_t)(t->bucket_size - 1), t->first, t->next);
+    HIP_CHECK(hipGetLastError());
+    *out = t;
+    return GPUE_OK;
+}
+
+// linear-probe lookup to the chain head, then chain walk (lookup_init,
+// join_hash_map_method.hpp:286-368): chains hold identical keys, so the walk
+// needs no further compare
+__device__ static inline uint32_t lc_lookup_head(uint32_t k, uint32_t log_bucket_size,
+                                                 uint32_t bucket_mask,
+                                                 const uint32_t* __restrict__ first,
+                                                 const uint32_t* __restrict__ build_keys) {
+    uint32_t hash = join_hash_u32(k, log_bucket_size + LC_FP_BITS);
+    uint32_t fp = hash << (32 - LC_FP_BITS);
+    uint32_t b = hash >> LC_FP_BITS;
+    uint32_t probe_times = 1;
+    for (;;) {
+        uint32_t cur = first[b];
+        if (cur == 0) return 0;
+        if ((cur & ~LC_DATA_MASK) == fp && build_keys[cur & LC_DATA_MASK] == k)
+            return cur & LC_DATA_MASK;
+        b = (b + probe_times) & bucket_mask;
+        probe_times++;
+    }
+}
+
+__global__ void k_probe_count_lc(const uint32_t* __restrict__ probe_keys, uint64_t n,
+                                 uint32_t log_bucket_size, uint32_t bucket_mask,
+                                 const uint32_t* __restrict__ first,
+                                 const uint32_t* __restrict__ next,
+                                 const uint32_t* __restrict__ build_keys,
+                                 uint32_t* __restrict__ row_counts) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint32_t b = lc_lookup_head(probe_keys[i], log_bucket_size, bucket_mask, first,
+                                    build_keys);
+        uint32_t c = 0;
+        while (b != 0) {
+            c++;
+            b = next[b];
+        }
+        row_counts[i] = c;
+    }
+}
+
+__global__ void k_probe_emit_lc(const uint32_t* __restrict__ probe_keys, uint64_t n,
+                                uint32_t log_bucket_size, uint32_t bucket_mask,
+                                const uint32_t* __restrict__ first,
+                                const uint32_t* __restrict__ next,
+                                const uint32_t* __restrict__ build_keys,
+                                const uint32_t* __restrict__ row_counts,
+                                const uint64_t* __restrict__ row_offsets,
+                                uint32_t* __restrict__ out_probe,
+                                uint32_t* __restrict__ out_build) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        if (row_counts[i] == 0) continue;
+        uint64_t pos = row_offsets[i];
+        uint32_t b = lc_lookup_head(probe_keys[i], log_bucket_size, bucket_mask, first,
+                                    build_keys);
+        while (b != 0) {
+            out_probe[pos] = (uint32_t)i;
+            out_build[pos] = b;
+            pos++;
+            b = next[b];
+        }
+    }
+}
+
 // probe with key-equality compare along the chain (lookup_init +
 // _probe_from_ht with AreKeysInChainIdentical == false)
 __global__ void k_probe_count_bc(const uint32_t* __restrict__ probe_keys, uint64_t n,
@@ -911,7 +1054,12 @@ int gpue_join_probe_emit_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* pro
     uint64_t* d_offsets = nullptr;
     HIP_CHECK(hipMalloc(&d_counts, n_rows * sizeof(uint32_t)));
     HIP_CHECK(hipMalloc(&d_bsums, (nb + 1) * sizeof(uint64_t)));
-    if (t->kind == gpue_join_table::BUCKET_CHAINED) {
+    if (t->kind == gpue_join_table::LINEAR_CHAINED) {
+        hipLaunchKernelGGL(k_probe_count_lc, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const uint32_t*)probe_keys->ptr, n_rows, t->log_bucket_size,
+                           (uint32_t)(t->bucket_size - 1), t->first, t->next, t->build_keys,
+                           d_counts);
+    } else if (t->kind == gpue_join_table::BUCKET_CHAINED) {
         hipLaunchKernelGGL(k_probe_count_bc, dim3(nb), dim3(BLOCK), 0, s->stream,
                            (const uint32_t*)probe_keys->ptr, n_rows, t->log_bucket_size,
                            t->first, t->next, t->build_keys, d_counts);
@@ -932,7 +1080,13 @@ int gpue_join_probe_emit_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* pro
         HIP_CHECK(hipMalloc(&d_offsets, n_rows * sizeof(uint64_t)));
         hipLaunchKernelGGL(k_scan_offsets, dim3(nb), dim3(BLOCK), 0, s->stream,
                            d_counts, n_rows, tile, d_bsums, d_offsets);
-        if (t->kind == gpue_join_table::BUCKET_CHAINED) {
+        if (t->kind == gpue_join_table::LINEAR_CHAINED) {
+            hipLaunchKernelGGL(k_probe_emit_lc, dim3(nb), dim3(BLOCK), 0, s->stream,
+                               (const uint32_t*)probe_keys->ptr, n_rows, t->log_bucket_size,
+                               (uint32_t)(t->bucket_size - 1), t->first, t->next,
+                               t->build_keys, d_counts, d_offsets,
+                               (uint32_t*)out_probe_idx->ptr, (uint32_t*)out_build_idx->ptr);
+        } else if (t->kind == gpue_join_table::BUCKET_CHAINED) {
             hipLaunchKernelGGL(k_probe_emit_bc, dim3(nb), dim3(BLOCK), 0, s->stream,
                                (const uint32_t*)probe_keys->ptr, n_rows, t->log_bucket_size,
                                t->first, t->next, t->build_keys, d_counts, d_offsets,

@@ -72,6 +72,7 @@ def _declare(lib):
         "gpue_join_build_payload_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_build_range_direct_i32": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_build_bucket_chained_u32": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
+        "gpue_join_build_linear_chained_u32": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_table_destroy": (None, [c_vp]),
         "gpue_join_table_minmax": (c_i32, [c_vp, ctypes.POINTER(c_i64), ctypes.POINTER(c_i64)]),
         "gpue_join_table_first_d2h": (c_i32, [c_vp, c_vp, c_u64]),
@@ -232,6 +233,12 @@ class Engine:
     def join_build_bucket_chained(self, keys: DBuf, row_count) -> JoinTable:
         h = c_vp()
         _ck(self._lib, self._lib.gpue_join_build_bucket_chained_u32(
+            self._h, keys._h, row_count, ctypes.byref(h)))
+        return JoinTable(self, h)
+
+    def join_build_linear_chained(self, keys: DBuf, row_count) -> JoinTable:
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_join_build_linear_chained_u32(
             self._h, keys._h, row_count, ctypes.byref(h)))
         return JoinTable(self, h)
 

@@ -132,6 +132,40 @@ def test_bucket_chained_join_parity(engine):
     t.destroy()
 
 
+def test_linear_chained_join_parity(engine):
+    """LINEAR_CHAINED (fp-packed, linear probing) vs the oracle restatement:
+    slot placement under concurrent collisions may differ from sequential
+    insertion, but the probe follows the same (fp,key) search — the match
+    multiset is identical."""
+    rng = np.random.default_rng(31)
+    build_keys = np.concatenate([[0], rng.integers(0, 2**30, 80_000)]).astype(np.uint32)
+    probe_keys = np.concatenate([build_keys[1:],
+                                 rng.integers(0, 2**30, 40_000).astype(np.uint32)])
+    rng.shuffle(probe_keys)
+    kb = engine.alloc(build_keys.nbytes)
+    kb.h2d(build_keys.view(np.int32))
+    t = engine.join_build_linear_chained(kb, len(build_keys) - 1)
+    pb = engine.alloc(probe_keys.nbytes)
+    pb.h2d(probe_keys.view(np.int32))
+    cnt = engine.join_probe_emit(t, pb, len(probe_keys))
+    op_buf = engine.alloc(max(cnt, 1) * 4)
+    ob_buf = engine.alloc(max(cnt, 1) * 4)
+    engine.join_probe_emit(t, pb, len(probe_keys), op_buf, ob_buf)
+    gop = op_buf.d2h(np.uint32, cnt)
+    gob = ob_buf.d2h(np.uint32, cnt)
+
+    lf, ln, ls, ll = orc.linear_chained_build(build_keys)
+    heads = orc.linear_chained_lookup(build_keys, probe_keys, lf, ls, ll)
+    eop, eob = orc.probe_emit(build_keys, ln, probe_keys, heads)
+    assert cnt == len(eop)
+    got = np.sort(gop.astype(np.uint64) << np.uint64(32) | gob.astype(np.uint64))
+    exp = np.sort(eop.astype(np.uint64) << np.uint64(32) | eob.astype(np.uint64))
+    assert np.array_equal(got, exp)
+    for b in (kb, pb, op_buf, ob_buf):
+        b.free()
+    t.destroy()
+
+
 def _build_date_table(engine, year_filter):
     datekey, dyear = gen.gen_dates()
     mn, mx, first = gen.build_date_dim_payload(year_filter)
