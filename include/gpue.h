@@ -227,6 +227,18 @@ int gpue_hash_agg_sum_u64(gpue_session* s, gpue_dbuf* keys /*u64*/, gpue_dbuf* v
                           gpue_dbuf* out_sums, gpue_dbuf* out_counts /*nullable*/,
                           uint64_t max_out, uint64_t* n_groups);
 
+/* Full aggregate-function states: SUM + COUNT + MIN + MAX (AVG = SUM/COUNT
+ * at finalize, reference exprs/agg/aggregate.h:136-269). */
+int gpue_hash_agg_stats_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uint64_t n,
+                            uint64_t capacity_hint, gpue_dbuf* out_keys, gpue_dbuf* out_sums,
+                            gpue_dbuf* out_counts, gpue_dbuf* out_mins, gpue_dbuf* out_maxs,
+                            uint64_t max_out, uint64_t* n_groups);
+/* Decimal SUM widened to int128 (exprs/agg/sum.h:181): lo/hi pair with an
+ * explicit carry, exact mod 2^128, order-independent. */
+int gpue_hash_agg_sum128_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uint64_t n,
+                             uint64_t capacity_hint, gpue_dbuf* out_keys, gpue_dbuf* out_lo,
+                             gpue_dbuf* out_hi, uint64_t max_out, uint64_t* n_groups);
+
 /* Row gather by index — the exchange sink's add_rows_selective analog
  * (exchange_sink_operator.cpp:670): materializes per-channel row slices at
  * the counting-sorted indexes before the RCCL all-to-all. */

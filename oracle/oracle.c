@@ -836,6 +836,85 @@ uint64_t orc_hash_agg_sum_u64(const uint64_t* keys, const int64_t* vals, uint64_
     return g;
 }
 
+/* Full aggregate-function state set (SUM/COUNT/MIN/MAX, exprs/agg/) and
+ * decimal SUM widening to int128 (exprs/agg/sum.h:181). */
+uint64_t orc_hash_agg_stats_u64(const uint64_t* keys, const int64_t* vals, uint64_t n,
+                                uint64_t* out_keys, int64_t* out_sums, int64_t* out_counts,
+                                int64_t* out_mins, int64_t* out_maxs, uint64_t max_out) {
+    uint64_t cap = 16;
+    while (cap < n * 2) cap <<= 1;
+    uint64_t* slots = (uint64_t*)malloc(cap * 8);
+    int64_t* sums = (int64_t*)calloc(cap, 8);
+    int64_t* counts = (int64_t*)calloc(cap, 8);
+    int64_t* mins = (int64_t*)malloc(cap * 8);
+    int64_t* maxs = (int64_t*)malloc(cap * 8);
+    memset(slots, 0xFF, cap * 8);
+    for (uint64_t i = 0; i < cap; i++) { mins[i] = INT64_MAX; maxs[i] = INT64_MIN; }
+    uint32_t mask = (uint32_t)(cap - 1);
+    for (uint64_t i = 0; i < n; i++) {
+        uint64_t k = keys[i];
+        uint32_t s = agg_hash_u64(k, mask);
+        for (;;) {
+            if (slots[s] == AGG_EMPTY) slots[s] = k;
+            if (slots[s] == k) {
+                sums[s] += vals[i];
+                counts[s]++;
+                if (vals[i] < mins[s]) mins[s] = vals[i];
+                if (vals[i] > maxs[s]) maxs[s] = vals[i];
+                break;
+            }
+            s = (s + 1) & mask;
+        }
+    }
+    uint64_t g = 0;
+    for (uint64_t s = 0; s < cap; s++) {
+        if (slots[s] == AGG_EMPTY) continue;
+        if (g >= max_out) { g = UINT64_MAX; break; }
+        out_keys[g] = slots[s];
+        out_sums[g] = sums[s];
+        out_counts[g] = counts[s];
+        out_mins[g] = mins[s];
+        out_maxs[g] = maxs[s];
+        g++;
+    }
+    free(slots); free(sums); free(counts); free(mins); free(maxs);
+    return g;
+}
+
+uint64_t orc_hash_agg_sum128_u64(const uint64_t* keys, const int64_t* vals, uint64_t n,
+                                 uint64_t* out_keys, uint64_t* out_lo, int64_t* out_hi,
+                                 uint64_t max_out) {
+    uint64_t cap = 16;
+    while (cap < n * 2) cap <<= 1;
+    uint64_t* slots = (uint64_t*)malloc(cap * 8);
+    __int128* sums = (__int128*)calloc(cap, 16);
+    memset(slots, 0xFF, cap * 8);
+    uint32_t mask = (uint32_t)(cap - 1);
+    for (uint64_t i = 0; i < n; i++) {
+        uint64_t k = keys[i];
+        uint32_t s = agg_hash_u64(k, mask);
+        for (;;) {
+            if (slots[s] == AGG_EMPTY) slots[s] = k;
+            if (slots[s] == k) {
+                sums[s] += (__int128)vals[i];
+                break;
+            }
+            s = (s + 1) & mask;
+        }
+    }
+    uint64_t g = 0;
+    for (uint64_t s = 0; s < cap; s++) {
+        if (slots[s] == AGG_EMPTY) continue;
+        if (g >= max_out) { g = UINT64_MAX; break; }
+        out_keys[g] = slots[s];
+        out_lo[g] = (uint64_t)(unsigned __int128)sums[s];
+        out_hi[g] = (int64_t)(sums[s] >> 64);
+        g++;
+    }
+    free(slots); free(sums);
+    return g;
+}
+
 /* Compute-only pipeline legs (columns pre-generated by the caller) — these
  * are what bench.py's cpu_baseline TIMES, so data generation stays outside
  * the measured region on both CPU and GPU. */

@@ -175,6 +175,13 @@ uint64_t orc_hash_agg_sum_u64(const uint64_t* keys, const int64_t* vals, uint64_
                               uint64_t* out_keys, int64_t* out_sums, int64_t* out_counts,
                               uint64_t max_out);
 
+uint64_t orc_hash_agg_stats_u64(const uint64_t* keys, const int64_t* vals, uint64_t n,
+                                uint64_t* out_keys, int64_t* out_sums, int64_t* out_counts,
+                                int64_t* out_mins, int64_t* out_maxs, uint64_t max_out);
+uint64_t orc_hash_agg_sum128_u64(const uint64_t* keys, const int64_t* vals, uint64_t n,
+                                 uint64_t* out_keys, uint64_t* out_lo, int64_t* out_hi,
+                                 uint64_t max_out);
+
 /* compute-only legs timed by bench.py's cpu_baseline (columns pre-generated) */
 int64_t orc_q1_kernel(const int32_t* od, const int32_t* ep, const int32_t* dc,
                       uint64_t n_rows, const uint32_t* dfirst, int64_t mn, int64_t mx,

@@ -99,6 +99,10 @@ def _decl(lib):
     lib.orc_q3_build_order_bits.argtypes = [c_vp, c_vp, c_u64, c_vp, c_i32, c_vp]
     lib.orc_q3_probe_agg.restype = c_u64
     lib.orc_q3_probe_agg.argtypes = [c_vp] * 4 + [c_u64, c_vp, c_i32, c_vp, c_vp, c_u64]
+    lib.orc_hash_agg_stats_u64.restype = c_u64
+    lib.orc_hash_agg_stats_u64.argtypes = [c_vp, c_vp, c_u64] + [c_vp] * 5 + [c_u64]
+    lib.orc_hash_agg_sum128_u64.restype = c_u64
+    lib.orc_hash_agg_sum128_u64.argtypes = [c_vp, c_vp, c_u64] + [c_vp] * 3 + [c_u64]
     lib.orc_hash_agg_sum_u64.restype = c_u64
     lib.orc_hash_agg_sum_u64.argtypes = [c_vp, c_vp, c_u64, c_vp, c_vp, c_vp, c_u64]
     lib.orc_q1_kernel.restype = c_i64
@@ -239,6 +243,24 @@ def q3_pipeline(seed, row_start, n, n_orders, n_custs, seg=1,
     assert g != 2**64 - 1
     order = np.argsort(ok[:g])
     return ok[:g][order].copy(), os_[:g][order].copy()
+
+
+def hash_agg_stats(keys, vals):
+    cap = max(len(keys), 16)
+    o = [np.empty(cap, np.uint64)] + [np.empty(cap, np.int64) for _ in range(4)]
+    g = load().orc_hash_agg_stats_u64(_p(keys), _p(vals), len(keys), *[_p(x) for x in o], cap)
+    assert g != 2**64 - 1
+    return tuple(x[:g].copy() for x in o)
+
+
+def hash_agg_sum128(keys, vals):
+    cap = max(len(keys), 16)
+    ok = np.empty(cap, np.uint64)
+    lo = np.empty(cap, np.uint64)
+    hi = np.empty(cap, np.int64)
+    g = load().orc_hash_agg_sum128_u64(_p(keys), _p(vals), len(keys), _p(ok), _p(lo), _p(hi), cap)
+    assert g != 2**64 - 1
+    return ok[:g].copy(), lo[:g].copy(), hi[:g].copy()
 
 
 def hash_agg_sum(keys: np.ndarray, vals: np.ndarray):

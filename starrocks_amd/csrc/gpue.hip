@@ -1785,6 +1785,229 @@ __global__ void k_hash_agg_sum(const uint64_t* __restrict__ keys,
 // convert_hash_map_to_chunk analog (aggregator.cpp:1742-1816): iterate the
 // table, emit keys + finalized states (emission order is table order —
 // results are a set, compared key-sorted)
+// Full aggregate-function state set (reference exprs/agg/: SUM, COUNT,
+// MIN, MAX; AVG = SUM/COUNT at finalize, aggregate.h:136-269) and the
+// decimal SUM widening to int128 (exprs/agg/sum.h:181: decimal64 inputs
+// accumulate in int128 — here as a lo/hi pair with an explicit carry, two
+// 64-bit atomics per update, exact mod 2^128 and order-independent).
+__global__ void k_agg_init_minmax(long long* __restrict__ mins, long long* __restrict__ maxs,
+                                  uint64_t cap) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t s = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; s < cap; s += stride) {
+        mins[s] = 0x7FFFFFFFFFFFFFFFll;
+        maxs[s] = 0x8000000000000000ll;
+    }
+}
+
+__global__ void k_hash_agg_stats(const uint64_t* __restrict__ keys,
+                                 const int64_t* __restrict__ vals, uint64_t n,
+                                 unsigned long long* __restrict__ slots,
+                                 unsigned long long* __restrict__ sums,
+                                 unsigned long long* __restrict__ counts,
+                                 long long* __restrict__ mins,
+                                 long long* __restrict__ maxs, uint64_t cap_mask) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        unsigned long long k = keys[i];
+        long long v = (long long)vals[i];
+        uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        for (;;) {
+            unsigned long long cur = slots[s];
+            if (cur == k) break;
+            if (cur == AGG_EMPTY) {
+                unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
+                if (old == AGG_EMPTY || old == k) break;
+            }
+            s = (s + 1) & cap_mask;
+        }
+        atomicAdd(&sums[s], (unsigned long long)v);
+        atomicAdd(&counts[s], 1ull);
+        atomicMin(&mins[s], v);
+        atomicMax(&maxs[s], v);
+    }
+}
+
+__global__ void k_hash_agg_sum128(const uint64_t* __restrict__ keys,
+                                  const int64_t* __restrict__ vals, uint64_t n,
+                                  unsigned long long* __restrict__ slots,
+                                  unsigned long long* __restrict__ lo,
+                                  unsigned long long* __restrict__ hi, uint64_t cap_mask) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        unsigned long long k = keys[i];
+        long long v = (long long)vals[i];
+        uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        for (;;) {
+            unsigned long long cur = slots[s];
+            if (cur == k) break;
+            if (cur == AGG_EMPTY) {
+                unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
+                if (old == AGG_EMPTY || old == k) break;
+            }
+            s = (s + 1) & cap_mask;
+        }
+        unsigned long long vlo = (unsigned long long)v;
+        long long vhi = v < 0 ? -1ll : 0ll; // sign extension of the int64 addend
+        unsigned long long old_lo = atomicAdd(&lo[s], vlo);
+        unsigned long long carry = (old_lo + vlo < old_lo) ? 1ull : 0ull;
+        atomicAdd(&hi[s], (unsigned long long)vhi + carry);
+    }
+}
+
+__global__ void k_hash_agg_emit_wide(const unsigned long long* __restrict__ slots,
+                                     const unsigned long long* __restrict__ a,
+                                     const unsigned long long* __restrict__ b,
+                                     const long long* __restrict__ c,
+                                     const long long* __restrict__ d, uint64_t cap,
+                                     unsigned long long* __restrict__ cursor,
+                                     uint64_t max_out, uint64_t* __restrict__ out_keys,
+                                     int64_t* __restrict__ out_a, int64_t* __restrict__ out_b,
+                                     int64_t* __restrict__ out_c, int64_t* __restrict__ out_d) {
+    // same tiled two-phase structure as k_hash_agg_emit
+    uint64_t nb = gridDim.x;
+    uint64_t tile = (cap + nb - 1) / nb;
+    uint64_t lo_ = (uint64_t)blockIdx.x * tile;
+    uint64_t hi_ = min(lo_ + tile, cap);
+    int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+    __shared__ uint64_t wsum[BLOCK / WAVE + 1];
+    uint64_t cnum = 0;
+    for (uint64_t s = lo_ + threadIdx.x; s < hi_; s += blockDim.x)
+        cnum += (slots[s] != AGG_EMPTY);
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        cnum += __shfl_down((unsigned long long)cnum, off, WAVE);
+    if (lane == 0) wsum[wid] = cnum;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        uint64_t t = 0;
+        for (int w = 0; w < BLOCK / WAVE; w++) t += wsum[w];
+        wsum[BLOCK / WAVE] = atomicAdd(cursor, (unsigned long long)t);
+    }
+    __syncthreads();
+    uint64_t offset = wsum[BLOCK / WAVE];
+    __shared__ uint64_t wbase[BLOCK / WAVE + 1];
+    for (uint64_t base = lo_; base < hi_; base += blockDim.x) {
+        uint64_t s = base + threadIdx.x;
+        bool has = (s < hi_) && (slots[s] != AGG_EMPTY);
+        uint64_t mask = __ballot(has);
+        uint32_t wcount = __popcll(mask);
+        if (lane == 0) wbase[wid] = wcount;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            uint64_t acc = offset;
+            for (int w = 0; w < BLOCK / WAVE; w++) {
+                uint64_t v = wbase[w];
+                wbase[w] = acc;
+                acc += v;
+            }
+            wbase[BLOCK / WAVE] = acc;
+        }
+        __syncthreads();
+        if (has) {
+            uint64_t pos = wbase[wid] + __popcll(mask & ((1ull << lane) - 1));
+            if (pos < max_out) {
+                out_keys[pos] = slots[s];
+                out_a[pos] = (int64_t)a[s];
+                if (out_b) out_b[pos] = (int64_t)b[s];
+                if (out_c) out_c[pos] = (int64_t)c[s];
+                if (out_d) out_d[pos] = (int64_t)d[s];
+            }
+        }
+        offset = wbase[BLOCK / WAVE];
+        __syncthreads();
+    }
+}
+
+extern "C" {
+int gpue_hash_agg_stats_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uint64_t n,
+                            uint64_t capacity_hint, gpue_dbuf* out_keys, gpue_dbuf* out_sums,
+                            gpue_dbuf* out_counts, gpue_dbuf* out_mins, gpue_dbuf* out_maxs,
+                            uint64_t max_out, uint64_t* n_groups);
+int gpue_hash_agg_sum128_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uint64_t n,
+                             uint64_t capacity_hint, gpue_dbuf* out_keys, gpue_dbuf* out_lo,
+                             gpue_dbuf* out_hi, uint64_t max_out, uint64_t* n_groups);
+}
+
+int gpue_hash_agg_stats_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uint64_t n,
+                            uint64_t capacity_hint, gpue_dbuf* out_keys, gpue_dbuf* out_sums,
+                            gpue_dbuf* out_counts, gpue_dbuf* out_mins, gpue_dbuf* out_maxs,
+                            uint64_t max_out, uint64_t* n_groups) {
+    ARG_CHECK(s && keys && vals && out_keys && out_sums && out_counts && out_mins &&
+              out_maxs && n_groups);
+    uint64_t cap = 16;
+    uint64_t want = capacity_hint ? capacity_hint : n * 2;
+    while (cap < want) cap <<= 1;
+    unsigned long long *d_slots = nullptr, *d_sums = nullptr, *d_counts = nullptr,
+                       *d_cursor = nullptr;
+    long long *d_mins = nullptr, *d_maxs = nullptr;
+    HIP_CHECK(hipMalloc(&d_slots, cap * 8));
+    HIP_CHECK(hipMalloc(&d_sums, cap * 8));
+    HIP_CHECK(hipMalloc(&d_counts, cap * 8));
+    HIP_CHECK(hipMalloc(&d_mins, cap * 8));
+    HIP_CHECK(hipMalloc(&d_maxs, cap * 8));
+    HIP_CHECK(hipMalloc(&d_cursor, 8));
+    HIP_CHECK(hipMemsetAsync(d_slots, 0xFF, cap * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_sums, 0, cap * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_counts, 0, cap * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_cursor, 0, 8, s->stream));
+    hipLaunchKernelGGL(k_agg_init_minmax, dim3(grid_for(cap)), dim3(BLOCK), 0, s->stream,
+                       d_mins, d_maxs, cap);
+    hipLaunchKernelGGL(k_hash_agg_stats, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+                       (const uint64_t*)keys->ptr, (const int64_t*)vals->ptr, n, d_slots,
+                       d_sums, d_counts, d_mins, d_maxs, cap - 1);
+    hipLaunchKernelGGL(k_hash_agg_emit_wide, dim3(grid_for(cap)), dim3(BLOCK), 0, s->stream,
+                       d_slots, d_sums, d_counts, d_mins, d_maxs, cap, d_cursor, max_out,
+                       (uint64_t*)out_keys->ptr, (int64_t*)out_sums->ptr,
+                       (int64_t*)out_counts->ptr, (int64_t*)out_mins->ptr,
+                       (int64_t*)out_maxs->ptr);
+    unsigned long long groups = 0;
+    HIP_CHECK(hipMemcpyAsync(&groups, d_cursor, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_slots); (void)hipFree(d_sums); (void)hipFree(d_counts);
+    (void)hipFree(d_mins); (void)hipFree(d_maxs); (void)hipFree(d_cursor);
+    *n_groups = groups;
+    if (groups > max_out) {
+        snprintf(g_err, sizeof(g_err), "agg_stats: %llu groups exceed max_out", groups);
+        return GPUE_ERR_ARG;
+    }
+    return GPUE_OK;
+}
+
+int gpue_hash_agg_sum128_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uint64_t n,
+                             uint64_t capacity_hint, gpue_dbuf* out_keys, gpue_dbuf* out_lo,
+                             gpue_dbuf* out_hi, uint64_t max_out, uint64_t* n_groups) {
+    ARG_CHECK(s && keys && vals && out_keys && out_lo && out_hi && n_groups);
+    uint64_t cap = 16;
+    uint64_t want = capacity_hint ? capacity_hint : n * 2;
+    while (cap < want) cap <<= 1;
+    unsigned long long *d_slots = nullptr, *d_lo = nullptr, *d_hi = nullptr,
+                       *d_cursor = nullptr;
+    HIP_CHECK(hipMalloc(&d_slots, cap * 8));
+    HIP_CHECK(hipMalloc(&d_lo, cap * 8));
+    HIP_CHECK(hipMalloc(&d_hi, cap * 8));
+    HIP_CHECK(hipMalloc(&d_cursor, 8));
+    HIP_CHECK(hipMemsetAsync(d_slots, 0xFF, cap * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_lo, 0, cap * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_hi, 0, cap * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_cursor, 0, 8, s->stream));
+    hipLaunchKernelGGL(k_hash_agg_sum128, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+                       (const uint64_t*)keys->ptr, (const int64_t*)vals->ptr, n, d_slots,
+                       d_lo, d_hi, cap - 1);
+    hipLaunchKernelGGL(k_hash_agg_emit_wide, dim3(grid_for(cap)), dim3(BLOCK), 0, s->stream,
+                       d_slots, d_lo, d_hi, nullptr, nullptr, cap, d_cursor, max_out,
+                       (uint64_t*)out_keys->ptr, (int64_t*)out_lo->ptr,
+                       (int64_t*)out_hi->ptr, nullptr, nullptr);
+    unsigned long long groups = 0;
+    HIP_CHECK(hipMemcpyAsync(&groups, d_cursor, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_slots); (void)hipFree(d_lo); (void)hipFree(d_hi); (void)hipFree(d_cursor);
+    *n_groups = groups;
+    if (groups > max_out) {
+        snprintf(g_err, sizeof(g_err), "agg_sum128: %llu groups exceed max_out", groups);
+        return GPUE_ERR_ARG;
+    }
+    return GPUE_OK;
+}
+
 // convert_hash_map_to_chunk analog (aggregator.cpp:1742-1816). Tiled
 // two-phase: each block counts its contiguous slot tile, takes ONE cursor
 // atomic for its base, then emits with ballot/prefix placement. A per-wave

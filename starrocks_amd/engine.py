@@ -98,6 +98,10 @@ def _declare(lib):
                                        c_vp, c_vp, c_u64, ctypes.POINTER(c_u64)]),
         "gpue_q3_probe_agg": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_vp, c_i32, c_u64,
                                      c_vp, c_vp, c_u64, ctypes.POINTER(c_u64)]),
+        "gpue_hash_agg_stats_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_u64] + [c_vp] * 5 +
+                                    [c_u64, ctypes.POINTER(c_u64)]),
+        "gpue_hash_agg_sum128_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_u64] + [c_vp] * 3 +
+                                     [c_u64, ctypes.POINTER(c_u64)]),
         "gpue_hash_agg_sum_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_u64, c_vp, c_vp, c_vp,
                                           c_u64, ctypes.POINTER(c_u64)]),
         "gpue_timer_start": (c_i32, [c_vp]),
@@ -350,6 +354,22 @@ class Engine:
         _ck(self._lib, self._lib.gpue_hash_agg_sum_u64(
             self._h, keys._h, vals._h, n, capacity_hint, out_keys._h, out_sums._h, oc,
             max_out or out_keys.nbytes // 8, ctypes.byref(g)))
+        return g.value
+
+    def hash_agg_stats_u64(self, keys, vals, n, out_keys, out_sums, out_counts,
+                           out_mins, out_maxs, max_out, capacity_hint=0):
+        g = c_u64()
+        _ck(self._lib, self._lib.gpue_hash_agg_stats_u64(
+            self._h, keys._h, vals._h, n, capacity_hint, out_keys._h, out_sums._h,
+            out_counts._h, out_mins._h, out_maxs._h, max_out, ctypes.byref(g)))
+        return g.value
+
+    def hash_agg_sum128_u64(self, keys, vals, n, out_keys, out_lo, out_hi, max_out,
+                            capacity_hint=0):
+        g = c_u64()
+        _ck(self._lib, self._lib.gpue_hash_agg_sum128_u64(
+            self._h, keys._h, vals._h, n, capacity_hint, out_keys._h, out_lo._h,
+            out_hi._h, max_out, ctypes.byref(g)))
         return g.value
 
     def gather_u32(self, inp: DBuf, idx: DBuf, n, out: DBuf):

@@ -391,6 +391,63 @@ def test_q3_pipeline_parity(engine):
         b.free()
 
 
+def test_hash_agg_stats_parity(engine):
+    """SUM/COUNT/MIN/MAX states + AVG finalize vs oracle — int64 bit-exact;
+    AVG is the double division of the exact sum/count pair (identical IEEE
+    operands => bit-identical; stated tolerance 1e-12 regardless)."""
+    rng = np.random.default_rng(41)
+    n = 1_000_000
+    keys = rng.integers(0, 10_000, n).astype(np.uint64)
+    vals = rng.integers(-10**12, 10**12, n).astype(np.int64)
+    kb = engine.alloc(keys.nbytes)
+    kb.h2d(keys)
+    vb = engine.alloc(vals.nbytes)
+    vb.h2d(vals)
+    outs = [engine.alloc(20_000 * 8) for _ in range(5)]
+    g = engine.hash_agg_stats_u64(kb, vb, n, *outs, 20_000)
+    gk = outs[0].d2h(np.uint64, g)
+    gs, gc, gmn, gmx = (o.d2h(np.int64, g) for o in outs[1:])
+    ek, es, ec, emn, emx = orc.hash_agg_stats(keys, vals)
+    assert g == len(ek)
+    o1, o2 = np.argsort(gk), np.argsort(ek)
+    assert np.array_equal(gk[o1], ek[o2])
+    assert np.array_equal(gs[o1], es[o2])
+    assert np.array_equal(gc[o1], ec[o2])
+    assert np.array_equal(gmn[o1], emn[o2])
+    assert np.array_equal(gmx[o1], emx[o2])
+    # AVG finalize (float path): double division of exact int64 pairs
+    gavg = gs[o1].astype(np.float64) / gc[o1]
+    eavg = es[o2].astype(np.float64) / ec[o2]
+    assert np.allclose(gavg, eavg, rtol=1e-12, atol=0)
+    for b in [kb, vb] + outs:
+        b.free()
+
+
+def test_hash_agg_sum128_parity(engine):
+    """Decimal128 SUM (int128 lo/hi with carry) vs oracle __int128 — exact."""
+    rng = np.random.default_rng(43)
+    n = 2_000_000
+    keys = rng.integers(0, 500, n).astype(np.uint64)
+    vals = rng.integers(-2**62, 2**62, n).astype(np.int64)
+    kb = engine.alloc(keys.nbytes)
+    kb.h2d(keys)
+    vb = engine.alloc(vals.nbytes)
+    vb.h2d(vals)
+    ok_b, lo_b, hi_b = (engine.alloc(1000 * 8) for _ in range(3))
+    g = engine.hash_agg_sum128_u64(kb, vb, n, ok_b, lo_b, hi_b, 1000)
+    gk = ok_b.d2h(np.uint64, g)
+    glo = lo_b.d2h(np.uint64, g)
+    ghi = hi_b.d2h(np.int64, g)
+    ek, elo, ehi = orc.hash_agg_sum128(keys, vals)
+    assert g == len(ek)
+    o1, o2 = np.argsort(gk), np.argsort(ek)
+    assert np.array_equal(gk[o1], ek[o2])
+    assert np.array_equal(glo[o1], elo[o2])
+    assert np.array_equal(ghi[o1], ehi[o2])
+    for b in (kb, vb, ok_b, lo_b, hi_b):
+        b.free()
+
+
 def test_partition_parity(engine):
     n, nch = 3_000_000, 8
     keys = engine.alloc(n * 4)
