@@ -382,6 +382,7 @@ enum { TAG_ORDERDATE = 1, TAG_EXTPRICE = 2, TAG_DISCOUNT = 3,
        TAG_SNAT = 13, TAG_SCITY = 14 };
 
 #define N_DAYS 2556
+#define AGG_EMPTY 0xFFFFFFFFFFFFFFFFull
 #define N_PARTS_SF100 1400000u
 #define N_SUPPS_SF100 200000u
 #define N_CUSTS_SF100 3000000u
@@ -759,27 +760,130 @@ void orc_q3_build_order_bits(const int32_t* o_custkey, const int32_t* o_orderdat
     }
 }
 
-/* fused lineitem probe + hash agg: returns group count (keys=l_orderkey,
- * sums=revenue scale-4) via orc_hash_agg_sum_u64 on the passing rows */
+/* fused lineitem probe + hash agg (keys=l_orderkey, sums=revenue scale-4).
+ * OpenMP: each thread owns the hash-partition h(key) %% nthreads of the
+ * group space — every key is aggregated by exactly one thread (the CPU
+ * two-level/partitioned agg idiom, agg_hash_variant.cpp:318) — then emits
+ * its partition's groups at a prefix offset. */
 uint64_t orc_q3_probe_agg(const int64_t* lk, const int64_t* ext, const int64_t* disc,
                           const int32_t* ship, uint64_t n, const uint8_t* order_bits,
                           int32_t ship_cutoff, uint64_t* out_keys, int64_t* out_sums,
                           uint64_t max_out) {
+#ifdef _OPENMP
+    int nt = omp_get_max_threads();
+#else
+    int nt = 1;
+#endif
     uint64_t* keys = (uint64_t*)malloc(n * 8);
     int64_t* vals = (int64_t*)malloc(n * 8);
     uint64_t m = 0;
-    for (uint64_t i = 0; i < n; i++) {
-        if (ship[i] <= ship_cutoff) continue;
-        uint64_t o = (uint64_t)lk[i] - 1;
-        if (!(order_bits[o / 8] & (1u << (o % 8)))) continue;
-        keys[m] = (uint64_t)lk[i];
-        vals[m] = ext[i] * (100 - disc[i]);
-        m++;
+#pragma omp parallel
+    {
+        /* filter pass: per-thread chunks, prefix offsets */
+#ifdef _OPENMP
+        int t = omp_get_thread_num();
+#else
+        int t = 0;
+#endif
+        static uint64_t tcounts[1024];
+        uint64_t chunk = (n + nt - 1) / nt;
+        uint64_t lo = (uint64_t)t * chunk, hi = lo + chunk < n ? lo + chunk : n;
+        if (lo > n) lo = n;
+        if (hi < lo) hi = lo;
+        uint64_t c = 0;
+        for (uint64_t i = lo; i < hi; i++) {
+            if (ship[i] <= ship_cutoff) continue;
+            uint64_t o = (uint64_t)lk[i] - 1;
+            if (order_bits[o / 8] & (1u << (o % 8))) c++;
+        }
+        tcounts[t + 1] = c;
+#pragma omp barrier
+#pragma omp single
+        {
+            tcounts[0] = 0;
+            for (int j = 1; j <= nt; j++) tcounts[j] += tcounts[j - 1];
+            m = tcounts[nt];
+        }
+        uint64_t w = tcounts[t];
+        for (uint64_t i = lo; i < hi; i++) {
+            if (ship[i] <= ship_cutoff) continue;
+            uint64_t o = (uint64_t)lk[i] - 1;
+            if (!(order_bits[o / 8] & (1u << (o % 8)))) continue;
+            keys[w] = (uint64_t)lk[i];
+            vals[w] = ext[i] * (100 - disc[i]);
+            w++;
+        }
     }
-    uint64_t g = orc_hash_agg_sum_u64(keys, vals, m, out_keys, out_sums, NULL, max_out);
+    /* partitioned aggregate: thread t owns keys with h(key) %% nt == t */
+    uint64_t cap = 16;
+    while (cap < m * 2 + 16) cap <<= 1;
+    uint64_t percap = cap / (nt ? nt : 1) + 16;
+    uint64_t* gcounts = (uint64_t*)calloc(nt + 1, sizeof(uint64_t));
+    uint64_t** tkeys = (uint64_t**)malloc(nt * sizeof(void*));
+    int64_t** tsums = (int64_t**)malloc(nt * sizeof(void*));
+    int overflow = 0;
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        int t = omp_get_thread_num();
+#else
+        int t = 0;
+#endif
+        uint64_t mycap = 16;
+        while (mycap < percap * 4) mycap <<= 1;
+        uint64_t* slots = (uint64_t*)malloc(mycap * 8);
+        int64_t* sums = (int64_t*)calloc(mycap, 8);
+        memset(slots, 0xFF, mycap * 8);
+        uint32_t mask = (uint32_t)(mycap - 1);
+        for (uint64_t i = 0; i < m; i++) {
+            uint64_t k = keys[i];
+            uint64_t h = k * 11400714819323198485ull;
+            if ((int)((h >> 48) % (uint64_t)nt) != t) continue;
+            uint32_t s = (uint32_t)(h >> 32) & mask;
+            for (;;) {
+                if (slots[s] == AGG_EMPTY) slots[s] = k;
+                if (slots[s] == k) {
+                    sums[s] += vals[i];
+                    break;
+                }
+                s = (s + 1) & mask;
+            }
+        }
+        uint64_t g = 0;
+        uint64_t* ok = (uint64_t*)malloc(mycap * 8);
+        int64_t* os = (int64_t*)malloc(mycap * 8);
+        for (uint64_t s = 0; s < mycap; s++) {
+            if (slots[s] == AGG_EMPTY) continue;
+            ok[g] = slots[s];
+            os[g] = sums[s];
+            g++;
+        }
+        gcounts[t + 1] = g;
+        tkeys[t] = ok;
+        tsums[t] = os;
+        free(slots);
+        free(sums);
+#pragma omp barrier
+#pragma omp single
+        {
+            for (int j = 1; j <= nt; j++) gcounts[j] += gcounts[j - 1];
+            if (gcounts[nt] > max_out) overflow = 1;
+        }
+        if (!overflow)
+            for (uint64_t j = 0; j < gcounts[t + 1] - gcounts[t]; j++) {
+                out_keys[gcounts[t] + j] = tkeys[t][j];
+                out_sums[gcounts[t] + j] = tsums[t][j];
+            }
+        free(tkeys[t]);
+        free(tsums[t]);
+    }
+    uint64_t total = overflow ? UINT64_MAX : gcounts[nt];
+    free(gcounts);
+    free(tkeys);
+    free(tsums);
     free(keys);
     free(vals);
-    return g;
+    return total;
 }
 
 /* ====================================================================== */
@@ -791,8 +895,6 @@ uint64_t orc_q3_probe_agg(const int64_t* lk, const int64_t* ext, const int64_t* 
  * iteration order, which is not part of the result (output compared as a
  * key-sorted set); restated here as open-addressing linear probing.        */
 /* ====================================================================== */
-
-#define AGG_EMPTY 0xFFFFFFFFFFFFFFFFull
 
 static inline uint32_t agg_hash_u64(uint64_t v, uint32_t mask) {
     /* multiplicative 64-bit (same family as JoinKeyHash<T,8>) */
