@@ -135,23 +135,35 @@ def cpu_baseline(workload, rows_full):
 
 
 def cpu_baseline_q3(rows_full):
-    """Oracle Q3 probe+agg leg, single-threaded restatement (the hash agg is
-    sequential); bounded sample."""
+    """Oracle Q3 compute leg (orders-bits build + probe + hash agg) on this
+    box's cores; generation untimed, matching the GPU timed region."""
     from oracle import pyoracle as orc
-    import numpy as np
-    sample = min(rows_full, 5_000_000)
-    n_orders, n_custs = 2_000_000, 200_000  # scaled dims for the bounded sample
+    sample = min(rows_full, 20_000_000)
+    n_orders, n_custs = 8_000_000, 800_000  # dims scaled with the sample
+    mkt = orc.gen_cust_mkt16(SEED, n_custs)
+    cbits = np.zeros((n_custs + 7) // 8, np.uint8)
+    orc.load().orc_q3_build_cust_bits(orc._p(mkt), n_custs, orc.mkt_literal(1),
+                                      orc._p(cbits))
+    oc, od_ = orc.gen_orders_q3(SEED, n_orders, n_custs)
+    obits = np.zeros((n_orders + 7) // 8, np.uint8)
+    lk, ext, disc, ship = orc.gen_lineitem_q3(SEED, 0, sample, n_orders)
+    ok = np.empty(sample, np.uint64)
+    os_ = np.empty(sample, np.int64)
     t0 = time.perf_counter()
     passes = 0
     while time.perf_counter() - t0 < 10.0:
-        orc.q3_pipeline(SEED, 0, sample, n_orders, n_custs)
+        orc.load().orc_q3_build_order_bits(orc._p(oc), orc._p(od_), n_orders,
+                                           orc._p(cbits), Q3_CUTOFF, orc._p(obits))
+        orc.load().orc_q3_probe_agg(orc._p(lk), orc._p(ext), orc._p(disc), orc._p(ship),
+                                    sample, orc._p(obits), Q3_CUTOFF, orc._p(ok),
+                                    orc._p(os_), sample)
         passes += 1
     dt = time.perf_counter() - t0
     return {"value": round(passes * sample / dt, 1), "unit": "rows/s",
             "cores": os.cpu_count(), "kind": "port",
             "sample": f"{passes} passes over {sample} rows, dims scaled to "
-                      f"{n_orders}/{n_custs} ({dt:.1f}s, OMP hash-partitioned oracle "
-                      "incl. generation+build)"}
+                      f"{n_orders}/{n_custs} ({dt:.1f}s, OMP partition-scatter oracle; "
+                      "generation untimed)"}
 
 
 def read_pmc_traffic(workload):

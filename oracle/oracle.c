@@ -814,10 +814,56 @@ uint64_t orc_q3_probe_agg(const int64_t* lk, const int64_t* ext, const int64_t* 
             w++;
         }
     }
-    /* partitioned aggregate: thread t owns keys with h(key) %% nt == t */
-    uint64_t cap = 16;
-    while (cap < m * 2 + 16) cap <<= 1;
-    uint64_t percap = cap / (nt ? nt : 1) + 16;
+    /* partitioned aggregate: scatter the filtered pairs into nt hash
+     * partitions (parallel histogram + prefix + scatter), then thread t
+     * aggregates its contiguous partition with a private map — each key
+     * handled by exactly one thread, no locks, no cross-scan. */
+    uint64_t* pkeys = (uint64_t*)malloc((m + 1) * 8);
+    int64_t* pvals = (int64_t*)malloc((m + 1) * 8);
+    uint64_t* poff = (uint64_t*)calloc((size_t)nt * nt + nt + 1, sizeof(uint64_t));
+    uint64_t* pstart = (uint64_t*)calloc(nt + 1, sizeof(uint64_t));
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        int t = omp_get_thread_num();
+#else
+        int t = 0;
+#endif
+        uint64_t chunk = (m + nt - 1) / nt;
+        uint64_t lo = (uint64_t)t * chunk, hi = lo + chunk < m ? lo + chunk : m;
+        if (lo > m) lo = m;
+        if (hi < lo) hi = lo;
+        uint64_t* hist = poff + (size_t)t * nt;
+        for (uint64_t i = lo; i < hi; i++) {
+            uint64_t h = keys[i] * 11400714819323198485ull;
+            hist[(h >> 48) % (uint64_t)nt]++;
+        }
+#pragma omp barrier
+#pragma omp single
+        {
+            /* offsets: partition-major, then chunk order within partition */
+            uint64_t acc = 0;
+            for (int p = 0; p < nt; p++) {
+                pstart[p] = acc;
+                for (int c = 0; c < nt; c++) {
+                    uint64_t v = poff[(size_t)c * nt + p];
+                    poff[(size_t)c * nt + p] = acc;
+                    acc += v;
+                }
+            }
+            pstart[nt] = acc;
+        }
+        uint64_t* cur = (uint64_t*)malloc(nt * 8);
+        memcpy(cur, poff + (size_t)t * nt, nt * 8);
+        for (uint64_t i = lo; i < hi; i++) {
+            uint64_t h = keys[i] * 11400714819323198485ull;
+            uint64_t p = (h >> 48) % (uint64_t)nt;
+            uint64_t w = cur[p]++;
+            pkeys[w] = keys[i];
+            pvals[w] = vals[i];
+        }
+        free(cur);
+    }
     uint64_t* gcounts = (uint64_t*)calloc(nt + 1, sizeof(uint64_t));
     uint64_t** tkeys = (uint64_t**)malloc(nt * sizeof(void*));
     int64_t** tsums = (int64_t**)malloc(nt * sizeof(void*));
@@ -829,29 +875,28 @@ uint64_t orc_q3_probe_agg(const int64_t* lk, const int64_t* ext, const int64_t* 
 #else
         int t = 0;
 #endif
+        uint64_t lo = pstart[t], hi = pstart[t + 1];
         uint64_t mycap = 16;
-        while (mycap < percap * 4) mycap <<= 1;
+        while (mycap < (hi - lo) * 2 + 16) mycap <<= 1;
         uint64_t* slots = (uint64_t*)malloc(mycap * 8);
         int64_t* sums = (int64_t*)calloc(mycap, 8);
         memset(slots, 0xFF, mycap * 8);
         uint32_t mask = (uint32_t)(mycap - 1);
-        for (uint64_t i = 0; i < m; i++) {
-            uint64_t k = keys[i];
-            uint64_t h = k * 11400714819323198485ull;
-            if ((int)((h >> 48) % (uint64_t)nt) != t) continue;
-            uint32_t s = (uint32_t)(h >> 32) & mask;
+        for (uint64_t i = lo; i < hi; i++) {
+            uint64_t k = pkeys[i];
+            uint32_t s = (uint32_t)((k * 11400714819323198485ull) >> 32) & mask;
             for (;;) {
                 if (slots[s] == AGG_EMPTY) slots[s] = k;
                 if (slots[s] == k) {
-                    sums[s] += vals[i];
+                    sums[s] += pvals[i];
                     break;
                 }
                 s = (s + 1) & mask;
             }
         }
         uint64_t g = 0;
-        uint64_t* ok = (uint64_t*)malloc(mycap * 8);
-        int64_t* os = (int64_t*)malloc(mycap * 8);
+        uint64_t* ok = (uint64_t*)malloc((mycap ? mycap : 1) * 8);
+        int64_t* os = (int64_t*)malloc((mycap ? mycap : 1) * 8);
         for (uint64_t s = 0; s < mycap; s++) {
             if (slots[s] == AGG_EMPTY) continue;
             ok[g] = slots[s];
@@ -881,6 +926,10 @@ uint64_t orc_q3_probe_agg(const int64_t* lk, const int64_t* ext, const int64_t* 
     free(gcounts);
     free(tkeys);
     free(tsums);
+    free(pkeys);
+    free(pvals);
+    free(poff);
+    free(pstart);
     free(keys);
     free(vals);
     return total;
