@@ -245,6 +245,12 @@ int gpue_hash_agg_sum128_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, 
 int gpue_gather_u32(gpue_session* s, gpue_dbuf* in, gpue_dbuf* idx, uint64_t n,
                     gpue_dbuf* out);
 
+/* ---- TopN ----
+ * ORDER BY value DESC LIMIT k (reference exec/chunks_sorter_topn.cpp):
+ * deterministic (value, key) lexicographic descending; k <= 16. */
+int gpue_topk_i64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uint64_t n,
+                  int k, uint64_t* out_keys, int64_t* out_vals);
+
 /* ---- event timing on the session stream (bench roofline evidence) ---- */
 int gpue_timer_start(gpue_session* s);
 int gpue_timer_stop(gpue_session* s, float* ms_out);

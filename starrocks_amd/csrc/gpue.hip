@@ -2224,7 +2224,11 @@ __global__ void k_q3_order_bits(const int32_t* __restrict__ ocust,
     }
 }
 
-// fused lineitem filter + orders semi-probe + hash-agg insert
+// fused lineitem filter + orders semi-probe + hash-agg insert.
+// Two-phase per pair of rows: ship filter (streamed, 16 B/lane via longlong2)
+// gates the orders-bitset gather; both rows' gathers issue back-to-back
+// before either insert (guide §5 item 4(c)). COUNT is not in Q3's select
+// list — one CAS-claim + one atomicAdd per passing row.
 __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
                                const int64_t* __restrict__ ext,
                                const int64_t* __restrict__ disc,
@@ -2233,15 +2237,14 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
                                unsigned long long* __restrict__ slots,
                                unsigned long long* __restrict__ sums,
                                unsigned long long* __restrict__ counts, uint64_t cap_mask) {
+    (void)counts;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
-        if (ship[i] <= ship_cutoff) continue;
-        unsigned long long k = (unsigned long long)lk[i];
-        uint64_t o = k - 1;
-        if (!((order_bits[o >> 5] >> (o & 31)) & 1u)) continue;
-        unsigned long long v = (unsigned long long)(ext[i] * (100 - disc[i]));
+    const uint64_t n2 = n / 2;
+    const longlong2* __restrict__ lk2 = (const longlong2*)lk;
+    const longlong2* __restrict__ ext2 = (const longlong2*)ext;
+    const longlong2* __restrict__ disc2 = (const longlong2*)disc;
+    auto insert = [&](unsigned long long k, unsigned long long v) {
         uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
-        (void)counts; // COUNT is not in Q3's select list — one CAS + one add
         for (;;) {
             unsigned long long cur = slots[s];
             if (cur == k) { atomicAdd(&sums[s], v); break; }
@@ -2251,6 +2254,33 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
             }
             s = (s + 1) & cap_mask;
         }
+    };
+    for (uint64_t p = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; p < n2; p += stride) {
+        int2 sh = ((const int2*)ship)[p];
+        longlong2 k2 = lk2[p];
+        // phase 1: issue both bitset word gathers unconditionally-ish (the
+        // ship filter has ~50 % selectivity; batching beats branch-serial)
+        uint64_t o0 = (uint64_t)k2.x - 1, o1 = (uint64_t)k2.y - 1;
+        uint32_t w0 = order_bits[o0 >> 5] >> (o0 & 31);
+        uint32_t w1 = order_bits[o1 >> 5] >> (o1 & 31);
+        bool p0 = (sh.x > ship_cutoff) & (w0 & 1u);
+        bool p1 = (sh.y > ship_cutoff) & (w1 & 1u);
+        if (p0 | p1) {
+            longlong2 e2 = ext2[p];
+            longlong2 d2 = disc2[p];
+            if (p0) insert((unsigned long long)k2.x,
+                           (unsigned long long)(e2.x * (100 - d2.x)));
+            if (p1) insert((unsigned long long)k2.y,
+                           (unsigned long long)(e2.y * (100 - d2.y)));
+        }
+    }
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (uint64_t i = n2 * 2 + tid; i < n; i += stride) {
+        if (ship[i] <= ship_cutoff) continue;
+        unsigned long long k = (unsigned long long)lk[i];
+        uint64_t o = k - 1;
+        if (!((order_bits[o >> 5] >> (o & 31)) & 1u)) continue;
+        insert(k, (unsigned long long)(ext[i] * (100 - disc[i])));
     }
 }
 
@@ -2431,6 +2461,121 @@ int gpue_gather_u32(gpue_session* s, gpue_dbuf* in, gpue_dbuf* idx, uint64_t n,
                        (const uint32_t*)in->ptr, (const uint32_t*)idx->ptr, n,
                        (uint32_t*)out->ptr);
     HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
+// ---------------------------------------------------------------------------
+// TopN — ORDER BY value DESC LIMIT k (reference exec/chunks_sorter_topn.cpp;
+// SSB/TPC-H result shapes end in ORDER BY + LIMIT).
+// Order: (value, key) lexicographic descending — deterministic under value
+// ties. Per-thread register top-k -> per-block LDS merge -> host merges the
+// per-block candidates (nb × k entries, tiny).
+// ---------------------------------------------------------------------------
+static constexpr int TOPK_MAX = 16;
+
+__global__ void k_topk_block(const uint64_t* __restrict__ keys,
+                             const int64_t* __restrict__ vals, uint64_t n, int k,
+                             uint64_t* __restrict__ blk_keys,
+                             int64_t* __restrict__ blk_vals) {
+    int64_t tv[TOPK_MAX];
+    uint64_t tk[TOPK_MAX];
+    for (int j = 0; j < k; j++) {
+        tv[j] = INT64_MIN;
+        tk[j] = 0;
+    }
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        int64_t v = vals[i];
+        uint64_t ky = keys[i];
+        if (v < tv[k - 1] || (v == tv[k - 1] && ky <= tk[k - 1])) continue;
+        int j = k - 1;
+        while (j > 0 && (tv[j - 1] < v || (tv[j - 1] == v && tk[j - 1] < ky))) {
+            tv[j] = tv[j - 1];
+            tk[j] = tk[j - 1];
+            j--;
+        }
+        tv[j] = v;
+        tk[j] = ky;
+    }
+    __shared__ int64_t sv[BLOCK * TOPK_MAX];
+    __shared__ uint64_t sk[BLOCK * TOPK_MAX];
+    for (int j = 0; j < k; j++) {
+        sv[threadIdx.x * k + j] = tv[j];
+        sk[threadIdx.x * k + j] = tk[j];
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        // merge the block's blockDim×k candidates into k
+        int64_t bv[TOPK_MAX];
+        uint64_t bk[TOPK_MAX];
+        for (int j = 0; j < k; j++) {
+            bv[j] = INT64_MIN;
+            bk[j] = 0;
+        }
+        for (int t = 0; t < (int)blockDim.x * k; t++) {
+            int64_t v = sv[t];
+            uint64_t ky = sk[t];
+            if (v < bv[k - 1] || (v == bv[k - 1] && ky <= bk[k - 1])) continue;
+            int j = k - 1;
+            while (j > 0 && (bv[j - 1] < v || (bv[j - 1] == v && bk[j - 1] < ky))) {
+                bv[j] = bv[j - 1];
+                bk[j] = bk[j - 1];
+                j--;
+            }
+            bv[j] = v;
+            bk[j] = ky;
+        }
+        for (int j = 0; j < k; j++) {
+            blk_vals[(uint64_t)blockIdx.x * k + j] = bv[j];
+            blk_keys[(uint64_t)blockIdx.x * k + j] = bk[j];
+        }
+    }
+}
+
+extern "C" int gpue_topk_i64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uint64_t n,
+                             int k, uint64_t* out_keys, int64_t* out_vals);
+int gpue_topk_i64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uint64_t n, int k,
+                  uint64_t* out_keys, int64_t* out_vals) {
+    ARG_CHECK(s && keys && vals && out_keys && out_vals && k >= 1 && k <= TOPK_MAX);
+    ARG_CHECK(keys->bytes >= n * 8 && vals->bytes >= n * 8);
+    uint32_t nb = grid_for(n);
+    uint64_t* d_bk = nullptr;
+    int64_t* d_bv = nullptr;
+    HIP_CHECK(hipMalloc(&d_bk, (uint64_t)nb * k * 8));
+    HIP_CHECK(hipMalloc(&d_bv, (uint64_t)nb * k * 8));
+    hipLaunchKernelGGL(k_topk_block, dim3(nb), dim3(BLOCK), 0, s->stream,
+                       (const uint64_t*)keys->ptr, (const int64_t*)vals->ptr, n, k,
+                       d_bk, d_bv);
+    uint64_t* h_bk = (uint64_t*)malloc((uint64_t)nb * k * 8);
+    int64_t* h_bv = (int64_t*)malloc((uint64_t)nb * k * 8);
+    HIP_CHECK(hipMemcpyAsync(h_bk, d_bk, (uint64_t)nb * k * 8, hipMemcpyDeviceToHost,
+                             s->stream));
+    HIP_CHECK(hipMemcpyAsync(h_bv, d_bv, (uint64_t)nb * k * 8, hipMemcpyDeviceToHost,
+                             s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    // final k-way selection over nb×k host candidates
+    for (int j = 0; j < k; j++) {
+        out_vals[j] = INT64_MIN;
+        out_keys[j] = 0;
+    }
+    for (uint64_t t = 0; t < (uint64_t)nb * k; t++) {
+        int64_t v = h_bv[t];
+        uint64_t ky = h_bk[t];
+        if (v < out_vals[k - 1] || (v == out_vals[k - 1] && ky <= out_keys[k - 1])) continue;
+        int j = k - 1;
+        while (j > 0 && (out_vals[j - 1] < v ||
+                         (out_vals[j - 1] == v && out_keys[j - 1] < ky))) {
+            out_vals[j] = out_vals[j - 1];
+            out_keys[j] = out_keys[j - 1];
+            j--;
+        }
+        out_vals[j] = v;
+        out_keys[j] = ky;
+    }
+    free(h_bk);
+    free(h_bv);
+    (void)hipFree(d_bk);
+    (void)hipFree(d_bv);
     return GPUE_OK;
 }
 

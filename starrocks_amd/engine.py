@@ -104,6 +104,7 @@ def _declare(lib):
                                      [c_u64, ctypes.POINTER(c_u64)]),
         "gpue_hash_agg_sum_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_u64, c_vp, c_vp, c_vp,
                                           c_u64, ctypes.POINTER(c_u64)]),
+        "gpue_topk_i64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp]),
         "gpue_timer_start": (c_i32, [c_vp]),
         "gpue_timer_stop": (c_i32, [c_vp, ctypes.POINTER(ctypes.c_float)]),
     }
@@ -380,6 +381,14 @@ class Engine:
         _ck(self._lib, self._lib.gpue_partition_i32(self._h, keys._h, n, num_channels,
                                                     sp.ctypes.data_as(c_vp), row_indexes._h))
         return sp
+
+    def topk_i64(self, keys: DBuf, vals: DBuf, n, k):
+        ok = np.zeros(k, np.uint64)
+        ov = np.zeros(k, np.int64)
+        _ck(self._lib, self._lib.gpue_topk_i64(self._h, keys._h, vals._h, n, k,
+                                               ok.ctypes.data_as(c_vp),
+                                               ov.ctypes.data_as(c_vp)))
+        return ok, ov
 
     # ---- stream event timing (bench) ----
     def timer_start(self):

@@ -448,6 +448,52 @@ def test_hash_agg_sum128_parity(engine):
         b.free()
 
 
+def test_topk_parity(engine):
+    """TopN (ORDER BY value DESC LIMIT k): deterministic (value,key)
+    lexicographic order vs numpy, including value ties."""
+    rng = np.random.default_rng(47)
+    n = 3_000_000
+    keys = rng.integers(1, 2**40, n).astype(np.uint64)
+    vals = rng.integers(0, 50_000, n).astype(np.int64)  # many ties
+    kb = engine.alloc(keys.nbytes)
+    kb.h2d(keys)
+    vb = engine.alloc(vals.nbytes)
+    vb.h2d(vals)
+    for k in (1, 10, 16):
+        gk, gv = engine.topk_i64(kb, vb, n, k)
+        order = np.lexsort((keys, vals))[::-1][:k]  # value desc, key desc tie-break
+        assert np.array_equal(gv, vals[order])
+        assert np.array_equal(gk, keys[order])
+    kb.free()
+    vb.free()
+
+
+def test_q3_with_topn(engine):
+    """Q3's full result shape: join + high-cardinality agg + ORDER BY revenue
+    DESC LIMIT 10 — top-10 of the group sums vs numpy over the oracle's."""
+    n, n_orders, n_custs = 2_000_000, 150_000, 30_000
+    mkt = engine.alloc(n_custs * 16)
+    engine.gen_cust_mkt16(SEED, n_custs, mkt)
+    cbits = engine.alloc((n_custs + 31) // 32 * 4)
+    engine.bits_str16_eq(mkt, n_custs, orc.mkt_literal(1), cbits)
+    oc, od = engine.alloc(n_orders * 4), engine.alloc(n_orders * 4)
+    engine.gen_orders_q3(SEED, n_orders, n_custs, oc, od)
+    obits = engine.alloc((n_orders + 31) // 32 * 4)
+    engine.q3_order_bits(oc, od, n_orders, cbits, 19950315, obits)
+    lk, ext, disc = (engine.alloc(n * 8) for _ in range(3))
+    ship = engine.alloc(n * 4)
+    engine.gen_lineitem_q3(SEED, 0, n, n_orders, lk, ext, disc, ship)
+    ok_b, os_b = engine.alloc(n * 8), engine.alloc(n * 8)
+    g = engine.q3_probe_agg(lk, ext, disc, ship, n, obits, 19950315, ok_b, os_b, n)
+    tk, tv = engine.topk_i64(ok_b, os_b, g, 10)
+    ek, es = orc.q3_pipeline(SEED, 0, n, n_orders, n_custs)
+    order = np.lexsort((ek, es))[::-1][:10]
+    assert np.array_equal(tv, es[order])
+    assert np.array_equal(tk, ek[order])
+    for b in (mkt, cbits, oc, od, obits, lk, ext, disc, ship, ok_b, os_b):
+        b.free()
+
+
 def test_partition_parity(engine):
     n, nch = 3_000_000, 8
     keys = engine.alloc(n * 4)
