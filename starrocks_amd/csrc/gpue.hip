@@ -1368,6 +1368,77 @@ k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
         if (g[j] != 0) atomicAdd(&group_sums[j], g[j]);
 }
 
+// Experiment variant (GPUE_Q21_GLOB=1): no LDS group array — passing rows
+// (~0.8 %) atomicAdd straight into the 56 KB global/L2-resident group
+// buffer, letting the kernel run at streaming geometry (grid_stream, 256-
+// thread blocks, no LDS occupancy cap).
+__global__ void k_q21_star_agg_glob(const int32_t* __restrict__ pk,
+                                    const int32_t* __restrict__ sk,
+                                    const int32_t* __restrict__ od,
+                                    const int32_t* __restrict__ rv, uint64_t n,
+                                    const uint32_t* __restrict__ pbits, int64_t psmin,
+                                    uint64_t psint, const uint16_t* __restrict__ pfirst,
+                                    const uint32_t* __restrict__ sbits, int64_t ssmin,
+                                    uint64_t ssint, const uint16_t* __restrict__ dfirst,
+                                    int64_t dmin,
+                                    unsigned long long* __restrict__ group_sums) {
+    const uint64_t n4 = n / 4;
+    const int4* __restrict__ pk4 = (const int4*)pk;
+    const int4* __restrict__ sk4 = (const int4*)sk;
+    const int4* __restrict__ od4 = (const int4*)od;
+    const int4* __restrict__ rv4 = (const int4*)rv;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    auto quad = [&](int4 p4, int4 s4, int4 o4, int4 r4) {
+        uint32_t pb[4], pin[4];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint32_t idx = (uint32_t)((&p4.x)[j] - psmin);
+            pin[j] = idx < psint;
+            uint32_t cidx = pin[j] ? idx : 0u;
+            pb[j] = pbits[cidx >> 5] >> (cidx & 31);
+        }
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            if (!(pin[j] & pb[j] & 1u)) continue;
+            uint32_t brand1 = pfirst[(&p4.x)[j] - 1];
+            uint32_t sidx = (uint32_t)((&s4.x)[j] - ssmin);
+            if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
+            uint32_t year1 = dfirst[(&o4.x)[j] - dmin];
+            atomicAdd(&group_sums[(year1 - 1) * 1000 + (brand1 - 1)],
+                      (unsigned long long)(int64_t)(&r4.x)[j]);
+        }
+    };
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i + stride < n4; i += 2 * stride) {
+        int4 pa = pk4[i], sa = sk4[i], oa = od4[i], ra = rv4[i];
+        uint64_t i2 = i + stride;
+        int4 pb_ = pk4[i2], sb = sk4[i2], ob = od4[i2], rb = rv4[i2];
+        quad(pa, sa, oa, ra);
+        quad(pb_, sb, ob, rb);
+    }
+    for (; i < n4; i += stride) quad(pk4[i], sk4[i], od4[i], rv4[i]);
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (uint64_t r = n4 * 4 + tid; r < n; r += stride) {
+        uint32_t pidx = (uint32_t)(pk[r] - psmin);
+        if (pidx >= psint || !((pbits[pidx >> 5] >> (pidx & 31)) & 1u)) continue;
+        uint32_t brand1 = pfirst[pk[r] - 1];
+        uint32_t sidx = (uint32_t)(sk[r] - ssmin);
+        if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
+        uint32_t year1 = dfirst[od[r] - dmin];
+        atomicAdd(&group_sums[(year1 - 1) * 1000 + (brand1 - 1)],
+                  (unsigned long long)(int64_t)rv[r]);
+    }
+}
+
+static bool q21_glob() {
+    static int v = -1;
+    if (v < 0) {
+        const char* e = getenv("GPUE_Q21_GLOB");
+        v = (e && atoi(e)) ? 1 : 0;
+    }
+    return v == 1;
+}
+
 extern "C" int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts,
                                        gpue_join_table* supps, gpue_join_table* dates,
                                        gpue_dbuf* pk, gpue_dbuf* sk, gpue_dbuf* od,
@@ -1380,6 +1451,16 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
     ARG_CHECK(parts->min_key == 1 && supps->min_key == 1);
     ARG_CHECK(parts->first16 && supps->first16 && dates->first16);
     HIP_CHECK(hipMemsetAsync(group_sums->ptr, 0, NG_Q21 * sizeof(int64_t), s->stream));
+    if (q21_glob()) {
+        hipLaunchKernelGGL(k_q21_star_agg_glob, dim3(grid_stream(n)), dim3(BLOCK), 0,
+                           s->stream, (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
+                           (const int32_t*)od->ptr, (const int32_t*)rv->ptr, n,
+                           parts->bitset, parts->set_min,
+                           (uint64_t)(parts->set_max - parts->set_min + 1), parts->first16,
+                           supps->bitset, supps->set_min,
+                           (uint64_t)(supps->set_max - supps->set_min + 1), dates->first16,
+                           dates->min_key, (unsigned long long*)group_sums->ptr);
+    } else
     hipLaunchKernelGGL(k_q21_star_agg, dim3(env_cap("GPUE_GRID_WIDE", 512)), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)od->ptr, (const int32_t*)rv->ptr, n,
