@@ -200,9 +200,6 @@ def main():
         tdist.init_process_group("nccl")
         dist = tdist
 
-    if wl == "q3" and world > 1:
-        raise SystemExit("q3 multi-GPU (partitioned high-cardinality agg merge) lands in "
-                         "round 2 — run q3 at N=1 or use q43 for the partitioned mode")
     eng = Engine(local_rank)
     rows = args.rows or {"q1": SF10_ROWS, "q21": SF100_ROWS, "q43": SF100_ROWS,
                          "q3": Q3_ROWS_PER_GPU}[wl]
@@ -258,9 +255,10 @@ def main():
         oc, od_ = eng.alloc(Q3_N_ORDERS * 4), eng.alloc(Q3_N_ORDERS * 4)
         eng.gen_orders_q3(SEED, Q3_N_ORDERS, Q3_N_CUSTS, oc, od_)
         obits = eng.alloc((Q3_N_ORDERS + 31) // 32 * 4)
-        lk, ext, disc = (eng.alloc(rows * 8) for _ in range(3))
-        ship = eng.alloc(rows * 4)
-        eng.gen_lineitem_q3(SEED, row_start, rows, Q3_N_ORDERS, lk, ext, disc, ship)
+        if world == 1:
+            lk, ext, disc = (eng.alloc(rows * 8) for _ in range(3))
+            ship = eng.alloc(rows * 4)
+            eng.gen_lineitem_q3(SEED, row_start, rows, Q3_N_ORDERS, lk, ext, disc, ship)
         max_out = 64_000_000
         ok_b, os_b = eng.alloc(max_out * 8), eng.alloc(max_out * 8)
         agg_tab = eng.agg_table_create(64_000_000)  # persistent; reset per pass
@@ -273,12 +271,67 @@ def main():
             return eng.q3_probe_agg_t(lk, ext, disc, ship, rows, obits, Q3_CUTOFF,
                                       agg_tab, ok_b, os_b, max_out)
 
-        def step():
-            # orders build pass (the hash-join BUILD phase: scan orders,
-            # apply date + customer filters) + probe
-            eng.q3_order_bits(oc, od_, Q3_N_ORDERS, cbits, Q3_CUTOFF, obits)
-            g = kernel_only()
-            return np.array([g], np.int64)
+        if world == 1:
+            def step():
+                # orders build pass (the hash-join BUILD phase: scan orders,
+                # apply date + customer filters) + probe
+                eng.q3_order_bits(oc, od_, Q3_N_ORDERS, cbits, Q3_CUTOFF, obits)
+                g = kernel_only()
+                return np.array([g], np.int64)
+        else:
+            # hash-partitioned mode on l_orderkey (config 5
+            # "genuinely requires partitioning"): after the all-to-all every
+            # rank's received rows carry orderkeys the partition function
+            # routes to it, so the HIGH-CARDINALITY group space is DISJOINT
+            # across ranks — local aggregation is final, no merge.
+            # Columns move between gpue kernels and RCCL in the same torch
+            # tensors (wrap_ptr).
+            cols_t = ([torch.empty(rows, dtype=torch.int64, device="cuda")
+                       for _ in range(3)] +
+                      [torch.empty(rows, dtype=torch.int32, device="cuda")])
+            lk, ext, disc = (eng.wrap_ptr(t.data_ptr(), rows * 8) for t in cols_t[:3])
+            ship = eng.wrap_ptr(cols_t[3].data_ptr(), rows * 4)
+            eng.gen_lineitem_q3(SEED, row_start, rows, Q3_N_ORDERS, lk, ext, disc, ship)
+            eng.sync()
+            send_t = ([torch.empty(rows, dtype=torch.int64, device="cuda")
+                       for _ in range(3)] +
+                      [torch.empty(rows, dtype=torch.int32, device="cuda")])
+            send = ([eng.wrap_ptr(t.data_ptr(), rows * 8) for t in send_t[:3]] +
+                    [eng.wrap_ptr(send_t[3].data_ptr(), rows * 4)])
+            ridx = eng.alloc(rows * 4)
+            sp = eng.partition_i64(lk, rows, world, ridx)
+            in_splits = np.diff(sp).astype(np.int64)
+            in_t = torch.from_numpy(in_splits).cuda()
+            out_t = torch.empty(world, dtype=torch.int64, device="cuda")
+            dist.all_to_all_single(out_t, in_t)
+            out_splits = out_t.cpu().numpy()
+            n_recv = max(int(out_splits.sum()), 1)
+            recv_t = ([torch.empty(n_recv, dtype=torch.int64, device="cuda")
+                       for _ in range(3)] +
+                      [torch.empty(n_recv, dtype=torch.int32, device="cuda")])
+            recv = ([eng.wrap_ptr(t.data_ptr(), n_recv * 8) for t in recv_t[:3]] +
+                    [eng.wrap_ptr(recv_t[3].data_ptr(), n_recv * 4)])
+            in_list = [int(x) for x in in_splits]
+            out_list = [int(x) for x in out_splits]
+
+            def kernel_only():
+                return eng.q3_probe_agg_t(recv[0], recv[1], recv[2], recv[3], n_recv,
+                                          obits, Q3_CUTOFF, agg_tab, ok_b, os_b, max_out)
+
+            def step():
+                eng.partition_i64(lk, rows, world, ridx)
+                for j, (c, s_) in enumerate(zip((lk, ext, disc, ship), send)):
+                    if j < 3:
+                        eng.gather_u64(c, ridx, rows, s_)
+                    else:
+                        eng.gather_u32(c, ridx, rows, s_)
+                eng.sync()
+                for st, rt in zip(send_t, recv_t):
+                    dist.all_to_all_single(rt[:n_recv], st, out_list, in_list)
+                torch.cuda.synchronize()
+                eng.q3_order_bits(oc, od_, Q3_N_ORDERS, cbits, Q3_CUTOFF, obits)
+                g = kernel_only()
+                return np.array([g], np.int64)
     else:  # q43
         acc, acc_t = make_acc(800)
         if world == 1:
@@ -342,6 +395,10 @@ def main():
 
     def run_step():
         if dist is not None:
+            if wl == "q3":
+                # partitioned high-cardinality agg: group spaces are disjoint
+                # across ranks — local aggregation IS the final result
+                return step()
             kernel_only() if wl != "q43" else step_partitioned_kernel()
             eng.sync()
             # phase1 -> phase2 aggregate merge (agg_hash_variant.h merge_batch
@@ -394,7 +451,7 @@ def main():
     for _ in range(R):
         kernel_only()
     kernel_ms = eng.timer_stop() / R
-    kernel_rows = rows if not (wl == "q43" and world > 1) else max(n_recv, 1)
+    kernel_rows = rows if not (wl in ("q43", "q3") and world > 1) else max(n_recv, 1)
     algo_bytes = BYTES_PER_ROW[wl] * kernel_rows
     achieved_gbps = algo_bytes / (kernel_ms / 1e3) / 1e9
     roofline = {"bound": "hbm", "achieved": round(achieved_gbps, 1), "peak": HBM_PEAK_GBPS,
@@ -425,7 +482,7 @@ def main():
                 "rows_per_gpu": rows,
                 "seed": SEED,
                 "parallelism": (f"dp{world}-hash-partitioned-alltoall"
-                                if wl == "q43" and world > 1
+                                if wl in ("q43", "q3") and world > 1
                                 else f"dp{world}-replicated-dims"),
             },
             "roofline": roofline,

@@ -244,6 +244,11 @@ int gpue_hash_agg_sum128_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, 
  * the counting-sorted indexes before the RCCL all-to-all. */
 int gpue_gather_u32(gpue_session* s, gpue_dbuf* in, gpue_dbuf* idx, uint64_t n,
                     gpue_dbuf* out);
+int gpue_gather_u64(gpue_session* s, gpue_dbuf* in, gpue_dbuf* idx, uint64_t n,
+                    gpue_dbuf* out);
+/* BIGINT-key partition (FNV over the 8 LE bytes of an int64 key column) */
+int gpue_partition_i64(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t num_channels,
+                       uint64_t* start_points_out, gpue_dbuf* row_indexes_out);
 
 /* ---- TopN ----
  * ORDER BY value DESC LIMIT k (reference exec/chunks_sorter_topn.cpp):

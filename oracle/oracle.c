@@ -351,6 +351,17 @@ void orc_partition_channel_u32(const uint32_t* keys, uint64_t n, uint32_t num_ch
     }
 }
 
+/* BIGINT variant: FNV over the 8 LE bytes (reference fnv_hash applied to an
+ * int64 key column) + ReduceOp */
+void orc_partition_channel_u64(const uint64_t* keys, uint64_t n, uint32_t num_channels,
+                               uint32_t* channel_ids) {
+#pragma omp parallel for schedule(static)
+    for (uint64_t i = 0; i < n; i++) {
+        uint32_t h = orc_fnv_hash(&keys[i], 8, 0x811C9DC5u);
+        channel_ids[i] = (uint32_t)(((uint64_t)h * num_channels) >> 32);
+    }
+}
+
 /* counting-sort row layout (exchange_sink_operator.cpp:629-660): forward
  * count, prefix-sum, then reverse iteration emit so each channel's rows stay
  * in ascending source order. */

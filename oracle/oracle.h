@@ -89,6 +89,8 @@ void orc_partition_channel_u32(const uint32_t* keys, uint64_t n, uint32_t num_ch
                                uint32_t* channel_ids);
 /* counting-sort row layout: start_points[ch+1] sizes then reverse emit
  * (exchange_sink_operator.cpp:629-660). row_indexes gets source row per slot. */
+void orc_partition_channel_u64(const uint64_t* keys, uint64_t n, uint32_t num_channels,
+                               uint32_t* channel_ids);
 void orc_partition_counting_sort(const uint32_t* channel_ids, uint64_t n,
                                  uint32_t num_channels, uint64_t* start_points,
                                  uint32_t* row_indexes);

@@ -78,6 +78,7 @@ def _decl(lib):
     lib.orc_filter_i64_lt_mt.restype = c_u64
     lib.orc_filter_i64_lt_mt.argtypes = [c_vp, c_u64, c_i64, c_vp]
     lib.orc_partition_channel_u32.argtypes = [c_vp, c_u64, u, c_vp]
+    lib.orc_partition_channel_u64.argtypes = [c_vp, c_u64, u, c_vp]
     lib.orc_partition_counting_sort.argtypes = [c_vp, c_u64, u, c_vp, c_vp]
     lib.orc_q1_pipeline.restype = c_i64
     lib.orc_q1_pipeline.argtypes = [c_u64, c_u64, c_u64, c_i32, c_i32, ctypes.POINTER(c_u64)]
@@ -339,6 +340,12 @@ def linear_chained_lookup(build_keys: np.ndarray, probe_keys: np.ndarray, first,
 def partition_channels(keys: np.ndarray, num_channels: int) -> np.ndarray:
     out = np.zeros(len(keys), np.uint32)
     load().orc_partition_channel_u32(_p(keys), len(keys), num_channels, _p(out))
+    return out
+
+
+def partition_channels_u64(keys: np.ndarray, num_channels: int) -> np.ndarray:
+    out = np.zeros(len(keys), np.uint32)
+    load().orc_partition_channel_u64(_p(keys), len(keys), num_channels, _p(out))
     return out
 
 

@@ -1634,6 +1634,13 @@ __device__ static inline uint32_t fnv_u32(uint32_t key, uint32_t seed) {
     return h;
 }
 
+__device__ static inline uint32_t fnv_u64(uint64_t key, uint32_t seed) {
+    uint32_t h = seed;
+    #pragma unroll
+    for (int b = 0; b < 8; b++) h = (((uint32_t)(key >> (8 * b)) & 0xFFu) ^ h) * 16777619u;
+    return h;
+}
+
 static constexpr uint32_t MAX_CH = 64;
 
 __global__ void k_partition_hist(const uint32_t* __restrict__ keys, uint64_t n, uint64_t tile,
@@ -1666,6 +1673,82 @@ __global__ void k_partition_emit(const uint32_t* __restrict__ keys, uint64_t n, 
         uint64_t pos = atomicAdd((unsigned long long*)&cursor[ch], 1ull);
         row_indexes[pos] = (uint32_t)i;
     }
+}
+
+__global__ void k_partition_hist64(const uint64_t* __restrict__ keys, uint64_t n,
+                                   uint64_t tile, uint32_t nch,
+                                   uint32_t* __restrict__ block_hist) {
+    __shared__ uint32_t h[MAX_CH];
+    for (uint32_t c = threadIdx.x; c < nch; c += blockDim.x) h[c] = 0;
+    __syncthreads();
+    uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
+    for (uint64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        uint32_t ch = (uint32_t)(((uint64_t)fnv_u64(keys[i], 0x811C9DC5u) * nch) >> 32);
+        atomicAdd(&h[ch], 1u);
+    }
+    __syncthreads();
+    for (uint32_t c = threadIdx.x; c < nch; c += blockDim.x)
+        block_hist[(uint64_t)blockIdx.x * nch + c] = h[c];
+}
+
+__global__ void k_partition_emit64(const uint64_t* __restrict__ keys, uint64_t n,
+                                   uint64_t tile, uint32_t nch,
+                                   const uint64_t* __restrict__ block_offsets,
+                                   uint32_t* __restrict__ row_indexes) {
+    __shared__ uint64_t cursor[MAX_CH];
+    for (uint32_t c = threadIdx.x; c < nch; c += blockDim.x)
+        cursor[c] = block_offsets[(uint64_t)blockIdx.x * nch + c];
+    __syncthreads();
+    uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
+    for (uint64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        uint32_t ch = (uint32_t)(((uint64_t)fnv_u64(keys[i], 0x811C9DC5u) * nch) >> 32);
+        uint64_t pos = atomicAdd((unsigned long long*)&cursor[ch], 1ull);
+        row_indexes[pos] = (uint32_t)i;
+    }
+}
+
+// BIGINT-key variant (FNV over the 8 LE bytes, reference fnv_hash on an
+// int64 key column)
+extern "C" int gpue_partition_i64(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t nch,
+                                  uint64_t* start_points_out, gpue_dbuf* row_indexes_out);
+int gpue_partition_i64(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t nch,
+                       uint64_t* start_points_out, gpue_dbuf* row_indexes_out) {
+    ARG_CHECK(s && keys && start_points_out && row_indexes_out);
+    ARG_CHECK(nch >= 1 && nch <= MAX_CH);
+    ARG_CHECK(keys->bytes >= n * 8 && row_indexes_out->bytes >= n * 4);
+    uint32_t nb = grid_for(n);
+    uint64_t tile = (n + nb - 1) / nb;
+    uint32_t* d_hist = nullptr;
+    uint64_t* d_off = nullptr;
+    HIP_CHECK(hipMalloc(&d_hist, (uint64_t)nb * nch * sizeof(uint32_t)));
+    hipLaunchKernelGGL(k_partition_hist64, dim3(nb), dim3(BLOCK), 0, s->stream,
+                       (const uint64_t*)keys->ptr, n, tile, nch, d_hist);
+    uint32_t* h_hist = (uint32_t*)malloc((uint64_t)nb * nch * sizeof(uint32_t));
+    uint64_t* h_off = (uint64_t*)malloc((uint64_t)nb * nch * sizeof(uint64_t));
+    HIP_CHECK(hipMemcpyAsync(h_hist, d_hist, (uint64_t)nb * nch * sizeof(uint32_t),
+                             hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    uint64_t acc = 0;
+    for (uint32_t c = 0; c < nch; c++) {
+        start_points_out[c] = acc;
+        for (uint32_t b = 0; b < nb; b++) {
+            h_off[(uint64_t)b * nch + c] = acc;
+            acc += h_hist[(uint64_t)b * nch + c];
+        }
+    }
+    start_points_out[nch] = acc;
+    HIP_CHECK(hipMalloc(&d_off, (uint64_t)nb * nch * sizeof(uint64_t)));
+    HIP_CHECK(hipMemcpyAsync(d_off, h_off, (uint64_t)nb * nch * sizeof(uint64_t),
+                             hipMemcpyHostToDevice, s->stream));
+    hipLaunchKernelGGL(k_partition_emit64, dim3(nb), dim3(BLOCK), 0, s->stream,
+                       (const uint64_t*)keys->ptr, n, tile, nch, d_off,
+                       (uint32_t*)row_indexes_out->ptr);
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_hist);
+    (void)hipFree(d_off);
+    free(h_hist);
+    free(h_off);
+    return GPUE_OK;
 }
 
 int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t nch,
@@ -2505,6 +2588,27 @@ __global__ void k_gather_u32(const uint32_t* __restrict__ in,
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
         out[i] = in[idx[i]];
+}
+
+__global__ void k_gather_u64(const uint64_t* __restrict__ in,
+                             const uint32_t* __restrict__ idx, uint64_t n,
+                             uint64_t* __restrict__ out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+        out[i] = in[idx[i]];
+}
+
+extern "C" int gpue_gather_u64(gpue_session* s, gpue_dbuf* in, gpue_dbuf* idx, uint64_t n,
+                               gpue_dbuf* out);
+int gpue_gather_u64(gpue_session* s, gpue_dbuf* in, gpue_dbuf* idx, uint64_t n,
+                    gpue_dbuf* out) {
+    ARG_CHECK(s && in && idx && out);
+    ARG_CHECK(idx->bytes >= n * 4 && out->bytes >= n * 8);
+    hipLaunchKernelGGL(k_gather_u64, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+                       (const uint64_t*)in->ptr, (const uint32_t*)idx->ptr, n,
+                       (uint64_t*)out->ptr);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
 }
 
 extern "C" int gpue_gather_u32(gpue_session* s, gpue_dbuf* in, gpue_dbuf* idx, uint64_t n,

@@ -87,6 +87,8 @@ def _declare(lib):
         "gpue_q21_star_agg_async": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_partition_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
         "gpue_gather_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
+        "gpue_gather_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
+        "gpue_partition_i64": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
         "gpue_gen_lineitem_q3": (c_i32, [c_vp, c_u64, c_u64, c_u64, c_u64] + [c_vp] * 4),
         "gpue_gen_orders_q3": (c_i32, [c_vp, c_u64, c_u64, c_u32, c_vp, c_vp]),
         "gpue_gen_cust_mkt16": (c_i32, [c_vp, c_u64, c_u32, c_vp]),
@@ -375,6 +377,15 @@ class Engine:
 
     def gather_u32(self, inp: DBuf, idx: DBuf, n, out: DBuf):
         _ck(self._lib, self._lib.gpue_gather_u32(self._h, inp._h, idx._h, n, out._h))
+
+    def gather_u64(self, inp: DBuf, idx: DBuf, n, out: DBuf):
+        _ck(self._lib, self._lib.gpue_gather_u64(self._h, inp._h, idx._h, n, out._h))
+
+    def partition_i64(self, keys: DBuf, n, num_channels, row_indexes: DBuf) -> np.ndarray:
+        sp = np.zeros(num_channels + 1, dtype=np.uint64)
+        _ck(self._lib, self._lib.gpue_partition_i64(self._h, keys._h, n, num_channels,
+                                                    sp.ctypes.data_as(c_vp), row_indexes._h))
+        return sp
 
     def partition(self, keys: DBuf, n, num_channels, row_indexes: DBuf) -> np.ndarray:
         sp = np.zeros(num_channels + 1, dtype=np.uint64)

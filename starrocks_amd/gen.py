@@ -147,6 +147,22 @@ def fnv_u32(keys: np.ndarray) -> np.ndarray:
         return h
 
 
+def fnv_u64(keys: np.ndarray) -> np.ndarray:
+    """FNV over the 8 LE bytes of each uint64 key (BIGINT columns)."""
+    with np.errstate(over="ignore"):
+        h = np.full(keys.shape, 0x811C9DC5, np.uint32)
+        k = keys.astype(np.uint64)
+        for b in range(8):
+            byte = ((k >> np.uint64(8 * b)) & np.uint64(0xFF)).astype(np.uint32)
+            h = (byte ^ h) * np.uint32(16777619)
+        return h
+
+
+def partition_channels_u64(keys: np.ndarray, num_channels: int) -> np.ndarray:
+    return ((fnv_u64(keys).astype(np.uint64) * np.uint64(num_channels))
+            >> np.uint64(32)).astype(np.uint32)
+
+
 def partition_channels(keys: np.ndarray, num_channels: int) -> np.ndarray:
     """ReduceOp channel assignment (shuffler.h:71-86): (fnv(key)*n)>>32."""
     return ((fnv_u32(keys).astype(np.uint64) * np.uint64(num_channels))

@@ -43,6 +43,51 @@ def test_q3_mkt_dictionary():
         assert len(s) == 16
 
 
+def test_q3_partitioned_execution_math():
+    """Simulate bench.py's N>1 hash-partitioned q3 exactly: rows route by
+    fnv64(l_orderkey) channels; each simulated rank aggregates only its
+    received rows; the group spaces are DISJOINT across ranks and their
+    union equals the unpartitioned result bit-exactly."""
+    from starrocks_amd import gen
+    n, world = 400_000, 4
+    lk, ext, disc, ship = orc.gen_lineitem_q3(SEED, 0, n, N_ORDERS)
+    row_channel = gen.partition_channels_u64(lk.astype(np.uint64), world)
+    # reference-pinned channel function: oracle vs numpy restatement
+    assert np.array_equal(row_channel, orc.partition_channels_u64(lk.astype(np.uint64), world))
+
+    mkt = orc.gen_cust_mkt16(SEED, N_CUSTS)
+    cbits = np.zeros((N_CUSTS + 7) // 8, np.uint8)
+    orc.load().orc_q3_build_cust_bits(orc._p(mkt), N_CUSTS, orc.mkt_literal(1), orc._p(cbits))
+    oc, od = orc.gen_orders_q3(SEED, N_ORDERS, N_CUSTS)
+    obits = np.zeros((N_ORDERS + 7) // 8, np.uint8)
+    orc.load().orc_q3_build_order_bits(orc._p(oc), orc._p(od), N_ORDERS, orc._p(cbits),
+                                       19950315, orc._p(obits))
+    merged = {}
+    seen_keys_per_rank = []
+    for r in range(world):
+        sel = row_channel == r
+        cap = max(int(sel.sum()), 16)
+        ok = np.empty(cap, np.uint64)
+        os_ = np.empty(cap, np.int64)
+        g = orc.load().orc_q3_probe_agg(
+            orc._p(np.ascontiguousarray(lk[sel])), orc._p(np.ascontiguousarray(ext[sel])),
+            orc._p(np.ascontiguousarray(disc[sel])), orc._p(np.ascontiguousarray(ship[sel])),
+            int(sel.sum()), orc._p(obits), 19950315, orc._p(ok), orc._p(os_), cap)
+        assert g != 2**64 - 1
+        seen_keys_per_rank.append(set(ok[:g].tolist()))
+        for k, s in zip(ok[:g].tolist(), os_[:g].tolist()):
+            merged[k] = merged.get(k, 0) + s
+    # disjointness: each group key appears on exactly one rank
+    for a in range(world):
+        for b in range(a + 1, world):
+            assert not (seen_keys_per_rank[a] & seen_keys_per_rank[b])
+    wk, ws = orc.q3_pipeline(SEED, 0, n, N_ORDERS, N_CUSTS)
+    mk = np.array(sorted(merged), np.uint64)
+    ms = np.array([merged[k] for k in sorted(merged)], np.int64)
+    assert np.array_equal(mk, wk)
+    assert np.array_equal(ms, ws)
+
+
 def test_q3_sharded_merge():
     """Lineitem row-sharding merges bit-exactly (group-keyed sum merge)."""
     n = 400_000
