@@ -192,15 +192,21 @@ def main():
 
     dist = None
     torch = None
+    # GPUE_DIST_BACKEND=gloo lets the N>1 code path be smoke-tested with two
+    # ranks sharing one GPU (gloo supports cuda-tensor all_reduce; the real
+    # runs use RCCL). Device = local_rank modulo visible devices — identical
+    # to local_rank on a full node.
+    backend = os.environ.get("GPUE_DIST_BACKEND", "nccl")
+    dev = local_rank % max(Engine.device_count(), 1)
     if world > 1:
         import torch as _torch
         import torch.distributed as tdist
         torch = _torch
-        torch.cuda.set_device(local_rank)
-        tdist.init_process_group("nccl")
+        torch.cuda.set_device(dev)
+        tdist.init_process_group(backend)
         dist = tdist
 
-    eng = Engine(local_rank)
+    eng = Engine(dev)
     rows = args.rows or {"q1": SF10_ROWS, "q21": SF100_ROWS, "q43": SF100_ROWS,
                          "q3": Q3_ROWS_PER_GPU}[wl]
     row_start = rank * rows  # weak scaling: each rank owns its shard
