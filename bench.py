@@ -197,14 +197,19 @@ def main():
     # runs use RCCL). Device = local_rank modulo visible devices — identical
     # to local_rank on a full node.
     backend = os.environ.get("GPUE_DIST_BACKEND", "nccl")
-    dev = local_rank % max(Engine.device_count(), 1)
     if world > 1:
+        # torch's HIP context must initialize BEFORE the ctypes engine loads
+        # the HIP runtime (the reverse order leaves torch.cuda device-less
+        # under torchrun)
         import torch as _torch
         import torch.distributed as tdist
         torch = _torch
+        dev = local_rank % max(torch.cuda.device_count(), 1)
         torch.cuda.set_device(dev)
         tdist.init_process_group(backend)
         dist = tdist
+    else:
+        dev = local_rank % max(Engine.device_count(), 1)
 
     eng = Engine(dev)
     rows = args.rows or {"q1": SF10_ROWS, "q21": SF100_ROWS, "q43": SF100_ROWS,
