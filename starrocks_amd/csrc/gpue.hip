@@ -86,6 +86,10 @@ struct gpue_session {
     hipEvent_t ev_start, ev_stop;
     // cached SSB date dimension (datekey per day index), built lazily
     int32_t* d_datekey = nullptr; // device, N_DAYS entries
+    // pinned bounce buffer: small result reads (the per-step accumulator
+    // pull) pay ~5 µs instead of a pageable-copy ~20 µs
+    void* pinned = nullptr;
+    static constexpr uint64_t PINNED_BYTES = 65536;
 };
 
 struct gpue_dbuf {
@@ -143,12 +147,14 @@ int gpue_session_create(int device_index, gpue_session** out) {
     HIP_CHECK(hipStreamCreate(&s->stream));
     HIP_CHECK(hipEventCreate(&s->ev_start));
     HIP_CHECK(hipEventCreate(&s->ev_stop));
+    HIP_CHECK(hipHostMalloc(&s->pinned, gpue_session::PINNED_BYTES));
     *out = s;
     return GPUE_OK;
 }
 
 void gpue_session_destroy(gpue_session* s) {
     if (!s) return;
+    if (s->pinned) (void)hipHostFree(s->pinned);
     if (s->d_datekey) (void)hipFree(s->d_datekey);
     (void)hipEventDestroy(s->ev_start);
     (void)hipEventDestroy(s->ev_stop);
@@ -206,6 +212,13 @@ int gpue_dbuf_h2d(gpue_dbuf* b, const void* src, uint64_t bytes, uint64_t dst_of
 
 int gpue_dbuf_d2h(gpue_dbuf* b, void* dst, uint64_t bytes, uint64_t src_off) {
     ARG_CHECK(b && dst && src_off + bytes <= b->bytes);
+    if (bytes <= gpue_session::PINNED_BYTES && b->s->pinned) {
+        HIP_CHECK(hipMemcpyAsync(b->s->pinned, (char*)b->ptr + src_off, bytes,
+                                 hipMemcpyDeviceToHost, b->s->stream));
+        HIP_CHECK(hipStreamSynchronize(b->s->stream));
+        memcpy(dst, b->s->pinned, bytes);
+        return GPUE_OK;
+    }
     HIP_CHECK(hipMemcpyAsync(dst, (char*)b->ptr + src_off, bytes,
                              hipMemcpyDeviceToHost, b->s->stream));
     HIP_CHECK(hipStreamSynchronize(b->s->stream));
