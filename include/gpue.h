@@ -86,6 +86,10 @@ int gpue_gen_lineorder_q43(gpue_session* s, uint64_t seed, uint64_t row_start, u
  * values v < theta; output order bit-identical to the reference loop. */
 int gpue_scan_filter_i64_lt(gpue_session* s, gpue_dbuf* in, uint64_t n, int64_t theta,
                             gpue_dbuf* out, uint64_t* out_count);
+/* Single-pass variant (decoupled lookback): one coalesced read + one
+ * coalesced write per element — the reference's algorithmic byte count. */
+int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64_t theta,
+                               gpue_dbuf* out, uint64_t* out_count);
 
 /* ---- hash-join build ----
  * Replaces JoinHashTable::build with the RANGE_DIRECT_MAPPING method the

@@ -483,6 +483,145 @@ __global__ void k_filter_emit(const int64_t* __restrict__ in, uint64_t n, int64_
     }
 }
 
+// ---------------------------------------------------------------------------
+// Single-pass ordered compaction — decoupled lookback. Each block claims
+// consecutive tiles via an atomic ticket (a claimed tile's predecessor was
+// claimed earlier by a resident block => forward progress); the tile's
+// {flag,count} pack into ONE 64-bit atomic word, so publish/consume need no
+// separate fence (all consumed data is in-word; guide §6 G16 concerns
+// multi-word hand-offs). Passing values stage compactly in LDS, one
+// coalesced read + one coalesced write per element — the reference's
+// algorithmic byte count exactly. Spins are bounded: on timeout the kernel
+// sets an error flag and exits (host returns GPUE_ERR_HIP).
+// ---------------------------------------------------------------------------
+static constexpr uint32_t FILT_ITEMS = 8;                     // rows per thread per tile
+static constexpr uint32_t FILT_TILE = BLOCK * FILT_ITEMS;     // 2048 rows
+static constexpr unsigned long long FILT_AGG = 1ull << 62;    // aggregate available
+static constexpr unsigned long long FILT_PREFIX = 2ull << 62; // inclusive prefix available
+static constexpr unsigned long long FILT_CNT_MASK = (1ull << 62) - 1;
+
+__global__ void k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, int64_t theta,
+                                  int64_t* __restrict__ out,
+                                  unsigned long long* __restrict__ tile_desc,
+                                  unsigned long long* __restrict__ ticket,
+                                  unsigned long long* __restrict__ total_out,
+                                  unsigned long long* __restrict__ error_out) {
+    __shared__ int64_t stage[FILT_TILE];
+    __shared__ uint32_t chunk_base[1];
+    __shared__ uint64_t sh_tile;
+    __shared__ unsigned long long sh_excl;
+    const uint64_t n_tiles = (n + FILT_TILE - 1) / FILT_TILE;
+    int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+    __shared__ uint32_t wbase[BLOCK / WAVE + 1];
+    for (;;) {
+        if (threadIdx.x == 0) sh_tile = atomicAdd(ticket, 1ull);
+        __syncthreads();
+        uint64_t t = sh_tile;
+        if (t >= n_tiles) return;
+        uint64_t lo = t * FILT_TILE;
+        uint64_t hi = min(lo + FILT_TILE, n);
+        // phase 1: ballot-compact passing values into LDS, preserving order
+        if (threadIdx.x == 0) chunk_base[0] = 0;
+        __syncthreads();
+        for (uint64_t base = lo; base < hi; base += blockDim.x) {
+            uint64_t i = base + threadIdx.x;
+            int64_t v = (i < hi) ? in[i] : 0;
+            bool pred = (i < hi) && (v < theta);
+            uint64_t mask = __ballot(pred);
+            uint32_t wcount = __popcll(mask);
+            if (lane == 0) wbase[wid] = wcount;
+            __syncthreads();
+            if (threadIdx.x == 0) {
+                uint32_t acc = chunk_base[0];
+                for (int w = 0; w < BLOCK / WAVE; w++) {
+                    uint32_t c = wbase[w];
+                    wbase[w] = acc;
+                    acc += c;
+                }
+                chunk_base[0] = acc;
+            }
+            __syncthreads();
+            if (pred)
+                stage[wbase[wid] + __popcll(mask & ((1ull << lane) - 1))] = v;
+            __syncthreads();
+        }
+        uint32_t tile_count = chunk_base[0];
+        // phase 2 (thread 0): publish aggregate, look back, publish prefix
+        if (threadIdx.x == 0) {
+            if (t == 0) {
+                __hip_atomic_store(&tile_desc[0], FILT_PREFIX | tile_count,
+                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                sh_excl = 0;
+                if (t == n_tiles - 1) *total_out = tile_count;
+            } else {
+                __hip_atomic_store(&tile_desc[t], FILT_AGG | tile_count,
+                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                unsigned long long running = 0;
+                int64_t j = (int64_t)t - 1;
+                uint64_t spins = 0;
+                while (j >= 0) {
+                    unsigned long long d = __hip_atomic_load(&tile_desc[j], __ATOMIC_RELAXED,
+                                                             __HIP_MEMORY_SCOPE_AGENT);
+                    unsigned long long flag = d & ~FILT_CNT_MASK;
+                    if (flag == FILT_PREFIX) {
+                        running += d & FILT_CNT_MASK;
+                        break;
+                    }
+                    if (flag == FILT_AGG) {
+                        running += d & FILT_CNT_MASK;
+                        j--;
+                        continue;
+                    }
+                    if (++spins > (1ull << 31)) { // bounded: flag error, bail out
+                        atomicOr(error_out, 1ull);
+                        running = 0;
+                        break;
+                    }
+                }
+                __hip_atomic_store(&tile_desc[t], FILT_PREFIX | (running + tile_count),
+                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                sh_excl = running;
+                if (t == n_tiles - 1) *total_out = running + tile_count;
+            }
+        }
+        __syncthreads();
+        // phase 3: coalesced write of the staged values at the tile's offset
+        uint64_t off = sh_excl;
+        for (uint32_t j = threadIdx.x; j < tile_count; j += blockDim.x)
+            out[off + j] = stage[j];
+        __syncthreads();
+    }
+}
+
+extern "C" int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n,
+                                          int64_t theta, gpue_dbuf* out, uint64_t* out_count);
+int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64_t theta,
+                               gpue_dbuf* out, uint64_t* out_count) {
+    ARG_CHECK(s && in && out && out_count && in->bytes >= n * 8 && n > 0);
+    uint64_t n_tiles = (n + FILT_TILE - 1) / FILT_TILE;
+    unsigned long long* d_desc = nullptr;
+    unsigned long long* d_misc = nullptr; // {ticket, total, error}
+    HIP_CHECK(hipMalloc(&d_desc, n_tiles * 8));
+    HIP_CHECK(hipMalloc(&d_misc, 3 * 8));
+    HIP_CHECK(hipMemsetAsync(d_desc, 0, n_tiles * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_misc, 0, 3 * 8, s->stream));
+    uint32_t nb = (uint32_t)(n_tiles < 2048 ? n_tiles : 2048);
+    hipLaunchKernelGGL(k_filter_lookback, dim3(nb), dim3(BLOCK), 0, s->stream,
+                       (const int64_t*)in->ptr, n, theta, (int64_t*)out->ptr, d_desc,
+                       d_misc, d_misc + 1, d_misc + 2);
+    unsigned long long h_misc[3];
+    HIP_CHECK(hipMemcpyAsync(h_misc, d_misc, 3 * 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_desc);
+    (void)hipFree(d_misc);
+    if (h_misc[2]) {
+        snprintf(g_err, sizeof(g_err), "filter lookback spin timeout (predecessor stuck)");
+        return GPUE_ERR_HIP;
+    }
+    *out_count = h_misc[1];
+    return GPUE_OK;
+}
+
 int gpue_scan_filter_i64_lt(gpue_session* s, gpue_dbuf* in, uint64_t n, int64_t theta,
                             gpue_dbuf* out, uint64_t* out_count) {
     ARG_CHECK(s && in && out && out_count && in->bytes >= n * 8);

@@ -69,6 +69,7 @@ def _declare(lib):
         "gpue_gen_lineorder_q21": (c_i32, [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp, c_vp, c_vp]),
         "gpue_gen_lineorder_q43": (c_i32, [c_vp, c_u64, c_u64, c_u64] + [c_vp] * 6),
         "gpue_scan_filter_i64_lt": (c_i32, [c_vp, c_vp, c_u64, c_i64, c_vp, ctypes.POINTER(c_u64)]),
+        "gpue_scan_filter_i64_lt_sp": (c_i32, [c_vp, c_vp, c_u64, c_i64, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_join_build_payload_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_build_range_direct_i32": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_build_bucket_chained_u32": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
@@ -223,6 +224,12 @@ class Engine:
         cnt = c_u64()
         _ck(self._lib, self._lib.gpue_scan_filter_i64_lt(self._h, inp._h, n, theta, out._h,
                                                          ctypes.byref(cnt)))
+        return cnt.value
+
+    def scan_filter_i64_lt_sp(self, inp: DBuf, n, theta, out: DBuf) -> int:
+        cnt = c_u64()
+        _ck(self._lib, self._lib.gpue_scan_filter_i64_lt_sp(self._h, inp._h, n, theta,
+                                                            out._h, ctypes.byref(cnt)))
         return cnt.value
 
     def join_build_payload(self, keys: DBuf, payloads: DBuf, n_rows) -> JoinTable:

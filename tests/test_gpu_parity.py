@@ -66,6 +66,26 @@ def test_filter_parity_bit_exact_ordered(engine, n, sel):
     out.free()
 
 
+@pytest.mark.parametrize("n,sel", [(1_000_000, 0.1), (10_000_000, 0.5), (999_999, 0.01),
+                                   (50_000_000, 0.3), (4096, 1.0), (1000, 0.0)])
+def test_filter_singlepass_parity(engine, n, sel):
+    """Single-pass decoupled-lookback compaction: bit-exact INCLUDING order
+    vs the oracle (and therefore vs the two-pass kernel)."""
+    inp = engine.alloc(n * 8)
+    engine.gen_i64(inp, SEED, 3, 0, n)
+    data = inp.d2h(np.int64, n)
+    theta = int(np.quantile(data.astype(np.float64), sel)) if 0 < sel < 1 else (
+        np.iinfo(np.int64).max if sel >= 1 else np.iinfo(np.int64).min)
+    out = engine.alloc(n * 8)
+    cnt = engine.scan_filter_i64_lt_sp(inp, n, theta, out)
+    expect = orc.filter_i64_lt(data, theta)
+    assert cnt == len(expect)
+    got = out.d2h(np.int64, cnt) if cnt else np.empty(0, np.int64)
+    assert np.array_equal(got, expect)
+    inp.free()
+    out.free()
+
+
 def test_join_build_probe_emit_parity(engine):
     """Range-direct build + probe emit vs oracle: multiset of (probe,build)."""
     rng = np.random.default_rng(5)
