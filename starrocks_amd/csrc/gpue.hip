@@ -506,53 +506,63 @@ __global__ void k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, in
                                   unsigned long long* __restrict__ ticket,
                                   unsigned long long* __restrict__ total_out,
                                   unsigned long long* __restrict__ error_out) {
-    __shared__ int64_t stage[FILT_TILE];
-    __shared__ uint32_t chunk_base[1];
+    // thread t owns the ITEMS consecutive rows [lo + 8t, lo + 8t + 8): output
+    // order == thread order == row order; ONE block scan per tile.
     __shared__ uint64_t sh_tile;
     __shared__ unsigned long long sh_excl;
+    __shared__ uint32_t wsum[BLOCK / WAVE];
     const uint64_t n_tiles = (n + FILT_TILE - 1) / FILT_TILE;
     int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
-    __shared__ uint32_t wbase[BLOCK / WAVE + 1];
     for (;;) {
         if (threadIdx.x == 0) sh_tile = atomicAdd(ticket, 1ull);
         __syncthreads();
         uint64_t t = sh_tile;
         if (t >= n_tiles) return;
         uint64_t lo = t * FILT_TILE;
-        uint64_t hi = min(lo + FILT_TILE, n);
-        // phase 1: ballot-compact passing values into LDS, preserving order
-        if (threadIdx.x == 0) chunk_base[0] = 0;
-        __syncthreads();
-        for (uint64_t base = lo; base < hi; base += blockDim.x) {
-            uint64_t i = base + threadIdx.x;
-            int64_t v = (i < hi) ? in[i] : 0;
-            bool pred = (i < hi) && (v < theta);
-            uint64_t mask = __ballot(pred);
-            uint32_t wcount = __popcll(mask);
-            if (lane == 0) wbase[wid] = wcount;
-            __syncthreads();
-            if (threadIdx.x == 0) {
-                uint32_t acc = chunk_base[0];
-                for (int w = 0; w < BLOCK / WAVE; w++) {
-                    uint32_t c = wbase[w];
-                    wbase[w] = acc;
-                    acc += c;
-                }
-                chunk_base[0] = acc;
+        uint64_t hi = min(lo + (uint64_t)FILT_TILE, n);
+        uint64_t my = lo + (uint64_t)threadIdx.x * FILT_ITEMS;
+        int64_t v[FILT_ITEMS];
+        uint32_t pm = 0; // predicate bitmask over my 8 rows
+        if (my + FILT_ITEMS <= hi) {
+            const longlong2* p2 = (const longlong2*)(in + my);
+            #pragma unroll
+            for (int j = 0; j < (int)FILT_ITEMS / 2; j++) {
+                longlong2 w = p2[j];
+                v[2 * j] = w.x;
+                v[2 * j + 1] = w.y;
             }
-            __syncthreads();
-            if (pred)
-                stage[wbase[wid] + __popcll(mask & ((1ull << lane) - 1))] = v;
-            __syncthreads();
+            #pragma unroll
+            for (int j = 0; j < (int)FILT_ITEMS; j++) pm |= (v[j] < theta) << j;
+        } else {
+            for (int j = 0; j < (int)FILT_ITEMS; j++) {
+                uint64_t i = my + j;
+                v[j] = (i < hi) ? in[i] : 0;
+                pm |= ((i < hi) && (v[j] < theta)) << j;
+            }
         }
-        uint32_t tile_count = chunk_base[0];
-        // phase 2 (thread 0): publish aggregate, look back, publish prefix
+        uint32_t c = __popc(pm);
+        // block exclusive scan of per-thread counts: wave shfl-scan + wave
+        // totals combined by thread 0
+        uint32_t pre = c;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            uint32_t up = __shfl_up(pre, off, WAVE);
+            if (lane >= off) pre += up;
+        }
+        uint32_t wave_total = __shfl(pre, WAVE - 1, WAVE);
+        uint32_t my_excl = pre - c;
+        if (lane == WAVE - 1) wsum[wid] = pre; // inclusive wave total
+        __syncthreads();
+        uint32_t wave_base = 0;
+        for (int w = 0; w < wid; w++) wave_base += wsum[w];
+        uint32_t tile_count = 0;
+        for (int w = 0; w < BLOCK / WAVE; w++) tile_count += wsum[w];
+        (void)wave_total;
+        // lookback (thread 0)
         if (threadIdx.x == 0) {
             if (t == 0) {
                 __hip_atomic_store(&tile_desc[0], FILT_PREFIX | tile_count,
                                    __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                 sh_excl = 0;
-                if (t == n_tiles - 1) *total_out = tile_count;
             } else {
                 __hip_atomic_store(&tile_desc[t], FILT_AGG | tile_count,
                                    __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
@@ -581,14 +591,14 @@ __global__ void k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, in
                 __hip_atomic_store(&tile_desc[t], FILT_PREFIX | (running + tile_count),
                                    __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                 sh_excl = running;
-                if (t == n_tiles - 1) *total_out = running + tile_count;
             }
+            if (t == n_tiles - 1) *total_out = sh_excl + tile_count;
         }
         __syncthreads();
-        // phase 3: coalesced write of the staged values at the tile's offset
-        uint64_t off = sh_excl;
-        for (uint32_t j = threadIdx.x; j < tile_count; j += blockDim.x)
-            out[off + j] = stage[j];
+        uint64_t w = sh_excl + wave_base + my_excl;
+        #pragma unroll
+        for (int j = 0; j < (int)FILT_ITEMS; j++)
+            if (pm & (1u << j)) out[w++] = v[j];
         __syncthreads();
     }
 }
