@@ -513,11 +513,15 @@ __global__ void k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, in
     __shared__ uint32_t wsum[BLOCK / WAVE];
     const uint64_t n_tiles = (n + FILT_TILE - 1) / FILT_TILE;
     int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
-    for (;;) {
-        if (threadIdx.x == 0) sh_tile = atomicAdd(ticket, 1ull);
-        __syncthreads();
-        uint64_t t = sh_tile;
-        if (t >= n_tiles) return;
+    (void)ticket;
+    (void)sh_tile;
+    // static ascending tiles: the host sizes the grid to be FULLY RESIDENT
+    // (occupancy-queried), so tile t's predecessor is always executing on
+    // another resident block — no ticket atomic (a single ticket address
+    // serializes ~12 ns × n_tiles: measured 6 ms of a 7 ms launch at 488 K
+    // tiles). The bounded spin turns any residency miscount into an error
+    // return instead of a hang.
+    for (uint64_t t = blockIdx.x; t < n_tiles; t += gridDim.x) {
         uint64_t lo = t * FILT_TILE;
         uint64_t hi = min(lo + (uint64_t)FILT_TILE, n);
         uint64_t my = lo + (uint64_t)threadIdx.x * FILT_ITEMS;
@@ -610,12 +614,23 @@ int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64
     ARG_CHECK(s && in && out && out_count && in->bytes >= n * 8 && n > 0);
     uint64_t n_tiles = (n + FILT_TILE - 1) / FILT_TILE;
     unsigned long long* d_desc = nullptr;
-    unsigned long long* d_misc = nullptr; // {ticket, total, error}
+    unsigned long long* d_misc = nullptr; // {ticket(unused), total, error}
     HIP_CHECK(hipMalloc(&d_desc, n_tiles * 8));
     HIP_CHECK(hipMalloc(&d_misc, 3 * 8));
     HIP_CHECK(hipMemsetAsync(d_desc, 0, n_tiles * 8, s->stream));
     HIP_CHECK(hipMemsetAsync(d_misc, 0, 3 * 8, s->stream));
-    uint32_t nb = (uint32_t)(n_tiles < 2048 ? n_tiles : 2048);
+    // fully-resident grid for the static-assignment lookback (see kernel
+    // comment); the occupancy API can over-report by one block per CU on
+    // SGPR-heavy 256-thread kernels (MI355X_MICROARCH.md) — subtract one
+    int blocks_per_cu = 0;
+    HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(&blocks_per_cu,
+                                                           k_filter_lookback, BLOCK, 0));
+    if (blocks_per_cu > 1) blocks_per_cu -= 1;
+    if (blocks_per_cu < 1) blocks_per_cu = 1;
+    hipDeviceProp_t props;
+    HIP_CHECK(hipGetDeviceProperties(&props, s->device));
+    uint64_t resident = (uint64_t)props.multiProcessorCount * blocks_per_cu;
+    uint32_t nb = (uint32_t)(n_tiles < resident ? n_tiles : resident);
     hipLaunchKernelGGL(k_filter_lookback, dim3(nb), dim3(BLOCK), 0, s->stream,
                        (const int64_t*)in->ptr, n, theta, (int64_t*)out->ptr, d_desc,
                        d_misc, d_misc + 1, d_misc + 2);
