@@ -561,42 +561,64 @@ __global__ void k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, in
         uint32_t tile_count = 0;
         for (int w = 0; w < BLOCK / WAVE; w++) tile_count += wsum[w];
         (void)wave_total;
-        // lookback (thread 0)
-        if (threadIdx.x == 0) {
+        // WAVE-PARALLEL lookback (wave 0): 64 predecessor descriptors per
+        // step — a serial walk scans O(grid) aggregates per tile and
+        // measured 7.5 ms at 488 K tiles
+        if (wid == 0) {
             if (t == 0) {
-                __hip_atomic_store(&tile_desc[0], FILT_PREFIX | tile_count,
-                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                sh_excl = 0;
+                if (lane == 0) {
+                    __hip_atomic_store(&tile_desc[0], FILT_PREFIX | tile_count,
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    sh_excl = 0;
+                }
             } else {
-                __hip_atomic_store(&tile_desc[t], FILT_AGG | tile_count,
-                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                if (lane == 0)
+                    __hip_atomic_store(&tile_desc[t], FILT_AGG | tile_count,
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
                 unsigned long long running = 0;
-                int64_t j = (int64_t)t - 1;
+                int64_t base = (int64_t)t - 1; // lane l inspects base - l
                 uint64_t spins = 0;
-                while (j >= 0) {
-                    unsigned long long d = __hip_atomic_load(&tile_desc[j], __ATOMIC_RELAXED,
-                                                             __HIP_MEMORY_SCOPE_AGENT);
+                for (;;) {
+                    int64_t idx = base - lane;
+                    // before tile 0: a virtual PREFIX of 0
+                    unsigned long long d =
+                        idx >= 0 ? __hip_atomic_load(&tile_desc[idx], __ATOMIC_RELAXED,
+                                                     __HIP_MEMORY_SCOPE_AGENT)
+                                 : FILT_PREFIX;
                     unsigned long long flag = d & ~FILT_CNT_MASK;
-                    if (flag == FILT_PREFIX) {
-                        running += d & FILT_CNT_MASK;
+                    uint64_t prefix_mask = __ballot(flag == FILT_PREFIX);
+                    uint64_t invalid_mask = __ballot(flag == 0);
+                    int first_prefix = prefix_mask ? (__ffsll(prefix_mask) - 1) : WAVE;
+                    int first_invalid = invalid_mask ? (__ffsll(invalid_mask) - 1) : WAVE;
+                    if (first_prefix < first_invalid) {
+                        unsigned long long contrib =
+                            (lane <= first_prefix) ? (d & FILT_CNT_MASK) : 0;
+                        for (int off = WAVE / 2; off > 0; off >>= 1)
+                            contrib += __shfl_down(contrib, off, WAVE);
+                        running += __shfl(contrib, 0, WAVE);
                         break;
                     }
-                    if (flag == FILT_AGG) {
-                        running += d & FILT_CNT_MASK;
-                        j--;
+                    if (first_invalid == WAVE) { // full window of aggregates
+                        unsigned long long contrib = d & FILT_CNT_MASK;
+                        for (int off = WAVE / 2; off > 0; off >>= 1)
+                            contrib += __shfl_down(contrib, off, WAVE);
+                        running += __shfl(contrib, 0, WAVE);
+                        base -= WAVE;
                         continue;
                     }
-                    if (++spins > (1ull << 31)) { // bounded: flag error, bail out
-                        atomicOr(error_out, 1ull);
+                    if (++spins > (1ull << 28)) { // bounded: flag error, bail out
+                        if (lane == 0) atomicOr(error_out, 1ull);
                         running = 0;
                         break;
                     }
                 }
-                __hip_atomic_store(&tile_desc[t], FILT_PREFIX | (running + tile_count),
-                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                sh_excl = running;
+                if (lane == 0) {
+                    __hip_atomic_store(&tile_desc[t], FILT_PREFIX | (running + tile_count),
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    sh_excl = running;
+                }
             }
-            if (t == n_tiles - 1) *total_out = sh_excl + tile_count;
+            if (lane == 0 && t == n_tiles - 1) *total_out = sh_excl + tile_count;
         }
         __syncthreads();
         uint64_t w = sh_excl + wave_base + my_excl;
