@@ -588,8 +588,8 @@ __global__ void k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, in
                     unsigned long long flag = d & ~FILT_CNT_MASK;
                     uint64_t prefix_mask = __ballot(flag == FILT_PREFIX);
                     uint64_t invalid_mask = __ballot(flag == 0);
-                    int first_prefix = prefix_mask ? (__ffsll(prefix_mask) - 1) : WAVE;
-                    int first_invalid = invalid_mask ? (__ffsll(invalid_mask) - 1) : WAVE;
+                    int first_prefix = prefix_mask ? (__ffsll((unsigned long long)prefix_mask) - 1) : WAVE;
+                    int first_invalid = invalid_mask ? (__ffsll((unsigned long long)invalid_mask) - 1) : WAVE;
                     if (first_prefix < first_invalid) {
                         unsigned long long contrib =
                             (lane <= first_prefix) ? (d & FILT_CNT_MASK) : 0;
