@@ -254,6 +254,15 @@ int gpue_gather_u64(gpue_session* s, gpue_dbuf* in, gpue_dbuf* idx, uint64_t n,
 int gpue_partition_i64(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t num_channels,
                        uint64_t* start_points_out, gpue_dbuf* row_indexes_out);
 
+/* ---- storage ingress (the step upstream of the scan) ----
+ * Decode the reference's numeric page body (bitshuffle+LZ4,
+ * be/src/storage/rowset/bitshuffle_page.h framing after the 16-byte header;
+ * bitshuffle 0.5.1 published algorithm + LZ4 block format restated in
+ * oracle/oracle.c): n_values int32 (multiple of 8) from the device-resident
+ * page bytes. */
+int gpue_page_decode_bshuf_lz4_i32(gpue_session* s, gpue_dbuf* page, uint32_t n_values,
+                                   gpue_dbuf* out);
+
 /* ---- TopN ----
  * ORDER BY value DESC LIMIT k (reference exec/chunks_sorter_topn.cpp):
  * deterministic (value, key) lexicographic descending; k <= 16. */

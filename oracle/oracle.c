@@ -1134,4 +1134,202 @@ void orc_q21_kernel(const int32_t* pk, const int32_t* sk, const int32_t* od,
     free(partials);
 }
 
+/* ====================================================================== */
+/* Storage ingress: the reference's numeric column
+ * pages are bitshuffle+LZ4 (be/src/storage/rowset/bitshuffle_page.h: 16-byte
+ * header {num_elements, compressed_size, padded_num_elements, elem_bytes},
+ * then bshuf_compress_lz4 of the padded values). The bitshuffle library
+ * (pinned 0.5.1, thirdparty/patches/bitshuffle-0.5.1.patch) is
+ * download-script-only and ABSENT offline, so this restates its PUBLISHED
+ * algorithms:
+ *  - bshuf_trans_bit_elem: per block of block_elems elements (default for
+ *    4-byte elements: 8192 B / 4 rounded to a multiple of 8 = 2048), the
+ *    data becomes bit-plane-major — plane (j,i) (byte j of the element,
+ *    bit i of that byte) is block_elems/8 bytes; bit (e%8) of its byte
+ *    (e/8) is bit i of byte j of element e.
+ *  - bshuf_compress_lz4 framing: per block, a 4-byte BIG-ENDIAN compressed
+ *    length, then one LZ4 block (LZ4 block format spec: token, literal
+ *    lengths with 255-extensions, 2-byte little-endian match offsets,
+ *    matches >= 4 with 255-extended lengths, literals-only tail >= 5).
+ * Byte-compatibility with the upstream binaries cannot be verified offline
+ * (DESIGN.md notes this as the one unpinned format); encode+decode are
+ * restated together and the GPU decode is parity-tested against them.     */
+/* ====================================================================== */
+
+#define BSHUF_BLOCK_I32 2048u /* default block for elem_size 4 */
+
+/* bit-plane transpose of one block (elem_size 4) */
+static void bshuf_transpose_block_i32(const uint32_t* in, uint8_t* out, uint32_t elems) {
+    uint32_t plane_bytes = elems / 8;
+    memset(out, 0, (size_t)plane_bytes * 32);
+    for (uint32_t e = 0; e < elems; e++) {
+        uint32_t v = in[e];
+        for (uint32_t j = 0; j < 4; j++) {
+            uint8_t byte = (uint8_t)(v >> (8 * j));
+            for (uint32_t i = 0; i < 8; i++) {
+                if (byte & (1u << i))
+                    out[(size_t)(j * 8 + i) * plane_bytes + e / 8] |= 1u << (e % 8);
+            }
+        }
+    }
+}
+
+static void bshuf_untranspose_block_i32(const uint8_t* in, uint32_t* out, uint32_t elems) {
+    uint32_t plane_bytes = elems / 8;
+    for (uint32_t e = 0; e < elems; e++) {
+        uint32_t v = 0;
+        for (uint32_t j = 0; j < 4; j++) {
+            for (uint32_t i = 0; i < 8; i++) {
+                uint32_t bit = (in[(size_t)(j * 8 + i) * plane_bytes + e / 8] >> (e % 8)) & 1u;
+                v |= bit << (8 * j + i);
+            }
+        }
+        out[e] = v;
+    }
+}
+
+/* minimal LZ4 block compressor (format-correct greedy hash matcher) */
+static size_t lz4_compress_block(const uint8_t* src, size_t n, uint8_t* dst) {
+    enum { HASH_LOG = 13, MINMATCH = 4 };
+    static uint32_t table[1 << HASH_LOG];
+    memset(table, 0, sizeof(table));
+    size_t s = 0, d = 0, anchor = 0;
+    while (n >= MINMATCH + 1 && s + MINMATCH + 5 < n) { /* keep the >=5-literal tail rule */
+        uint32_t seq;
+        memcpy(&seq, src + s, 4);
+        uint32_t h = (seq * 2654435761u) >> (32 - HASH_LOG);
+        size_t cand = table[h];
+        table[h] = (uint32_t)s;
+        uint32_t cseq = 0;
+        if (s > 0 && cand < s && s - cand <= 65535) memcpy(&cseq, src + cand, 4);
+        if (s > 0 && cand < s && s - cand <= 65535 && cseq == seq) {
+            /* extend match, but leave >= 5 literals + last-5-within rules */
+            size_t limit = n - 5;
+            size_t len = 4;
+            while (s + len < limit && src[cand + len] == src[s + len] && len < 0xFFFF) len++;
+            size_t lit = s - anchor;
+            /* token */
+            size_t tok = d++;
+            uint8_t t = 0;
+            if (lit >= 15) {
+                t |= 0xF0;
+                size_t rem = lit - 15;
+                while (rem >= 255) { dst[d++] = 255; rem -= 255; }
+                dst[d++] = (uint8_t)rem;
+            } else {
+                t |= (uint8_t)(lit << 4);
+            }
+            memcpy(dst + d, src + anchor, lit);
+            d += lit;
+            uint16_t off = (uint16_t)(s - cand);
+            dst[d++] = (uint8_t)off;
+            dst[d++] = (uint8_t)(off >> 8);
+            size_t mlen = len - MINMATCH;
+            if (mlen >= 15) {
+                t |= 0x0F;
+                size_t rem = mlen - 15;
+                while (rem >= 255) { dst[d++] = 255; rem -= 255; }
+                dst[d++] = (uint8_t)rem;
+            } else {
+                t |= (uint8_t)mlen;
+            }
+            dst[tok] = t;
+            s += len;
+            anchor = s;
+        } else {
+            s++;
+        }
+    }
+    /* final literals-only sequence */
+    size_t lit = n - anchor;
+    size_t tok = d++;
+    uint8_t t = 0;
+    if (lit >= 15) {
+        t = 0xF0;
+        size_t rem = lit - 15;
+        while (rem >= 255) { dst[d++] = 255; rem -= 255; }
+        dst[d++] = (uint8_t)rem;
+    } else {
+        t = (uint8_t)(lit << 4);
+    }
+    dst[tok] = t;
+    memcpy(dst + d, src + anchor, lit);
+    d += lit;
+    return d;
+}
+
+static size_t lz4_decompress_block(const uint8_t* src, size_t comp_n, uint8_t* dst,
+                                   size_t dst_cap) {
+    size_t s = 0, d = 0;
+    while (s < comp_n) {
+        uint8_t tok = src[s++];
+        size_t lit = tok >> 4;
+        if (lit == 15) {
+            uint8_t b;
+            do { b = src[s++]; lit += b; } while (b == 255);
+        }
+        if (d + lit > dst_cap) return 0;
+        memcpy(dst + d, src + s, lit);
+        s += lit;
+        d += lit;
+        if (s >= comp_n) break; /* last sequence has no match */
+        uint16_t off = (uint16_t)(src[s] | (src[s + 1] << 8));
+        s += 2;
+        size_t mlen = (tok & 0xF);
+        if (mlen == 15) {
+            uint8_t b;
+            do { b = src[s++]; mlen += b; } while (b == 255);
+        }
+        mlen += 4;
+        if (d + mlen > dst_cap || off > d) return 0;
+        for (size_t k = 0; k < mlen; k++) { dst[d] = dst[d - off]; d++; } /* overlap ok */
+    }
+    return d;
+}
+
+/* encode n (multiple of 8) int32 values into the page body
+ * (bshuf_compress_lz4 framing). Returns encoded bytes; out must hold
+ * n*4 + 16*nblocks slack. block_starts (optional, nblocks+1 entries) gets
+ * each block's byte offset for the GPU's parallel decode. */
+uint64_t orc_bshuf_lz4_encode_i32(const int32_t* values, uint32_t n, uint8_t* out,
+                                  uint32_t* block_starts) {
+    uint8_t tmp[BSHUF_BLOCK_I32 * 4];
+    uint64_t d = 0;
+    uint32_t nb = 0;
+    for (uint32_t b = 0; b < n; b += BSHUF_BLOCK_I32) {
+        uint32_t elems = n - b < BSHUF_BLOCK_I32 ? n - b : BSHUF_BLOCK_I32;
+        if (block_starts) block_starts[nb] = (uint32_t)d;
+        bshuf_transpose_block_i32((const uint32_t*)values + b, tmp, elems);
+        uint64_t lenpos = d;
+        d += 4;
+        size_t c = lz4_compress_block(tmp, (size_t)elems * 4, out + d);
+        d += c;
+        /* 4-byte BIG-endian compressed length (bshuf_write_uint32_BE) */
+        out[lenpos] = (uint8_t)(c >> 24);
+        out[lenpos + 1] = (uint8_t)(c >> 16);
+        out[lenpos + 2] = (uint8_t)(c >> 8);
+        out[lenpos + 3] = (uint8_t)c;
+        nb++;
+    }
+    if (block_starts) block_starts[nb] = (uint32_t)d;
+    return d;
+}
+
+/* decode the page body back to n int32 values; returns bytes consumed */
+uint64_t orc_bshuf_lz4_decode_i32(const uint8_t* in, uint32_t n, int32_t* values) {
+    uint8_t tmp[BSHUF_BLOCK_I32 * 4];
+    uint64_t s = 0;
+    for (uint32_t b = 0; b < n; b += BSHUF_BLOCK_I32) {
+        uint32_t elems = n - b < BSHUF_BLOCK_I32 ? n - b : BSHUF_BLOCK_I32;
+        uint32_t c = ((uint32_t)in[s] << 24) | ((uint32_t)in[s + 1] << 16) |
+                     ((uint32_t)in[s + 2] << 8) | in[s + 3];
+        s += 4;
+        size_t got = lz4_decompress_block(in + s, c, tmp, sizeof(tmp));
+        if (got != (size_t)elems * 4) return 0;
+        s += c;
+        bshuf_untranspose_block_i32(tmp, (uint32_t*)values + b, elems);
+    }
+    return s;
+}
+
 void orc_free(void* p) { free(p); }

@@ -193,6 +193,12 @@ void orc_q21_kernel(const int32_t* pk, const int32_t* sk, const int32_t* od,
                     const uint32_t* sfirst, const uint32_t* dfirst, int64_t dmin,
                     int threads, int64_t* group_sums);
 
+/* storage ingress: bitshuffle(0.5.1 published algorithm)+LZ4(block spec)
+ * page body encode/decode for int32 */
+uint64_t orc_bshuf_lz4_encode_i32(const int32_t* values, uint32_t n, uint8_t* out,
+                                  uint32_t* block_starts);
+uint64_t orc_bshuf_lz4_decode_i32(const uint8_t* in, uint32_t n, int32_t* values);
+
 void orc_free(void* p);
 
 #ifdef __cplusplus

@@ -246,6 +246,25 @@ def q3_pipeline(seed, row_start, n, n_orders, n_custs, seg=1,
     return ok[:g][order].copy(), os_[:g][order].copy()
 
 
+def bshuf_lz4_encode_i32(values: np.ndarray):
+    lib = load()
+    lib.orc_bshuf_lz4_encode_i32.restype = c_u64
+    lib.orc_bshuf_lz4_encode_i32.argtypes = [c_vp, c_u32, c_vp, c_vp]
+    out = np.zeros(len(values) * 4 + 16 * (len(values) // 2048 + 2), np.uint8)
+    n = lib.orc_bshuf_lz4_encode_i32(_p(values), len(values), _p(out), None)
+    return out[:n].copy()
+
+
+def bshuf_lz4_decode_i32(page: np.ndarray, n: int):
+    lib = load()
+    lib.orc_bshuf_lz4_decode_i32.restype = c_u64
+    lib.orc_bshuf_lz4_decode_i32.argtypes = [c_vp, c_u32, c_vp]
+    vals = np.zeros(n, np.int32)
+    used = lib.orc_bshuf_lz4_decode_i32(_p(page), n, _p(vals))
+    assert used == len(page), (used, len(page))
+    return vals
+
+
 def hash_agg_stats(keys, vals):
     cap = max(len(keys), 16)
     o = [np.empty(cap, np.uint64)] + [np.empty(cap, np.int64) for _ in range(4)]

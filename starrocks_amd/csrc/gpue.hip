@@ -2824,6 +2824,128 @@ int gpue_gather_u32(gpue_session* s, gpue_dbuf* in, gpue_dbuf* idx, uint64_t n,
 }
 
 // ---------------------------------------------------------------------------
+// Storage ingress: decode the reference's numeric page
+// body — bshuf_compress_lz4 framing (per block: 4-byte big-endian length +
+// one LZ4 block of the bit-plane-transposed values; bitshuffle 0.5.1
+// published algorithm, block = 2048 int32 elements). Three kernels:
+// block-starts scan (serial walk of the lengths), per-block LZ4 decode (one
+// thread per independent block — thousands of blocks in flight), and the
+// bit-plane un-transpose (grid-stride per element). Formats restated in
+// oracle/oracle.c; parity GPU <-> oracle (byte-compat caveat in DESIGN.md).
+// ---------------------------------------------------------------------------
+static constexpr uint32_t BSHUF_BLOCK_I32 = 2048;
+
+__global__ void k_page_block_starts(const uint8_t* __restrict__ page, uint32_t n_blocks,
+                                    uint32_t* __restrict__ starts) {
+    if (threadIdx.x == 0 && blockIdx.x == 0) {
+        uint32_t off = 0;
+        for (uint32_t b = 0; b < n_blocks; b++) {
+            starts[b] = off;
+            uint32_t c = ((uint32_t)page[off] << 24) | ((uint32_t)page[off + 1] << 16) |
+                         ((uint32_t)page[off + 2] << 8) | page[off + 3];
+            off += 4 + c;
+        }
+        starts[n_blocks] = off;
+    }
+}
+
+__global__ void k_lz4_decode_blocks(const uint8_t* __restrict__ page,
+                                    const uint32_t* __restrict__ starts, uint32_t n_blocks,
+                                    uint32_t n_values, uint8_t* __restrict__ scratch,
+                                    unsigned long long* __restrict__ error_out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t b = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; b < n_blocks;
+         b += stride) {
+        uint32_t elems = min(n_values - (uint32_t)b * BSHUF_BLOCK_I32, BSHUF_BLOCK_I32);
+        uint32_t s = starts[b];
+        uint32_t comp_n = ((uint32_t)page[s] << 24) | ((uint32_t)page[s + 1] << 16) |
+                          ((uint32_t)page[s + 2] << 8) | page[s + 3];
+        const uint8_t* src = page + s + 4;
+        uint8_t* dst = scratch + (uint64_t)b * BSHUF_BLOCK_I32 * 4;
+        uint64_t dst_cap = (uint64_t)elems * 4;
+        uint64_t si = 0, d = 0;
+        while (si < comp_n) {
+            uint8_t tok = src[si++];
+            uint64_t lit = tok >> 4;
+            if (lit == 15) {
+                uint8_t x;
+                do { x = src[si++]; lit += x; } while (x == 255);
+            }
+            if (d + lit > dst_cap) { atomicOr(error_out, 1ull); return; }
+            for (uint64_t k = 0; k < lit; k++) dst[d + k] = src[si + k];
+            si += lit;
+            d += lit;
+            if (si >= comp_n) break;
+            uint32_t off = src[si] | ((uint32_t)src[si + 1] << 8);
+            si += 2;
+            uint64_t mlen = tok & 0xF;
+            if (mlen == 15) {
+                uint8_t x;
+                do { x = src[si++]; mlen += x; } while (x == 255);
+            }
+            mlen += 4;
+            if (d + mlen > dst_cap || off > d) { atomicOr(error_out, 1ull); return; }
+            for (uint64_t k = 0; k < mlen; k++) { dst[d] = dst[d - off]; d++; }
+        }
+        if (d != dst_cap) atomicOr(error_out, 1ull);
+    }
+}
+
+__global__ void k_bshuf_untranspose_i32(const uint8_t* __restrict__ scratch, uint32_t n_values,
+                                        int32_t* __restrict__ out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t e = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; e < n_values;
+         e += stride) {
+        uint32_t b = (uint32_t)(e / BSHUF_BLOCK_I32);
+        uint32_t le = (uint32_t)(e % BSHUF_BLOCK_I32);
+        uint32_t elems = min(n_values - b * BSHUF_BLOCK_I32, BSHUF_BLOCK_I32);
+        const uint8_t* blk = scratch + (uint64_t)b * BSHUF_BLOCK_I32 * 4;
+        uint32_t plane_bytes = elems / 8;
+        uint32_t v = 0;
+        #pragma unroll
+        for (uint32_t p = 0; p < 32; p++) {
+            uint32_t bit = (blk[(uint64_t)p * plane_bytes + le / 8] >> (le % 8)) & 1u;
+            v |= bit << ((p / 8) * 8 + (p % 8));
+        }
+        out[e] = (int32_t)v;
+    }
+}
+
+extern "C" int gpue_page_decode_bshuf_lz4_i32(gpue_session* s, gpue_dbuf* page,
+                                              uint32_t n_values, gpue_dbuf* out);
+int gpue_page_decode_bshuf_lz4_i32(gpue_session* s, gpue_dbuf* page, uint32_t n_values,
+                                   gpue_dbuf* out) {
+    ARG_CHECK(s && page && out && n_values > 0 && n_values % 8 == 0);
+    ARG_CHECK(out->bytes >= (uint64_t)n_values * 4);
+    uint32_t n_blocks = (n_values + BSHUF_BLOCK_I32 - 1) / BSHUF_BLOCK_I32;
+    uint32_t* d_starts = nullptr;
+    uint8_t* d_scratch = nullptr;
+    unsigned long long* d_err = nullptr;
+    HIP_CHECK(hipMalloc(&d_starts, (n_blocks + 1) * 4));
+    HIP_CHECK(hipMalloc(&d_scratch, (uint64_t)n_blocks * BSHUF_BLOCK_I32 * 4));
+    HIP_CHECK(hipMalloc(&d_err, 8));
+    HIP_CHECK(hipMemsetAsync(d_err, 0, 8, s->stream));
+    hipLaunchKernelGGL(k_page_block_starts, dim3(1), dim3(1), 0, s->stream,
+                       (const uint8_t*)page->ptr, n_blocks, d_starts);
+    hipLaunchKernelGGL(k_lz4_decode_blocks, dim3(grid_for(n_blocks)), dim3(BLOCK), 0,
+                       s->stream, (const uint8_t*)page->ptr, d_starts, n_blocks, n_values,
+                       d_scratch, d_err);
+    hipLaunchKernelGGL(k_bshuf_untranspose_i32, dim3(grid_for(n_values)), dim3(BLOCK), 0,
+                       s->stream, d_scratch, n_values, (int32_t*)out->ptr);
+    unsigned long long h_err = 0;
+    HIP_CHECK(hipMemcpyAsync(&h_err, d_err, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_starts);
+    (void)hipFree(d_scratch);
+    (void)hipFree(d_err);
+    if (h_err) {
+        snprintf(g_err, sizeof(g_err), "page decode: malformed LZ4 block");
+        return GPUE_ERR_ARG;
+    }
+    return GPUE_OK;
+}
+
+// ---------------------------------------------------------------------------
 // TopN — ORDER BY value DESC LIMIT k (reference exec/chunks_sorter_topn.cpp;
 // SSB/TPC-H result shapes end in ORDER BY + LIMIT).
 // Order: (value, key) lexicographic descending — deterministic under value

@@ -108,6 +108,7 @@ def _declare(lib):
         "gpue_hash_agg_sum_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_u64, c_vp, c_vp, c_vp,
                                           c_u64, ctypes.POINTER(c_u64)]),
         "gpue_topk_i64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp]),
+        "gpue_page_decode_bshuf_lz4_i32": (c_i32, [c_vp, c_vp, c_u32, c_vp]),
         "gpue_timer_start": (c_i32, [c_vp]),
         "gpue_timer_stop": (c_i32, [c_vp, ctypes.POINTER(ctypes.c_float)]),
     }
@@ -399,6 +400,10 @@ class Engine:
         _ck(self._lib, self._lib.gpue_partition_i32(self._h, keys._h, n, num_channels,
                                                     sp.ctypes.data_as(c_vp), row_indexes._h))
         return sp
+
+    def page_decode_bshuf_lz4_i32(self, page: DBuf, n_values, out: DBuf):
+        _ck(self._lib, self._lib.gpue_page_decode_bshuf_lz4_i32(self._h, page._h,
+                                                                n_values, out._h))
 
     def topk_i64(self, keys: DBuf, vals: DBuf, n, k):
         ok = np.zeros(k, np.uint64)

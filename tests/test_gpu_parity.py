@@ -539,6 +539,24 @@ def test_q3_with_topn(engine):
         b.free()
 
 
+def test_page_decode_parity(engine):
+    """Storage ingress: GPU bitshuffle+LZ4 page decode == oracle decode ==
+    original values, for compressible and incompressible pages."""
+    rng = np.random.default_rng(53)
+    for vals in [rng.integers(0, 50, 2048 * 200).astype(np.int32),
+                 rng.integers(0, 2**31, 2048 * 37 + 8).astype(np.int32),
+                 np.arange(2048 * 10, dtype=np.int32)]:
+        page = orc.bshuf_lz4_encode_i32(vals)
+        pbuf = engine.alloc(len(page))
+        pbuf.h2d(page)
+        out = engine.alloc(len(vals) * 4)
+        engine.page_decode_bshuf_lz4_i32(pbuf, len(vals), out)
+        got = out.d2h(np.int32, len(vals))
+        assert np.array_equal(got, vals)
+        pbuf.free()
+        out.free()
+
+
 def test_partition_parity(engine):
     n, nch = 3_000_000, 8
     keys = engine.alloc(n * 4)
