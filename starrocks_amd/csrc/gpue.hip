@@ -2024,6 +2024,100 @@ __global__ void k_ub_sum1(const int4* __restrict__ a, uint64_t n4,
     if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(out, (unsigned long long)sum);
 }
 
+// q21 decomposition probes: phase-1 only (pk stream + part-bitset gather)
+// and streams-only (4 columns, no probe) — isolates which resource bounds
+// the star-agg kernels
+__global__ __launch_bounds__(BLOCK_Q21) void
+k_ub_q21_phase1(const int4* __restrict__ pk4, uint64_t n4,
+                const uint32_t* __restrict__ pbits, int64_t psmin, uint64_t psint,
+                unsigned long long* __restrict__ out) {
+    uint64_t acc = 0;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i + stride < n4; i += 2 * stride) {
+        int4 pa = pk4[i];
+        int4 pb_ = pk4[i + stride];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint32_t idx = (uint32_t)((&pa.x)[j] - psmin);
+            uint32_t cidx = idx < psint ? idx : 0u;
+            acc += pbits[cidx >> 5] >> (cidx & 31) & 1u;
+        }
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint32_t idx = (uint32_t)((&pb_.x)[j] - psmin);
+            uint32_t cidx = idx < psint ? idx : 0u;
+            acc += pbits[cidx >> 5] >> (cidx & 31) & 1u;
+        }
+    }
+    for (; i < n4; i += stride) {
+        int4 p4 = pk4[i];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint32_t idx = (uint32_t)((&p4.x)[j] - psmin);
+            uint32_t cidx = idx < psint ? idx : 0u;
+            acc += pbits[cidx >> 5] >> (cidx & 31) & 1u;
+        }
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        acc += __shfl_down((unsigned long long)acc, off, WAVE);
+    if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(out, (unsigned long long)acc);
+}
+
+__global__ __launch_bounds__(BLOCK_Q21) void
+k_ub_sum4(const int4* __restrict__ a, const int4* __restrict__ b,
+          const int4* __restrict__ c, const int4* __restrict__ d, uint64_t n4,
+          unsigned long long* __restrict__ out) {
+    int64_t sum = 0;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+        int4 x = a[i], y = b[i], z = c[i], w = d[i];
+        sum += (int64_t)x.x + x.y + x.z + x.w + y.x + y.y + y.z + y.w +
+               (int64_t)z.x + z.y + z.z + z.w + w.x + w.y + w.z + w.w;
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        sum += __shfl_down((long long)sum, off, WAVE);
+    if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(out, (unsigned long long)sum);
+}
+
+extern "C" int gpue_ubench_q21(gpue_session* s, int which, gpue_dbuf* pk, gpue_dbuf* sk,
+                               gpue_dbuf* od, gpue_dbuf* rv, gpue_dbuf* pbits,
+                               int64_t psmin, uint64_t psint, uint64_t n, int grid1024,
+                               int reps, float* ms_out);
+int gpue_ubench_q21(gpue_session* s, int which, gpue_dbuf* pk, gpue_dbuf* sk, gpue_dbuf* od,
+                    gpue_dbuf* rv, gpue_dbuf* pbits, int64_t psmin, uint64_t psint,
+                    uint64_t n, int grid1024, int reps, float* ms_out) {
+    ARG_CHECK(s && pk && ms_out && reps > 0);
+    uint64_t n4 = n / 4;
+    unsigned long long* d_out = nullptr;
+    HIP_CHECK(hipMalloc(&d_out, 8));
+    HIP_CHECK(hipMemsetAsync(d_out, 0, 8, s->stream));
+    for (int pass = 0; pass < 2; pass++) {
+        if (pass == 1) HIP_CHECK(hipEventRecord(s->ev_start, s->stream));
+        int iters = pass == 0 ? 1 : reps;
+        for (int r = 0; r < iters; r++) {
+            if (which == 0) {
+                ARG_CHECK(pbits);
+                hipLaunchKernelGGL(k_ub_q21_phase1, dim3(grid1024), dim3(BLOCK_Q21), 0,
+                                   s->stream, (const int4*)pk->ptr, n4, (const uint32_t*)pbits->ptr,
+                                   psmin, psint, d_out);
+            } else {
+                ARG_CHECK(sk && od && rv);
+                hipLaunchKernelGGL(k_ub_sum4, dim3(grid1024), dim3(BLOCK_Q21), 0, s->stream,
+                                   (const int4*)pk->ptr, (const int4*)sk->ptr,
+                                   (const int4*)od->ptr, (const int4*)rv->ptr, n4, d_out);
+            }
+        }
+    }
+    HIP_CHECK(hipEventRecord(s->ev_stop, s->stream));
+    HIP_CHECK(hipEventSynchronize(s->ev_stop));
+    float ms = 0;
+    HIP_CHECK(hipEventElapsedTime(&ms, s->ev_start, s->ev_stop));
+    (void)hipFree(d_out);
+    *ms_out = ms / reps;
+    return GPUE_OK;
+}
+
 extern "C" int gpue_ubench(gpue_session* s, int which, gpue_dbuf* a, gpue_dbuf* b,
                            gpue_dbuf* c, uint64_t n_i32, int reps, float* ms_out);
 int gpue_ubench(gpue_session* s, int which, gpue_dbuf* a, gpue_dbuf* b, gpue_dbuf* c,
