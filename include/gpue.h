@@ -159,6 +159,15 @@ int gpue_q21_star_agg(gpue_session* s, gpue_join_table* parts, gpue_join_table* 
                       gpue_dbuf* lo_orderdate, gpue_dbuf* lo_revenue, uint64_t n_rows,
                       int64_t* group_sums_out);
 
+/* Two-stream pipelined variant: K_A (part runtime-filter probe + brand
+ * payload -> u16 per row, brand_scratch) overlaps K_B (remaining streams +
+ * supplier/date probes + group agg) across n_chunks chunks — the two legs
+ * measured fully additive inside one kernel. */
+int gpue_q21_star_agg_pipe(gpue_session* s, gpue_join_table* parts, gpue_join_table* supps,
+                           gpue_join_table* dates, gpue_dbuf* lo_partkey, gpue_dbuf* lo_suppkey,
+                           gpue_dbuf* lo_orderdate, gpue_dbuf* lo_revenue, uint64_t n_rows,
+                           gpue_dbuf* brand_scratch /* n_rows × u16 */, gpue_dbuf* group_sums,
+                           int n_chunks);
 /* Async bench variant of the star aggregate (7000 × int64 device buffer). */
 int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_table* supps,
                             gpue_join_table* dates, gpue_dbuf* lo_partkey, gpue_dbuf* lo_suppkey,

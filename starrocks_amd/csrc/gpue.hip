@@ -90,6 +90,10 @@ struct gpue_session {
     // pull) pay ~5 µs instead of a pageable-copy ~20 µs
     void* pinned = nullptr;
     static constexpr uint64_t PINNED_BYTES = 65536;
+    // second stream + event pool for pipelined two-kernel operators
+    hipStream_t stream2 = nullptr;
+    static constexpr int N_CHUNK_EVENTS = 32;
+    hipEvent_t chunk_ev[N_CHUNK_EVENTS] = {};
 };
 
 struct gpue_dbuf {
@@ -148,6 +152,9 @@ int gpue_session_create(int device_index, gpue_session** out) {
     HIP_CHECK(hipEventCreate(&s->ev_start));
     HIP_CHECK(hipEventCreate(&s->ev_stop));
     HIP_CHECK(hipHostMalloc(&s->pinned, gpue_session::PINNED_BYTES));
+    HIP_CHECK(hipStreamCreate(&s->stream2));
+    for (int i = 0; i < gpue_session::N_CHUNK_EVENTS; i++)
+        HIP_CHECK(hipEventCreateWithFlags(&s->chunk_ev[i], hipEventDisableTiming));
     *out = s;
     return GPUE_OK;
 }
@@ -155,6 +162,9 @@ int gpue_session_create(int device_index, gpue_session** out) {
 void gpue_session_destroy(gpue_session* s) {
     if (!s) return;
     if (s->pinned) (void)hipHostFree(s->pinned);
+    for (int i = 0; i < gpue_session::N_CHUNK_EVENTS; i++)
+        if (s->chunk_ev[i]) (void)hipEventDestroy(s->chunk_ev[i]);
+    if (s->stream2) (void)hipStreamDestroy(s->stream2);
     if (s->d_datekey) (void)hipFree(s->d_datekey);
     (void)hipEventDestroy(s->ev_start);
     (void)hipEventDestroy(s->ev_stop);
@@ -1627,6 +1637,154 @@ __global__ void k_q21_star_agg_glob(const int32_t* __restrict__ pk,
         atomicAdd(&group_sums[(year1 - 1) * 1000 + (brand1 - 1)],
                   (unsigned long long)(int64_t)rv[r]);
     }
+}
+
+// ---------------------------------------------------------------------------
+// Two-stream pipelined q21 (GPUE_Q21_PIPE=1): the fused kernel's streaming
+// leg (1.57 ms) and its part-probe gather leg (2.32 ms) measured fully
+// ADDITIVE (profiles/q21_decomp.log) — so split them into an operator pair
+// and overlap across chunks: K_A (pk stream + part runtime-filter probe +
+// brand payload gather -> u16 per row) runs on stream2 for chunk i while
+// K_B (sk/od/rv streams + brand u16 + supplier/date probes + LDS group agg)
+// consumes chunk i-1 on the session stream, linked by events.
+// ---------------------------------------------------------------------------
+__global__ void k_q21_pass_payload(const int32_t* __restrict__ pk, uint64_t n,
+                                   const uint32_t* __restrict__ pbits, int64_t psmin,
+                                   uint64_t psint, const uint16_t* __restrict__ pfirst,
+                                   uint16_t* __restrict__ brand_out) {
+    const uint64_t n4 = n / 4;
+    const int4* __restrict__ pk4 = (const int4*)pk;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i + stride < n4; i += 2 * stride) {
+        int4 pa = pk4[i];
+        int4 pb_ = pk4[i + stride];
+        ushort4 oa, ob;
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint32_t idx = (uint32_t)((&pa.x)[j] - psmin);
+            bool in = idx < psint;
+            uint32_t cidx = in ? idx : 0u;
+            bool pass = in & ((pbits[cidx >> 5] >> (cidx & 31)) & 1u);
+            (&oa.x)[j] = pass ? pfirst[(&pa.x)[j] - 1] : (uint16_t)0;
+        }
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint32_t idx = (uint32_t)((&pb_.x)[j] - psmin);
+            bool in = idx < psint;
+            uint32_t cidx = in ? idx : 0u;
+            bool pass = in & ((pbits[cidx >> 5] >> (cidx & 31)) & 1u);
+            (&ob.x)[j] = pass ? pfirst[(&pb_.x)[j] - 1] : (uint16_t)0;
+        }
+        ((ushort4*)brand_out)[i] = oa;
+        ((ushort4*)brand_out)[i + stride] = ob;
+    }
+    for (; i < n4; i += stride) {
+        int4 p4 = pk4[i];
+        ushort4 o;
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint32_t idx = (uint32_t)((&p4.x)[j] - psmin);
+            bool in = idx < psint;
+            uint32_t cidx = in ? idx : 0u;
+            bool pass = in & ((pbits[cidx >> 5] >> (cidx & 31)) & 1u);
+            (&o.x)[j] = pass ? pfirst[(&p4.x)[j] - 1] : (uint16_t)0;
+        }
+        ((ushort4*)brand_out)[i] = o;
+    }
+    // tail rows (n % 4)
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (uint64_t r = n4 * 4 + tid; r < n; r += stride) {
+        uint32_t idx = (uint32_t)(pk[r] - psmin);
+        bool in = idx < psint;
+        uint32_t cidx = in ? idx : 0u;
+        bool pass = in & ((pbits[cidx >> 5] >> (cidx & 31)) & 1u);
+        brand_out[r] = pass ? pfirst[pk[r] - 1] : (uint16_t)0;
+    }
+}
+
+__global__ __launch_bounds__(BLOCK_Q21) void
+k_q21_phase2(const uint16_t* __restrict__ brand, const int32_t* __restrict__ sk,
+             const int32_t* __restrict__ od, const int32_t* __restrict__ rv, uint64_t n,
+             const uint32_t* __restrict__ sbits, int64_t ssmin, uint64_t ssint,
+             const uint16_t* __restrict__ dfirst, int64_t dmin,
+             unsigned long long* __restrict__ group_sums) {
+    __shared__ unsigned long long g[NG_Q21];
+    for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x) g[j] = 0;
+    __syncthreads();
+    const uint64_t n4 = n / 4;
+    const ushort4* __restrict__ br4 = (const ushort4*)brand;
+    const int4* __restrict__ sk4 = (const int4*)sk;
+    const int4* __restrict__ od4 = (const int4*)od;
+    const int4* __restrict__ rv4 = (const int4*)rv;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+        ushort4 b4 = br4[i];
+        int4 s4 = sk4[i];
+        int4 o4 = od4[i];
+        int4 r4 = rv4[i];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint16_t brand1 = (&b4.x)[j];
+            if (brand1 == 0) continue;
+            uint32_t sidx = (uint32_t)((&s4.x)[j] - ssmin);
+            if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
+            uint32_t year1 = dfirst[(&o4.x)[j] - dmin];
+            atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)],
+                      (unsigned long long)(int64_t)(&r4.x)[j]);
+        }
+    }
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (uint64_t r = n4 * 4 + tid; r < n; r += stride) {
+        uint16_t brand1 = brand[r];
+        if (brand1 == 0) continue;
+        uint32_t sidx = (uint32_t)(sk[r] - ssmin);
+        if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
+        uint32_t year1 = dfirst[od[r] - dmin];
+        atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)], (unsigned long long)(int64_t)rv[r]);
+    }
+    __syncthreads();
+    for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x)
+        if (g[j] != 0) atomicAdd(&group_sums[j], g[j]);
+}
+
+extern "C" int gpue_q21_star_agg_pipe(gpue_session* s, gpue_join_table* parts,
+                                      gpue_join_table* supps, gpue_join_table* dates,
+                                      gpue_dbuf* pk, gpue_dbuf* sk, gpue_dbuf* od,
+                                      gpue_dbuf* rv, uint64_t n, gpue_dbuf* brand_scratch,
+                                      gpue_dbuf* group_sums, int n_chunks);
+int gpue_q21_star_agg_pipe(gpue_session* s, gpue_join_table* parts, gpue_join_table* supps,
+                           gpue_join_table* dates, gpue_dbuf* pk, gpue_dbuf* sk,
+                           gpue_dbuf* od, gpue_dbuf* rv, uint64_t n,
+                           gpue_dbuf* brand_scratch, gpue_dbuf* group_sums, int n_chunks) {
+    ARG_CHECK(s && parts && supps && dates && pk && sk && od && rv && brand_scratch &&
+              group_sums);
+    ARG_CHECK(parts->first16 && parts->bitset && supps->bitset && dates->first16);
+    ARG_CHECK(brand_scratch->bytes >= n * 2);
+    ARG_CHECK(n_chunks >= 1 && n_chunks <= gpue_session::N_CHUNK_EVENTS);
+    HIP_CHECK(hipMemsetAsync(group_sums->ptr, 0, NG_Q21 * sizeof(int64_t), s->stream));
+    uint64_t chunk = ((n / n_chunks) + 3) & ~3ull; // 4-row aligned chunks
+    for (int c = 0; c < n_chunks; c++) {
+        uint64_t lo = (uint64_t)c * chunk;
+        if (lo >= n) break;
+        uint64_t len = min(chunk, n - lo);
+        hipLaunchKernelGGL(k_q21_pass_payload, dim3(grid_for(len)), dim3(BLOCK), 0,
+                           s->stream2, (const int32_t*)pk->ptr + lo, len, parts->bitset,
+                           parts->set_min, (uint64_t)(parts->set_max - parts->set_min + 1),
+                           parts->first16, (uint16_t*)brand_scratch->ptr + lo);
+        HIP_CHECK(hipEventRecord(s->chunk_ev[c], s->stream2));
+        HIP_CHECK(hipStreamWaitEvent(s->stream, s->chunk_ev[c], 0));
+        hipLaunchKernelGGL(k_q21_phase2, dim3(env_cap("GPUE_GRID_WIDE", 512)),
+                           dim3(BLOCK_Q21), 0, s->stream,
+                           (const uint16_t*)brand_scratch->ptr + lo,
+                           (const int32_t*)sk->ptr + lo, (const int32_t*)od->ptr + lo,
+                           (const int32_t*)rv->ptr + lo, len, supps->bitset,
+                           supps->set_min, (uint64_t)(supps->set_max - supps->set_min + 1),
+                           dates->first16, dates->min_key,
+                           (unsigned long long*)group_sums->ptr);
+    }
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
 }
 
 static bool q21_glob() {

@@ -86,6 +86,7 @@ def _declare(lib):
         "gpue_dbuf_ptr": (c_i32, [c_vp, ctypes.POINTER(c_vp)]),
         "gpue_q43_star_agg_async": (c_i32, [c_vp] * 11 + [c_u64, c_vp]),
         "gpue_q21_star_agg_async": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
+        "gpue_q21_star_agg_pipe": (c_i32, [c_vp] * 8 + [c_u64, c_vp, c_vp, c_i32]),
         "gpue_partition_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
         "gpue_gather_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_gather_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
@@ -301,6 +302,12 @@ class Engine:
     def q1_join_sum_async(self, dates: JoinTable, od: DBuf, ep: DBuf, dc: DBuf, n, acc: DBuf):
         _ck(self._lib, self._lib.gpue_q1_join_sum_async(self._h, dates._h, od._h, ep._h,
                                                         dc._h, n, acc._h))
+
+    def q21_star_agg_pipe(self, parts, supps, dates, pk, sk, od, rv, n,
+                          brand_scratch: DBuf, group_sums: DBuf, n_chunks=8):
+        _ck(self._lib, self._lib.gpue_q21_star_agg_pipe(
+            self._h, parts._h, supps._h, dates._h, pk._h, sk._h, od._h, rv._h, n,
+            brand_scratch._h, group_sums._h, n_chunks))
 
     def q21_star_agg_async(self, parts, supps, dates, pk, sk, od, rv, n, group_sums: DBuf):
         _ck(self._lib, self._lib.gpue_q21_star_agg_async(self._h, parts._h, supps._h, dates._h,
