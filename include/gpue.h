@@ -134,6 +134,15 @@ int gpue_join_table_first_d2h(gpue_join_table* t, uint32_t* dst, uint64_t n_entr
 int gpue_join_probe_emit_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
                              uint64_t n_rows, gpue_dbuf* out_probe_idx,
                              gpue_dbuf* out_build_idx, uint64_t* match_count);
+/* Per-join-type probe (join_hash_map.h:228-333): mode 0 INNER, 1 LEFT_SEMI,
+ * 2 LEFT_ANTI (unmatched rows, build index 0 = NULL sentinel), 3 LEFT_OUTER. */
+int gpue_join_probe_emit_mode_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
+                                  uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
+                                  gpue_dbuf* out_build_idx, uint64_t* match_count);
+/* RIGHT SEMI (anti=0) / RIGHT ANTI (anti=1): matched/unmatched BUILD rows. */
+int gpue_join_probe_right_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
+                              uint64_t n_rows, int anti, gpue_dbuf* out_build_idx,
+                              uint64_t* count);
 
 /* ---- fused probe + aggregate pipelines ----
  * Replace the probe -> output gather -> Aggregator::update_batch chain

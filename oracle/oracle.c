@@ -288,6 +288,67 @@ uint64_t orc_probe_emit_u32(const uint32_t* build_keys, const uint32_t* next,
     return m;
 }
 
+/* Per-join-type probe variants (reference join_hash_map.h:228-333,
+ * _probe_from_ht_for_left_{outer,semi,anti}_join in join_hash_map.hpp):
+ * LEFT_SEMI emits each matching probe row once (first match's build index);
+ * LEFT_ANTI emits unmatched probe rows with build index 0 (the NULL
+ * sentinel row); LEFT_OUTER emits all pairs, or (i, 0) when unmatched.
+ * mode: 0 INNER, 1 LEFT_SEMI, 2 LEFT_ANTI, 3 LEFT_OUTER. */
+uint64_t orc_probe_emit_mode_u32(const uint32_t* build_keys, const uint32_t* next,
+                                 const uint32_t* probe_keys, const uint32_t* heads,
+                                 uint32_t probe_rows, int mode,
+                                 uint32_t* out_probe_idx, uint32_t* out_build_idx) {
+    uint64_t m = 0;
+    for (uint32_t i = 0; i < probe_rows; i++) {
+        uint32_t b = heads[i];
+        uint32_t c = 0;
+        while (b != 0) {
+            if (build_keys[b] == probe_keys[i]) {
+                if (mode == 0 || mode == 3) {
+                    out_probe_idx[m] = i;
+                    out_build_idx[m] = b;
+                    m++;
+                } else if (mode == 1 && c == 0) {
+                    out_probe_idx[m] = i;
+                    out_build_idx[m] = b;
+                    m++;
+                }
+                c++;
+                if (mode == 1 || mode == 2) break;
+            }
+            b = next[b];
+        }
+        if (c == 0 && (mode == 2 || mode == 3)) {
+            out_probe_idx[m] = i;
+            out_build_idx[m] = 0;
+            m++;
+        }
+    }
+    return m;
+}
+
+/* RIGHT SEMI/ANTI (join_hash_map.hpp _probe_from_ht_for_right_*): the probe
+ * pass marks matched BUILD rows; the output is the matched (semi) or
+ * unmatched (anti) build rows, 1-based. Returns count. */
+uint64_t orc_probe_right_u32(const uint32_t* build_keys, const uint32_t* next,
+                             uint32_t build_rows, const uint32_t* probe_keys,
+                             const uint32_t* heads, uint32_t probe_rows, int anti,
+                             uint32_t* out_build_idx) {
+    uint8_t* matched = (uint8_t*)calloc(build_rows + 1, 1);
+    for (uint32_t i = 0; i < probe_rows; i++) {
+        uint32_t b = heads[i];
+        while (b != 0) {
+            if (build_keys[b] == probe_keys[i]) matched[b] = 1;
+            b = next[b];
+        }
+    }
+    uint64_t m = 0;
+    for (uint32_t j = 1; j <= build_rows; j++)
+        if ((anti && !matched[j]) || (!anti && matched[j])) out_build_idx[m++] = j;
+    free(matched);
+    return m;
+}
+
 /* ====================================================================== */
 /* Predicate filter                                                        */
 /* ====================================================================== */

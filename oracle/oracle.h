@@ -75,6 +75,17 @@ uint64_t orc_probe_emit_u32(const uint32_t* build_keys, const uint32_t* next,
                             uint32_t probe_rows, int collision_free,
                             uint32_t* out_probe_idx, uint32_t* out_build_idx);
 
+/* per-join-type probe emits (join_hash_map.h:228-333): mode 0 INNER,
+ * 1 LEFT_SEMI, 2 LEFT_ANTI, 3 LEFT_OUTER */
+uint64_t orc_probe_emit_mode_u32(const uint32_t* build_keys, const uint32_t* next,
+                                 const uint32_t* probe_keys, const uint32_t* heads,
+                                 uint32_t probe_rows, int mode,
+                                 uint32_t* out_probe_idx, uint32_t* out_build_idx);
+uint64_t orc_probe_right_u32(const uint32_t* build_keys, const uint32_t* next,
+                             uint32_t build_rows, const uint32_t* probe_keys,
+                             const uint32_t* heads, uint32_t probe_rows, int anti,
+                             uint32_t* out_build_idx);
+
 /* ---- predicate filter (base/simd/filter.h:26-38 + chunk_predicate_evaluator.cpp:31-80) ----
  * stable stream compaction of int64 values where v < theta; returns count */
 uint64_t orc_filter_i64_lt(const int64_t* in, uint64_t n, int64_t theta, int64_t* out);

@@ -310,6 +310,28 @@ def range_direct_lookup(probe_keys: np.ndarray, min_value, max_value, first: np.
     return heads
 
 
+def probe_emit_mode(build_keys, nxt, probe_keys, heads, mode):
+    lib = load()
+    lib.orc_probe_emit_mode_u32.restype = c_u64
+    lib.orc_probe_emit_mode_u32.argtypes = [c_vp, c_vp, c_vp, c_vp, c_u32, c_i32, c_vp, c_vp]
+    cap = max(len(probe_keys) * 8, 1024)
+    op = np.empty(cap, np.uint32)
+    ob = np.empty(cap, np.uint32)
+    m = lib.orc_probe_emit_mode_u32(_p(build_keys), _p(nxt), _p(probe_keys), _p(heads),
+                                    len(probe_keys), mode, _p(op), _p(ob))
+    return op[:m].copy(), ob[:m].copy()
+
+
+def probe_right(build_keys, nxt, probe_keys, heads, anti):
+    lib = load()
+    lib.orc_probe_right_u32.restype = c_u64
+    lib.orc_probe_right_u32.argtypes = [c_vp, c_vp, c_u32, c_vp, c_vp, c_u32, c_i32, c_vp]
+    out = np.empty(len(build_keys), np.uint32)
+    m = lib.orc_probe_right_u32(_p(build_keys), _p(nxt), len(build_keys) - 1,
+                                _p(probe_keys), _p(heads), len(probe_keys), anti, _p(out))
+    return out[:m].copy()
+
+
 def probe_emit(build_keys: np.ndarray, nxt: np.ndarray, probe_keys: np.ndarray,
                heads: np.ndarray, collision_free=False):
     cap = max(len(probe_keys) * 8, 1024)

@@ -77,6 +77,17 @@ static inline uint32_t grid_stream(uint64_t n) {
     return grid_capped(n, cap);
 }
 
+// per-join-type count transform (reference join_hash_map.h:228-333):
+// mode 0 INNER, 1 LEFT_SEMI, 2 LEFT_ANTI, 3 LEFT_OUTER
+__device__ static inline uint32_t join_mode_count(uint32_t c, int mode) {
+    switch (mode) {
+    case 1: return c ? 1u : 0u;
+    case 2: return c ? 0u : 1u;
+    case 3: return c ? c : 1u;
+    default: return c;
+    }
+}
+
 // ---------------------------------------------------------------------------
 // session / buffers
 // ---------------------------------------------------------------------------
@@ -1083,7 +1094,7 @@ __global__ void k_probe_count_lc(const uint32_t* __restrict__ probe_keys, uint64
                                  uint32_t log_bucket_size, uint32_t bucket_mask,
                                  const uint32_t* __restrict__ first,
                                  const uint32_t* __restrict__ next,
-                                 const uint32_t* __restrict__ build_keys,
+                                 const uint32_t* __restrict__ build_keys, int mode,
                                  uint32_t* __restrict__ row_counts) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
@@ -1094,7 +1105,7 @@ __global__ void k_probe_count_lc(const uint32_t* __restrict__ probe_keys, uint64
             c++;
             b = next[b];
         }
-        row_counts[i] = c;
+        row_counts[i] = join_mode_count(c, mode);
     }
 }
 
@@ -1102,7 +1113,7 @@ __global__ void k_probe_emit_lc(const uint32_t* __restrict__ probe_keys, uint64_
                                 uint32_t log_bucket_size, uint32_t bucket_mask,
                                 const uint32_t* __restrict__ first,
                                 const uint32_t* __restrict__ next,
-                                const uint32_t* __restrict__ build_keys,
+                                const uint32_t* __restrict__ build_keys, int mode,
                                 const uint32_t* __restrict__ row_counts,
                                 const uint64_t* __restrict__ row_offsets,
                                 uint32_t* __restrict__ out_probe,
@@ -1113,10 +1124,17 @@ __global__ void k_probe_emit_lc(const uint32_t* __restrict__ probe_keys, uint64_
         uint64_t pos = row_offsets[i];
         uint32_t b = lc_lookup_head(probe_keys[i], log_bucket_size, bucket_mask, first,
                                     build_keys);
+        if (b == 0) { // unmatched emit (ANTI/OUTER)
+            out_probe[pos] = (uint32_t)i;
+            out_build[pos] = 0;
+            continue;
+        }
+        if (mode == 2) continue;
         while (b != 0) {
             out_probe[pos] = (uint32_t)i;
             out_build[pos] = b;
             pos++;
+            if (mode == 1) break;
             b = next[b];
         }
     }
@@ -1127,7 +1145,7 @@ __global__ void k_probe_emit_lc(const uint32_t* __restrict__ probe_keys, uint64_
 __global__ void k_probe_count_bc(const uint32_t* __restrict__ probe_keys, uint64_t n,
                                  uint32_t log_bucket_size, const uint32_t* __restrict__ first,
                                  const uint32_t* __restrict__ next,
-                                 const uint32_t* __restrict__ build_keys,
+                                 const uint32_t* __restrict__ build_keys, int mode,
                                  uint32_t* __restrict__ row_counts) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
@@ -1138,14 +1156,14 @@ __global__ void k_probe_count_bc(const uint32_t* __restrict__ probe_keys, uint64
             c += (build_keys[b] == k);
             b = next[b];
         }
-        row_counts[i] = c;
+        row_counts[i] = join_mode_count(c, mode);
     }
 }
 
 __global__ void k_probe_emit_bc(const uint32_t* __restrict__ probe_keys, uint64_t n,
                                 uint32_t log_bucket_size, const uint32_t* __restrict__ first,
                                 const uint32_t* __restrict__ next,
-                                const uint32_t* __restrict__ build_keys,
+                                const uint32_t* __restrict__ build_keys, int mode,
                                 const uint32_t* __restrict__ row_counts,
                                 const uint64_t* __restrict__ row_offsets,
                                 uint32_t* __restrict__ out_probe,
@@ -1156,13 +1174,22 @@ __global__ void k_probe_emit_bc(const uint32_t* __restrict__ probe_keys, uint64_
         uint32_t k = probe_keys[i];
         uint64_t pos = row_offsets[i];
         uint32_t b = first[join_hash_u32(k, log_bucket_size)];
+        bool any = false;
         while (b != 0) {
             if (build_keys[b] == k) {
-                out_probe[pos] = (uint32_t)i;
-                out_build[pos] = b;
-                pos++;
+                any = true;
+                if (mode != 2) {
+                    out_probe[pos] = (uint32_t)i;
+                    out_build[pos] = b;
+                    pos++;
+                }
+                if (mode == 1 || mode == 2) break;
             }
             b = next[b];
+        }
+        if (!any && (mode == 2 || mode == 3)) {
+            out_probe[pos] = (uint32_t)i;
+            out_build[pos] = 0;
         }
     }
 }
@@ -1198,7 +1225,7 @@ __global__ void k_block_sums_u32(const uint32_t* __restrict__ v, uint64_t n, uin
 // that simplification.
 __global__ void k_probe_count_rd(const int32_t* __restrict__ probe_keys, uint64_t n,
                                  int64_t mn, int64_t mx, const uint32_t* __restrict__ first,
-                                 const uint32_t* __restrict__ next,
+                                 const uint32_t* __restrict__ next, int mode,
                                  uint32_t* __restrict__ row_counts) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
@@ -1211,13 +1238,13 @@ __global__ void k_probe_count_rd(const int32_t* __restrict__ probe_keys, uint64_
                 b = next[b];
             }
         }
-        row_counts[i] = c;
+        row_counts[i] = join_mode_count(c, mode);
     }
 }
 
 __global__ void k_probe_emit_rd(const int32_t* __restrict__ probe_keys, uint64_t n,
-                                int64_t mn, const uint32_t* __restrict__ first,
-                                const uint32_t* __restrict__ next,
+                                int64_t mn, int64_t mx, const uint32_t* __restrict__ first,
+                                const uint32_t* __restrict__ next, int mode,
                                 const uint32_t* __restrict__ row_counts,
                                 const uint64_t* __restrict__ row_offsets,
                                 uint32_t* __restrict__ out_probe,
@@ -1226,11 +1253,19 @@ __global__ void k_probe_emit_rd(const int32_t* __restrict__ probe_keys, uint64_t
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         if (row_counts[i] == 0) continue;
         uint64_t pos = row_offsets[i];
-        uint32_t b = first[probe_keys[i] - mn];
+        int32_t k = probe_keys[i];
+        uint32_t b = (k >= mn && k <= mx) ? first[k - mn] : 0;
+        if (b == 0) { // only reachable for ANTI/OUTER: the unmatched emit
+            out_probe[pos] = (uint32_t)i;
+            out_build[pos] = 0;
+            continue;
+        }
+        if (mode == 2) continue; // matched row contributes nothing to ANTI
         while (b != 0) {
             out_probe[pos] = (uint32_t)i;
             out_build[pos] = b;
             pos++;
+            if (mode == 1) break; // SEMI: first match only
             b = next[b];
         }
     }
@@ -1251,10 +1286,23 @@ __global__ void k_scan_offsets(const uint32_t* __restrict__ counts, uint64_t n, 
     }
 }
 
+extern "C" int gpue_join_probe_emit_mode_i32(gpue_session* s, gpue_join_table* t,
+                                             gpue_dbuf* probe_keys, uint64_t n_rows, int mode,
+                                             gpue_dbuf* out_probe_idx,
+                                             gpue_dbuf* out_build_idx, uint64_t* match_count);
+
 int gpue_join_probe_emit_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
                              uint64_t n_rows, gpue_dbuf* out_probe_idx,
                              gpue_dbuf* out_build_idx, uint64_t* match_count) {
+    return gpue_join_probe_emit_mode_i32(s, t, probe_keys, n_rows, 0, out_probe_idx,
+                                         out_build_idx, match_count);
+}
+
+int gpue_join_probe_emit_mode_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
+                                  uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
+                                  gpue_dbuf* out_build_idx, uint64_t* match_count) {
     ARG_CHECK(s && t && probe_keys && match_count && t->next != nullptr);
+    ARG_CHECK(mode >= 0 && mode <= 3);
     ARG_CHECK(probe_keys->bytes >= n_rows * 4);
     uint32_t nb = grid_for(n_rows);
     uint64_t tile = (n_rows + nb - 1) / nb;
@@ -1267,15 +1315,15 @@ int gpue_join_probe_emit_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* pro
         hipLaunchKernelGGL(k_probe_count_lc, dim3(nb), dim3(BLOCK), 0, s->stream,
                            (const uint32_t*)probe_keys->ptr, n_rows, t->log_bucket_size,
                            (uint32_t)(t->bucket_size - 1), t->first, t->next, t->build_keys,
-                           d_counts);
+                           mode, d_counts);
     } else if (t->kind == gpue_join_table::BUCKET_CHAINED) {
         hipLaunchKernelGGL(k_probe_count_bc, dim3(nb), dim3(BLOCK), 0, s->stream,
                            (const uint32_t*)probe_keys->ptr, n_rows, t->log_bucket_size,
-                           t->first, t->next, t->build_keys, d_counts);
+                           t->first, t->next, t->build_keys, mode, d_counts);
     } else
     hipLaunchKernelGGL(k_probe_count_rd, dim3(nb), dim3(BLOCK), 0, s->stream,
                        (const int32_t*)probe_keys->ptr, n_rows, t->min_key, t->max_key,
-                       t->first, t->next, d_counts);
+                       t->first, t->next, mode, d_counts);
     hipLaunchKernelGGL(k_block_sums_u32, dim3(nb), dim3(BLOCK), 0, s->stream,
                        d_counts, n_rows, tile, d_bsums);
     hipLaunchKernelGGL(k_scan_small, dim3(1), dim3(1), 0, s->stream, d_bsums, nb);
@@ -1293,21 +1341,136 @@ int gpue_join_probe_emit_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* pro
             hipLaunchKernelGGL(k_probe_emit_lc, dim3(nb), dim3(BLOCK), 0, s->stream,
                                (const uint32_t*)probe_keys->ptr, n_rows, t->log_bucket_size,
                                (uint32_t)(t->bucket_size - 1), t->first, t->next,
-                               t->build_keys, d_counts, d_offsets,
+                               t->build_keys, mode, d_counts, d_offsets,
                                (uint32_t*)out_probe_idx->ptr, (uint32_t*)out_build_idx->ptr);
         } else if (t->kind == gpue_join_table::BUCKET_CHAINED) {
             hipLaunchKernelGGL(k_probe_emit_bc, dim3(nb), dim3(BLOCK), 0, s->stream,
                                (const uint32_t*)probe_keys->ptr, n_rows, t->log_bucket_size,
-                               t->first, t->next, t->build_keys, d_counts, d_offsets,
+                               t->first, t->next, t->build_keys, mode, d_counts, d_offsets,
                                (uint32_t*)out_probe_idx->ptr, (uint32_t*)out_build_idx->ptr);
         } else
         hipLaunchKernelGGL(k_probe_emit_rd, dim3(nb), dim3(BLOCK), 0, s->stream,
-                           (const int32_t*)probe_keys->ptr, n_rows, t->min_key, t->first,
-                           t->next, d_counts, d_offsets,
+                           (const int32_t*)probe_keys->ptr, n_rows, t->min_key, t->max_key,
+                           t->first, t->next, mode, d_counts, d_offsets,
                            (uint32_t*)out_probe_idx->ptr, (uint32_t*)out_build_idx->ptr);
         HIP_CHECK(hipStreamSynchronize(s->stream));
         (void)hipFree(d_offsets);
     }
+    (void)hipFree(d_counts);
+    (void)hipFree(d_bsums);
+    return GPUE_OK;
+}
+
+// RIGHT SEMI/ANTI (join_hash_map.hpp _probe_from_ht_for_right_*): the probe
+// pass marks matched BUILD rows; output = matched (semi) / unmatched (anti)
+// build rows.
+__global__ void k_right_mark_rd(const int32_t* __restrict__ probe_keys, uint64_t n,
+                                int64_t mn, int64_t mx, const uint32_t* __restrict__ first,
+                                const uint32_t* __restrict__ next,
+                                uint32_t* __restrict__ matched_bits) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        int32_t k = probe_keys[i];
+        if (k < mn || k > mx) continue;
+        uint32_t b = first[k - mn];
+        while (b != 0) {
+            atomicOr(&matched_bits[b >> 5], 1u << (b & 31));
+            b = next[b];
+        }
+    }
+}
+
+__global__ void k_right_mark_keys(const uint32_t* __restrict__ probe_keys, uint64_t n,
+                                  int is_linear, uint32_t log_bucket_size,
+                                  uint32_t bucket_mask, const uint32_t* __restrict__ first,
+                                  const uint32_t* __restrict__ next,
+                                  const uint32_t* __restrict__ build_keys,
+                                  uint32_t* __restrict__ matched_bits) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint32_t k = probe_keys[i];
+        uint32_t b = is_linear
+                         ? lc_lookup_head(k, log_bucket_size, bucket_mask, first, build_keys)
+                         : first[join_hash_u32(k, log_bucket_size)];
+        while (b != 0) {
+            if (is_linear || build_keys[b] == k)
+                atomicOr(&matched_bits[b >> 5], 1u << (b & 31));
+            b = next[b];
+        }
+    }
+}
+
+__global__ void k_right_row_counts(const uint32_t* __restrict__ matched_bits,
+                                   uint64_t build_rows, int anti,
+                                   uint32_t* __restrict__ row_counts) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; j < build_rows;
+         j += stride) {
+        uint32_t b = (uint32_t)j + 1; // build rows are 1-based
+        uint32_t m = (matched_bits[b >> 5] >> (b & 31)) & 1u;
+        row_counts[j] = anti ? (1u - m) : m;
+    }
+}
+
+__global__ void k_right_emit(const uint32_t* __restrict__ row_counts,
+                             const uint64_t* __restrict__ row_offsets, uint64_t build_rows,
+                             uint32_t* __restrict__ out_build) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; j < build_rows;
+         j += stride)
+        if (row_counts[j]) out_build[row_offsets[j]] = (uint32_t)j + 1;
+}
+
+extern "C" int gpue_join_probe_right_i32(gpue_session* s, gpue_join_table* t,
+                                         gpue_dbuf* probe_keys, uint64_t n_rows, int anti,
+                                         gpue_dbuf* out_build_idx, uint64_t* count);
+int gpue_join_probe_right_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
+                              uint64_t n_rows, int anti, gpue_dbuf* out_build_idx,
+                              uint64_t* count) {
+    ARG_CHECK(s && t && probe_keys && count && t->next != nullptr);
+    uint64_t nwords = (t->row_count + 1 + 31) / 32;
+    uint32_t* d_bits = nullptr;
+    uint32_t* d_counts = nullptr;
+    uint64_t* d_bsums = nullptr;
+    uint64_t* d_offsets = nullptr;
+    HIP_CHECK(hipMalloc(&d_bits, nwords * 4));
+    HIP_CHECK(hipMemsetAsync(d_bits, 0, nwords * 4, s->stream));
+    if (t->kind == gpue_join_table::RANGE_DIRECT) {
+        hipLaunchKernelGGL(k_right_mark_rd, dim3(grid_for(n_rows)), dim3(BLOCK), 0, s->stream,
+                           (const int32_t*)probe_keys->ptr, n_rows, t->min_key, t->max_key,
+                           t->first, t->next, d_bits);
+    } else {
+        hipLaunchKernelGGL(k_right_mark_keys, dim3(grid_for(n_rows)), dim3(BLOCK), 0,
+                           s->stream, (const uint32_t*)probe_keys->ptr, n_rows,
+                           t->kind == gpue_join_table::LINEAR_CHAINED ? 1 : 0,
+                           t->log_bucket_size, (uint32_t)(t->bucket_size - 1), t->first,
+                           t->next, t->build_keys, d_bits);
+    }
+    uint64_t br = t->row_count;
+    uint32_t nb = grid_for(br);
+    uint64_t tile = (br + nb - 1) / nb;
+    HIP_CHECK(hipMalloc(&d_counts, br * 4));
+    HIP_CHECK(hipMalloc(&d_bsums, (nb + 1) * 8));
+    hipLaunchKernelGGL(k_right_row_counts, dim3(nb), dim3(BLOCK), 0, s->stream, d_bits, br,
+                       anti, d_counts);
+    hipLaunchKernelGGL(k_block_sums_u32, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts, br,
+                       tile, d_bsums);
+    hipLaunchKernelGGL(k_scan_small, dim3(1), dim3(1), 0, s->stream, d_bsums, nb);
+    uint64_t total = 0;
+    HIP_CHECK(hipMemcpyAsync(&total, d_bsums + nb, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    *count = total;
+    if (out_build_idx && total > 0) {
+        ARG_CHECK(out_build_idx->bytes >= total * 4);
+        HIP_CHECK(hipMalloc(&d_offsets, br * 8));
+        hipLaunchKernelGGL(k_scan_offsets, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts, br,
+                           tile, d_bsums, d_offsets);
+        hipLaunchKernelGGL(k_right_emit, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts,
+                           d_offsets, br, (uint32_t*)out_build_idx->ptr);
+        HIP_CHECK(hipStreamSynchronize(s->stream));
+        (void)hipFree(d_offsets);
+    }
+    (void)hipFree(d_bits);
     (void)hipFree(d_counts);
     (void)hipFree(d_bsums);
     return GPUE_OK;

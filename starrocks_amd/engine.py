@@ -78,6 +78,8 @@ def _declare(lib):
         "gpue_join_table_minmax": (c_i32, [c_vp, ctypes.POINTER(c_i64), ctypes.POINTER(c_i64)]),
         "gpue_join_table_first_d2h": (c_i32, [c_vp, c_vp, c_u64]),
         "gpue_join_probe_emit_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp, c_vp, ctypes.POINTER(c_u64)]),
+        "gpue_join_probe_emit_mode_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp, ctypes.POINTER(c_u64)]),
+        "gpue_join_probe_right_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_q1_join_sum": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_u64,
                                      ctypes.POINTER(c_i64), ctypes.POINTER(c_u64)]),
         "gpue_q21_star_agg": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
@@ -265,6 +267,22 @@ class Engine:
         ob = out_build._h if out_build else None
         _ck(self._lib, self._lib.gpue_join_probe_emit_i32(self._h, table._h, probe_keys._h,
                                                           n_rows, op, ob, ctypes.byref(cnt)))
+        return cnt.value
+
+    def join_probe_emit_mode(self, table, probe_keys, n_rows, mode,
+                             out_probe=None, out_build=None) -> int:
+        cnt = c_u64()
+        op = out_probe._h if out_probe else None
+        ob = out_build._h if out_build else None
+        _ck(self._lib, self._lib.gpue_join_probe_emit_mode_i32(
+            self._h, table._h, probe_keys._h, n_rows, mode, op, ob, ctypes.byref(cnt)))
+        return cnt.value
+
+    def join_probe_right(self, table, probe_keys, n_rows, anti, out_build=None) -> int:
+        cnt = c_u64()
+        ob = out_build._h if out_build else None
+        _ck(self._lib, self._lib.gpue_join_probe_right_i32(
+            self._h, table._h, probe_keys._h, n_rows, anti, ob, ctypes.byref(cnt)))
         return cnt.value
 
     def q1_join_sum(self, dates: JoinTable, od: DBuf, ep: DBuf, dc: DBuf, n):

@@ -581,6 +581,60 @@ def test_page_decode_feeds_join(engine):
     dates.destroy()
 
 
+def test_join_modes_parity(engine):
+    """LEFT SEMI/ANTI/OUTER + RIGHT SEMI/ANTI probe variants vs oracle, over
+    both RANGE_DIRECT and BUCKET_CHAINED tables."""
+    rng = np.random.default_rng(61)
+    build_keys = np.concatenate([[0], rng.integers(100, 4000, 30_000)]).astype(np.uint32)
+    probe_keys = rng.integers(0, 5000, 200_000).astype(np.uint32)
+    kb = engine.alloc(build_keys.nbytes)
+    kb.h2d(build_keys.view(np.int32))
+    pb = engine.alloc(probe_keys.nbytes)
+    pb.h2d(probe_keys.view(np.int32))
+    mn, mx = int(build_keys[1:].min()), int(build_keys[1:].max())
+
+    for kind in ("rd", "bc"):
+        if kind == "rd":
+            t = engine.join_build_range_direct(kb, len(build_keys) - 1)
+            efirst, enxt = orc.range_direct_build(build_keys.view(np.int32), mn, mx)
+            heads = orc.range_direct_lookup(probe_keys.view(np.int32), mn, mx, efirst)
+        else:
+            t = engine.join_build_bucket_chained(kb, len(build_keys) - 1)
+            efirst, enxt, bs, log = orc.bucket_chained_build(build_keys)
+            heads = orc.bucket_chained_lookup(probe_keys, efirst, bs, log)
+        for mode in (0, 1, 2, 3):
+            cnt = engine.join_probe_emit_mode(t, pb, len(probe_keys), mode)
+            op_b = engine.alloc(max(cnt, 1) * 4)
+            ob_b = engine.alloc(max(cnt, 1) * 4)
+            engine.join_probe_emit_mode(t, pb, len(probe_keys), mode, op_b, ob_b)
+            gop = op_b.d2h(np.uint32, cnt)
+            gob = ob_b.d2h(np.uint32, cnt)
+            eop, eob = orc.probe_emit_mode(build_keys, enxt, probe_keys, heads, mode)
+            assert cnt == len(eop), (kind, mode, cnt, len(eop))
+            if mode == 1:
+                # SEMI's emitted build row is chain-order-dependent: the probe
+                # row set + key equality are the contract
+                assert sorted(gop.tolist()) == sorted(eop.tolist())
+                assert (build_keys[gob] == probe_keys[gop]).all()
+            else:
+                got = np.sort(gop.astype(np.uint64) << np.uint64(32) | gob.astype(np.uint64))
+                exp = np.sort(eop.astype(np.uint64) << np.uint64(32) | eob.astype(np.uint64))
+                assert np.array_equal(got, exp), (kind, mode)
+            op_b.free()
+            ob_b.free()
+        for anti in (0, 1):
+            cnt = engine.join_probe_right(t, pb, len(probe_keys), anti)
+            ob_b = engine.alloc(max(cnt, 1) * 4)
+            engine.join_probe_right(t, pb, len(probe_keys), anti, ob_b)
+            got = np.sort(ob_b.d2h(np.uint32, cnt))
+            exp = np.sort(orc.probe_right(build_keys, enxt, probe_keys, heads, anti))
+            assert np.array_equal(got, exp), (kind, anti)
+            ob_b.free()
+        t.destroy()
+    kb.free()
+    pb.free()
+
+
 def test_partition_parity(engine):
     n, nch = 3_000_000, 8
     keys = engine.alloc(n * 4)
