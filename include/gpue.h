@@ -139,6 +139,20 @@ int gpue_join_probe_emit_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* pro
 int gpue_join_probe_emit_mode_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
                                   uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
                                   gpue_dbuf* out_build_idx, uint64_t* match_count);
+/* Nullable variants (NullableColumn is_nulls path,
+ * join_hash_map_method.hpp:56-120): null build rows never enter a chain;
+ * null probe rows match nothing (ANTI/OUTER emit them as unmatched). */
+int gpue_join_build_bucket_chained_nulls_u32(gpue_session* s, gpue_dbuf* keys,
+                                             gpue_dbuf* is_nulls /*u8, 1-based*/,
+                                             uint64_t row_count, gpue_join_table** out);
+int gpue_join_probe_emit_nulls_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
+                                   gpue_dbuf* probe_nulls /*u8*/, uint64_t n_rows, int mode,
+                                   gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                                   uint64_t* match_count);
+/* Multi-column key packing (SERIALIZED_FIXED_SIZE_BIGINT,
+ * join_hash_map_helper.h:112-136): two int32 key columns -> one 8-byte key. */
+int gpue_pack_keys_2xi32(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b, uint64_t n,
+                         gpue_dbuf* out);
 /* RIGHT SEMI (anti=0) / RIGHT ANTI (anti=1): matched/unmatched BUILD rows. */
 int gpue_join_probe_right_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
                               uint64_t n_rows, int anti, gpue_dbuf* out_build_idx,

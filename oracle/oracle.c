@@ -176,6 +176,45 @@ void orc_bucket_chained_lookup_u32(const uint32_t* probe_keys, uint32_t probe_ro
         heads[i] = first[orc_join_hash_u32(probe_keys[i], log_bucket_size)];
 }
 
+/* Nullable variants (construct_hash_table / lookup_init is_nulls paths,
+ * join_hash_map_method.hpp:56-85,101-120): null build rows are skipped
+ * (next=0 — the row never enters a chain); null probe rows get chain head 0
+ * (SIMDGather::gather with nulls, base/simd/gather.h:82-113). */
+void orc_bucket_chained_build_nulls_u32(const uint32_t* keys, const uint8_t* is_nulls,
+                                        uint32_t row_count, uint32_t* first, uint32_t* next,
+                                        uint32_t bucket_size, uint32_t log_bucket_size) {
+    (void)bucket_size;
+    const uint32_t num_rows = row_count + 1;
+    for (uint32_t i = 0; i < num_rows; i++)
+        next[i] = orc_join_hash_u32(keys[i], log_bucket_size);
+    for (uint32_t i = 0; i < num_rows; i++) {
+        if (i >= 1 && is_nulls[i] == 0) {
+            uint32_t b = next[i];
+            next[i] = first[b];
+            first[b] = i;
+        } else {
+            next[i] = 0;
+        }
+    }
+}
+
+void orc_bucket_chained_lookup_nulls_u32(const uint32_t* probe_keys, const uint8_t* is_nulls,
+                                         uint32_t probe_rows, const uint32_t* first,
+                                         uint32_t bucket_size, uint32_t log_bucket_size,
+                                         uint32_t* heads) {
+    (void)bucket_size;
+    for (uint32_t i = 0; i < probe_rows; i++)
+        heads[i] = is_nulls[i] ? 0 : first[orc_join_hash_u32(probe_keys[i], log_bucket_size)];
+}
+
+/* SERIALIZED_FIXED_SIZE key packing (join_hash_map_helper.h:112-136
+ * serialize_fixed_size_key_column + serialize_batch_at_interval): two int32
+ * key columns packed little-endian into one 8-byte key. */
+void orc_pack_keys_2xi32(const int32_t* a, const int32_t* b, uint64_t n, uint64_t* out) {
+    for (uint64_t i = 0; i < n; i++)
+        out[i] = (uint64_t)(uint32_t)a[i] | ((uint64_t)(uint32_t)b[i] << 32);
+}
+
 /* TLinearChainedJoinHashMap (join_hash_map_method.h:118-150,
  * join_hash_map_method.hpp:125-368): FP_BITS=8; hash in space
  * bucket_size<<8 / log+8; first[b] = (hash<<24 fp) | 24-bit row index;

@@ -47,6 +47,15 @@ void orc_bucket_chained_lookup_u32(const uint32_t* probe_keys, uint32_t probe_ro
                                    const uint32_t* first, uint32_t bucket_size,
                                    uint32_t log_bucket_size, uint32_t* heads);
 
+void orc_bucket_chained_build_nulls_u32(const uint32_t* keys, const uint8_t* is_nulls,
+                                        uint32_t row_count, uint32_t* first, uint32_t* next,
+                                        uint32_t bucket_size, uint32_t log_bucket_size);
+void orc_bucket_chained_lookup_nulls_u32(const uint32_t* probe_keys, const uint8_t* is_nulls,
+                                         uint32_t probe_rows, const uint32_t* first,
+                                         uint32_t bucket_size, uint32_t log_bucket_size,
+                                         uint32_t* heads);
+void orc_pack_keys_2xi32(const int32_t* a, const int32_t* b, uint64_t n, uint64_t* out);
+
 /* TLinearChained (fp-packed) build+lookup: join_hash_map_method.hpp:125-368,
  * join_hash_map_method.h:118-150 (FP_BITS=8, fp|24-bit index packed in first) */
 void orc_linear_chained_build_u32(const uint32_t* keys, uint32_t row_count,

@@ -310,6 +310,39 @@ def range_direct_lookup(probe_keys: np.ndarray, min_value, max_value, first: np.
     return heads
 
 
+def bucket_chained_build_nulls(keys_1based, nulls_1based):
+    lib = load()
+    lib.orc_bucket_chained_build_nulls_u32.argtypes = [c_vp, c_vp, c_u32, c_vp, c_vp,
+                                                       c_u32, c_u32]
+    row_count = len(keys_1based) - 1
+    bucket_size = lib.orc_calc_bucket_size(row_count + 1)
+    log = int(bucket_size - 1).bit_length()
+    first = np.zeros(bucket_size, np.uint32)
+    nxt = np.zeros(row_count + 1, np.uint32)
+    lib.orc_bucket_chained_build_nulls_u32(_p(keys_1based), _p(nulls_1based), row_count,
+                                           _p(first), _p(nxt), bucket_size, log)
+    return first, nxt, bucket_size, log
+
+
+def bucket_chained_lookup_nulls(probe_keys, probe_nulls, first, bucket_size, log):
+    lib = load()
+    lib.orc_bucket_chained_lookup_nulls_u32.argtypes = [c_vp, c_vp, c_u32, c_vp, c_u32,
+                                                        c_u32, c_vp]
+    heads = np.zeros(len(probe_keys), np.uint32)
+    lib.orc_bucket_chained_lookup_nulls_u32(_p(probe_keys), _p(probe_nulls),
+                                            len(probe_keys), _p(first), bucket_size, log,
+                                            _p(heads))
+    return heads
+
+
+def pack_keys_2xi32(a, b):
+    lib = load()
+    lib.orc_pack_keys_2xi32.argtypes = [c_vp, c_vp, c_u64, c_vp]
+    out = np.zeros(len(a), np.uint64)
+    lib.orc_pack_keys_2xi32(_p(a), _p(b), len(a), _p(out))
+    return out
+
+
 def probe_emit_mode(build_keys, nxt, probe_keys, heads, mode):
     lib = load()
     lib.orc_probe_emit_mode_u32.restype = c_u64

@@ -956,12 +956,77 @@ __global__ void k_build_bucket_chained(const uint32_t* __restrict__ keys, uint64
         next[i] = atomicExch(&first[join_hash_u32(keys[i], log_bucket_size)], (uint32_t)i);
 }
 
+__global__ void k_build_bucket_chained_nulls(const uint32_t* __restrict__ keys,
+                                             const uint8_t* __restrict__ is_nulls,
+                                             uint64_t row_count, uint32_t log_bucket_size,
+                                             uint32_t* __restrict__ first,
+                                             uint32_t* __restrict__ next) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i <= row_count;
+         i += stride) {
+        if (is_nulls[i]) continue; // null build rows never enter a chain
+        next[i] = atomicExch(&first[join_hash_u32(keys[i], log_bucket_size)], (uint32_t)i);
+    }
+}
+
+// pack two int32 key columns into one 8-byte key (the selector's
+// SERIALIZED_FIXED_SIZE_BIGINT path, join_hash_map_helper.h:112-136)
+__global__ void k_pack_keys_2xi32(const int32_t* __restrict__ a,
+                                  const int32_t* __restrict__ b, uint64_t n,
+                                  uint64_t* __restrict__ out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+        out[i] = (uint64_t)(uint32_t)a[i] | ((uint64_t)(uint32_t)b[i] << 32);
+}
+
+extern "C" int gpue_pack_keys_2xi32(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b, uint64_t n,
+                                    gpue_dbuf* out);
+int gpue_pack_keys_2xi32(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b, uint64_t n,
+                         gpue_dbuf* out) {
+    ARG_CHECK(s && a && b && out && out->bytes >= n * 8);
+    hipLaunchKernelGGL(k_pack_keys_2xi32, dim3(grid_stream(n)), dim3(BLOCK), 0, s->stream,
+                       (const int32_t*)a->ptr, (const int32_t*)b->ptr, n,
+                       (uint64_t*)out->ptr);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
 // calc_bucket_size (join_hash_map_helper.h:68-78): NormalizeCapacity(n+(n-1)/4)+1
 static uint32_t calc_bucket_size(uint32_t size) {
     uint64_t expect = (uint64_t)size + (size - 1) / 4;
     if (expect >= (1ull << 31)) return 1u << 31;
     uint64_t norm = expect ? (~0ull >> __builtin_clzll(expect)) : 1;
     return (uint32_t)(norm + 1);
+}
+
+extern "C" int gpue_join_build_bucket_chained_nulls_u32(gpue_session* s, gpue_dbuf* keys,
+                                                        gpue_dbuf* is_nulls,
+                                                        uint64_t row_count,
+                                                        gpue_join_table** out);
+int gpue_join_build_bucket_chained_nulls_u32(gpue_session* s, gpue_dbuf* keys,
+                                             gpue_dbuf* is_nulls, uint64_t row_count,
+                                             gpue_join_table** out) {
+    ARG_CHECK(s && keys && is_nulls && out && row_count > 0 && row_count + 1 < (1ull << 31));
+    ARG_CHECK(keys->bytes >= (row_count + 1) * 4 && is_nulls->bytes >= row_count + 1);
+    gpue_join_table* t = new gpue_join_table();
+    t->s = s;
+    t->kind = gpue_join_table::BUCKET_CHAINED;
+    t->row_count = row_count;
+    t->bucket_size = calc_bucket_size((uint32_t)(row_count + 1));
+    t->log_bucket_size = (uint32_t)__builtin_ctzll(t->bucket_size);
+    HIP_CHECK(hipMalloc(&t->first, t->bucket_size * sizeof(uint32_t)));
+    HIP_CHECK(hipMalloc(&t->next, (row_count + 1) * sizeof(uint32_t)));
+    HIP_CHECK(hipMalloc(&t->build_keys, (row_count + 1) * sizeof(uint32_t)));
+    HIP_CHECK(hipMemsetAsync(t->first, 0, t->bucket_size * sizeof(uint32_t), s->stream));
+    HIP_CHECK(hipMemsetAsync(t->next, 0, (row_count + 1) * sizeof(uint32_t), s->stream));
+    HIP_CHECK(hipMemcpyAsync(t->build_keys, keys->ptr, (row_count + 1) * sizeof(uint32_t),
+                             hipMemcpyDeviceToDevice, s->stream));
+    hipLaunchKernelGGL(k_build_bucket_chained_nulls, dim3(grid_for(row_count)), dim3(BLOCK),
+                       0, s->stream, t->build_keys, (const uint8_t*)is_nulls->ptr, row_count,
+                       t->log_bucket_size, t->first, t->next);
+    HIP_CHECK(hipGetLastError());
+    *out = t;
+    return GPUE_OK;
 }
 
 extern "C" int gpue_join_build_bucket_chained_u32(gpue_session* s, gpue_dbuf* keys,
@@ -1352,6 +1417,117 @@ int gpue_join_probe_emit_mode_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf
         hipLaunchKernelGGL(k_probe_emit_rd, dim3(nb), dim3(BLOCK), 0, s->stream,
                            (const int32_t*)probe_keys->ptr, n_rows, t->min_key, t->max_key,
                            t->first, t->next, mode, d_counts, d_offsets,
+                           (uint32_t*)out_probe_idx->ptr, (uint32_t*)out_build_idx->ptr);
+        HIP_CHECK(hipStreamSynchronize(s->stream));
+        (void)hipFree(d_offsets);
+    }
+    (void)hipFree(d_counts);
+    (void)hipFree(d_bsums);
+    return GPUE_OK;
+}
+
+// Nullable probe (lookup_init is_nulls path + per-type semantics): a null
+// probe key matches nothing — INNER/SEMI emit nothing for it, ANTI/OUTER
+// emit the unmatched (i, 0) row.
+__global__ void k_probe_count_bc_nulls(const uint32_t* __restrict__ probe_keys,
+                                       const uint8_t* __restrict__ is_nulls, uint64_t n,
+                                       uint32_t log_bucket_size,
+                                       const uint32_t* __restrict__ first,
+                                       const uint32_t* __restrict__ next,
+                                       const uint32_t* __restrict__ build_keys, int mode,
+                                       uint32_t* __restrict__ row_counts) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint32_t c = 0;
+        if (!is_nulls[i]) {
+            uint32_t k = probe_keys[i];
+            uint32_t b = first[join_hash_u32(k, log_bucket_size)];
+            while (b != 0) {
+                c += (build_keys[b] == k);
+                b = next[b];
+            }
+        }
+        row_counts[i] = join_mode_count(c, mode);
+    }
+}
+
+__global__ void k_probe_emit_bc_nulls(const uint32_t* __restrict__ probe_keys,
+                                      const uint8_t* __restrict__ is_nulls, uint64_t n,
+                                      uint32_t log_bucket_size,
+                                      const uint32_t* __restrict__ first,
+                                      const uint32_t* __restrict__ next,
+                                      const uint32_t* __restrict__ build_keys, int mode,
+                                      const uint32_t* __restrict__ row_counts,
+                                      const uint64_t* __restrict__ row_offsets,
+                                      uint32_t* __restrict__ out_probe,
+                                      uint32_t* __restrict__ out_build) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        if (row_counts[i] == 0) continue;
+        uint64_t pos = row_offsets[i];
+        bool any = false;
+        if (!is_nulls[i]) {
+            uint32_t k = probe_keys[i];
+            uint32_t b = first[join_hash_u32(k, log_bucket_size)];
+            while (b != 0) {
+                if (build_keys[b] == k) {
+                    any = true;
+                    if (mode != 2) {
+                        out_probe[pos] = (uint32_t)i;
+                        out_build[pos] = b;
+                        pos++;
+                    }
+                    if (mode == 1 || mode == 2) break;
+                }
+                b = next[b];
+            }
+        }
+        if (!any && (mode == 2 || mode == 3)) {
+            out_probe[pos] = (uint32_t)i;
+            out_build[pos] = 0;
+        }
+    }
+}
+
+extern "C" int gpue_join_probe_emit_nulls_i32(gpue_session* s, gpue_join_table* t,
+                                              gpue_dbuf* probe_keys, gpue_dbuf* probe_nulls,
+                                              uint64_t n_rows, int mode,
+                                              gpue_dbuf* out_probe_idx,
+                                              gpue_dbuf* out_build_idx, uint64_t* match_count);
+int gpue_join_probe_emit_nulls_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
+                                   gpue_dbuf* probe_nulls, uint64_t n_rows, int mode,
+                                   gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                                   uint64_t* match_count) {
+    ARG_CHECK(s && t && probe_keys && probe_nulls && match_count);
+    ARG_CHECK(t->kind == gpue_join_table::BUCKET_CHAINED && t->next != nullptr);
+    ARG_CHECK(probe_nulls->bytes >= n_rows);
+    uint32_t nb = grid_for(n_rows);
+    uint64_t tile = (n_rows + nb - 1) / nb;
+    uint32_t* d_counts = nullptr;
+    uint64_t* d_bsums = nullptr;
+    uint64_t* d_offsets = nullptr;
+    HIP_CHECK(hipMalloc(&d_counts, n_rows * sizeof(uint32_t)));
+    HIP_CHECK(hipMalloc(&d_bsums, (nb + 1) * sizeof(uint64_t)));
+    hipLaunchKernelGGL(k_probe_count_bc_nulls, dim3(nb), dim3(BLOCK), 0, s->stream,
+                       (const uint32_t*)probe_keys->ptr, (const uint8_t*)probe_nulls->ptr,
+                       n_rows, t->log_bucket_size, t->first, t->next, t->build_keys, mode,
+                       d_counts);
+    hipLaunchKernelGGL(k_block_sums_u32, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts,
+                       n_rows, tile, d_bsums);
+    hipLaunchKernelGGL(k_scan_small, dim3(1), dim3(1), 0, s->stream, d_bsums, nb);
+    uint64_t total = 0;
+    HIP_CHECK(hipMemcpyAsync(&total, d_bsums + nb, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    *match_count = total;
+    if (out_probe_idx && out_build_idx && total > 0) {
+        ARG_CHECK(out_probe_idx->bytes >= total * 4 && out_build_idx->bytes >= total * 4);
+        HIP_CHECK(hipMalloc(&d_offsets, n_rows * sizeof(uint64_t)));
+        hipLaunchKernelGGL(k_scan_offsets, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts,
+                           n_rows, tile, d_bsums, d_offsets);
+        hipLaunchKernelGGL(k_probe_emit_bc_nulls, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const uint32_t*)probe_keys->ptr,
+                           (const uint8_t*)probe_nulls->ptr, n_rows, t->log_bucket_size,
+                           t->first, t->next, t->build_keys, mode, d_counts, d_offsets,
                            (uint32_t*)out_probe_idx->ptr, (uint32_t*)out_build_idx->ptr);
         HIP_CHECK(hipStreamSynchronize(s->stream));
         (void)hipFree(d_offsets);

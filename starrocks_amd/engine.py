@@ -80,6 +80,9 @@ def _declare(lib):
         "gpue_join_probe_emit_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_join_probe_emit_mode_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_join_probe_right_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, ctypes.POINTER(c_u64)]),
+        "gpue_join_build_bucket_chained_nulls_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
+        "gpue_join_probe_emit_nulls_i32": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp, ctypes.POINTER(c_u64)]),
+        "gpue_pack_keys_2xi32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_q1_join_sum": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_u64,
                                      ctypes.POINTER(c_i64), ctypes.POINTER(c_u64)]),
         "gpue_q21_star_agg": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
@@ -277,6 +280,25 @@ class Engine:
         _ck(self._lib, self._lib.gpue_join_probe_emit_mode_i32(
             self._h, table._h, probe_keys._h, n_rows, mode, op, ob, ctypes.byref(cnt)))
         return cnt.value
+
+    def join_build_bucket_chained_nulls(self, keys: DBuf, nulls: DBuf, row_count):
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_join_build_bucket_chained_nulls_u32(
+            self._h, keys._h, nulls._h, row_count, ctypes.byref(h)))
+        return JoinTable(self, h)
+
+    def join_probe_emit_nulls(self, table, probe_keys, probe_nulls, n_rows, mode,
+                              out_probe=None, out_build=None) -> int:
+        cnt = c_u64()
+        op = out_probe._h if out_probe else None
+        ob = out_build._h if out_build else None
+        _ck(self._lib, self._lib.gpue_join_probe_emit_nulls_i32(
+            self._h, table._h, probe_keys._h, probe_nulls._h, n_rows, mode, op, ob,
+            ctypes.byref(cnt)))
+        return cnt.value
+
+    def pack_keys_2xi32(self, a: DBuf, b: DBuf, n, out: DBuf):
+        _ck(self._lib, self._lib.gpue_pack_keys_2xi32(self._h, a._h, b._h, n, out._h))
 
     def join_probe_right(self, table, probe_keys, n_rows, anti, out_build=None) -> int:
         cnt = c_u64()

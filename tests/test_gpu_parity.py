@@ -635,6 +635,86 @@ def test_join_modes_parity(engine):
     pb.free()
 
 
+def test_nullable_join_parity(engine):
+    """Nullable build + probe (JoinBuildProbeFuncNullable semantics,
+    reference join_hash_map_test.cpp:1190-1240): null build rows never
+    match; null probe rows match nothing (ANTI/OUTER emit them unmatched)."""
+    rng = np.random.default_rng(67)
+    nbuild, nprobe = 20_000, 150_000
+    build_keys = np.concatenate([[0], rng.integers(0, 2000, nbuild)]).astype(np.uint32)
+    build_nulls = np.concatenate([[0], (rng.random(nbuild) < 0.3)]).astype(np.uint8)
+    probe_keys = rng.integers(0, 2500, nprobe).astype(np.uint32)
+    probe_nulls = (rng.random(nprobe) < 0.2).astype(np.uint8)
+
+    kb = engine.alloc(build_keys.nbytes)
+    kb.h2d(build_keys.view(np.int32))
+    nb = engine.alloc(build_nulls.nbytes)
+    nb.h2d(build_nulls)
+    t = engine.join_build_bucket_chained_nulls(kb, nb, nbuild)
+    pb = engine.alloc(probe_keys.nbytes)
+    pb.h2d(probe_keys.view(np.int32))
+    pn = engine.alloc(probe_nulls.nbytes)
+    pn.h2d(probe_nulls)
+
+    first, nxt, bs, log = orc.bucket_chained_build_nulls(build_keys, build_nulls)
+    heads = orc.bucket_chained_lookup_nulls(probe_keys, probe_nulls, first, bs, log)
+    for mode in (0, 1, 2, 3):
+        cnt = engine.join_probe_emit_nulls(t, pb, pn, nprobe, mode)
+        op_b = engine.alloc(max(cnt, 1) * 4)
+        ob_b = engine.alloc(max(cnt, 1) * 4)
+        engine.join_probe_emit_nulls(t, pb, pn, nprobe, mode, op_b, ob_b)
+        gop = op_b.d2h(np.uint32, cnt)
+        gob = ob_b.d2h(np.uint32, cnt)
+        eop, eob = orc.probe_emit_mode(build_keys, nxt, probe_keys, heads, mode)
+        assert cnt == len(eop), (mode, cnt, len(eop))
+        if mode == 1:
+            assert sorted(gop.tolist()) == sorted(eop.tolist())
+            assert (build_keys[gob] == probe_keys[gop]).all()
+        else:
+            got = np.sort(gop.astype(np.uint64) << np.uint64(32) | gob.astype(np.uint64))
+            exp = np.sort(eop.astype(np.uint64) << np.uint64(32) | eob.astype(np.uint64))
+            assert np.array_equal(got, exp), mode
+        op_b.free()
+        ob_b.free()
+    for b in (kb, nb, pb, pn):
+        b.free()
+    t.destroy()
+
+
+def test_packed_multikey_groupby(engine):
+    """Multi-column GROUP BY via SERIALIZED_FIXED_SIZE packing: pack two i32
+    key columns to u64 on device, aggregate, compare with a numpy groupby
+    over the column PAIR."""
+    rng = np.random.default_rng(71)
+    n = 2_000_000
+    a = rng.integers(0, 50, n).astype(np.int32)
+    b = rng.integers(0, 40, n).astype(np.int32)
+    v = rng.integers(-10**9, 10**9, n).astype(np.int64)
+    ab = engine.alloc(n * 4)
+    ab.h2d(a)
+    bb = engine.alloc(n * 4)
+    bb.h2d(b)
+    keys = engine.alloc(n * 8)
+    engine.pack_keys_2xi32(ab, bb, n, keys)
+    # device packing == oracle packing
+    assert np.array_equal(keys.d2h(np.uint64, n), orc.pack_keys_2xi32(a, b))
+    vb = engine.alloc(n * 8)
+    vb.h2d(v)
+    outs = [engine.alloc(4000 * 8) for _ in range(2)]
+    g = engine.hash_agg_sum_u64(keys, vb, n, outs[0], outs[1], None, 4000)
+    gk = outs[0].d2h(np.uint64, g)
+    gs = outs[1].d2h(np.int64, g)
+    packed = a.astype(np.uint64) | (b.astype(np.uint64) << np.uint64(32))
+    uk, inv = np.unique(packed, return_inverse=True)
+    es = np.zeros(len(uk), np.int64)
+    np.add.at(es, inv, v)
+    order = np.argsort(gk)
+    assert np.array_equal(gk[order], uk)
+    assert np.array_equal(gs[order], es)
+    for x in [ab, bb, keys, vb] + outs:
+        x.free()
+
+
 def test_partition_parity(engine):
     n, nch = 3_000_000, 8
     keys = engine.alloc(n * 4)
