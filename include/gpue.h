@@ -153,6 +153,20 @@ int gpue_join_probe_emit_nulls_i32(gpue_session* s, gpue_join_table* t, gpue_dbu
  * join_hash_map_helper.h:112-136): two int32 key columns -> one 8-byte key. */
 int gpue_pack_keys_2xi32(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b, uint64_t n,
                          gpue_dbuf* out);
+/* SERIALIZED_VARCHAR / Slice keys (the selector's remaining constructor
+ * branch, join_hash_map.cpp:269-281 -> JoinKeyHash<Slice>,
+ * join_hash_map_helper.h:57-64: crc_hash_32(bytes,len,CRC_HASH_SEED1) masked
+ * to the bucket count; probe verifies with a byte compare along the chain).
+ * Columns are BinaryColumn-shaped: a bytes buffer + uint32 offsets
+ * (binary_column.h:458-459). Build rows are 1-based with row 0 the empty
+ * sentinel, so offsets has row_count+2 entries; probe rows are 0-based with
+ * n_rows+1 entries. */
+int gpue_join_build_varchar(gpue_session* s, gpue_dbuf* bytes, gpue_dbuf* offsets,
+                            uint64_t row_count, gpue_join_table** out);
+int gpue_join_probe_emit_varchar(gpue_session* s, gpue_join_table* t, gpue_dbuf* pbytes,
+                                 gpue_dbuf* poffsets, uint64_t n_rows,
+                                 gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                                 uint64_t* match_count);
 /* RIGHT SEMI (anti=0) / RIGHT ANTI (anti=1): matched/unmatched BUILD rows. */
 int gpue_join_probe_right_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
                               uint64_t n_rows, int anti, gpue_dbuf* out_build_idx,

@@ -176,6 +176,49 @@ void orc_bucket_chained_lookup_u32(const uint32_t* probe_keys, uint32_t probe_ro
         heads[i] = first[orc_join_hash_u32(probe_keys[i], log_bucket_size)];
 }
 
+/* SERIALIZED_VARCHAR / Slice keys (the selector's last constructor branch,
+ * join_hash_table.cpp:215-217): JoinKeyHash<Slice> = crc_hash_32(bytes, len,
+ * 0x811C9DC5) & (bucket_size-1) (join_hash_map_helper.h:57-64); chains walk
+ * with a byte compare. Columns are BinaryColumn-shaped: bytes + uint32
+ * offsets (binary_column.h:458-459), rows 1-based with row 0 = empty
+ * sentinel. */
+void orc_slice_build_u32(const uint8_t* bytes, const uint32_t* offsets, uint32_t row_count,
+                         uint32_t* first, uint32_t* next, uint32_t bucket_size,
+                         uint32_t log_bucket_size) {
+    (void)log_bucket_size;
+    for (uint32_t i = 1; i <= row_count; i++) {
+        uint32_t len = offsets[i + 1] - offsets[i];
+        uint32_t b = orc_crc_hash_32(bytes + offsets[i], (int32_t)len, 0x811C9DC5u) &
+                     (bucket_size - 1);
+        next[i] = first[b];
+        first[b] = i;
+    }
+}
+
+uint64_t orc_slice_probe_emit(const uint8_t* bbytes, const uint32_t* boffsets,
+                              const uint32_t* next, uint32_t bucket_size,
+                              const uint32_t* first, const uint8_t* pbytes,
+                              const uint32_t* poffsets, uint32_t probe_rows,
+                              uint32_t* out_probe_idx, uint32_t* out_build_idx) {
+    uint64_t m = 0;
+    for (uint32_t i = 0; i < probe_rows; i++) {
+        uint32_t len = poffsets[i + 1] - poffsets[i];
+        uint32_t b = orc_crc_hash_32(pbytes + poffsets[i], (int32_t)len, 0x811C9DC5u) &
+                     (bucket_size - 1);
+        uint32_t j = first[b];
+        while (j != 0) {
+            uint32_t blen = boffsets[j + 1] - boffsets[j];
+            if (blen == len && memcmp(bbytes + boffsets[j], pbytes + poffsets[i], len) == 0) {
+                out_probe_idx[m] = i;
+                out_build_idx[m] = j;
+                m++;
+            }
+            j = next[j];
+        }
+    }
+    return m;
+}
+
 /* Nullable variants (construct_hash_table / lookup_init is_nulls paths,
  * join_hash_map_method.hpp:56-85,101-120): null build rows are skipped
  * (next=0 — the row never enters a chain); null probe rows get chain head 0

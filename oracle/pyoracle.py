@@ -71,6 +71,9 @@ def _decl(lib):
     lib.orc_linear_chained_lookup_u32.argtypes = [c_vp, c_vp, u, c_vp, u, u, c_vp]
     lib.orc_range_direct_build_i32.argtypes = [c_vp, u, c_i64, c_vp, c_vp]
     lib.orc_range_direct_lookup_i32.argtypes = [c_vp, c_u64, c_i64, c_i64, c_vp, c_vp]
+    lib.orc_slice_build_u32.argtypes = [c_vp, c_vp, u, c_vp, c_vp, u, u]
+    lib.orc_slice_probe_emit.restype = c_u64
+    lib.orc_slice_probe_emit.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, c_vp, u, c_vp, c_vp]
     lib.orc_probe_emit_u32.restype = c_u64
     lib.orc_probe_emit_u32.argtypes = [c_vp, c_vp, c_vp, c_vp, u, c_i32, c_vp, c_vp]
     lib.orc_filter_i64_lt.restype = c_u64
@@ -127,6 +130,25 @@ def fnv_hash(data: bytes, seed: int) -> int:
 
 def join_hash_slice(data: bytes, num_buckets: int) -> int:
     return load().orc_join_hash_slice(data, len(data), num_buckets)
+
+
+def slice_join(bbytes, boffsets, row_count, pbytes, poffsets, probe_rows, max_out):
+    """Build a Slice-key chained table over build rows 1..row_count and emit
+    all (probe_idx, build_idx) pairs. bbytes/pbytes uint8 arrays;
+    boffsets uint32[row_count+2] (row 0 = empty sentinel), poffsets
+    uint32[probe_rows+1]."""
+    lib = load()
+    bucket_size = lib.orc_calc_bucket_size(row_count + 1)
+    log = int(bucket_size).bit_length() - 1
+    first = np.zeros(bucket_size, np.uint32)
+    nxt = np.zeros(row_count + 1, np.uint32)
+    lib.orc_slice_build_u32(_p(bbytes), _p(boffsets), row_count, _p(first), _p(nxt),
+                            bucket_size, log)
+    op = np.empty(max_out, np.uint32)
+    ob = np.empty(max_out, np.uint32)
+    m = lib.orc_slice_probe_emit(_p(bbytes), _p(boffsets), _p(nxt), bucket_size, _p(first),
+                                 _p(pbytes), _p(poffsets), probe_rows, _p(op), _p(ob))
+    return op[:m], ob[:m]
 
 
 def q1_pipeline(seed, row_start, n_rows, year, threads=0):

@@ -134,10 +134,12 @@ struct gpue_join_table {
     // method discriminator — the GPU analog of JoinHashMapSelector's choice
     // (reference join_hash_table.cpp:164-344)
     enum Kind { PAYLOAD = 0, RANGE_DIRECT = 1, BUCKET_CHAINED = 2,
-                LINEAR_CHAINED = 3 } kind = PAYLOAD;
+                LINEAR_CHAINED = 3, VARCHAR = 4 } kind = PAYLOAD;
     uint32_t log_bucket_size = 0;
     uint32_t* build_keys = nullptr; // chained methods keep the build keys for the
                                     // probe-side equality check (1-based, row 0 sentinel)
+    uint8_t* key_bytes = nullptr;   // VARCHAR: BinaryColumn bytes + uint32 offsets
+    uint32_t* key_offsets = nullptr;
 };
 
 int gpue_device_count(int* out) {
@@ -918,6 +920,8 @@ void gpue_join_table_destroy(gpue_join_table* t) {
     if (t->bitset) (void)hipFree(t->bitset);
     if (t->first16) (void)hipFree(t->first16);
     if (t->build_keys) (void)hipFree(t->build_keys);
+    if (t->key_bytes) (void)hipFree(t->key_bytes);
+    if (t->key_offsets) (void)hipFree(t->key_offsets);
     delete t;
 }
 
@@ -1417,6 +1421,179 @@ int gpue_join_probe_emit_mode_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf
         hipLaunchKernelGGL(k_probe_emit_rd, dim3(nb), dim3(BLOCK), 0, s->stream,
                            (const int32_t*)probe_keys->ptr, n_rows, t->min_key, t->max_key,
                            t->first, t->next, mode, d_counts, d_offsets,
+                           (uint32_t*)out_probe_idx->ptr, (uint32_t*)out_build_idx->ptr);
+        HIP_CHECK(hipStreamSynchronize(s->stream));
+        (void)hipFree(d_offsets);
+    }
+    (void)hipFree(d_counts);
+    (void)hipFree(d_bsums);
+    return GPUE_OK;
+}
+
+// ---------------------------------------------------------------------------
+// SERIALIZED_VARCHAR / Slice keys (the selector's last constructor branch):
+// JoinKeyHash<Slice> = crc_hash_32(bytes, len, 0x811C9DC5) & (bucket_size-1)
+// (join_hash_map_helper.h:57-64; CRC32-C + phmap_mix<4>, bitwise device
+// form identical to the oracle's table form), chains walked with a byte
+// compare (hash + verify). Columns are BinaryColumn-shaped: bytes + uint32
+// offsets (binary_column.h:458-459), rows 1-based with row 0 = empty.
+// ---------------------------------------------------------------------------
+__device__ static inline uint32_t crc_hash_32_dev(const uint8_t* p, uint32_t len,
+                                                  uint32_t seed) {
+    uint32_t h = seed;
+    for (uint32_t i = 0; i < len; i++) {
+        h ^= p[i];
+        #pragma unroll
+        for (int k = 0; k < 8; k++)
+            h = (h >> 1) ^ (0x82F63B78u & (uint32_t) - (int)(h & 1));
+    }
+    uint64_t l = (uint64_t)h * 0xcc9e2d51ull;
+    return (uint32_t)(l ^ (l >> 32));
+}
+
+__global__ void k_build_varchar(const uint8_t* __restrict__ bytes,
+                                const uint32_t* __restrict__ offsets, uint64_t row_count,
+                                uint32_t bucket_mask, uint32_t* __restrict__ first,
+                                uint32_t* __restrict__ next) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i <= row_count;
+         i += stride) {
+        uint32_t len = offsets[i + 1] - offsets[i];
+        uint32_t b = crc_hash_32_dev(bytes + offsets[i], len, 0x811C9DC5u) & bucket_mask;
+        next[i] = atomicExch(&first[b], (uint32_t)i);
+    }
+}
+
+__device__ static inline bool slice_eq(const uint8_t* a, uint32_t alen, const uint8_t* b,
+                                       uint32_t blen) {
+    if (alen != blen) return false;
+    for (uint32_t k = 0; k < alen; k++)
+        if (a[k] != b[k]) return false;
+    return true;
+}
+
+__global__ void k_probe_count_vc(const uint8_t* __restrict__ pbytes,
+                                 const uint32_t* __restrict__ poffsets, uint64_t n,
+                                 uint32_t bucket_mask, const uint32_t* __restrict__ first,
+                                 const uint32_t* __restrict__ next,
+                                 const uint8_t* __restrict__ bbytes,
+                                 const uint32_t* __restrict__ boffsets,
+                                 uint32_t* __restrict__ row_counts) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint32_t len = poffsets[i + 1] - poffsets[i];
+        uint32_t b = crc_hash_32_dev(pbytes + poffsets[i], len, 0x811C9DC5u) & bucket_mask;
+        uint32_t j = first[b];
+        uint32_t c = 0;
+        while (j != 0) {
+            c += slice_eq(bbytes + boffsets[j], boffsets[j + 1] - boffsets[j],
+                          pbytes + poffsets[i], len);
+            j = next[j];
+        }
+        row_counts[i] = c;
+    }
+}
+
+__global__ void k_probe_emit_vc(const uint8_t* __restrict__ pbytes,
+                                const uint32_t* __restrict__ poffsets, uint64_t n,
+                                uint32_t bucket_mask, const uint32_t* __restrict__ first,
+                                const uint32_t* __restrict__ next,
+                                const uint8_t* __restrict__ bbytes,
+                                const uint32_t* __restrict__ boffsets,
+                                const uint32_t* __restrict__ row_counts,
+                                const uint64_t* __restrict__ row_offsets,
+                                uint32_t* __restrict__ out_probe,
+                                uint32_t* __restrict__ out_build) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        if (row_counts[i] == 0) continue;
+        uint64_t pos = row_offsets[i];
+        uint32_t len = poffsets[i + 1] - poffsets[i];
+        uint32_t b = crc_hash_32_dev(pbytes + poffsets[i], len, 0x811C9DC5u) & bucket_mask;
+        uint32_t j = first[b];
+        while (j != 0) {
+            if (slice_eq(bbytes + boffsets[j], boffsets[j + 1] - boffsets[j],
+                         pbytes + poffsets[i], len)) {
+                out_probe[pos] = (uint32_t)i;
+                out_build[pos] = j;
+                pos++;
+            }
+            j = next[j];
+        }
+    }
+}
+
+extern "C" {
+int gpue_join_build_varchar(gpue_session* s, gpue_dbuf* bytes, gpue_dbuf* offsets,
+                            uint64_t row_count, gpue_join_table** out);
+int gpue_join_probe_emit_varchar(gpue_session* s, gpue_join_table* t, gpue_dbuf* pbytes,
+                                 gpue_dbuf* poffsets, uint64_t n_rows,
+                                 gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                                 uint64_t* match_count);
+}
+
+int gpue_join_build_varchar(gpue_session* s, gpue_dbuf* bytes, gpue_dbuf* offsets,
+                            uint64_t row_count, gpue_join_table** out) {
+    ARG_CHECK(s && bytes && offsets && out && row_count > 0);
+    ARG_CHECK(offsets->bytes >= (row_count + 2) * 4);
+    gpue_join_table* t = new gpue_join_table();
+    t->s = s;
+    t->kind = gpue_join_table::VARCHAR;
+    t->row_count = row_count;
+    t->bucket_size = calc_bucket_size((uint32_t)(row_count + 1));
+    t->log_bucket_size = (uint32_t)__builtin_ctzll(t->bucket_size);
+    HIP_CHECK(hipMalloc(&t->first, t->bucket_size * sizeof(uint32_t)));
+    HIP_CHECK(hipMalloc(&t->next, (row_count + 1) * sizeof(uint32_t)));
+    HIP_CHECK(hipMalloc(&t->key_bytes, bytes->bytes));
+    HIP_CHECK(hipMalloc(&t->key_offsets, (row_count + 2) * sizeof(uint32_t)));
+    HIP_CHECK(hipMemsetAsync(t->first, 0, t->bucket_size * sizeof(uint32_t), s->stream));
+    HIP_CHECK(hipMemsetAsync(t->next, 0, (row_count + 1) * sizeof(uint32_t), s->stream));
+    HIP_CHECK(hipMemcpyAsync(t->key_bytes, bytes->ptr, bytes->bytes,
+                             hipMemcpyDeviceToDevice, s->stream));
+    HIP_CHECK(hipMemcpyAsync(t->key_offsets, offsets->ptr, (row_count + 2) * 4,
+                             hipMemcpyDeviceToDevice, s->stream));
+    hipLaunchKernelGGL(k_build_varchar, dim3(grid_for(row_count)), dim3(BLOCK), 0, s->stream,
+                       t->key_bytes, t->key_offsets, row_count,
+                       (uint32_t)(t->bucket_size - 1), t->first, t->next);
+    HIP_CHECK(hipGetLastError());
+    *out = t;
+    return GPUE_OK;
+}
+
+int gpue_join_probe_emit_varchar(gpue_session* s, gpue_join_table* t, gpue_dbuf* pbytes,
+                                 gpue_dbuf* poffsets, uint64_t n_rows,
+                                 gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                                 uint64_t* match_count) {
+    ARG_CHECK(s && t && pbytes && poffsets && match_count);
+    ARG_CHECK(t->kind == gpue_join_table::VARCHAR);
+    ARG_CHECK(poffsets->bytes >= (n_rows + 1) * 4);
+    uint32_t nb = grid_for(n_rows);
+    uint64_t tile = (n_rows + nb - 1) / nb;
+    uint32_t* d_counts = nullptr;
+    uint64_t* d_bsums = nullptr;
+    uint64_t* d_offsets = nullptr;
+    HIP_CHECK(hipMalloc(&d_counts, n_rows * 4));
+    HIP_CHECK(hipMalloc(&d_bsums, (nb + 1) * 8));
+    hipLaunchKernelGGL(k_probe_count_vc, dim3(nb), dim3(BLOCK), 0, s->stream,
+                       (const uint8_t*)pbytes->ptr, (const uint32_t*)poffsets->ptr, n_rows,
+                       (uint32_t)(t->bucket_size - 1), t->first, t->next, t->key_bytes,
+                       t->key_offsets, d_counts);
+    hipLaunchKernelGGL(k_block_sums_u32, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts,
+                       n_rows, tile, d_bsums);
+    hipLaunchKernelGGL(k_scan_small, dim3(1), dim3(1), 0, s->stream, d_bsums, nb);
+    uint64_t total = 0;
+    HIP_CHECK(hipMemcpyAsync(&total, d_bsums + nb, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    *match_count = total;
+    if (out_probe_idx && out_build_idx && total > 0) {
+        ARG_CHECK(out_probe_idx->bytes >= total * 4 && out_build_idx->bytes >= total * 4);
+        HIP_CHECK(hipMalloc(&d_offsets, n_rows * 8));
+        hipLaunchKernelGGL(k_scan_offsets, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts,
+                           n_rows, tile, d_bsums, d_offsets);
+        hipLaunchKernelGGL(k_probe_emit_vc, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const uint8_t*)pbytes->ptr, (const uint32_t*)poffsets->ptr,
+                           n_rows, (uint32_t)(t->bucket_size - 1), t->first, t->next,
+                           t->key_bytes, t->key_offsets, d_counts, d_offsets,
                            (uint32_t*)out_probe_idx->ptr, (uint32_t*)out_build_idx->ptr);
         HIP_CHECK(hipStreamSynchronize(s->stream));
         (void)hipFree(d_offsets);

@@ -113,6 +113,9 @@ def _declare(lib):
                                      [c_u64, ctypes.POINTER(c_u64)]),
         "gpue_hash_agg_sum_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_u64, c_vp, c_vp, c_vp,
                                           c_u64, ctypes.POINTER(c_u64)]),
+        "gpue_join_build_varchar": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
+        "gpue_join_probe_emit_varchar": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_vp, c_vp,
+                                                 ctypes.POINTER(c_u64)]),
         "gpue_topk_i64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp]),
         "gpue_page_decode_bshuf_lz4_i32": (c_i32, [c_vp, c_vp, c_u32, c_vp]),
         "gpue_timer_start": (c_i32, [c_vp]),
@@ -295,6 +298,21 @@ class Engine:
         _ck(self._lib, self._lib.gpue_join_probe_emit_nulls_i32(
             self._h, table._h, probe_keys._h, probe_nulls._h, n_rows, mode, op, ob,
             ctypes.byref(cnt)))
+        return cnt.value
+
+    def join_build_varchar(self, bytes_: DBuf, offsets: DBuf, row_count) -> JoinTable:
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_join_build_varchar(
+            self._h, bytes_._h, offsets._h, row_count, ctypes.byref(h)))
+        return JoinTable(self, h)
+
+    def join_probe_emit_varchar(self, table, pbytes: DBuf, poffsets: DBuf, n_rows,
+                                out_probe=None, out_build=None) -> int:
+        cnt = c_u64()
+        op = out_probe._h if out_probe else None
+        ob = out_build._h if out_build else None
+        _ck(self._lib, self._lib.gpue_join_probe_emit_varchar(
+            self._h, table._h, pbytes._h, poffsets._h, n_rows, op, ob, ctypes.byref(cnt)))
         return cnt.value
 
     def pack_keys_2xi32(self, a: DBuf, b: DBuf, n, out: DBuf):

@@ -733,3 +733,51 @@ def test_partition_parity(engine):
         assert len(np.unique(rows)) == len(rows)
     keys.free()
     ri.free()
+
+
+def _varchar_cols(rng, n, card, one_based):
+    """BinaryColumn-shaped (bytes, offsets) with duplicate-heavy strings."""
+    pool = [f"city_{i:04d}".encode() + b"x" * int(rng.integers(0, 9)) for i in range(card)]
+    rows = [pool[int(i)] for i in rng.integers(0, card, n)]
+    if one_based:
+        rows = [b""] + rows  # row 0 = empty sentinel
+    offsets = np.zeros(len(rows) + 1, np.uint32)
+    np.cumsum([len(r) for r in rows], out=offsets[1:])
+    return np.frombuffer(b"".join(rows), np.uint8).copy(), offsets, rows
+
+
+def test_varchar_join_parity(engine):
+    """SERIALIZED_VARCHAR / Slice keys (join_hash_map.cpp:269-281): GPU
+    crc-hash chained build + byte-verify probe vs the oracle restatement —
+    match-pair multisets equal; the Slice hash itself is pinned by the
+    ref-shim crc cross-check in test_oracle_golden."""
+    rng = np.random.default_rng(77)
+    n_build, n_probe, card = 40_000, 200_000, 700
+    bb, bo, brows = _varchar_cols(rng, n_build, card, one_based=True)
+    pb_, po, prows = _varchar_cols(rng, n_probe, 2 * card, one_based=False)
+
+    eop, eob = orc.slice_join(bb, bo, n_build, pb_, po, n_probe, 64_000_000)
+
+    d_bb, d_bo = engine.alloc(bb.nbytes), engine.alloc(bo.nbytes)
+    d_bb.h2d(bb)
+    d_bo.h2d(bo)
+    t = engine.join_build_varchar(d_bb, d_bo, n_build)
+    d_pb, d_po = engine.alloc(pb_.nbytes), engine.alloc(po.nbytes)
+    d_pb.h2d(pb_)
+    d_po.h2d(po)
+    cnt = engine.join_probe_emit_varchar(t, d_pb, d_po, n_probe)
+    assert cnt == len(eop)
+    op_buf, ob_buf = engine.alloc(max(cnt, 1) * 4), engine.alloc(max(cnt, 1) * 4)
+    cnt2 = engine.join_probe_emit_varchar(t, d_pb, d_po, n_probe, op_buf, ob_buf)
+    assert cnt2 == cnt
+    gop = op_buf.d2h(np.uint32, cnt)
+    gob = ob_buf.d2h(np.uint32, cnt)
+    # every emitted pair joins equal strings
+    for i in rng.integers(0, cnt, 200):
+        assert prows[gop[i]] == brows[gob[i]]
+    got = np.sort(gop.astype(np.uint64) << np.uint64(32) | gob.astype(np.uint64))
+    exp = np.sort(eop.astype(np.uint64) << np.uint64(32) | eob.astype(np.uint64))
+    assert np.array_equal(got, exp)
+    for b in (d_bb, d_bo, d_pb, d_po, op_buf, ob_buf):
+        b.free()
+    t.destroy()

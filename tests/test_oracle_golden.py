@@ -216,3 +216,25 @@ def test_q21_pipeline_small_vs_numpy():
     expect = np.zeros(7000, np.int64)
     np.add.at(expect, gid, rv[ok].astype(np.int64))
     assert np.array_equal(gs, expect)
+
+
+def test_slice_join_oracle_vs_brute():
+    """Slice-key chained join (orc_slice_build_u32/orc_slice_probe_emit) vs a
+    dict-of-strings brute force: identical match-pair multisets."""
+    rng = np.random.default_rng(31)
+    pool = [f"s{i}".encode() + b"y" * int(rng.integers(0, 7)) for i in range(150)]
+    brows = [b""] + [pool[int(i)] for i in rng.integers(0, 150, 3000)]
+    prows = [pool[int(i) % 150] if i % 4 else b"missing" for i in rng.integers(0, 600, 9000)]
+    bo = np.zeros(len(brows) + 1, np.uint32)
+    np.cumsum([len(r) for r in brows], out=bo[1:])
+    po = np.zeros(len(prows) + 1, np.uint32)
+    np.cumsum([len(r) for r in prows], out=po[1:])
+    bb = np.frombuffer(b"".join(brows), np.uint8).copy()
+    pb = np.frombuffer(b"".join(prows), np.uint8).copy()
+    op, ob = orc.slice_join(bb, bo, len(brows) - 1, pb, po, len(prows), 10_000_000)
+    index = {}
+    for j in range(1, len(brows)):
+        index.setdefault(brows[j], []).append(j)
+    expect = sorted((i, j) for i, s in enumerate(prows) for j in index.get(s, []))
+    assert sorted(zip(op.tolist(), ob.tolist())) == expect
+    assert len(expect) > 0
