@@ -51,6 +51,13 @@ void gpue_dbuf_free(gpue_dbuf* b);
 int gpue_dbuf_h2d(gpue_dbuf* b, const void* src, uint64_t bytes, uint64_t dst_off);
 int gpue_dbuf_d2h(gpue_dbuf* b, void* dst, uint64_t bytes, uint64_t src_off);
 int gpue_dbuf_memset(gpue_dbuf* b, int value, uint64_t bytes);
+int gpue_dbuf_d2d(gpue_dbuf* src, gpue_dbuf* dst, uint64_t bytes, uint64_t src_off,
+                  uint64_t dst_off);
+/* SUM(a[i]*b[i]) + row count accumulated into acc (i64[2]) — the agg sink
+ * update for the chunked (unfused) config-2 plan (Aggregator::update_batch
+ * with sum<int64> states, reference be/src/exprs/agg/sum.h:45-181). */
+int gpue_sum_prod_u32(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b, uint64_t n,
+                      gpue_dbuf* acc);
 /* Wrap external device memory (e.g. a torch tensor's data_ptr) so kernels
  * operate in place and torch.distributed (RCCL) moves the same buffers —
  * the all-to-all leg of configs 4-5. Caller keeps ownership. */

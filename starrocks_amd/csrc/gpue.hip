@@ -254,6 +254,57 @@ int gpue_dbuf_memset(gpue_dbuf* b, int value, uint64_t bytes) {
     return GPUE_OK;
 }
 
+extern "C" int gpue_dbuf_d2d(gpue_dbuf* src, gpue_dbuf* dst, uint64_t bytes,
+                             uint64_t src_off, uint64_t dst_off);
+int gpue_dbuf_d2d(gpue_dbuf* src, gpue_dbuf* dst, uint64_t bytes, uint64_t src_off,
+                  uint64_t dst_off) {
+    ARG_CHECK(src && dst && src->bytes >= src_off + bytes && dst->bytes >= dst_off + bytes);
+    HIP_CHECK(hipMemcpy((uint8_t*)dst->ptr + dst_off, (const uint8_t*)src->ptr + src_off,
+                        bytes, hipMemcpyDeviceToDevice));
+    return GPUE_OK;
+}
+
+__global__ void k_add_u64(unsigned long long* p, unsigned long long v) { *p += v; }
+
+// SUM(a[i]*b[i]) + COUNT over a gathered match chunk — the agg sink's update
+// for the chunked (unfused) config-2 plan: Aggregator::update_batch with a
+// sum<int64> state (reference be/src/exprs/agg/sum.h:45-181). Block-reduce in
+// LDS, one atomic per block.
+__global__ void k_sum_prod_u32(const uint32_t* __restrict__ a,
+                               const uint32_t* __restrict__ b, uint64_t n,
+                               int64_t* __restrict__ acc) {
+    __shared__ long long red[BLOCK / 64];
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    long long local = 0;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+        local += (long long)a[i] * b[i];
+    for (int off = 32; off > 0; off >>= 1)
+        local += __shfl_down(local, off, 64);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = local;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        long long s = 0;
+        for (int w = 0; w < BLOCK / 64; w++) s += red[w];
+        atomicAdd((unsigned long long*)acc, (unsigned long long)s);
+    }
+}
+
+extern "C" int gpue_sum_prod_u32(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b, uint64_t n,
+                                 gpue_dbuf* acc /*i64[2]: sum, count*/);
+int gpue_sum_prod_u32(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b, uint64_t n,
+                      gpue_dbuf* acc) {
+    ARG_CHECK(s && a && b && acc && acc->bytes >= 16);
+    ARG_CHECK(a->bytes >= n * 4 && b->bytes >= n * 4);
+    if (n > 0)
+        hipLaunchKernelGGL(k_sum_prod_u32, dim3(grid_stream(n)), dim3(BLOCK), 0, s->stream,
+                           (const uint32_t*)a->ptr, (const uint32_t*)b->ptr, n,
+                           (int64_t*)acc->ptr);
+    hipLaunchKernelGGL(k_add_u64, dim3(1), dim3(1), 0, s->stream,
+                       (unsigned long long*)acc->ptr + 1, (unsigned long long)n);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
 // ---------------------------------------------------------------------------
 // deterministic synthetic generator — splitmix64 finalizer, counter-based.
 // MUST stay identical to oracle/oracle.c sm64/orc_gen_u64 and the numpy

@@ -113,6 +113,8 @@ def _declare(lib):
                                      [c_u64, ctypes.POINTER(c_u64)]),
         "gpue_hash_agg_sum_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_u64, c_vp, c_vp, c_vp,
                                           c_u64, ctypes.POINTER(c_u64)]),
+        "gpue_dbuf_d2d": (c_i32, [c_vp, c_vp, c_u64, c_u64, c_u64]),
+        "gpue_sum_prod_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_join_build_varchar": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_probe_emit_varchar": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_vp, c_vp,
                                                  ctypes.POINTER(c_u64)]),
@@ -447,6 +449,12 @@ class Engine:
             self._h, keys._h, vals._h, n, capacity_hint, out_keys._h, out_lo._h,
             out_hi._h, max_out, ctypes.byref(g)))
         return g.value
+
+    def dbuf_d2d(self, src: DBuf, dst: DBuf, nbytes, src_off=0, dst_off=0):
+        _ck(self._lib, self._lib.gpue_dbuf_d2d(src._h, dst._h, nbytes, src_off, dst_off))
+
+    def sum_prod_u32(self, a: DBuf, b: DBuf, n, acc: DBuf):
+        _ck(self._lib, self._lib.gpue_sum_prod_u32(self._h, a._h, b._h, n, acc._h))
 
     def gather_u32(self, inp: DBuf, idx: DBuf, n, out: DBuf):
         _ck(self._lib, self._lib.gpue_gather_u32(self._h, inp._h, idx._h, n, out._h))

@@ -781,3 +781,15 @@ def test_varchar_join_parity(engine):
     for b in (d_bb, d_bo, d_pb, d_po, op_buf, ob_buf):
         b.free()
     t.destroy()
+
+
+def test_q1_operator_pipeline_parity(engine):
+    """The chunked operator plan (source -> build -> probe+gather -> agg sink,
+    starrocks_amd/pipeline.py mirroring operator.h:141-144 push/pull) produces
+    the SAME bit-exact (sum, count) as the fused q1 kernel and the oracle —
+    the §8b boundary exercised end to end over multiple bounded chunks."""
+    from starrocks_amd.pipeline import q1_operator_pipeline
+    n = 3_000_000
+    got = q1_operator_pipeline(engine, SEED, n, chunk_rows=700_000)
+    expect = orc.q1_pipeline(SEED, 0, n, 1993)
+    assert got == expect
