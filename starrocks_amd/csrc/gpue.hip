@@ -3441,6 +3441,47 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
     }
 }
 
+// Decomposition instrument for q3's roofline attribution:
+// legs bitmask accumulates per-leg cost — 1: ship stream, 2: +lk read &
+// order-bits gather, 4: +ext/disc product, 8: +hash-table insert (the full
+// kernel). Local accumulate + one atomic per block prevents DCE.
+__global__ void k_q3_legs(const int64_t* __restrict__ lk,
+                          const int64_t* __restrict__ ext,
+                          const int64_t* __restrict__ disc,
+                          const int32_t* __restrict__ ship, uint64_t n,
+                          const uint32_t* __restrict__ order_bits, int32_t ship_cutoff,
+                          int legs, unsigned long long* __restrict__ slots,
+                          unsigned long long* __restrict__ sums, uint64_t cap_mask,
+                          unsigned long long* __restrict__ sink) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    unsigned long long acc = 0;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        if (ship[i] <= ship_cutoff) continue;
+        acc++;
+        if (!(legs & 2)) continue;
+        unsigned long long k = (unsigned long long)lk[i];
+        uint64_t o = k - 1;
+        if (!((order_bits[o >> 5] >> (o & 31)) & 1u)) continue;
+        acc++;
+        if (!(legs & 4)) continue;
+        unsigned long long v = (unsigned long long)(ext[i] * (100 - disc[i]));
+        acc += v & 1;
+        if (!(legs & 8)) continue;
+        uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        for (;;) {
+            unsigned long long cur = slots[s];
+            if (cur == k) { atomicAdd(&sums[s], v); break; }
+            if (cur == AGG_EMPTY) {
+                unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
+                if (old == AGG_EMPTY || old == k) { atomicAdd(&sums[s], v); break; }
+            }
+            s = (s + 1) & cap_mask;
+        }
+    }
+    for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
+    if (threadIdx.x == 0) atomicAdd(sink, acc);
+}
+
 extern "C" {
 int gpue_gen_lineitem_q3(gpue_session* s, uint64_t seed, uint64_t row_start, uint64_t n,
                          uint64_t n_orders, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
@@ -3456,6 +3497,26 @@ int gpue_q3_probe_agg(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf*
                       gpue_dbuf* ship, uint64_t n, gpue_dbuf* order_bits, int32_t ship_cutoff,
                       uint64_t capacity_hint, gpue_dbuf* out_keys, gpue_dbuf* out_sums,
                       uint64_t max_out, uint64_t* n_groups);
+}
+
+extern "C" int gpue_q3_decomp(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext,
+                              gpue_dbuf* disc, gpue_dbuf* ship, uint64_t n,
+                              gpue_dbuf* order_bits, int32_t ship_cutoff, int legs,
+                              gpue_agg_table* at, gpue_dbuf* sink);
+int gpue_q3_decomp(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
+                   gpue_dbuf* ship, uint64_t n, gpue_dbuf* order_bits, int32_t ship_cutoff,
+                   int legs, gpue_agg_table* at, gpue_dbuf* sink) {
+    ARG_CHECK(s && lk && ext && disc && ship && order_bits && sink && sink->bytes >= 8);
+    ARG_CHECK(!(legs & 8) || at);
+    hipLaunchKernelGGL(k_q3_legs, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))),
+                       dim3(BLOCK), 0, s->stream, (const int64_t*)lk->ptr,
+                       (const int64_t*)ext->ptr, (const int64_t*)disc->ptr,
+                       (const int32_t*)ship->ptr, n, (const uint32_t*)order_bits->ptr,
+                       ship_cutoff, legs, at ? at->slots : nullptr,
+                       at ? at->sums : nullptr, at ? at->cap - 1 : 0,
+                       (unsigned long long*)sink->ptr);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
 }
 
 int gpue_gen_lineitem_q3(gpue_session* s, uint64_t seed, uint64_t row_start, uint64_t n,

@@ -114,6 +114,8 @@ def _declare(lib):
         "gpue_hash_agg_sum_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_u64, c_vp, c_vp, c_vp,
                                           c_u64, ctypes.POINTER(c_u64)]),
         "gpue_dbuf_d2d": (c_i32, [c_vp, c_vp, c_u64, c_u64, c_u64]),
+        "gpue_q3_decomp": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_vp, c_i32, c_i32,
+                                   c_vp, c_vp]),
         "gpue_sum_prod_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_join_build_varchar": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_probe_emit_varchar": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_vp, c_vp,
@@ -424,6 +426,11 @@ class Engine:
             self._h, lk._h, ext._h, disc._h, ship._h, n, order_bits._h, ship_cutoff,
             capacity_hint, out_keys._h, out_sums._h, max_out, ctypes.byref(g)))
         return g.value
+
+    def q3_decomp(self, lk, ext, disc, ship, n, order_bits, ship_cutoff, legs, at, sink):
+        _ck(self._lib, self._lib.gpue_q3_decomp(
+            self._h, lk._h, ext._h, disc._h, ship._h, n, order_bits._h, ship_cutoff, legs,
+            at, sink._h))
 
     def hash_agg_sum_u64(self, keys: DBuf, vals: DBuf, n, out_keys: DBuf, out_sums: DBuf,
                          out_counts: DBuf = None, max_out=0, capacity_hint=0):

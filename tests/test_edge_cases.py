@@ -1,0 +1,94 @@
+"""Edge cases the reference's own join/agg tests cover (SURVEY.md §8c:
+join_hash_map_test.cpp exercises empty tables, one-row tables, all-duplicate
+keys, and boundary key values): oracle-side here, GPU-side in
+test_edge_cases_gpu.py with the same constructions.
+"""
+
+import numpy as np
+
+from oracle import pyoracle as orc
+
+SEED = 42
+
+
+def test_single_row_build():
+    """One build row (bucket_size = NormalizeCapacity(2+...)+1 floor)."""
+    first, nxt, bs, log = orc.bucket_chained_build(np.array([0, 77], np.uint32))
+    heads = orc.bucket_chained_lookup(np.array([77, 78, 0], np.uint32), first, bs, log)
+    assert heads[0] == 1 and heads[1] == 0 or heads[1] != 1
+    op, ob = orc.probe_emit(np.array([0, 77], np.uint32), nxt,
+                            np.array([77, 78, 0], np.uint32), heads)
+    assert list(zip(op.tolist(), ob.tolist())) == [(0, 1)]
+
+
+def test_all_duplicate_build_keys_chain():
+    """100K identical keys -> one 100K-deep chain; every probe of that key
+    emits all 100K build rows (the reference's chain-walk resumability case)."""
+    n = 100_000
+    keys = np.concatenate([[0], np.full(n, 12345)]).astype(np.uint32)
+    first, nxt, bs, log = orc.bucket_chained_build(keys)
+    probe = np.array([12345, 999], np.uint32)
+    heads = orc.bucket_chained_lookup(probe, first, bs, log)
+    # emit count (n) far exceeds pyoracle.probe_emit's 8x-probe-rows cap
+    # heuristic, so size the out buffers explicitly
+    op = np.empty(n + 16, np.uint32)
+    ob = np.empty(n + 16, np.uint32)
+    m = orc.load().orc_probe_emit_u32(orc._p(keys), orc._p(nxt), orc._p(probe),
+                                      orc._p(heads), len(probe), 0, orc._p(op), orc._p(ob))
+    assert m == n
+    op, ob = op[:m], ob[:m]
+    assert (op == 0).all()
+    assert sorted(ob.tolist()) == list(range(1, n + 1))
+
+
+def test_extreme_key_values():
+    """Keys at the u32 boundaries hash and join correctly."""
+    bkeys = np.concatenate([[0], [0, 1, 0x7FFFFFFF, 0x80000000, 0xFFFFFFFF]]).astype(np.uint32)
+    first, nxt, bs, log = orc.bucket_chained_build(bkeys)
+    probe = np.array([0xFFFFFFFF, 0x80000000, 0x7FFFFFFF, 1, 0, 2], np.uint32)
+    heads = orc.bucket_chained_lookup(probe, first, bs, log)
+    op, ob = orc.probe_emit(bkeys, nxt, probe, heads)
+    got = sorted(zip(op.tolist(), ob.tolist()))
+    assert got == [(0, 5), (1, 4), (2, 3), (3, 2), (4, 1)]
+
+
+def test_empty_probe():
+    first, nxt, bs, log = orc.bucket_chained_build(np.array([0, 5, 6], np.uint32))
+    heads = orc.bucket_chained_lookup(np.empty(0, np.uint32), first, bs, log)
+    op, ob = orc.probe_emit(np.array([0, 5, 6], np.uint32), nxt,
+                            np.empty(0, np.uint32), heads)
+    assert len(op) == 0 and len(ob) == 0
+
+
+def test_varchar_empty_string_keys():
+    """Build rows holding EMPTY strings are real rows (distinct from the row-0
+    sentinel, which never enters a chain): an empty probe string matches
+    exactly the empty build rows."""
+    brows = [b"", b"", b"abc", b""]  # build rows 1,2,4 empty; row 3 = "abc"
+    rows = [b""] + brows  # row 0 sentinel
+    bo = np.zeros(len(rows) + 1, np.uint32)
+    np.cumsum([len(r) for r in rows], out=bo[1:])
+    bb = np.frombuffer(b"".join(rows), np.uint8).copy() if b"".join(rows) else np.zeros(1, np.uint8)
+    prows = [b"", b"abc", b"zzz"]
+    po = np.zeros(len(prows) + 1, np.uint32)
+    pb = np.frombuffer(b"".join(prows), np.uint8).copy()
+    np.cumsum([len(r) for r in prows], out=po[1:])
+    op, ob = orc.slice_join(bb, bo, len(brows), pb, po, len(prows), 1000)
+    got = sorted(zip(op.tolist(), ob.tolist()))
+    # probe 0 ("") matches build rows 1,2,4 (the empty rows); probe 1 matches row 3
+    assert got == [(0, 1), (0, 2), (0, 4), (1, 3)]
+
+
+def test_q1_no_passing_rows():
+    """A year outside the dim range: zero matches, zero sum."""
+    s, cnt = orc.q1_pipeline(SEED, 0, 100_000, 1888)
+    assert (s, cnt) == (0, 0)
+
+
+def test_hash_agg_single_group():
+    """All rows in one group (worst-case atomic contention shape)."""
+    n = 50_000
+    keys = np.full(n, 42, np.uint64)
+    vals = np.arange(n, dtype=np.int64)
+    ok, os_, oc = orc.hash_agg_sum(keys, vals)
+    assert len(ok) == 1 and ok[0] == 42 and os_[0] == vals.sum() and oc[0] == n

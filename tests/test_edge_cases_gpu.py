@@ -1,0 +1,129 @@
+"""GPU mirrors of tests/test_edge_cases.py (SURVEY.md §8c edge cases):
+same constructions through the C-ABI on the MI355X."""
+
+import numpy as np
+import pytest
+
+from oracle import pyoracle as orc
+
+pytestmark = pytest.mark.gpu
+
+SEED = 42
+
+
+def _emit(engine, t, probe, expect_pairs):
+    pb = engine.alloc(max(probe.nbytes, 4))
+    if len(probe):
+        pb.h2d(probe)
+    cnt = engine.join_probe_emit(t, pb, len(probe))
+    assert cnt == len(expect_pairs)
+    if cnt:
+        op_buf, ob_buf = engine.alloc(cnt * 4), engine.alloc(cnt * 4)
+        engine.join_probe_emit(t, pb, len(probe), op_buf, ob_buf)
+        got = sorted(zip(op_buf.d2h(np.uint32, cnt).tolist(),
+                         ob_buf.d2h(np.uint32, cnt).tolist()))
+        assert got == sorted(expect_pairs)
+        op_buf.free()
+        ob_buf.free()
+    pb.free()
+
+
+def test_single_row_build_gpu(engine):
+    keys = np.array([0, 77], np.uint32)
+    kb = engine.alloc(keys.nbytes)
+    kb.h2d(keys)
+    t = engine.join_build_bucket_chained(kb, 1)
+    _emit(engine, t, np.array([77, 78, 0], np.uint32), [(0, 1)])
+    _emit(engine, t, np.empty(0, np.uint32), [])
+    kb.free()
+    t.destroy()
+
+
+def test_all_duplicate_build_keys_chain_gpu(engine):
+    n = 100_000
+    keys = np.concatenate([[0], np.full(n, 12345)]).astype(np.uint32)
+    kb = engine.alloc(keys.nbytes)
+    kb.h2d(keys)
+    t = engine.join_build_bucket_chained(kb, n)
+    _emit(engine, t, np.array([12345, 999], np.uint32),
+          [(0, j) for j in range(1, n + 1)])
+    kb.free()
+    t.destroy()
+
+
+def test_extreme_key_values_gpu(engine):
+    bkeys = np.concatenate([[0], [0, 1, 0x7FFFFFFF, 0x80000000, 0xFFFFFFFF]]).astype(np.uint32)
+    kb = engine.alloc(bkeys.nbytes)
+    kb.h2d(bkeys)
+    t = engine.join_build_bucket_chained(kb, 5)
+    _emit(engine, t, np.array([0xFFFFFFFF, 0x80000000, 0x7FFFFFFF, 1, 0, 2], np.uint32),
+          [(0, 5), (1, 4), (2, 3), (3, 2), (4, 1)])
+    kb.free()
+    t.destroy()
+
+
+def test_varchar_empty_string_keys_gpu(engine):
+    brows = [b"", b"", b"abc", b""]
+    rows = [b""] + brows
+    bo = np.zeros(len(rows) + 1, np.uint32)
+    np.cumsum([len(r) for r in rows], out=bo[1:])
+    bb = np.frombuffer(b"".join(rows), np.uint8).copy()
+    prows = [b"", b"abc", b"zzz"]
+    po = np.zeros(len(prows) + 1, np.uint32)
+    np.cumsum([len(r) for r in prows], out=po[1:])
+    pb = np.frombuffer(b"".join(prows), np.uint8).copy()
+    d_bb, d_bo = engine.alloc(max(bb.nbytes, 4)), engine.alloc(bo.nbytes)
+    d_bb.h2d(bb)
+    d_bo.h2d(bo)
+    t = engine.join_build_varchar(d_bb, d_bo, len(brows))
+    d_pb, d_po = engine.alloc(pb.nbytes), engine.alloc(po.nbytes)
+    d_pb.h2d(pb)
+    d_po.h2d(po)
+    cnt = engine.join_probe_emit_varchar(t, d_pb, d_po, len(prows))
+    assert cnt == 4
+    op_buf, ob_buf = engine.alloc(cnt * 4), engine.alloc(cnt * 4)
+    engine.join_probe_emit_varchar(t, d_pb, d_po, len(prows), op_buf, ob_buf)
+    got = sorted(zip(op_buf.d2h(np.uint32, cnt).tolist(),
+                     ob_buf.d2h(np.uint32, cnt).tolist()))
+    assert got == [(0, 1), (0, 2), (0, 4), (1, 3)]
+    for b in (d_bb, d_bo, d_pb, d_po, op_buf, ob_buf):
+        b.free()
+    t.destroy()
+
+
+def test_q1_no_passing_rows_gpu(engine):
+    """Payload table whose payloads are ALL zero (year matches no dim row):
+    the fused kernel returns (0, 0) like the oracle."""
+    n = 1_000_000
+    cols = [engine.alloc(n * 4) for _ in range(3)]
+    engine.gen_lineorder_q1(SEED, 0, n, *cols)
+    from starrocks_amd import gen
+    datekey, dyear = gen.gen_dates()
+    dpay = np.zeros(len(datekey), np.uint32)
+    kb = engine.alloc(datekey.nbytes)
+    kb.h2d(datekey.astype(np.int32))
+    pbuf = engine.alloc(dpay.nbytes)
+    pbuf.h2d(dpay)
+    t = engine.join_build_payload(kb, pbuf, len(datekey))
+    s, cnt = engine.q1_join_sum(t, *cols, n)
+    assert (s, cnt) == (0, 0)
+    for b in cols + [kb, pbuf]:
+        b.free()
+    t.destroy()
+
+
+def test_hash_agg_single_group_gpu(engine):
+    n = 500_000
+    keys = np.full(n, 42, np.uint64)
+    vals = np.arange(n, dtype=np.int64)
+    kb, vb = engine.alloc(n * 8), engine.alloc(n * 8)
+    kb.h2d(keys)
+    vb.h2d(vals)
+    ok_b, os_b, oc_b = (engine.alloc(16 * 8) for _ in range(3))
+    g = engine.hash_agg_sum_u64(kb, vb, n, ok_b, os_b, oc_b, max_out=16)
+    assert g == 1
+    assert ok_b.d2h(np.uint64, 1)[0] == 42
+    assert os_b.d2h(np.int64, 1)[0] == vals.sum()
+    assert oc_b.d2h(np.int64, 1)[0] == n
+    for b in (kb, vb, ok_b, os_b, oc_b):
+        b.free()
