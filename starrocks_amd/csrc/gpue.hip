@@ -2916,6 +2916,8 @@ int gpue_agg_table_create(gpue_session* s, uint64_t capacity, gpue_agg_table** o
     return GPUE_OK;
 }
 
+extern "C" int gpue_agg_table_reset(gpue_agg_table* t);
+
 void gpue_agg_table_destroy(gpue_agg_table* t) {
     if (!t) return;
     (void)hipFree(t->slots);
@@ -2931,6 +2933,11 @@ static int agg_table_reset(gpue_agg_table* t, bool with_counts = true) {
     if (with_counts) HIP_CHECK(hipMemsetAsync(t->counts, 0, t->cap * 8, t->s->stream));
     HIP_CHECK(hipMemsetAsync(t->cursor, 0, 8, t->s->stream));
     return GPUE_OK;
+}
+
+int gpue_agg_table_reset(gpue_agg_table* t) {
+    ARG_CHECK(t);
+    return agg_table_reset(t);
 }
 
 __global__ void k_hash_agg_sum(const uint64_t* __restrict__ keys,
@@ -3468,7 +3475,7 @@ __global__ void k_q3_legs(const int64_t* __restrict__ lk,
         acc += v & 1;
         if (!(legs & 8)) continue;
         uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
-        for (;;) {
+        for (uint64_t tries = 0; tries <= cap_mask; tries++) {  // bounded: full table => stop, not hang
             unsigned long long cur = slots[s];
             if (cur == k) { atomicAdd(&sums[s], v); break; }
             if (cur == AGG_EMPTY) {

@@ -103,6 +103,7 @@ def _declare(lib):
         "gpue_q3_order_bits": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp, c_i32, c_vp]),
         "gpue_agg_table_create": (c_i32, [c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_agg_table_destroy": (None, [c_vp]),
+        "gpue_agg_table_reset": (c_i32, [c_vp]),
         "gpue_q3_probe_agg_t": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_vp, c_i32, c_vp,
                                        c_vp, c_vp, c_u64, ctypes.POINTER(c_u64)]),
         "gpue_q3_probe_agg": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_vp, c_i32, c_u64,
@@ -407,6 +408,9 @@ class Engine:
         h = c_vp()
         _ck(self._lib, self._lib.gpue_agg_table_create(self._h, capacity, ctypes.byref(h)))
         return h
+
+    def agg_table_reset(self, h):
+        _ck(self._lib, self._lib.gpue_agg_table_reset(h))
 
     def agg_table_destroy(self, h):
         self._lib.gpue_agg_table_destroy(h)
