@@ -3430,11 +3430,16 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
     (void)counts;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
-        if (ship[i] <= ship_cutoff) continue;
-        unsigned long long k = (unsigned long long)lk[i];
+        // non-temporal loads on the four streamed columns: the decomp
+        // instrument (tools/q3_decomp.py) shows the order-bits
+        // random gather dominates (2.1 of 3.3 ms); streaming around L2 buys a
+        // small (~1.4%) but consistent win by leaving L2 to the bitset.
+        if (__builtin_nontemporal_load(ship + i) <= ship_cutoff) continue;
+        unsigned long long k = (unsigned long long)__builtin_nontemporal_load(lk + i);
         uint64_t o = k - 1;
         if (!((order_bits[o >> 5] >> (o & 31)) & 1u)) continue;
-        unsigned long long v = (unsigned long long)(ext[i] * (100 - disc[i]));
+        unsigned long long v = (unsigned long long)(__builtin_nontemporal_load(ext + i) *
+                                                    (100 - __builtin_nontemporal_load(disc + i)));
         uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
         for (;;) {
             unsigned long long cur = slots[s];
@@ -3452,6 +3457,7 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
 // legs bitmask accumulates per-leg cost — 1: ship stream, 2: +lk read &
 // order-bits gather, 4: +ext/disc product, 8: +hash-table insert (the full
 // kernel). Local accumulate + one atomic per block prevents DCE.
+template <bool NT>  // NT: non-temporal stream loads — leave L2 to the bitset
 __global__ void k_q3_legs(const int64_t* __restrict__ lk,
                           const int64_t* __restrict__ ext,
                           const int64_t* __restrict__ disc,
@@ -3463,15 +3469,19 @@ __global__ void k_q3_legs(const int64_t* __restrict__ lk,
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     unsigned long long acc = 0;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
-        if (ship[i] <= ship_cutoff) continue;
+        int32_t sh = NT ? __builtin_nontemporal_load(ship + i) : ship[i];
+        if (sh <= ship_cutoff) continue;
         acc++;
         if (!(legs & 2)) continue;
-        unsigned long long k = (unsigned long long)lk[i];
+        unsigned long long k =
+            (unsigned long long)(NT ? __builtin_nontemporal_load(lk + i) : lk[i]);
         uint64_t o = k - 1;
         if (!((order_bits[o >> 5] >> (o & 31)) & 1u)) continue;
         acc++;
         if (!(legs & 4)) continue;
-        unsigned long long v = (unsigned long long)(ext[i] * (100 - disc[i]));
+        int64_t e_ = NT ? __builtin_nontemporal_load(ext + i) : ext[i];
+        int64_t d_ = NT ? __builtin_nontemporal_load(disc + i) : disc[i];
+        unsigned long long v = (unsigned long long)(e_ * (100 - d_));
         acc += v & 1;
         if (!(legs & 8)) continue;
         uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
@@ -3515,7 +3525,8 @@ int gpue_q3_decomp(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* di
                    int legs, gpue_agg_table* at, gpue_dbuf* sink) {
     ARG_CHECK(s && lk && ext && disc && ship && order_bits && sink && sink->bytes >= 8);
     ARG_CHECK(!(legs & 8) || at);
-    hipLaunchKernelGGL(k_q3_legs, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))),
+    hipLaunchKernelGGL((legs & 16) ? k_q3_legs<true> : k_q3_legs<false>,
+                       dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))),
                        dim3(BLOCK), 0, s->stream, (const int64_t*)lk->ptr,
                        (const int64_t*)ext->ptr, (const int64_t*)disc->ptr,
                        (const int32_t*)ship->ptr, n, (const uint32_t*)order_bits->ptr,

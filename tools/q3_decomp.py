@@ -60,8 +60,9 @@ def main():
 
     lines = []
     prev = None
-    names = {1: "ship stream", 3: "+lk & obits gather", 7: "+ext*disc", 15: "+hash insert"}
-    for legs in (1, 3, 7, 15):
+    names = {1: "ship stream", 3: "+lk & obits gather", 7: "+ext*disc", 15: "+hash insert",
+             17: "NT ship stream", 19: "NT +lk & obits", 23: "NT +ext*disc", 31: "NT full"}
+    for legs in (1, 3, 7, 15, 17, 19, 23, 31):
         for _ in range(WARMUP):
             if legs & 8:
                 e.agg_table_reset(at)
@@ -74,6 +75,8 @@ def main():
                 e.agg_table_reset(at)
             e.q3_decomp(lk, ext, disc, ship, N, obits, CUTOFF, legs, at, sink)
         ms = e.timer_stop() / STEPS
+        if legs == 17:
+            prev = None
         delta = "" if prev is None else f"  (+{ms - prev:.3f} ms)"
         lines.append(f"legs={legs:2d} {names[legs]:<20s} {ms:8.3f} ms/pass{delta}")
         prev = ms
