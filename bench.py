@@ -421,6 +421,16 @@ def main():
     eng.sync()
     log(f"[bench] setup {time.perf_counter()-t_setup:.1f}s")
 
+    # hipGraph step replay (include/gpue.h graph API): capture the step's
+    # async launch sequence once, replay per step — removes per-launch host
+    # overhead (~14 us/step, i.e. ~9% of an SF10-sized step). q3 is excluded:
+    # its step does a device->host count readback inside q3_probe_agg_t.
+    if wl != "q3" and os.environ.get("GPUE_NO_GRAPH") != "1":
+        _graph = eng.graph_capture(kernel_only)
+
+        def kernel_only():
+            eng.graph_launch(_graph)
+
     def run_step():
         if dist is not None:
             if wl == "q3":

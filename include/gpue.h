@@ -58,6 +58,18 @@ int gpue_dbuf_d2d(gpue_dbuf* src, gpue_dbuf* dst, uint64_t bytes, uint64_t src_o
  * with sum<int64> states, reference be/src/exprs/agg/sum.h:45-181). */
 int gpue_sum_prod_u32(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b, uint64_t n,
                       gpue_dbuf* acc);
+
+/* ---- hipGraph step replay ----
+ * Capture the async launch sequence of one step between begin/end, then
+ * replay it with a single hipGraphLaunch per step. Removes the per-launch
+ * host cost that dominates small (SF10-sized) steps; the CDNA-native analog
+ * of the reference driver's time-slice batching. Only stream-async entries
+ * (the *_async pipelines, memsets) may sit inside a capture. */
+typedef struct gpue_graph gpue_graph;
+int gpue_graph_begin(gpue_session* s);
+int gpue_graph_end(gpue_session* s, gpue_graph** out);
+int gpue_graph_launch(gpue_session* s, gpue_graph* g);
+void gpue_graph_destroy(gpue_graph* g);
 /* Wrap external device memory (e.g. a torch tensor's data_ptr) so kernels
  * operate in place and torch.distributed (RCCL) moves the same buffers —
  * the all-to-all leg of configs 4-5. Caller keeps ownership. */

@@ -306,6 +306,65 @@ int gpue_sum_prod_u32(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b, uint64_t n,
 }
 
 // ---------------------------------------------------------------------------
+// hipGraph step replay: capture a step's launch sequence once, replay it per
+// step with a single hipGraphLaunch — removes the per-launch host cost that
+// dominates small (SF10-sized) steps. The CDNA-native analog of the
+// reference's driver time-slice batching.
+// ---------------------------------------------------------------------------
+struct gpue_graph {
+    hipGraph_t graph = nullptr;
+    hipGraphExec_t exec = nullptr;
+};
+
+extern "C" {
+int gpue_graph_begin(gpue_session* s);
+int gpue_graph_end(gpue_session* s, gpue_graph** out);
+int gpue_graph_launch(gpue_session* s, gpue_graph* g);
+void gpue_graph_destroy(gpue_graph* g);
+}
+
+int gpue_graph_begin(gpue_session* s) {
+    ARG_CHECK(s);
+    HIP_CHECK(hipStreamBeginCapture(s->stream, hipStreamCaptureModeThreadLocal));
+    return GPUE_OK;
+}
+
+int gpue_graph_end(gpue_session* s, gpue_graph** out) {
+    ARG_CHECK(s && out);
+    gpue_graph* g = new gpue_graph();
+    hipError_t err = hipStreamEndCapture(s->stream, &g->graph);
+    if (err != hipSuccess) {
+        delete g;
+        snprintf(g_err, sizeof(g_err), "hipStreamEndCapture failed: %s",
+                 hipGetErrorString(err));
+        return GPUE_ERR_HIP;
+    }
+    err = hipGraphInstantiate(&g->exec, g->graph, nullptr, nullptr, 0);
+    if (err != hipSuccess) {
+        (void)hipGraphDestroy(g->graph);
+        delete g;
+        snprintf(g_err, sizeof(g_err), "hipGraphInstantiate failed: %s",
+                 hipGetErrorString(err));
+        return GPUE_ERR_HIP;
+    }
+    *out = g;
+    return GPUE_OK;
+}
+
+int gpue_graph_launch(gpue_session* s, gpue_graph* g) {
+    ARG_CHECK(s && g && g->exec);
+    HIP_CHECK(hipGraphLaunch(g->exec, s->stream));
+    return GPUE_OK;
+}
+
+void gpue_graph_destroy(gpue_graph* g) {
+    if (!g) return;
+    if (g->exec) (void)hipGraphExecDestroy(g->exec);
+    if (g->graph) (void)hipGraphDestroy(g->graph);
+    delete g;
+}
+
+// ---------------------------------------------------------------------------
 // deterministic synthetic generator — splitmix64 finalizer, counter-based.
 // MUST stay identical to oracle/oracle.c sm64/orc_gen_u64 and the numpy
 // restatement in starrocks_amd/gen.py.

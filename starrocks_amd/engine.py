@@ -118,6 +118,10 @@ def _declare(lib):
         "gpue_q3_decomp": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_vp, c_i32, c_i32,
                                    c_vp, c_vp]),
         "gpue_sum_prod_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
+        "gpue_graph_begin": (c_i32, [c_vp]),
+        "gpue_graph_end": (c_i32, [c_vp, ctypes.POINTER(c_vp)]),
+        "gpue_graph_launch": (c_i32, [c_vp, c_vp]),
+        "gpue_graph_destroy": (None, [c_vp]),
         "gpue_join_build_varchar": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_probe_emit_varchar": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_vp, c_vp,
                                                  ctypes.POINTER(c_u64)]),
@@ -460,6 +464,28 @@ class Engine:
             self._h, keys._h, vals._h, n, capacity_hint, out_keys._h, out_lo._h,
             out_hi._h, max_out, ctypes.byref(g)))
         return g.value
+
+    def graph_capture(self, fn):
+        """Capture fn()'s async launches as a hipGraph; returns the exec
+        handle for graph_launch. fn must only enqueue stream-async work."""
+        _ck(self._lib, self._lib.gpue_graph_begin(self._h))
+        try:
+            fn()
+        except BaseException:
+            h = c_vp()
+            self._lib.gpue_graph_end(self._h, ctypes.byref(h))  # abort capture
+            if h:
+                self._lib.gpue_graph_destroy(h)
+            raise
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_graph_end(self._h, ctypes.byref(h)))
+        return h
+
+    def graph_launch(self, g):
+        _ck(self._lib, self._lib.gpue_graph_launch(self._h, g))
+
+    def graph_destroy(self, g):
+        self._lib.gpue_graph_destroy(g)
 
     def dbuf_d2d(self, src: DBuf, dst: DBuf, nbytes, src_off=0, dst_off=0):
         _ck(self._lib, self._lib.gpue_dbuf_d2d(src._h, dst._h, nbytes, src_off, dst_off))

@@ -793,3 +793,34 @@ def test_q1_operator_pipeline_parity(engine):
     got = q1_operator_pipeline(engine, SEED, n, chunk_rows=700_000)
     expect = orc.q1_pipeline(SEED, 0, n, 1993)
     assert got == expect
+
+
+def test_graph_replay_q1(engine):
+    """hipGraph capture/replay of the q1 step (include/gpue.h graph API):
+    two replays produce the same bit-exact (sum, count) as the direct call
+    and the oracle."""
+    n = 2_000_000
+    cols = [engine.alloc(n * 4) for _ in range(3)]
+    engine.gen_lineorder_q1(SEED, 0, n, *cols)
+    from starrocks_amd import gen
+    datekey, dyear = gen.gen_dates()
+    dpay = np.where(dyear == 1993, dyear - 1992 + 1, 0).astype(np.uint32)
+    kb = engine.alloc(datekey.nbytes)
+    kb.h2d(datekey.astype(np.int32))
+    pb = engine.alloc(dpay.nbytes)
+    pb.h2d(dpay)
+    t = engine.join_build_payload(kb, pb, len(datekey))
+    acc = engine.alloc(16)
+    g = engine.graph_capture(
+        lambda: engine.q1_join_sum_async(t, cols[0], cols[1], cols[2], n, acc))
+    expect = orc.q1_pipeline(SEED, 0, n, 1993)
+    for _ in range(2):
+        engine.graph_launch(g)
+        got = acc.d2h(np.int64, 2)
+        assert (int(got[0]), int(got[1])) == expect
+    engine.graph_destroy(g)
+    direct = engine.q1_join_sum(t, *cols, n)
+    assert direct == expect
+    for b in cols + [kb, pb, acc]:
+        b.free()
+    t.destroy()
