@@ -421,11 +421,13 @@ def main():
     eng.sync()
     log(f"[bench] setup {time.perf_counter()-t_setup:.1f}s")
 
-    # hipGraph step replay (include/gpue.h graph API): capture the step's
-    # async launch sequence once, replay per step — removes per-launch host
-    # overhead (~14 us/step, i.e. ~9% of an SF10-sized step). q3 is excluded:
-    # its step does a device->host count readback inside q3_probe_agg_t.
-    if wl != "q3" and os.environ.get("GPUE_NO_GRAPH") != "1":
+    # hipGraph step replay (include/gpue.h graph API) — measured NEGATIVE for
+    # these 2-node steps (memset+kernel): 0.160 vs 0.154 ms/step on the SF10
+    # line; hipGraphLaunch costs more than the two direct async launches it
+    # replaces. Kept opt-in (GPUE_GRAPH=1) for multi-kernel
+    # steps where the node count amortizes it. q3 is excluded: its step does
+    # a device->host count readback inside q3_probe_agg_t.
+    if wl != "q3" and os.environ.get("GPUE_GRAPH") == "1":
         _graph = eng.graph_capture(kernel_only)
 
         def kernel_only():
