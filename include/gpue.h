@@ -186,6 +186,12 @@ int gpue_join_probe_emit_varchar(gpue_session* s, gpue_join_table* t, gpue_dbuf*
                                  gpue_dbuf* poffsets, uint64_t n_rows,
                                  gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
                                  uint64_t* match_count);
+/* Per-join-type Slice probe: mode 0 INNER, 1 LEFT_SEMI, 2 LEFT_ANTI,
+ * 3 LEFT_OUTER (join_hash_map.h:228-333 semantics as for the i32 paths). */
+int gpue_join_probe_emit_varchar_mode(gpue_session* s, gpue_join_table* t, gpue_dbuf* pbytes,
+                                      gpue_dbuf* poffsets, uint64_t n_rows, int mode,
+                                      gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                                      uint64_t* match_count);
 /* RIGHT SEMI (anti=0) / RIGHT ANTI (anti=1): matched/unmatched BUILD rows. */
 int gpue_join_probe_right_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
                               uint64_t n_rows, int anti, gpue_dbuf* out_build_idx,

@@ -219,6 +219,43 @@ uint64_t orc_slice_probe_emit(const uint8_t* bbytes, const uint32_t* boffsets,
     return m;
 }
 
+/* Slice-key per-join-type probe (join_hash_map.h:228-333 semantics over the
+ * crc-hashed chained table above): mode 0 INNER, 1 LEFT_SEMI (first match
+ * only), 2 LEFT_ANTI (unmatched probe rows, build 0 sentinel), 3 LEFT_OUTER. */
+uint64_t orc_slice_probe_emit_mode(const uint8_t* bbytes, const uint32_t* boffsets,
+                                   const uint32_t* next, uint32_t bucket_size,
+                                   const uint32_t* first, const uint8_t* pbytes,
+                                   const uint32_t* poffsets, uint32_t probe_rows, int mode,
+                                   uint32_t* out_probe_idx, uint32_t* out_build_idx) {
+    uint64_t m = 0;
+    for (uint32_t i = 0; i < probe_rows; i++) {
+        uint32_t len = poffsets[i + 1] - poffsets[i];
+        uint32_t b = orc_crc_hash_32(pbytes + poffsets[i], (int32_t)len, 0x811C9DC5u) &
+                     (bucket_size - 1);
+        uint32_t j = first[b];
+        uint32_t c = 0;
+        while (j != 0) {
+            uint32_t blen = boffsets[j + 1] - boffsets[j];
+            if (blen == len && memcmp(bbytes + boffsets[j], pbytes + poffsets[i], len) == 0) {
+                if (mode == 0 || mode == 3 || (mode == 1 && c == 0)) {
+                    out_probe_idx[m] = i;
+                    out_build_idx[m] = j;
+                    m++;
+                }
+                c++;
+                if (mode == 1 || mode == 2) break;
+            }
+            j = next[j];
+        }
+        if (c == 0 && (mode == 2 || mode == 3)) {
+            out_probe_idx[m] = i;
+            out_build_idx[m] = 0;
+            m++;
+        }
+    }
+    return m;
+}
+
 /* Nullable variants (construct_hash_table / lookup_init is_nulls paths,
  * join_hash_map_method.hpp:56-85,101-120): null build rows are skipped
  * (next=0 — the row never enters a chain); null probe rows get chain head 0

@@ -50,6 +50,11 @@ void orc_bucket_chained_lookup_u32(const uint32_t* probe_keys, uint32_t probe_ro
 void orc_slice_build_u32(const uint8_t* bytes, const uint32_t* offsets, uint32_t row_count,
                          uint32_t* first, uint32_t* next, uint32_t bucket_size,
                          uint32_t log_bucket_size);
+uint64_t orc_slice_probe_emit_mode(const uint8_t* bbytes, const uint32_t* boffsets,
+                                   const uint32_t* next, uint32_t bucket_size,
+                                   const uint32_t* first, const uint8_t* pbytes,
+                                   const uint32_t* poffsets, uint32_t probe_rows, int mode,
+                                   uint32_t* out_probe_idx, uint32_t* out_build_idx);
 uint64_t orc_slice_probe_emit(const uint8_t* bbytes, const uint32_t* boffsets,
                               const uint32_t* next, uint32_t bucket_size,
                               const uint32_t* first, const uint8_t* pbytes,

@@ -74,6 +74,9 @@ def _decl(lib):
     lib.orc_slice_build_u32.argtypes = [c_vp, c_vp, u, c_vp, c_vp, u, u]
     lib.orc_slice_probe_emit.restype = c_u64
     lib.orc_slice_probe_emit.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, c_vp, u, c_vp, c_vp]
+    lib.orc_slice_probe_emit_mode.restype = c_u64
+    lib.orc_slice_probe_emit_mode.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, c_vp, u,
+                                              c_i32, c_vp, c_vp]
     lib.orc_probe_emit_u32.restype = c_u64
     lib.orc_probe_emit_u32.argtypes = [c_vp, c_vp, c_vp, c_vp, u, c_i32, c_vp, c_vp]
     lib.orc_filter_i64_lt.restype = c_u64
@@ -148,6 +151,24 @@ def slice_join(bbytes, boffsets, row_count, pbytes, poffsets, probe_rows, max_ou
     ob = np.empty(max_out, np.uint32)
     m = lib.orc_slice_probe_emit(_p(bbytes), _p(boffsets), _p(nxt), bucket_size, _p(first),
                                  _p(pbytes), _p(poffsets), probe_rows, _p(op), _p(ob))
+    return op[:m], ob[:m]
+
+
+def slice_join_mode(bbytes, boffsets, row_count, pbytes, poffsets, probe_rows, mode,
+                    max_out):
+    """Per-join-type Slice probe: mode 0 INNER, 1 SEMI, 2 ANTI, 3 OUTER."""
+    lib = load()
+    bucket_size = lib.orc_calc_bucket_size(row_count + 1)
+    log = int(bucket_size).bit_length() - 1
+    first = np.zeros(bucket_size, np.uint32)
+    nxt = np.zeros(row_count + 1, np.uint32)
+    lib.orc_slice_build_u32(_p(bbytes), _p(boffsets), row_count, _p(first), _p(nxt),
+                            bucket_size, log)
+    op = np.empty(max_out, np.uint32)
+    ob = np.empty(max_out, np.uint32)
+    m = lib.orc_slice_probe_emit_mode(_p(bbytes), _p(boffsets), _p(nxt), bucket_size,
+                                      _p(first), _p(pbytes), _p(poffsets), probe_rows,
+                                      mode, _p(op), _p(ob))
     return op[:m], ob[:m]
 
 

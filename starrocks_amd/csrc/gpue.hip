@@ -1587,7 +1587,7 @@ __global__ void k_probe_count_vc(const uint8_t* __restrict__ pbytes,
                                  uint32_t bucket_mask, const uint32_t* __restrict__ first,
                                  const uint32_t* __restrict__ next,
                                  const uint8_t* __restrict__ bbytes,
-                                 const uint32_t* __restrict__ boffsets,
+                                 const uint32_t* __restrict__ boffsets, int mode,
                                  uint32_t* __restrict__ row_counts) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
@@ -1600,7 +1600,7 @@ __global__ void k_probe_count_vc(const uint8_t* __restrict__ pbytes,
                           pbytes + poffsets[i], len);
             j = next[j];
         }
-        row_counts[i] = c;
+        row_counts[i] = join_mode_count(c, mode);
     }
 }
 
@@ -1610,7 +1610,7 @@ __global__ void k_probe_emit_vc(const uint8_t* __restrict__ pbytes,
                                 const uint32_t* __restrict__ next,
                                 const uint8_t* __restrict__ bbytes,
                                 const uint32_t* __restrict__ boffsets,
-                                const uint32_t* __restrict__ row_counts,
+                                int mode, const uint32_t* __restrict__ row_counts,
                                 const uint64_t* __restrict__ row_offsets,
                                 uint32_t* __restrict__ out_probe,
                                 uint32_t* __restrict__ out_build) {
@@ -1621,14 +1621,23 @@ __global__ void k_probe_emit_vc(const uint8_t* __restrict__ pbytes,
         uint32_t len = poffsets[i + 1] - poffsets[i];
         uint32_t b = crc_hash_32_dev(pbytes + poffsets[i], len, 0x811C9DC5u) & bucket_mask;
         uint32_t j = first[b];
+        uint32_t c = 0;
         while (j != 0) {
             if (slice_eq(bbytes + boffsets[j], boffsets[j + 1] - boffsets[j],
                          pbytes + poffsets[i], len)) {
-                out_probe[pos] = (uint32_t)i;
-                out_build[pos] = j;
-                pos++;
+                if (mode == 0 || mode == 3 || (mode == 1 && c == 0)) {
+                    out_probe[pos] = (uint32_t)i;
+                    out_build[pos] = j;
+                    pos++;
+                }
+                c++;
+                if (mode == 1 || mode == 2) break;
             }
             j = next[j];
+        }
+        if (c == 0 && (mode == 2 || mode == 3)) { // unmatched emit, build 0 = NULL
+            out_probe[pos] = (uint32_t)i;
+            out_build[pos] = 0;
         }
     }
 }
@@ -1640,6 +1649,10 @@ int gpue_join_probe_emit_varchar(gpue_session* s, gpue_join_table* t, gpue_dbuf*
                                  gpue_dbuf* poffsets, uint64_t n_rows,
                                  gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
                                  uint64_t* match_count);
+int gpue_join_probe_emit_varchar_mode(gpue_session* s, gpue_join_table* t, gpue_dbuf* pbytes,
+                                      gpue_dbuf* poffsets, uint64_t n_rows, int mode,
+                                      gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                                      uint64_t* match_count);
 }
 
 int gpue_join_build_varchar(gpue_session* s, gpue_dbuf* bytes, gpue_dbuf* offsets,
@@ -1674,7 +1687,16 @@ int gpue_join_probe_emit_varchar(gpue_session* s, gpue_join_table* t, gpue_dbuf*
                                  gpue_dbuf* poffsets, uint64_t n_rows,
                                  gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
                                  uint64_t* match_count) {
+    return gpue_join_probe_emit_varchar_mode(s, t, pbytes, poffsets, n_rows, 0,
+                                             out_probe_idx, out_build_idx, match_count);
+}
+
+int gpue_join_probe_emit_varchar_mode(gpue_session* s, gpue_join_table* t, gpue_dbuf* pbytes,
+                                      gpue_dbuf* poffsets, uint64_t n_rows, int mode,
+                                      gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                                      uint64_t* match_count) {
     ARG_CHECK(s && t && pbytes && poffsets && match_count);
+    ARG_CHECK(mode >= 0 && mode <= 3);
     ARG_CHECK(t->kind == gpue_join_table::VARCHAR);
     ARG_CHECK(poffsets->bytes >= (n_rows + 1) * 4);
     uint32_t nb = grid_for(n_rows);
@@ -1687,7 +1709,7 @@ int gpue_join_probe_emit_varchar(gpue_session* s, gpue_join_table* t, gpue_dbuf*
     hipLaunchKernelGGL(k_probe_count_vc, dim3(nb), dim3(BLOCK), 0, s->stream,
                        (const uint8_t*)pbytes->ptr, (const uint32_t*)poffsets->ptr, n_rows,
                        (uint32_t)(t->bucket_size - 1), t->first, t->next, t->key_bytes,
-                       t->key_offsets, d_counts);
+                       t->key_offsets, mode, d_counts);
     hipLaunchKernelGGL(k_block_sums_u32, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts,
                        n_rows, tile, d_bsums);
     hipLaunchKernelGGL(k_scan_small, dim3(1), dim3(1), 0, s->stream, d_bsums, nb);
@@ -1703,7 +1725,7 @@ int gpue_join_probe_emit_varchar(gpue_session* s, gpue_join_table* t, gpue_dbuf*
         hipLaunchKernelGGL(k_probe_emit_vc, dim3(nb), dim3(BLOCK), 0, s->stream,
                            (const uint8_t*)pbytes->ptr, (const uint32_t*)poffsets->ptr,
                            n_rows, (uint32_t)(t->bucket_size - 1), t->first, t->next,
-                           t->key_bytes, t->key_offsets, d_counts, d_offsets,
+                           t->key_bytes, t->key_offsets, mode, d_counts, d_offsets,
                            (uint32_t*)out_probe_idx->ptr, (uint32_t*)out_build_idx->ptr);
         HIP_CHECK(hipStreamSynchronize(s->stream));
         (void)hipFree(d_offsets);

@@ -125,6 +125,8 @@ def _declare(lib):
         "gpue_join_build_varchar": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_probe_emit_varchar": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_vp, c_vp,
                                                  ctypes.POINTER(c_u64)]),
+        "gpue_join_probe_emit_varchar_mode": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_i32,
+                                                      c_vp, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_topk_i64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp]),
         "gpue_page_decode_bshuf_lz4_i32": (c_i32, [c_vp, c_vp, c_u32, c_vp]),
         "gpue_timer_start": (c_i32, [c_vp]),
@@ -322,6 +324,16 @@ class Engine:
         ob = out_build._h if out_build else None
         _ck(self._lib, self._lib.gpue_join_probe_emit_varchar(
             self._h, table._h, pbytes._h, poffsets._h, n_rows, op, ob, ctypes.byref(cnt)))
+        return cnt.value
+
+    def join_probe_emit_varchar_mode(self, table, pbytes: DBuf, poffsets: DBuf, n_rows,
+                                     mode, out_probe=None, out_build=None) -> int:
+        cnt = c_u64()
+        op = out_probe._h if out_probe else None
+        ob = out_build._h if out_build else None
+        _ck(self._lib, self._lib.gpue_join_probe_emit_varchar_mode(
+            self._h, table._h, pbytes._h, poffsets._h, n_rows, mode, op, ob,
+            ctypes.byref(cnt)))
         return cnt.value
 
     def pack_keys_2xi32(self, a: DBuf, b: DBuf, n, out: DBuf):
