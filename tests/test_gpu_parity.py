@@ -862,3 +862,25 @@ def test_varchar_join_modes_parity(engine):
     for b in (d_bb, d_bo, d_pb, d_po):
         b.free()
     t.destroy()
+
+
+def test_cabi_cpp_demo(engine):
+    """The standalone C++ caller (examples/cabi_q1.cpp, linked against
+    libgpue.so with no Python in the loop) produces the oracle's exact
+    (sum, count) for the config-2 plan."""
+    import json
+    import os
+    import subprocess
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    exe = os.path.join(repo, "examples", "cabi_q1")
+    if not os.path.exists(exe):
+        subprocess.run(["g++", "-O2", f"-I{os.path.join(repo, 'include')}",
+                        os.path.join(repo, "examples", "cabi_q1.cpp"),
+                        f"-L{os.path.join(repo, 'starrocks_amd')}", "-lgpue",
+                        "-Wl,-rpath,$ORIGIN/../starrocks_amd", "-o", exe], check=True)
+    n = 1_500_000
+    out = subprocess.run([exe, str(n)], capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr
+    got = json.loads(out.stdout.strip())
+    es, ecnt = orc.q1_pipeline(42, 0, n, 1993)
+    assert (got["sum"], got["count"]) == (es, ecnt)
