@@ -192,6 +192,20 @@ int gpue_join_probe_emit_varchar_mode(gpue_session* s, gpue_join_table* t, gpue_
                                       gpue_dbuf* poffsets, uint64_t n_rows, int mode,
                                       gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
                                       uint64_t* match_count);
+/* ---- SimdBlockFilter runtime filter ----
+ * The reference's split-block bloom (runtime_filter.h:79-232, upstream
+ * fastfilter_cpp): 32-byte buckets of 8 uint32 lanes, one bit per lane from
+ * (key*SALT[i])>>27; inserted hash = phmap_mix<8>(value) for integer keys
+ * (runtime_filter.h:1271-1276). directory must hold 32<<log_num_buckets
+ * bytes, where log_num_buckets = max(1, ceil(log2(n))-5)
+ * (runtime_filter.cpp:26-36). Build is atomicOr — the directory is
+ * bit-identical to the reference's serial build. test writes one u8 per row
+ * (1 = maybe-member): the scan-side early prune pushed to probe operators
+ * (operator.h:188-199). */
+int gpue_sbf_build_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n,
+                       int32_t log_num_buckets, gpue_dbuf* directory);
+int gpue_sbf_test_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, gpue_dbuf* directory,
+                      int32_t log_num_buckets, gpue_dbuf* out);
 /* RIGHT SEMI (anti=0) / RIGHT ANTI (anti=1): matched/unmatched BUILD rows. */
 int gpue_join_probe_right_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
                               uint64_t n_rows, int anti, gpue_dbuf* out_build_idx,

@@ -256,6 +256,62 @@ uint64_t orc_slice_probe_emit_mode(const uint8_t* bbytes, const uint32_t* boffse
     return m;
 }
 
+/* SimdBlockFilter — the reference's split-block bloom runtime filter
+ * (runtime_filter.h:79-232, runtime_filter.cpp:26-36,114-124; upstream
+ * fastfilter_cpp simd-block.h, "Cache-, Hash- and Space-Efficient Bloom
+ * Filters"). Buckets are 8 x uint32 (32 B); SALT per runtime_filter.h:58.
+ * init(n): log_num_buckets = max(1, ceil(log2(max(n,1))) - 5) and the
+ * directory holds 2^log_num_buckets buckets, zeroed.
+ * The inserted hash for integer keys is phmap_mix<8>(std::hash<T>(v))
+ * (runtime_filter.h:1271-1276); std::hash<T> for integral T is the
+ * sign-extended value itself (libstdc++). */
+static const uint32_t ORC_SBF_SALT[8] = {0x47b6137bu, 0x44974d91u, 0x8824ad5bu,
+                                         0xa2b7289du, 0x705495c7u, 0x2df1424bu,
+                                         0x9efc4947u, 0x5c6bfb31u};
+
+uint64_t orc_phmap_mix8(uint64_t a) {
+    const uint64_t k = 0xde5fb9d2630458e9ull;
+    unsigned __int128 p = (unsigned __int128)a * k;
+    return (uint64_t)(p >> 64) + (uint64_t)p;
+}
+
+int32_t orc_sbf_log_num_buckets(uint64_t nums) {
+    if (nums < 1) nums = 1;
+    int32_t log_heap_space = 0;
+    while ((1ull << log_heap_space) < nums) log_heap_space++; /* ceil(log2) */
+    return log_heap_space - 5 > 1 ? log_heap_space - 5 : 1;
+}
+
+void orc_sbf_insert_hash(uint32_t* directory, int32_t log_num_buckets, uint64_t h) {
+    uint32_t bucket = (uint32_t)(h & ((1ull << log_num_buckets) - 1));
+    uint32_t key = (uint32_t)(h >> log_num_buckets);
+    for (int i = 0; i < 8; i++)
+        directory[bucket * 8 + i] |= 1u << ((key * ORC_SBF_SALT[i]) >> 27);
+}
+
+int orc_sbf_test_hash(const uint32_t* directory, int32_t log_num_buckets, uint64_t h) {
+    uint32_t bucket = (uint32_t)(h & ((1ull << log_num_buckets) - 1));
+    uint32_t key = (uint32_t)(h >> log_num_buckets);
+    for (int i = 0; i < 8; i++)
+        if (!(directory[bucket * 8 + i] & (1u << ((key * ORC_SBF_SALT[i]) >> 27))))
+            return 0;
+    return 1;
+}
+
+void orc_sbf_build_i32(const int32_t* keys, uint64_t n, uint32_t* directory,
+                       int32_t log_num_buckets) {
+    for (uint64_t i = 0; i < n; i++)
+        orc_sbf_insert_hash(directory, log_num_buckets,
+                            orc_phmap_mix8((uint64_t)(int64_t)keys[i]));
+}
+
+void orc_sbf_test_i32(const int32_t* keys, uint64_t n, const uint32_t* directory,
+                      int32_t log_num_buckets, uint8_t* out) {
+    for (uint64_t i = 0; i < n; i++)
+        out[i] = (uint8_t)orc_sbf_test_hash(directory, log_num_buckets,
+                                            orc_phmap_mix8((uint64_t)(int64_t)keys[i]));
+}
+
 /* Nullable variants (construct_hash_table / lookup_init is_nulls paths,
  * join_hash_map_method.hpp:56-85,101-120): null build rows are skipped
  * (next=0 — the row never enters a chain); null probe rows get chain head 0

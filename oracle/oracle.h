@@ -50,6 +50,14 @@ void orc_bucket_chained_lookup_u32(const uint32_t* probe_keys, uint32_t probe_ro
 void orc_slice_build_u32(const uint8_t* bytes, const uint32_t* offsets, uint32_t row_count,
                          uint32_t* first, uint32_t* next, uint32_t bucket_size,
                          uint32_t log_bucket_size);
+uint64_t orc_phmap_mix8(uint64_t a);
+int32_t orc_sbf_log_num_buckets(uint64_t nums);
+void orc_sbf_insert_hash(uint32_t* directory, int32_t log_num_buckets, uint64_t h);
+int orc_sbf_test_hash(const uint32_t* directory, int32_t log_num_buckets, uint64_t h);
+void orc_sbf_build_i32(const int32_t* keys, uint64_t n, uint32_t* directory,
+                       int32_t log_num_buckets);
+void orc_sbf_test_i32(const int32_t* keys, uint64_t n, const uint32_t* directory,
+                      int32_t log_num_buckets, uint8_t* out);
 uint64_t orc_slice_probe_emit_mode(const uint8_t* bbytes, const uint32_t* boffsets,
                                    const uint32_t* next, uint32_t bucket_size,
                                    const uint32_t* first, const uint8_t* pbytes,

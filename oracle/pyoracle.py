@@ -74,6 +74,12 @@ def _decl(lib):
     lib.orc_slice_build_u32.argtypes = [c_vp, c_vp, u, c_vp, c_vp, u, u]
     lib.orc_slice_probe_emit.restype = c_u64
     lib.orc_slice_probe_emit.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, c_vp, u, c_vp, c_vp]
+    lib.orc_phmap_mix8.restype = c_u64
+    lib.orc_phmap_mix8.argtypes = [c_u64]
+    lib.orc_sbf_log_num_buckets.restype = c_i32
+    lib.orc_sbf_log_num_buckets.argtypes = [c_u64]
+    lib.orc_sbf_build_i32.argtypes = [c_vp, c_u64, c_vp, c_i32]
+    lib.orc_sbf_test_i32.argtypes = [c_vp, c_u64, c_vp, c_i32, c_vp]
     lib.orc_slice_probe_emit_mode.restype = c_u64
     lib.orc_slice_probe_emit_mode.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, c_vp, u,
                                               c_i32, c_vp, c_vp]
@@ -472,3 +478,19 @@ def partition_counting_sort(channel_ids: np.ndarray, num_channels: int):
     load().orc_partition_counting_sort(_p(channel_ids), len(channel_ids), num_channels,
                                        _p(sp), _p(ri))
     return sp, ri
+
+
+def sbf_build(keys_i32: np.ndarray):
+    """SimdBlockFilter build (runtime_filter.cpp:26-36): returns (directory,
+    log_num_buckets)."""
+    lib = load()
+    log = lib.orc_sbf_log_num_buckets(len(keys_i32))
+    directory = np.zeros((1 << log) * 8, np.uint32)
+    lib.orc_sbf_build_i32(_p(keys_i32), len(keys_i32), _p(directory), log)
+    return directory, log
+
+
+def sbf_test(keys_i32: np.ndarray, directory: np.ndarray, log: int) -> np.ndarray:
+    out = np.zeros(len(keys_i32), np.uint8)
+    load().orc_sbf_test_i32(_p(keys_i32), len(keys_i32), _p(directory), log, _p(out))
+    return out

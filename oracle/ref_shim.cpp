@@ -21,4 +21,7 @@ unsigned ref_fnv_hash(const void* d, int n, unsigned seed) {
 unsigned ref_xorshift32(unsigned x) {
     return starrocks::HashUtil::xorshift32(x);
 }
+unsigned long long ref_phmap_mix8(unsigned long long a) {
+    return starrocks::phmap_mix<8>()(a);
+}
 }

@@ -1735,6 +1735,93 @@ int gpue_join_probe_emit_varchar_mode(gpue_session* s, gpue_join_table* t, gpue_
     return GPUE_OK;
 }
 
+// ---------------------------------------------------------------------------
+// SimdBlockFilter — the reference's split-block bloom runtime filter
+// (runtime_filter.h:79-232, runtime_filter.cpp:26-36,114-124; upstream
+// fastfilter_cpp). 32 B buckets of 8 lanes; one bit per lane from
+// (key*SALT[i])>>27; inserted hash = phmap_mix<8>(value) for integer keys
+// (runtime_filter.h:1271-1276). Build uses atomicOr (order-independent, so
+// the directory is BIT-IDENTICAL to the oracle's serial build); probe is the
+// scan-side early-prune the reference pushes to scan operators
+// (operator.h:188-199). Oracle restatement: oracle.c orc_sbf_*.
+// ---------------------------------------------------------------------------
+__constant__ static const uint32_t SBF_SALT[8] = {0x47b6137bu, 0x44974d91u, 0x8824ad5bu,
+                                                  0xa2b7289du, 0x705495c7u, 0x2df1424bu,
+                                                  0x9efc4947u, 0x5c6bfb31u};
+
+__device__ static inline uint64_t phmap_mix8_dev(uint64_t a) {
+    const uint64_t k = 0xde5fb9d2630458e9ull;
+    return a * k + __umul64hi(a, k);
+}
+
+__global__ void k_sbf_build_i32(const int32_t* __restrict__ keys, uint64_t n,
+                                int32_t log_num_buckets, uint32_t* __restrict__ dir) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint64_t h = phmap_mix8_dev((uint64_t)(int64_t)keys[i]);
+        uint32_t bucket = (uint32_t)(h & ((1ull << log_num_buckets) - 1));
+        uint32_t key = (uint32_t)(h >> log_num_buckets);
+        #pragma unroll
+        for (int j = 0; j < 8; j++)
+            atomicOr(&dir[bucket * 8 + j], 1u << ((key * SBF_SALT[j]) >> 27));
+    }
+}
+
+__global__ void k_sbf_test_i32(const int32_t* __restrict__ keys, uint64_t n,
+                               const uint32_t* __restrict__ dir, int32_t log_num_buckets,
+                               uint8_t* __restrict__ out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint64_t h = phmap_mix8_dev((uint64_t)(int64_t)keys[i]);
+        uint32_t bucket = (uint32_t)(h & ((1ull << log_num_buckets) - 1));
+        uint32_t key = (uint32_t)(h >> log_num_buckets);
+        const uint32_t* b = &dir[bucket * 8];
+        uint4 lo = *(const uint4*)b;
+        uint4 hi = *(const uint4*)(b + 4);
+        uint32_t pass = 1;
+        const uint32_t* lanes = &lo.x;
+        #pragma unroll
+        for (int j = 0; j < 4; j++)
+            pass &= (lanes[j] >> ((key * SBF_SALT[j]) >> 27)) & 1u;
+        const uint32_t* lanes2 = &hi.x;
+        #pragma unroll
+        for (int j = 0; j < 4; j++)
+            pass &= (lanes2[j] >> ((key * SBF_SALT[j + 4]) >> 27)) & 1u;
+        out[i] = (uint8_t)pass;
+    }
+}
+
+extern "C" {
+int gpue_sbf_build_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n,
+                       int32_t log_num_buckets, gpue_dbuf* directory);
+int gpue_sbf_test_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, gpue_dbuf* directory,
+                      int32_t log_num_buckets, gpue_dbuf* out);
+}
+
+int gpue_sbf_build_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, int32_t log_num_buckets,
+                       gpue_dbuf* directory) {
+    ARG_CHECK(s && keys && directory && log_num_buckets >= 1 && log_num_buckets < 32);
+    ARG_CHECK(directory->bytes >= (32ull << log_num_buckets));
+    ARG_CHECK(keys->bytes >= n * 4);
+    HIP_CHECK(hipMemsetAsync(directory->ptr, 0, 32ull << log_num_buckets, s->stream));
+    hipLaunchKernelGGL(k_sbf_build_i32, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+                       (const int32_t*)keys->ptr, n, log_num_buckets,
+                       (uint32_t*)directory->ptr);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
+int gpue_sbf_test_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, gpue_dbuf* directory,
+                      int32_t log_num_buckets, gpue_dbuf* out) {
+    ARG_CHECK(s && keys && directory && out && log_num_buckets >= 1 && log_num_buckets < 32);
+    ARG_CHECK(keys->bytes >= n * 4 && out->bytes >= n);
+    hipLaunchKernelGGL(k_sbf_test_i32, dim3(grid_stream(n)), dim3(BLOCK), 0, s->stream,
+                       (const int32_t*)keys->ptr, n, (const uint32_t*)directory->ptr,
+                       log_num_buckets, (uint8_t*)out->ptr);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
 // Nullable probe (lookup_init is_nulls path + per-type semantics): a null
 // probe key matches nothing — INNER/SEMI emit nothing for it, ANTI/OUTER
 // emit the unmatched (i, 0) row.

@@ -127,6 +127,8 @@ def _declare(lib):
                                                  ctypes.POINTER(c_u64)]),
         "gpue_join_probe_emit_varchar_mode": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_i32,
                                                       c_vp, c_vp, ctypes.POINTER(c_u64)]),
+        "gpue_sbf_build_i32": (c_i32, [c_vp, c_vp, c_u64, c_i32, c_vp]),
+        "gpue_sbf_test_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_i32, c_vp]),
         "gpue_topk_i64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp]),
         "gpue_page_decode_bshuf_lz4_i32": (c_i32, [c_vp, c_vp, c_u32, c_vp]),
         "gpue_timer_start": (c_i32, [c_vp]),
@@ -526,6 +528,14 @@ class Engine:
     def page_decode_bshuf_lz4_i32(self, page: DBuf, n_values, out: DBuf):
         _ck(self._lib, self._lib.gpue_page_decode_bshuf_lz4_i32(self._h, page._h,
                                                                 n_values, out._h))
+
+    def sbf_build(self, keys: DBuf, n, log_num_buckets, directory: DBuf):
+        _ck(self._lib, self._lib.gpue_sbf_build_i32(self._h, keys._h, n, log_num_buckets,
+                                                    directory._h))
+
+    def sbf_test(self, keys: DBuf, n, directory: DBuf, log_num_buckets, out: DBuf):
+        _ck(self._lib, self._lib.gpue_sbf_test_i32(self._h, keys._h, n, directory._h,
+                                                   log_num_buckets, out._h))
 
     def topk_i64(self, keys: DBuf, vals: DBuf, n, k):
         ok = np.zeros(k, np.uint64)

@@ -884,3 +884,50 @@ def test_cabi_cpp_demo(engine):
     got = json.loads(out.stdout.strip())
     es, ecnt = orc.q1_pipeline(42, 0, n, 1993)
     assert (got["sum"], got["count"]) == (es, ecnt)
+
+
+def test_sbf_runtime_filter_parity(engine):
+    """SimdBlockFilter (split-block bloom runtime filter, runtime_filter.h:
+    79-232): the GPU atomicOr build produces a directory BIT-IDENTICAL to the
+    oracle's serial build (lane ORs are order-independent), membership tests
+    match per row, and probing it before a join (the reference's scan-side
+    early prune) never changes the join result."""
+    rng = np.random.default_rng(19)
+    n_build, n_probe = 200_000, 1_000_000
+    bkeys = rng.choice(20_000_000, n_build, replace=False).astype(np.int32)
+    pkeys = rng.integers(0, 40_000_000, n_probe).astype(np.int32)
+
+    directory, log = orc.sbf_build(bkeys)
+    kb = engine.alloc(bkeys.nbytes)
+    kb.h2d(bkeys)
+    d_dir = engine.alloc(directory.nbytes)
+    engine.sbf_build(kb, n_build, log, d_dir)
+    assert np.array_equal(d_dir.d2h(np.uint32, len(directory)), directory)
+
+    pb = engine.alloc(pkeys.nbytes)
+    pb.h2d(pkeys)
+    d_out = engine.alloc(n_probe)
+    engine.sbf_test(pb, n_probe, d_dir, log, d_out)
+    got = d_out.d2h(np.uint8, n_probe)
+    assert np.array_equal(got, orc.sbf_test(pkeys, directory, log))
+
+    # no false negatives: every build key tests positive
+    d_out2 = engine.alloc(n_build)
+    engine.sbf_test(kb, n_build, d_dir, log, d_out2)
+    assert d_out2.d2h(np.uint8, n_build).all()
+
+    # integration: join with SBF pre-prune == join without (result unchanged)
+    bkeys1 = np.concatenate([[0], bkeys]).astype(np.uint32)
+    kb1 = engine.alloc(bkeys1.nbytes)
+    kb1.h2d(bkeys1)
+    t = engine.join_build_bucket_chained(kb1, n_build)
+    full = engine.join_probe_emit(t, pb, n_probe)
+    kept = pkeys[got.astype(bool)]
+    pb2 = engine.alloc(max(kept.nbytes, 4))
+    pb2.h2d(kept)
+    pruned = engine.join_probe_emit(t, pb2, len(kept))
+    assert pruned == full  # the prune dropped only non-matching rows
+    assert len(kept) < n_probe  # and it did drop rows
+    for b in (kb, kb1, d_dir, pb, pb2, d_out, d_out2):
+        b.free()
+    t.destroy()

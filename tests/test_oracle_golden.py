@@ -238,3 +238,41 @@ def test_slice_join_oracle_vs_brute():
     expect = sorted((i, j) for i, s in enumerate(prows) for j in index.get(s, []))
     assert sorted(zip(op.tolist(), ob.tolist())) == expect
     assert len(expect) > 0
+
+
+def test_phmap_mix8_vs_reference_binary():
+    """orc_phmap_mix8 (the SimdBlockFilter insert hash for integer keys,
+    runtime_filter.h:1271-1276) against the reference's own phmap_mix<8>
+    compiled from its headers (oracle/_ref)."""
+    ref = orc.load_ref()
+    if ref is None:
+        pytest.skip("reference shim not built")
+    import ctypes
+    ref.ref_phmap_mix8.restype = ctypes.c_uint64
+    ref.ref_phmap_mix8.argtypes = [ctypes.c_uint64]
+    rng = np.random.default_rng(3)
+    for a in rng.integers(0, 2**63, 500, dtype=np.uint64).tolist() + [0, 1, 2**64 - 1]:
+        assert orc.load().orc_phmap_mix8(a) == ref.ref_phmap_mix8(a)
+
+
+def test_sbf_filter_properties():
+    """Split-block bloom: no false negatives ever; false-positive rate at the
+    reference's sizing (~32 bits/key budget) stays low; sizing formula
+    matches runtime_filter.cpp:26-31."""
+    lib = orc.load()
+    assert lib.orc_sbf_log_num_buckets(1) == 1
+    assert lib.orc_sbf_log_num_buckets(64) == 1
+    assert lib.orc_sbf_log_num_buckets(65) == 2
+    assert lib.orc_sbf_log_num_buckets(200_000) == 13
+    rng = np.random.default_rng(17)
+    members = rng.choice(5_000_000, 100_000, replace=False).astype(np.int32)
+    directory, log = orc.sbf_build(members)
+    assert sbf_all_pass(members, directory, log)
+    non = np.setdiff1d(rng.integers(5_000_000, 50_000_000, 100_000).astype(np.int32),
+                       members)
+    fp = orc.sbf_test(non, directory, log).mean()
+    assert fp < 0.05, fp
+
+
+def sbf_all_pass(keys, directory, log):
+    return bool(orc.sbf_test(keys, directory, log).all())
