@@ -2287,7 +2287,10 @@ k_q21_star_agg(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
         }
     };
     // (a 2-deep pk prefetch pipeline measured 9 % SLOWER here — the extra
-    // registers/branches cost more than the overlapped latency buys)
+    // registers/branches cost more than the overlapped latency buys;
+    // q43-style deferred stream loads behind the part mask measured 2565 vs
+    // 2792 GB/s — q21 is LATENCY-bound, so shortening the issue window by
+    // loading all four streams up front beats saving sk/od/rv lines)
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     for (; i + stride < n4; i += 2 * stride) {
         int4 pa = pk4[i], sa = sk4[i], oa = od4[i], ra = rv4[i];
@@ -2624,18 +2627,28 @@ k_q43_star_agg(const int32_t* __restrict__ ck, const int32_t* __restrict__ sk,
     const int4* __restrict__ rv4 = (const int4*)rv;
     const int4* __restrict__ sc4 = (const int4*)sc;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    auto quad = [&](int4 c4, int4 s4, int4 p4, int4 o4, int4 r4, int4 x4) {
-        uint32_t pb[4], pin[4];
+    // The part bitset (1/25 selective) gates everything: test it from pk
+    // alone, and only LOAD the other five streams for quads with a survivor.
+    // A 64 B line holds 16 rows; P(no survivor in a line) = (24/25)^16 = 0.52,
+    // so about half the ck/sk/od/rv/sc lines are never fetched (GPUE env
+    // A/B-measured).
+    auto quad_mask = [&](int4 p4) -> uint32_t {
+        uint32_t m = 0;
         #pragma unroll
         for (int j = 0; j < 4; j++) {
             uint32_t idx = (uint32_t)((&p4.x)[j] - psmin);
-            pin[j] = idx < psint;
-            uint32_t cidx = pin[j] ? idx : 0u;
-            pb[j] = pbits[cidx >> 5] >> (cidx & 31);
+            uint32_t cidx = idx < psint ? idx : 0u;
+            m |= ((idx < psint) & (pbits[cidx >> 5] >> (cidx & 31)) & 1u) << j;
         }
+        return m;
+    };
+    // rv/sc feed only the final accumulate: after all four filters the pass
+    // rate is ~1e-4, so load them per surviving ROW as scalars — those two
+    // streams are then almost never fetched at all.
+    auto quad_rest = [&](uint32_t m, uint64_t q, int4 c4, int4 s4, int4 p4, int4 o4) {
         #pragma unroll
         for (int j = 0; j < 4; j++) {
-            if (!(pin[j] & pb[j] & 1u)) continue;
+            if (!((m >> j) & 1u)) continue;
             uint32_t sidx = (uint32_t)((&s4.x)[j] - ssmin);
             if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
             uint32_t cidx = (uint32_t)((&c4.x)[j] - csmin);
@@ -2644,19 +2657,25 @@ k_q43_star_agg(const int32_t* __restrict__ ck, const int32_t* __restrict__ sk,
             if (dpay == 0) continue;
             uint32_t ppay = pfirst[(&p4.x)[j] - 1];
             uint32_t spay = sfirst[(&s4.x)[j] - 1];
+            uint64_t r = q * 4 + j;
             atomicAdd(&g[(dpay - 1) * 400 + (spay - 1) * 40 + (ppay - 1)],
-                      (unsigned long long)((int64_t)(&r4.x)[j] - (&x4.x)[j]));
+                      (unsigned long long)((int64_t)rv[r] - sc[r]));
         }
     };
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     for (; i + stride < n4; i += 2 * stride) {
-        int4 ca = ck4[i], sa = sk4[i], pa = pk4[i], oa = od4[i], ra = rv4[i], xa = sc4[i];
+        int4 pa = pk4[i];
         uint64_t i2 = i + stride;
-        int4 cb = ck4[i2], sb = sk4[i2], pb_ = pk4[i2], ob = od4[i2], rb = rv4[i2], xb = sc4[i2];
-        quad(ca, sa, pa, oa, ra, xa);
-        quad(cb, sb, pb_, ob, rb, xb);
+        int4 pb_ = pk4[i2];
+        uint32_t ma = quad_mask(pa), mb = quad_mask(pb_);
+        if (ma) quad_rest(ma, i, ck4[i], sk4[i], pa, od4[i]);
+        if (mb) quad_rest(mb, i2, ck4[i2], sk4[i2], pb_, od4[i2]);
     }
-    for (; i < n4; i += stride) quad(ck4[i], sk4[i], pk4[i], od4[i], rv4[i], sc4[i]);
+    for (; i < n4; i += stride) {
+        int4 p4 = pk4[i];
+        uint32_t m = quad_mask(p4);
+        if (m) quad_rest(m, i, ck4[i], sk4[i], p4, od4[i]);
+    }
     // scalar tail
     uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     for (uint64_t r = n4 * 4 + tid; r < n; r += stride) {
@@ -2693,7 +2712,7 @@ int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs, gpue_join_t
     ARG_CHECK(custs->bitset && supps->bitset && parts->bitset);
     ARG_CHECK(custs->min_key == 1 && supps->min_key == 1 && parts->min_key == 1);
     HIP_CHECK(hipMemsetAsync(group_sums->ptr, 0, NG_Q43 * sizeof(int64_t), s->stream));
-    hipLaunchKernelGGL(k_q43_star_agg, dim3(env_cap("GPUE_GRID_WIDE", 512)), dim3(BLOCK_Q21), 0, s->stream,
+    hipLaunchKernelGGL(k_q43_star_agg, dim3(env_cap("GPUE_GRID_Q43", 256)), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)ck->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)pk->ptr, (const int32_t*)od->ptr,
                        (const int32_t*)rv->ptr, (const int32_t*)sc->ptr, n,
