@@ -3554,6 +3554,34 @@ __global__ void k_bits_str16_eq(const ulonglong2* __restrict__ col, uint64_t n,
 // Each lane evaluates 4 consecutive orders (int4 loads); 8-lane groups pack
 // their 32 predicate bits into one word via shuffles — no atomics, each
 // 32-bit bitset word written exactly once.
+// Ballot variant (A/B via GPUE_Q3_OBITS_V2) — measured NEGATIVE: 8.3 ms vs
+// 2.0 ms for the quad kernel at 450 M orders. One order per thread leaves a
+// single outstanding cust-bitset gather per lane where the int4 quad form
+// keeps four in flight — the pass is gather-latency-bound, so
+// memory-level parallelism beats the simpler ballot assembly. Kept (off by
+// default) as the documented experiment.
+__global__ void k_q3_order_bits_ballot(const int32_t* __restrict__ ocust,
+                                       const int32_t* __restrict__ odate, uint64_t n_orders,
+                                       const uint32_t* __restrict__ cust_bits,
+                                       int32_t cutoff, uint32_t* __restrict__ bits) {
+    int lane = threadIdx.x & (WAVE - 1);
+    uint64_t wave_id = ((uint64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+    uint64_t total_waves = (uint64_t)gridDim.x * blockDim.x / WAVE;
+    for (uint64_t wb = wave_id * WAVE; wb < n_orders; wb += total_waves * WAVE) {
+        uint64_t o = wb + lane;
+        bool pass = false;
+        if (o < n_orders && odate[o] < cutoff) {
+            uint32_t c = (uint32_t)ocust[o] - 1;
+            pass = (cust_bits[c >> 5] >> (c & 31)) & 1u;
+        }
+        unsigned long long mask = __ballot(pass);
+        if (lane == 0) {
+            bits[wb >> 5] = (uint32_t)mask;
+            if (wb + 32 < n_orders) bits[(wb >> 5) + 1] = (uint32_t)(mask >> 32);
+        }
+    }
+}
+
 __global__ void k_q3_order_bits(const int32_t* __restrict__ ocust,
                                 const int32_t* __restrict__ odate, uint64_t n_orders,
                                 const uint32_t* __restrict__ cust_bits, int32_t cutoff,
@@ -3779,6 +3807,13 @@ int gpue_q3_order_bits(gpue_session* s, gpue_dbuf* ocust, gpue_dbuf* odate, uint
     ARG_CHECK(s && ocust && odate && cust_bits && order_bits);
     ARG_CHECK(order_bits->bytes >= (n_orders + 31) / 32 * 4);
     HIP_CHECK(hipMemsetAsync(order_bits->ptr, 0, (n_orders + 31) / 32 * 4, s->stream));
+    if (env_cap("GPUE_Q3_OBITS_V2", 0) == 1)
+        hipLaunchKernelGGL(k_q3_order_bits_ballot, dim3(grid_stream(n_orders)), dim3(BLOCK),
+                           0, s->stream, (const int32_t*)ocust->ptr,
+                           (const int32_t*)odate->ptr, n_orders,
+                           (const uint32_t*)cust_bits->ptr, cutoff,
+                           (uint32_t*)order_bits->ptr);
+    else
     hipLaunchKernelGGL(k_q3_order_bits, dim3(grid_for(n_orders)), dim3(BLOCK), 0, s->stream,
                        (const int32_t*)ocust->ptr, (const int32_t*)odate->ptr, n_orders,
                        (const uint32_t*)cust_bits->ptr, cutoff, (uint32_t*)order_bits->ptr);
