@@ -268,6 +268,12 @@ int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs, gpue_join_t
  * ascend by source row). start_points has num_channels+1 entries. */
 int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t num_channels,
                        uint64_t* start_points_out, gpue_dbuf* row_indexes_out);
+/* Multi-column partition key: the sink seeds FNV_SEED then CHAINS fnv_hash
+ * per partition column, each column seeding with the running hash
+ * (exchange_sink_operator.cpp:611-617). Two-int32-column variant. */
+int gpue_partition_2xi32(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b, uint64_t n,
+                         uint32_t num_channels, uint64_t* start_points_out,
+                         gpue_dbuf* row_indexes_out);
 
 /* ---- config 5: TPC-H Q3-shaped ----
  * lineitem ⋈ orders ⋈ customer; c_mktsegment = 16-byte dictionary string

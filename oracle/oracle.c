@@ -578,6 +578,18 @@ uint64_t orc_filter_i64_lt_mt(const int64_t* in, uint64_t n, int64_t theta, int6
  * default _exchange_hash_function_version=0 ⇒ fnv,
  * InternalService.thrift:378) + Shuffler HASH_PARTITIONED non-compat
  * ReduceOp (shuffler.h:71-86, hash_util.hpp:251-262). */
+/* Multi-column partition key: the exchange sink seeds FNV_SEED then CHAINS
+ * fnv_hash per partition column, each column seeding with the running hash
+ * (exchange_sink_operator.cpp:611-617). Two-int32 variant. */
+void orc_partition_channel_2xi32(const int32_t* a, const int32_t* b, uint64_t n,
+                                 uint32_t num_channels, uint32_t* out) {
+    for (uint64_t i = 0; i < n; i++) {
+        uint32_t h = orc_fnv_hash(&a[i], 4, 0x811C9DC5u);
+        h = orc_fnv_hash(&b[i], 4, h);
+        out[i] = (uint32_t)(((uint64_t)h * num_channels) >> 32);
+    }
+}
+
 void orc_partition_channel_u32(const uint32_t* keys, uint64_t n, uint32_t num_channels,
                                uint32_t* channel_ids) {
 #pragma omp parallel for schedule(static)

@@ -126,6 +126,8 @@ uint64_t orc_filter_i64_lt_mt(const int64_t* in, uint64_t n, int64_t theta, int6
  * hash_util.hpp:120-262) ----
  * hashes: per-row FNV over the 4-byte key with running seed FNV_SEED, then
  * channel = ReduceOp(hash, num_channels) = (hash * n) >> 32 (HASH_PARTITIONED). */
+void orc_partition_channel_2xi32(const int32_t* a, const int32_t* b, uint64_t n,
+                                 uint32_t num_channels, uint32_t* out);
 void orc_partition_channel_u32(const uint32_t* keys, uint64_t n, uint32_t num_channels,
                                uint32_t* channel_ids);
 /* counting-sort row layout: start_points[ch+1] sizes then reverse emit

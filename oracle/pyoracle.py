@@ -90,6 +90,7 @@ def _decl(lib):
     lib.orc_filter_i64_lt_mt.restype = c_u64
     lib.orc_filter_i64_lt_mt.argtypes = [c_vp, c_u64, c_i64, c_vp]
     lib.orc_partition_channel_u32.argtypes = [c_vp, c_u64, u, c_vp]
+    lib.orc_partition_channel_2xi32.argtypes = [c_vp, c_vp, c_u64, u, c_vp]
     lib.orc_partition_channel_u64.argtypes = [c_vp, c_u64, u, c_vp]
     lib.orc_partition_counting_sort.argtypes = [c_vp, c_u64, u, c_vp, c_vp]
     lib.orc_q1_pipeline.restype = c_i64

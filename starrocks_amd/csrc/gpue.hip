@@ -2786,6 +2786,45 @@ __global__ void k_partition_emit(const uint32_t* __restrict__ keys, uint64_t n, 
     }
 }
 
+// Multi-column partition key: the exchange sink seeds FNV_SEED then CHAINS
+// fnv_hash per partition column, each column using the running hash as its
+// seed (exchange_sink_operator.cpp:611-617). Two-int32 variant.
+__global__ void k_partition_hist_2xi32(const int32_t* __restrict__ a,
+                                       const int32_t* __restrict__ b, uint64_t n,
+                                       uint64_t tile, uint32_t nch,
+                                       uint32_t* __restrict__ block_hist) {
+    __shared__ uint32_t h[MAX_CH];
+    for (uint32_t c = threadIdx.x; c < nch; c += blockDim.x) h[c] = 0;
+    __syncthreads();
+    uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
+    for (uint64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        uint32_t hash = fnv_u32((uint32_t)b[i], fnv_u32((uint32_t)a[i], 0x811C9DC5u));
+        uint32_t ch = (uint32_t)(((uint64_t)hash * nch) >> 32);
+        atomicAdd(&h[ch], 1u);
+    }
+    __syncthreads();
+    for (uint32_t c = threadIdx.x; c < nch; c += blockDim.x)
+        block_hist[(uint64_t)blockIdx.x * nch + c] = h[c];
+}
+
+__global__ void k_partition_emit_2xi32(const int32_t* __restrict__ a,
+                                       const int32_t* __restrict__ b, uint64_t n,
+                                       uint64_t tile, uint32_t nch,
+                                       const uint64_t* __restrict__ block_offsets,
+                                       uint32_t* __restrict__ row_indexes) {
+    __shared__ uint64_t cursor[MAX_CH];
+    for (uint32_t c = threadIdx.x; c < nch; c += blockDim.x)
+        cursor[c] = block_offsets[(uint64_t)blockIdx.x * nch + c];
+    __syncthreads();
+    uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
+    for (uint64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        uint32_t hash = fnv_u32((uint32_t)b[i], fnv_u32((uint32_t)a[i], 0x811C9DC5u));
+        uint32_t ch = (uint32_t)(((uint64_t)hash * nch) >> 32);
+        uint64_t pos = atomicAdd((unsigned long long*)&cursor[ch], 1ull);
+        row_indexes[pos] = (uint32_t)i;
+    }
+}
+
 __global__ void k_partition_hist64(const uint64_t* __restrict__ keys, uint64_t n,
                                    uint64_t tile, uint32_t nch,
                                    uint32_t* __restrict__ block_hist) {
@@ -2903,6 +2942,50 @@ int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t nc
     return GPUE_OK;
 }
 
+
+extern "C" int gpue_partition_2xi32(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b,
+                                    uint64_t n, uint32_t nch, uint64_t* start_points_out,
+                                    gpue_dbuf* row_indexes_out);
+int gpue_partition_2xi32(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b, uint64_t n,
+                         uint32_t nch, uint64_t* start_points_out,
+                         gpue_dbuf* row_indexes_out) {
+    ARG_CHECK(s && a && b && start_points_out && row_indexes_out);
+    ARG_CHECK(nch >= 1 && nch <= MAX_CH);
+    ARG_CHECK(a->bytes >= n * 4 && b->bytes >= n * 4 && row_indexes_out->bytes >= n * 4);
+    uint32_t nb = grid_for(n);
+    uint64_t tile = (n + nb - 1) / nb;
+    uint32_t* d_hist = nullptr;
+    uint64_t* d_off = nullptr;
+    HIP_CHECK(hipMalloc(&d_hist, (uint64_t)nb * nch * sizeof(uint32_t)));
+    hipLaunchKernelGGL(k_partition_hist_2xi32, dim3(nb), dim3(BLOCK), 0, s->stream,
+                       (const int32_t*)a->ptr, (const int32_t*)b->ptr, n, tile, nch, d_hist);
+    uint32_t* h_hist = (uint32_t*)malloc((uint64_t)nb * nch * sizeof(uint32_t));
+    uint64_t* h_off = (uint64_t*)malloc((uint64_t)nb * nch * sizeof(uint64_t));
+    HIP_CHECK(hipMemcpyAsync(h_hist, d_hist, (uint64_t)nb * nch * sizeof(uint32_t),
+                             hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    uint64_t acc = 0;
+    for (uint32_t c = 0; c < nch; c++) {
+        start_points_out[c] = acc;
+        for (uint32_t blk = 0; blk < nb; blk++) {
+            h_off[(uint64_t)blk * nch + c] = acc;
+            acc += h_hist[(uint64_t)blk * nch + c];
+        }
+    }
+    start_points_out[nch] = acc;
+    HIP_CHECK(hipMalloc(&d_off, (uint64_t)nb * nch * sizeof(uint64_t)));
+    HIP_CHECK(hipMemcpyAsync(d_off, h_off, (uint64_t)nb * nch * sizeof(uint64_t),
+                             hipMemcpyHostToDevice, s->stream));
+    hipLaunchKernelGGL(k_partition_emit_2xi32, dim3(nb), dim3(BLOCK), 0, s->stream,
+                       (const int32_t*)a->ptr, (const int32_t*)b->ptr, n, tile, nch, d_off,
+                       (uint32_t*)row_indexes_out->ptr);
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_hist);
+    (void)hipFree(d_off);
+    free(h_hist);
+    free(h_off);
+    return GPUE_OK;
+}
 
 // ---------------------------------------------------------------------------
 // streaming microbenchmarks — establish the achievable ceilings the fused

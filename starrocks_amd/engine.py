@@ -96,6 +96,7 @@ def _declare(lib):
         "gpue_gather_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_gather_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_partition_i64": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
+        "gpue_partition_2xi32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
         "gpue_gen_lineitem_q3": (c_i32, [c_vp, c_u64, c_u64, c_u64, c_u64] + [c_vp] * 4),
         "gpue_gen_orders_q3": (c_i32, [c_vp, c_u64, c_u64, c_u32, c_vp, c_vp]),
         "gpue_gen_cust_mkt16": (c_i32, [c_vp, c_u64, c_u32, c_vp]),
@@ -523,6 +524,13 @@ class Engine:
         sp = np.zeros(num_channels + 1, dtype=np.uint64)
         _ck(self._lib, self._lib.gpue_partition_i32(self._h, keys._h, n, num_channels,
                                                     sp.ctypes.data_as(c_vp), row_indexes._h))
+        return sp
+
+    def partition_2xi32(self, a: DBuf, b: DBuf, n, num_channels,
+                        row_indexes: DBuf) -> np.ndarray:
+        sp = np.zeros(num_channels + 1, dtype=np.uint64)
+        _ck(self._lib, self._lib.gpue_partition_2xi32(
+            self._h, a._h, b._h, n, num_channels, sp.ctypes.data_as(c_vp), row_indexes._h))
         return sp
 
     def page_decode_bshuf_lz4_i32(self, page: DBuf, n_values, out: DBuf):

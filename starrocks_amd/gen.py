@@ -163,6 +163,23 @@ def partition_channels_u64(keys: np.ndarray, num_channels: int) -> np.ndarray:
             >> np.uint64(32)).astype(np.uint32)
 
 
+def fnv_u32_seeded(keys: np.ndarray, seeds: np.ndarray) -> np.ndarray:
+    """fnv over 4 LE bytes with a per-row running seed (the exchange sink's
+    chained multi-column hash, exchange_sink_operator.cpp:611-617)."""
+    with np.errstate(over="ignore"):
+        h = seeds.astype(np.uint32).copy()
+        k = keys.astype(np.uint32)
+        for b in range(4):
+            byte = ((k >> np.uint32(8 * b)) & np.uint32(0xFF))
+            h = (byte ^ h) * np.uint32(16777619)
+        return h
+
+
+def partition_channels_2xi32(a: np.ndarray, b: np.ndarray, num_channels: int) -> np.ndarray:
+    h = fnv_u32_seeded(b, fnv_u32(a))
+    return ((h.astype(np.uint64) * np.uint64(num_channels)) >> np.uint64(32)).astype(np.uint32)
+
+
 def partition_channels(keys: np.ndarray, num_channels: int) -> np.ndarray:
     """ReduceOp channel assignment (shuffler.h:71-86): (fnv(key)*n)>>32."""
     return ((fnv_u32(keys).astype(np.uint64) * np.uint64(num_channels))

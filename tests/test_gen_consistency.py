@@ -49,3 +49,21 @@ def test_lineorder_q21_columns():
         assert np.array_equal(x, y)
     pk = a[0]
     assert pk.min() >= 1 and pk.max() <= gen.N_PARTS_SF100
+
+
+def test_partition_channels_2xi32_oracle_vs_numpy():
+    """Chained 2-column FNV partition (exchange_sink_operator.cpp:611-617):
+    oracle C vs numpy restatement, bit-exact."""
+    import numpy as np
+    from oracle import pyoracle as orc
+    from starrocks_amd import gen
+    rng = np.random.default_rng(21)
+    a = rng.integers(-2**31, 2**31, 30_000).astype(np.int32)
+    b = rng.integers(-2**31, 2**31, 30_000).astype(np.int32)
+    for nch in (2, 5, 8, 64):
+        out = np.zeros(len(a), np.uint32)
+        orc.load().orc_partition_channel_2xi32(orc._p(a), orc._p(b), len(a), nch, orc._p(out))
+        assert np.array_equal(out, gen.partition_channels_2xi32(a, b, nch))
+        # chain property: equals fnv(b, seed=fnv(a, FNV_SEED)) reduced
+        h = gen.fnv_u32_seeded(b, gen.fnv_u32(a))
+        assert np.array_equal(out, ((h.astype(np.uint64) * nch) >> np.uint64(32)).astype(np.uint32))

@@ -931,3 +931,27 @@ def test_sbf_runtime_filter_parity(engine):
     for b in (kb, kb1, d_dir, pb, pb2, d_out, d_out2):
         b.free()
     t.destroy()
+
+
+def test_partition_2xi32_parity(engine):
+    """Two-column chained-FNV partition: channel sizes + per-channel row SETS
+    match the oracle (within-channel order is block-atomic nondeterministic,
+    as for the single-column kernels)."""
+    rng = np.random.default_rng(29)
+    n, nch = 2_000_000, 8
+    a = rng.integers(-2**31, 2**31, n).astype(np.int32)
+    b = rng.integers(-2**31, 2**31, n).astype(np.int32)
+    da, db = engine.alloc(a.nbytes), engine.alloc(b.nbytes)
+    da.h2d(a)
+    db.h2d(b)
+    ridx = engine.alloc(n * 4)
+    sp = engine.partition_2xi32(da, db, n, nch, ridx)
+    ch = gen.partition_channels_2xi32(a, b, nch)
+    expect_counts = np.bincount(ch, minlength=nch)
+    assert np.array_equal(np.diff(sp.astype(np.int64)), expect_counts)
+    got = ridx.d2h(np.uint32, n)
+    for c in range(nch):
+        seg = got[int(sp[c]):int(sp[c + 1])]
+        assert np.array_equal(np.sort(seg), np.flatnonzero(ch == c).astype(np.uint32))
+    for buf in (da, db, ridx):
+        buf.free()
