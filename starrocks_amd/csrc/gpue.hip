@@ -3711,6 +3711,62 @@ __global__ void k_q3_order_bits(const int32_t* __restrict__ ocust,
     }
 }
 
+// Branchless double-quad variant (A/B via GPUE_Q3_OBITS_V3): the customer-
+// bit gathers issue UNCONDITIONALLY with a pass-clamped index (the same
+// guide §5 4(c) pattern as k_q1_join_sum_bitset — an if around a load emits
+// per-element branches + vmcnt waits), and two quads run per iteration for
+// 8 gathers in flight per lane.
+__global__ void k_q3_order_bits_bl(const int32_t* __restrict__ ocust,
+                                   const int32_t* __restrict__ odate, uint64_t n_orders,
+                                   const uint32_t* __restrict__ cust_bits, int32_t cutoff,
+                                   uint32_t* __restrict__ bits) {
+    uint64_t n4 = (n_orders + 3) / 4;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    int lane = threadIdx.x & (WAVE - 1);
+    auto quad_bits = [&](uint64_t q) -> uint32_t {
+        uint32_t b4 = 0;
+        uint64_t base = q * 4;
+        if (base + 4 <= n_orders) {
+            int4 d4 = ((const int4*)odate)[q];
+            int4 c4 = ((const int4*)ocust)[q];
+            uint32_t pass[4], w[4];
+            #pragma unroll
+            for (int j = 0; j < 4; j++) {
+                pass[j] = (&d4.x)[j] < cutoff;
+                uint32_t c = pass[j] ? (uint32_t)(&c4.x)[j] - 1 : 0u;
+                w[j] = cust_bits[c >> 5] >> (c & 31);
+            }
+            #pragma unroll
+            for (int j = 0; j < 4; j++) b4 |= (w[j] & pass[j] & 1u) << j;
+        } else {
+            for (int j = 0; j < 4; j++) {
+                uint64_t o = base + j;
+                if (o < n_orders && odate[o] < cutoff) {
+                    uint32_t c = (uint32_t)ocust[o] - 1;
+                    b4 |= ((cust_bits[c >> 5] >> (c & 31)) & 1u) << j;
+                }
+            }
+        }
+        return b4;
+    };
+    for (uint64_t q = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; q - lane < n4;
+         q += 2 * stride) {
+        uint64_t q2 = q + stride;
+        uint32_t b4a = (q < n4) ? quad_bits(q) : 0u;
+        uint32_t b4b = (q2 < n4) ? quad_bits(q2) : 0u;
+        uint32_t wa = 0, wb = 0;
+        #pragma unroll
+        for (int k = 0; k < 8; k++) {
+            wa |= (uint32_t)__shfl((int)b4a, (lane & ~7) + k, WAVE) << (4 * k);
+            wb |= (uint32_t)__shfl((int)b4b, (lane & ~7) + k, WAVE) << (4 * k);
+        }
+        if (q < n4 && (lane & 7) == 0) bits[q / 8] = wa;
+        if (q2 - lane < n4) {
+            if (q2 < n4 && (lane & 7) == 0) bits[q2 / 8] = wb;
+        }
+    }
+}
+
 // fused lineitem filter + orders semi-probe + hash-agg insert.
 // Scalar per-row form: a 16 B/lane two-row variant with batched bitset
 // gathers measured 1188 vs 1759 GB/s — the ship filter (~50 % selective)
@@ -3890,7 +3946,12 @@ int gpue_q3_order_bits(gpue_session* s, gpue_dbuf* ocust, gpue_dbuf* odate, uint
     ARG_CHECK(s && ocust && odate && cust_bits && order_bits);
     ARG_CHECK(order_bits->bytes >= (n_orders + 31) / 32 * 4);
     HIP_CHECK(hipMemsetAsync(order_bits->ptr, 0, (n_orders + 31) / 32 * 4, s->stream));
-    if (env_cap("GPUE_Q3_OBITS_V2", 0) == 1)
+    if (env_cap("GPUE_Q3_OBITS_V3", 0) == 1)
+        hipLaunchKernelGGL(k_q3_order_bits_bl, dim3(grid_for(n_orders)), dim3(BLOCK), 0,
+                           s->stream, (const int32_t*)ocust->ptr, (const int32_t*)odate->ptr,
+                           n_orders, (const uint32_t*)cust_bits->ptr, cutoff,
+                           (uint32_t*)order_bits->ptr);
+    else if (env_cap("GPUE_Q3_OBITS_V2", 0) == 1)
         hipLaunchKernelGGL(k_q3_order_bits_ballot, dim3(grid_stream(n_orders)), dim3(BLOCK),
                            0, s->stream, (const int32_t*)ocust->ptr,
                            (const int32_t*)odate->ptr, n_orders,
