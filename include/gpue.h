@@ -192,6 +192,17 @@ int gpue_join_probe_emit_varchar_mode(gpue_session* s, gpue_join_table* t, gpue_
                                       gpue_dbuf* poffsets, uint64_t n_rows, int mode,
                                       gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
                                       uint64_t* match_count);
+/* Nullable Slice keys (is_nulls semantics as the fixed-size paths:
+ * join_hash_map_method.hpp:56-120 — null build rows never chain, null probe
+ * rows match nothing; ANTI/OUTER emit them unmatched). is_nulls u8, build
+ * side 1-based. */
+int gpue_join_build_varchar_nulls(gpue_session* s, gpue_dbuf* bytes, gpue_dbuf* offsets,
+                                  gpue_dbuf* is_nulls, uint64_t row_count,
+                                  gpue_join_table** out);
+int gpue_join_probe_emit_varchar_nulls(gpue_session* s, gpue_join_table* t, gpue_dbuf* pbytes,
+                                       gpue_dbuf* poffsets, gpue_dbuf* probe_nulls,
+                                       uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
+                                       gpue_dbuf* out_build_idx, uint64_t* match_count);
 /* ---- SimdBlockFilter runtime filter ----
  * The reference's split-block bloom (runtime_filter.h:79-232, upstream
  * fastfilter_cpp): 32-byte buckets of 8 uint32 lanes, one bit per lane from

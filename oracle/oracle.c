@@ -312,6 +312,65 @@ void orc_sbf_test_i32(const int32_t* keys, uint64_t n, const uint32_t* directory
                                             orc_phmap_mix8((uint64_t)(int64_t)keys[i]));
 }
 
+/* Nullable Slice keys (same is_nulls semantics as the fixed-size paths:
+ * null build rows never enter a chain, null probe rows match nothing and
+ * surface as unmatched for ANTI/OUTER). */
+void orc_slice_build_nulls_u32(const uint8_t* bytes, const uint32_t* offsets,
+                               const uint8_t* is_nulls, uint32_t row_count, uint32_t* first,
+                               uint32_t* next, uint32_t bucket_size,
+                               uint32_t log_bucket_size) {
+    (void)log_bucket_size;
+    for (uint32_t i = 1; i <= row_count; i++) {
+        if (is_nulls[i]) {
+            next[i] = 0;
+            continue;
+        }
+        uint32_t len = offsets[i + 1] - offsets[i];
+        uint32_t b = orc_crc_hash_32(bytes + offsets[i], (int32_t)len, 0x811C9DC5u) &
+                     (bucket_size - 1);
+        next[i] = first[b];
+        first[b] = i;
+    }
+}
+
+uint64_t orc_slice_probe_emit_nulls(const uint8_t* bbytes, const uint32_t* boffsets,
+                                    const uint32_t* next, uint32_t bucket_size,
+                                    const uint32_t* first, const uint8_t* pbytes,
+                                    const uint32_t* poffsets, const uint8_t* probe_nulls,
+                                    uint32_t probe_rows, int mode, uint32_t* out_probe_idx,
+                                    uint32_t* out_build_idx) {
+    uint64_t m = 0;
+    for (uint32_t i = 0; i < probe_rows; i++) {
+        uint32_t c = 0;
+        if (!probe_nulls[i]) {
+            uint32_t len = poffsets[i + 1] - poffsets[i];
+            uint32_t b = orc_crc_hash_32(pbytes + poffsets[i], (int32_t)len, 0x811C9DC5u) &
+                         (bucket_size - 1);
+            uint32_t j = first[b];
+            while (j != 0) {
+                uint32_t blen = boffsets[j + 1] - boffsets[j];
+                if (blen == len &&
+                    memcmp(bbytes + boffsets[j], pbytes + poffsets[i], len) == 0) {
+                    if (mode == 0 || mode == 3 || (mode == 1 && c == 0)) {
+                        out_probe_idx[m] = i;
+                        out_build_idx[m] = j;
+                        m++;
+                    }
+                    c++;
+                    if (mode == 1 || mode == 2) break;
+                }
+                j = next[j];
+            }
+        }
+        if (c == 0 && (mode == 2 || mode == 3)) {
+            out_probe_idx[m] = i;
+            out_build_idx[m] = 0;
+            m++;
+        }
+    }
+    return m;
+}
+
 /* Nullable variants (construct_hash_table / lookup_init is_nulls paths,
  * join_hash_map_method.hpp:56-85,101-120): null build rows are skipped
  * (next=0 — the row never enters a chain); null probe rows get chain head 0

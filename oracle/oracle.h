@@ -58,6 +58,16 @@ void orc_sbf_build_i32(const int32_t* keys, uint64_t n, uint32_t* directory,
                        int32_t log_num_buckets);
 void orc_sbf_test_i32(const int32_t* keys, uint64_t n, const uint32_t* directory,
                       int32_t log_num_buckets, uint8_t* out);
+void orc_slice_build_nulls_u32(const uint8_t* bytes, const uint32_t* offsets,
+                               const uint8_t* is_nulls, uint32_t row_count, uint32_t* first,
+                               uint32_t* next, uint32_t bucket_size,
+                               uint32_t log_bucket_size);
+uint64_t orc_slice_probe_emit_nulls(const uint8_t* bbytes, const uint32_t* boffsets,
+                                    const uint32_t* next, uint32_t bucket_size,
+                                    const uint32_t* first, const uint8_t* pbytes,
+                                    const uint32_t* poffsets, const uint8_t* probe_nulls,
+                                    uint32_t probe_rows, int mode, uint32_t* out_probe_idx,
+                                    uint32_t* out_build_idx);
 uint64_t orc_slice_probe_emit_mode(const uint8_t* bbytes, const uint32_t* boffsets,
                                    const uint32_t* next, uint32_t bucket_size,
                                    const uint32_t* first, const uint8_t* pbytes,

@@ -80,6 +80,10 @@ def _decl(lib):
     lib.orc_sbf_log_num_buckets.argtypes = [c_u64]
     lib.orc_sbf_build_i32.argtypes = [c_vp, c_u64, c_vp, c_i32]
     lib.orc_sbf_test_i32.argtypes = [c_vp, c_u64, c_vp, c_i32, c_vp]
+    lib.orc_slice_build_nulls_u32.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, u, u]
+    lib.orc_slice_probe_emit_nulls.restype = c_u64
+    lib.orc_slice_probe_emit_nulls.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, c_vp, c_vp,
+                                               u, c_i32, c_vp, c_vp]
     lib.orc_slice_probe_emit_mode.restype = c_u64
     lib.orc_slice_probe_emit_mode.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, c_vp, u,
                                               c_i32, c_vp, c_vp]
@@ -495,3 +499,22 @@ def sbf_test(keys_i32: np.ndarray, directory: np.ndarray, log: int) -> np.ndarra
     out = np.zeros(len(keys_i32), np.uint8)
     load().orc_sbf_test_i32(_p(keys_i32), len(keys_i32), _p(directory), log, _p(out))
     return out
+
+
+def slice_join_nulls(bbytes, boffsets, bnulls, row_count, pbytes, poffsets, pnulls,
+                     probe_rows, mode, max_out):
+    """Nullable Slice-key join: null build rows never chain; null probe rows
+    match nothing (ANTI/OUTER emit them unmatched)."""
+    lib = load()
+    bucket_size = lib.orc_calc_bucket_size(row_count + 1)
+    log = int(bucket_size).bit_length() - 1
+    first = np.zeros(bucket_size, np.uint32)
+    nxt = np.zeros(row_count + 1, np.uint32)
+    lib.orc_slice_build_nulls_u32(_p(bbytes), _p(boffsets), _p(bnulls), row_count,
+                                  _p(first), _p(nxt), bucket_size, log)
+    op = np.empty(max_out, np.uint32)
+    ob = np.empty(max_out, np.uint32)
+    m = lib.orc_slice_probe_emit_nulls(_p(bbytes), _p(boffsets), _p(nxt), bucket_size,
+                                       _p(first), _p(pbytes), _p(poffsets), _p(pnulls),
+                                       probe_rows, mode, _p(op), _p(ob))
+    return op[:m], ob[:m]

@@ -140,6 +140,7 @@ struct gpue_join_table {
                                     // probe-side equality check (1-based, row 0 sentinel)
     uint8_t* key_bytes = nullptr;   // VARCHAR: BinaryColumn bytes + uint32 offsets
     uint32_t* key_offsets = nullptr;
+    uint8_t* key_nulls = nullptr;   // VARCHAR nullable: is_nulls (1-based), or null
 };
 
 int gpue_device_count(int* out) {
@@ -1032,6 +1033,7 @@ void gpue_join_table_destroy(gpue_join_table* t) {
     if (t->build_keys) (void)hipFree(t->build_keys);
     if (t->key_bytes) (void)hipFree(t->key_bytes);
     if (t->key_offsets) (void)hipFree(t->key_offsets);
+    if (t->key_nulls) (void)hipFree(t->key_nulls);
     delete t;
 }
 
@@ -1562,12 +1564,14 @@ __device__ static inline uint32_t crc_hash_32_dev(const uint8_t* p, uint32_t len
 }
 
 __global__ void k_build_varchar(const uint8_t* __restrict__ bytes,
-                                const uint32_t* __restrict__ offsets, uint64_t row_count,
-                                uint32_t bucket_mask, uint32_t* __restrict__ first,
-                                uint32_t* __restrict__ next) {
+                                const uint32_t* __restrict__ offsets,
+                                const uint8_t* __restrict__ is_nulls /*may be null*/,
+                                uint64_t row_count, uint32_t bucket_mask,
+                                uint32_t* __restrict__ first, uint32_t* __restrict__ next) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i <= row_count;
          i += stride) {
+        if (is_nulls && is_nulls[i]) { next[i] = 0; continue; }
         uint32_t len = offsets[i + 1] - offsets[i];
         uint32_t b = crc_hash_32_dev(bytes + offsets[i], len, 0x811C9DC5u) & bucket_mask;
         next[i] = atomicExch(&first[b], (uint32_t)i);
@@ -1587,13 +1591,14 @@ __global__ void k_probe_count_vc(const uint8_t* __restrict__ pbytes,
                                  uint32_t bucket_mask, const uint32_t* __restrict__ first,
                                  const uint32_t* __restrict__ next,
                                  const uint8_t* __restrict__ bbytes,
-                                 const uint32_t* __restrict__ boffsets, int mode,
-                                 uint32_t* __restrict__ row_counts) {
+                                 const uint32_t* __restrict__ boffsets,
+                                 const uint8_t* __restrict__ pnulls /*may be null*/,
+                                 int mode, uint32_t* __restrict__ row_counts) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         uint32_t len = poffsets[i + 1] - poffsets[i];
         uint32_t b = crc_hash_32_dev(pbytes + poffsets[i], len, 0x811C9DC5u) & bucket_mask;
-        uint32_t j = first[b];
+        uint32_t j = (pnulls && pnulls[i]) ? 0u : first[b];
         uint32_t c = 0;
         while (j != 0) {
             c += slice_eq(bbytes + boffsets[j], boffsets[j + 1] - boffsets[j],
@@ -1610,6 +1615,7 @@ __global__ void k_probe_emit_vc(const uint8_t* __restrict__ pbytes,
                                 const uint32_t* __restrict__ next,
                                 const uint8_t* __restrict__ bbytes,
                                 const uint32_t* __restrict__ boffsets,
+                                const uint8_t* __restrict__ pnulls /*may be null*/,
                                 int mode, const uint32_t* __restrict__ row_counts,
                                 const uint64_t* __restrict__ row_offsets,
                                 uint32_t* __restrict__ out_probe,
@@ -1620,7 +1626,7 @@ __global__ void k_probe_emit_vc(const uint8_t* __restrict__ pbytes,
         uint64_t pos = row_offsets[i];
         uint32_t len = poffsets[i + 1] - poffsets[i];
         uint32_t b = crc_hash_32_dev(pbytes + poffsets[i], len, 0x811C9DC5u) & bucket_mask;
-        uint32_t j = first[b];
+        uint32_t j = (pnulls && pnulls[i]) ? 0u : first[b];
         uint32_t c = 0;
         while (j != 0) {
             if (slice_eq(bbytes + boffsets[j], boffsets[j + 1] - boffsets[j],
@@ -1653,10 +1659,23 @@ int gpue_join_probe_emit_varchar_mode(gpue_session* s, gpue_join_table* t, gpue_
                                       gpue_dbuf* poffsets, uint64_t n_rows, int mode,
                                       gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
                                       uint64_t* match_count);
+int gpue_join_probe_emit_varchar_nulls(gpue_session* s, gpue_join_table* t, gpue_dbuf* pbytes,
+                                       gpue_dbuf* poffsets, gpue_dbuf* pnulls,
+                                       uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
+                                       gpue_dbuf* out_build_idx, uint64_t* match_count);
+int gpue_join_build_varchar_nulls(gpue_session* s, gpue_dbuf* bytes, gpue_dbuf* offsets,
+                                  gpue_dbuf* is_nulls, uint64_t row_count,
+                                  gpue_join_table** out);
 }
 
 int gpue_join_build_varchar(gpue_session* s, gpue_dbuf* bytes, gpue_dbuf* offsets,
                             uint64_t row_count, gpue_join_table** out) {
+    return gpue_join_build_varchar_nulls(s, bytes, offsets, nullptr, row_count, out);
+}
+
+int gpue_join_build_varchar_nulls(gpue_session* s, gpue_dbuf* bytes, gpue_dbuf* offsets,
+                                  gpue_dbuf* is_nulls, uint64_t row_count,
+                                  gpue_join_table** out) {
     ARG_CHECK(s && bytes && offsets && out && row_count > 0);
     ARG_CHECK(offsets->bytes >= (row_count + 2) * 4);
     gpue_join_table* t = new gpue_join_table();
@@ -1675,8 +1694,14 @@ int gpue_join_build_varchar(gpue_session* s, gpue_dbuf* bytes, gpue_dbuf* offset
                              hipMemcpyDeviceToDevice, s->stream));
     HIP_CHECK(hipMemcpyAsync(t->key_offsets, offsets->ptr, (row_count + 2) * 4,
                              hipMemcpyDeviceToDevice, s->stream));
+    if (is_nulls) {
+        ARG_CHECK(is_nulls->bytes >= row_count + 1);
+        HIP_CHECK(hipMalloc(&t->key_nulls, row_count + 1));
+        HIP_CHECK(hipMemcpyAsync(t->key_nulls, is_nulls->ptr, row_count + 1,
+                                 hipMemcpyDeviceToDevice, s->stream));
+    }
     hipLaunchKernelGGL(k_build_varchar, dim3(grid_for(row_count)), dim3(BLOCK), 0, s->stream,
-                       t->key_bytes, t->key_offsets, row_count,
+                       t->key_bytes, t->key_offsets, t->key_nulls, row_count,
                        (uint32_t)(t->bucket_size - 1), t->first, t->next);
     HIP_CHECK(hipGetLastError());
     *out = t;
@@ -1695,6 +1720,14 @@ int gpue_join_probe_emit_varchar_mode(gpue_session* s, gpue_join_table* t, gpue_
                                       gpue_dbuf* poffsets, uint64_t n_rows, int mode,
                                       gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
                                       uint64_t* match_count) {
+    return gpue_join_probe_emit_varchar_nulls(s, t, pbytes, poffsets, nullptr, n_rows, mode,
+                                              out_probe_idx, out_build_idx, match_count);
+}
+
+int gpue_join_probe_emit_varchar_nulls(gpue_session* s, gpue_join_table* t, gpue_dbuf* pbytes,
+                                       gpue_dbuf* poffsets, gpue_dbuf* pnulls,
+                                       uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
+                                       gpue_dbuf* out_build_idx, uint64_t* match_count) {
     ARG_CHECK(s && t && pbytes && poffsets && match_count);
     ARG_CHECK(mode >= 0 && mode <= 3);
     ARG_CHECK(t->kind == gpue_join_table::VARCHAR);
@@ -1709,7 +1742,8 @@ int gpue_join_probe_emit_varchar_mode(gpue_session* s, gpue_join_table* t, gpue_
     hipLaunchKernelGGL(k_probe_count_vc, dim3(nb), dim3(BLOCK), 0, s->stream,
                        (const uint8_t*)pbytes->ptr, (const uint32_t*)poffsets->ptr, n_rows,
                        (uint32_t)(t->bucket_size - 1), t->first, t->next, t->key_bytes,
-                       t->key_offsets, mode, d_counts);
+                       t->key_offsets, pnulls ? (const uint8_t*)pnulls->ptr : nullptr,
+                       mode, d_counts);
     hipLaunchKernelGGL(k_block_sums_u32, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts,
                        n_rows, tile, d_bsums);
     hipLaunchKernelGGL(k_scan_small, dim3(1), dim3(1), 0, s->stream, d_bsums, nb);
@@ -1725,8 +1759,10 @@ int gpue_join_probe_emit_varchar_mode(gpue_session* s, gpue_join_table* t, gpue_
         hipLaunchKernelGGL(k_probe_emit_vc, dim3(nb), dim3(BLOCK), 0, s->stream,
                            (const uint8_t*)pbytes->ptr, (const uint32_t*)poffsets->ptr,
                            n_rows, (uint32_t)(t->bucket_size - 1), t->first, t->next,
-                           t->key_bytes, t->key_offsets, mode, d_counts, d_offsets,
-                           (uint32_t*)out_probe_idx->ptr, (uint32_t*)out_build_idx->ptr);
+                           t->key_bytes, t->key_offsets,
+                           pnulls ? (const uint8_t*)pnulls->ptr : nullptr, mode, d_counts,
+                           d_offsets, (uint32_t*)out_probe_idx->ptr,
+                           (uint32_t*)out_build_idx->ptr);
         HIP_CHECK(hipStreamSynchronize(s->stream));
         (void)hipFree(d_offsets);
     }

@@ -128,6 +128,11 @@ def _declare(lib):
                                                  ctypes.POINTER(c_u64)]),
         "gpue_join_probe_emit_varchar_mode": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_i32,
                                                       c_vp, c_vp, ctypes.POINTER(c_u64)]),
+        "gpue_join_build_varchar_nulls": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64,
+                                                  ctypes.POINTER(c_vp)]),
+        "gpue_join_probe_emit_varchar_nulls": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_u64,
+                                                       c_i32, c_vp, c_vp,
+                                                       ctypes.POINTER(c_u64)]),
         "gpue_sbf_build_i32": (c_i32, [c_vp, c_vp, c_u64, c_i32, c_vp]),
         "gpue_sbf_test_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_i32, c_vp]),
         "gpue_topk_i64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp]),
@@ -327,6 +332,23 @@ class Engine:
         ob = out_build._h if out_build else None
         _ck(self._lib, self._lib.gpue_join_probe_emit_varchar(
             self._h, table._h, pbytes._h, poffsets._h, n_rows, op, ob, ctypes.byref(cnt)))
+        return cnt.value
+
+    def join_build_varchar_nulls(self, bytes_: DBuf, offsets: DBuf, nulls: DBuf,
+                                 row_count) -> JoinTable:
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_join_build_varchar_nulls(
+            self._h, bytes_._h, offsets._h, nulls._h, row_count, ctypes.byref(h)))
+        return JoinTable(self, h)
+
+    def join_probe_emit_varchar_nulls(self, table, pbytes, poffsets, pnulls, n_rows, mode,
+                                      out_probe=None, out_build=None) -> int:
+        cnt = c_u64()
+        op = out_probe._h if out_probe else None
+        ob = out_build._h if out_build else None
+        _ck(self._lib, self._lib.gpue_join_probe_emit_varchar_nulls(
+            self._h, table._h, pbytes._h, poffsets._h, pnulls._h, n_rows, mode, op, ob,
+            ctypes.byref(cnt)))
         return cnt.value
 
     def join_probe_emit_varchar_mode(self, table, pbytes: DBuf, poffsets: DBuf, n_rows,

@@ -137,3 +137,36 @@ def test_slice_join_modes_vs_brute():
             assert all(brows[j - 1] == prows[p] for p, j in got)
         else:
             assert got == sorted(expect)
+
+
+def test_slice_join_nulls_vs_brute():
+    """Nullable Slice keys: null build rows invisible, null probe rows
+    unmatched (emitted for ANTI/OUTER with build 0), per
+    join_hash_map_method.hpp:56-120 semantics."""
+    rng = np.random.default_rng(41)
+    pool = [f"v{i}".encode() for i in range(30)]
+    brows = [pool[int(i)] for i in rng.integers(0, 30, 200)]
+    bnulls = np.concatenate([[0], rng.integers(0, 2, 200)]).astype(np.uint8)
+    prows = [pool[int(i) % 30] if i % 3 else b"nope" for i in rng.integers(0, 90, 400)]
+    pnulls = rng.integers(0, 2, 400).astype(np.uint8)
+    bb, bo = _vc_cols(brows, True)
+    pb, po = _vc_cols(prows, False)
+    index = {}
+    for j, r in enumerate(brows, start=1):
+        if not bnulls[j]:
+            index.setdefault(r, []).append(j)
+    for mode in (0, 2, 3):
+        op, ob = orc.slice_join_nulls(bb, bo, bnulls, len(brows), pb, po, pnulls,
+                                      len(prows), mode, 500_000)
+        got = sorted(zip(op.tolist(), ob.tolist()))
+        expect = []
+        for i, r in enumerate(prows):
+            hits = [] if pnulls[i] else index.get(r, [])
+            if mode == 0:
+                expect += [(i, j) for j in hits]
+            elif mode == 2:
+                if not hits:
+                    expect.append((i, 0))
+            else:
+                expect += [(i, j) for j in hits] if hits else [(i, 0)]
+        assert got == sorted(expect), mode

@@ -955,3 +955,42 @@ def test_partition_2xi32_parity(engine):
         assert np.array_equal(np.sort(seg), np.flatnonzero(ch == c).astype(np.uint32))
     for buf in (da, db, ridx):
         buf.free()
+
+
+def test_varchar_join_nulls_parity(engine):
+    """Nullable Slice-key join vs the oracle restatement across modes."""
+    rng = np.random.default_rng(43)
+    n_build, n_probe, card = 15_000, 80_000, 400
+    bb, bo, brows = _varchar_cols(rng, n_build, card, one_based=True)
+    pb_, po, prows = _varchar_cols(rng, n_probe, 2 * card, one_based=False)
+    bnulls = np.concatenate([[0], rng.integers(0, 2, n_build)]).astype(np.uint8)
+    pnulls = rng.integers(0, 2, n_probe).astype(np.uint8)
+    d_bb, d_bo = engine.alloc(bb.nbytes), engine.alloc(bo.nbytes)
+    d_bb.h2d(bb)
+    d_bo.h2d(bo)
+    d_bn = engine.alloc(bnulls.nbytes)
+    d_bn.h2d(bnulls)
+    t = engine.join_build_varchar_nulls(d_bb, d_bo, d_bn, n_build)
+    d_pb, d_po = engine.alloc(pb_.nbytes), engine.alloc(po.nbytes)
+    d_pb.h2d(pb_)
+    d_po.h2d(po)
+    d_pn = engine.alloc(pnulls.nbytes)
+    d_pn.h2d(pnulls)
+    for mode in (0, 2, 3):
+        eop, eob = orc.slice_join_nulls(bb, bo, bnulls, n_build, pb_, po, pnulls,
+                                        n_probe, mode, 64_000_000)
+        cnt = engine.join_probe_emit_varchar_nulls(t, d_pb, d_po, d_pn, n_probe, mode)
+        assert cnt == len(eop), mode
+        op_buf, ob_buf = engine.alloc(max(cnt, 1) * 4), engine.alloc(max(cnt, 1) * 4)
+        engine.join_probe_emit_varchar_nulls(t, d_pb, d_po, d_pn, n_probe, mode,
+                                             op_buf, ob_buf)
+        gop = op_buf.d2h(np.uint32, cnt)
+        gob = ob_buf.d2h(np.uint32, cnt)
+        got = np.sort(gop.astype(np.uint64) << np.uint64(32) | gob.astype(np.uint64))
+        exp = np.sort(eop.astype(np.uint64) << np.uint64(32) | eob.astype(np.uint64))
+        assert np.array_equal(got, exp), mode
+        op_buf.free()
+        ob_buf.free()
+    for b in (d_bb, d_bo, d_bn, d_pb, d_po, d_pn):
+        b.free()
+    t.destroy()
