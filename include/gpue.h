@@ -203,6 +203,15 @@ int gpue_join_probe_emit_varchar_nulls(gpue_session* s, gpue_join_table* t, gpue
                                        gpue_dbuf* poffsets, gpue_dbuf* probe_nulls,
                                        uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
                                        gpue_dbuf* out_build_idx, uint64_t* match_count);
+/* Dictionary-encoded binary page decode (binary_dict_page.cpp:229-280): the
+ * data page's int32 codewords (bitshuffle layer via
+ * gpue_page_decode_bshuf_lz4_i32) index the dict page's distinct strings
+ * (binary_plain_page.h string_at_index); output is a BinaryColumn (bytes +
+ * uint32 offsets, 0-based rows). Call with out_* NULL for the byte count. */
+int gpue_dict_decode_binary(gpue_session* s, gpue_dbuf* dict_bytes, gpue_dbuf* dict_offsets,
+                            gpue_dbuf* codes, uint64_t n, gpue_dbuf* out_bytes,
+                            gpue_dbuf* out_offsets, uint64_t* total_bytes);
+
 /* ---- SimdBlockFilter runtime filter ----
  * The reference's split-block bloom (runtime_filter.h:79-232, upstream
  * fastfilter_cpp): 32-byte buckets of 8 uint32 lanes, one bit per lane from

@@ -371,6 +371,27 @@ uint64_t orc_slice_probe_emit_nulls(const uint8_t* bbytes, const uint32_t* boffs
     return m;
 }
 
+/* Dictionary-encoded binary page decode (storage/rowset/binary_dict_page.cpp:
+ * 229-280): the data page holds int32 codewords (bitshuffle-encoded on disk —
+ * orc_page_decode_bshuf_lz4 above decodes that layer); each code indexes the
+ * dict page's distinct strings (binary_plain_page.h string_at_index). Output
+ * is BinaryColumn-shaped bytes + uint32 offsets (0-based rows). Returns total
+ * output bytes. */
+uint64_t orc_dict_decode_binary(const uint8_t* dict_bytes, const uint32_t* dict_offsets,
+                                const int32_t* codes, uint64_t n, uint8_t* out_bytes,
+                                uint32_t* out_offsets) {
+    uint64_t pos = 0;
+    for (uint64_t i = 0; i < n; i++) {
+        uint32_t c = (uint32_t)codes[i];
+        uint32_t len = dict_offsets[c + 1] - dict_offsets[c];
+        out_offsets[i] = (uint32_t)pos;
+        memcpy(out_bytes + pos, dict_bytes + dict_offsets[c], len);
+        pos += len;
+    }
+    out_offsets[n] = (uint32_t)pos;
+    return pos;
+}
+
 /* Nullable variants (construct_hash_table / lookup_init is_nulls paths,
  * join_hash_map_method.hpp:56-85,101-120): null build rows are skipped
  * (next=0 — the row never enters a chain); null probe rows get chain head 0

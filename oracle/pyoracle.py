@@ -80,6 +80,8 @@ def _decl(lib):
     lib.orc_sbf_log_num_buckets.argtypes = [c_u64]
     lib.orc_sbf_build_i32.argtypes = [c_vp, c_u64, c_vp, c_i32]
     lib.orc_sbf_test_i32.argtypes = [c_vp, c_u64, c_vp, c_i32, c_vp]
+    lib.orc_dict_decode_binary.restype = c_u64
+    lib.orc_dict_decode_binary.argtypes = [c_vp, c_vp, c_vp, c_u64, c_vp, c_vp]
     lib.orc_slice_build_nulls_u32.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, u, u]
     lib.orc_slice_probe_emit_nulls.restype = c_u64
     lib.orc_slice_probe_emit_nulls.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, c_vp, c_vp,
@@ -518,3 +520,15 @@ def slice_join_nulls(bbytes, boffsets, bnulls, row_count, pbytes, poffsets, pnul
                                        _p(first), _p(pbytes), _p(poffsets), _p(pnulls),
                                        probe_rows, mode, _p(op), _p(ob))
     return op[:m], ob[:m]
+
+
+def dict_decode_binary(dict_bytes, dict_offsets, codes):
+    """binary_dict_page.cpp:229-280 — codes -> BinaryColumn (bytes, offsets)."""
+    n = len(codes)
+    lens = dict_offsets[np.asarray(codes) + 1] - dict_offsets[np.asarray(codes)]
+    out_bytes = np.zeros(max(int(lens.sum()), 1), np.uint8)
+    out_offsets = np.zeros(n + 1, np.uint32)
+    total = load().orc_dict_decode_binary(_p(dict_bytes), _p(dict_offsets),
+                                          _p(np.ascontiguousarray(codes, np.int32)), n,
+                                          _p(out_bytes), _p(out_offsets))
+    return out_bytes[:total], out_offsets

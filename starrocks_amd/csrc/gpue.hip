@@ -1543,6 +1543,85 @@ int gpue_join_probe_emit_mode_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf
 }
 
 // ---------------------------------------------------------------------------
+// Dictionary-encoded binary page decode (binary_dict_page.cpp:229-280): the
+// data page's int32 codewords (bitshuffle layer handled by
+// gpue_page_decode_bshuf_lz4_i32) index the dict page's distinct strings;
+// output is a BinaryColumn (bytes + uint32 offsets). Two-phase: per-row
+// lengths -> scan -> parallel byte copy.
+// ---------------------------------------------------------------------------
+__global__ void k_dict_code_lengths(const int32_t* __restrict__ codes, uint64_t n,
+                                    const uint32_t* __restrict__ doff,
+                                    uint32_t* __restrict__ lens) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint32_t c = (uint32_t)codes[i];
+        lens[i] = doff[c + 1] - doff[c];
+    }
+}
+
+__global__ void k_dict_emit(const int32_t* __restrict__ codes, uint64_t n,
+                            const uint8_t* __restrict__ dbytes,
+                            const uint32_t* __restrict__ doff,
+                            const uint64_t* __restrict__ row_offsets, uint64_t total,
+                            uint8_t* __restrict__ out_bytes,
+                            uint32_t* __restrict__ out_offsets) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (tid == 0) out_offsets[n] = (uint32_t)total;
+    for (uint64_t i = tid; i < n; i += stride) {
+        uint32_t c = (uint32_t)codes[i];
+        uint32_t len = doff[c + 1] - doff[c];
+        uint64_t pos = row_offsets[i];
+        out_offsets[i] = (uint32_t)pos;
+        const uint8_t* src = dbytes + doff[c];
+        for (uint32_t k = 0; k < len; k++) out_bytes[pos + k] = src[k];
+    }
+}
+
+extern "C" int gpue_dict_decode_binary(gpue_session* s, gpue_dbuf* dict_bytes,
+                                       gpue_dbuf* dict_offsets, gpue_dbuf* codes,
+                                       uint64_t n, gpue_dbuf* out_bytes,
+                                       gpue_dbuf* out_offsets, uint64_t* total_bytes);
+int gpue_dict_decode_binary(gpue_session* s, gpue_dbuf* dict_bytes, gpue_dbuf* dict_offsets,
+                            gpue_dbuf* codes, uint64_t n, gpue_dbuf* out_bytes,
+                            gpue_dbuf* out_offsets, uint64_t* total_bytes) {
+    ARG_CHECK(s && dict_bytes && dict_offsets && codes && total_bytes);
+    ARG_CHECK(codes->bytes >= n * 4);
+    uint32_t nb = grid_for(n);
+    uint64_t tile = (n + nb - 1) / nb;
+    uint32_t* d_lens = nullptr;
+    uint64_t* d_bsums = nullptr;
+    uint64_t* d_off = nullptr;
+    HIP_CHECK(hipMalloc(&d_lens, n * 4));
+    HIP_CHECK(hipMalloc(&d_bsums, (nb + 1) * 8));
+    hipLaunchKernelGGL(k_dict_code_lengths, dim3(nb), dim3(BLOCK), 0, s->stream,
+                       (const int32_t*)codes->ptr, n, (const uint32_t*)dict_offsets->ptr,
+                       d_lens);
+    hipLaunchKernelGGL(k_block_sums_u32, dim3(nb), dim3(BLOCK), 0, s->stream, d_lens, n,
+                       tile, d_bsums);
+    hipLaunchKernelGGL(k_scan_small, dim3(1), dim3(1), 0, s->stream, d_bsums, nb);
+    uint64_t total = 0;
+    HIP_CHECK(hipMemcpyAsync(&total, d_bsums + nb, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    *total_bytes = total;
+    if (out_bytes && out_offsets) {
+        ARG_CHECK(out_bytes->bytes >= total && out_offsets->bytes >= (n + 1) * 4);
+        HIP_CHECK(hipMalloc(&d_off, n * 8));
+        hipLaunchKernelGGL(k_scan_offsets, dim3(nb), dim3(BLOCK), 0, s->stream, d_lens, n,
+                           tile, d_bsums, d_off);
+        hipLaunchKernelGGL(k_dict_emit, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const int32_t*)codes->ptr, n, (const uint8_t*)dict_bytes->ptr,
+                           (const uint32_t*)dict_offsets->ptr, d_off, total,
+                           (uint8_t*)out_bytes->ptr, (uint32_t*)out_offsets->ptr);
+        HIP_CHECK(hipStreamSynchronize(s->stream));
+        (void)hipFree(d_off);
+    }
+    (void)hipFree(d_lens);
+    (void)hipFree(d_bsums);
+    return GPUE_OK;
+}
+
+// ---------------------------------------------------------------------------
 // SERIALIZED_VARCHAR / Slice keys (the selector's last constructor branch):
 // JoinKeyHash<Slice> = crc_hash_32(bytes, len, 0x811C9DC5) & (bucket_size-1)
 // (join_hash_map_helper.h:57-64; CRC32-C + phmap_mix<4>, bitwise device

@@ -133,6 +133,8 @@ def _declare(lib):
         "gpue_join_probe_emit_varchar_nulls": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_u64,
                                                        c_i32, c_vp, c_vp,
                                                        ctypes.POINTER(c_u64)]),
+        "gpue_dict_decode_binary": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_vp, c_vp,
+                                            ctypes.POINTER(c_u64)]),
         "gpue_sbf_build_i32": (c_i32, [c_vp, c_vp, c_u64, c_i32, c_vp]),
         "gpue_sbf_test_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_i32, c_vp]),
         "gpue_topk_i64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp]),
@@ -558,6 +560,15 @@ class Engine:
     def page_decode_bshuf_lz4_i32(self, page: DBuf, n_values, out: DBuf):
         _ck(self._lib, self._lib.gpue_page_decode_bshuf_lz4_i32(self._h, page._h,
                                                                 n_values, out._h))
+
+    def dict_decode_binary(self, dict_bytes: DBuf, dict_offsets: DBuf, codes: DBuf, n,
+                           out_bytes: DBuf = None, out_offsets: DBuf = None) -> int:
+        t = c_u64()
+        ob = out_bytes._h if out_bytes else None
+        oo = out_offsets._h if out_offsets else None
+        _ck(self._lib, self._lib.gpue_dict_decode_binary(
+            self._h, dict_bytes._h, dict_offsets._h, codes._h, n, ob, oo, ctypes.byref(t)))
+        return t.value
 
     def sbf_build(self, keys: DBuf, n, log_num_buckets, directory: DBuf):
         _ck(self._lib, self._lib.gpue_sbf_build_i32(self._h, keys._h, n, log_num_buckets,

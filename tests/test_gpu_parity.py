@@ -994,3 +994,49 @@ def test_varchar_join_nulls_parity(engine):
     for b in (d_bb, d_bo, d_bn, d_pb, d_po, d_pn):
         b.free()
     t.destroy()
+
+
+def test_dict_decode_and_varchar_join_chain(engine):
+    """Storage ingress chain for string columns: bitshuffle+LZ4 codes page
+    decode -> dict decode (binary_dict_page.cpp:229-280) -> BinaryColumn ->
+    varchar join, each stage bit-exact vs the oracle."""
+    rng = np.random.default_rng(53)
+    words = [f"w{i:03d}".encode() + b"q" * int(rng.integers(0, 5)) for i in range(200)]
+    doff = np.zeros(201, np.uint32)
+    np.cumsum([len(w) for w in words], out=doff[1:])
+    dbytes = np.frombuffer(b"".join(words), np.uint8).copy()
+    n = 100_000
+    codes = rng.integers(0, 200, n).astype(np.int32)
+    # encode the codes page with the oracle encoder, decode it on GPU
+    page = orc.bshuf_lz4_encode_i32(codes)
+    d_page = engine.alloc(page.nbytes)
+    d_page.h2d(page)
+    d_codes = engine.alloc(n * 4)
+    engine.page_decode_bshuf_lz4_i32(d_page, n, d_codes)
+    assert np.array_equal(d_codes.d2h(np.int32, n), codes)
+    # dict decode on GPU vs oracle
+    d_db, d_do = engine.alloc(dbytes.nbytes), engine.alloc(doff.nbytes)
+    d_db.h2d(dbytes)
+    d_do.h2d(doff)
+    total = engine.dict_decode_binary(d_db, d_do, d_codes, n)
+    eb, eo = orc.dict_decode_binary(dbytes, doff, codes)
+    assert total == len(eb)
+    d_ob, d_oo = engine.alloc(total), engine.alloc((n + 1) * 4)
+    engine.dict_decode_binary(d_db, d_do, d_codes, n, d_ob, d_oo)
+    assert np.array_equal(d_ob.d2h(np.uint8, total), eb)
+    assert np.array_equal(d_oo.d2h(np.uint32, n + 1), eo)
+    # the decoded BinaryColumn probes a varchar build table (subset of words)
+    brows = [b""] + [words[i] for i in range(0, 200, 2)]
+    bo = np.zeros(len(brows) + 1, np.uint32)
+    np.cumsum([len(r) for r in brows], out=bo[1:])
+    bb = np.frombuffer(b"".join(brows), np.uint8).copy()
+    d_bb, d_bo = engine.alloc(bb.nbytes), engine.alloc(bo.nbytes)
+    d_bb.h2d(bb)
+    d_bo.h2d(bo)
+    t = engine.join_build_varchar(d_bb, d_bo, len(brows) - 1)
+    cnt = engine.join_probe_emit_varchar(t, d_ob, d_oo, n)
+    expect_cnt = int(np.sum(codes % 2 == 0))  # even codes are in the build set
+    assert cnt == expect_cnt
+    for b in (d_page, d_codes, d_db, d_do, d_ob, d_oo, d_bb, d_bo):
+        b.free()
+    t.destroy()
