@@ -203,6 +203,24 @@ int gpue_join_probe_emit_varchar_nulls(gpue_session* s, gpue_join_table* t, gpue
                                        gpue_dbuf* poffsets, gpue_dbuf* probe_nulls,
                                        uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
                                        gpue_dbuf* out_build_idx, uint64_t* match_count);
+/* ---- streaming pre-aggregation building blocks ----
+ * The AUTO-mode streaming agg (aggregate_streaming_sink_operator.cpp:224-310;
+ * thresholds aggregator.h:175-178: LowReduction 0.2, HighReduction 0.9,
+ * StableLimit 5) over the persistent gpue_agg_table. push = build_hash_map +
+ * compute_batch_agg_states (cnts NULL: each row counts 1; non-NULL: rows are
+ * pre-aggregated partials, merge_batch semantics aggregate.h:158-168;
+ * update_only=1 aggregates only rows whose group exists and marks the rest
+ * in miss_mask — the selective pre-agg form); probe_hits =
+ * build_hash_map_with_selection's hit count; emit = convert_hash_map_to_chunk. */
+int gpue_hash_agg_push_u64(gpue_session* s, gpue_agg_table* at, gpue_dbuf* keys,
+                           gpue_dbuf* vals, gpue_dbuf* cnts, uint64_t n, int update_only,
+                           gpue_dbuf* miss_mask, uint64_t* hits_out);
+int gpue_hash_agg_probe_hits_u64(gpue_session* s, gpue_agg_table* at, gpue_dbuf* keys,
+                                 uint64_t n, uint64_t* hits_out);
+int gpue_hash_agg_emit_u64(gpue_session* s, gpue_agg_table* at, gpue_dbuf* out_keys,
+                           gpue_dbuf* out_sums, gpue_dbuf* out_counts, uint64_t max_out,
+                           uint64_t* n_groups);
+
 /* Dictionary-encoded binary page decode (binary_dict_page.cpp:229-280): the
  * data page's int32 codewords (bitshuffle layer via
  * gpue_page_decode_bshuf_lz4_i32) index the dict page's distinct strings

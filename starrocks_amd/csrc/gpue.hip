@@ -3696,6 +3696,148 @@ int gpue_hash_agg_sum_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uin
 }
 
 // ---------------------------------------------------------------------------
+// Streaming pre-aggregation building blocks (aggregate_streaming_sink_
+// operator.cpp:224-310 AUTO policy; the host-side state machine lives in
+// starrocks_amd/pipeline.py StreamingAggOperator). The persistent
+// gpue_agg_table is the operator's hash map:
+//  - push: build_hash_map + compute_batch_agg_states (insert or update);
+//    cnts == null means each row contributes count 1, otherwise the row is a
+//    pre-aggregated partial (merge_batch semantics, exprs/agg/aggregate.h:
+//    158-168) contributing its own count;
+//  - probe: build_hash_map_with_selection's hit count (rows whose group
+//    already exists — SIMD::count_zero(streaming_selection));
+//  - update-only: _push_chunk_by_selective_preaggregation — aggregate rows
+//    whose group exists, mark the rest for pass-through.
+// ---------------------------------------------------------------------------
+__global__ void k_hash_agg_push(const uint64_t* __restrict__ keys,
+                                const int64_t* __restrict__ vals,
+                                const int64_t* __restrict__ cnts, uint64_t n,
+                                unsigned long long* __restrict__ slots,
+                                unsigned long long* __restrict__ sums,
+                                unsigned long long* __restrict__ counts, uint64_t cap_mask,
+                                int update_only, uint8_t* __restrict__ miss_mask,
+                                unsigned long long* __restrict__ hits_out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    unsigned long long local_hits = 0;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        unsigned long long k = keys[i];
+        unsigned long long v = vals ? (unsigned long long)vals[i] : 0ull;
+        unsigned long long c = cnts ? (unsigned long long)cnts[i] : 1ull;
+        uint64_t slot = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        for (;;) {
+            unsigned long long cur = slots[slot];
+            if (cur == k) {
+                if (vals) {
+                    atomicAdd(&sums[slot], v);
+                    atomicAdd(&counts[slot], c);
+                }
+                local_hits++;
+                if (miss_mask) miss_mask[i] = 0;
+                break;
+            }
+            if (cur == AGG_EMPTY) {
+                if (update_only) { // new group: leave for pass-through
+                    if (miss_mask) miss_mask[i] = 1;
+                    break;
+                }
+                unsigned long long old = atomicCAS(&slots[slot], AGG_EMPTY, k);
+                if (old == AGG_EMPTY || old == k) {
+                    if (vals) {
+                        atomicAdd(&sums[slot], v);
+                        atomicAdd(&counts[slot], c);
+                    }
+                    if (old == k) local_hits++;
+                    if (miss_mask) miss_mask[i] = (old != k);
+                    break;
+                }
+                continue; // lost the claim to another key: re-read this slot
+            }
+            slot = (slot + 1) & cap_mask;
+        }
+    }
+    if (hits_out) {
+        for (int off = 32; off > 0; off >>= 1)
+            local_hits += __shfl_down(local_hits, off, WAVE);
+        if ((threadIdx.x & (WAVE - 1)) == 0 && local_hits)
+            atomicAdd(hits_out, local_hits);
+    }
+}
+
+extern "C" {
+int gpue_hash_agg_push_u64(gpue_session* s, gpue_agg_table* at, gpue_dbuf* keys,
+                           gpue_dbuf* vals, gpue_dbuf* cnts /*nullable*/, uint64_t n,
+                           int update_only, gpue_dbuf* miss_mask /*nullable u8*/,
+                           uint64_t* hits_out /*nullable*/);
+int gpue_hash_agg_probe_hits_u64(gpue_session* s, gpue_agg_table* at, gpue_dbuf* keys,
+                                 uint64_t n, uint64_t* hits_out);
+int gpue_hash_agg_emit_u64(gpue_session* s, gpue_agg_table* at, gpue_dbuf* out_keys,
+                           gpue_dbuf* out_sums, gpue_dbuf* out_counts /*nullable*/,
+                           uint64_t max_out, uint64_t* n_groups);
+}
+
+static int hash_agg_push_impl(gpue_session* s, gpue_agg_table* at, gpue_dbuf* keys,
+                              gpue_dbuf* vals, gpue_dbuf* cnts, uint64_t n, int update_only,
+                              gpue_dbuf* miss_mask, uint64_t* hits_out) {
+    ARG_CHECK(s && at && keys && keys->bytes >= n * 8);
+    ARG_CHECK(!vals || vals->bytes >= n * 8);
+    ARG_CHECK(!cnts || cnts->bytes >= n * 8);
+    ARG_CHECK(!miss_mask || miss_mask->bytes >= n);
+    unsigned long long* d_hits = nullptr;
+    if (hits_out) {
+        HIP_CHECK(hipMalloc(&d_hits, 8));
+        HIP_CHECK(hipMemsetAsync(d_hits, 0, 8, s->stream));
+    }
+    hipLaunchKernelGGL(k_hash_agg_push, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+                       (const uint64_t*)keys->ptr, vals ? (const int64_t*)vals->ptr : nullptr,
+                       cnts ? (const int64_t*)cnts->ptr : nullptr, n, at->slots, at->sums,
+                       at->counts, at->cap - 1, update_only,
+                       miss_mask ? (uint8_t*)miss_mask->ptr : nullptr, d_hits);
+    HIP_CHECK(hipGetLastError());
+    if (hits_out) {
+        unsigned long long h = 0;
+        HIP_CHECK(hipMemcpyAsync(&h, d_hits, 8, hipMemcpyDeviceToHost, s->stream));
+        HIP_CHECK(hipStreamSynchronize(s->stream));
+        (void)hipFree(d_hits);
+        *hits_out = h;
+    }
+    return GPUE_OK;
+}
+
+int gpue_hash_agg_push_u64(gpue_session* s, gpue_agg_table* at, gpue_dbuf* keys,
+                           gpue_dbuf* vals, gpue_dbuf* cnts, uint64_t n, int update_only,
+                           gpue_dbuf* miss_mask, uint64_t* hits_out) {
+    return hash_agg_push_impl(s, at, keys, vals, cnts, n, update_only, miss_mask, hits_out);
+}
+
+int gpue_hash_agg_probe_hits_u64(gpue_session* s, gpue_agg_table* at, gpue_dbuf* keys,
+                                 uint64_t n, uint64_t* hits_out) {
+    ARG_CHECK(hits_out);
+    // probe-only: update_only with no vals — counts untouched, misses unmarked
+    return hash_agg_push_impl(s, at, keys, nullptr, nullptr, n, 1, nullptr, hits_out);
+}
+
+int gpue_hash_agg_emit_u64(gpue_session* s, gpue_agg_table* at, gpue_dbuf* out_keys,
+                           gpue_dbuf* out_sums, gpue_dbuf* out_counts, uint64_t max_out,
+                           uint64_t* n_groups) {
+    ARG_CHECK(s && at && out_keys && out_sums && n_groups);
+    HIP_CHECK(hipMemsetAsync(at->cursor, 0, 8, s->stream));
+    hipLaunchKernelGGL(k_hash_agg_emit, dim3(grid_for(at->cap)), dim3(BLOCK), 0, s->stream,
+                       at->slots, at->sums, at->counts, at->cap, at->cursor, max_out,
+                       (uint64_t*)out_keys->ptr, (int64_t*)out_sums->ptr,
+                       out_counts ? (int64_t*)out_counts->ptr : nullptr);
+    unsigned long long groups = 0;
+    HIP_CHECK(hipMemcpyAsync(&groups, at->cursor, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    *n_groups = groups;
+    if (groups > max_out) {
+        snprintf(g_err, sizeof(g_err), "hash_agg_emit: %llu groups exceed max_out %llu",
+                 groups, (unsigned long long)max_out);
+        return GPUE_ERR_ARG;
+    }
+    return GPUE_OK;
+}
+
+// ---------------------------------------------------------------------------
 // Config 5 — TPC-H Q3-shaped: lineitem ⋈ orders ⋈ customer with a 16-byte
 // dictionary-string filter (SERIALIZED_FIXED_SIZE_LARGEINT packing: two u64
 // compares — reference join_hash_table.cpp:185-192), decimal revenue as

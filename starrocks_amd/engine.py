@@ -135,6 +135,12 @@ def _declare(lib):
                                                        ctypes.POINTER(c_u64)]),
         "gpue_dict_decode_binary": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_vp, c_vp,
                                             ctypes.POINTER(c_u64)]),
+        "gpue_hash_agg_push_u64": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_i32,
+                                           c_vp, ctypes.POINTER(c_u64)]),
+        "gpue_hash_agg_probe_hits_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64,
+                                                 ctypes.POINTER(c_u64)]),
+        "gpue_hash_agg_emit_u64": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_u64,
+                                           ctypes.POINTER(c_u64)]),
         "gpue_sbf_build_i32": (c_i32, [c_vp, c_vp, c_u64, c_i32, c_vp]),
         "gpue_sbf_test_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_i32, c_vp]),
         "gpue_topk_i64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp]),
@@ -560,6 +566,29 @@ class Engine:
     def page_decode_bshuf_lz4_i32(self, page: DBuf, n_values, out: DBuf):
         _ck(self._lib, self._lib.gpue_page_decode_bshuf_lz4_i32(self._h, page._h,
                                                                 n_values, out._h))
+
+    def hash_agg_push(self, at, keys: DBuf, vals: DBuf, n, cnts: DBuf = None,
+                      update_only=0, miss_mask: DBuf = None, want_hits=False):
+        hv = c_u64()
+        _ck(self._lib, self._lib.gpue_hash_agg_push_u64(
+            self._h, at, keys._h, vals._h if vals else None, cnts._h if cnts else None, n,
+            update_only, miss_mask._h if miss_mask else None,
+            ctypes.byref(hv) if want_hits else None))
+        return hv.value if want_hits else None
+
+    def hash_agg_probe_hits(self, at, keys: DBuf, n) -> int:
+        hv = c_u64()
+        _ck(self._lib, self._lib.gpue_hash_agg_probe_hits_u64(self._h, at, keys._h, n,
+                                                              ctypes.byref(hv)))
+        return hv.value
+
+    def hash_agg_emit(self, at, out_keys: DBuf, out_sums: DBuf, max_out,
+                      out_counts: DBuf = None) -> int:
+        g = c_u64()
+        _ck(self._lib, self._lib.gpue_hash_agg_emit_u64(
+            self._h, at, out_keys._h, out_sums._h, out_counts._h if out_counts else None,
+            max_out, ctypes.byref(g)))
+        return g.value
 
     def dict_decode_binary(self, dict_bytes: DBuf, dict_offsets: DBuf, codes: DBuf, n,
                            out_bytes: DBuf = None, out_offsets: DBuf = None) -> int:

@@ -257,6 +257,174 @@ class AggregateSinkOperator(Operator):
         return self._result(self._e, self._state)
 
 
+class StreamingAggOperator(Operator):
+    """AUTO-mode streaming pre-aggregation
+    (aggregate_streaming_sink_operator.cpp:224-310): per chunk, the state
+    machine INIT_PREAGG -> ADJUST -> {PASS_THROUGH | PREAGG |
+    SELECTIVE_PREAGG} decides between aggregating into the local table,
+    streaming the chunk through to the downstream (merge) aggregate, or the
+    selective mix (aggregate rows whose group already exists, stream the
+    rest). Thresholds are the reference's (aggregator.h:175-178,
+    aggregator.cpp:150-160): LowReduction 0.2, HighReduction 0.9,
+    StableLimit 5 consecutive chunks to leave ADJUST; hit count =
+    build_hash_map_with_selection's existing-group rows. At finishing the
+    local table streams out as pre-aggregated partials (cnt column set), the
+    merge_batch form (aggregate.h:158-168).
+
+    Chunks: {"n", "keys": DBuf u64, "vals": DBuf i64, "cnts": DBuf|None}.
+    """
+
+    LOW_REDUCTION = 0.2     # aggregator.h:175
+    HIGH_REDUCTION = 0.9    # aggregator.h:176
+    STABLE_LIMIT = 5        # aggregator.h:178
+
+    def __init__(self, engine, capacity=1 << 20, max_groups=1 << 20):
+        super().__init__()
+        self._e = engine
+        self._at = engine.agg_table_create(capacity)
+        engine.agg_table_reset(self._at)
+        self._max_groups = max_groups
+        self._out = []
+        self._state = "INIT_PREAGG"
+        self._counts = {"pass": 0, "preagg": 0, "selective": 0, "adjust": 0}
+        self._drained = False
+
+    def need_input(self):
+        return not self._finishing and not self._out
+
+    def has_output(self):
+        if self._finishing and not self._drained:
+            self._drain()
+        return bool(self._out)
+
+    def pull_chunk(self):
+        return self._out.pop(0)
+
+    def _preagg(self, c):
+        self._e.hash_agg_push(self._at, c["keys"], c["vals"], c["n"], cnts=c.get("cnts"))
+        for b in (c["keys"], c["vals"]):
+            b.free()
+
+    def _stream(self, c):
+        self._out.append(c)
+
+    def _selective(self, c):
+        import numpy as np
+        n = c["n"]
+        mask = self._e.alloc(n)
+        self._e.hash_agg_push(self._at, c["keys"], c["vals"], n, cnts=c.get("cnts"),
+                              update_only=1, miss_mask=mask)
+        m = mask.d2h(np.uint8, n)
+        miss = np.flatnonzero(m).astype(np.uint32)
+        mask.free()
+        if len(miss):
+            idx = self._e.alloc(miss.nbytes)
+            idx.h2d(miss)
+            ks = self._e.alloc(len(miss) * 8)
+            vs = self._e.alloc(len(miss) * 8)
+            self._e.gather_u64(c["keys"], idx, len(miss), ks)
+            self._e.gather_u64(c["vals"], idx, len(miss), vs)
+            idx.free()
+            self._out.append({"n": len(miss), "keys": ks, "vals": vs, "cnts": None})
+        for b in (c["keys"], c["vals"]):
+            b.free()
+
+    def push_chunk(self, c):
+        n = c["n"]
+        if self._state == "INIT_PREAGG":
+            # first chunks always pre-aggregate; leave for ADJUST once the
+            # table stops reducing (ht expansion heuristic simplified to the
+            # reduction test on the next chunk)
+            hits = self._e.hash_agg_probe_hits(self._at, c["keys"], n)
+            if self._counts["adjust"] == 0 or hits >= self.HIGH_REDUCTION * n:
+                self._counts["adjust"] += 1
+                self._preagg(c)
+                return
+            self._state = "ADJUST"
+        if self._state == "ADJUST":
+            hits = self._e.hash_agg_probe_hits(self._at, c["keys"], n)
+            if hits <= self.LOW_REDUCTION * n:
+                self._stream(c)
+                self._counts["pass"] += 1
+                self._counts["preagg"] = self._counts["selective"] = 0
+                if self._counts["pass"] >= self.STABLE_LIMIT:
+                    self._state = "PASS_THROUGH"
+            elif hits >= self.HIGH_REDUCTION * n:
+                self._preagg(c)
+                self._counts["preagg"] += 1
+                self._counts["pass"] = self._counts["selective"] = 0
+                if self._counts["preagg"] >= self.STABLE_LIMIT:
+                    self._state = "PREAGG"
+            else:
+                self._selective(c)
+                self._counts["selective"] += 1
+                self._counts["pass"] = self._counts["preagg"] = 0
+                if self._counts["selective"] >= self.STABLE_LIMIT:
+                    self._state = "SELECTIVE_PREAGG"
+        elif self._state == "PASS_THROUGH":
+            self._stream(c)
+        elif self._state == "SELECTIVE_PREAGG":
+            self._selective(c)
+        else:
+            self._preagg(c)
+
+    def _drain(self):
+        """Finishing: the local table streams out as pre-agg partials."""
+        self._drained = True
+        ok = self._e.alloc(self._max_groups * 8)
+        os_ = self._e.alloc(self._max_groups * 8)
+        oc = self._e.alloc(self._max_groups * 8)
+        g = self._e.hash_agg_emit(self._at, ok, os_, self._max_groups, out_counts=oc)
+        if g:
+            self._out.append({"n": g, "keys": ok, "vals": os_, "cnts": oc})
+        else:
+            for b in (ok, os_, oc):
+                b.free()
+
+    def close(self):
+        self._e.agg_table_destroy(self._at)
+        super().close()
+
+
+class FinalAggSink(Operator):
+    """The downstream (merge) aggregate: accepts raw rows AND pre-agg
+    partials (cnts set) into one table — merge_batch semantics."""
+
+    def __init__(self, engine, capacity=1 << 20, max_groups=1 << 20):
+        super().__init__()
+        self._e = engine
+        self._at = engine.agg_table_create(capacity)
+        engine.agg_table_reset(self._at)
+        self._max_groups = max_groups
+
+    def need_input(self):
+        return not self._finishing
+
+    def has_output(self):
+        return False
+
+    def push_chunk(self, c):
+        self._e.hash_agg_push(self._at, c["keys"], c["vals"], c["n"], cnts=c.get("cnts"))
+        for b in (c["keys"], c["vals"], c.get("cnts")):
+            if b is not None:
+                b.free()
+
+    def result(self):
+        import numpy as np
+        ok = self._e.alloc(self._max_groups * 8)
+        os_ = self._e.alloc(self._max_groups * 8)
+        oc = self._e.alloc(self._max_groups * 8)
+        g = self._e.hash_agg_emit(self._at, ok, os_, self._max_groups, out_counts=oc)
+        keys = ok.d2h(np.uint64, g)
+        sums = os_.d2h(np.int64, g)
+        cnts = oc.d2h(np.int64, g)
+        for b in (ok, os_, oc):
+            b.free()
+        self._e.agg_table_destroy(self._at)
+        order = np.argsort(keys)
+        return keys[order], sums[order], cnts[order]
+
+
 def q1_operator_pipeline(engine, seed, total_rows, year=1993,
                          chunk_rows=DEFAULT_CHUNK_ROWS, dim_chunk_rows=1000):
     """Config-2's plan as the reference would run it — dim source -> build

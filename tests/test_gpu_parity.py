@@ -1040,3 +1040,48 @@ def test_dict_decode_and_varchar_join_chain(engine):
     for b in (d_page, d_codes, d_db, d_do, d_ob, d_oo, d_bb, d_bo):
         b.free()
     t.destroy()
+
+
+def test_streaming_preagg_auto_policy(engine):
+    """AUTO streaming pre-agg (aggregate_streaming_sink_operator.cpp:224-310
+    restated in StreamingAggOperator): a low-cardinality phase drives the
+    operator to PREAGG, a high-cardinality phase drives it to PASS_THROUGH,
+    and the final merged (keys, sums, counts) equal the oracle's one-shot
+    aggregate over the whole input regardless of the route each chunk took."""
+    from starrocks_amd.pipeline import (StreamingAggOperator, FinalAggSink,
+                                        PipelineDriver, ChunkSourceOperator)
+    rng = np.random.default_rng(61)
+    chunks = []
+    # phase 1: 8 chunks over 100 groups (reduction ~1.0 -> PREAGG)
+    for _ in range(8):
+        chunks.append((rng.integers(1, 101, 40_000).astype(np.uint64),
+                       rng.integers(0, 1000, 40_000).astype(np.int64)))
+    # phase 2: 10 chunks of mostly-unique keys (reduction ~0 -> PASS_THROUGH)
+    for i in range(10):
+        chunks.append((rng.integers(10_000 + i * 10**7, 10_000 + (i + 1) * 10**7,
+                                    40_000).astype(np.uint64),
+                       rng.integers(0, 1000, 40_000).astype(np.int64)))
+
+    all_keys = np.concatenate([k for k, _ in chunks])
+    all_vals = np.concatenate([v for _, v in chunks])
+    ek, es, ec = orc.hash_agg_sum(all_keys, all_vals)
+    order = np.argsort(ek)
+    ek, es, ec = ek[order], es[order], ec[order]
+
+    def source(row_start, n):
+        k, v = chunks[row_start]
+        kb = engine.alloc(k.nbytes)
+        kb.h2d(k)
+        vb = engine.alloc(v.nbytes)
+        vb.h2d(v)
+        return {"n": len(k), "keys": kb, "vals": vb, "cnts": None}
+
+    sagg = StreamingAggOperator(engine)
+    sink = FinalAggSink(engine)
+    gk, gs, gc = PipelineDriver(
+        [ChunkSourceOperator(source, len(chunks), 1), sagg, sink]).process()
+    assert sagg._state in ("PASS_THROUGH", "ADJUST")
+    assert sagg._counts["pass"] >= 1  # phase 2 streamed
+    assert np.array_equal(gk, ek)
+    assert np.array_equal(gs, es)
+    assert np.array_equal(gc, ec)
