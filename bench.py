@@ -254,12 +254,30 @@ def main():
         eng.gen_lineorder_q1(SEED, row_start, rows, *cols)
         acc, acc_t = make_acc(2)
 
-        def kernel_only():
-            eng.q1_join_sum_async(tables["dates"], cols[0], cols[1], cols[2], rows, acc)
+        if world == 1 and os.environ.get("GPUE_Q1_NO_ACCUM") != "1":
+            # accumulating step (gpue_q1_join_sum_accum): the persistent acc
+            # is never reset — each step is one kernel launch, the per-step
+            # value is the host-side difference of successive readbacks
+            acc.h2d(np.zeros(2, np.int64))
+            prev = np.zeros(2, np.int64)
 
-        def step():
-            kernel_only()
-            return acc.d2h(np.int64, 2)
+            def kernel_only():
+                eng.q1_join_sum_accum(tables["dates"], cols[0], cols[1], cols[2], rows, acc)
+
+            def step():
+                nonlocal prev
+                kernel_only()
+                cur = acc.d2h(np.int64, 2)
+                out = cur - prev
+                prev = cur
+                return out
+        else:
+            def kernel_only():
+                eng.q1_join_sum_async(tables["dates"], cols[0], cols[1], cols[2], rows, acc)
+
+            def step():
+                kernel_only()
+                return acc.d2h(np.int64, 2)
     elif wl == "q21":
         cols = [eng.alloc(rows * 4) for _ in range(4)]
         eng.gen_lineorder_q21(SEED, row_start, rows, *cols)

@@ -203,6 +203,14 @@ int gpue_join_probe_emit_varchar_nulls(gpue_session* s, gpue_join_table* t, gpue
                                        gpue_dbuf* poffsets, gpue_dbuf* probe_nulls,
                                        uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
                                        gpue_dbuf* out_build_idx, uint64_t* match_count);
+/* Accumulating config-2 step: NO per-step accumulator reset — the caller
+ * keeps a persistent i64[2] acc and differences successive readbacks.
+ * Removes one launch from the 2-op step (SF10 is launch-cost-sensitive). */
+int gpue_q1_join_sum_accum(gpue_session* s, gpue_join_table* dates, gpue_dbuf* od,
+                           gpue_dbuf* ep, gpue_dbuf* dc, uint64_t n, gpue_dbuf* acc);
+
+typedef struct gpue_agg_table gpue_agg_table; /* defined with the agg section below */
+
 /* ---- streaming pre-aggregation building blocks ----
  * The AUTO-mode streaming agg (aggregate_streaming_sink_operator.cpp:224-310;
  * thresholds aggregator.h:175-178: LowReduction 0.2, HighReduction 0.9,
@@ -334,7 +342,6 @@ int gpue_q3_order_bits(gpue_session* s, gpue_dbuf* o_custkey, gpue_dbuf* o_order
                        gpue_dbuf* order_bits);
 /* persistent aggregate table (per-query hash map analog, reused across
  * passes so repeated executions pay reset, not allocation) */
-typedef struct gpue_agg_table gpue_agg_table;
 int gpue_agg_table_create(gpue_session* s, uint64_t capacity, gpue_agg_table** out);
 void gpue_agg_table_destroy(gpue_agg_table* t);
 /* fused lineitem filter + orders semi-probe + hash aggregate */

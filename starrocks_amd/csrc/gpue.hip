@@ -2310,6 +2310,22 @@ static void launch_q1(gpue_session* s, gpue_join_table* dates, const int32_t* od
     }
 }
 
+// Accumulating form: NO per-step memset — the caller keeps a persistent acc
+// and differences successive readbacks on the host. Removes one launch from
+// the 2-op step (the SF10 line is launch-overhead-sensitive, §4b).
+extern "C" int gpue_q1_join_sum_accum(gpue_session* s, gpue_join_table* dates, gpue_dbuf* od,
+                                      gpue_dbuf* ep, gpue_dbuf* dc, uint64_t n,
+                                      gpue_dbuf* acc);
+int gpue_q1_join_sum_accum(gpue_session* s, gpue_join_table* dates, gpue_dbuf* od,
+                           gpue_dbuf* ep, gpue_dbuf* dc, uint64_t n, gpue_dbuf* acc) {
+    ARG_CHECK(s && dates && od && ep && dc && acc && acc->bytes >= 16);
+    launch_q1(s, dates, (const int32_t*)od->ptr, (const int32_t*)ep->ptr,
+              (const int32_t*)dc->ptr, n, (unsigned long long*)acc->ptr,
+              (unsigned long long*)acc->ptr + 1);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
 int gpue_q1_join_sum(gpue_session* s, gpue_join_table* dates, gpue_dbuf* od, gpue_dbuf* ep,
                      gpue_dbuf* dc, uint64_t n, int64_t* sum_out, uint64_t* match_count_out) {
     ARG_CHECK(s && dates && od && ep && dc && sum_out && match_count_out);

@@ -87,6 +87,7 @@ def _declare(lib):
                                      ctypes.POINTER(c_i64), ctypes.POINTER(c_u64)]),
         "gpue_q21_star_agg": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_q1_join_sum_async": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
+        "gpue_q1_join_sum_accum": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_dbuf_wrap": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_dbuf_ptr": (c_i32, [c_vp, ctypes.POINTER(c_vp)]),
         "gpue_q43_star_agg_async": (c_i32, [c_vp] * 11 + [c_u64, c_vp]),
@@ -410,6 +411,10 @@ class Engine:
         _ck(self._lib, self._lib.gpue_q43_star_agg_async(
             self._h, custs._h, supps._h, parts._h, dates._h, ck._h, sk._h, pk._h,
             od._h, rv._h, sc._h, n, group_sums._h))
+
+    def q1_join_sum_accum(self, dates: JoinTable, od: DBuf, ep: DBuf, dc: DBuf, n, acc: DBuf):
+        _ck(self._lib, self._lib.gpue_q1_join_sum_accum(self._h, dates._h, od._h, ep._h,
+                                                        dc._h, n, acc._h))
 
     def q1_join_sum_async(self, dates: JoinTable, od: DBuf, ep: DBuf, dc: DBuf, n, acc: DBuf):
         _ck(self._lib, self._lib.gpue_q1_join_sum_async(self._h, dates._h, od._h, ep._h,
