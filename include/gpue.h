@@ -133,6 +133,13 @@ int gpue_join_build_bucket_chained_u32(gpue_session* s, gpue_dbuf* keys /*u32, 1
  * in first[], linear probing, same-key chains via next[]. */
 int gpue_join_build_linear_chained_u32(gpue_session* s, gpue_dbuf* keys /*u32, 1-based*/,
                                        uint64_t row_count, gpue_join_table** out);
+/* 8-byte (BIGINT) key bucket-chained variant — JoinKeyHash<8>
+ * (join_hash_map_helper.h:46-55), u64 build-key compares along chains. */
+int gpue_join_build_bucket_chained_u64(gpue_session* s, gpue_dbuf* keys /*u64, 1-based*/,
+                                       uint64_t row_count, gpue_join_table** out);
+int gpue_join_probe_emit_mode_u64(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
+                                  uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
+                                  gpue_dbuf* out_build_idx, uint64_t* match_count);
 /* Row-index variant: first/next chain structure exactly as the reference
  * builds it (chain order under duplicate keys is scatter-order, which on GPU
  * is nondeterministic — the emitted match multiset is identical). */

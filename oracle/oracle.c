@@ -176,6 +176,49 @@ void orc_bucket_chained_lookup_u32(const uint32_t* probe_keys, uint32_t probe_ro
         heads[i] = first[orc_join_hash_u32(probe_keys[i], log_bucket_size)];
 }
 
+/* 8-byte (BIGINT) key variants: JoinKeyHash<8> (join_hash_map_helper.h:46-54,
+ * multiplier 11400714819323198485 — pinned by the stats64 KATs), same
+ * chained structure. */
+void orc_bucket_chained_build_u64(const uint64_t* keys, uint32_t row_count,
+                                  uint32_t* first, uint32_t* next,
+                                  uint32_t bucket_size, uint32_t log_bucket_size) {
+    (void)bucket_size;
+    const uint32_t num_rows = row_count + 1;
+    for (uint32_t i = 1; i < num_rows; i++)
+        next[i] = orc_join_hash_u64(keys[i], log_bucket_size);
+    for (uint32_t i = 1; i < num_rows; i++) {
+        uint32_t b = next[i];
+        next[i] = first[b];
+        first[b] = i;
+    }
+}
+
+void orc_bucket_chained_lookup_u64(const uint64_t* probe_keys, uint32_t probe_rows,
+                                   const uint32_t* first, uint32_t bucket_size,
+                                   uint32_t log_bucket_size, uint32_t* heads) {
+    (void)bucket_size;
+    for (uint32_t i = 0; i < probe_rows; i++)
+        heads[i] = first[orc_join_hash_u64(probe_keys[i], log_bucket_size)];
+}
+
+uint64_t orc_probe_emit_u64(const uint64_t* build_keys, const uint32_t* next,
+                            const uint64_t* probe_keys, const uint32_t* heads,
+                            uint32_t probe_rows, int collision_free,
+                            uint32_t* out_probe_idx, uint32_t* out_build_idx) {
+    uint64_t m = 0;
+    for (uint32_t i = 0; i < probe_rows; i++) {
+        for (uint32_t b = heads[i]; b != 0; b = next[b]) {
+            if (collision_free || build_keys[b] == probe_keys[i]) {
+                out_probe_idx[m] = i;
+                out_build_idx[m] = b;
+                m++;
+                if (collision_free) break;
+            }
+        }
+    }
+    return m;
+}
+
 /* SERIALIZED_VARCHAR / Slice keys (the selector's last constructor branch,
  * join_hash_table.cpp:215-217): JoinKeyHash<Slice> = crc_hash_32(bytes, len,
  * 0x811C9DC5) & (bucket_size-1) (join_hash_map_helper.h:57-64); chains walk

@@ -47,6 +47,16 @@ void orc_bucket_chained_lookup_u32(const uint32_t* probe_keys, uint32_t probe_ro
                                    const uint32_t* first, uint32_t bucket_size,
                                    uint32_t log_bucket_size, uint32_t* heads);
 
+void orc_bucket_chained_build_u64(const uint64_t* keys, uint32_t row_count,
+                                  uint32_t* first, uint32_t* next,
+                                  uint32_t bucket_size, uint32_t log_bucket_size);
+void orc_bucket_chained_lookup_u64(const uint64_t* probe_keys, uint32_t probe_rows,
+                                   const uint32_t* first, uint32_t bucket_size,
+                                   uint32_t log_bucket_size, uint32_t* heads);
+uint64_t orc_probe_emit_u64(const uint64_t* build_keys, const uint32_t* next,
+                            const uint64_t* probe_keys, const uint32_t* heads,
+                            uint32_t probe_rows, int collision_free,
+                            uint32_t* out_probe_idx, uint32_t* out_build_idx);
 void orc_slice_build_u32(const uint8_t* bytes, const uint32_t* offsets, uint32_t row_count,
                          uint32_t* first, uint32_t* next, uint32_t bucket_size,
                          uint32_t log_bucket_size);

@@ -71,6 +71,10 @@ def _decl(lib):
     lib.orc_linear_chained_lookup_u32.argtypes = [c_vp, c_vp, u, c_vp, u, u, c_vp]
     lib.orc_range_direct_build_i32.argtypes = [c_vp, u, c_i64, c_vp, c_vp]
     lib.orc_range_direct_lookup_i32.argtypes = [c_vp, c_u64, c_i64, c_i64, c_vp, c_vp]
+    lib.orc_bucket_chained_build_u64.argtypes = [c_vp, u, c_vp, c_vp, u, u]
+    lib.orc_bucket_chained_lookup_u64.argtypes = [c_vp, u, c_vp, u, u, c_vp]
+    lib.orc_probe_emit_u64.restype = c_u64
+    lib.orc_probe_emit_u64.argtypes = [c_vp, c_vp, c_vp, c_vp, u, c_i32, c_vp, c_vp]
     lib.orc_slice_build_u32.argtypes = [c_vp, c_vp, u, c_vp, c_vp, u, u]
     lib.orc_slice_probe_emit.restype = c_u64
     lib.orc_slice_probe_emit.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, c_vp, u, c_vp, c_vp]
@@ -532,3 +536,23 @@ def dict_decode_binary(dict_bytes, dict_offsets, codes):
                                           _p(np.ascontiguousarray(codes, np.int32)), n,
                                           _p(out_bytes), _p(out_offsets))
     return out_bytes[:total], out_offsets
+
+
+def bucket_chained_join_u64(build_keys_1based, probe_keys, max_out):
+    """8-byte-key chained join (JoinKeyHash<8>): emit all match pairs."""
+    lib = load()
+    row_count = len(build_keys_1based) - 1
+    bucket_size = lib.orc_calc_bucket_size(row_count + 1)
+    log = int(bucket_size - 1).bit_length()
+    first = np.zeros(bucket_size, np.uint32)
+    nxt = np.zeros(row_count + 1, np.uint32)
+    lib.orc_bucket_chained_build_u64(_p(build_keys_1based), row_count, _p(first), _p(nxt),
+                                     bucket_size, log)
+    heads = np.zeros(len(probe_keys), np.uint32)
+    lib.orc_bucket_chained_lookup_u64(_p(probe_keys), len(probe_keys), _p(first),
+                                      bucket_size, log, _p(heads))
+    op = np.empty(max_out, np.uint32)
+    ob = np.empty(max_out, np.uint32)
+    m = lib.orc_probe_emit_u64(_p(build_keys_1based), _p(nxt), _p(probe_keys), _p(heads),
+                               len(probe_keys), 0, _p(op), _p(ob))
+    return op[:m], ob[:m]

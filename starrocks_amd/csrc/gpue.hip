@@ -134,13 +134,14 @@ struct gpue_join_table {
     // method discriminator — the GPU analog of JoinHashMapSelector's choice
     // (reference join_hash_table.cpp:164-344)
     enum Kind { PAYLOAD = 0, RANGE_DIRECT = 1, BUCKET_CHAINED = 2,
-                LINEAR_CHAINED = 3, VARCHAR = 4 } kind = PAYLOAD;
+                LINEAR_CHAINED = 3, VARCHAR = 4, BUCKET_CHAINED64 = 5 } kind = PAYLOAD;
     uint32_t log_bucket_size = 0;
     uint32_t* build_keys = nullptr; // chained methods keep the build keys for the
                                     // probe-side equality check (1-based, row 0 sentinel)
     uint8_t* key_bytes = nullptr;   // VARCHAR: BinaryColumn bytes + uint32 offsets
     uint32_t* key_offsets = nullptr;
     uint8_t* key_nulls = nullptr;   // VARCHAR nullable: is_nulls (1-based), or null
+    uint64_t* build_keys64 = nullptr; // BUCKET_CHAINED64: 8-byte build keys
 };
 
 int gpue_device_count(int* out) {
@@ -1034,6 +1035,7 @@ void gpue_join_table_destroy(gpue_join_table* t) {
     if (t->key_bytes) (void)hipFree(t->key_bytes);
     if (t->key_offsets) (void)hipFree(t->key_offsets);
     if (t->key_nulls) (void)hipFree(t->key_nulls);
+    if (t->build_keys64) (void)hipFree(t->build_keys64);
     delete t;
 }
 
@@ -1070,6 +1072,79 @@ __global__ void k_build_bucket_chained(const uint32_t* __restrict__ keys, uint64
     for (uint64_t i = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i <= row_count;
          i += stride)
         next[i] = atomicExch(&first[join_hash_u32(keys[i], log_bucket_size)], (uint32_t)i);
+}
+
+// 8-byte (BIGINT) key variants — JoinKeyHash<8> (join_hash_map_helper.h:
+// 46-55, multiplier 11400714819323198485; pinned by the oracle's stats64
+// KATs), same chained structure with u64 build-key compares.
+__device__ static inline uint32_t join_hash_u64_dev(uint64_t v, uint32_t num_log_buckets) {
+    v ^= v >> (64 - num_log_buckets);
+    return (uint32_t)((v * 11400714819323198485ull) >> (64 - num_log_buckets));
+}
+
+__global__ void k_build_bucket_chained64(const uint64_t* __restrict__ keys,
+                                         uint64_t row_count, uint32_t log_bucket_size,
+                                         uint32_t* __restrict__ first,
+                                         uint32_t* __restrict__ next) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i <= row_count;
+         i += stride)
+        next[i] = atomicExch(&first[join_hash_u64_dev(keys[i], log_bucket_size)],
+                             (uint32_t)i);
+}
+
+__global__ void k_probe_count_bc64(const uint64_t* __restrict__ probe_keys, uint64_t n,
+                                   uint32_t log_bucket_size,
+                                   const uint32_t* __restrict__ first,
+                                   const uint32_t* __restrict__ next,
+                                   const uint64_t* __restrict__ build_keys, int mode,
+                                   uint32_t* __restrict__ row_counts) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint64_t k = probe_keys[i];
+        uint32_t b = first[join_hash_u64_dev(k, log_bucket_size)];
+        uint32_t c = 0;
+        while (b != 0) {
+            c += (build_keys[b] == k);
+            b = next[b];
+        }
+        row_counts[i] = join_mode_count(c, mode);
+    }
+}
+
+__global__ void k_probe_emit_bc64(const uint64_t* __restrict__ probe_keys, uint64_t n,
+                                  uint32_t log_bucket_size,
+                                  const uint32_t* __restrict__ first,
+                                  const uint32_t* __restrict__ next,
+                                  const uint64_t* __restrict__ build_keys, int mode,
+                                  const uint32_t* __restrict__ row_counts,
+                                  const uint64_t* __restrict__ row_offsets,
+                                  uint32_t* __restrict__ out_probe,
+                                  uint32_t* __restrict__ out_build) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        if (row_counts[i] == 0) continue;
+        uint64_t pos = row_offsets[i];
+        uint64_t k = probe_keys[i];
+        uint32_t b = first[join_hash_u64_dev(k, log_bucket_size)];
+        uint32_t c = 0;
+        while (b != 0) {
+            if (build_keys[b] == k) {
+                if (mode == 0 || mode == 3 || (mode == 1 && c == 0)) {
+                    out_probe[pos] = (uint32_t)i;
+                    out_build[pos] = b;
+                    pos++;
+                }
+                c++;
+                if (mode == 1 || mode == 2) break;
+            }
+            b = next[b];
+        }
+        if (c == 0 && (mode == 2 || mode == 3)) {
+            out_probe[pos] = (uint32_t)i;
+            out_build[pos] = 0;
+        }
+    }
 }
 
 __global__ void k_build_bucket_chained_nulls(const uint32_t* __restrict__ keys,
@@ -1169,6 +1244,89 @@ int gpue_join_build_bucket_chained_u32(gpue_session* s, gpue_dbuf* keys, uint64_
                        t->next);
     HIP_CHECK(hipGetLastError());
     *out = t;
+    return GPUE_OK;
+}
+
+// forward decls: the count->scan->emit helpers are defined with the probe
+// section further down this file
+__global__ void k_block_sums_u32(const uint32_t* __restrict__ counts, uint64_t n,
+                                 uint64_t tile, uint64_t* __restrict__ block_sums);
+__global__ void k_scan_small(uint64_t* __restrict__ sums, uint64_t n_blocks);
+__global__ void k_scan_offsets(const uint32_t* __restrict__ counts, uint64_t n, uint64_t tile,
+                               const uint64_t* __restrict__ block_offsets,
+                               uint64_t* __restrict__ offsets);
+
+extern "C" {
+int gpue_join_build_bucket_chained_u64(gpue_session* s, gpue_dbuf* keys /*u64, 1-based*/,
+                                       uint64_t row_count, gpue_join_table** out);
+int gpue_join_probe_emit_mode_u64(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
+                                  uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
+                                  gpue_dbuf* out_build_idx, uint64_t* match_count);
+}
+
+int gpue_join_build_bucket_chained_u64(gpue_session* s, gpue_dbuf* keys, uint64_t row_count,
+                                       gpue_join_table** out) {
+    ARG_CHECK(s && keys && out && row_count > 0 && row_count + 1 < (1ull << 31));
+    ARG_CHECK(keys->bytes >= (row_count + 1) * 8);
+    gpue_join_table* t = new gpue_join_table();
+    t->s = s;
+    t->kind = gpue_join_table::BUCKET_CHAINED64;
+    t->row_count = row_count;
+    t->bucket_size = calc_bucket_size((uint32_t)(row_count + 1));
+    t->log_bucket_size = (uint32_t)__builtin_ctzll(t->bucket_size);
+    HIP_CHECK(hipMalloc(&t->first, t->bucket_size * sizeof(uint32_t)));
+    HIP_CHECK(hipMalloc(&t->next, (row_count + 1) * sizeof(uint32_t)));
+    HIP_CHECK(hipMalloc(&t->build_keys64, (row_count + 1) * sizeof(uint64_t)));
+    HIP_CHECK(hipMemsetAsync(t->first, 0, t->bucket_size * sizeof(uint32_t), s->stream));
+    HIP_CHECK(hipMemsetAsync(t->next, 0, (row_count + 1) * sizeof(uint32_t), s->stream));
+    HIP_CHECK(hipMemcpyAsync(t->build_keys64, keys->ptr, (row_count + 1) * sizeof(uint64_t),
+                             hipMemcpyDeviceToDevice, s->stream));
+    hipLaunchKernelGGL(k_build_bucket_chained64, dim3(grid_for(row_count)), dim3(BLOCK), 0,
+                       s->stream, t->build_keys64, row_count, t->log_bucket_size, t->first,
+                       t->next);
+    HIP_CHECK(hipGetLastError());
+    *out = t;
+    return GPUE_OK;
+}
+
+int gpue_join_probe_emit_mode_u64(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
+                                  uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
+                                  gpue_dbuf* out_build_idx, uint64_t* match_count) {
+    ARG_CHECK(s && t && probe_keys && match_count);
+    ARG_CHECK(t->kind == gpue_join_table::BUCKET_CHAINED64);
+    ARG_CHECK(mode >= 0 && mode <= 3);
+    ARG_CHECK(probe_keys->bytes >= n_rows * 8);
+    uint32_t nb = grid_for(n_rows);
+    uint64_t tile = (n_rows + nb - 1) / nb;
+    uint32_t* d_counts = nullptr;
+    uint64_t* d_bsums = nullptr;
+    uint64_t* d_offsets = nullptr;
+    HIP_CHECK(hipMalloc(&d_counts, n_rows * 4));
+    HIP_CHECK(hipMalloc(&d_bsums, (nb + 1) * 8));
+    hipLaunchKernelGGL(k_probe_count_bc64, dim3(nb), dim3(BLOCK), 0, s->stream,
+                       (const uint64_t*)probe_keys->ptr, n_rows, t->log_bucket_size,
+                       t->first, t->next, t->build_keys64, mode, d_counts);
+    hipLaunchKernelGGL(k_block_sums_u32, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts,
+                       n_rows, tile, d_bsums);
+    hipLaunchKernelGGL(k_scan_small, dim3(1), dim3(1), 0, s->stream, d_bsums, nb);
+    uint64_t total = 0;
+    HIP_CHECK(hipMemcpyAsync(&total, d_bsums + nb, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    *match_count = total;
+    if (out_probe_idx && out_build_idx && total > 0) {
+        ARG_CHECK(out_probe_idx->bytes >= total * 4 && out_build_idx->bytes >= total * 4);
+        HIP_CHECK(hipMalloc(&d_offsets, n_rows * 8));
+        hipLaunchKernelGGL(k_scan_offsets, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts,
+                           n_rows, tile, d_bsums, d_offsets);
+        hipLaunchKernelGGL(k_probe_emit_bc64, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const uint64_t*)probe_keys->ptr, n_rows, t->log_bucket_size,
+                           t->first, t->next, t->build_keys64, mode, d_counts, d_offsets,
+                           (uint32_t*)out_probe_idx->ptr, (uint32_t*)out_build_idx->ptr);
+        HIP_CHECK(hipStreamSynchronize(s->stream));
+        (void)hipFree(d_offsets);
+    }
+    (void)hipFree(d_counts);
+    (void)hipFree(d_bsums);
     return GPUE_OK;
 }
 

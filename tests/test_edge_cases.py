@@ -170,3 +170,21 @@ def test_slice_join_nulls_vs_brute():
             else:
                 expect += [(i, j) for j in hits] if hits else [(i, 0)]
         assert got == sorted(expect), mode
+
+
+def test_bucket_chained_u64_oracle_vs_brute():
+    """8-byte-key chained join (JoinKeyHash<8>, join_hash_map_helper.h:46-55)
+    vs brute force over keys spanning the full u64 range."""
+    rng = np.random.default_rng(67)
+    bkeys = np.concatenate([[0], rng.integers(0, 2**63, 2000, dtype=np.uint64) * 2 + 1,
+                            rng.integers(0, 300, 1000, dtype=np.uint64)]).astype(np.uint64)
+    probe = np.concatenate([rng.integers(0, 300, 2000, dtype=np.uint64),
+                            bkeys[1:50]]).astype(np.uint64)
+    op, ob = orc.bucket_chained_join_u64(bkeys, probe, 10_000_000)
+    index = {}
+    for j in range(1, len(bkeys)):
+        index.setdefault(int(bkeys[j]), []).append(j)
+    expect = sorted((i, j) for i, k in enumerate(probe.tolist())
+                    for j in index.get(int(k), []))
+    assert sorted(zip(op.tolist(), ob.tolist())) == expect
+    assert len(expect) > 0
