@@ -1117,3 +1117,26 @@ def test_bucket_chained_u64_join_parity(engine):
     for b in (kb, pb, op_buf, ob_buf):
         b.free()
     t.destroy()
+
+
+def test_distinct_count_via_agg_claims(engine):
+    """COUNT(DISTINCT col): agg-table insert where distinct = rows - existing
+    hits (each first-insertion claims exactly once under concurrent CAS —
+    the agg_hash_set semantics of the reference's distinct aggregate)."""
+    rng = np.random.default_rng(73)
+    n = 2_000_000
+    keys = rng.integers(1, 300_000, n).astype(np.uint64)
+    expect = len(np.unique(keys))
+    at = engine.agg_table_create(1 << 20)
+    engine.agg_table_reset(at)
+    kb = engine.alloc(keys.nbytes)
+    kb.h2d(keys)
+    hits = engine.hash_agg_push(at, kb, None, n, want_hits=True)
+    assert n - hits == expect
+    # and the emitted group count agrees
+    ok, os_ = engine.alloc(expect * 8), engine.alloc(expect * 8)
+    g = engine.hash_agg_emit(at, ok, os_, expect)
+    assert g == expect
+    for b in (kb, ok, os_):
+        b.free()
+    engine.agg_table_destroy(at)
