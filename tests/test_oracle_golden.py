@@ -276,3 +276,14 @@ def test_sbf_filter_properties():
 
 def sbf_all_pass(keys, directory, log):
     return bool(orc.sbf_test(keys, directory, log).all())
+
+
+def test_golden_fixture_phmap_mix8():
+    """Committed phmap_mix<8> vectors (generated from the reference's own
+    headers via oracle/_ref) — the GPU box pins the SimdBlockFilter insert
+    hash against these without /root/reference."""
+    with open(os.path.join(GOLDEN, "hash_kats.json")) as f:
+        kats = json.load(f)
+    lib = orc.load()
+    for item in kats["phmap_mix8"]:
+        assert lib.orc_phmap_mix8(item["input"]) == item["expect"]
