@@ -259,6 +259,10 @@ int gpue_sbf_build_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n,
                        int32_t log_num_buckets, gpue_dbuf* directory);
 int gpue_sbf_test_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, gpue_dbuf* directory,
                       int32_t log_num_buckets, gpue_dbuf* out);
+/* RIGHT SEMI (anti=0) / RIGHT ANTI (anti=1) over Slice keys. */
+int gpue_join_probe_right_varchar(gpue_session* s, gpue_join_table* t, gpue_dbuf* pbytes,
+                                  gpue_dbuf* poffsets, uint64_t n_rows, int anti,
+                                  gpue_dbuf* out_build_idx, uint64_t* count);
 /* RIGHT SEMI (anti=0) / RIGHT ANTI (anti=1): matched/unmatched BUILD rows. */
 int gpue_join_probe_right_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
                               uint64_t n_rows, int anti, gpue_dbuf* out_build_idx,

@@ -355,6 +355,32 @@ void orc_sbf_test_i32(const int32_t* keys, uint64_t n, const uint32_t* directory
                                             orc_phmap_mix8((uint64_t)(int64_t)keys[i]));
 }
 
+/* RIGHT SEMI/ANTI over Slice keys (join_hash_map.hpp
+ * _probe_from_ht_for_right_* semantics): mark matched build rows, output
+ * matched (semi) or unmatched (anti) build rows 1-based. */
+uint64_t orc_slice_probe_right(const uint8_t* bbytes, const uint32_t* boffsets,
+                               const uint32_t* next, uint32_t bucket_size,
+                               const uint32_t* first, uint32_t build_rows,
+                               const uint8_t* pbytes, const uint32_t* poffsets,
+                               uint32_t probe_rows, int anti, uint32_t* out_build_idx) {
+    uint8_t* matched = (uint8_t*)calloc(build_rows + 1, 1);
+    for (uint32_t i = 0; i < probe_rows; i++) {
+        uint32_t len = poffsets[i + 1] - poffsets[i];
+        uint32_t b = orc_crc_hash_32(pbytes + poffsets[i], (int32_t)len, 0x811C9DC5u) &
+                     (bucket_size - 1);
+        for (uint32_t j = first[b]; j != 0; j = next[j]) {
+            uint32_t blen = boffsets[j + 1] - boffsets[j];
+            if (blen == len && memcmp(bbytes + boffsets[j], pbytes + poffsets[i], len) == 0)
+                matched[j] = 1;
+        }
+    }
+    uint64_t m = 0;
+    for (uint32_t j = 1; j <= build_rows; j++)
+        if ((anti && !matched[j]) || (!anti && matched[j])) out_build_idx[m++] = j;
+    free(matched);
+    return m;
+}
+
 /* Nullable Slice keys (same is_nulls semantics as the fixed-size paths:
  * null build rows never enter a chain, null probe rows match nothing and
  * surface as unmatched for ANTI/OUTER). */

@@ -78,6 +78,11 @@ uint64_t orc_slice_probe_emit_nulls(const uint8_t* bbytes, const uint32_t* boffs
                                     const uint32_t* poffsets, const uint8_t* probe_nulls,
                                     uint32_t probe_rows, int mode, uint32_t* out_probe_idx,
                                     uint32_t* out_build_idx);
+uint64_t orc_slice_probe_right(const uint8_t* bbytes, const uint32_t* boffsets,
+                               const uint32_t* next, uint32_t bucket_size,
+                               const uint32_t* first, uint32_t build_rows,
+                               const uint8_t* pbytes, const uint32_t* poffsets,
+                               uint32_t probe_rows, int anti, uint32_t* out_build_idx);
 uint64_t orc_dict_decode_binary(const uint8_t* dict_bytes, const uint32_t* dict_offsets,
                                 const int32_t* codes, uint64_t n, uint8_t* out_bytes,
                                 uint32_t* out_offsets);

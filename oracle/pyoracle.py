@@ -90,6 +90,9 @@ def _decl(lib):
     lib.orc_slice_probe_emit_nulls.restype = c_u64
     lib.orc_slice_probe_emit_nulls.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, c_vp, c_vp,
                                                u, c_i32, c_vp, c_vp]
+    lib.orc_slice_probe_right.restype = c_u64
+    lib.orc_slice_probe_right.argtypes = [c_vp, c_vp, c_vp, u, c_vp, u, c_vp, c_vp, u,
+                                          c_i32, c_vp]
     lib.orc_slice_probe_emit_mode.restype = c_u64
     lib.orc_slice_probe_emit_mode.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, c_vp, u,
                                               c_i32, c_vp, c_vp]
@@ -556,3 +559,19 @@ def bucket_chained_join_u64(build_keys_1based, probe_keys, max_out):
     m = lib.orc_probe_emit_u64(_p(build_keys_1based), _p(nxt), _p(probe_keys), _p(heads),
                                len(probe_keys), 0, _p(op), _p(ob))
     return op[:m], ob[:m]
+
+
+def slice_probe_right(bbytes, boffsets, row_count, pbytes, poffsets, probe_rows, anti):
+    """RIGHT SEMI/ANTI over Slice keys: matched/unmatched build rows."""
+    lib = load()
+    bucket_size = lib.orc_calc_bucket_size(row_count + 1)
+    log = int(bucket_size).bit_length() - 1
+    first = np.zeros(bucket_size, np.uint32)
+    nxt = np.zeros(row_count + 1, np.uint32)
+    lib.orc_slice_build_u32(_p(bbytes), _p(boffsets), row_count, _p(first), _p(nxt),
+                            bucket_size, log)
+    out = np.empty(row_count, np.uint32)
+    m = lib.orc_slice_probe_right(_p(bbytes), _p(boffsets), _p(nxt), bucket_size, _p(first),
+                                  row_count, _p(pbytes), _p(poffsets), probe_rows, anti,
+                                  _p(out))
+    return np.sort(out[:m])

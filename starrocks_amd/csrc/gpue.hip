@@ -2095,6 +2095,79 @@ int gpue_sbf_test_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, gpue_dbuf* d
     return GPUE_OK;
 }
 
+// RIGHT SEMI/ANTI over Slice keys: probe pass marks matched BUILD rows
+// (atomicOr into a bitset), then compact matched/unmatched build rows.
+__global__ void k_probe_mark_vc(const uint8_t* __restrict__ pbytes,
+                                const uint32_t* __restrict__ poffsets, uint64_t n,
+                                uint32_t bucket_mask, const uint32_t* __restrict__ first,
+                                const uint32_t* __restrict__ next,
+                                const uint8_t* __restrict__ bbytes,
+                                const uint32_t* __restrict__ boffsets,
+                                uint32_t* __restrict__ matched_bits) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint32_t len = poffsets[i + 1] - poffsets[i];
+        uint32_t b = crc_hash_32_dev(pbytes + poffsets[i], len, 0x811C9DC5u) & bucket_mask;
+        for (uint32_t j = first[b]; j != 0; j = next[j])
+            if (slice_eq(bbytes + boffsets[j], boffsets[j + 1] - boffsets[j],
+                         pbytes + poffsets[i], len))
+                atomicOr(&matched_bits[j >> 5], 1u << (j & 31));
+    }
+}
+
+extern "C" int gpue_join_probe_right_varchar(gpue_session* s, gpue_join_table* t,
+                                             gpue_dbuf* pbytes, gpue_dbuf* poffsets,
+                                             uint64_t n_rows, int anti,
+                                             gpue_dbuf* out_build_idx, uint64_t* count);
+__global__ void k_right_emit_bits(const uint32_t* __restrict__ matched_bits,
+                                  uint64_t build_rows, int anti,
+                                  unsigned long long* __restrict__ cursor,
+                                  uint32_t* __restrict__ out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t j = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; j <= build_rows;
+         j += stride) {
+        uint32_t hit = (matched_bits[j >> 5] >> (j & 31)) & 1u;
+        if ((anti && !hit) || (!anti && hit)) {
+            unsigned long long pos = atomicAdd(cursor, 1ull);
+            if (out) out[pos] = (uint32_t)j;
+        }
+    }
+}
+
+int gpue_join_probe_right_varchar(gpue_session* s, gpue_join_table* t, gpue_dbuf* pbytes,
+                                  gpue_dbuf* poffsets, uint64_t n_rows, int anti,
+                                  gpue_dbuf* out_build_idx, uint64_t* count) {
+    ARG_CHECK(s && t && pbytes && poffsets && count);
+    ARG_CHECK(t->kind == gpue_join_table::VARCHAR);
+    uint64_t nwords = (t->row_count + 1 + 31) / 32;
+    uint32_t* d_bits = nullptr;
+    unsigned long long* d_cursor = nullptr;
+    HIP_CHECK(hipMalloc(&d_bits, nwords * 4));
+    HIP_CHECK(hipMalloc(&d_cursor, 8));
+    HIP_CHECK(hipMemsetAsync(d_bits, 0, nwords * 4, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_cursor, 0, 8, s->stream));
+    hipLaunchKernelGGL(k_probe_mark_vc, dim3(grid_for(n_rows)), dim3(BLOCK), 0, s->stream,
+                       (const uint8_t*)pbytes->ptr, (const uint32_t*)poffsets->ptr, n_rows,
+                       (uint32_t)(t->bucket_size - 1), t->first, t->next, t->key_bytes,
+                       t->key_offsets, d_bits);
+    if (out_build_idx) {
+        ARG_CHECK(out_build_idx->bytes >= t->row_count * 4);
+        hipLaunchKernelGGL(k_right_emit_bits, dim3(grid_for(t->row_count)), dim3(BLOCK), 0,
+                           s->stream, d_bits, t->row_count, anti, d_cursor,
+                           (uint32_t*)out_build_idx->ptr);
+    } else {
+        hipLaunchKernelGGL(k_right_emit_bits, dim3(grid_for(t->row_count)), dim3(BLOCK), 0,
+                           s->stream, d_bits, t->row_count, anti, d_cursor, nullptr);
+    }
+    unsigned long long c = 0;
+    HIP_CHECK(hipMemcpyAsync(&c, d_cursor, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_bits);
+    (void)hipFree(d_cursor);
+    *count = c;
+    return GPUE_OK;
+}
+
 // Nullable probe (lookup_init is_nulls path + per-type semantics): a null
 // probe key matches nothing — INNER/SEMI emit nothing for it, ANTI/OUTER
 // emit the unmatched (i, 0) row.

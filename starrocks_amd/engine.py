@@ -82,6 +82,8 @@ def _declare(lib):
         "gpue_join_table_first_d2h": (c_i32, [c_vp, c_vp, c_u64]),
         "gpue_join_probe_emit_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_join_probe_emit_mode_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp, ctypes.POINTER(c_u64)]),
+        "gpue_join_probe_right_varchar": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_i32,
+                                                  c_vp, ctypes.POINTER(c_u64)]),
         "gpue_join_probe_right_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_join_build_bucket_chained_nulls_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_probe_emit_nulls_i32": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp, ctypes.POINTER(c_u64)]),
@@ -390,6 +392,14 @@ class Engine:
 
     def pack_keys_2xi32(self, a: DBuf, b: DBuf, n, out: DBuf):
         _ck(self._lib, self._lib.gpue_pack_keys_2xi32(self._h, a._h, b._h, n, out._h))
+
+    def join_probe_right_varchar(self, table, pbytes, poffsets, n_rows, anti,
+                                 out_build=None) -> int:
+        cnt = c_u64()
+        ob = out_build._h if out_build else None
+        _ck(self._lib, self._lib.gpue_join_probe_right_varchar(
+            self._h, table._h, pbytes._h, poffsets._h, n_rows, anti, ob, ctypes.byref(cnt)))
+        return cnt.value
 
     def join_probe_right(self, table, probe_keys, n_rows, anti, out_build=None) -> int:
         cnt = c_u64()

@@ -1140,3 +1140,30 @@ def test_distinct_count_via_agg_claims(engine):
     for b in (kb, ok, os_):
         b.free()
     engine.agg_table_destroy(at)
+
+
+def test_varchar_probe_right_parity(engine):
+    """RIGHT SEMI/ANTI over Slice keys: matched/unmatched build-row sets
+    equal the oracle's."""
+    rng = np.random.default_rng(79)
+    n_build, n_probe, card = 30_000, 90_000, 800
+    bb, bo, brows = _varchar_cols(rng, n_build, card, one_based=True)
+    pb_, po, prows = _varchar_cols(rng, n_probe, card // 2, one_based=False)
+    d_bb, d_bo = engine.alloc(bb.nbytes), engine.alloc(bo.nbytes)
+    d_bb.h2d(bb)
+    d_bo.h2d(bo)
+    t = engine.join_build_varchar(d_bb, d_bo, n_build)
+    d_pb, d_po = engine.alloc(pb_.nbytes), engine.alloc(po.nbytes)
+    d_pb.h2d(pb_)
+    d_po.h2d(po)
+    for anti in (0, 1):
+        expect = orc.slice_probe_right(bb, bo, n_build, pb_, po, n_probe, anti)
+        ob = engine.alloc(n_build * 4)
+        cnt = engine.join_probe_right_varchar(t, d_pb, d_po, n_probe, anti, ob)
+        assert cnt == len(expect), anti
+        got = np.sort(ob.d2h(np.uint32, cnt))
+        assert np.array_equal(got, expect), anti
+        ob.free()
+    for b in (d_bb, d_bo, d_pb, d_po):
+        b.free()
+    t.destroy()
