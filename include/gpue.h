@@ -110,6 +110,17 @@ int gpue_scan_filter_i64_lt(gpue_session* s, gpue_dbuf* in, uint64_t n, int64_t 
 int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64_t theta,
                                gpue_dbuf* out, uint64_t* out_count);
 
+/* Multi-conjunct predicate evaluation with the reference's eager-prune
+ * strategy (chunk_predicate_evaluator.cpp:31-80: AND-merge per conjunct,
+ * all-true skip, all-false short-circuit, compact all columns when zeros
+ * exceed max(0.8*rows, 1024)). preds are (col_index, op, lo, hi) arrays;
+ * op: 0 EQ(lo), 1 LT(hi), 2 BETWEEN[lo,hi]. Columns compact stably in
+ * place; out_rows = survivors. */
+int gpue_eval_conjuncts_i32(gpue_session* s, gpue_dbuf** cols, int n_cols, uint64_t n_rows,
+                            const int32_t* pred_col, const int32_t* pred_op,
+                            const int32_t* pred_lo, const int32_t* pred_hi, int n_preds,
+                            uint64_t* out_rows);
+
 /* ---- hash-join build ----
  * Replaces JoinHashTable::build with the RANGE_DIRECT_MAPPING method the
  * selector takes for dense int keys (reference

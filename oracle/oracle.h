@@ -83,6 +83,10 @@ uint64_t orc_slice_probe_right(const uint8_t* bbytes, const uint32_t* boffsets,
                                const uint32_t* first, uint32_t build_rows,
                                const uint8_t* pbytes, const uint32_t* poffsets,
                                uint32_t probe_rows, int anti, uint32_t* out_build_idx);
+uint64_t orc_eval_conjuncts_i32(int32_t** cols, int n_cols, uint64_t n_rows,
+                                const int32_t* pred_col, const int32_t* pred_op,
+                                const int32_t* pred_lo, const int32_t* pred_hi,
+                                int n_preds);
 uint64_t orc_dict_decode_binary(const uint8_t* dict_bytes, const uint32_t* dict_offsets,
                                 const int32_t* codes, uint64_t n, uint8_t* out_bytes,
                                 uint32_t* out_offsets);

@@ -84,6 +84,9 @@ def _decl(lib):
     lib.orc_sbf_log_num_buckets.argtypes = [c_u64]
     lib.orc_sbf_build_i32.argtypes = [c_vp, c_u64, c_vp, c_i32]
     lib.orc_sbf_test_i32.argtypes = [c_vp, c_u64, c_vp, c_i32, c_vp]
+    lib.orc_eval_conjuncts_i32.restype = c_u64
+    lib.orc_eval_conjuncts_i32.argtypes = [c_vp, c_i32, c_u64, c_vp, c_vp, c_vp, c_vp,
+                                           c_i32]
     lib.orc_dict_decode_binary.restype = c_u64
     lib.orc_dict_decode_binary.argtypes = [c_vp, c_vp, c_vp, c_u64, c_vp, c_vp]
     lib.orc_slice_build_nulls_u32.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, u, u]
@@ -575,3 +578,18 @@ def slice_probe_right(bbytes, boffsets, row_count, pbytes, poffsets, probe_rows,
                                   row_count, _p(pbytes), _p(poffsets), probe_rows, anti,
                                   _p(out))
     return np.sort(out[:m])
+
+
+def eval_conjuncts(cols, preds):
+    """Eager-prune multi-conjunct filter (chunk_predicate_evaluator.cpp:31-80).
+    cols: list of int32 arrays (modified in place); preds: (col,op,lo,hi)."""
+    import ctypes as ct
+    n = len(cols[0])
+    ptrs = (c_vp * len(cols))(*[c.ctypes.data_as(c_vp).value for c in cols])
+    pc = np.array([p[0] for p in preds], np.int32)
+    po = np.array([p[1] for p in preds], np.int32)
+    pl = np.array([p[2] for p in preds], np.int32)
+    ph = np.array([p[3] for p in preds], np.int32)
+    m = load().orc_eval_conjuncts_i32(ptrs, len(cols), n, _p(pc), _p(po), _p(pl), _p(ph),
+                                      len(preds))
+    return m

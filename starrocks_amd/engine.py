@@ -68,6 +68,9 @@ def _declare(lib):
         "gpue_gen_lineorder_q1": (c_i32, [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp, c_vp]),
         "gpue_gen_lineorder_q21": (c_i32, [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp, c_vp, c_vp]),
         "gpue_gen_lineorder_q43": (c_i32, [c_vp, c_u64, c_u64, c_u64] + [c_vp] * 6),
+        "gpue_eval_conjuncts_i32": (c_i32, [c_vp, ctypes.POINTER(c_vp), c_i32, c_u64,
+                                            c_vp, c_vp, c_vp, c_vp, c_i32,
+                                            ctypes.POINTER(c_u64)]),
         "gpue_scan_filter_i64_lt": (c_i32, [c_vp, c_vp, c_u64, c_i64, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_scan_filter_i64_lt_sp": (c_i32, [c_vp, c_vp, c_u64, c_i64, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_join_build_payload_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
@@ -274,6 +277,21 @@ class Engine:
         _ck(self._lib, self._lib.gpue_scan_filter_i64_lt_sp(self._h, inp._h, n, theta,
                                                             out._h, ctypes.byref(cnt)))
         return cnt.value
+
+    def eval_conjuncts(self, cols, n_rows, preds) -> int:
+        """preds: list of (col_index, op, lo, hi); op 0 EQ / 1 LT / 2 BETWEEN.
+        Compacts cols stably in place; returns surviving rows."""
+        arr = (c_vp * len(cols))(*[c._h for c in cols])
+        pc = np.array([p[0] for p in preds], np.int32)
+        po = np.array([p[1] for p in preds], np.int32)
+        pl = np.array([p[2] for p in preds], np.int32)
+        ph = np.array([p[3] for p in preds], np.int32)
+        out = c_u64()
+        _ck(self._lib, self._lib.gpue_eval_conjuncts_i32(
+            self._h, arr, len(cols), n_rows, pc.ctypes.data_as(c_vp),
+            po.ctypes.data_as(c_vp), pl.ctypes.data_as(c_vp), ph.ctypes.data_as(c_vp),
+            len(preds), ctypes.byref(out)))
+        return out.value
 
     def join_build_payload(self, keys: DBuf, payloads: DBuf, n_rows) -> JoinTable:
         h = c_vp()

@@ -188,3 +188,26 @@ def test_bucket_chained_u64_oracle_vs_brute():
                     for j in index.get(int(k), []))
     assert sorted(zip(op.tolist(), ob.tolist())) == expect
     assert len(expect) > 0
+
+
+def test_eval_conjuncts_eager_prune_oracle():
+    """Eager-prune conjunct evaluation == plain numpy AND-filter; SSB Q1.1's
+    WHERE shape (d between, discount between, quantity <)."""
+    rng = np.random.default_rng(83)
+    n = 200_000
+    od = rng.integers(19920101, 19990101, n).astype(np.int32)
+    dc = rng.integers(0, 11, n).astype(np.int32)
+    qt = rng.integers(1, 51, n).astype(np.int32)
+    keep = ((od >= 19930101) & (od <= 19931231) & (dc >= 1) & (dc <= 3) & (qt < 25))
+    expect = [od[keep], dc[keep], qt[keep]]
+    cols = [od.copy(), dc.copy(), qt.copy()]
+    m = orc.eval_conjuncts(cols, [(0, 2, 19930101, 19931231), (1, 2, 1, 3), (2, 1, 0, 25)])
+    assert m == int(keep.sum())
+    for got, exp in zip(cols, expect):
+        assert np.array_equal(got[:m], exp)
+    # all-false short-circuit
+    cols2 = [od.copy(), dc.copy(), qt.copy()]
+    assert orc.eval_conjuncts(cols2, [(1, 0, 99, 0)]) == 0
+    # all-true skip (no compaction, full count)
+    cols3 = [od.copy(), dc.copy(), qt.copy()]
+    assert orc.eval_conjuncts(cols3, [(2, 1, 0, 100)]) == n

@@ -1167,3 +1167,31 @@ def test_varchar_probe_right_parity(engine):
     for b in (d_bb, d_bo, d_pb, d_po):
         b.free()
     t.destroy()
+
+
+def test_eval_conjuncts_parity(engine):
+    """GPU eager-prune conjunct evaluation vs the oracle: same survivors,
+    same stable order, across prune-triggering and non-triggering
+    selectivities (chunk_predicate_evaluator.cpp:31-80)."""
+    rng = np.random.default_rng(89)
+    n = 3_000_000
+    od = rng.integers(19920101, 19990101, n).astype(np.int32)
+    dc = rng.integers(0, 11, n).astype(np.int32)
+    qt = rng.integers(1, 51, n).astype(np.int32)
+    for preds in ([(0, 2, 19930101, 19931231), (1, 2, 1, 3), (2, 1, 0, 25)],  # sparse -> prunes
+                  [(2, 1, 0, 45), (1, 2, 0, 9)],                               # dense -> no prune
+                  [(1, 0, 99, 0)],                                             # all-false
+                  [(2, 1, 0, 100)]):                                           # all-true
+        cols_np = [od.copy(), dc.copy(), qt.copy()]
+        em = orc.eval_conjuncts(cols_np, preds)
+        dcols = []
+        for a in (od, dc, qt):
+            b = engine.alloc(a.nbytes)
+            b.h2d(a)
+            dcols.append(b)
+        gm = engine.eval_conjuncts(dcols, n, preds)
+        assert gm == em, preds
+        for b, exp in zip(dcols, cols_np):
+            if gm:
+                assert np.array_equal(b.d2h(np.int32, gm), exp[:gm])
+            b.free()
