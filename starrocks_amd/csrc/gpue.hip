@@ -4496,6 +4496,85 @@ __global__ void k_q3_legs(const int64_t* __restrict__ lk,
     if (threadIdx.x == 0) atomicAdd(sink, acc);
 }
 
+// ---------------------------------------------------------------------------
+// Partitioned q3 probe (round-2 lever 2 measured early): the 56 MB order-
+// bits bitset misses L2 at 90% (profiles/r01f_tcc_hit_rates.txt), so the
+// probe pays L3/HBM latency per gather. Two passes restore locality:
+//   A) stream ship/lk/ext/disc; passing rows (ship filter) partition by
+//      orderkey RANGE into nparts slices (the local-exchange repartition
+//      idiom, local_exchange.h:71-260, keyed by range not hash so each
+//      slice's bitset window is contiguous), materializing (key u32,
+//      v = ext*(100-disc) i64);
+//   B) one launch per partition probes its L2-resident bitset window and
+//      inserts into the shared CAS agg table.
+// Results are identical to the fused kernel (same inserts, order-free).
+// ---------------------------------------------------------------------------
+__global__ void k_q3_part_hist(const int64_t* __restrict__ lk,
+                               const int32_t* __restrict__ ship, uint64_t n,
+                               int32_t ship_cutoff, uint64_t slice, uint32_t nparts,
+                               uint64_t tile, uint32_t* __restrict__ block_hist) {
+    extern __shared__ uint32_t h[];
+    for (uint32_t c = threadIdx.x; c < nparts; c += blockDim.x) h[c] = 0;
+    __syncthreads();
+    uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
+    for (uint64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        if (__builtin_nontemporal_load(ship + i) <= ship_cutoff) continue;
+        uint32_t p = (uint32_t)(((uint64_t)__builtin_nontemporal_load(lk + i) - 1) / slice);
+        atomicAdd(&h[p], 1u);
+    }
+    __syncthreads();
+    for (uint32_t c = threadIdx.x; c < nparts; c += blockDim.x)
+        block_hist[(uint64_t)blockIdx.x * nparts + c] = h[c];
+}
+
+__global__ void k_q3_part_emit(const int64_t* __restrict__ lk,
+                               const int64_t* __restrict__ ext,
+                               const int64_t* __restrict__ disc,
+                               const int32_t* __restrict__ ship, uint64_t n,
+                               int32_t ship_cutoff, uint64_t slice, uint32_t nparts,
+                               uint64_t tile, const uint64_t* __restrict__ block_offsets,
+                               uint32_t* __restrict__ out_keys,
+                               long long* __restrict__ out_vals) {
+    extern __shared__ uint64_t cur[];
+    for (uint32_t c = threadIdx.x; c < nparts; c += blockDim.x)
+        cur[c] = block_offsets[(uint64_t)blockIdx.x * nparts + c];
+    __syncthreads();
+    uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
+    for (uint64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        if (__builtin_nontemporal_load(ship + i) <= ship_cutoff) continue;
+        uint64_t k = (uint64_t)__builtin_nontemporal_load(lk + i);
+        uint32_t p = (uint32_t)((k - 1) / slice);
+        uint64_t pos = atomicAdd((unsigned long long*)&cur[p], 1ull);
+        out_keys[pos] = (uint32_t)k;
+        out_vals[pos] = __builtin_nontemporal_load(ext + i) *
+                        (100 - __builtin_nontemporal_load(disc + i));
+    }
+}
+
+__global__ void k_q3_probe_slice(const uint32_t* __restrict__ keys,
+                                 const long long* __restrict__ vals, uint64_t n,
+                                 const uint32_t* __restrict__ order_bits,
+                                 unsigned long long* __restrict__ slots,
+                                 unsigned long long* __restrict__ sums, uint64_t cap_mask) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        unsigned long long k = keys[i];
+        uint64_t o = k - 1;
+        if (!((order_bits[o >> 5] >> (o & 31)) & 1u)) continue;
+        unsigned long long v = (unsigned long long)vals[i];
+        uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        for (;;) {
+            unsigned long long cur = slots[s];
+            if (cur == k) { atomicAdd(&sums[s], v); break; }
+            if (cur == AGG_EMPTY) {
+                unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
+                if (old == AGG_EMPTY || old == k) { atomicAdd(&sums[s], v); break; }
+            }
+            s = (s + 1) & cap_mask;
+        }
+    }
+}
+
 extern "C" {
 int gpue_gen_lineitem_q3(gpue_session* s, uint64_t seed, uint64_t row_start, uint64_t n,
                          uint64_t n_orders, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
@@ -4635,6 +4714,91 @@ int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbu
     *n_groups = groups;
     if (groups > max_out) {
         snprintf(g_err, sizeof(g_err), "q3: %llu groups exceed max_out %llu", groups,
+                 (unsigned long long)max_out);
+        return GPUE_ERR_ARG;
+    }
+    return GPUE_OK;
+}
+
+extern "C" int gpue_q3_probe_agg_part(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext,
+                                      gpue_dbuf* disc, gpue_dbuf* ship, uint64_t n,
+                                      uint64_t n_orders, gpue_dbuf* order_bits,
+                                      int32_t ship_cutoff, gpue_agg_table* at,
+                                      gpue_dbuf* keys_scratch, gpue_dbuf* vals_scratch,
+                                      uint32_t nparts, gpue_dbuf* out_keys,
+                                      gpue_dbuf* out_sums, uint64_t max_out,
+                                      uint64_t* n_groups);
+int gpue_q3_probe_agg_part(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
+                           gpue_dbuf* ship, uint64_t n, uint64_t n_orders,
+                           gpue_dbuf* order_bits, int32_t ship_cutoff, gpue_agg_table* at,
+                           gpue_dbuf* keys_scratch, gpue_dbuf* vals_scratch, uint32_t nparts,
+                           gpue_dbuf* out_keys, gpue_dbuf* out_sums, uint64_t max_out,
+                           uint64_t* n_groups) {
+    ARG_CHECK(s && lk && ext && disc && ship && order_bits && at && keys_scratch &&
+              vals_scratch && out_keys && out_sums && n_groups);
+    ARG_CHECK(nparts >= 1 && nparts <= 256 && n_orders > 0);
+    ARG_CHECK(keys_scratch->bytes >= n * 4 && vals_scratch->bytes >= n * 8);
+    int rc = agg_table_reset(at, /*with_counts=*/false);
+    if (rc != GPUE_OK) return rc;
+    uint64_t slice = (n_orders + nparts - 1) / nparts;
+    uint32_t nb = grid_capped(n, env_cap("GPUE_GRID_Q3P", 512));
+    uint64_t tile = (n + nb - 1) / nb;
+    uint32_t* d_hist = nullptr;
+    uint64_t* d_off = nullptr;
+    HIP_CHECK(hipMalloc(&d_hist, (uint64_t)nb * nparts * 4));
+    hipLaunchKernelGGL(k_q3_part_hist, dim3(nb), dim3(BLOCK), nparts * 4, s->stream,
+                       (const int64_t*)lk->ptr, (const int32_t*)ship->ptr, n, ship_cutoff,
+                       slice, nparts, tile, d_hist);
+    uint32_t* h_hist = (uint32_t*)malloc((uint64_t)nb * nparts * 4);
+    uint64_t* h_off = (uint64_t*)malloc((uint64_t)nb * nparts * 8);
+    uint64_t* pstart = (uint64_t*)malloc((nparts + 1) * 8);
+    HIP_CHECK(hipMemcpyAsync(h_hist, d_hist, (uint64_t)nb * nparts * 4,
+                             hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    uint64_t acc = 0;
+    for (uint32_t c = 0; c < nparts; c++) {
+        pstart[c] = acc;
+        for (uint32_t b = 0; b < nb; b++) {
+            h_off[(uint64_t)b * nparts + c] = acc;
+            acc += h_hist[(uint64_t)b * nparts + c];
+        }
+    }
+    pstart[nparts] = acc;
+    HIP_CHECK(hipMalloc(&d_off, (uint64_t)nb * nparts * 8));
+    HIP_CHECK(hipMemcpyAsync(d_off, h_off, (uint64_t)nb * nparts * 8, hipMemcpyHostToDevice,
+                             s->stream));
+    hipLaunchKernelGGL(k_q3_part_emit, dim3(nb), dim3(BLOCK), nparts * 8, s->stream,
+                       (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
+                       (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n, ship_cutoff,
+                       slice, nparts, tile, d_off, (uint32_t*)keys_scratch->ptr,
+                       (long long*)vals_scratch->ptr);
+    // pass B: one launch per partition — all CUs probe ONE L2-resident
+    // bitset window at a time
+    for (uint32_t p = 0; p < nparts; p++) {
+        uint64_t cnt = pstart[p + 1] - pstart[p];
+        if (!cnt) continue;
+        hipLaunchKernelGGL(k_q3_probe_slice,
+                           dim3(grid_capped(cnt, env_cap("GPUE_GRID_Q3S", MAX_GRID))),
+                           dim3(BLOCK), 0, s->stream,
+                           (const uint32_t*)keys_scratch->ptr + pstart[p],
+                           (const long long*)vals_scratch->ptr + pstart[p], cnt,
+                           (const uint32_t*)order_bits->ptr, at->slots, at->sums,
+                           at->cap - 1);
+    }
+    hipLaunchKernelGGL(k_hash_agg_emit, dim3(grid_for(at->cap)), dim3(BLOCK), 0, s->stream,
+                       at->slots, at->sums, at->counts, at->cap, at->cursor, max_out,
+                       (uint64_t*)out_keys->ptr, (int64_t*)out_sums->ptr, nullptr);
+    unsigned long long groups = 0;
+    HIP_CHECK(hipMemcpyAsync(&groups, at->cursor, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_hist);
+    (void)hipFree(d_off);
+    free(h_hist);
+    free(h_off);
+    free(pstart);
+    *n_groups = groups;
+    if (groups > max_out) {
+        snprintf(g_err, sizeof(g_err), "q3p: %llu groups exceed max_out %llu", groups,
                  (unsigned long long)max_out);
         return GPUE_ERR_ARG;
     }

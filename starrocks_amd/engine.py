@@ -114,6 +114,9 @@ def _declare(lib):
         "gpue_agg_table_create": (c_i32, [c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_agg_table_destroy": (None, [c_vp]),
         "gpue_agg_table_reset": (c_i32, [c_vp]),
+        "gpue_q3_probe_agg_part": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_u64, c_vp,
+                                           c_i32, c_vp, c_vp, c_vp, c_u32, c_vp, c_vp,
+                                           c_u64, ctypes.POINTER(c_u64)]),
         "gpue_q3_probe_agg_t": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_vp, c_i32, c_vp,
                                        c_vp, c_vp, c_u64, ctypes.POINTER(c_u64)]),
         "gpue_q3_probe_agg": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_vp, c_i32, c_u64,
@@ -535,6 +538,16 @@ class Engine:
         _ck(self._lib, self._lib.gpue_q3_decomp(
             self._h, lk._h, ext._h, disc._h, ship._h, n, order_bits._h, ship_cutoff, legs,
             at, sink._h))
+
+    def q3_probe_agg_part(self, lk, ext, disc, ship, n, n_orders, order_bits,
+                          ship_cutoff, at, keys_scratch, vals_scratch, nparts,
+                          out_keys, out_sums, max_out) -> int:
+        g = c_u64()
+        _ck(self._lib, self._lib.gpue_q3_probe_agg_part(
+            self._h, lk._h, ext._h, disc._h, ship._h, n, n_orders, order_bits._h,
+            ship_cutoff, at, keys_scratch._h, vals_scratch._h, nparts, out_keys._h,
+            out_sums._h, max_out, ctypes.byref(g)))
+        return g.value
 
     def hash_agg_sum_u64(self, keys: DBuf, vals: DBuf, n, out_keys: DBuf, out_sums: DBuf,
                          out_counts: DBuf = None, max_out=0, capacity_hint=0):

@@ -1195,3 +1195,37 @@ def test_eval_conjuncts_parity(engine):
             if gm:
                 assert np.array_equal(b.d2h(np.int32, gm), exp[:gm])
             b.free()
+
+
+def test_q3_partitioned_probe_parity(engine):
+    """Range-partitioned q3 probe (pass A partition + pass B L2-resident
+    slice probes) produces the SAME groups/sums as the fused kernel and the
+    oracle."""
+    n, n_orders, n_custs = 4_000_000, 2_000_000, 200_000
+    mkt = engine.alloc(n_custs * 16)
+    engine.gen_cust_mkt16(SEED, n_custs, mkt)
+    cbits = engine.alloc((n_custs + 31) // 32 * 4)
+    engine.bits_str16_eq(mkt, n_custs, orc.mkt_literal(1), cbits)
+    oc, od = engine.alloc(n_orders * 4), engine.alloc(n_orders * 4)
+    engine.gen_orders_q3(SEED, n_orders, n_custs, oc, od)
+    obits = engine.alloc((n_orders + 31) // 32 * 4)
+    engine.q3_order_bits(oc, od, n_orders, cbits, 19950315, obits)
+    lk, ext, disc = (engine.alloc(n * 8) for _ in range(3))
+    ship = engine.alloc(n * 4)
+    engine.gen_lineitem_q3(SEED, 0, n, n_orders, lk, ext, disc, ship)
+    at = engine.agg_table_create(1 << 22)
+    ks, vs = engine.alloc(n * 4), engine.alloc(n * 8)
+    max_out = 2_000_000
+    ok_b, os_b = engine.alloc(max_out * 8), engine.alloc(max_out * 8)
+    g = engine.q3_probe_agg_part(lk, ext, disc, ship, n, n_orders, obits, 19950315,
+                                 at, ks, vs, 32, ok_b, os_b, max_out)
+    gk = ok_b.d2h(np.uint64, g)
+    gs = os_b.d2h(np.int64, g)
+    order = np.argsort(gk)
+    ek, es = orc.q3_pipeline(SEED, 0, n, n_orders, n_custs)
+    assert g == len(ek)
+    assert np.array_equal(gk[order], ek)
+    assert np.array_equal(gs[order], es)
+    for b in (mkt, cbits, oc, od, obits, lk, ext, disc, ship, ks, vs, ok_b, os_b):
+        b.free()
+    engine.agg_table_destroy(at)
