@@ -309,13 +309,25 @@ def main():
         ok_b, os_b = eng.alloc(max_out * 8), eng.alloc(max_out * 8)
         agg_tab = eng.agg_table_create(64_000_000)  # persistent; reset per pass
 
-        def kernel_only():
-            # fused lineitem filter + orders semi-probe + hash aggregate
-            # (per pass: agg-table reset + probe + emit — the whole
-            # probe-side query pass; the table itself persists like the
-            # CPU's reused chunk allocations)
-            return eng.q3_probe_agg_t(lk, ext, disc, ship, rows, obits, Q3_CUTOFF,
-                                      agg_tab, ok_b, os_b, max_out)
+        nparts = int(os.environ.get("GPUE_Q3_PART", "0"))
+        if nparts and world == 1:
+            # range-partitioned probe (DESIGN.md §4b L2-locality lever):
+            # pass A partitions ship-passing rows by orderkey range, pass B
+            # probes one L2-resident bitset window per partition
+            ks_scr, vs_scr = eng.alloc(rows * 4), eng.alloc(rows * 8)
+
+            def kernel_only():
+                return eng.q3_probe_agg_part(lk, ext, disc, ship, rows, Q3_N_ORDERS,
+                                             obits, Q3_CUTOFF, agg_tab, ks_scr, vs_scr,
+                                             nparts, ok_b, os_b, max_out)
+        else:
+            def kernel_only():
+                # fused lineitem filter + orders semi-probe + hash aggregate
+                # (per pass: agg-table reset + probe + emit — the whole
+                # probe-side query pass; the table itself persists like the
+                # CPU's reused chunk allocations)
+                return eng.q3_probe_agg_t(lk, ext, disc, ship, rows, obits, Q3_CUTOFF,
+                                          agg_tab, ok_b, os_b, max_out)
 
         if world == 1:
             def step():
