@@ -211,3 +211,28 @@ def test_eval_conjuncts_eager_prune_oracle():
     # all-true skip (no compaction, full count)
     cols3 = [od.copy(), dc.copy(), qt.copy()]
     assert orc.eval_conjuncts(cols3, [(2, 1, 0, 100)]) == n
+
+
+def test_full_outer_composition():
+    """FULL OUTER JOIN == LEFT OUTER emit (mode 3) ∪ RIGHT ANTI build rows
+    (probe side NULL) — the reference's full-outer = probe pass + unmatched
+    build scan (join_hash_map.h:228-333 list). Verified against brute force."""
+    rng = np.random.default_rng(97)
+    build = rng.integers(0, 40, 150).astype(np.uint32)
+    probe = rng.integers(0, 60, 200).astype(np.uint32)
+    bkeys = np.concatenate([[0], build]).astype(np.uint32)
+    first, nxt, bs, log = orc.bucket_chained_build(bkeys)
+    heads = orc.bucket_chained_lookup(probe, first, bs, log)
+    op, ob = orc.probe_emit_mode(bkeys, nxt, probe, heads, 3)
+    unmatched_build = orc.probe_right(bkeys, nxt, probe, heads, 1)
+    got = sorted([(int(p), int(b)) for p, b in zip(op, ob)] +
+                 [(-1, int(j)) for j in unmatched_build])  # -1 = NULL probe side
+    bset = set(build.tolist())
+    pset = set(probe.tolist())
+    expect = []
+    for i, k in enumerate(probe.tolist()):
+        hits = [j for j in range(1, len(bkeys)) if build[j - 1] == k]
+        expect += [(i, j) for j in hits] if hits else [(i, 0)]
+    expect += [(-1, j) for j in range(1, len(bkeys)) if build[j - 1] not in pset]
+    assert got == sorted(expect)
+    assert any(p == -1 for p, _ in got) and any(b == 0 for _, b in got)
