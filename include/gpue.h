@@ -120,6 +120,10 @@ int gpue_eval_conjuncts_i32(gpue_session* s, gpue_dbuf** cols, int n_cols, uint6
                             const int32_t* pred_col, const int32_t* pred_op,
                             const int32_t* pred_lo, const int32_t* pred_hi, int n_preds,
                             uint64_t* out_rows);
+int gpue_eval_conjuncts_i64(gpue_session* s, gpue_dbuf** cols, int n_cols, uint64_t n_rows,
+                            const int32_t* pred_col, const int32_t* pred_op,
+                            const int64_t* pred_lo, const int64_t* pred_hi, int n_preds,
+                            uint64_t* out_rows);
 
 /* ---- hash-join build ----
  * Replaces JoinHashTable::build with the RANGE_DIRECT_MAPPING method the

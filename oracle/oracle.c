@@ -739,6 +739,61 @@ uint64_t orc_eval_conjuncts_i32(int32_t** cols, int n_cols, uint64_t n_rows,
     return n;
 }
 
+/* i64-column variant of the eager-prune conjunct evaluator. */
+uint64_t orc_eval_conjuncts_i64(int64_t** cols, int n_cols, uint64_t n_rows,
+                                const int32_t* pred_col, const int32_t* pred_op,
+                                const int64_t* pred_lo, const int64_t* pred_hi,
+                                int n_preds) {
+    uint64_t n = n_rows;
+    uint8_t* filter = (uint8_t*)malloc(n ? n : 1);
+    memset(filter, 1, n ? n : 1);
+    uint64_t zero_count = 0;
+    for (int p = 0; p < n_preds; p++) {
+        const int64_t* c = cols[pred_col[p]];
+        int op = pred_op[p];
+        int64_t lo = pred_lo[p], hi = pred_hi[p];
+        uint64_t true_count = 0;
+        for (uint64_t i = 0; i < n; i++) {
+            int pass = op == 0 ? (c[i] == lo) : op == 1 ? (c[i] < hi)
+                                              : (c[i] >= lo && c[i] <= hi);
+            true_count += pass;
+        }
+        if (true_count == n) continue;
+        if (true_count == 0) { free(filter); return 0; }
+        for (uint64_t i = 0; i < n; i++) {
+            int pass = op == 0 ? (c[i] == lo) : op == 1 ? (c[i] < hi)
+                                              : (c[i] >= lo && c[i] <= hi);
+            filter[i] = (uint8_t)(filter[i] & pass);
+        }
+        zero_count = 0;
+        for (uint64_t i = 0; i < n; i++) zero_count += (filter[i] == 0);
+        uint64_t prune_threshold = n * 8 / 10 > 1024 ? n * 8 / 10 : 1024;
+        if (zero_count > prune_threshold) {
+            uint64_t w = 0;
+            for (uint64_t i = 0; i < n; i++)
+                if (filter[i]) {
+                    for (int k = 0; k < n_cols; k++) cols[k][w] = cols[k][i];
+                    w++;
+                }
+            n = w;
+            if (n == 0) { free(filter); return 0; }
+            memset(filter, 1, n);
+            zero_count = 0;
+        }
+    }
+    if (zero_count) {
+        uint64_t w = 0;
+        for (uint64_t i = 0; i < n; i++)
+            if (filter[i]) {
+                for (int k = 0; k < n_cols; k++) cols[k][w] = cols[k][i];
+                w++;
+            }
+        n = w;
+    }
+    free(filter);
+    return n;
+}
+
 /* eval_conjuncts + Column::filter_range stream compaction
  * (be/src/exprs/chunk_predicate_evaluator.cpp:31-80,
  *  be/src/base/simd/filter.h:26-38): stable, order-preserving. */

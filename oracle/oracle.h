@@ -87,6 +87,10 @@ uint64_t orc_eval_conjuncts_i32(int32_t** cols, int n_cols, uint64_t n_rows,
                                 const int32_t* pred_col, const int32_t* pred_op,
                                 const int32_t* pred_lo, const int32_t* pred_hi,
                                 int n_preds);
+uint64_t orc_eval_conjuncts_i64(int64_t** cols, int n_cols, uint64_t n_rows,
+                                const int32_t* pred_col, const int32_t* pred_op,
+                                const int64_t* pred_lo, const int64_t* pred_hi,
+                                int n_preds);
 uint64_t orc_dict_decode_binary(const uint8_t* dict_bytes, const uint32_t* dict_offsets,
                                 const int32_t* codes, uint64_t n, uint8_t* out_bytes,
                                 uint32_t* out_offsets);

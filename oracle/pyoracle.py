@@ -84,6 +84,9 @@ def _decl(lib):
     lib.orc_sbf_log_num_buckets.argtypes = [c_u64]
     lib.orc_sbf_build_i32.argtypes = [c_vp, c_u64, c_vp, c_i32]
     lib.orc_sbf_test_i32.argtypes = [c_vp, c_u64, c_vp, c_i32, c_vp]
+    lib.orc_eval_conjuncts_i64.restype = c_u64
+    lib.orc_eval_conjuncts_i64.argtypes = [c_vp, c_i32, c_u64, c_vp, c_vp, c_vp, c_vp,
+                                           c_i32]
     lib.orc_eval_conjuncts_i32.restype = c_u64
     lib.orc_eval_conjuncts_i32.argtypes = [c_vp, c_i32, c_u64, c_vp, c_vp, c_vp, c_vp,
                                            c_i32]
@@ -593,3 +596,14 @@ def eval_conjuncts(cols, preds):
     m = load().orc_eval_conjuncts_i32(ptrs, len(cols), n, _p(pc), _p(po), _p(pl), _p(ph),
                                       len(preds))
     return m
+
+
+def eval_conjuncts_i64(cols, preds):
+    n = len(cols[0])
+    ptrs = (c_vp * len(cols))(*[c.ctypes.data_as(c_vp).value for c in cols])
+    pc = np.array([p[0] for p in preds], np.int32)
+    po = np.array([p[1] for p in preds], np.int32)
+    pl = np.array([p[2] for p in preds], np.int64)
+    ph = np.array([p[3] for p in preds], np.int64)
+    return load().orc_eval_conjuncts_i64(ptrs, len(cols), n, _p(pc), _p(po), _p(pl),
+                                         _p(ph), len(preds))

@@ -1838,6 +1838,28 @@ __global__ void k_mask_emit_idx(const uint8_t* __restrict__ filter, uint64_t n,
     }
 }
 
+__global__ void k_eval_pred_i64(const int64_t* __restrict__ col, uint64_t n, int op,
+                                int64_t lo, int64_t hi, uint8_t* __restrict__ filter,
+                                unsigned long long* __restrict__ true_count) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    unsigned long long local = 0;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        int64_t v = col[i];
+        uint8_t pass = op == 0 ? (v == lo) : op == 1 ? (v < hi) : (v >= lo && v <= hi);
+        uint8_t merged = filter[i] & pass;
+        filter[i] = merged;
+        local += merged;
+    }
+    for (int off = 32; off > 0; off >>= 1)
+        local += __shfl_down(local, off, WAVE);
+    if ((threadIdx.x & (WAVE - 1)) == 0 && local)
+        atomicAdd(true_count, local);
+}
+
+__global__ void k_gather_u64(const uint64_t* __restrict__ in,
+                             const uint32_t* __restrict__ idx, uint64_t n,
+                             uint64_t* __restrict__ out);
+
 extern "C" int gpue_eval_conjuncts_i32(gpue_session* s, gpue_dbuf** cols, int n_cols,
                                        uint64_t n_rows, const int32_t* pred_col,
                                        const int32_t* pred_op, const int32_t* pred_lo,
@@ -1900,6 +1922,83 @@ int gpue_eval_conjuncts_i32(gpue_session* s, gpue_dbuf** cols, int n_cols, uint6
         // NOTE: tc counts MERGED survivors (filter & pass) — a conjunct whose
         // own column is all-true over already-filtered rows leaves tc ==
         // previous survivor count, matching the reference's merge+count_zero
+        if (tc == 0) { n = 0; break; }
+        zero_count = n - tc;
+        uint64_t prune_threshold = n * 8 / 10 > 1024 ? n * 8 / 10 : 1024;
+        if (zero_count > prune_threshold) {
+            int rc = compact_all(tc);
+            if (rc != GPUE_OK) return rc;
+        }
+    }
+    if (n > 0 && zero_count > 0) {
+        int rc = compact_all(n - zero_count);
+        if (rc != GPUE_OK) return rc;
+    }
+    (void)hipFree(d_filter);
+    (void)hipFree(d_tc);
+    *out_rows = n;
+    return GPUE_OK;
+}
+
+extern "C" int gpue_eval_conjuncts_i64(gpue_session* s, gpue_dbuf** cols, int n_cols,
+                                       uint64_t n_rows, const int32_t* pred_col,
+                                       const int32_t* pred_op, const int64_t* pred_lo,
+                                       const int64_t* pred_hi, int n_preds,
+                                       uint64_t* out_rows);
+int gpue_eval_conjuncts_i64(gpue_session* s, gpue_dbuf** cols, int n_cols, uint64_t n_rows,
+                            const int32_t* pred_col, const int32_t* pred_op,
+                            const int64_t* pred_lo, const int64_t* pred_hi, int n_preds,
+                            uint64_t* out_rows) {
+    ARG_CHECK(s && cols && n_cols > 0 && pred_col && pred_op && pred_lo && pred_hi &&
+              out_rows);
+    uint64_t n = n_rows;
+    uint8_t* d_filter = nullptr;
+    unsigned long long* d_tc = nullptr;
+    HIP_CHECK(hipMalloc(&d_filter, n_rows ? n_rows : 1));
+    HIP_CHECK(hipMalloc(&d_tc, 8));
+    HIP_CHECK(hipMemsetAsync(d_filter, 1, n_rows ? n_rows : 1, s->stream));
+    uint64_t zero_count = 0;
+
+    auto compact_all = [&](uint64_t keep) -> int {
+        uint32_t nb = grid_for(n);
+        uint64_t tile = (n + nb - 1) / nb;
+        uint64_t* d_bs = nullptr;
+        uint32_t* d_idx = nullptr;
+        int64_t* d_tmp = nullptr;
+        HIP_CHECK(hipMalloc(&d_bs, (nb + 1) * 8));
+        HIP_CHECK(hipMalloc(&d_idx, keep * 4));
+        HIP_CHECK(hipMalloc(&d_tmp, keep * 8));
+        hipLaunchKernelGGL(k_mask_block_counts, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           d_filter, n, tile, d_bs);
+        hipLaunchKernelGGL(k_scan_small, dim3(1), dim3(1), 0, s->stream, d_bs, nb);
+        hipLaunchKernelGGL(k_mask_emit_idx, dim3(nb), dim3(BLOCK), 0, s->stream, d_filter,
+                           n, tile, d_bs, d_idx);
+        for (int k = 0; k < n_cols; k++) {
+            hipLaunchKernelGGL(k_gather_u64, dim3(grid_stream(keep)), dim3(BLOCK), 0,
+                               s->stream, (const uint64_t*)cols[k]->ptr, d_idx, keep,
+                               (uint64_t*)d_tmp);
+            HIP_CHECK(hipMemcpyAsync(cols[k]->ptr, d_tmp, keep * 8,
+                                     hipMemcpyDeviceToDevice, s->stream));
+        }
+        HIP_CHECK(hipStreamSynchronize(s->stream));
+        (void)hipFree(d_bs);
+        (void)hipFree(d_idx);
+        (void)hipFree(d_tmp);
+        n = keep;
+        HIP_CHECK(hipMemsetAsync(d_filter, 1, n ? n : 1, s->stream));
+        zero_count = 0;
+        return GPUE_OK;
+    };
+
+    for (int p = 0; p < n_preds && n > 0; p++) {
+        ARG_CHECK(pred_col[p] >= 0 && pred_col[p] < n_cols);
+        HIP_CHECK(hipMemsetAsync(d_tc, 0, 8, s->stream));
+        hipLaunchKernelGGL(k_eval_pred_i64, dim3(grid_stream(n)), dim3(BLOCK), 0, s->stream,
+                           (const int64_t*)cols[pred_col[p]]->ptr, n, pred_op[p],
+                           pred_lo[p], pred_hi[p], d_filter, d_tc);
+        unsigned long long tc = 0;
+        HIP_CHECK(hipMemcpyAsync(&tc, d_tc, 8, hipMemcpyDeviceToHost, s->stream));
+        HIP_CHECK(hipStreamSynchronize(s->stream));
         if (tc == 0) { n = 0; break; }
         zero_count = n - tc;
         uint64_t prune_threshold = n * 8 / 10 > 1024 ? n * 8 / 10 : 1024;

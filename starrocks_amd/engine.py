@@ -71,6 +71,9 @@ def _declare(lib):
         "gpue_eval_conjuncts_i32": (c_i32, [c_vp, ctypes.POINTER(c_vp), c_i32, c_u64,
                                             c_vp, c_vp, c_vp, c_vp, c_i32,
                                             ctypes.POINTER(c_u64)]),
+        "gpue_eval_conjuncts_i64": (c_i32, [c_vp, ctypes.POINTER(c_vp), c_i32, c_u64,
+                                            c_vp, c_vp, c_vp, c_vp, c_i32,
+                                            ctypes.POINTER(c_u64)]),
         "gpue_scan_filter_i64_lt": (c_i32, [c_vp, c_vp, c_u64, c_i64, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_scan_filter_i64_lt_sp": (c_i32, [c_vp, c_vp, c_u64, c_i64, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_join_build_payload_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
@@ -291,6 +294,19 @@ class Engine:
         ph = np.array([p[3] for p in preds], np.int32)
         out = c_u64()
         _ck(self._lib, self._lib.gpue_eval_conjuncts_i32(
+            self._h, arr, len(cols), n_rows, pc.ctypes.data_as(c_vp),
+            po.ctypes.data_as(c_vp), pl.ctypes.data_as(c_vp), ph.ctypes.data_as(c_vp),
+            len(preds), ctypes.byref(out)))
+        return out.value
+
+    def eval_conjuncts_i64(self, cols, n_rows, preds) -> int:
+        arr = (c_vp * len(cols))(*[c._h for c in cols])
+        pc = np.array([p[0] for p in preds], np.int32)
+        po = np.array([p[1] for p in preds], np.int32)
+        pl = np.array([p[2] for p in preds], np.int64)
+        ph = np.array([p[3] for p in preds], np.int64)
+        out = c_u64()
+        _ck(self._lib, self._lib.gpue_eval_conjuncts_i64(
             self._h, arr, len(cols), n_rows, pc.ctypes.data_as(c_vp),
             po.ctypes.data_as(c_vp), pl.ctypes.data_as(c_vp), ph.ctypes.data_as(c_vp),
             len(preds), ctypes.byref(out)))

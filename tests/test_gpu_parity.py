@@ -1229,3 +1229,25 @@ def test_q3_partitioned_probe_parity(engine):
     for b in (mkt, cbits, oc, od, obits, lk, ext, disc, ship, ks, vs, ok_b, os_b):
         b.free()
     engine.agg_table_destroy(at)
+
+
+def test_eval_conjuncts_i64_parity(engine):
+    """i64-column conjunct evaluation (the q3 lineitem column type) vs the
+    oracle: survivors and stable order equal."""
+    rng = np.random.default_rng(101)
+    n = 2_000_000
+    a = rng.integers(0, 10**12, n).astype(np.int64)
+    b = rng.integers(0, 100, n).astype(np.int64)
+    preds = [(0, 2, 10**11, 5 * 10**11), (1, 1, 0, 30)]
+    cols_np = [a.copy(), b.copy()]
+    em = orc.eval_conjuncts_i64(cols_np, preds)
+    dcols = []
+    for arr in (a, b):
+        buf = engine.alloc(arr.nbytes)
+        buf.h2d(arr)
+        dcols.append(buf)
+    gm = engine.eval_conjuncts_i64(dcols, n, preds)
+    assert gm == em
+    for buf, exp in zip(dcols, cols_np):
+        assert np.array_equal(buf.d2h(np.int64, gm), exp[:gm])
+        buf.free()
