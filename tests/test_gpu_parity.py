@@ -1251,3 +1251,33 @@ def test_eval_conjuncts_i64_parity(engine):
     for buf, exp in zip(dcols, cols_np):
         assert np.array_equal(buf.d2h(np.int64, gm), exp[:gm])
         buf.free()
+
+
+def test_q1_accum_step_soak(engine):
+    """The accumulating q1 step (persistent acc + host diff, bench default at
+    N=1): 300 consecutive steps each produce the exact oracle (sum, count)
+    as a difference of readbacks — no drift, no race."""
+    n = 1_000_000
+    cols = [engine.alloc(n * 4) for _ in range(3)]
+    engine.gen_lineorder_q1(SEED, 0, n, *cols)
+    from starrocks_amd import gen
+    datekey, dyear = gen.gen_dates()
+    dpay = np.where(dyear == 1993, dyear - 1992 + 1, 0).astype(np.uint32)
+    kb = engine.alloc(datekey.nbytes)
+    kb.h2d(datekey.astype(np.int32))
+    pb = engine.alloc(dpay.nbytes)
+    pb.h2d(dpay)
+    t = engine.join_build_payload(kb, pb, len(datekey))
+    acc = engine.alloc(16)
+    acc.h2d(np.zeros(2, np.int64))
+    expect = orc.q1_pipeline(SEED, 0, n, 1993)
+    prev = np.zeros(2, np.int64)
+    for step in range(300):
+        engine.q1_join_sum_accum(t, cols[0], cols[1], cols[2], n, acc)
+        cur = acc.d2h(np.int64, 2)
+        d = cur - prev
+        prev = cur
+        assert (int(d[0]), int(d[1])) == expect, step
+    for b in cols + [kb, pb, acc]:
+        b.free()
+    t.destroy()
