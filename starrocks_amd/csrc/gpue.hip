@@ -4612,18 +4612,24 @@ __global__ void k_q3_part_hist(const int64_t* __restrict__ lk,
                                const int32_t* __restrict__ ship, uint64_t n,
                                int32_t ship_cutoff, uint64_t slice, uint32_t nparts,
                                uint64_t tile, uint32_t* __restrict__ block_hist) {
-    extern __shared__ uint32_t h[];
-    for (uint32_t c = threadIdx.x; c < nparts; c += blockDim.x) h[c] = 0;
+    // wave-PRIVATE counters: same-address LDS atomic serialization measured
+    // 4x the stream bound (profiles/q3_partitioned_probe_r01.txt); each wave
+    // owns its own nparts counters, written out per (block, wave, partition)
+    // so the emit can reserve per-wave ranges with no atomics at all.
+    extern __shared__ uint32_t h[]; // [waves_per_block][nparts]
+    const int wid = threadIdx.x / WAVE;
+    const int nw = blockDim.x / WAVE;
+    for (uint32_t c = threadIdx.x; c < (uint32_t)nw * nparts; c += blockDim.x) h[c] = 0;
     __syncthreads();
     uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
     for (uint64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
         if (__builtin_nontemporal_load(ship + i) <= ship_cutoff) continue;
         uint32_t p = (uint32_t)(((uint64_t)__builtin_nontemporal_load(lk + i) - 1) / slice);
-        atomicAdd(&h[p], 1u);
+        atomicAdd(&h[wid * nparts + p], 1u);
     }
     __syncthreads();
-    for (uint32_t c = threadIdx.x; c < nparts; c += blockDim.x)
-        block_hist[(uint64_t)blockIdx.x * nparts + c] = h[c];
+    for (uint32_t c = threadIdx.x; c < (uint32_t)nw * nparts; c += blockDim.x)
+        block_hist[(uint64_t)blockIdx.x * nw * nparts + c] = h[c];
 }
 
 __global__ void k_q3_part_emit(const int64_t* __restrict__ lk,
@@ -4631,22 +4637,64 @@ __global__ void k_q3_part_emit(const int64_t* __restrict__ lk,
                                const int64_t* __restrict__ disc,
                                const int32_t* __restrict__ ship, uint64_t n,
                                int32_t ship_cutoff, uint64_t slice, uint32_t nparts,
-                               uint64_t tile, const uint64_t* __restrict__ block_offsets,
+                               uint64_t tile, const uint64_t* __restrict__ wave_offsets,
                                uint32_t* __restrict__ out_keys,
                                long long* __restrict__ out_vals) {
-    extern __shared__ uint64_t cur[];
-    for (uint32_t c = threadIdx.x; c < nparts; c += blockDim.x)
-        cur[c] = block_offsets[(uint64_t)blockIdx.x * nparts + c];
+    // per-(block,wave,partition) ranges reserved by the host scan: each
+    // wave bumps its own LDS cursors — contention only among 64 lanes
+    extern __shared__ uint64_t cur[]; // [waves_per_block][nparts]
+    const int wid = threadIdx.x / WAVE;
+    const int nw = blockDim.x / WAVE;
+    for (uint32_t c = threadIdx.x; c < (uint32_t)nw * nparts; c += blockDim.x)
+        cur[c] = wave_offsets[(uint64_t)blockIdx.x * nw * nparts + c];
     __syncthreads();
     uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
     for (uint64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
         if (__builtin_nontemporal_load(ship + i) <= ship_cutoff) continue;
         uint64_t k = (uint64_t)__builtin_nontemporal_load(lk + i);
         uint32_t p = (uint32_t)((k - 1) / slice);
-        uint64_t pos = atomicAdd((unsigned long long*)&cur[p], 1ull);
+        uint64_t pos = atomicAdd((unsigned long long*)&cur[wid * nparts + p], 1ull);
         out_keys[pos] = (uint32_t)k;
         out_vals[pos] = __builtin_nontemporal_load(ext + i) *
                         (100 - __builtin_nontemporal_load(disc + i));
+    }
+}
+
+// fused pass B: every block is statically assigned ONE partition
+// (block_part/block_base arrays) — one launch instead of nparts small
+// launch-bound ones; consecutive blocks share a partition so each XCD's L2
+// sees one bitset window at a time.
+__global__ void k_q3_probe_slices_fused(const uint32_t* __restrict__ keys,
+                                        const long long* __restrict__ vals,
+                                        const uint32_t* __restrict__ block_part,
+                                        const uint64_t* __restrict__ pstart,
+                                        const uint32_t* __restrict__ blocks_per_part,
+                                        const uint32_t* __restrict__ first_block_of_part,
+                                        const uint32_t* __restrict__ order_bits,
+                                        unsigned long long* __restrict__ slots,
+                                        unsigned long long* __restrict__ sums,
+                                        uint64_t cap_mask) {
+    uint32_t p = block_part[blockIdx.x];
+    uint64_t base = pstart[p];
+    uint64_t cnt = pstart[p + 1] - base;
+    uint32_t nb_p = blocks_per_part[p];
+    uint32_t rel = blockIdx.x - first_block_of_part[p];
+    uint64_t stride = (uint64_t)nb_p * blockDim.x;
+    for (uint64_t i = (uint64_t)rel * blockDim.x + threadIdx.x; i < cnt; i += stride) {
+        unsigned long long k = keys[base + i];
+        uint64_t o = k - 1;
+        if (!((order_bits[o >> 5] >> (o & 31)) & 1u)) continue;
+        unsigned long long v = (unsigned long long)vals[base + i];
+        uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        for (;;) {
+            unsigned long long cur_ = slots[s];
+            if (cur_ == k) { atomicAdd(&sums[s], v); break; }
+            if (cur_ == AGG_EMPTY) {
+                unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
+                if (old == AGG_EMPTY || old == k) { atomicAdd(&sums[s], v); break; }
+            }
+            s = (s + 1) & cap_mask;
+        }
     }
 }
 
@@ -4842,47 +4890,79 @@ int gpue_q3_probe_agg_part(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_
     uint64_t slice = (n_orders + nparts - 1) / nparts;
     uint32_t nb = grid_capped(n, env_cap("GPUE_GRID_Q3P", 512));
     uint64_t tile = (n + nb - 1) / nb;
+    const uint32_t nw = BLOCK / WAVE; // waves per block
+    uint64_t hist_len = (uint64_t)nb * nw * nparts;
     uint32_t* d_hist = nullptr;
     uint64_t* d_off = nullptr;
-    HIP_CHECK(hipMalloc(&d_hist, (uint64_t)nb * nparts * 4));
-    hipLaunchKernelGGL(k_q3_part_hist, dim3(nb), dim3(BLOCK), nparts * 4, s->stream,
+    HIP_CHECK(hipMalloc(&d_hist, hist_len * 4));
+    hipLaunchKernelGGL(k_q3_part_hist, dim3(nb), dim3(BLOCK), nw * nparts * 4, s->stream,
                        (const int64_t*)lk->ptr, (const int32_t*)ship->ptr, n, ship_cutoff,
                        slice, nparts, tile, d_hist);
-    uint32_t* h_hist = (uint32_t*)malloc((uint64_t)nb * nparts * 4);
-    uint64_t* h_off = (uint64_t*)malloc((uint64_t)nb * nparts * 8);
+    uint32_t* h_hist = (uint32_t*)malloc(hist_len * 4);
+    uint64_t* h_off = (uint64_t*)malloc(hist_len * 8);
     uint64_t* pstart = (uint64_t*)malloc((nparts + 1) * 8);
-    HIP_CHECK(hipMemcpyAsync(h_hist, d_hist, (uint64_t)nb * nparts * 4,
-                             hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipMemcpyAsync(h_hist, d_hist, hist_len * 4, hipMemcpyDeviceToHost,
+                             s->stream));
     HIP_CHECK(hipStreamSynchronize(s->stream));
     uint64_t acc = 0;
-    for (uint32_t c = 0; c < nparts; c++) {
-        pstart[c] = acc;
-        for (uint32_t b = 0; b < nb; b++) {
-            h_off[(uint64_t)b * nparts + c] = acc;
-            acc += h_hist[(uint64_t)b * nparts + c];
-        }
+    for (uint32_t p = 0; p < nparts; p++) {
+        pstart[p] = acc;
+        for (uint32_t b = 0; b < nb; b++)
+            for (uint32_t w = 0; w < nw; w++) {
+                uint64_t idx = ((uint64_t)b * nw + w) * nparts + p;
+                h_off[idx] = acc;
+                acc += h_hist[idx];
+            }
     }
     pstart[nparts] = acc;
-    HIP_CHECK(hipMalloc(&d_off, (uint64_t)nb * nparts * 8));
-    HIP_CHECK(hipMemcpyAsync(d_off, h_off, (uint64_t)nb * nparts * 8, hipMemcpyHostToDevice,
-                             s->stream));
-    hipLaunchKernelGGL(k_q3_part_emit, dim3(nb), dim3(BLOCK), nparts * 8, s->stream,
+    HIP_CHECK(hipMalloc(&d_off, hist_len * 8));
+    HIP_CHECK(hipMemcpyAsync(d_off, h_off, hist_len * 8, hipMemcpyHostToDevice, s->stream));
+    hipLaunchKernelGGL(k_q3_part_emit, dim3(nb), dim3(BLOCK), nw * nparts * 8, s->stream,
                        (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
                        (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n, ship_cutoff,
                        slice, nparts, tile, d_off, (uint32_t*)keys_scratch->ptr,
                        (long long*)vals_scratch->ptr);
-    // pass B: one launch per partition — all CUs probe ONE L2-resident
-    // bitset window at a time
+    // fused pass B: assign contiguous block ranges per partition, one launch
+    uint32_t total_blocks = env_cap("GPUE_GRID_Q3S", 2048);
+    if (total_blocks < nparts) total_blocks = nparts;
+    uint32_t* bpp = (uint32_t*)malloc(nparts * 4);
+    uint32_t* fbp = (uint32_t*)malloc(nparts * 4);
+    uint64_t total_rows = acc ? acc : 1;
+    uint32_t assigned = 0;
     for (uint32_t p = 0; p < nparts; p++) {
         uint64_t cnt = pstart[p + 1] - pstart[p];
-        if (!cnt) continue;
-        hipLaunchKernelGGL(k_q3_probe_slice,
-                           dim3(grid_capped(cnt, env_cap("GPUE_GRID_Q3S", MAX_GRID))),
-                           dim3(BLOCK), 0, s->stream,
-                           (const uint32_t*)keys_scratch->ptr + pstart[p],
-                           (const long long*)vals_scratch->ptr + pstart[p], cnt,
-                           (const uint32_t*)order_bits->ptr, at->slots, at->sums,
+        uint32_t want = (uint32_t)((__uint128_t)cnt * total_blocks / total_rows);
+        bpp[p] = cnt ? (want ? want : 1u) : 0u;
+        assigned += bpp[p];
+    }
+    uint32_t nb2 = 0;
+    uint32_t* bpart = (uint32_t*)malloc((assigned ? assigned : 1) * 4);
+    for (uint32_t p = 0; p < nparts; p++) {
+        fbp[p] = nb2;
+        for (uint32_t b = 0; b < bpp[p]; b++) bpart[nb2++] = p;
+    }
+    if (nb2 > 0) {
+        uint32_t *d_bpart = nullptr, *d_bpp = nullptr, *d_fbp = nullptr;
+        uint64_t* d_pstart = nullptr;
+        HIP_CHECK(hipMalloc(&d_bpart, nb2 * 4));
+        HIP_CHECK(hipMalloc(&d_bpp, nparts * 4));
+        HIP_CHECK(hipMalloc(&d_fbp, nparts * 4));
+        HIP_CHECK(hipMalloc(&d_pstart, (nparts + 1) * 8));
+        HIP_CHECK(hipMemcpyAsync(d_bpart, bpart, nb2 * 4, hipMemcpyHostToDevice, s->stream));
+        HIP_CHECK(hipMemcpyAsync(d_bpp, bpp, nparts * 4, hipMemcpyHostToDevice, s->stream));
+        HIP_CHECK(hipMemcpyAsync(d_fbp, fbp, nparts * 4, hipMemcpyHostToDevice, s->stream));
+        HIP_CHECK(hipMemcpyAsync(d_pstart, pstart, (nparts + 1) * 8, hipMemcpyHostToDevice,
+                                 s->stream));
+        hipLaunchKernelGGL(k_q3_probe_slices_fused, dim3(nb2), dim3(BLOCK), 0, s->stream,
+                           (const uint32_t*)keys_scratch->ptr,
+                           (const long long*)vals_scratch->ptr, d_bpart, d_pstart, d_bpp,
+                           d_fbp, (const uint32_t*)order_bits->ptr, at->slots, at->sums,
                            at->cap - 1);
+        HIP_CHECK(hipStreamSynchronize(s->stream));
+        (void)hipFree(d_bpart);
+        (void)hipFree(d_bpp);
+        (void)hipFree(d_fbp);
+        (void)hipFree(d_pstart);
     }
     hipLaunchKernelGGL(k_hash_agg_emit, dim3(grid_for(at->cap)), dim3(BLOCK), 0, s->stream,
                        at->slots, at->sums, at->counts, at->cap, at->cursor, max_out,
@@ -4895,6 +4975,9 @@ int gpue_q3_probe_agg_part(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_
     free(h_hist);
     free(h_off);
     free(pstart);
+    free(bpp);
+    free(fbp);
+    free(bpart);
     *n_groups = groups;
     if (groups > max_out) {
         snprintf(g_err, sizeof(g_err), "q3p: %llu groups exceed max_out %llu", groups,
