@@ -4660,6 +4660,93 @@ __global__ void k_q3_part_emit(const int64_t* __restrict__ lk,
     }
 }
 
+// v3 emit — LDS-STAGED tile sort (the standard GPU radix-partition form):
+// each 1024-row tile is counted, prefix-summed, scattered into an LDS
+// staging area CONTIGUOUS BY PARTITION, then flushed segment-by-segment to
+// the block's per-partition global cursors — writes go out as ~384 B
+// coalesced runs instead of isolated 12 B scatters (the cost that sank v1
+// at 2.9 ms and v2 at worse; profiles/q3_partitioned_probe_r01.txt).
+static constexpr int Q3P_TILE = 1024; // rows per tile = BLOCK * 4
+__global__ void k_q3_part_emit_staged(const int64_t* __restrict__ lk,
+                                      const int64_t* __restrict__ ext,
+                                      const int64_t* __restrict__ disc,
+                                      const int32_t* __restrict__ ship, uint64_t n,
+                                      int32_t ship_cutoff, uint64_t slice, uint32_t nparts,
+                                      uint64_t tile,
+                                      const uint64_t* __restrict__ block_offsets,
+                                      uint32_t* __restrict__ out_keys,
+                                      long long* __restrict__ out_vals) {
+    __shared__ uint32_t stage_k[Q3P_TILE];
+    __shared__ long long stage_v[Q3P_TILE];
+    __shared__ uint32_t cnt[256], cnt2[256], tile_off[257];
+    __shared__ unsigned long long gcur[256];
+    for (uint32_t c = threadIdx.x; c < nparts; c += blockDim.x)
+        gcur[c] = block_offsets[(uint64_t)blockIdx.x * nparts + c];
+    __syncthreads();
+    uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
+    const int K = Q3P_TILE / BLOCK;
+    for (uint64_t t0 = lo; t0 < hi; t0 += Q3P_TILE) {
+        // 1) load this tile's rows into registers
+        uint32_t rk[K];
+        long long rv[K];
+        uint32_t rp[K], pass[K];
+        #pragma unroll
+        for (int j = 0; j < K; j++) {
+            uint64_t i = t0 + threadIdx.x + (uint64_t)j * blockDim.x;
+            pass[j] = 0;
+            if (i < hi && __builtin_nontemporal_load(ship + i) > ship_cutoff) {
+                uint64_t k = (uint64_t)__builtin_nontemporal_load(lk + i);
+                rk[j] = (uint32_t)k;
+                rp[j] = (uint32_t)((k - 1) / slice);
+                rv[j] = __builtin_nontemporal_load(ext + i) *
+                        (100 - __builtin_nontemporal_load(disc + i));
+                pass[j] = 1;
+            }
+        }
+        // 2) tile histogram
+        for (uint32_t c = threadIdx.x; c < nparts; c += blockDim.x) cnt[c] = cnt2[c] = 0;
+        __syncthreads();
+        #pragma unroll
+        for (int j = 0; j < K; j++)
+            if (pass[j]) atomicAdd(&cnt[rp[j]], 1u);
+        __syncthreads();
+        // 3) tile prefix (serial over <=256 partitions — trivial)
+        if (threadIdx.x == 0) {
+            uint32_t a = 0;
+            for (uint32_t p = 0; p < nparts; p++) {
+                tile_off[p] = a;
+                a += cnt[p];
+            }
+            tile_off[nparts] = a;
+        }
+        __syncthreads();
+        // 4) scatter into the partition-contiguous LDS staging area
+        #pragma unroll
+        for (int j = 0; j < K; j++)
+            if (pass[j]) {
+                uint32_t slot = tile_off[rp[j]] + atomicAdd(&cnt2[rp[j]], 1u);
+                stage_k[slot] = rk[j];
+                stage_v[slot] = rv[j];
+            }
+        __syncthreads();
+        // 5) coalesced flush: stage index -> partition via 8-step search
+        uint32_t total = tile_off[nparts];
+        for (uint32_t j = threadIdx.x; j < total; j += blockDim.x) {
+            uint32_t lo_p = 0, hi_p = nparts;
+            while (hi_p - lo_p > 1) {
+                uint32_t mid = (lo_p + hi_p) / 2;
+                if (tile_off[mid] <= j) lo_p = mid; else hi_p = mid;
+            }
+            uint64_t dst = gcur[lo_p] + (j - tile_off[lo_p]);
+            out_keys[dst] = stage_k[j];
+            out_vals[dst] = stage_v[j];
+        }
+        __syncthreads();
+        for (uint32_t c = threadIdx.x; c < nparts; c += blockDim.x) gcur[c] += cnt[c];
+        __syncthreads();
+    }
+}
+
 // fused pass B: every block is statically assigned ONE partition
 // (block_part/block_base arrays) — one launch instead of nparts small
 // launch-bound ones; consecutive blocks share a partition so each XCD's L2
@@ -4916,12 +5003,34 @@ int gpue_q3_probe_agg_part(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_
     }
     pstart[nparts] = acc;
     HIP_CHECK(hipMalloc(&d_off, hist_len * 8));
+    if (env_cap("GPUE_Q3P_STAGED", 1) == 1) {
+        // per-BLOCK offsets (partition-major): the staged emit keeps one
+        // cursor per (block, partition)
+        uint64_t* h_boff = (uint64_t*)malloc((uint64_t)nb * nparts * 8);
+        uint64_t acc2 = 0;
+        for (uint32_t p = 0; p < nparts; p++)
+            for (uint32_t b = 0; b < nb; b++) {
+                h_boff[(uint64_t)b * nparts + p] = acc2;
+                for (uint32_t w = 0; w < nw; w++)
+                    acc2 += h_hist[((uint64_t)b * nw + w) * nparts + p];
+            }
+        HIP_CHECK(hipMemcpyAsync(d_off, h_boff, (uint64_t)nb * nparts * 8,
+                                 hipMemcpyHostToDevice, s->stream));
+        HIP_CHECK(hipStreamSynchronize(s->stream));
+        free(h_boff);
+        hipLaunchKernelGGL(k_q3_part_emit_staged, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
+                           (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
+                           ship_cutoff, slice, nparts, tile, d_off,
+                           (uint32_t*)keys_scratch->ptr, (long long*)vals_scratch->ptr);
+    } else {
     HIP_CHECK(hipMemcpyAsync(d_off, h_off, hist_len * 8, hipMemcpyHostToDevice, s->stream));
     hipLaunchKernelGGL(k_q3_part_emit, dim3(nb), dim3(BLOCK), nw * nparts * 8, s->stream,
                        (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
                        (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n, ship_cutoff,
                        slice, nparts, tile, d_off, (uint32_t*)keys_scratch->ptr,
                        (long long*)vals_scratch->ptr);
+    }
     // fused pass B: assign contiguous block ranges per partition, one launch
     uint32_t total_blocks = env_cap("GPUE_GRID_Q3S", 2048);
     if (total_blocks < nparts) total_blocks = nparts;
