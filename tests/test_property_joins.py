@@ -9,12 +9,15 @@ the oracle on fixed seeds.
 import numpy as np
 from hypothesis import given, settings, strategies as st
 
+# derandomize: the driver's round gate must be deterministic — the same
+# example set runs every time (still hypothesis-shrunk coverage)
+
 from oracle import pyoracle as orc
 
 key_lists = st.lists(st.integers(0, 50), min_size=1, max_size=120)
 
 
-@settings(max_examples=60, deadline=None)
+@settings(max_examples=60, deadline=None, derandomize=True)
 @given(build=key_lists, probe=key_lists)
 def test_bucket_chained_inner_matches_brute(build, probe):
     bkeys = np.concatenate([[0], build]).astype(np.uint32)
@@ -27,7 +30,7 @@ def test_bucket_chained_inner_matches_brute(build, probe):
     assert sorted(zip(op.tolist(), ob.tolist())) == expect
 
 
-@settings(max_examples=40, deadline=None)
+@settings(max_examples=40, deadline=None, derandomize=True)
 @given(build=key_lists, probe=key_lists, mode=st.integers(0, 3))
 def test_probe_modes_row_counts(build, probe, mode):
     """Per-mode emitted row counts follow join_hash_map.h semantics for any
@@ -49,7 +52,7 @@ def test_probe_modes_row_counts(build, probe, mode):
         assert all(probe[i] not in bset for i in op.tolist())
 
 
-@settings(max_examples=40, deadline=None)
+@settings(max_examples=40, deadline=None, derandomize=True)
 @given(st.lists(st.tuples(st.integers(0, 30), st.integers(-10**6, 10**6)),
                 min_size=1, max_size=200))
 def test_hash_agg_sum_matches_numpy(rows):
@@ -67,7 +70,7 @@ def test_hash_agg_sum_matches_numpy(rows):
     assert np.array_equal(oc[order], cnts)
 
 
-@settings(max_examples=30, deadline=None)
+@settings(max_examples=30, deadline=None, derandomize=True)
 @given(st.lists(st.binary(min_size=0, max_size=12), min_size=1, max_size=60),
        st.lists(st.binary(min_size=0, max_size=12), min_size=1, max_size=60))
 def test_slice_join_matches_brute(brows, prows):
@@ -86,7 +89,7 @@ def test_slice_join_matches_brute(brows, prows):
     assert sorted(zip(op.tolist(), ob.tolist())) == expect
 
 
-@settings(max_examples=30, deadline=None)
+@settings(max_examples=30, deadline=None, derandomize=True)
 @given(st.lists(st.integers(-2**31, 2**31 - 1), min_size=1, max_size=300),
        st.integers(0, 4), st.integers(-100, 100), st.integers(-100, 100))
 def test_eval_conjuncts_single_pred(vals, ncols_extra, lo, hi):
