@@ -100,3 +100,45 @@ def test_eval_conjuncts_single_pred(vals, ncols_extra, lo, hi):
     assert m == int(keep.sum())
     for c in cols:
         assert np.array_equal(c[:m], col[keep])
+
+
+@settings(max_examples=40, deadline=None, derandomize=True)
+@given(build=key_lists, probe=key_lists)
+def test_three_methods_agree(build, probe):
+    """BUCKET_CHAINED, LINEAR_CHAINED and RANGE_DIRECT produce the same match
+    multiset for any key set in range — the selector's method choice
+    (join_hash_table.cpp:164-344) must never change results."""
+    bkeys = np.concatenate([[0], build]).astype(np.uint32)
+    pkeys = np.array(probe, np.uint32)
+    bf, bn, bs, bl = orc.bucket_chained_build(bkeys)
+    bh = orc.bucket_chained_lookup(pkeys, bf, bs, bl)
+    ref = sorted(zip(*[x.tolist() for x in orc.probe_emit(bkeys, bn, pkeys, bh)]))
+    lf, ln, ls, ll = orc.linear_chained_build(bkeys)
+    lh = orc.linear_chained_lookup(bkeys, pkeys, lf, ls, ll)
+    got_l = sorted(zip(*[x.tolist() for x in orc.probe_emit(bkeys, ln, pkeys, lh)]))
+    assert got_l == ref
+    mn, mx = int(min(build)), int(max(build))
+    rkeys = bkeys.view(np.int32)
+    rf, rn = orc.range_direct_build(rkeys, mn, mx)
+    rh = orc.range_direct_lookup(pkeys.view(np.int32), mn, mx, rf)
+    got_r = sorted(zip(*[x.tolist() for x in orc.probe_emit(bkeys, rn, pkeys, rh)]))
+    assert got_r == ref
+
+
+@settings(max_examples=30, deadline=None, derandomize=True)
+@given(st.lists(st.integers(0, 2**32 - 1), min_size=1, max_size=400),
+       st.integers(1, 16))
+def test_partition_counting_sort_stable(keys, nch):
+    """The counting-sort partition layout is STABLE (each channel's rows
+    ascend by source row — exchange_sink_operator.cpp:636-660) and channels
+    are exactly the ReduceOp assignment."""
+    from starrocks_amd import gen
+    k = np.array(keys, np.uint32)
+    ch = orc.partition_channels(k, nch)
+    assert np.array_equal(ch, gen.partition_channels(k, nch))
+    sp, ridx = orc.partition_counting_sort(ch, nch)
+    for c in range(nch):
+        seg = ridx[sp[c]:sp[c + 1]]
+        assert np.array_equal(seg, np.sort(seg))          # stability
+        assert all(ch[i] == c for i in seg.tolist())      # correct routing
+    assert sp[-1] == len(k)
