@@ -105,7 +105,16 @@ struct gpue_session {
     hipStream_t stream2 = nullptr;
     static constexpr int N_CHUNK_EVENTS = 32;
     hipEvent_t chunk_ev[N_CHUNK_EVENTS] = {};
+    // device error latch for hash-aggregate kernels: bit 0 =
+    // sentinel key (~0ull) seen in input — the reference aggregator accepts
+    // every key value, our open-addressing table reserves ~0ull, so a real
+    // ~0ull key must fail loudly, never silently drop its group; bit 1 =
+    // probe walk exhausted the table (capacity_hint < distinct keys) —
+    // bounded walk + error instead of an infinite-spin GPU hang
+    unsigned int* d_agg_err = nullptr;
 };
+#define AGG_ERR_SENTINEL 1u
+#define AGG_ERR_FULL 2u
 
 struct gpue_dbuf {
     gpue_session* s;
@@ -170,12 +179,31 @@ int gpue_session_create(int device_index, gpue_session** out) {
     HIP_CHECK(hipStreamCreate(&s->stream2));
     for (int i = 0; i < gpue_session::N_CHUNK_EVENTS; i++)
         HIP_CHECK(hipEventCreateWithFlags(&s->chunk_ev[i], hipEventDisableTiming));
+    HIP_CHECK(hipMalloc(&s->d_agg_err, sizeof(unsigned int)));
+    HIP_CHECK(hipMemset(s->d_agg_err, 0, sizeof(unsigned int)));
     *out = s;
     return GPUE_OK;
 }
 
+// Read + clear the session's aggregate-kernel error latch. Call sites sit
+// right after a stream sync that already waits for the aggregating kernel.
+static int agg_err_check(gpue_session* s, const char* what) {
+    unsigned int e = 0;
+    HIP_CHECK(hipMemcpy(&e, s->d_agg_err, sizeof(e), hipMemcpyDeviceToHost));
+    if (e == 0) return GPUE_OK;
+    HIP_CHECK(hipMemset(s->d_agg_err, 0, sizeof(e)));
+    snprintf(g_err, sizeof(g_err), "%s: %s%s%s", what,
+             (e & AGG_ERR_SENTINEL) ? "group key 0xFFFFFFFFFFFFFFFF collides with the "
+                                      "empty-slot sentinel" : "",
+             (e & (AGG_ERR_SENTINEL | AGG_ERR_FULL)) == (AGG_ERR_SENTINEL | AGG_ERR_FULL)
+                 ? "; " : "",
+             (e & AGG_ERR_FULL) ? "hash table full (capacity < distinct keys)" : "");
+    return GPUE_ERR_ARG;
+}
+
 void gpue_session_destroy(gpue_session* s) {
     if (!s) return;
+    if (s->d_agg_err) (void)hipFree(s->d_agg_err);
     if (s->pinned) (void)hipHostFree(s->pinned);
     for (int i = 0; i < gpue_session::N_CHUNK_EVENTS; i++)
         if (s->chunk_ev[i]) (void)hipEventDestroy(s->chunk_ev[i]);
@@ -3759,6 +3787,19 @@ int gpue_ubench(gpue_session* s, int which, gpue_dbuf* a, gpue_dbuf* b, gpue_dbu
 // ---------------------------------------------------------------------------
 static constexpr unsigned long long AGG_EMPTY = 0xFFFFFFFFFFFFFFFFull;
 
+// ADVICE r01 guards: a real group key equal to the empty-slot sentinel is
+// latched as an error (the row is skipped — the host fails the call, so the
+// partial results never escape); every open-addressing walk is bounded at
+// the table capacity and latches "table full" instead of spinning forever.
+__device__ static inline bool agg_key_is_sentinel(unsigned long long k,
+                                                  unsigned int* err) {
+    if (k == AGG_EMPTY) {
+        atomicOr(err, AGG_ERR_SENTINEL);
+        return true;
+    }
+    return false;
+}
+
 struct gpue_agg_table {
     gpue_session* s;
     uint64_t cap;
@@ -3815,12 +3856,14 @@ __global__ void k_hash_agg_sum(const uint64_t* __restrict__ keys,
                                unsigned long long* __restrict__ slots,
                                unsigned long long* __restrict__ sums,
                                unsigned long long* __restrict__ counts,
-                               uint64_t cap_mask) {
+                               uint64_t cap_mask, unsigned int* __restrict__ err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         unsigned long long k = keys[i];
+        if (agg_key_is_sentinel(k, err)) continue;
         unsigned long long v = (unsigned long long)vals[i];
         uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        uint64_t left = cap_mask + 1;
         for (;;) {
             unsigned long long cur = slots[s];
             if (cur == k) {
@@ -3837,6 +3880,7 @@ __global__ void k_hash_agg_sum(const uint64_t* __restrict__ keys,
                 }
             }
             s = (s + 1) & cap_mask;
+            if (--left == 0) { atomicOr(err, AGG_ERR_FULL); break; }
         }
     }
 }
@@ -3864,12 +3908,16 @@ __global__ void k_hash_agg_stats(const uint64_t* __restrict__ keys,
                                  unsigned long long* __restrict__ sums,
                                  unsigned long long* __restrict__ counts,
                                  long long* __restrict__ mins,
-                                 long long* __restrict__ maxs, uint64_t cap_mask) {
+                                 long long* __restrict__ maxs, uint64_t cap_mask,
+                                 unsigned int* __restrict__ err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         unsigned long long k = keys[i];
+        if (agg_key_is_sentinel(k, err)) continue;
         long long v = (long long)vals[i];
         uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        uint64_t left = cap_mask + 1;
+        bool ok = true;
         for (;;) {
             unsigned long long cur = slots[s];
             if (cur == k) break;
@@ -3878,7 +3926,9 @@ __global__ void k_hash_agg_stats(const uint64_t* __restrict__ keys,
                 if (old == AGG_EMPTY || old == k) break;
             }
             s = (s + 1) & cap_mask;
+            if (--left == 0) { atomicOr(err, AGG_ERR_FULL); ok = false; break; }
         }
+        if (!ok) continue;
         atomicAdd(&sums[s], (unsigned long long)v);
         atomicAdd(&counts[s], 1ull);
         atomicMin(&mins[s], v);
@@ -3890,12 +3940,16 @@ __global__ void k_hash_agg_sum128(const uint64_t* __restrict__ keys,
                                   const int64_t* __restrict__ vals, uint64_t n,
                                   unsigned long long* __restrict__ slots,
                                   unsigned long long* __restrict__ lo,
-                                  unsigned long long* __restrict__ hi, uint64_t cap_mask) {
+                                  unsigned long long* __restrict__ hi, uint64_t cap_mask,
+                                  unsigned int* __restrict__ err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         unsigned long long k = keys[i];
+        if (agg_key_is_sentinel(k, err)) continue;
         long long v = (long long)vals[i];
         uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        uint64_t left = cap_mask + 1;
+        bool ok = true;
         for (;;) {
             unsigned long long cur = slots[s];
             if (cur == k) break;
@@ -3904,7 +3958,9 @@ __global__ void k_hash_agg_sum128(const uint64_t* __restrict__ keys,
                 if (old == AGG_EMPTY || old == k) break;
             }
             s = (s + 1) & cap_mask;
+            if (--left == 0) { atomicOr(err, AGG_ERR_FULL); ok = false; break; }
         }
+        if (!ok) continue;
         unsigned long long vlo = (unsigned long long)v;
         long long vhi = v < 0 ? -1ll : 0ll; // sign extension of the int64 addend
         unsigned long long old_lo = atomicAdd(&lo[s], vlo);
@@ -4012,7 +4068,7 @@ int gpue_hash_agg_stats_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, u
                        d_mins, d_maxs, cap);
     hipLaunchKernelGGL(k_hash_agg_stats, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
                        (const uint64_t*)keys->ptr, (const int64_t*)vals->ptr, n, d_slots,
-                       d_sums, d_counts, d_mins, d_maxs, cap - 1);
+                       d_sums, d_counts, d_mins, d_maxs, cap - 1, s->d_agg_err);
     hipLaunchKernelGGL(k_hash_agg_emit_wide, dim3(grid_for(cap)), dim3(BLOCK), 0, s->stream,
                        d_slots, d_sums, d_counts, d_mins, d_maxs, cap, d_cursor, max_out,
                        (uint64_t*)out_keys->ptr, (int64_t*)out_sums->ptr,
@@ -4024,6 +4080,8 @@ int gpue_hash_agg_stats_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, u
     (void)hipFree(d_slots); (void)hipFree(d_sums); (void)hipFree(d_counts);
     (void)hipFree(d_mins); (void)hipFree(d_maxs); (void)hipFree(d_cursor);
     *n_groups = groups;
+    int erc = agg_err_check(s, "agg_stats");
+    if (erc != GPUE_OK) return erc;
     if (groups > max_out) {
         snprintf(g_err, sizeof(g_err), "agg_stats: %llu groups exceed max_out", groups);
         return GPUE_ERR_ARG;
@@ -4050,7 +4108,7 @@ int gpue_hash_agg_sum128_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, 
     HIP_CHECK(hipMemsetAsync(d_cursor, 0, 8, s->stream));
     hipLaunchKernelGGL(k_hash_agg_sum128, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
                        (const uint64_t*)keys->ptr, (const int64_t*)vals->ptr, n, d_slots,
-                       d_lo, d_hi, cap - 1);
+                       d_lo, d_hi, cap - 1, s->d_agg_err);
     hipLaunchKernelGGL(k_hash_agg_emit_wide, dim3(grid_for(cap)), dim3(BLOCK), 0, s->stream,
                        d_slots, d_lo, d_hi, nullptr, nullptr, cap, d_cursor, max_out,
                        (uint64_t*)out_keys->ptr, (int64_t*)out_lo->ptr,
@@ -4060,6 +4118,8 @@ int gpue_hash_agg_sum128_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, 
     HIP_CHECK(hipStreamSynchronize(s->stream));
     (void)hipFree(d_slots); (void)hipFree(d_lo); (void)hipFree(d_hi); (void)hipFree(d_cursor);
     *n_groups = groups;
+    int erc = agg_err_check(s, "agg_sum128");
+    if (erc != GPUE_OK) return erc;
     if (groups > max_out) {
         snprintf(g_err, sizeof(g_err), "agg_sum128: %llu groups exceed max_out", groups);
         return GPUE_ERR_ARG;
@@ -4159,7 +4219,7 @@ int gpue_hash_agg_sum_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uin
     HIP_CHECK(hipMemsetAsync(d_cursor, 0, 8, s->stream));
     hipLaunchKernelGGL(k_hash_agg_sum, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
                        (const uint64_t*)keys->ptr, (const int64_t*)vals->ptr, n,
-                       d_slots, d_sums, d_counts, cap - 1);
+                       d_slots, d_sums, d_counts, cap - 1, s->d_agg_err);
     hipLaunchKernelGGL(k_hash_agg_emit, dim3(grid_for(cap)), dim3(BLOCK), 0, s->stream,
                        d_slots, d_sums, d_counts, cap, d_cursor, max_out,
                        (uint64_t*)out_keys->ptr, (int64_t*)out_sums->ptr,
@@ -4172,6 +4232,8 @@ int gpue_hash_agg_sum_u64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uin
     (void)hipFree(d_counts);
     (void)hipFree(d_cursor);
     *n_groups = groups;
+    int erc = agg_err_check(s, "hash_agg");
+    if (erc != GPUE_OK) return erc;
     if (groups > max_out) {
         snprintf(g_err, sizeof(g_err), "hash_agg: %llu groups exceed max_out %llu",
                  groups, (unsigned long long)max_out);
@@ -4201,14 +4263,17 @@ __global__ void k_hash_agg_push(const uint64_t* __restrict__ keys,
                                 unsigned long long* __restrict__ sums,
                                 unsigned long long* __restrict__ counts, uint64_t cap_mask,
                                 int update_only, uint8_t* __restrict__ miss_mask,
-                                unsigned long long* __restrict__ hits_out) {
+                                unsigned long long* __restrict__ hits_out,
+                                unsigned int* __restrict__ err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     unsigned long long local_hits = 0;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         unsigned long long k = keys[i];
+        if (agg_key_is_sentinel(k, err)) { if (miss_mask) miss_mask[i] = 1; continue; }
         unsigned long long v = vals ? (unsigned long long)vals[i] : 0ull;
         unsigned long long c = cnts ? (unsigned long long)cnts[i] : 1ull;
         uint64_t slot = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        uint64_t left = cap_mask + 1;
         for (;;) {
             unsigned long long cur = slots[slot];
             if (cur == k) {
@@ -4238,6 +4303,11 @@ __global__ void k_hash_agg_push(const uint64_t* __restrict__ keys,
                 continue; // lost the claim to another key: re-read this slot
             }
             slot = (slot + 1) & cap_mask;
+            if (--left == 0) {
+                atomicOr(err, AGG_ERR_FULL);
+                if (miss_mask) miss_mask[i] = 1;
+                break;
+            }
         }
     }
     if (hits_out) {
@@ -4276,7 +4346,8 @@ static int hash_agg_push_impl(gpue_session* s, gpue_agg_table* at, gpue_dbuf* ke
                        (const uint64_t*)keys->ptr, vals ? (const int64_t*)vals->ptr : nullptr,
                        cnts ? (const int64_t*)cnts->ptr : nullptr, n, at->slots, at->sums,
                        at->counts, at->cap - 1, update_only,
-                       miss_mask ? (uint8_t*)miss_mask->ptr : nullptr, d_hits);
+                       miss_mask ? (uint8_t*)miss_mask->ptr : nullptr, d_hits,
+                       s->d_agg_err);
     HIP_CHECK(hipGetLastError());
     if (hits_out) {
         unsigned long long h = 0;
@@ -4314,6 +4385,8 @@ int gpue_hash_agg_emit_u64(gpue_session* s, gpue_agg_table* at, gpue_dbuf* out_k
     HIP_CHECK(hipMemcpyAsync(&groups, at->cursor, 8, hipMemcpyDeviceToHost, s->stream));
     HIP_CHECK(hipStreamSynchronize(s->stream));
     *n_groups = groups;
+    int erc = agg_err_check(s, "hash_agg");
+    if (erc != GPUE_OK) return erc;
     if (groups > max_out) {
         snprintf(g_err, sizeof(g_err), "hash_agg_emit: %llu groups exceed max_out %llu",
                  groups, (unsigned long long)max_out);
@@ -4522,7 +4595,8 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
                                const uint32_t* __restrict__ order_bits, int32_t ship_cutoff,
                                unsigned long long* __restrict__ slots,
                                unsigned long long* __restrict__ sums,
-                               unsigned long long* __restrict__ counts, uint64_t cap_mask) {
+                               unsigned long long* __restrict__ counts, uint64_t cap_mask,
+                               unsigned int* __restrict__ err) {
     (void)counts;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
@@ -4537,6 +4611,7 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
         unsigned long long v = (unsigned long long)(__builtin_nontemporal_load(ext + i) *
                                                     (100 - __builtin_nontemporal_load(disc + i)));
         uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        uint64_t left = cap_mask + 1;
         for (;;) {
             unsigned long long cur = slots[s];
             if (cur == k) { atomicAdd(&sums[s], v); break; }
@@ -4545,6 +4620,7 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
                 if (old == AGG_EMPTY || old == k) { atomicAdd(&sums[s], v); break; }
             }
             s = (s + 1) & cap_mask;
+            if (--left == 0) { atomicOr(err, AGG_ERR_FULL); break; }
         }
     }
 }
@@ -4760,7 +4836,7 @@ __global__ void k_q3_probe_slices_fused(const uint32_t* __restrict__ keys,
                                         const uint32_t* __restrict__ order_bits,
                                         unsigned long long* __restrict__ slots,
                                         unsigned long long* __restrict__ sums,
-                                        uint64_t cap_mask) {
+                                        uint64_t cap_mask, unsigned int* __restrict__ err) {
     uint32_t p = block_part[blockIdx.x];
     uint64_t base = pstart[p];
     uint64_t cnt = pstart[p + 1] - base;
@@ -4773,6 +4849,7 @@ __global__ void k_q3_probe_slices_fused(const uint32_t* __restrict__ keys,
         if (!((order_bits[o >> 5] >> (o & 31)) & 1u)) continue;
         unsigned long long v = (unsigned long long)vals[base + i];
         uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        uint64_t left = cap_mask + 1;
         for (;;) {
             unsigned long long cur_ = slots[s];
             if (cur_ == k) { atomicAdd(&sums[s], v); break; }
@@ -4781,6 +4858,7 @@ __global__ void k_q3_probe_slices_fused(const uint32_t* __restrict__ keys,
                 if (old == AGG_EMPTY || old == k) { atomicAdd(&sums[s], v); break; }
             }
             s = (s + 1) & cap_mask;
+            if (--left == 0) { atomicOr(err, AGG_ERR_FULL); break; }
         }
     }
 }
@@ -4789,7 +4867,8 @@ __global__ void k_q3_probe_slice(const uint32_t* __restrict__ keys,
                                  const long long* __restrict__ vals, uint64_t n,
                                  const uint32_t* __restrict__ order_bits,
                                  unsigned long long* __restrict__ slots,
-                                 unsigned long long* __restrict__ sums, uint64_t cap_mask) {
+                                 unsigned long long* __restrict__ sums, uint64_t cap_mask,
+                                 unsigned int* __restrict__ err) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         unsigned long long k = keys[i];
@@ -4797,6 +4876,7 @@ __global__ void k_q3_probe_slice(const uint32_t* __restrict__ keys,
         if (!((order_bits[o >> 5] >> (o & 31)) & 1u)) continue;
         unsigned long long v = (unsigned long long)vals[i];
         uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        uint64_t left = cap_mask + 1;
         for (;;) {
             unsigned long long cur = slots[s];
             if (cur == k) { atomicAdd(&sums[s], v); break; }
@@ -4805,6 +4885,7 @@ __global__ void k_q3_probe_slice(const uint32_t* __restrict__ keys,
                 if (old == AGG_EMPTY || old == k) { atomicAdd(&sums[s], v); break; }
             }
             s = (s + 1) & cap_mask;
+            if (--left == 0) { atomicOr(err, AGG_ERR_FULL); break; }
         }
     }
 }
@@ -4938,7 +5019,7 @@ int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbu
                        (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
                        (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
                        (const uint32_t*)order_bits->ptr, ship_cutoff, at->slots, at->sums,
-                       at->counts, at->cap - 1);
+                       at->counts, at->cap - 1, s->d_agg_err);
     hipLaunchKernelGGL(k_hash_agg_emit, dim3(grid_for(at->cap)), dim3(BLOCK), 0, s->stream,
                        at->slots, at->sums, at->counts, at->cap, at->cursor, max_out,
                        (uint64_t*)out_keys->ptr, (int64_t*)out_sums->ptr, nullptr);
@@ -4946,6 +5027,8 @@ int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbu
     HIP_CHECK(hipMemcpyAsync(&groups, at->cursor, 8, hipMemcpyDeviceToHost, s->stream));
     HIP_CHECK(hipStreamSynchronize(s->stream));
     *n_groups = groups;
+    int erc = agg_err_check(s, "q3");
+    if (erc != GPUE_OK) return erc;
     if (groups > max_out) {
         snprintf(g_err, sizeof(g_err), "q3: %llu groups exceed max_out %llu", groups,
                  (unsigned long long)max_out);
@@ -5066,7 +5149,7 @@ int gpue_q3_probe_agg_part(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_
                            (const uint32_t*)keys_scratch->ptr,
                            (const long long*)vals_scratch->ptr, d_bpart, d_pstart, d_bpp,
                            d_fbp, (const uint32_t*)order_bits->ptr, at->slots, at->sums,
-                           at->cap - 1);
+                           at->cap - 1, s->d_agg_err);
         HIP_CHECK(hipStreamSynchronize(s->stream));
         (void)hipFree(d_bpart);
         (void)hipFree(d_bpp);
@@ -5118,7 +5201,7 @@ int gpue_q3_probe_agg(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf*
                        (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
                        (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
                        (const uint32_t*)order_bits->ptr, ship_cutoff, d_slots, d_sums,
-                       d_counts, cap - 1);
+                       d_counts, cap - 1, s->d_agg_err);
     hipLaunchKernelGGL(k_hash_agg_emit, dim3(grid_for(cap)), dim3(BLOCK), 0, s->stream,
                        d_slots, d_sums, d_counts, cap, d_cursor, max_out,
                        (uint64_t*)out_keys->ptr, (int64_t*)out_sums->ptr, nullptr);
@@ -5130,6 +5213,8 @@ int gpue_q3_probe_agg(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf*
     (void)hipFree(d_counts);
     (void)hipFree(d_cursor);
     *n_groups = groups;
+    int erc = agg_err_check(s, "q3");
+    if (erc != GPUE_OK) return erc;
     if (groups > max_out) {
         snprintf(g_err, sizeof(g_err), "q3: %llu groups exceed max_out %llu", groups,
                  (unsigned long long)max_out);

@@ -222,3 +222,18 @@ def build_part_dim_payload(seed: int, n_parts: int, category: int) -> np.ndarray
 def build_supp_dim_payload(seed: int, n_supps: int, region: int) -> np.ndarray:
     skeys = np.arange(1, n_supps + 1, dtype=np.uint64)
     return (supp_region(seed, skeys) == region).astype(np.uint32)
+
+
+# TPC-H c_mktsegment 5-value dictionary, 16-byte space-padded fixed strings
+# (the SERIALIZED_FIXED_SIZE ≤16 B key packing of join_key_constructor.h:40-153
+# applied to the Q3 string literal). Product-path copy of the constant the
+# oracle also restates (oracle.c MKT_SEGMENTS) — kept here so the product path
+# has zero oracle imports (oracle/ is test infrastructure only);
+# tests/test_gen_consistency.py asserts the two stay identical.
+MKT_SEGMENTS = [b"AUTOMOBILE      ", b"BUILDING        ", b"FURNITURE       ",
+                b"MACHINERY       ", b"HOUSEHOLD       "]
+
+
+def mkt_literal(idx: int = 1) -> bytes:
+    """16-byte mktsegment literal (Q3 filters c_mktsegment = 'BUILDING')."""
+    return MKT_SEGMENTS[idx]

@@ -127,3 +127,46 @@ def test_hash_agg_single_group_gpu(engine):
     assert oc_b.d2h(np.int64, 1)[0] == n
     for b in (kb, vb, ok_b, os_b, oc_b):
         b.free()
+
+
+def test_hash_agg_sentinel_key_errors_gpu(engine):
+    """a real group key equal to the empty-slot sentinel (~0ull)
+    must raise, not silently drop the group (the reference aggregator
+    accepts every key value, agg_hash_map.h:112-290)."""
+    from starrocks_amd.engine import GpueError
+    n = 1024
+    keys = np.arange(1, n + 1, dtype=np.uint64)
+    keys[100] = 0xFFFFFFFFFFFFFFFF
+    vals = np.ones(n, np.int64)
+    kb, vb = engine.alloc(n * 8), engine.alloc(n * 8)
+    kb.h2d(keys)
+    vb.h2d(vals)
+    ok_b, os_b = engine.alloc(4096 * 8), engine.alloc(4096 * 8)
+    with pytest.raises(GpueError, match="sentinel"):
+        engine.hash_agg_sum_u64(kb, vb, n, ok_b, os_b, max_out=4096)
+    # the latch clears on read: a clean call on the same session succeeds
+    keys[100] = 100
+    kb.h2d(keys)
+    g = engine.hash_agg_sum_u64(kb, vb, n, ok_b, os_b, max_out=4096)
+    assert g == n - 1  # key 100 appears twice
+    for b in (kb, vb, ok_b, os_b):
+        b.free()
+
+
+def test_hash_agg_capacity_overflow_errors_gpu(engine):
+    """capacity_hint < distinct keys must return an error, never
+    spin forever (bounded probe walk)."""
+    from starrocks_amd.engine import GpueError
+    n = 4096
+    keys = np.arange(1, n + 1, dtype=np.uint64)
+    vals = np.ones(n, np.int64)
+    kb, vb = engine.alloc(n * 8), engine.alloc(n * 8)
+    kb.h2d(keys)
+    vb.h2d(vals)
+    ok_b, os_b = engine.alloc(n * 8), engine.alloc(n * 8)
+    with pytest.raises(GpueError, match="full"):
+        engine.hash_agg_sum_u64(kb, vb, n, ok_b, os_b, max_out=n, capacity_hint=64)
+    g = engine.hash_agg_sum_u64(kb, vb, n, ok_b, os_b, max_out=n)
+    assert g == n
+    for b in (kb, vb, ok_b, os_b):
+        b.free()

@@ -67,3 +67,12 @@ def test_partition_channels_2xi32_oracle_vs_numpy():
         # chain property: equals fnv(b, seed=fnv(a, FNV_SEED)) reduced
         h = gen.fnv_u32_seeded(b, gen.fnv_u32(a))
         assert np.array_equal(out, ((h.astype(np.uint64) * nch) >> np.uint64(32)).astype(np.uint32))
+
+
+def test_mkt_segments_match_oracle():
+    """gen.MKT_SEGMENTS (product path) must equal the oracle's restatement of
+    the TPC-H c_mktsegment dictionary (oracle.c MKT_SEGMENTS) — the product
+    path may not import oracle/ (DESIGN.md §2), so the constant lives twice."""
+    for i in range(5):
+        assert gen.mkt_literal(i) == orc.mkt_literal(i)
+        assert len(gen.mkt_literal(i)) == 16
