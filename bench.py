@@ -48,6 +48,18 @@ def log(msg):
 
 
 def build_payload_table(eng, keys_i32, payload_u32):
+    # Plan step: run the reference's JoinHashMapSelector on this build side
+    # (join_hash_table.cpp:164-344, restated as gpue_join_select_method). The
+    # SSB dims are dense unique int keys, so the decision must land in the
+    # RANGE_DIRECT family — whose MI355X fused form is the payload table
+    # (first[key-min] holds payload+1). Any other decision
+    # means the workload changed and the fused kernel no longer applies.
+    m = eng.join_select_method(0 /*ONE_KEY*/, 1 /*LT_INT*/, len(keys_i32),
+                               int(keys_i32.min()), int(keys_i32.max()))
+    if eng.JM_NAMES[m] not in ("RANGE_DIRECT", "RANGE_DIRECT_SET",
+                               "DENSE_RANGE_DIRECT", "DIRECT"):
+        raise RuntimeError(f"selector chose {eng.JM_NAMES[m]}; the fused "
+                           "payload-table path requires a direct-mapped plan")
     k = eng.alloc(keys_i32.nbytes)
     k.h2d(keys_i32)
     p = eng.alloc(payload_u32.nbytes)

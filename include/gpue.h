@@ -160,6 +160,70 @@ int gpue_join_probe_emit_mode_u64(gpue_session* s, gpue_join_table* t, gpue_dbuf
  * is nondeterministic — the emitted match multiset is identical). */
 int gpue_join_build_range_direct_i32(gpue_session* s, gpue_dbuf* keys /*i32, 1-based row 0 sentinel*/,
                                      uint64_t row_count, gpue_join_table** out);
+
+/* ---- JoinHashMapSelector (join_hash_table.cpp:164-344) --------------------
+ * The reference's automatic key-constructor + map-method decision, restated
+ * as pure host functions. Probe-mode values double as the join_type input.
+ */
+#define GPUE_JOIN_INNER 0
+#define GPUE_JOIN_LEFT_SEMI 1
+#define GPUE_JOIN_LEFT_ANTI 2
+#define GPUE_JOIN_LEFT_OUTER 3
+/* key-constructor classes (JoinKeyConstructorUnaryType, :164-229) */
+#define GPUE_KEYCON_ONE_KEY 0
+#define GPUE_KEYCON_ONE_KEY_VARCHAR 1
+#define GPUE_KEYCON_FIXED_INT 2      /* SERIALIZED_FIXED_SIZE_INT, packed <= 4 B */
+#define GPUE_KEYCON_FIXED_BIGINT 3   /* packed <= 8 B */
+#define GPUE_KEYCON_FIXED_LARGEINT 4 /* packed <= 16 B */
+#define GPUE_KEYCON_SERIALIZED_VARCHAR 5
+/* map methods (JoinHashMapMethodType, join_hash_map_method.h) */
+#define GPUE_JM_DIRECT 0
+#define GPUE_JM_RANGE_DIRECT 1
+#define GPUE_JM_RANGE_DIRECT_SET 2
+#define GPUE_JM_DENSE_RANGE_DIRECT 3
+#define GPUE_JM_LINEAR_CHAINED 4
+#define GPUE_JM_LINEAR_CHAINED_SET 5
+#define GPUE_JM_BUCKET_CHAINED 6
+/* logical-type classes for the method decision */
+#define GPUE_LT_TINY 0    /* BOOLEAN/TINYINT/SMALLINT -> DIRECT_MAPPING (:239) */
+#define GPUE_LT_INT 1
+#define GPUE_LT_BIGINT 2
+#define GPUE_LT_OTHER 3   /* other fixed-width (largeint/decimal/date...) */
+#define GPUE_LT_VARCHAR 4
+
+/* _determine_key_constructor (:164-229). fixed_sizes[i]: key column i's
+ * fixed byte width; for varchar keys pass _get_binary_column_max_size's
+ * result (1..16 when the fixed-size-string opt applies, else 0).
+ * packed_bytes_out: total packed width (0 when not fixed-packed). */
+int gpue_join_select_key_constructor(int num_keys, const int32_t* fixed_sizes,
+                                     const uint8_t* null_safe,
+                                     int enable_fixed_size_string,
+                                     int32_t* packed_bytes_out);
+/* single-VARCHAR refinement (:178-194) */
+int gpue_join_select_varchar_constructor(int32_t max_size, int enable_fixed_size_string);
+/* _determine_hash_map_method (:231-256) + range-direct (:270-321) + linear
+ * (:323-344) gates. min/max_value: the single int key's bounds over build
+ * rows (ignored unless the range-direct gate applies). l2_size/l3_size:
+ * reference reads CpuInfo L2 and halves L3 itself (:295-296). Returns a
+ * GPUE_JM_* value. */
+int gpue_join_select_method(int key_constructor, int lt_class, uint64_t row_count,
+                            int64_t min_value, int64_t max_value, int mode,
+                            int with_other_conjunct, int enable_range_direct,
+                            int enable_linear_chained, uint64_t l2_size,
+                            uint64_t l3_size);
+/* Auto build for a single i32 key column (keys 1-based, row 0 sentinel):
+ * computes the build keys' min/max on device, runs the selector with the
+ * reference's default-on session flags, and builds the matching GPU table.
+ * l2_size/l3_size 0 -> MI355X defaults (4 MiB XCD L2 / 256 MiB Infinity
+ * Cache). method_out (nullable) receives the GPUE_JM_* decision. Mapping to
+ * physical layouts: DIRECT/RANGE_DIRECT/RANGE_DIRECT_SET/DENSE map onto the
+ * u32 first[] direct-mapped table (the 1-bit SET and 2-bit DENSE packings
+ * are CPU-cache idioms; with interval < 2^32 the u32 array is HBM-resident
+ * and probes in ONE load — DESIGN.md (S)3), LINEAR_CHAINED(+SET) onto the
+ * fp-packed linear-probed table, BUCKET_CHAINED onto first/next chains. */
+int gpue_join_build_auto_i32(gpue_session* s, gpue_dbuf* keys, uint64_t row_count,
+                             int mode, int with_other_conjunct, uint64_t l2_size,
+                             uint64_t l3_size, gpue_join_table** out, int* method_out);
 void gpue_join_table_destroy(gpue_join_table* t);
 int gpue_join_table_minmax(gpue_join_table* t, int64_t* min_out, int64_t* max_out);
 /* d2h the first[] array (tests) */

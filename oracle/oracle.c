@@ -101,6 +101,148 @@ uint32_t orc_calc_bucket_size(uint32_t size) {
     return (uint32_t)(norm + 1);
 }
 
+
+/* ====================================================================== */
+/* JoinHashMapSelector restatement (join_hash_table.cpp:164-344):          */
+/* _determine_key_constructor (:164-229) picks how the join key column(s)  */
+/* are materialized; _determine_hash_map_method (:231-256) +               */
+/* _try_use_range_direct_mapping (:270-321) + _try_use_linear_chained      */
+/* (:323-344) pick the map. Session flags default TRUE                     */
+/* (SessionVariable.java:2060-2067); cache sizes are parameters (the       */
+/* reference reads CpuInfo L2 / half-L3 at :295-296).                      */
+/* ====================================================================== */
+
+/* key-constructor results (JoinKeyConstructorUnaryType classes) */
+enum {
+    ORC_KEYCON_ONE_KEY = 0,            /* single fixed-width key, used as-is */
+    ORC_KEYCON_ONE_KEY_VARCHAR = 1,    /* single Slice key (TYPE_VARCHAR) */
+    ORC_KEYCON_FIXED_INT = 2,          /* packed <= 4 B  (SERIALIZED_FIXED_SIZE_INT) */
+    ORC_KEYCON_FIXED_BIGINT = 3,       /* packed <= 8 B  */
+    ORC_KEYCON_FIXED_LARGEINT = 4,     /* packed <= 16 B */
+    ORC_KEYCON_SERIALIZED_VARCHAR = 5, /* variable-length serialization */
+};
+/* hash-map methods (JoinHashMapMethodType) */
+enum {
+    ORC_JM_DIRECT = 0,
+    ORC_JM_RANGE_DIRECT = 1,
+    ORC_JM_RANGE_DIRECT_SET = 2,
+    ORC_JM_DENSE_RANGE_DIRECT = 3,
+    ORC_JM_LINEAR_CHAINED = 4,
+    ORC_JM_LINEAR_CHAINED_SET = 5,
+    ORC_JM_BUCKET_CHAINED = 6,
+};
+/* logical-type classes relevant to the method decision */
+enum {
+    ORC_LT_TINY = 0,    /* BOOLEAN/TINYINT/SMALLINT -> DIRECT_MAPPING */
+    ORC_LT_INT = 1,     /* TYPE_INT */
+    ORC_LT_BIGINT = 2,  /* TYPE_BIGINT */
+    ORC_LT_OTHER = 3,   /* other fixed types (largeint/date/decimal/...) */
+    ORC_LT_VARCHAR = 4,
+};
+
+/* _determine_key_constructor (join_hash_table.cpp:164-229).
+ * fixed_sizes[i]: bytes of key column i when fixed-width; for varchar keys
+ * the caller passes _get_binary_column_max_size's result (1..16 when the
+ * fixed-size-string optimization applies, else 0). null_safe[i]: the key is
+ * a null-safe equal (<=>), which keeps a null byte in the packing (:211).
+ * Returns the constructor class; *packed_bytes_out = the packed key width
+ * (serialized_fixed_size_key_bytes total), 0 when not fixed-packed. */
+int orc_join_select_key_constructor(int num_keys, const int32_t* fixed_sizes,
+                                    const uint8_t* null_safe,
+                                    int enable_fixed_size_string,
+                                    int32_t* packed_bytes_out) {
+    *packed_bytes_out = 0;
+    if (num_keys == 1 && !null_safe[0]) { /* :175 */
+        int32_t sz = fixed_sizes[0];
+        if (sz > 0) {
+            /* single varchar with the fixed-size-string opt ON takes the
+             * packed path (:178-194); a fixed-width type stays ONE_KEY */
+            *packed_bytes_out = sz;
+            return ORC_KEYCON_ONE_KEY;
+        }
+        return ORC_KEYCON_ONE_KEY_VARCHAR;
+    }
+    /* multi-key (or null-safe single): sum fixed widths (+1 null byte per
+     * null-safe key); any un-fixable varchar forces full serialization
+     * (:199-216) */
+    int64_t total = 0;
+    for (int i = 0; i < num_keys; i++) {
+        int32_t cur = fixed_sizes[i];
+        if (cur <= 0) return ORC_KEYCON_SERIALIZED_VARCHAR;
+        cur += null_safe[i] ? 1 : 0;
+        total += cur;
+    }
+    if (total > 16) { // > 16 B never fixed-packs; packed width unused
+        return ORC_KEYCON_SERIALIZED_VARCHAR;
+    }
+    *packed_bytes_out = (int32_t)total;
+    if (total <= 4) return ORC_KEYCON_FIXED_INT;       /* :219 */
+    if (total <= 8) return ORC_KEYCON_FIXED_BIGINT;    /* :222 */
+    if (total <= 16) return ORC_KEYCON_FIXED_LARGEINT; /* :225 */
+    return ORC_KEYCON_SERIALIZED_VARCHAR;              /* :229 */
+}
+
+/* single-VARCHAR refinement of the above (:178-194): when the one
+ * non-null-safe key is varchar and max_size in (0,16], it packs fixed. */
+int orc_join_select_varchar_constructor(int32_t max_size,
+                                        int enable_fixed_size_string) {
+    if (!enable_fixed_size_string || max_size <= 0) return ORC_KEYCON_ONE_KEY_VARCHAR;
+    if (max_size <= 4) return ORC_KEYCON_FIXED_INT;
+    if (max_size <= 8) return ORC_KEYCON_FIXED_BIGINT;
+    if (max_size <= 16) return ORC_KEYCON_FIXED_LARGEINT;
+    return ORC_KEYCON_ONE_KEY_VARCHAR;
+}
+
+/* _determine_hash_map_method (:231-256) with _try_use_range_direct_mapping
+ * (:270-321) and _try_use_linear_chained (:323-344). mode: 0 INNER, 1
+ * LEFT_SEMI, 2 LEFT_ANTI, 3 LEFT_OUTER, 4 RIGHT_SEMI, 5 RIGHT_ANTI (the
+ * gpue probe-mode encoding). min/max_value: the single int key's bounds
+ * over build rows 1..row_count (ignored unless the range-direct gate
+ * applies). l2_size/l3_size: CpuInfo::get_l2_cache_size() and FULL L3 (the
+ * reference halves L3 itself at :295). */
+int orc_join_select_method(int key_constructor, int lt_class, uint64_t row_count,
+                           int64_t min_value, int64_t max_value, int mode,
+                           int with_other_conjunct, int enable_range_direct,
+                           int enable_linear_chained, uint64_t l2_size,
+                           uint64_t l3_size) {
+    if (lt_class == ORC_LT_TINY) return ORC_JM_DIRECT; /* :239 */
+    const int semi_or_anti_no_conj = (mode == 1 || mode == 2) && !with_other_conjunct;
+    uint64_t rc_plus_1 = row_count + 1;
+    const uint64_t bucket_size =
+        orc_calc_bucket_size(rc_plus_1 > 0xFFFFFFFFull ? 0xFFFFFFFFu : (uint32_t)rc_plus_1);
+
+    if (key_constructor == ORC_KEYCON_ONE_KEY &&
+        (lt_class == ORC_LT_INT || lt_class == ORC_LT_BIGINT)) { /* :242 */
+        if (enable_range_direct && row_count > 0) { /* :274 */
+            /* overflow guards (:283-291) */
+            if (!(min_value == INT64_MIN && max_value == INT64_MAX)) {
+                uint64_t interval = (uint64_t)max_value - (uint64_t)min_value + 1;
+                if (interval < 0xFFFFFFFFull) { /* :288 value_interval < UINT32_MAX */
+                    if (semi_or_anti_no_conj) { /* :301 */
+                        uint64_t memory = (interval + 7) / 8;
+                        /* one bit vs 8 bytes of first+next (:303) */
+                        if (memory <= bucket_size * 64 || memory <= l3_size / 2)
+                            return ORC_JM_RANGE_DIRECT_SET;
+                    } else {
+                        if (interval <= bucket_size || interval <= l2_size) /* :307 */
+                            return ORC_JM_RANGE_DIRECT;
+                        /* 2-bit dense groups + u32/row vs bucket-chained + 10%
+                         * headroom (:310-317) */
+                        if (interval / 4 + row_count * 4 <=
+                            (bucket_size + bucket_size / 10) * 4)
+                            return ORC_JM_DENSE_RANGE_DIRECT;
+                    }
+                }
+            }
+        }
+    }
+    /* _try_use_linear_chained (:323-344): bucket count must fit the 24-bit
+     * fp-packed index (join_hash_map_method.h:135-139 DATA_MASK) */
+    if (enable_linear_chained && bucket_size <= 0xFFFFFFu)
+        return semi_or_anti_no_conj ? ORC_JM_LINEAR_CHAINED_SET : ORC_JM_LINEAR_CHAINED;
+    return ORC_JM_BUCKET_CHAINED; /* _get_fallback_method :258-263 (asof out of scope) */
+}
+
 /* ====================================================================== */
 /* Deterministic synthetic data — splitmix64 finalizer, counter-based.     */
 /* Shared definition with the HIP generator (csrc/gpue_kernels.hip) and    */

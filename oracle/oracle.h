@@ -26,6 +26,19 @@ uint32_t orc_join_hash_u64(uint64_t v, uint32_t num_log_buckets);
 uint32_t orc_join_hash_slice(const void* p, int32_t n, uint32_t num_buckets);
 uint32_t orc_calc_bucket_size(uint32_t size);
 
+/* JoinHashMapSelector restatement (join_hash_table.cpp:164-344); enums and
+ * parameter semantics documented at the definitions in oracle.c */
+int orc_join_select_key_constructor(int num_keys, const int32_t* fixed_sizes,
+                                    const uint8_t* null_safe,
+                                    int enable_fixed_size_string,
+                                    int32_t* packed_bytes_out);
+int orc_join_select_varchar_constructor(int32_t max_size, int enable_fixed_size_string);
+int orc_join_select_method(int key_constructor, int lt_class, uint64_t row_count,
+                           int64_t min_value, int64_t max_value, int mode,
+                           int with_other_conjunct, int enable_range_direct,
+                           int enable_linear_chained, uint64_t l2_size,
+                           uint64_t l3_size);
+
 /* ---- deterministic synthetic data (stateless splitmix64 finalizer) ---- */
 uint64_t orc_gen_u64(uint64_t seed, uint64_t tag, uint64_t i);
 void orc_gen_u32_mod(uint64_t seed, uint64_t tag, uint64_t start, uint64_t n,

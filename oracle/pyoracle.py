@@ -62,6 +62,13 @@ def _decl(lib):
     lib.orc_join_hash_slice.argtypes = [c_vp, c_i32, u]
     lib.orc_calc_bucket_size.restype = u
     lib.orc_calc_bucket_size.argtypes = [u]
+    lib.orc_join_select_key_constructor.restype = c_i32
+    lib.orc_join_select_key_constructor.argtypes = [c_i32, c_vp, c_vp, c_i32, c_vp]
+    lib.orc_join_select_varchar_constructor.restype = c_i32
+    lib.orc_join_select_varchar_constructor.argtypes = [c_i32, c_i32]
+    lib.orc_join_select_method.restype = c_i32
+    lib.orc_join_select_method.argtypes = [c_i32, c_i32, c_u64, c_i64, c_i64, c_i32,
+                                           c_i32, c_i32, c_i32, c_u64, c_u64]
     lib.orc_gen_u64.restype = c_u64
     lib.orc_gen_u64.argtypes = [c_u64, c_u64, c_u64]
     lib.orc_gen_dates.argtypes = [c_i32, c_vp, c_vp]
@@ -607,3 +614,36 @@ def eval_conjuncts_i64(cols, preds):
     ph = np.array([p[3] for p in preds], np.int64)
     return load().orc_eval_conjuncts_i64(ptrs, len(cols), n, _p(pc), _p(po), _p(pl),
                                          _p(ph), len(preds))
+
+
+# --- JoinHashMapSelector restatement (join_hash_table.cpp:164-344) ---
+JM_NAMES = {0: "DIRECT", 1: "RANGE_DIRECT", 2: "RANGE_DIRECT_SET",
+            3: "DENSE_RANGE_DIRECT", 4: "LINEAR_CHAINED",
+            5: "LINEAR_CHAINED_SET", 6: "BUCKET_CHAINED"}
+KEYCON_NAMES = {0: "ONE_KEY", 1: "ONE_KEY_VARCHAR", 2: "FIXED_INT",
+                3: "FIXED_BIGINT", 4: "FIXED_LARGEINT", 5: "SERIALIZED_VARCHAR"}
+
+
+def join_select_key_constructor(fixed_sizes, null_safe=None,
+                                enable_fixed_size_string=1):
+    n = len(fixed_sizes)
+    fs = np.asarray(fixed_sizes, np.int32)
+    ns = np.asarray(null_safe if null_safe is not None else [0] * n, np.uint8)
+    pb = c_i32()
+    kc = load().orc_join_select_key_constructor(n, _p(fs), _p(ns),
+                                                enable_fixed_size_string,
+                                                ctypes.byref(pb))
+    return kc, pb.value
+
+
+def join_select_varchar_constructor(max_size, enable_fixed_size_string=1):
+    return load().orc_join_select_varchar_constructor(max_size, enable_fixed_size_string)
+
+
+def join_select_method(key_constructor, lt_class, row_count, min_value, max_value,
+                       mode=0, with_other_conjunct=0, enable_range_direct=1,
+                       enable_linear_chained=1, l2_size=0, l3_size=0):
+    return load().orc_join_select_method(key_constructor, lt_class, row_count,
+                                         min_value, max_value, mode,
+                                         with_other_conjunct, enable_range_direct,
+                                         enable_linear_chained, l2_size, l3_size)

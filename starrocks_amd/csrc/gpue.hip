@@ -1218,6 +1218,109 @@ static uint32_t calc_bucket_size(uint32_t size) {
     return (uint32_t)(norm + 1);
 }
 
+// ---------------------------------------------------------------------------
+// JoinHashMapSelector restatement (reference join_hash_table.cpp:164-344) —
+// the automatic key-constructor + map-method decision the reference makes at
+// JoinHashTable::build time. Pure host logic; enum values in gpue.h. The
+// decision is the REFERENCE's (CPU cache-size thresholds as parameters, the
+// session flags defaulting true per SessionVariable.java:2060-2067);
+// gpue_join_build_auto_i32 below maps the decision onto the GPU table kinds.
+// ---------------------------------------------------------------------------
+extern "C" {
+int gpue_join_select_key_constructor(int num_keys, const int32_t* fixed_sizes,
+                                     const uint8_t* null_safe,
+                                     int enable_fixed_size_string,
+                                     int32_t* packed_bytes_out);
+int gpue_join_select_varchar_constructor(int32_t max_size, int enable_fixed_size_string);
+int gpue_join_select_method(int key_constructor, int lt_class, uint64_t row_count,
+                            int64_t min_value, int64_t max_value, int mode,
+                            int with_other_conjunct, int enable_range_direct,
+                            int enable_linear_chained, uint64_t l2_size,
+                            uint64_t l3_size);
+}
+
+// _determine_key_constructor (join_hash_table.cpp:164-229)
+int gpue_join_select_key_constructor(int num_keys, const int32_t* fixed_sizes,
+                                     const uint8_t* null_safe,
+                                     int enable_fixed_size_string,
+                                     int32_t* packed_bytes_out) {
+    (void)enable_fixed_size_string;
+    *packed_bytes_out = 0;
+    if (num_keys == 1 && !null_safe[0]) { // :175 single non-null-safe key
+        int32_t sz = fixed_sizes[0];
+        if (sz > 0) {
+            *packed_bytes_out = sz;
+            return GPUE_KEYCON_ONE_KEY;
+        }
+        return GPUE_KEYCON_ONE_KEY_VARCHAR;
+    }
+    // :199-229 multi-key / null-safe: pack fixed widths (+1 B per null-safe)
+    int64_t total = 0;
+    for (int i = 0; i < num_keys; i++) {
+        int32_t cur = fixed_sizes[i];
+        if (cur <= 0) return GPUE_KEYCON_SERIALIZED_VARCHAR;
+        cur += null_safe[i] ? 1 : 0;
+        total += cur;
+    }
+    if (total > 16) { // > 16 B never fixed-packs; packed width unused
+        return GPUE_KEYCON_SERIALIZED_VARCHAR;
+    }
+    *packed_bytes_out = (int32_t)total;
+    if (total <= 4) return GPUE_KEYCON_FIXED_INT;
+    if (total <= 8) return GPUE_KEYCON_FIXED_BIGINT;
+    if (total <= 16) return GPUE_KEYCON_FIXED_LARGEINT;
+    return GPUE_KEYCON_SERIALIZED_VARCHAR;
+}
+
+// single-varchar fixed-size-string refinement (:178-194 via
+// _get_binary_column_max_size :121-162)
+int gpue_join_select_varchar_constructor(int32_t max_size, int enable_fixed_size_string) {
+    if (!enable_fixed_size_string || max_size <= 0) return GPUE_KEYCON_ONE_KEY_VARCHAR;
+    if (max_size <= 4) return GPUE_KEYCON_FIXED_INT;
+    if (max_size <= 8) return GPUE_KEYCON_FIXED_BIGINT;
+    if (max_size <= 16) return GPUE_KEYCON_FIXED_LARGEINT;
+    return GPUE_KEYCON_ONE_KEY_VARCHAR;
+}
+
+// _determine_hash_map_method (:231-256) + _try_use_range_direct_mapping
+// (:270-321) + _try_use_linear_chained (:323-344)
+int gpue_join_select_method(int key_constructor, int lt_class, uint64_t row_count,
+                            int64_t min_value, int64_t max_value, int mode,
+                            int with_other_conjunct, int enable_range_direct,
+                            int enable_linear_chained, uint64_t l2_size,
+                            uint64_t l3_size) {
+    if (lt_class == GPUE_LT_TINY) return GPUE_JM_DIRECT; // :239
+    const bool semi_or_anti_no_conj =
+        (mode == GPUE_JOIN_LEFT_SEMI || mode == GPUE_JOIN_LEFT_ANTI) && !with_other_conjunct;
+    uint64_t rc1 = row_count + 1;
+    const uint64_t bucket_size =
+        calc_bucket_size(rc1 > 0xFFFFFFFFull ? 0xFFFFFFFFu : (uint32_t)rc1);
+    if (key_constructor == GPUE_KEYCON_ONE_KEY &&
+        (lt_class == GPUE_LT_INT || lt_class == GPUE_LT_BIGINT)) { // :242
+        if (enable_range_direct && row_count > 0 &&
+            !(min_value == INT64_MIN && max_value == INT64_MAX)) { // :283 overflow guard
+            uint64_t interval = (uint64_t)max_value - (uint64_t)min_value + 1;
+            if (interval < 0xFFFFFFFFull) { // :288
+                if (semi_or_anti_no_conj) { // :301 one bit vs 8 B first+next
+                    uint64_t memory = (interval + 7) / 8;
+                    if (memory <= bucket_size * 64 || memory <= l3_size / 2)
+                        return GPUE_JM_RANGE_DIRECT_SET;
+                } else {
+                    if (interval <= bucket_size || interval <= l2_size) // :307
+                        return GPUE_JM_RANGE_DIRECT;
+                    if (interval / 4 + row_count * 4 <=
+                        (bucket_size + bucket_size / 10) * 4) // :310-317
+                        return GPUE_JM_DENSE_RANGE_DIRECT;
+                }
+            }
+        }
+    }
+    // :323-344 — 24-bit fp-packed bucket index cap (join_hash_map_method.h:135)
+    if (enable_linear_chained && bucket_size <= 0xFFFFFFu)
+        return semi_or_anti_no_conj ? GPUE_JM_LINEAR_CHAINED_SET : GPUE_JM_LINEAR_CHAINED;
+    return GPUE_JM_BUCKET_CHAINED; // fallback :258-263 (asof out of scope)
+}
+
 extern "C" int gpue_join_build_bucket_chained_nulls_u32(gpue_session* s, gpue_dbuf* keys,
                                                         gpue_dbuf* is_nulls,
                                                         uint64_t row_count,
@@ -1434,6 +1537,46 @@ int gpue_join_build_linear_chained_u32(gpue_session* s, gpue_dbuf* keys, uint64_
     HIP_CHECK(hipGetLastError());
     *out = t;
     return GPUE_OK;
+}
+
+// ---------------------------------------------------------------------------
+// Auto build (JoinHashTable::build's selector step, join_hash_table.cpp:
+// 350-380 -> JoinHashMapSelector): min/max the build keys on device, decide
+// with the reference's rules, build the matching GPU table. Decision->layout
+// mapping documented in gpue.h.
+// ---------------------------------------------------------------------------
+extern "C" int gpue_join_build_auto_i32(gpue_session* s, gpue_dbuf* keys,
+                                        uint64_t row_count, int mode,
+                                        int with_other_conjunct, uint64_t l2_size,
+                                        uint64_t l3_size, gpue_join_table** out,
+                                        int* method_out);
+int gpue_join_build_auto_i32(gpue_session* s, gpue_dbuf* keys, uint64_t row_count,
+                             int mode, int with_other_conjunct, uint64_t l2_size,
+                             uint64_t l3_size, gpue_join_table** out, int* method_out) {
+    ARG_CHECK(s && keys && out && row_count > 0);
+    ARG_CHECK(keys->bytes >= (row_count + 1) * 4);
+    if (l2_size == 0) l2_size = 4ull << 20;   // MI355X XCD L2
+    if (l3_size == 0) l3_size = 256ull << 20; // Infinity Cache
+    int64_t mn = 0, mx = 0;
+    int rc = join_table_minmax(s, (const int32_t*)keys->ptr + 1, row_count, &mn, &mx);
+    if (rc != GPUE_OK) return rc;
+    int method = gpue_join_select_method(GPUE_KEYCON_ONE_KEY, GPUE_LT_INT, row_count,
+                                         mn, mx, mode, with_other_conjunct,
+                                         /*enable_range_direct=*/1,
+                                         /*enable_linear_chained=*/1, l2_size, l3_size);
+    if (method_out) *method_out = method;
+    switch (method) {
+        case GPUE_JM_DIRECT:
+        case GPUE_JM_RANGE_DIRECT:
+        case GPUE_JM_RANGE_DIRECT_SET:
+        case GPUE_JM_DENSE_RANGE_DIRECT:
+            return gpue_join_build_range_direct_i32(s, keys, row_count, out);
+        case GPUE_JM_LINEAR_CHAINED:
+        case GPUE_JM_LINEAR_CHAINED_SET:
+            return gpue_join_build_linear_chained_u32(s, keys, row_count, out);
+        default:
+            return gpue_join_build_bucket_chained_u32(s, keys, row_count, out);
+    }
 }
 
 // linear-probe lookup to the chain head, then chain walk (lookup_init,

@@ -63,6 +63,14 @@ def _declare(lib):
         "gpue_dbuf_h2d": (c_i32, [c_vp, c_vp, c_u64, c_u64]),
         "gpue_dbuf_d2h": (c_i32, [c_vp, c_vp, c_u64, c_u64]),
         "gpue_dbuf_memset": (c_i32, [c_vp, c_i32, c_u64]),
+        "gpue_join_select_key_constructor": (c_i32, [c_i32, c_vp, c_vp, c_i32,
+                                                     ctypes.POINTER(c_i32)]),
+        "gpue_join_select_varchar_constructor": (c_i32, [c_i32, c_i32]),
+        "gpue_join_select_method": (c_i32, [c_i32, c_i32, c_u64, c_i64, c_i64, c_i32,
+                                            c_i32, c_i32, c_i32, c_u64, c_u64]),
+        "gpue_join_build_auto_i32": (c_i32, [c_vp, c_vp, c_u64, c_i32, c_i32, c_u64,
+                                             c_u64, ctypes.POINTER(c_vp),
+                                             ctypes.POINTER(c_i32)]),
         "gpue_gen_u32_mod": (c_i32, [c_vp, c_vp, c_u64, c_u64, c_u64, c_u64, c_u32, c_u32]),
         "gpue_gen_i64": (c_i32, [c_vp, c_vp, c_u64, c_u64, c_u64, c_u64]),
         "gpue_gen_lineorder_q1": (c_i32, [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp, c_vp]),
@@ -311,6 +319,53 @@ class Engine:
             po.ctypes.data_as(c_vp), pl.ctypes.data_as(c_vp), ph.ctypes.data_as(c_vp),
             len(preds), ctypes.byref(out)))
         return out.value
+
+    # --- JoinHashMapSelector restatement (join_hash_table.cpp:164-344) ---
+    JM_NAMES = {0: "DIRECT", 1: "RANGE_DIRECT", 2: "RANGE_DIRECT_SET",
+                3: "DENSE_RANGE_DIRECT", 4: "LINEAR_CHAINED",
+                5: "LINEAR_CHAINED_SET", 6: "BUCKET_CHAINED"}
+    KEYCON_NAMES = {0: "ONE_KEY", 1: "ONE_KEY_VARCHAR", 2: "FIXED_INT",
+                    3: "FIXED_BIGINT", 4: "FIXED_LARGEINT", 5: "SERIALIZED_VARCHAR"}
+
+    def join_select_key_constructor(self, fixed_sizes, null_safe=None,
+                                    enable_fixed_size_string=1):
+        """_determine_key_constructor (:164-229). fixed_sizes: bytes per key
+        column (varchar: its max_size when fixable, else 0). Returns
+        (constructor, packed_bytes)."""
+        n = len(fixed_sizes)
+        fs = (c_i32 * n)(*fixed_sizes)
+        ns = (ctypes.c_uint8 * n)(*(null_safe or [0] * n))
+        pb = c_i32()
+        kc = self._lib.gpue_join_select_key_constructor(
+            n, fs, ns, enable_fixed_size_string, ctypes.byref(pb))
+        return kc, pb.value
+
+    def join_select_varchar_constructor(self, max_size, enable_fixed_size_string=1):
+        return self._lib.gpue_join_select_varchar_constructor(
+            max_size, enable_fixed_size_string)
+
+    def join_select_method(self, key_constructor, lt_class, row_count, min_value,
+                           max_value, mode=0, with_other_conjunct=0,
+                           enable_range_direct=1, enable_linear_chained=1,
+                           l2_size=0, l3_size=0):
+        """_determine_hash_map_method (:231-344). Pure decision — returns a
+        GPUE_JM_* value (JM_NAMES)."""
+        return self._lib.gpue_join_select_method(
+            key_constructor, lt_class, row_count, min_value, max_value, mode,
+            with_other_conjunct, enable_range_direct, enable_linear_chained,
+            l2_size, l3_size)
+
+    def join_build_auto(self, keys: DBuf, row_count, mode=0, with_other_conjunct=0,
+                        l2_size=0, l3_size=0):
+        """Selector-driven build (JoinHashTable::build :350-380): min/max on
+        device -> reference decision -> matching GPU table. Returns
+        (JoinTable, chosen GPUE_JM_* method)."""
+        h = c_vp()
+        m = c_i32()
+        _ck(self._lib, self._lib.gpue_join_build_auto_i32(
+            self._h, keys._h, row_count, mode, with_other_conjunct, l2_size, l3_size,
+            ctypes.byref(h), ctypes.byref(m)))
+        return JoinTable(self, h), m.value
 
     def join_build_payload(self, keys: DBuf, payloads: DBuf, n_rows) -> JoinTable:
         h = c_vp()
