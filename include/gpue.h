@@ -402,6 +402,16 @@ int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs, gpue_join_t
  * shuffler.h:71-86): per-row FNV hash -> ReduceOp channel -> counting-sort
  * row layout, bit-identical to the reference's (stable: each channel's rows
  * ascend by source row). start_points has num_channels+1 entries. */
+/* Version-1 exchange hash (xxh3) partition — the
+ * `_exchange_hash_function_version == 1` branch of
+ * exchange_sink_operator.cpp:604-610 (XXH3_64bits_withSeed per key value,
+ * seed XXH3_SEED_32, truncated u32; restated from the published XXH3 spec
+ * and pinned to python-xxhash vectors in tests/golden/xxh3_kats.json).
+ * FNV (gpue_partition_i32) stays the reference's backward-compatible
+ * default. */
+int gpue_partition_xxh3_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n,
+                            uint32_t num_channels, uint64_t* start_points_out,
+                            gpue_dbuf* row_indexes_out);
 int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t num_channels,
                        uint64_t* start_points_out, gpue_dbuf* row_indexes_out);
 /* Multi-column partition key: the sink seeds FNV_SEED then CHAINS fnv_hash

@@ -1018,6 +1018,73 @@ void orc_partition_channel_u64(const uint64_t* keys, uint64_t n, uint32_t num_ch
     }
 }
 
+/* ====================================================================== */
+/* XXH3-64 exchange hash, version 1 (exchange_sink_operator.cpp:604-610:    */
+/* `_exchange_hash_function_version == 1` uses Column::xxh3_hash =          */
+/* HashUtil::xx_hash3_64 = XXH3_64bits_withSeed, chained per key column     */
+/* from HashUtil::XXH3_SEED_32 = 0x9E3779B1, truncated to u32 per column    */
+/* hop — column_hash.cpp:65-68,473-476, hash_util.hpp:116,126).             */
+/* The thirdparty xxHash library is absent offline, so the 4-to-8-byte      */
+/* input path (the only one fixed-width key columns hit) is restated from   */
+/* the PUBLISHED XXH3 spec (xxHash v0.8.x, BSD-2): XXH3_len_4to8_64b with   */
+/* the default kSecret. Pinned against vectors generated by the published   */
+/* python `xxhash` module (tests/golden/xxh3_kats.json + generator).        */
+/* ====================================================================== */
+
+/* default kSecret bytes 8..23 as two LE u64 words (XXH3 spec) */
+#define XXH3_SECRET8 0x1cad21f72c81017cull
+#define XXH3_SECRET16 0xdb979083e96dd4deull
+#define XXH3_M2 0x9FB21C651E98DF25ull
+
+static inline uint64_t xxh3_rrmxmx(uint64_t h, uint64_t len) {
+    h ^= ((h << 49) | (h >> 15)) ^ ((h << 24) | (h >> 40));
+    h *= XXH3_M2;
+    h ^= (h >> 35) + len;
+    h *= XXH3_M2;
+    return h ^ (h >> 28);
+}
+
+uint64_t orc_xxh3_64_4to8(const void* data, int32_t len, uint64_t seed) {
+    uint32_t s32 = (uint32_t)seed;
+    uint32_t swapped = ((s32 & 0xFFu) << 24) | ((s32 & 0xFF00u) << 8) |
+                       ((s32 >> 8) & 0xFF00u) | (s32 >> 24);
+    seed ^= (uint64_t)swapped << 32;
+    uint32_t in1, in2;
+    memcpy(&in1, data, 4);
+    memcpy(&in2, (const uint8_t*)data + len - 4, 4);
+    uint64_t bitflip = (XXH3_SECRET8 ^ XXH3_SECRET16) - seed;
+    uint64_t input64 = (uint64_t)in2 + ((uint64_t)in1 << 32);
+    return xxh3_rrmxmx(input64 ^ bitflip, (uint64_t)len);
+}
+
+#define XXH3_SEED_32 0x9E3779B1u /* hash_util.hpp:126 */
+
+/* version-1 exchange hash over one i32 key column: hashes[] carries the
+ * chained per-column value (init to XXH3_SEED_32 before the first column —
+ * exchange_sink_operator.cpp:606-609) */
+void orc_xxh3_hash_i32(const int32_t* col, uint64_t n, uint32_t* hashes) {
+#pragma omp parallel for schedule(static)
+    for (uint64_t i = 0; i < n; i++)
+        hashes[i] = (uint32_t)orc_xxh3_64_4to8(&col[i], 4, hashes[i]);
+}
+
+void orc_xxh3_hash_i64(const int64_t* col, uint64_t n, uint32_t* hashes) {
+#pragma omp parallel for schedule(static)
+    for (uint64_t i = 0; i < n; i++)
+        hashes[i] = (uint32_t)orc_xxh3_64_4to8(&col[i], 8, hashes[i]);
+}
+
+/* single-column xxh3 partition: seed init + hash + ReduceOp channel
+ * (shuffler.h:71-86 — the Shuffler is hash-version agnostic) */
+void orc_partition_channel_xxh3_u32(const uint32_t* keys, uint64_t n,
+                                    uint32_t num_channels, uint32_t* channel_ids) {
+#pragma omp parallel for schedule(static)
+    for (uint64_t i = 0; i < n; i++) {
+        uint32_t h = (uint32_t)orc_xxh3_64_4to8(&keys[i], 4, XXH3_SEED_32);
+        channel_ids[i] = (uint32_t)(((uint64_t)h * num_channels) >> 32);
+    }
+}
+
 /* counting-sort row layout (exchange_sink_operator.cpp:629-660): forward
  * count, prefix-sum, then reverse iteration emit so each channel's rows stay
  * in ascending source order. */

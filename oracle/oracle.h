@@ -297,3 +297,11 @@ void orc_free(void* p);
 }
 #endif
 #endif
+
+/* XXH3-64 exchange hash version 1 (exchange_sink_operator.cpp:604-610;
+ * restated from the published XXH3 spec, pinned to python-xxhash vectors) */
+uint64_t orc_xxh3_64_4to8(const void* data, int32_t len, uint64_t seed);
+void orc_xxh3_hash_i32(const int32_t* col, uint64_t n, uint32_t* hashes);
+void orc_xxh3_hash_i64(const int64_t* col, uint64_t n, uint32_t* hashes);
+void orc_partition_channel_xxh3_u32(const uint32_t* keys, uint64_t n,
+                                    uint32_t num_channels, uint32_t* channel_ids);

@@ -69,6 +69,11 @@ def _decl(lib):
     lib.orc_join_select_method.restype = c_i32
     lib.orc_join_select_method.argtypes = [c_i32, c_i32, c_u64, c_i64, c_i64, c_i32,
                                            c_i32, c_i32, c_i32, c_u64, c_u64]
+    lib.orc_xxh3_64_4to8.restype = c_u64
+    lib.orc_xxh3_64_4to8.argtypes = [c_vp, c_i32, c_u64]
+    lib.orc_xxh3_hash_i32.argtypes = [c_vp, c_u64, c_vp]
+    lib.orc_xxh3_hash_i64.argtypes = [c_vp, c_u64, c_vp]
+    lib.orc_partition_channel_xxh3_u32.argtypes = [c_vp, c_u64, u, c_vp]
     lib.orc_gen_u64.restype = c_u64
     lib.orc_gen_u64.argtypes = [c_u64, c_u64, c_u64]
     lib.orc_gen_dates.argtypes = [c_i32, c_vp, c_vp]
@@ -647,3 +652,30 @@ def join_select_method(key_constructor, lt_class, row_count, min_value, max_valu
                                          min_value, max_value, mode,
                                          with_other_conjunct, enable_range_direct,
                                          enable_linear_chained, l2_size, l3_size)
+
+
+XXH3_SEED_32 = 0x9E3779B1
+
+
+def xxh3_64(data: bytes, seed: int = 0) -> int:
+    """XXH3-64 for 4-8 byte inputs (the exchange hash's per-value call)."""
+    return load().orc_xxh3_64_4to8(data, len(data), seed)
+
+
+def xxh3_exchange_hash_i32(cols) -> np.ndarray:
+    """Version-1 exchange hash over i32 key columns: seed XXH3_SEED_32,
+    chained per column, u32 truncation per hop
+    (exchange_sink_operator.cpp:604-610)."""
+    n = len(cols[0])
+    h = np.full(n, XXH3_SEED_32, np.uint32)
+    for col in cols:
+        c = np.ascontiguousarray(col, np.int32)
+        load().orc_xxh3_hash_i32(_p(c), n, _p(h))
+    return h
+
+
+def partition_channels_xxh3(keys: np.ndarray, num_channels: int) -> np.ndarray:
+    ch = np.empty(len(keys), np.uint32)
+    load().orc_partition_channel_xxh3_u32(_p(np.ascontiguousarray(keys, np.uint32)),
+                                          len(keys), num_channels, _p(ch))
+    return ch

@@ -3507,6 +3507,45 @@ int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs, gpue_join_t
 // emission order — the per-channel row SET is what routing correctness needs
 // (SURVEY.md §8c).
 // ---------------------------------------------------------------------------
+// XXH3-64 4-to-8-byte path (exchange hash version 1,
+// exchange_sink_operator.cpp:604-610; HashUtil::xx_hash3_64 =
+// XXH3_64bits_withSeed). Restated from the published XXH3 spec (xxHash
+// v0.8.x): XXH3_len_4to8_64b with the default kSecret words 8..23. Pinned
+// to python-xxhash vectors via the oracle (tests/golden/xxh3_kats.json).
+#define XXH3_SECRET8 0x1cad21f72c81017cull
+#define XXH3_SECRET16 0xdb979083e96dd4deull
+#define XXH3_M2 0x9FB21C651E98DF25ull
+#define XXH3_SEED_32 0x9E3779B1u // HashUtil::XXH3_SEED_32 (hash_util.hpp:126)
+
+__device__ static inline uint64_t xxh3_rrmxmx(uint64_t h, uint64_t len) {
+    h ^= ((h << 49) | (h >> 15)) ^ ((h << 24) | (h >> 40));
+    h *= XXH3_M2;
+    h ^= (h >> 35) + len;
+    h *= XXH3_M2;
+    return h ^ (h >> 28);
+}
+
+// 4-byte input (one i32 key), seed = running per-column hash
+__device__ static inline uint64_t xxh3_64_u32(uint32_t v, uint64_t seed) {
+    uint32_t s32 = (uint32_t)seed;
+    uint32_t sw = __builtin_bswap32(s32);
+    seed ^= (uint64_t)sw << 32;
+    uint64_t input64 = (uint64_t)v + ((uint64_t)v << 32); // in1 == in2 at len 4
+    uint64_t bitflip = (XXH3_SECRET8 ^ XXH3_SECRET16) - seed;
+    return xxh3_rrmxmx(input64 ^ bitflip, 4);
+}
+
+// 8-byte input (one i64 key)
+__device__ static inline uint64_t xxh3_64_u64(uint64_t v, uint64_t seed) {
+    uint32_t s32 = (uint32_t)seed;
+    uint32_t sw = __builtin_bswap32(s32);
+    seed ^= (uint64_t)sw << 32;
+    uint32_t in1 = (uint32_t)v, in2 = (uint32_t)(v >> 32);
+    uint64_t input64 = (uint64_t)in2 + ((uint64_t)in1 << 32);
+    uint64_t bitflip = (XXH3_SECRET8 ^ XXH3_SECRET16) - seed;
+    return xxh3_rrmxmx(input64 ^ bitflip, 8);
+}
+
 __device__ static inline uint32_t fnv_u32(uint32_t key, uint32_t seed) {
     uint32_t h = seed;
     #pragma unroll
@@ -3525,6 +3564,7 @@ __device__ static inline uint32_t fnv_u64(uint64_t key, uint32_t seed) {
 
 static constexpr uint32_t MAX_CH = 64;
 
+template <int HV> // 0 = FNV (default/back-compat), 1 = xxh3 (version 1)
 __global__ void k_partition_hist(const uint32_t* __restrict__ keys, uint64_t n, uint64_t tile,
                                  uint32_t nch, uint32_t* __restrict__ block_hist) {
     __shared__ uint32_t h[MAX_CH];
@@ -3532,7 +3572,8 @@ __global__ void k_partition_hist(const uint32_t* __restrict__ keys, uint64_t n, 
     __syncthreads();
     uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
     for (uint64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
-        uint32_t hash = fnv_u32(keys[i], 0x811C9DC5u);
+        uint32_t hash = HV ? (uint32_t)xxh3_64_u32(keys[i], XXH3_SEED_32)
+                           : fnv_u32(keys[i], 0x811C9DC5u);
         uint32_t ch = (uint32_t)(((uint64_t)hash * nch) >> 32);
         atomicAdd(&h[ch], 1u);
     }
@@ -3541,6 +3582,7 @@ __global__ void k_partition_hist(const uint32_t* __restrict__ keys, uint64_t n, 
         block_hist[(uint64_t)blockIdx.x * nch + c] = h[c];
 }
 
+template <int HV>
 __global__ void k_partition_emit(const uint32_t* __restrict__ keys, uint64_t n, uint64_t tile,
                                  uint32_t nch, const uint64_t* __restrict__ block_offsets,
                                  uint32_t* __restrict__ row_indexes) {
@@ -3550,7 +3592,8 @@ __global__ void k_partition_emit(const uint32_t* __restrict__ keys, uint64_t n, 
     __syncthreads();
     uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
     for (uint64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
-        uint32_t hash = fnv_u32(keys[i], 0x811C9DC5u);
+        uint32_t hash = HV ? (uint32_t)xxh3_64_u32(keys[i], XXH3_SEED_32)
+                           : fnv_u32(keys[i], 0x811C9DC5u);
         uint32_t ch = (uint32_t)(((uint64_t)hash * nch) >> 32);
         uint64_t pos = atomicAdd((unsigned long long*)&cursor[ch], 1ull);
         row_indexes[pos] = (uint32_t)i;
@@ -3672,8 +3715,9 @@ int gpue_partition_i64(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t nc
     return GPUE_OK;
 }
 
-int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t nch,
-                       uint64_t* start_points_out, gpue_dbuf* row_indexes_out) {
+static int partition_i32_impl(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t nch,
+                              uint64_t* start_points_out, gpue_dbuf* row_indexes_out,
+                              int hash_version) {
     ARG_CHECK(s && keys && start_points_out && row_indexes_out);
     ARG_CHECK(nch >= 1 && nch <= MAX_CH);
     ARG_CHECK(keys->bytes >= n * 4 && row_indexes_out->bytes >= n * 4);
@@ -3682,8 +3726,12 @@ int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t nc
     uint32_t* d_hist = nullptr;
     uint64_t* d_off = nullptr;
     HIP_CHECK(hipMalloc(&d_hist, (uint64_t)nb * nch * sizeof(uint32_t)));
-    hipLaunchKernelGGL(k_partition_hist, dim3(nb), dim3(BLOCK), 0, s->stream,
-                       (const uint32_t*)keys->ptr, n, tile, nch, d_hist);
+    if (hash_version == 1)
+        hipLaunchKernelGGL(k_partition_hist<1>, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const uint32_t*)keys->ptr, n, tile, nch, d_hist);
+    else
+        hipLaunchKernelGGL(k_partition_hist<0>, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const uint32_t*)keys->ptr, n, tile, nch, d_hist);
     // host scan in channel-major order -> per-(block,channel) start offsets
     uint32_t* h_hist = (uint32_t*)malloc((uint64_t)nb * nch * sizeof(uint32_t));
     uint64_t* h_off = (uint64_t*)malloc((uint64_t)nb * nch * sizeof(uint64_t));
@@ -3702,15 +3750,35 @@ int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t nc
     HIP_CHECK(hipMalloc(&d_off, (uint64_t)nb * nch * sizeof(uint64_t)));
     HIP_CHECK(hipMemcpyAsync(d_off, h_off, (uint64_t)nb * nch * sizeof(uint64_t),
                              hipMemcpyHostToDevice, s->stream));
-    hipLaunchKernelGGL(k_partition_emit, dim3(nb), dim3(BLOCK), 0, s->stream,
-                       (const uint32_t*)keys->ptr, n, tile, nch, d_off,
-                       (uint32_t*)row_indexes_out->ptr);
+    if (hash_version == 1)
+        hipLaunchKernelGGL(k_partition_emit<1>, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const uint32_t*)keys->ptr, n, tile, nch, d_off,
+                           (uint32_t*)row_indexes_out->ptr);
+    else
+        hipLaunchKernelGGL(k_partition_emit<0>, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const uint32_t*)keys->ptr, n, tile, nch, d_off,
+                           (uint32_t*)row_indexes_out->ptr);
     HIP_CHECK(hipStreamSynchronize(s->stream));
     (void)hipFree(d_hist);
     (void)hipFree(d_off);
     free(h_hist);
     free(h_off);
     return GPUE_OK;
+}
+
+int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t nch,
+                       uint64_t* start_points_out, gpue_dbuf* row_indexes_out) {
+    return partition_i32_impl(s, keys, n, nch, start_points_out, row_indexes_out, 0);
+}
+
+// version-1 exchange hash (xxh3) partition — exchange_sink_operator.cpp:
+// 604-610's `_exchange_hash_function_version == 1` branch
+extern "C" int gpue_partition_xxh3_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n,
+                                       uint32_t nch, uint64_t* start_points_out,
+                                       gpue_dbuf* row_indexes_out);
+int gpue_partition_xxh3_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t nch,
+                            uint64_t* start_points_out, gpue_dbuf* row_indexes_out) {
+    return partition_i32_impl(s, keys, n, nch, start_points_out, row_indexes_out, 1);
 }
 
 

@@ -113,6 +113,7 @@ def _declare(lib):
         "gpue_q21_star_agg_async": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_q21_star_agg_pipe": (c_i32, [c_vp] * 8 + [c_u64, c_vp, c_vp, c_i32]),
         "gpue_partition_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
+        "gpue_partition_xxh3_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
         "gpue_gather_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_gather_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_partition_i64": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
@@ -689,6 +690,14 @@ class Engine:
         sp = np.zeros(num_channels + 1, dtype=np.uint64)
         _ck(self._lib, self._lib.gpue_partition_i32(self._h, keys._h, n, num_channels,
                                                     sp.ctypes.data_as(c_vp), row_indexes._h))
+        return sp
+
+    def partition_xxh3(self, keys: DBuf, n, num_channels, row_indexes: DBuf) -> np.ndarray:
+        """Version-1 exchange hash (xxh3) partition
+        (exchange_sink_operator.cpp:604-610; FNV stays the default)."""
+        sp = np.zeros(num_channels + 1, dtype=np.uint64)
+        _ck(self._lib, self._lib.gpue_partition_xxh3_i32(
+            self._h, keys._h, n, num_channels, sp.ctypes.data_as(c_vp), row_indexes._h))
         return sp
 
     def partition_2xi32(self, a: DBuf, b: DBuf, n, num_channels,
