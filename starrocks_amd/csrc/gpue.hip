@@ -138,6 +138,8 @@ struct gpue_join_table {
     int64_t set_min = 0, set_max = -1; // bounding range of keys with payload != 0
     uint16_t* first16 = nullptr; // 16-bit payload copy when all payloads < 65536:
                                  // halves the random-gather footprint (L2 per XCD is 4 MiB)
+    uint32_t* prefilter = nullptr; // 2^19-bit (64 KB) fold of `bitset` for
+                                   // LDS-resident prefiltering (k_q21_star_agg_pf)
     uint64_t bucket_size = 0;
     uint64_t row_count = 0;
     // method discriminator — the GPU analog of JoinHashMapSelector's choice
@@ -912,6 +914,21 @@ __global__ void k_set_bounds(const uint32_t* __restrict__ first, uint64_t interv
     }
 }
 
+// Fold the passing-key bitset into the fixed 2^19-bit LDS prefilter
+// (conservative: a fold-set bit means "some key with this masked index
+// passes"). Built once per table; read-only at probe time.
+__global__ void k_build_prefilter(const uint32_t* __restrict__ bitset, uint64_t set_interval,
+                                  uint32_t* __restrict__ prefilter) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < set_interval;
+         i += stride) {
+        if ((bitset[i >> 5] >> (i & 31)) & 1u) {
+            uint32_t f = (uint32_t)i & ((1u << 19) - 1);
+            atomicOr(&prefilter[f >> 5], 1u << (f & 31));
+        }
+    }
+}
+
 __global__ void k_derive_bitset(const uint32_t* __restrict__ first, uint64_t offset,
                                 uint64_t interval, uint32_t* __restrict__ bitset) {
     // bit j of word w == (first[offset + w*32 + j] != 0)
@@ -1013,6 +1030,10 @@ int gpue_join_build_payload_i32(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* pay
     HIP_CHECK(hipMalloc(&t->bitset, nwords * sizeof(uint32_t)));
     hipLaunchKernelGGL(k_derive_bitset, dim3(grid_for(nwords)), dim3(BLOCK), 0, s->stream,
                        t->first, set_off, set_interval, t->bitset);
+    HIP_CHECK(hipMalloc(&t->prefilter, (1u << 19) / 8));
+    HIP_CHECK(hipMemsetAsync(t->prefilter, 0, (1u << 19) / 8, s->stream));
+    hipLaunchKernelGGL(k_build_prefilter, dim3(grid_for(set_interval)), dim3(BLOCK), 0,
+                       s->stream, t->bitset, set_interval, t->prefilter);
     uint32_t* d_ovf = nullptr;
     HIP_CHECK(hipMalloc(&d_ovf, sizeof(uint32_t)));
     HIP_CHECK(hipMemsetAsync(d_ovf, 0, sizeof(uint32_t), s->stream));
@@ -1059,6 +1080,7 @@ void gpue_join_table_destroy(gpue_join_table* t) {
     if (t->next) (void)hipFree(t->next);
     if (t->bitset) (void)hipFree(t->bitset);
     if (t->first16) (void)hipFree(t->first16);
+    if (t->prefilter) (void)hipFree(t->prefilter);
     if (t->build_keys) (void)hipFree(t->build_keys);
     if (t->key_bytes) (void)hipFree(t->key_bytes);
     if (t->key_offsets) (void)hipFree(t->key_offsets);
@@ -3149,6 +3171,101 @@ __global__ void k_q21_star_agg_glob(const int32_t* __restrict__ pk,
 }
 
 // ---------------------------------------------------------------------------
+// LDS-prefilter q21 (round-2 item 3 — the stream/gather overlap lever).
+// Measured state of the fused kernel: the 4-column stream leg
+// (1.57 ms, HBM-bound) and the per-row part-bitset L2 gather leg (2.32 ms)
+// are fully ADDITIVE, and TCC counters put the gathers at an 87% L2 hit rate
+// — the leg is bound by the TA's DIVERGENT-ADDRESS request rate (~1
+// lane/cycle), not by misses. The fix is to answer the per-row part filter
+// from LDS, whose random-access rate is bank-parallel (~20x the TA's rate
+// for divergent words):
+//   - the build side folds the part bitset (175 KB over the passing-key
+//     range) into a 2^19-bit = 64 KB PREFILTER by index masking; a clear bit
+//     is definitely-not-in-category, a set bit is "maybe" (fold factor ~2.7
+//     keys/bit at SF100 => ~10% maybes at a 4% true rate). This is the same
+//     conservative-filter idea as the reference's SimdBlockFilter runtime
+//     filter (runtime_filter.h:79), applied to the scan's dim filter.
+//   - each block stages the prefilter in LDS (64 KB) next to the 56 KB group
+//     array (120 KB/block, 1 block/CU, 16 waves — streams saturate HBM from
+//     4 waves/CU, cf. the q1 kernel) and tests all four quad keys with LDS
+//     reads. Only "maybe" rows (~10%) gather the u16 brand payload from L2,
+//     whose zero value encodes the exact category filter (payload tables:
+//     first[key-min] = brand+1 | 0) — so the exact 175 KB
+//     bitset is never probed at all and the TA leg shrinks ~10x.
+// Parity: identical emissions — prefilter false => bitset false => brand 0.
+// ---------------------------------------------------------------------------
+static constexpr int PF_LOG2 = 19;                      // 2^19 bits = 64 KB
+static constexpr uint32_t PF_WORDS = (1u << PF_LOG2) / 32;
+static constexpr uint32_t PF_MASK = (1u << PF_LOG2) - 1;
+
+__global__ __launch_bounds__(BLOCK_Q21) void
+k_q21_star_agg_pf(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
+                  const int32_t* __restrict__ od, const int32_t* __restrict__ rv,
+                  uint64_t n, const uint32_t* __restrict__ prefilter, int64_t psmin,
+                  uint64_t psint, const uint16_t* __restrict__ pfirst,
+                  const uint32_t* __restrict__ sbits, int64_t ssmin, uint64_t ssint,
+                  const uint16_t* __restrict__ dfirst, int64_t dmin,
+                  unsigned long long* __restrict__ group_sums) {
+    __shared__ uint32_t pf[PF_WORDS];        // 64 KB folded part filter
+    __shared__ unsigned long long g[NG_Q21]; // 56 KB group sums
+    for (uint32_t w = threadIdx.x; w < PF_WORDS; w += blockDim.x) pf[w] = prefilter[w];
+    for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x) g[j] = 0;
+    __syncthreads();
+    const uint64_t n4 = n / 4;
+    const int4* __restrict__ pk4 = (const int4*)pk;
+    const int4* __restrict__ sk4 = (const int4*)sk;
+    const int4* __restrict__ od4 = (const int4*)od;
+    const int4* __restrict__ rv4 = (const int4*)rv;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    auto quad = [&](int4 p4, int4 s4, int4 o4, int4 r4) {
+        uint32_t maybe[4];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint32_t idx = (uint32_t)((&p4.x)[j] - psmin);
+            bool in = idx < psint;
+            uint32_t fidx = (in ? idx : 0u) & PF_MASK;
+            maybe[j] = in & (pf[fidx >> 5] >> (fidx & 31)) & 1u;
+        }
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            if (!maybe[j]) continue;
+            uint32_t brand1 = pfirst[(&p4.x)[j] - 1]; // 0 = fails the category filter
+            if (!brand1) continue;
+            uint32_t sidx = (uint32_t)((&s4.x)[j] - ssmin);
+            if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
+            uint32_t year1 = dfirst[(&o4.x)[j] - dmin];
+            atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)],
+                      (unsigned long long)(int64_t)(&r4.x)[j]);
+        }
+    };
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i + stride < n4; i += 2 * stride) {
+        int4 pa = pk4[i], sa = sk4[i], oa = od4[i], ra = rv4[i];
+        uint64_t i2 = i + stride;
+        int4 pb_ = pk4[i2], sb = sk4[i2], ob = od4[i2], rb = rv4[i2];
+        quad(pa, sa, oa, ra);
+        quad(pb_, sb, ob, rb);
+    }
+    for (; i < n4; i += stride) quad(pk4[i], sk4[i], od4[i], rv4[i]);
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (uint64_t r = n4 * 4 + tid; r < n; r += stride) {
+        uint32_t idx = (uint32_t)(pk[r] - psmin);
+        if (idx >= psint) continue;
+        uint32_t fidx = idx & PF_MASK;
+        if (!((pf[fidx >> 5] >> (fidx & 31)) & 1u)) continue;
+        uint32_t brand1 = pfirst[pk[r] - 1];
+        if (!brand1) continue;
+        uint32_t sidx = (uint32_t)(sk[r] - ssmin);
+        if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
+        uint32_t year1 = dfirst[od[r] - dmin];
+        atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)], (unsigned long long)(int64_t)rv[r]);
+    }
+    __syncthreads();
+    for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x)
+        if (g[j] != 0) atomicAdd(&group_sums[j], g[j]);
+}
+
+// ---------------------------------------------------------------------------
 // Two-stream pipelined q21 (GPUE_Q21_PIPE=1): the fused kernel's streaming
 // leg (1.57 ms) and its part-probe gather leg (2.32 ms) measured fully
 // ADDITIVE (profiles/q21_decomp.log) — so split them into an operator pair
@@ -3305,6 +3422,17 @@ static bool q21_glob() {
     return v == 1;
 }
 
+// GPUE_Q21_PF: 1 (default) = LDS-prefilter kernel, 0 = the round-1 fused
+// L2-bitset kernel
+static bool q21_pf() {
+    static int v = -1;
+    if (v < 0) {
+        const char* e = getenv("GPUE_Q21_PF");
+        v = (e && !atoi(e)) ? 0 : 1;
+    }
+    return v == 1;
+}
+
 extern "C" int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts,
                                        gpue_join_table* supps, gpue_join_table* dates,
                                        gpue_dbuf* pk, gpue_dbuf* sk, gpue_dbuf* od,
@@ -3317,6 +3445,18 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
     ARG_CHECK(parts->min_key == 1 && supps->min_key == 1);
     ARG_CHECK(parts->first16 && supps->first16 && dates->first16);
     HIP_CHECK(hipMemsetAsync(group_sums->ptr, 0, NG_Q21 * sizeof(int64_t), s->stream));
+    if (q21_pf() && parts->prefilter) {
+        hipLaunchKernelGGL(k_q21_star_agg_pf, dim3(env_cap("GPUE_GRID_PF", 256)),
+                           dim3(BLOCK_Q21), 0, s->stream, (const int32_t*)pk->ptr,
+                           (const int32_t*)sk->ptr, (const int32_t*)od->ptr,
+                           (const int32_t*)rv->ptr, n, parts->prefilter, parts->set_min,
+                           (uint64_t)(parts->set_max - parts->set_min + 1), parts->first16,
+                           supps->bitset, supps->set_min,
+                           (uint64_t)(supps->set_max - supps->set_min + 1), dates->first16,
+                           dates->min_key, (unsigned long long*)group_sums->ptr);
+        HIP_CHECK(hipGetLastError());
+        return GPUE_OK;
+    }
     if (q21_glob()) {
         hipLaunchKernelGGL(k_q21_star_agg_glob, dim3(grid_stream(n)), dim3(BLOCK), 0,
                            s->stream, (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
