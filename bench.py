@@ -579,8 +579,13 @@ def main():
                 "traffic": traffic}
 
     if rank == 0:
-        cb = cpu_baseline(wl, rows) if world == 1 and wl != "q3" else (
-            cpu_baseline_q3(rows) if world == 1 else None)
+        # GPUE_SKIP_CPU_BASELINE=1: omit the ~10 s oracle leg during kernel
+        # sweeps; the driver's default run always measures it
+        if os.environ.get("GPUE_SKIP_CPU_BASELINE") == "1":
+            cb = None
+        else:
+            cb = cpu_baseline(wl, rows) if world == 1 and wl != "q3" else (
+                cpu_baseline_q3(rows) if world == 1 else None)
         result = {
             "metric": "ssb_join_agg_rows_per_sec",
             "value": round(value, 1),
