@@ -54,7 +54,7 @@ def build_payload_table(eng, keys_i32, payload_u32):
     # RANGE_DIRECT family — whose MI355X fused form is the payload table
     # (first[key-min] holds payload+1). Any other decision
     # means the workload changed and the fused kernel no longer applies.
-    m = eng.join_select_method(0 /*ONE_KEY*/, 1 /*LT_INT*/, len(keys_i32),
+    m = eng.join_select_method(0, 1, len(keys_i32),  # ONE_KEY constructor, LT_INT
                                int(keys_i32.min()), int(keys_i32.max()))
     if eng.JM_NAMES[m] not in ("RANGE_DIRECT", "RANGE_DIRECT_SET",
                                "DENSE_RANGE_DIRECT", "DIRECT"):

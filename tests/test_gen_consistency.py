@@ -76,3 +76,14 @@ def test_mkt_segments_match_oracle():
     for i in range(5):
         assert gen.mkt_literal(i) == orc.mkt_literal(i)
         assert len(gen.mkt_literal(i)) == 16
+
+
+def test_bench_script_compiles():
+    """bench.py is never imported by the CPU suite, so a syntax error would
+    only surface on the GPU box — compile it here."""
+    import ast
+    import os
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for fname in ("bench.py", "__graft_entry__.py"):
+        with open(os.path.join(root, fname)) as f:
+            ast.parse(f.read(), filename=fname)
