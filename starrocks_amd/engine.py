@@ -348,9 +348,11 @@ class Engine:
     def join_select_method(self, key_constructor, lt_class, row_count, min_value,
                            max_value, mode=0, with_other_conjunct=0,
                            enable_range_direct=1, enable_linear_chained=1,
-                           l2_size=0, l3_size=0):
+                           l2_size=4 << 20, l3_size=256 << 20):
         """_determine_hash_map_method (:231-344). Pure decision — returns a
-        GPUE_JM_* value (JM_NAMES)."""
+        GPUE_JM_* value (JM_NAMES). Cache-size defaults are the MI355X
+        residency tiers (4 MiB XCD L2 / 256 MiB Infinity Cache), matching
+        gpue_join_build_auto_i32; the reference reads CpuInfo here."""
         return self._lib.gpue_join_select_method(
             key_constructor, lt_class, row_count, min_value, max_value, mode,
             with_other_conjunct, enable_range_direct, enable_linear_chained,

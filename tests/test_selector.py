@@ -244,3 +244,16 @@ def test_join_build_auto_semi_set(engine):
     kb.free()
     pb.free()
     t.destroy()
+
+
+def test_bench_dims_select_range_direct_on_mi355x_defaults():
+    """The four SSB dim builds in bench.py route through the selector with
+    MI355X cache defaults (4 MiB XCD L2) — all must land in the
+    direct-mapped family or the fused payload-table path would refuse."""
+    L2M, L3M = 4 << 20, 256 << 20
+    dims = [(2556, 19920101, 19981230), (1_400_000, 1, 1_400_000),
+            (200_000, 1, 200_000), (3_000_000, 1, 3_000_000)]
+    for rc, mn, mx in dims:
+        m = orc.join_select_method(ONE_KEY, LT_INT, rc, mn, mx, INNER, 0,
+                                   l2_size=L2M, l3_size=L3M)
+        assert m == RANGE_DIRECT, (rc, orc.JM_NAMES[m])
