@@ -3198,6 +3198,10 @@ static constexpr int PF_LOG2 = 19;                      // 2^19 bits = 64 KB
 static constexpr uint32_t PF_WORDS = (1u << PF_LOG2) / 32;
 static constexpr uint32_t PF_MASK = (1u << PF_LOG2) - 1;
 
+// LDSG: group sums in LDS (56 KB, 1 block/CU) vs global atomics (frees the
+// LDS for a 2nd block -> 32 waves/CU). NT: non-temporal stream loads (leave
+// L2 to the u16 payload the maybe-rows gather).
+template <bool LDSG, bool NT>
 __global__ __launch_bounds__(BLOCK_Q21) void
 k_q21_star_agg_pf(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
                   const int32_t* __restrict__ od, const int32_t* __restrict__ rv,
@@ -3207,9 +3211,10 @@ k_q21_star_agg_pf(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk
                   const uint16_t* __restrict__ dfirst, int64_t dmin,
                   unsigned long long* __restrict__ group_sums) {
     __shared__ uint32_t pf[PF_WORDS];        // 64 KB folded part filter
-    __shared__ unsigned long long g[NG_Q21]; // 56 KB group sums
+    __shared__ unsigned long long g[LDSG ? NG_Q21 : 1]; // 56 KB group sums
     for (uint32_t w = threadIdx.x; w < PF_WORDS; w += blockDim.x) pf[w] = prefilter[w];
-    for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x) g[j] = 0;
+    if (LDSG)
+        for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x) g[j] = 0;
     __syncthreads();
     const uint64_t n4 = n / 4;
     const int4* __restrict__ pk4 = (const int4*)pk;
@@ -3234,15 +3239,28 @@ k_q21_star_agg_pf(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk
             uint32_t sidx = (uint32_t)((&s4.x)[j] - ssmin);
             if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
             uint32_t year1 = dfirst[(&o4.x)[j] - dmin];
-            atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)],
-                      (unsigned long long)(int64_t)(&r4.x)[j]);
+            unsigned long long* dst = LDSG ? &g[(year1 - 1) * 1000 + (brand1 - 1)]
+                                           : &group_sums[(year1 - 1) * 1000 + (brand1 - 1)];
+            atomicAdd(dst, (unsigned long long)(int64_t)(&r4.x)[j]);
         }
+    };
+    auto ld4 = [&](const int4* p, uint64_t i) {
+        if (!NT) return p[i];
+        // the builtin rejects HIP vector types; two u64 NT loads keep the
+        // 16-B dwordx4-equivalent width
+        const uint64_t* q = (const uint64_t*)(p + i);
+        uint64_t lo = __builtin_nontemporal_load(q);
+        uint64_t hi = __builtin_nontemporal_load(q + 1);
+        int4 v;
+        v.x = (int32_t)lo; v.y = (int32_t)(lo >> 32);
+        v.z = (int32_t)hi; v.w = (int32_t)(hi >> 32);
+        return v;
     };
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     for (; i + stride < n4; i += 2 * stride) {
-        int4 pa = pk4[i], sa = sk4[i], oa = od4[i], ra = rv4[i];
+        int4 pa = ld4(pk4, i), sa = ld4(sk4, i), oa = ld4(od4, i), ra = ld4(rv4, i);
         uint64_t i2 = i + stride;
-        int4 pb_ = pk4[i2], sb = sk4[i2], ob = od4[i2], rb = rv4[i2];
+        int4 pb_ = ld4(pk4, i2), sb = ld4(sk4, i2), ob = ld4(od4, i2), rb = ld4(rv4, i2);
         quad(pa, sa, oa, ra);
         quad(pb_, sb, ob, rb);
     }
@@ -3258,11 +3276,15 @@ k_q21_star_agg_pf(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk
         uint32_t sidx = (uint32_t)(sk[r] - ssmin);
         if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
         uint32_t year1 = dfirst[od[r] - dmin];
-        atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)], (unsigned long long)(int64_t)rv[r]);
+        unsigned long long* dst = LDSG ? &g[(year1 - 1) * 1000 + (brand1 - 1)]
+                                       : &group_sums[(year1 - 1) * 1000 + (brand1 - 1)];
+        atomicAdd(dst, (unsigned long long)(int64_t)rv[r]);
     }
-    __syncthreads();
-    for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x)
-        if (g[j] != 0) atomicAdd(&group_sums[j], g[j]);
+    if (LDSG) {
+        __syncthreads();
+        for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x)
+            if (g[j] != 0) atomicAdd(&group_sums[j], g[j]);
+    }
 }
 
 // ---------------------------------------------------------------------------
@@ -3446,14 +3468,26 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
     ARG_CHECK(parts->first16 && supps->first16 && dates->first16);
     HIP_CHECK(hipMemsetAsync(group_sums->ptr, 0, NG_Q21 * sizeof(int64_t), s->stream));
     if (q21_pf() && parts->prefilter) {
-        hipLaunchKernelGGL(k_q21_star_agg_pf, dim3(env_cap("GPUE_GRID_PF", 256)),
-                           dim3(BLOCK_Q21), 0, s->stream, (const int32_t*)pk->ptr,
-                           (const int32_t*)sk->ptr, (const int32_t*)od->ptr,
-                           (const int32_t*)rv->ptr, n, parts->prefilter, parts->set_min,
-                           (uint64_t)(parts->set_max - parts->set_min + 1), parts->first16,
-                           supps->bitset, supps->set_min,
-                           (uint64_t)(supps->set_max - supps->set_min + 1), dates->first16,
-                           dates->min_key, (unsigned long long*)group_sums->ptr);
+        const char* e = getenv("GPUE_Q21_PF");
+        int mode = e ? atoi(e) : 1;
+        const char* nt = getenv("GPUE_Q21_PF_NT");
+        bool use_nt = nt && atoi(nt);
+        int def_grid = mode == 2 ? 512 : 256; // global-group variant fits 2 blocks/CU
+        auto launch = [&](auto kern, int grid) {
+            hipLaunchKernelGGL(kern, dim3(env_cap("GPUE_GRID_PF", grid)),
+                               dim3(BLOCK_Q21), 0, s->stream, (const int32_t*)pk->ptr,
+                               (const int32_t*)sk->ptr, (const int32_t*)od->ptr,
+                               (const int32_t*)rv->ptr, n, parts->prefilter, parts->set_min,
+                               (uint64_t)(parts->set_max - parts->set_min + 1),
+                               parts->first16, supps->bitset, supps->set_min,
+                               (uint64_t)(supps->set_max - supps->set_min + 1),
+                               dates->first16, dates->min_key,
+                               (unsigned long long*)group_sums->ptr);
+        };
+        if (mode == 2 && use_nt) launch(k_q21_star_agg_pf<false, true>, def_grid);
+        else if (mode == 2) launch(k_q21_star_agg_pf<false, false>, def_grid);
+        else if (use_nt) launch(k_q21_star_agg_pf<true, true>, def_grid);
+        else launch(k_q21_star_agg_pf<true, false>, def_grid);
         HIP_CHECK(hipGetLastError());
         return GPUE_OK;
     }
