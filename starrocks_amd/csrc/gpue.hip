@@ -3470,8 +3470,11 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
     if (q21_pf() && parts->prefilter) {
         const char* e = getenv("GPUE_Q21_PF");
         int mode = e ? atoi(e) : 1;
+        // NT streams measured 2.267 vs 2.575 ms (frac 0.54 vs 0.47,
+        // gpurun_out/q21_var_sweep.log r02) — default on; the global-group
+        // 2-block variant (mode 2) measured 3.2 ms and stays for evidence
         const char* nt = getenv("GPUE_Q21_PF_NT");
-        bool use_nt = nt && atoi(nt);
+        bool use_nt = !nt || atoi(nt);
         int def_grid = mode == 2 ? 512 : 256; // global-group variant fits 2 blocks/CU
         auto launch = [&](auto kern, int grid) {
             hipLaunchKernelGGL(kern, dim3(env_cap("GPUE_GRID_PF", grid)),
