@@ -109,12 +109,33 @@ def build_dim_tables(eng, workload, rank=0, world=1):
 
 
 def cpu_affinity_cores():
-    """The cores this process may actually run on — os.cpu_count() lies under
-    container CPU quotas/affinity masks (VERDICT r01 weak #2)."""
+    """The CPU budget this process actually has. os.cpu_count() and even
+    sched_getaffinity lie under container CPU QUOTAS (cgroup cpu.max): the
+    GPU box reports 256 CPUs with a full affinity mask, but the oracle's
+    thread sweep (profiles/r02_cpu_diag.txt) peaks at 32 threads
+    (5.8 Grows/s) and collapses 26x at 256 — a >8x-oversubscribed OMP team
+    spends its time preempted at barriers. Use the cgroup quota when one is
+    set; otherwise the affinity count."""
     try:
-        return len(os.sched_getaffinity(0))
+        cores = len(os.sched_getaffinity(0))
     except AttributeError:
-        return os.cpu_count()
+        cores = os.cpu_count()
+    try:  # cgroup v2
+        with open("/sys/fs/cgroup/cpu.max") as f:
+            quota, period = f.read().split()
+            if quota != "max":
+                cores = min(cores, max(1, int(int(quota) / int(period))))
+    except OSError:
+        try:  # cgroup v1
+            with open("/sys/fs/cgroup/cpu/cpu.cfs_quota_us") as f:
+                quota = int(f.read())
+            with open("/sys/fs/cgroup/cpu/cpu.cfs_period_us") as f:
+                period = int(f.read())
+            if quota > 0:
+                cores = min(cores, max(1, quota // period))
+        except OSError:
+            pass
+    return cores
 
 
 def cpu_baseline(workload, rows_full):
@@ -149,20 +170,33 @@ def cpu_baseline(workload, rows_full):
         mn, _, dfirst = gen.build_date_dim_q43()
         run = lambda: orc.q43_kernel(ck, sk, pk, od, rv, sc, cfirst, sfirst,
                                      pfirst, dfirst, mn, threads=threads)
-    run()  # warmup: page-fault the dims, spin up the OMP team (untimed)
-    t0 = time.perf_counter()
-    passes = 0
-    while time.perf_counter() - t0 < 10.0:
-        run()
-        passes += 1
-    dt = time.perf_counter() - t0
-    rate = passes * sample / dt
+    # Thread-count calibration (profiles/r02_cpu_diag.txt): the box reports
+    # 256 CPUs but quota/contention makes big OMP teams collapse (peak at 32
+    # threads, 26x slower at 256). Probe a few candidates briefly and keep
+    # the fastest — the reported `cores` is what was actually used.
+    def rate_at(t, seconds):
+        nonlocal threads
+        threads = t
+        run()  # warm team (untimed)
+        t0 = time.perf_counter()
+        p = 0
+        while time.perf_counter() - t0 < seconds:
+            run()
+            p += 1
+        return p * sample / (time.perf_counter() - t0)
+
+    candidates = sorted({cores, 128, 64, 32, 16}, reverse=True)
+    candidates = [t for t in candidates if t <= cores] or [cores]
+    best = max(candidates, key=lambda t: rate_at(t, 0.8))
+    rate = rate_at(best, 8.0)
     gbps = rate * BYTES_PER_ROW[workload] / 1e9
-    return {"value": round(rate, 1), "unit": "rows/s", "cores": cores, "kind": "port",
-            "sample": f"{passes} passes over {sample} rows ({dt:.1f}s, oracle -O3 "
-                      f"-fopenmp, OMP_NUM_THREADS={threads}, sched_getaffinity={cores} "
-                      f"of cpu_count={os.cpu_count()}, achieved {gbps:.1f} GB/s "
-                      f"algorithmic, generation untimed, 1 warmup pass)"}
+    return {"value": round(rate, 1), "unit": "rows/s", "cores": best, "kind": "port",
+            "sample": f"oracle -O3 -fopenmp over {sample} rows, OMP threads "
+                      f"calibrated over {candidates} -> {best} "
+                      f"(sched_getaffinity={cores}, cpu_count={os.cpu_count()}; "
+                      f"large teams collapse under the box's CPU quota — "
+                      f"profiles/r02_cpu_diag.txt), achieved {gbps:.1f} GB/s "
+                      f"algorithmic, generation untimed"}
 
 
 def cpu_baseline_q3(rows_full):
@@ -181,6 +215,9 @@ def cpu_baseline_q3(rows_full):
     ok = np.empty(sample, np.uint64)
     os_ = np.empty(sample, np.int64)
     cores = cpu_affinity_cores()
+    # same quota-collapse mitigation as cpu_baseline (profiles/r02_cpu_diag.txt)
+    threads = min(cores, 32)
+    orc.load().orc_set_threads(threads)
 
     def run():
         orc.load().orc_q3_build_order_bits(orc._p(oc), orc._p(od_), n_orders,
@@ -198,11 +235,13 @@ def cpu_baseline_q3(rows_full):
     dt = time.perf_counter() - t0
     rate = passes * sample / dt
     gbps = rate * BYTES_PER_ROW["q3"] / 1e9
-    return {"value": round(rate, 1), "unit": "rows/s", "cores": cores, "kind": "port",
+    return {"value": round(rate, 1), "unit": "rows/s", "cores": threads, "kind": "port",
             "sample": f"{passes} passes over {sample} rows, dims scaled to "
                       f"{n_orders}/{n_custs} ({dt:.1f}s, OMP partition-scatter oracle, "
-                      f"sched_getaffinity={cores} of cpu_count={os.cpu_count()}, "
-                      f"achieved {gbps:.1f} GB/s algorithmic; generation untimed)"}
+                      f"{threads} threads of sched_getaffinity={cores}, "
+                      f"cpu_count={os.cpu_count()} — large teams collapse under "
+                      f"the box's CPU quota, profiles/r02_cpu_diag.txt; achieved "
+                      f"{gbps:.1f} GB/s algorithmic; generation untimed)"}
 
 
 def read_pmc_traffic(workload):

@@ -1044,6 +1044,14 @@ static inline uint64_t xxh3_rrmxmx(uint64_t h, uint64_t len) {
     return h ^ (h >> 28);
 }
 
+void orc_set_threads(int n) {
+#ifdef _OPENMP
+    if (n > 0) omp_set_num_threads(n);
+#else
+    (void)n;
+#endif
+}
+
 uint64_t orc_xxh3_64_4to8(const void* data, int32_t len, uint64_t seed) {
     uint32_t s32 = (uint32_t)seed;
     uint32_t swapped = ((s32 & 0xFFu) << 24) | ((s32 & 0xFF00u) << 8) |

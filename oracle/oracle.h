@@ -305,3 +305,5 @@ void orc_xxh3_hash_i32(const int32_t* col, uint64_t n, uint32_t* hashes);
 void orc_xxh3_hash_i64(const int64_t* col, uint64_t n, uint32_t* hashes);
 void orc_partition_channel_xxh3_u32(const uint32_t* keys, uint64_t n,
                                     uint32_t num_channels, uint32_t* channel_ids);
+/* set the OMP team size for the parameterless kernels (q3 legs) */
+void orc_set_threads(int n);

@@ -69,6 +69,7 @@ def _decl(lib):
     lib.orc_join_select_method.restype = c_i32
     lib.orc_join_select_method.argtypes = [c_i32, c_i32, c_u64, c_i64, c_i64, c_i32,
                                            c_i32, c_i32, c_i32, c_u64, c_u64]
+    lib.orc_set_threads.argtypes = [c_i32]
     lib.orc_xxh3_64_4to8.restype = c_u64
     lib.orc_xxh3_64_4to8.argtypes = [c_vp, c_i32, c_u64]
     lib.orc_xxh3_hash_i32.argtypes = [c_vp, c_u64, c_vp]
