@@ -3552,11 +3552,18 @@ int gpue_q21_star_agg(gpue_session* s, gpue_join_table* parts, gpue_join_table* 
 // ---------------------------------------------------------------------------
 static constexpr int NG_Q43 = 800;
 
+// PF: stage the 64 KB folded part prefilter in LDS (same structure as
+// k_q21_star_agg_pf) and answer the per-row part test there; only ~10%
+// "maybe" rows confirm against the exact L2 bitset, so the quad mask stays
+// EXACT and the deferred-stream-load savings (P(line skipped) = 0.52) are
+// preserved while the TA divergent-request leg shrinks ~10x. 64+6.4 KB LDS
+// -> two 1024-thread blocks/CU (32 waves).
+template <bool PF>
 __global__ __launch_bounds__(BLOCK_Q21) void
 k_q43_star_agg(const int32_t* __restrict__ ck, const int32_t* __restrict__ sk,
                const int32_t* __restrict__ pk, const int32_t* __restrict__ od,
                const int32_t* __restrict__ rv, const int32_t* __restrict__ sc,
-               uint64_t n,
+               uint64_t n, const uint32_t* __restrict__ prefilter,
                const uint32_t* __restrict__ cbits, int64_t csmin, uint64_t csint,
                const uint32_t* __restrict__ sbits, int64_t ssmin, uint64_t ssint,
                const uint16_t* __restrict__ sfirst,
@@ -3565,6 +3572,9 @@ k_q43_star_agg(const int32_t* __restrict__ ck, const int32_t* __restrict__ sk,
                const uint16_t* __restrict__ dfirst, int64_t dmin,
                unsigned long long* __restrict__ group_sums) {
     __shared__ unsigned long long g[NG_Q43];
+    __shared__ uint32_t pf[PF ? PF_WORDS : 1];
+    if (PF)
+        for (uint32_t w = threadIdx.x; w < PF_WORDS; w += blockDim.x) pf[w] = prefilter[w];
     for (int j = threadIdx.x; j < NG_Q43; j += blockDim.x) g[j] = 0;
     __syncthreads();
     const uint64_t n4 = n / 4;
@@ -3586,7 +3596,13 @@ k_q43_star_agg(const int32_t* __restrict__ ck, const int32_t* __restrict__ sk,
         for (int j = 0; j < 4; j++) {
             uint32_t idx = (uint32_t)((&p4.x)[j] - psmin);
             uint32_t cidx = idx < psint ? idx : 0u;
-            m |= ((idx < psint) & (pbits[cidx >> 5] >> (cidx & 31)) & 1u) << j;
+            if (PF) {
+                uint32_t fidx = cidx & PF_MASK;
+                bool maybe = (idx < psint) & (pf[fidx >> 5] >> (fidx & 31)) & 1u;
+                if (maybe) m |= ((pbits[cidx >> 5] >> (cidx & 31)) & 1u) << j;
+            } else {
+                m |= ((idx < psint) & (pbits[cidx >> 5] >> (cidx & 31)) & 1u) << j;
+            }
         }
         return m;
     };
@@ -3660,10 +3676,15 @@ int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs, gpue_join_t
     ARG_CHECK(custs->bitset && supps->bitset && parts->bitset);
     ARG_CHECK(custs->min_key == 1 && supps->min_key == 1 && parts->min_key == 1);
     HIP_CHECK(hipMemsetAsync(group_sums->ptr, 0, NG_Q43 * sizeof(int64_t), s->stream));
-    hipLaunchKernelGGL(k_q43_star_agg, dim3(env_cap("GPUE_GRID_Q43", 256)), dim3(BLOCK_Q21), 0, s->stream,
+    const char* pfe = getenv("GPUE_Q43_PF");
+    bool use_pf = (!pfe || atoi(pfe)) && parts->prefilter;
+    int def_grid = use_pf ? 512 : 256; // PF fits 2 blocks/CU
+    auto kern = use_pf ? k_q43_star_agg<true> : k_q43_star_agg<false>;
+    hipLaunchKernelGGL(kern, dim3(env_cap("GPUE_GRID_Q43", def_grid)), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)ck->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)pk->ptr, (const int32_t*)od->ptr,
                        (const int32_t*)rv->ptr, (const int32_t*)sc->ptr, n,
+                       parts->prefilter,
                        custs->bitset, custs->set_min,
                        (uint64_t)(custs->set_max - custs->set_min + 1),
                        supps->bitset, supps->set_min,
