@@ -513,3 +513,19 @@ int gpue_timer_stop(gpue_session* s, float* ms_out);
 }
 #endif
 #endif
+
+/* ---- chunked-exchange overlap support (SURVEY.md §7 hard part (d)) ------
+ * The session's HIP stream as an opaque pointer (wrap as a torch
+ * ExternalStream to event-order RCCL collectives against engine kernels),
+ * plus accumulate-only forms of the probe+agg steps so a received row-block
+ * can be probed while the next block's all-to-all is in flight — the analog
+ * of the reference's overlapped SinkBuffer (sink_buffer.cpp:533-536). */
+void* gpue_session_stream(gpue_session* s);
+int gpue_q43_star_agg_accum_async(gpue_session* s, gpue_join_table* custs,
+                                  gpue_join_table* supps, gpue_join_table* parts,
+                                  gpue_join_table* dates, gpue_dbuf* ck, gpue_dbuf* sk,
+                                  gpue_dbuf* pk, gpue_dbuf* od, gpue_dbuf* rv,
+                                  gpue_dbuf* sc, uint64_t n, gpue_dbuf* group_sums);
+int gpue_q3_probe_accum(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
+                        gpue_dbuf* ship, uint64_t n, gpue_dbuf* order_bits,
+                        int32_t ship_cutoff, gpue_agg_table* at);

@@ -223,6 +223,14 @@ int gpue_sync(gpue_session* s) {
     return GPUE_OK;
 }
 
+// The session's HIP stream as an opaque pointer — lets the host wrap it as
+// a torch ExternalStream so RCCL collectives on a second stream can be
+// event-ordered against engine kernels without full-device syncs (the
+// exchange/compute overlap of SURVEY.md (S)7 hard part (d); the reference
+// overlaps its sink the same way, sink_buffer.cpp:533-536).
+extern "C" void* gpue_session_stream(gpue_session* s);
+void* gpue_session_stream(gpue_session* s) { return s ? (void*)s->stream : nullptr; }
+
 int gpue_dbuf_alloc(gpue_session* s, uint64_t bytes, gpue_dbuf** out) {
     ARG_CHECK(s && out && bytes > 0);
     HIP_CHECK(hipSetDevice(s->device));
@@ -3666,6 +3674,12 @@ extern "C" int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs,
                                        gpue_join_table* dates, gpue_dbuf* ck, gpue_dbuf* sk,
                                        gpue_dbuf* pk, gpue_dbuf* od, gpue_dbuf* rv,
                                        gpue_dbuf* sc, uint64_t n, gpue_dbuf* group_sums);
+extern "C" int gpue_q43_star_agg_accum_async(gpue_session* s, gpue_join_table* custs,
+                                             gpue_join_table* supps, gpue_join_table* parts,
+                                             gpue_join_table* dates, gpue_dbuf* ck,
+                                             gpue_dbuf* sk, gpue_dbuf* pk, gpue_dbuf* od,
+                                             gpue_dbuf* rv, gpue_dbuf* sc, uint64_t n,
+                                             gpue_dbuf* group_sums);
 int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs, gpue_join_table* supps,
                             gpue_join_table* parts, gpue_join_table* dates, gpue_dbuf* ck,
                             gpue_dbuf* sk, gpue_dbuf* pk, gpue_dbuf* od, gpue_dbuf* rv,
@@ -3676,6 +3690,22 @@ int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs, gpue_join_t
     ARG_CHECK(custs->bitset && supps->bitset && parts->bitset);
     ARG_CHECK(custs->min_key == 1 && supps->min_key == 1 && parts->min_key == 1);
     HIP_CHECK(hipMemsetAsync(group_sums->ptr, 0, NG_Q43 * sizeof(int64_t), s->stream));
+    return gpue_q43_star_agg_accum_async(s, custs, supps, parts, dates, ck, sk, pk, od,
+                                         rv, sc, n, group_sums);
+}
+
+// accumulate-only form (no zeroing): the chunked-exchange pipeline probes
+// each received row-block as it lands, summing into the same group buffer
+int gpue_q43_star_agg_accum_async(gpue_session* s, gpue_join_table* custs,
+                                  gpue_join_table* supps, gpue_join_table* parts,
+                                  gpue_join_table* dates, gpue_dbuf* ck, gpue_dbuf* sk,
+                                  gpue_dbuf* pk, gpue_dbuf* od, gpue_dbuf* rv,
+                                  gpue_dbuf* sc, uint64_t n, gpue_dbuf* group_sums) {
+    ARG_CHECK(s && custs && supps && parts && dates && ck && sk && pk && od && rv && sc);
+    ARG_CHECK(group_sums && group_sums->bytes >= NG_Q43 * sizeof(int64_t));
+    ARG_CHECK(supps->first16 && parts->first16 && dates->first16);
+    ARG_CHECK(custs->bitset && supps->bitset && parts->bitset);
+    ARG_CHECK(custs->min_key == 1 && supps->min_key == 1 && parts->min_key == 1);
     const char* pfe = getenv("GPUE_Q43_PF");
     bool use_pf = (!pfe || atoi(pfe)) && parts->prefilter;
     int def_grid = use_pf ? 512 : 256; // PF fits 2 blocks/CU
@@ -5443,6 +5473,27 @@ int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbu
                  (unsigned long long)max_out);
         return GPUE_ERR_ARG;
     }
+    return GPUE_OK;
+}
+
+// probe-accumulate only (no table reset, no emit, async): the chunked
+// exchange pipeline probes each received block as it lands; callers reset
+// once per step (gpue_agg_table_reset) and emit once at the end
+// (gpue_hash_agg_emit_u64)
+extern "C" int gpue_q3_probe_accum(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext,
+                                   gpue_dbuf* disc, gpue_dbuf* ship, uint64_t n,
+                                   gpue_dbuf* order_bits, int32_t ship_cutoff,
+                                   gpue_agg_table* at);
+int gpue_q3_probe_accum(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
+                        gpue_dbuf* ship, uint64_t n, gpue_dbuf* order_bits,
+                        int32_t ship_cutoff, gpue_agg_table* at) {
+    ARG_CHECK(s && lk && ext && disc && ship && order_bits && at);
+    hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))), dim3(BLOCK), 0, s->stream,
+                       (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
+                       (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
+                       (const uint32_t*)order_bits->ptr, ship_cutoff, at->slots, at->sums,
+                       at->counts, at->cap - 1, s->d_agg_err);
+    HIP_CHECK(hipGetLastError());
     return GPUE_OK;
 }
 

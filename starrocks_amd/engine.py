@@ -113,6 +113,9 @@ def _declare(lib):
         "gpue_q21_star_agg_async": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_q21_star_agg_pipe": (c_i32, [c_vp] * 8 + [c_u64, c_vp, c_vp, c_i32]),
         "gpue_partition_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
+        "gpue_session_stream": (c_vp, [c_vp]),
+        "gpue_q43_star_agg_accum_async": (c_i32, [c_vp] * 5 + [c_vp] * 6 + [c_u64, c_vp]),
+        "gpue_q3_probe_accum": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_vp, c_i32, c_vp]),
         "gpue_partition_xxh3_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
         "gpue_gather_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_gather_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
@@ -693,6 +696,22 @@ class Engine:
         _ck(self._lib, self._lib.gpue_partition_i32(self._h, keys._h, n, num_channels,
                                                     sp.ctypes.data_as(c_vp), row_indexes._h))
         return sp
+
+    def stream_ptr(self) -> int:
+        """The session's HIP stream (for torch.cuda.ExternalStream wrapping —
+        event-orders RCCL collectives against engine kernels)."""
+        return int(self._lib.gpue_session_stream(self._h) or 0)
+
+    def q43_star_agg_accum_async(self, custs, supps, parts, dates, ck, sk, pk, od,
+                                 rv, sc, n, group_sums: DBuf):
+        _ck(self._lib, self._lib.gpue_q43_star_agg_accum_async(
+            self._h, custs._h, supps._h, parts._h, dates._h, ck._h, sk._h, pk._h,
+            od._h, rv._h, sc._h, n, group_sums._h))
+
+    def q3_probe_accum(self, lk, ext, disc, ship, n, order_bits, ship_cutoff, at):
+        _ck(self._lib, self._lib.gpue_q3_probe_accum(
+            self._h, lk._h, ext._h, disc._h, ship._h, n, order_bits._h,
+            ship_cutoff, at))
 
     def partition_xxh3(self, keys: DBuf, n, num_channels, row_indexes: DBuf) -> np.ndarray:
         """Version-1 exchange hash (xxh3) partition
