@@ -244,7 +244,12 @@ def cpu_baseline_q3(rows_full):
                       f"{gbps:.1f} GB/s algorithmic; generation untimed)"}
 
 
-def read_pmc_traffic(workload):
+def read_pmc_traffic(workload, rows, world):
+    """Committed PMC-counted bytes per launch — only valid for the standard
+    N=1 config the counters were collected on."""
+    if world != 1 or rows != {"q1": SF10_ROWS, "q21": SF100_ROWS,
+                              "q43": SF100_ROWS, "q3": Q3_ROWS_PER_GPU}[workload]:
+        return None
     p = os.path.join(REPO, "profiles", f"pmc_{workload}.json")
     if os.path.exists(p):
         with open(p) as f:
@@ -692,7 +697,7 @@ def main():
     kernel_rows = rows if not (wl in ("q43", "q3") and world > 1) else max(n_recv, 1)
     algo_bytes = BYTES_PER_ROW[wl] * kernel_rows
     achieved_gbps = algo_bytes / (kernel_ms / 1e3) / 1e9
-    traffic = read_pmc_traffic(wl)
+    traffic = read_pmc_traffic(wl, rows, world)
     # Honest presentation: `achieved`/`frac` are
     # kernel-only HIP-event timing of ALGORITHMIC bytes; `achieved_step`/
     # `frac_step` divide the same bytes by the full step (incl. result read);

@@ -505,15 +505,6 @@ int gpue_page_decode_bshuf_lz4_i32(gpue_session* s, gpue_dbuf* page, uint32_t n_
 int gpue_topk_i64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uint64_t n,
                   int k, uint64_t* out_keys, int64_t* out_vals);
 
-/* ---- event timing on the session stream (bench roofline evidence) ---- */
-int gpue_timer_start(gpue_session* s);
-int gpue_timer_stop(gpue_session* s, float* ms_out);
-
-#ifdef __cplusplus
-}
-#endif
-#endif
-
 /* ---- chunked-exchange overlap support (SURVEY.md §7 hard part (d)) ------
  * The session's HIP stream as an opaque pointer (wrap as a torch
  * ExternalStream to event-order RCCL collectives against engine kernels),
@@ -529,3 +520,25 @@ int gpue_q43_star_agg_accum_async(gpue_session* s, gpue_join_table* custs,
 int gpue_q3_probe_accum(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
                         gpue_dbuf* ship, uint64_t n, gpue_dbuf* order_bits,
                         int32_t ship_cutoff, gpue_agg_table* at);
+
+/* ---- pinned double-buffered H2D ingest ----
+ * north_star's "columnar batches pinned and streamed to HBM": two
+ * hipHostMalloc staging buffers on the session's second stream; while chunk
+ * k DMAs pinned->HBM, the host fills the other buffer — the morsel-driven
+ * async-io shape of the reference's scan operator
+ * (be/src/exec/pipeline/scan/scan_operator.h:40). */
+typedef struct gpue_ingest gpue_ingest;
+int gpue_ingest_create(gpue_session* s, uint64_t chunk_bytes, gpue_ingest** out);
+int gpue_ingest_push(gpue_ingest* g, const void* host, uint64_t bytes, gpue_dbuf* dst,
+                     uint64_t dst_off);
+int gpue_ingest_sync(gpue_ingest* g);
+void gpue_ingest_destroy(gpue_ingest* g);
+
+/* ---- event timing on the session stream (bench roofline evidence) ---- */
+int gpue_timer_start(gpue_session* s);
+int gpue_timer_stop(gpue_session* s, float* ms_out);
+
+#ifdef __cplusplus
+}
+#endif
+#endif

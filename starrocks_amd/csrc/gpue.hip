@@ -257,6 +257,80 @@ int gpue_dbuf_wrap(gpue_session* s, void* device_ptr, uint64_t bytes, gpue_dbuf*
     return GPUE_OK;
 }
 
+// ---------------------------------------------------------------------------
+// Pinned double-buffered H2D ingest — north_star's "columnar batches pinned
+// and streamed to HBM"; the morsel-driven async-io shape of the reference's
+// scan operator (be/src/exec/pipeline/scan/scan_operator.h:40: each morsel
+// is staged and shipped while the next is prepared). Two hipHostMalloc
+// staging buffers on the session's second stream: while chunk k DMAs
+// pinned->HBM, the host memcpys chunk k+1 into the other buffer. Pageable
+// single-shot h2d measured 56.5 GB/s on this path; pinned
+// DMA is the PCIe-rate fix.
+// ---------------------------------------------------------------------------
+struct gpue_ingest {
+    gpue_session* s;
+    void* pin[2];
+    uint64_t chunk_bytes;
+    hipEvent_t done[2]; // DMA completion per staging buffer
+    int cur;
+};
+
+extern "C" {
+int gpue_ingest_create(gpue_session* s, uint64_t chunk_bytes, gpue_ingest** out);
+int gpue_ingest_push(gpue_ingest* g, const void* host, uint64_t bytes, gpue_dbuf* dst,
+                     uint64_t dst_off);
+int gpue_ingest_sync(gpue_ingest* g);
+void gpue_ingest_destroy(gpue_ingest* g);
+}
+
+int gpue_ingest_create(gpue_session* s, uint64_t chunk_bytes, gpue_ingest** out) {
+    ARG_CHECK(s && out && chunk_bytes >= 4096);
+    gpue_ingest* g = new gpue_ingest{s, {nullptr, nullptr}, chunk_bytes, {}, 0};
+    for (int i = 0; i < 2; i++) {
+        HIP_CHECK(hipHostMalloc(&g->pin[i], chunk_bytes));
+        HIP_CHECK(hipEventCreateWithFlags(&g->done[i], hipEventDisableTiming));
+        HIP_CHECK(hipEventRecord(g->done[i], s->stream2)); // initially signalled
+    }
+    *out = g;
+    return GPUE_OK;
+}
+
+int gpue_ingest_push(gpue_ingest* g, const void* host, uint64_t bytes, gpue_dbuf* dst,
+                     uint64_t dst_off) {
+    ARG_CHECK(g && host && dst && dst_off + bytes <= dst->bytes);
+    const uint8_t* src = (const uint8_t*)host;
+    uint64_t off = 0;
+    while (off < bytes) {
+        uint64_t sz = bytes - off < g->chunk_bytes ? bytes - off : g->chunk_bytes;
+        int c = g->cur;
+        // wait for this staging buffer's previous DMA before refilling it —
+        // the OTHER buffer's DMA proceeds concurrently with this memcpy
+        HIP_CHECK(hipEventSynchronize(g->done[c]));
+        memcpy(g->pin[c], src + off, sz);
+        HIP_CHECK(hipMemcpyAsync((uint8_t*)dst->ptr + dst_off + off, g->pin[c], sz,
+                                 hipMemcpyHostToDevice, g->s->stream2));
+        HIP_CHECK(hipEventRecord(g->done[c], g->s->stream2));
+        g->cur ^= 1;
+        off += sz;
+    }
+    return GPUE_OK;
+}
+
+int gpue_ingest_sync(gpue_ingest* g) {
+    ARG_CHECK(g);
+    HIP_CHECK(hipStreamSynchronize(g->s->stream2));
+    return GPUE_OK;
+}
+
+void gpue_ingest_destroy(gpue_ingest* g) {
+    if (!g) return;
+    for (int i = 0; i < 2; i++) {
+        if (g->pin[i]) (void)hipHostFree(g->pin[i]);
+        if (g->done[i]) (void)hipEventDestroy(g->done[i]);
+    }
+    delete g;
+}
+
 // expose the device pointer (for torch interop / sub-buffer views)
 extern "C" int gpue_dbuf_ptr(gpue_dbuf* b, void** out);
 int gpue_dbuf_ptr(gpue_dbuf* b, void** out) {

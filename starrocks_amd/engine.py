@@ -114,6 +114,10 @@ def _declare(lib):
         "gpue_q21_star_agg_pipe": (c_i32, [c_vp] * 8 + [c_u64, c_vp, c_vp, c_i32]),
         "gpue_partition_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
         "gpue_session_stream": (c_vp, [c_vp]),
+        "gpue_ingest_create": (c_i32, [c_vp, c_u64, ctypes.POINTER(c_vp)]),
+        "gpue_ingest_push": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_u64]),
+        "gpue_ingest_sync": (c_i32, [c_vp]),
+        "gpue_ingest_destroy": (None, [c_vp]),
         "gpue_q43_star_agg_accum_async": (c_i32, [c_vp] * 5 + [c_vp] * 6 + [c_u64, c_vp]),
         "gpue_q3_probe_accum": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_vp, c_i32, c_vp]),
         "gpue_partition_xxh3_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
@@ -712,6 +716,24 @@ class Engine:
         _ck(self._lib, self._lib.gpue_q3_probe_accum(
             self._h, lk._h, ext._h, disc._h, ship._h, n, order_bits._h,
             ship_cutoff, at))
+
+    def ingest_create(self, chunk_bytes=32 << 20):
+        """Pinned double-buffered H2D ingest (scan_operator.h:40 morsel
+        shape); push host arrays with ingest_push, then ingest_sync."""
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_ingest_create(self._h, chunk_bytes,
+                                                    ctypes.byref(h)))
+        return h
+
+    def ingest_push(self, ing, arr: np.ndarray, dst: DBuf, dst_off=0):
+        _ck(self._lib, self._lib.gpue_ingest_push(
+            ing, arr.ctypes.data_as(c_vp), arr.nbytes, dst._h, dst_off))
+
+    def ingest_sync(self, ing):
+        _ck(self._lib, self._lib.gpue_ingest_sync(ing))
+
+    def ingest_destroy(self, ing):
+        self._lib.gpue_ingest_destroy(ing)
 
     def partition_xxh3(self, keys: DBuf, n, num_channels, row_indexes: DBuf) -> np.ndarray:
         """Version-1 exchange hash (xxh3) partition
