@@ -529,6 +529,10 @@ int gpue_q3_probe_accum(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbu
  * (be/src/exec/pipeline/scan/scan_operator.h:40). */
 typedef struct gpue_ingest gpue_ingest;
 int gpue_ingest_create(gpue_session* s, uint64_t chunk_bytes, gpue_ingest** out);
+/* page-locked host memory for producers that fill batches in place (zero
+ * staging copy; DMA at full PCIe rate) */
+int gpue_pinned_alloc(gpue_session* s, uint64_t bytes, void** host_ptr);
+void gpue_pinned_free(void* host_ptr);
 int gpue_ingest_push(gpue_ingest* g, const void* host, uint64_t bytes, gpue_dbuf* dst,
                      uint64_t dst_off);
 int gpue_ingest_sync(gpue_ingest* g);

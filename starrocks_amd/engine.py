@@ -115,6 +115,8 @@ def _declare(lib):
         "gpue_partition_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
         "gpue_session_stream": (c_vp, [c_vp]),
         "gpue_ingest_create": (c_i32, [c_vp, c_u64, ctypes.POINTER(c_vp)]),
+        "gpue_pinned_alloc": (c_i32, [c_vp, c_u64, ctypes.POINTER(c_vp)]),
+        "gpue_pinned_free": (None, [c_vp]),
         "gpue_ingest_push": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_u64]),
         "gpue_ingest_sync": (c_i32, [c_vp]),
         "gpue_ingest_destroy": (None, [c_vp]),
@@ -716,6 +718,15 @@ class Engine:
         _ck(self._lib, self._lib.gpue_q3_probe_accum(
             self._h, lk._h, ext._h, disc._h, ship._h, n, order_bits._h,
             ship_cutoff, at))
+
+    def pinned_alloc(self, nbytes) -> int:
+        """Page-locked host allocation (returns raw host pointer)."""
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_pinned_alloc(self._h, nbytes, ctypes.byref(h)))
+        return h.value
+
+    def pinned_free(self, host_ptr):
+        self._lib.gpue_pinned_free(c_vp(host_ptr))
 
     def ingest_create(self, chunk_bytes=32 << 20):
         """Pinned double-buffered H2D ingest (scan_operator.h:40 morsel
