@@ -521,6 +521,16 @@ int gpue_q3_probe_accum(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbu
                         gpue_dbuf* ship, uint64_t n, gpue_dbuf* order_bits,
                         int32_t ship_cutoff, gpue_agg_table* at);
 
+/* ---- growable agg table ----
+ * The MI355X analog of the reference's two-level conversion
+ * (Aggregator::try_convert_to_two_level_map, aggregator.cpp:1237-1241;
+ * agg_hash_variant.cpp:318): size() reads the claimed-group counter;
+ * ensure() doubles + rehashes on device until a push of additional_rows can
+ * never overflow (load factor kept <= 5/8). Call before each chunk push,
+ * as the reference checks before each chunk. */
+int gpue_agg_table_size(gpue_session* s, gpue_agg_table* t, uint64_t* n_groups);
+int gpue_agg_table_ensure(gpue_session* s, gpue_agg_table* t, uint64_t additional_rows);
+
 /* ---- pinned double-buffered H2D ingest ----
  * north_star's "columnar batches pinned and streamed to HBM": two
  * hipHostMalloc staging buffers on the session's second stream; while chunk

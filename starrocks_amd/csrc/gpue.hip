@@ -4352,6 +4352,7 @@ struct gpue_agg_table {
     unsigned long long* sums;
     unsigned long long* counts;
     unsigned long long* cursor;
+    unsigned long long* n_groups; // device count of claimed slots (grow trigger)
 };
 
 extern "C" {
@@ -4363,11 +4364,14 @@ int gpue_agg_table_create(gpue_session* s, uint64_t capacity, gpue_agg_table** o
     ARG_CHECK(s && out && capacity >= 16);
     uint64_t cap = 16;
     while (cap < capacity) cap <<= 1;
-    gpue_agg_table* t = new gpue_agg_table{s, cap, nullptr, nullptr, nullptr, nullptr};
+    gpue_agg_table* t = new gpue_agg_table{s, cap, nullptr, nullptr, nullptr, nullptr,
+                                           nullptr};
     HIP_CHECK(hipMalloc(&t->slots, cap * 8));
     HIP_CHECK(hipMalloc(&t->sums, cap * 8));
     HIP_CHECK(hipMalloc(&t->counts, cap * 8));
     HIP_CHECK(hipMalloc(&t->cursor, 8));
+    HIP_CHECK(hipMalloc(&t->n_groups, 8));
+    HIP_CHECK(hipMemsetAsync(t->n_groups, 0, 8, s->stream));
     *out = t;
     return GPUE_OK;
 }
@@ -4380,6 +4384,7 @@ void gpue_agg_table_destroy(gpue_agg_table* t) {
     (void)hipFree(t->sums);
     (void)hipFree(t->counts);
     (void)hipFree(t->cursor);
+    (void)hipFree(t->n_groups);
     delete t;
 }
 
@@ -4388,12 +4393,87 @@ static int agg_table_reset(gpue_agg_table* t, bool with_counts = true) {
     HIP_CHECK(hipMemsetAsync(t->sums, 0, t->cap * 8, t->s->stream));
     if (with_counts) HIP_CHECK(hipMemsetAsync(t->counts, 0, t->cap * 8, t->s->stream));
     HIP_CHECK(hipMemsetAsync(t->cursor, 0, 8, t->s->stream));
+    HIP_CHECK(hipMemsetAsync(t->n_groups, 0, 8, t->s->stream));
     return GPUE_OK;
 }
 
 int gpue_agg_table_reset(gpue_agg_table* t) {
     ARG_CHECK(t);
     return agg_table_reset(t);
+}
+
+// ---------------------------------------------------------------------------
+// Growable table — the MI355X analog of the reference's two-level
+// conversion (Aggregator::try_convert_to_two_level_map, aggregator.cpp:
+// 1237-1241; AggHashMapVariant::convert_to_two_level, agg_hash_variant.cpp:
+// 318): when the map crosses a size threshold mid-stream, the reference
+// swaps in a 16-way phmap parallel map to keep rehash/lock costs bounded.
+// Our open-addressing CAS table has no locks, so the property to preserve
+// is "a chunk push never overflows": callers check before each chunk
+// (gpue_agg_table_ensure) and the table doubles + rehashes on device. Keys
+// in the old table are unique, so the rehash claim never contends on
+// payload writes and the per-slot stores need no atomics.
+// ---------------------------------------------------------------------------
+__global__ void k_agg_rehash(const unsigned long long* __restrict__ os,
+                             const unsigned long long* __restrict__ osum,
+                             const unsigned long long* __restrict__ ocnt, uint64_t ocap,
+                             unsigned long long* __restrict__ ns,
+                             unsigned long long* __restrict__ nsum,
+                             unsigned long long* __restrict__ ncnt, uint64_t ncap_mask) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < ocap;
+         i += stride) {
+        unsigned long long k = os[i];
+        if (k == AGG_EMPTY) continue;
+        uint64_t s = ((k * 11400714819323198485ull) >> 32) & ncap_mask;
+        while (atomicCAS(&ns[s], AGG_EMPTY, k) != AGG_EMPTY) s = (s + 1) & ncap_mask;
+        nsum[s] = osum[i];
+        ncnt[s] = ocnt[i];
+    }
+}
+
+extern "C" {
+int gpue_agg_table_size(gpue_session* s, gpue_agg_table* t, uint64_t* n_groups);
+int gpue_agg_table_ensure(gpue_session* s, gpue_agg_table* t, uint64_t additional_rows);
+}
+
+int gpue_agg_table_size(gpue_session* s, gpue_agg_table* t, uint64_t* n_groups) {
+    ARG_CHECK(s && t && n_groups);
+    unsigned long long g = 0;
+    HIP_CHECK(hipMemcpyAsync(&g, t->n_groups, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    *n_groups = g;
+    return GPUE_OK;
+}
+
+int gpue_agg_table_ensure(gpue_session* s, gpue_agg_table* t, uint64_t additional_rows) {
+    ARG_CHECK(s && t);
+    uint64_t groups = 0;
+    int rc = gpue_agg_table_size(s, t, &groups);
+    if (rc != GPUE_OK) return rc;
+    // worst case every incoming row is a new group; keep load factor <= 5/8
+    uint64_t need = groups + additional_rows;
+    if (need + need / 2 <= t->cap) return GPUE_OK;
+    uint64_t ncap = t->cap;
+    while (need + need / 2 > ncap) ncap <<= 1;
+    unsigned long long *ns = nullptr, *nsum = nullptr, *ncnt = nullptr;
+    HIP_CHECK(hipMalloc(&ns, ncap * 8));
+    HIP_CHECK(hipMalloc(&nsum, ncap * 8));
+    HIP_CHECK(hipMalloc(&ncnt, ncap * 8));
+    HIP_CHECK(hipMemsetAsync(ns, 0xFF, ncap * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(nsum, 0, ncap * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(ncnt, 0, ncap * 8, s->stream));
+    hipLaunchKernelGGL(k_agg_rehash, dim3(grid_for(t->cap)), dim3(BLOCK), 0, s->stream,
+                       t->slots, t->sums, t->counts, t->cap, ns, nsum, ncnt, ncap - 1);
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(t->slots);
+    (void)hipFree(t->sums);
+    (void)hipFree(t->counts);
+    t->slots = ns;
+    t->sums = nsum;
+    t->counts = ncnt;
+    t->cap = ncap;
+    return GPUE_OK;
 }
 
 __global__ void k_hash_agg_sum(const uint64_t* __restrict__ keys,
@@ -4809,7 +4889,8 @@ __global__ void k_hash_agg_push(const uint64_t* __restrict__ keys,
                                 unsigned long long* __restrict__ counts, uint64_t cap_mask,
                                 int update_only, uint8_t* __restrict__ miss_mask,
                                 unsigned long long* __restrict__ hits_out,
-                                unsigned int* __restrict__ err) {
+                                unsigned int* __restrict__ err,
+                                unsigned long long* __restrict__ ngroups) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     unsigned long long local_hits = 0;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
@@ -4842,6 +4923,7 @@ __global__ void k_hash_agg_push(const uint64_t* __restrict__ keys,
                         atomicAdd(&counts[slot], c);
                     }
                     if (old == k) local_hits++;
+                    else if (ngroups) atomicAdd(ngroups, 1ull);
                     if (miss_mask) miss_mask[i] = (old != k);
                     break;
                 }
@@ -4892,7 +4974,7 @@ static int hash_agg_push_impl(gpue_session* s, gpue_agg_table* at, gpue_dbuf* ke
                        cnts ? (const int64_t*)cnts->ptr : nullptr, n, at->slots, at->sums,
                        at->counts, at->cap - 1, update_only,
                        miss_mask ? (uint8_t*)miss_mask->ptr : nullptr, d_hits,
-                       s->d_agg_err);
+                       s->d_agg_err, at->n_groups);
     HIP_CHECK(hipGetLastError());
     if (hits_out) {
         unsigned long long h = 0;
@@ -5141,7 +5223,8 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
                                unsigned long long* __restrict__ slots,
                                unsigned long long* __restrict__ sums,
                                unsigned long long* __restrict__ counts, uint64_t cap_mask,
-                               unsigned int* __restrict__ err) {
+                               unsigned int* __restrict__ err,
+                               unsigned long long* __restrict__ ngroups) {
     (void)counts;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
@@ -5162,7 +5245,11 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
             if (cur == k) { atomicAdd(&sums[s], v); break; }
             if (cur == AGG_EMPTY) {
                 unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
-                if (old == AGG_EMPTY || old == k) { atomicAdd(&sums[s], v); break; }
+                if (old == AGG_EMPTY || old == k) {
+                    if (old == AGG_EMPTY && ngroups) atomicAdd(ngroups, 1ull);
+                    atomicAdd(&sums[s], v);
+                    break;
+                }
             }
             s = (s + 1) & cap_mask;
             if (--left == 0) { atomicOr(err, AGG_ERR_FULL); break; }
@@ -5381,7 +5468,8 @@ __global__ void k_q3_probe_slices_fused(const uint32_t* __restrict__ keys,
                                         const uint32_t* __restrict__ order_bits,
                                         unsigned long long* __restrict__ slots,
                                         unsigned long long* __restrict__ sums,
-                                        uint64_t cap_mask, unsigned int* __restrict__ err) {
+                                        uint64_t cap_mask, unsigned int* __restrict__ err,
+                                        unsigned long long* __restrict__ ngroups) {
     uint32_t p = block_part[blockIdx.x];
     uint64_t base = pstart[p];
     uint64_t cnt = pstart[p + 1] - base;
@@ -5400,7 +5488,11 @@ __global__ void k_q3_probe_slices_fused(const uint32_t* __restrict__ keys,
             if (cur_ == k) { atomicAdd(&sums[s], v); break; }
             if (cur_ == AGG_EMPTY) {
                 unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
-                if (old == AGG_EMPTY || old == k) { atomicAdd(&sums[s], v); break; }
+                if (old == AGG_EMPTY || old == k) {
+                    if (old == AGG_EMPTY && ngroups) atomicAdd(ngroups, 1ull);
+                    atomicAdd(&sums[s], v);
+                    break;
+                }
             }
             s = (s + 1) & cap_mask;
             if (--left == 0) { atomicOr(err, AGG_ERR_FULL); break; }
@@ -5413,7 +5505,8 @@ __global__ void k_q3_probe_slice(const uint32_t* __restrict__ keys,
                                  const uint32_t* __restrict__ order_bits,
                                  unsigned long long* __restrict__ slots,
                                  unsigned long long* __restrict__ sums, uint64_t cap_mask,
-                                 unsigned int* __restrict__ err) {
+                                 unsigned int* __restrict__ err,
+                                 unsigned long long* __restrict__ ngroups) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         unsigned long long k = keys[i];
@@ -5427,7 +5520,11 @@ __global__ void k_q3_probe_slice(const uint32_t* __restrict__ keys,
             if (cur == k) { atomicAdd(&sums[s], v); break; }
             if (cur == AGG_EMPTY) {
                 unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
-                if (old == AGG_EMPTY || old == k) { atomicAdd(&sums[s], v); break; }
+                if (old == AGG_EMPTY || old == k) {
+                    if (old == AGG_EMPTY && ngroups) atomicAdd(ngroups, 1ull);
+                    atomicAdd(&sums[s], v);
+                    break;
+                }
             }
             s = (s + 1) & cap_mask;
             if (--left == 0) { atomicOr(err, AGG_ERR_FULL); break; }
@@ -5564,7 +5661,7 @@ int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbu
                        (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
                        (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
                        (const uint32_t*)order_bits->ptr, ship_cutoff, at->slots, at->sums,
-                       at->counts, at->cap - 1, s->d_agg_err);
+                       at->counts, at->cap - 1, s->d_agg_err, at->n_groups);
     hipLaunchKernelGGL(k_hash_agg_emit, dim3(grid_for(at->cap)), dim3(BLOCK), 0, s->stream,
                        at->slots, at->sums, at->counts, at->cap, at->cursor, max_out,
                        (uint64_t*)out_keys->ptr, (int64_t*)out_sums->ptr, nullptr);
@@ -5598,7 +5695,7 @@ int gpue_q3_probe_accum(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbu
                        (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
                        (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
                        (const uint32_t*)order_bits->ptr, ship_cutoff, at->slots, at->sums,
-                       at->counts, at->cap - 1, s->d_agg_err);
+                       at->counts, at->cap - 1, s->d_agg_err, at->n_groups);
     HIP_CHECK(hipGetLastError());
     return GPUE_OK;
 }
@@ -5715,7 +5812,7 @@ int gpue_q3_probe_agg_part(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_
                            (const uint32_t*)keys_scratch->ptr,
                            (const long long*)vals_scratch->ptr, d_bpart, d_pstart, d_bpp,
                            d_fbp, (const uint32_t*)order_bits->ptr, at->slots, at->sums,
-                           at->cap - 1, s->d_agg_err);
+                           at->cap - 1, s->d_agg_err, at->n_groups);
         HIP_CHECK(hipStreamSynchronize(s->stream));
         (void)hipFree(d_bpart);
         (void)hipFree(d_bpp);
@@ -5767,7 +5864,7 @@ int gpue_q3_probe_agg(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf*
                        (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
                        (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
                        (const uint32_t*)order_bits->ptr, ship_cutoff, d_slots, d_sums,
-                       d_counts, cap - 1, s->d_agg_err);
+                       d_counts, cap - 1, s->d_agg_err, nullptr);
     hipLaunchKernelGGL(k_hash_agg_emit, dim3(grid_for(cap)), dim3(BLOCK), 0, s->stream,
                        d_slots, d_sums, d_counts, cap, d_cursor, max_out,
                        (uint64_t*)out_keys->ptr, (int64_t*)out_sums->ptr, nullptr);

@@ -135,6 +135,8 @@ def _declare(lib):
         "gpue_agg_table_create": (c_i32, [c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_agg_table_destroy": (None, [c_vp]),
         "gpue_agg_table_reset": (c_i32, [c_vp]),
+        "gpue_agg_table_size": (c_i32, [c_vp, c_vp, ctypes.POINTER(c_u64)]),
+        "gpue_agg_table_ensure": (c_i32, [c_vp, c_vp, c_u64]),
         "gpue_q3_probe_agg_part": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_u64, c_vp,
                                            c_i32, c_vp, c_vp, c_vp, c_u32, c_vp, c_vp,
                                            c_u64, ctypes.POINTER(c_u64)]),
@@ -597,6 +599,18 @@ class Engine:
 
     def agg_table_reset(self, h):
         _ck(self._lib, self._lib.gpue_agg_table_reset(h))
+
+    def agg_table_size(self, h) -> int:
+        """Claimed group count (device counter over the claim sites)."""
+        g = c_u64()
+        _ck(self._lib, self._lib.gpue_agg_table_size(self._h, h, ctypes.byref(g)))
+        return g.value
+
+    def agg_table_ensure(self, h, additional_rows):
+        """Grow + rehash so a push of additional_rows can never overflow —
+        the reference's try_convert_to_two_level_map check before each chunk
+        (aggregator.cpp:1237-1241)."""
+        _ck(self._lib, self._lib.gpue_agg_table_ensure(self._h, h, additional_rows))
 
     def agg_table_destroy(self, h):
         self._lib.gpue_agg_table_destroy(h)

@@ -301,6 +301,10 @@ class StreamingAggOperator(Operator):
         return self._out.pop(0)
 
     def _preagg(self, c):
+        # reference shape: try_convert_to_two_level_map runs before each
+        # chunk (aggregator.cpp:1237-1241) — here the table grows + rehashes
+        # so the push can never overflow (gpue_agg_table_ensure)
+        self._e.agg_table_ensure(self._at, c["n"])
         self._e.hash_agg_push(self._at, c["keys"], c["vals"], c["n"], cnts=c.get("cnts"))
         for b in (c["keys"], c["vals"]):
             b.free()

@@ -193,3 +193,50 @@ def test_pinned_ingest_parity_gpu(engine):
     a = dev.d2h(np.int32, n)
     b = streamed.d2h(np.int32, n)
     assert np.array_equal(a, b)
+
+
+def test_agg_table_grow_gpu(engine):
+    """Growable agg table (the reference's two-level-conversion property,
+    aggregator.cpp:1237-1241): start tiny, push far more distinct groups
+    than the initial capacity in chunks with ensure() between — results
+    must equal a one-shot aggregate and the table must have grown."""
+    at = engine.agg_table_create(16)
+    engine.agg_table_reset(at)
+    rng = np.random.default_rng(5)
+    n, chunks = 200_000, 8
+    keys = rng.integers(1, 50_000, n).astype(np.uint64)
+    vals = rng.integers(-1000, 1000, n).astype(np.int64)
+    per = n // chunks
+    for c in range(chunks):
+        sl = slice(c * per, (c + 1) * per)
+        kb, vb = engine.alloc(per * 8), engine.alloc(per * 8)
+        kb.h2d(keys[sl])
+        vb.h2d(vals[sl])
+        engine.agg_table_ensure(at, per)
+        engine.hash_agg_push(at, kb, vb, per)
+        kb.free()
+        vb.free()
+    distinct = len(np.unique(keys))
+    assert engine.agg_table_size(at) == distinct
+    ok_b = engine.alloc(distinct * 8)
+    os_b = engine.alloc(distinct * 8)
+    oc_b = engine.alloc(distinct * 8)
+    g = engine.hash_agg_emit(at, ok_b, os_b, distinct, oc_b)
+    assert g == distinct
+    got_k = ok_b.d2h(np.uint64, g)
+    got_s = os_b.d2h(np.int64, g)
+    got_c = oc_b.d2h(np.int64, g)
+    order = np.argsort(got_k)
+    import collections
+    sums = collections.defaultdict(int)
+    cnts = collections.defaultdict(int)
+    for k, v in zip(keys.tolist(), vals.tolist()):
+        sums[k] += v
+        cnts[k] += 1
+    ek = np.array(sorted(sums), np.uint64)
+    assert np.array_equal(got_k[order], ek)
+    assert np.array_equal(got_s[order], np.array([sums[k] for k in ek.tolist()]))
+    assert np.array_equal(got_c[order], np.array([cnts[k] for k in ek.tolist()]))
+    for b in (ok_b, os_b, oc_b):
+        b.free()
+    engine.agg_table_destroy(at)
