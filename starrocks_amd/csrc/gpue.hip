@@ -4893,6 +4893,7 @@ __global__ void k_hash_agg_push(const uint64_t* __restrict__ keys,
                                 unsigned long long* __restrict__ ngroups) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     unsigned long long local_hits = 0;
+    unsigned long long local_claims = 0;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         unsigned long long k = keys[i];
         if (agg_key_is_sentinel(k, err)) { if (miss_mask) miss_mask[i] = 1; continue; }
@@ -4923,7 +4924,7 @@ __global__ void k_hash_agg_push(const uint64_t* __restrict__ keys,
                         atomicAdd(&counts[slot], c);
                     }
                     if (old == k) local_hits++;
-                    else if (ngroups) atomicAdd(ngroups, 1ull);
+                    else local_claims++;
                     if (miss_mask) miss_mask[i] = (old != k);
                     break;
                 }
@@ -4943,6 +4944,10 @@ __global__ void k_hash_agg_push(const uint64_t* __restrict__ keys,
         if ((threadIdx.x & (WAVE - 1)) == 0 && local_hits)
             atomicAdd(hits_out, local_hits);
     }
+    for (int off = 32; off > 0; off >>= 1)
+        local_claims += __shfl_down(local_claims, off, WAVE);
+    if ((threadIdx.x & (WAVE - 1)) == 0 && local_claims && ngroups)
+        atomicAdd(ngroups, local_claims);
 }
 
 extern "C" {
@@ -5226,6 +5231,7 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
                                unsigned int* __restrict__ err,
                                unsigned long long* __restrict__ ngroups) {
     (void)counts;
+    unsigned long long local_claims = 0; // one atomic per wave at kernel end
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         // non-temporal loads on the four streamed columns: the decomp
@@ -5246,7 +5252,7 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
             if (cur == AGG_EMPTY) {
                 unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
                 if (old == AGG_EMPTY || old == k) {
-                    if (old == AGG_EMPTY && ngroups) atomicAdd(ngroups, 1ull);
+                    if (old == AGG_EMPTY) local_claims++;
                     atomicAdd(&sums[s], v);
                     break;
                 }
@@ -5255,6 +5261,12 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
             if (--left == 0) { atomicOr(err, AGG_ERR_FULL); break; }
         }
     }
+    // same-address claim-counter traffic wave-reduced (the round-1 cursor
+    // lesson: per-item same-line atomics serialize at the TCC)
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        local_claims += __shfl_down(local_claims, off, WAVE);
+    if ((threadIdx.x & (WAVE - 1)) == 0 && local_claims && ngroups)
+        atomicAdd(ngroups, local_claims);
 }
 
 // Decomposition instrument for q3's roofline attribution:
@@ -5470,6 +5482,7 @@ __global__ void k_q3_probe_slices_fused(const uint32_t* __restrict__ keys,
                                         unsigned long long* __restrict__ sums,
                                         uint64_t cap_mask, unsigned int* __restrict__ err,
                                         unsigned long long* __restrict__ ngroups) {
+    unsigned long long local_claims = 0;
     uint32_t p = block_part[blockIdx.x];
     uint64_t base = pstart[p];
     uint64_t cnt = pstart[p + 1] - base;
@@ -5489,7 +5502,7 @@ __global__ void k_q3_probe_slices_fused(const uint32_t* __restrict__ keys,
             if (cur_ == AGG_EMPTY) {
                 unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
                 if (old == AGG_EMPTY || old == k) {
-                    if (old == AGG_EMPTY && ngroups) atomicAdd(ngroups, 1ull);
+                    if (old == AGG_EMPTY) local_claims++;
                     atomicAdd(&sums[s], v);
                     break;
                 }
@@ -5498,6 +5511,10 @@ __global__ void k_q3_probe_slices_fused(const uint32_t* __restrict__ keys,
             if (--left == 0) { atomicOr(err, AGG_ERR_FULL); break; }
         }
     }
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        local_claims += __shfl_down(local_claims, off, WAVE);
+    if ((threadIdx.x & (WAVE - 1)) == 0 && local_claims && ngroups)
+        atomicAdd(ngroups, local_claims);
 }
 
 __global__ void k_q3_probe_slice(const uint32_t* __restrict__ keys,
@@ -5507,6 +5524,7 @@ __global__ void k_q3_probe_slice(const uint32_t* __restrict__ keys,
                                  unsigned long long* __restrict__ sums, uint64_t cap_mask,
                                  unsigned int* __restrict__ err,
                                  unsigned long long* __restrict__ ngroups) {
+    unsigned long long local_claims = 0;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         unsigned long long k = keys[i];
@@ -5521,7 +5539,7 @@ __global__ void k_q3_probe_slice(const uint32_t* __restrict__ keys,
             if (cur == AGG_EMPTY) {
                 unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
                 if (old == AGG_EMPTY || old == k) {
-                    if (old == AGG_EMPTY && ngroups) atomicAdd(ngroups, 1ull);
+                    if (old == AGG_EMPTY) local_claims++;
                     atomicAdd(&sums[s], v);
                     break;
                 }
@@ -5530,6 +5548,10 @@ __global__ void k_q3_probe_slice(const uint32_t* __restrict__ keys,
             if (--left == 0) { atomicOr(err, AGG_ERR_FULL); break; }
         }
     }
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        local_claims += __shfl_down(local_claims, off, WAVE);
+    if ((threadIdx.x & (WAVE - 1)) == 0 && local_claims && ngroups)
+        atomicAdd(ngroups, local_claims);
 }
 
 extern "C" {
