@@ -3590,9 +3590,13 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
         const char* nt = getenv("GPUE_Q21_PF_NT");
         bool use_nt = !nt || atoi(nt);
         int def_grid = mode == 2 ? 512 : 256; // global-group variant fits 2 blocks/CU
+        // GPUE_Q21_PF_TPB: threads per block (the kernel is blockDim-agnostic;
+        // 120 KB LDS pins it at 1 block/CU regardless, so TPB sets waves/CU:
+        // 1024 -> 16, 512 -> 8, 256 -> 4 — the q1 streaming geometry)
+        int tpb = env_cap("GPUE_Q21_PF_TPB", BLOCK_Q21);
         auto launch = [&](auto kern, int grid) {
             hipLaunchKernelGGL(kern, dim3(env_cap("GPUE_GRID_PF", grid)),
-                               dim3(BLOCK_Q21), 0, s->stream, (const int32_t*)pk->ptr,
+                               dim3(tpb), 0, s->stream, (const int32_t*)pk->ptr,
                                (const int32_t*)sk->ptr, (const int32_t*)od->ptr,
                                (const int32_t*)rv->ptr, n, parts->prefilter, parts->set_min,
                                (uint64_t)(parts->set_max - parts->set_min + 1),
