@@ -246,7 +246,11 @@ int gpue_join_probe_emit_mode_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf
                                   gpue_dbuf* out_build_idx, uint64_t* match_count);
 /* Nullable variants (NullableColumn is_nulls path,
  * join_hash_map_method.hpp:56-120): null build rows never enter a chain;
- * null probe rows match nothing (ANTI/OUTER emit them as unmatched). */
+ * null probe rows match nothing (ANTI/OUTER emit them as unmatched).
+ * mode 6 = NULL_AWARE_LEFT_ANTI (the NOT IN lowering, hash_joiner.cpp:97,
+ * join_hash_map.hpp:1225-1240): as LEFT_ANTI but null probe rows are
+ * EXCLUDED from the output — NULL NOT IN (...) is never true. Also accepted
+ * by the varchar nulls probe. */
 int gpue_join_build_bucket_chained_nulls_u32(gpue_session* s, gpue_dbuf* keys,
                                              gpue_dbuf* is_nulls /*u8, 1-based*/,
                                              uint64_t row_count, gpue_join_table** out);

@@ -2378,6 +2378,12 @@ __global__ void k_probe_count_vc(const uint8_t* __restrict__ pbytes,
                                  int mode, uint32_t* __restrict__ row_counts) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        if (mode == 6 && pnulls && pnulls[i]) {
+            // NULL_AWARE_LEFT_ANTI (NOT IN): null probe rows excluded
+            // (join_hash_map.hpp:1225-1240)
+            row_counts[i] = 0;
+            continue;
+        }
         uint32_t len = poffsets[i + 1] - poffsets[i];
         uint32_t b = crc_hash_32_dev(pbytes + poffsets[i], len, 0x811C9DC5u) & bucket_mask;
         uint32_t j = (pnulls && pnulls[i]) ? 0u : first[b];
@@ -2387,7 +2393,7 @@ __global__ void k_probe_count_vc(const uint8_t* __restrict__ pbytes,
                           pbytes + poffsets[i], len);
             j = next[j];
         }
-        row_counts[i] = join_mode_count(c, mode);
+        row_counts[i] = join_mode_count(c, mode == 6 ? 2 : mode);
     }
 }
 
@@ -2511,7 +2517,7 @@ int gpue_join_probe_emit_varchar_nulls(gpue_session* s, gpue_join_table* t, gpue
                                        uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
                                        gpue_dbuf* out_build_idx, uint64_t* match_count) {
     ARG_CHECK(s && t && pbytes && poffsets && match_count);
-    ARG_CHECK(mode >= 0 && mode <= 3);
+    ARG_CHECK((mode >= 0 && mode <= 3) || mode == 6);
     ARG_CHECK(t->kind == gpue_join_table::VARCHAR);
     ARG_CHECK(poffsets->bytes >= (n_rows + 1) * 4);
     uint32_t nb = grid_for(n_rows);
@@ -2542,7 +2548,8 @@ int gpue_join_probe_emit_varchar_nulls(gpue_session* s, gpue_join_table* t, gpue
                            (const uint8_t*)pbytes->ptr, (const uint32_t*)poffsets->ptr,
                            n_rows, (uint32_t)(t->bucket_size - 1), t->first, t->next,
                            t->key_bytes, t->key_offsets,
-                           pnulls ? (const uint8_t*)pnulls->ptr : nullptr, mode, d_counts,
+                           pnulls ? (const uint8_t*)pnulls->ptr : nullptr,
+                           mode == 6 ? 2 : mode, d_counts,
                            d_offsets, (uint32_t*)out_probe_idx->ptr,
                            (uint32_t*)out_build_idx->ptr);
         HIP_CHECK(hipStreamSynchronize(s->stream));
@@ -2733,8 +2740,15 @@ __global__ void k_probe_count_bc_nulls(const uint32_t* __restrict__ probe_keys,
                 c += (build_keys[b] == k);
                 b = next[b];
             }
+        } else if (mode == 6) {
+            // NULL_AWARE_LEFT_ANTI (the NOT IN lowering,
+            // join_hash_map.hpp:1225-1240): null probe rows are EXCLUDED —
+            // NULL NOT IN (...) is never true — where plain LEFT_ANTI emits
+            // them as unmatched
+            row_counts[i] = 0;
+            continue;
         }
-        row_counts[i] = join_mode_count(c, mode);
+        row_counts[i] = join_mode_count(c, mode == 6 ? 2 : mode);
     }
 }
 
@@ -2749,6 +2763,7 @@ __global__ void k_probe_emit_bc_nulls(const uint32_t* __restrict__ probe_keys,
                                       uint32_t* __restrict__ out_probe,
                                       uint32_t* __restrict__ out_build) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    if (mode == 6) mode = 2; // null-aware anti: null rows already have count 0
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         if (row_counts[i] == 0) continue;
         uint64_t pos = row_offsets[i];

@@ -1281,3 +1281,80 @@ def test_q1_accum_step_soak(engine):
     for b in cols + [kb, pb, acc]:
         b.free()
     t.destroy()
+
+
+def test_null_aware_left_anti_join_parity(engine):
+    """NULL_AWARE_LEFT_ANTI (the NOT IN lowering, mode 6): identical to
+    LEFT_ANTI except null probe rows are EXCLUDED from the output — NULL
+    NOT IN (...) is never true (join_hash_map.hpp:1225-1240; hash_joiner.cpp
+    :97 maps NOT-IN anti joins to this type). Compared against a direct
+    numpy statement of that rule and cross-checked against plain LEFT_ANTI."""
+    rng = np.random.default_rng(91)
+    nbuild, nprobe = 10_000, 120_000
+    build_keys = np.concatenate([[0], rng.integers(0, 1500, nbuild)]).astype(np.uint32)
+    build_nulls = np.concatenate([[0], (rng.random(nbuild) < 0.2)]).astype(np.uint8)
+    probe_keys = rng.integers(0, 2000, nprobe).astype(np.uint32)
+    probe_nulls = (rng.random(nprobe) < 0.25).astype(np.uint8)
+
+    kb = engine.alloc(build_keys.nbytes)
+    kb.h2d(build_keys.view(np.int32))
+    nb = engine.alloc(build_nulls.nbytes)
+    nb.h2d(build_nulls)
+    t = engine.join_build_bucket_chained_nulls(kb, nb, nbuild)
+    pb = engine.alloc(probe_keys.nbytes)
+    pb.h2d(probe_keys.view(np.int32))
+    pn = engine.alloc(probe_nulls.nbytes)
+    pn.h2d(probe_nulls)
+
+    # expected: non-null probe rows whose key is absent from the NON-NULL
+    # build rows
+    present = set(build_keys[1:][build_nulls[1:] == 0].tolist())
+    exp = np.flatnonzero((probe_nulls == 0) &
+                         ~np.isin(probe_keys, np.fromiter(present, np.uint32)))
+
+    cnt = engine.join_probe_emit_nulls(t, pb, pn, nprobe, 6)
+    assert cnt == len(exp)
+    op_b = engine.alloc(max(cnt, 1) * 4)
+    ob_b = engine.alloc(max(cnt, 1) * 4)
+    engine.join_probe_emit_nulls(t, pb, pn, nprobe, 6, op_b, ob_b)
+    gop = np.sort(op_b.d2h(np.uint32, cnt))
+    assert np.array_equal(gop, exp.astype(np.uint32))
+    assert (ob_b.d2h(np.uint32, cnt) == 0).all()  # anti: build idx 0 sentinel
+
+    # plain LEFT_ANTI (mode 2) additionally emits every null probe row
+    cnt2 = engine.join_probe_emit_nulls(t, pb, pn, nprobe, 2)
+    assert cnt2 == cnt + int(probe_nulls.sum())
+    for b in (kb, nb, pb, pn, op_b, ob_b):
+        b.free()
+    t.destroy()
+
+
+def test_null_aware_left_anti_varchar_parity(engine):
+    """mode 6 over Slice keys (the serialized-varchar constructor path)."""
+    rng = np.random.default_rng(92)
+    bb, boff, brows = _varchar_cols(rng, 3000, 40, one_based=True)
+    pb_, poff, prows = _varchar_cols(rng, 20_000, 55, one_based=False)
+    pnulls = (rng.random(20_000) < 0.3).astype(np.uint8)
+    b_bytes = engine.alloc(max(bb.nbytes, 1))
+    b_bytes.h2d(bb)
+    b_off = engine.alloc(boff.nbytes)
+    b_off.h2d(boff)
+    t = engine.join_build_varchar(b_bytes, b_off, 3000)
+    p_bytes = engine.alloc(max(pb_.nbytes, 1))
+    p_bytes.h2d(pb_)
+    p_off = engine.alloc(poff.nbytes)
+    p_off.h2d(poff)
+    p_n = engine.alloc(pnulls.nbytes)
+    p_n.h2d(pnulls)
+    present = set(brows[1:])
+    exp = np.array([i for i in range(20_000)
+                    if not pnulls[i] and prows[i] not in present], np.uint32)
+    cnt = engine.join_probe_emit_varchar_nulls(t, p_bytes, p_off, p_n, 20_000, 6)
+    assert cnt == len(exp)
+    op_b = engine.alloc(max(cnt, 1) * 4)
+    ob_b = engine.alloc(max(cnt, 1) * 4)
+    engine.join_probe_emit_varchar_nulls(t, p_bytes, p_off, p_n, 20_000, 6, op_b, ob_b)
+    assert np.array_equal(np.sort(op_b.d2h(np.uint32, cnt)), exp)
+    for b in (b_bytes, b_off, p_bytes, p_off, p_n, op_b, ob_b):
+        b.free()
+    t.destroy()
