@@ -3658,15 +3658,15 @@ int gpue_q21_star_agg(gpue_session* s, gpue_join_table* parts, gpue_join_table* 
     ARG_CHECK(parts->first16 && supps->first16 && dates->first16); // payloads < 65536
     unsigned long long* d_g = nullptr;
     HIP_CHECK(hipMalloc(&d_g, NG_Q21 * sizeof(unsigned long long)));
-    HIP_CHECK(hipMemsetAsync(d_g, 0, NG_Q21 * sizeof(unsigned long long), s->stream));
-    hipLaunchKernelGGL(k_q21_star_agg, dim3(env_cap("GPUE_GRID_WIDE", 512)), dim3(BLOCK_Q21), 0, s->stream,
-                       (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
-                       (const int32_t*)od->ptr, (const int32_t*)rv->ptr, n,
-                       parts->bitset, parts->set_min,
-                       (uint64_t)(parts->set_max - parts->set_min + 1), parts->first16,
-                       supps->bitset, supps->set_min,
-                       (uint64_t)(supps->set_max - supps->set_min + 1), dates->first16,
-                       dates->min_key, d_g);
+    // route through the same kernel dispatch as the bench path (LDS
+    // prefilter default, GPUE_Q21_PF/GPUE_Q21_GLOB variants) so parity
+    // tests exercise whatever kernel the bench actually runs
+    gpue_dbuf tmp{s, d_g, NG_Q21 * sizeof(unsigned long long), false};
+    int rc = gpue_q21_star_agg_async(s, parts, supps, dates, pk, sk, od, rv, n, &tmp);
+    if (rc != GPUE_OK) {
+        (void)hipFree(d_g);
+        return rc;
+    }
     HIP_CHECK(hipMemcpyAsync(group_sums_out, d_g, NG_Q21 * sizeof(int64_t),
                              hipMemcpyDeviceToHost, s->stream));
     HIP_CHECK(hipStreamSynchronize(s->stream));
