@@ -552,6 +552,13 @@ int gpue_ingest_push(gpue_ingest* g, const void* host, uint64_t bytes, gpue_dbuf
 int gpue_ingest_sync(gpue_ingest* g);
 void gpue_ingest_destroy(gpue_ingest* g);
 
+/* RLE page decode for int32 (storage ingress):
+ * storage/rowset/rle_page.h header + base/bit/rle_encoding.h's
+ * Parquet-style RLE/bit-pack hybrid at bit_width 32 (byte-aligned runs).
+ * Two-phase device decode: run-table scan + binary-search parallel fill. */
+int gpue_page_decode_rle_i32(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
+                             gpue_dbuf* out);
+
 /* ---- event timing on the session stream (bench roofline evidence) ---- */
 int gpue_timer_start(gpue_session* s);
 int gpue_timer_stop(gpue_session* s, float* ms_out);

@@ -1019,6 +1019,152 @@ void orc_partition_channel_u64(const uint64_t* keys, uint64_t n, uint32_t num_ch
 }
 
 /* ====================================================================== */
+/* RLE page codec for int32 (storage ingress row):     */
+/* storage/rowset/rle_page.h (header: 4-byte LE num_elements; body) over    */
+/* base/bit/rle_encoding.h's Parquet-style RLE/bit-pack hybrid (itself     */
+/* "based on Apache Doris/Impala" per the file header — a PUBLISHED        */
+/* algorithm). At bit_width = 32 (rle_page.h:82: SIZE_OF_TYPE*8 for        */
+/* non-bool types) every write is byte-aligned:                             */
+/*   repeated-run := varint(count<<1) + 4-byte LE value                     */
+/*   literal-run  := byte(num_groups<<1|1) + num_groups*8 LE u32 values     */
+/* Encoder run-breaking mirrors RleEncoder<T>::Put/FlushBufferedValues/     */
+/* Flush (rle_encoding.h:457-612): buffer 8 values; >=8 equal -> repeated   */
+/* run; literal indicator re-issued at 63 groups; final literal group       */
+/* zero-padded (the page header's count bounds the decode).                 */
+/* ====================================================================== */
+
+static inline uint64_t rle_put_varint(uint8_t* out, uint64_t pos, uint32_t v) {
+    while (v >= 0x80) {
+        out[pos++] = (uint8_t)(v | 0x80);
+        v >>= 7;
+    }
+    out[pos++] = (uint8_t)v;
+    return pos;
+}
+
+static inline uint64_t rle_get_varint(const uint8_t* in, uint64_t pos, uint32_t* v) {
+    uint32_t r = 0;
+    int shift = 0;
+    for (;;) {
+        uint8_t b = in[pos++];
+        r |= (uint32_t)(b & 0x7F) << shift;
+        if (!(b & 0x80)) break;
+        shift += 7;
+    }
+    *v = r;
+    return pos;
+}
+
+/* Encode n int32 values; out must hold 4 + n*4 + n/127 + 16 bytes.
+ * Returns bytes written (header included). */
+uint64_t orc_rle_page_encode_i32(const int32_t* values, uint32_t n, uint8_t* out) {
+    memcpy(out, &n, 4); /* RLE_PAGE_HEADER_SIZE num_elements (rle_page.h:57) */
+    uint64_t pos = 4;
+    uint32_t buffered[8];
+    int num_buffered = 0;
+    uint32_t literal_count = 0;
+    uint32_t repeat_count = 0;
+    int32_t current_value = 0;
+    int64_t lit_indicator = -1;
+
+    /* FlushLiteralRun (rle_encoding.h:490-516) */
+#define FLUSH_LITERAL(update)                                                   \
+    do {                                                                        \
+        if (lit_indicator < 0) { lit_indicator = (int64_t)pos; pos += 1; }      \
+        for (int b = 0; b < num_buffered; b++) {                                \
+            memcpy(out + pos, &buffered[b], 4);                                 \
+            pos += 4;                                                           \
+        }                                                                       \
+        num_buffered = 0;                                                       \
+        if (update) {                                                           \
+            out[lit_indicator] = (uint8_t)(((literal_count / 8) << 1) | 1);     \
+            lit_indicator = -1;                                                 \
+            literal_count = 0;                                                  \
+        }                                                                       \
+    } while (0)
+
+    /* FlushRepeatedRun (:518-537) */
+#define FLUSH_REPEATED()                                                        \
+    do {                                                                        \
+        pos = rle_put_varint(out, pos, (repeat_count << 1) | 0);                \
+        memcpy(out + pos, &current_value, 4);                                   \
+        pos += 4;                                                               \
+        num_buffered = 0;                                                       \
+        repeat_count = 0;                                                       \
+    } while (0)
+
+    for (uint32_t i = 0; i < n; i++) {
+        int32_t v = values[i];
+        if (repeat_count > 0 && v == current_value) {
+            repeat_count++;
+            if (repeat_count > 8) continue; /* long-run fast path (:463-468) */
+        } else {
+            if (repeat_count >= 8) FLUSH_REPEATED();
+            repeat_count = 1;
+            current_value = v;
+        }
+        buffered[num_buffered++] = (uint32_t)v;
+        if (num_buffered == 8) {
+            /* FlushBufferedValues (:540-570) */
+            if (repeat_count >= 8) {
+                num_buffered = 0;
+                if (literal_count != 0) FLUSH_LITERAL(1);
+            } else {
+                literal_count += 8;
+                if (literal_count / 8 + 1 >= (1 << 6)) FLUSH_LITERAL(1);
+                else FLUSH_LITERAL(0);
+                repeat_count = 0; /* :569 — trailing repeats joined the literal group */
+            }
+        }
+    }
+    /* Flush (:579-601) */
+    if (literal_count > 0 || repeat_count > 0 || num_buffered > 0) {
+        int all_repeat = literal_count == 0 &&
+                         (repeat_count == (uint32_t)num_buffered || num_buffered == 0);
+        if (repeat_count > 0 && all_repeat) {
+            FLUSH_REPEATED();
+        } else {
+            for (; num_buffered != 0 && num_buffered < 8; num_buffered++)
+                buffered[num_buffered] = 0;
+            literal_count += (uint32_t)num_buffered;
+            FLUSH_LITERAL(1);
+            repeat_count = 0;
+        }
+    }
+#undef FLUSH_LITERAL
+#undef FLUSH_REPEATED
+    return pos;
+}
+
+/* RleDecoder<T>::GetBatch (:426-452) at bit_width 32 */
+uint64_t orc_rle_page_decode_i32(const uint8_t* page, int32_t* values) {
+    uint32_t n;
+    memcpy(&n, page, 4);
+    uint64_t pos = 4;
+    uint32_t done = 0;
+    while (done < n) {
+        uint32_t indicator;
+        pos = rle_get_varint(page, pos, &indicator);
+        if (indicator & 1) { /* literal run of (indicator>>1)*8 values */
+            uint32_t cnt = (indicator >> 1) * 8;
+            for (uint32_t i = 0; i < cnt; i++) {
+                if (done < n) memcpy(&values[done], page + pos, 4);
+                done++;
+                pos += 4;
+            }
+            if (done > n) done = n; /* zero-padded tail group */
+        } else {
+            uint32_t cnt = indicator >> 1;
+            int32_t v;
+            memcpy(&v, page + pos, 4);
+            pos += 4;
+            for (uint32_t i = 0; i < cnt && done < n; i++) values[done++] = v;
+        }
+    }
+    return n;
+}
+
+/* ====================================================================== */
 /* XXH3-64 exchange hash, version 1 (exchange_sink_operator.cpp:604-610:    */
 /* `_exchange_hash_function_version == 1` uses Column::xxh3_hash =          */
 /* HashUtil::xx_hash3_64 = XXH3_64bits_withSeed, chained per key column     */

@@ -307,3 +307,8 @@ void orc_partition_channel_xxh3_u32(const uint32_t* keys, uint64_t n,
                                     uint32_t num_channels, uint32_t* channel_ids);
 /* set the OMP team size for the parameterless kernels (q3 legs) */
 void orc_set_threads(int n);
+
+/* RLE page codec for int32 (rle_page.h + base/bit/rle_encoding.h at
+ * bit_width 32; Parquet-style RLE/bit-pack hybrid, byte-aligned) */
+uint64_t orc_rle_page_encode_i32(const int32_t* values, uint32_t n, uint8_t* out);
+uint64_t orc_rle_page_decode_i32(const uint8_t* page, int32_t* values);

@@ -70,6 +70,10 @@ def _decl(lib):
     lib.orc_join_select_method.argtypes = [c_i32, c_i32, c_u64, c_i64, c_i64, c_i32,
                                            c_i32, c_i32, c_i32, c_u64, c_u64]
     lib.orc_set_threads.argtypes = [c_i32]
+    lib.orc_rle_page_encode_i32.restype = c_u64
+    lib.orc_rle_page_encode_i32.argtypes = [c_vp, u, c_vp]
+    lib.orc_rle_page_decode_i32.restype = c_u64
+    lib.orc_rle_page_decode_i32.argtypes = [c_vp, c_vp]
     lib.orc_xxh3_64_4to8.restype = c_u64
     lib.orc_xxh3_64_4to8.argtypes = [c_vp, c_i32, c_u64]
     lib.orc_xxh3_hash_i32.argtypes = [c_vp, c_u64, c_vp]
@@ -680,3 +684,17 @@ def partition_channels_xxh3(keys: np.ndarray, num_channels: int) -> np.ndarray:
     load().orc_partition_channel_xxh3_u32(_p(np.ascontiguousarray(keys, np.uint32)),
                                           len(keys), num_channels, _p(ch))
     return ch
+
+
+def rle_page_encode_i32(values: np.ndarray) -> np.ndarray:
+    """RLE page (rle_page.h header + rle_encoding.h hybrid at bit_width 32)."""
+    a = np.ascontiguousarray(values, np.int32)
+    out = np.zeros(4 + a.nbytes + len(a) // 8 + 64, np.uint8)
+    nb = load().orc_rle_page_encode_i32(_p(a), len(a), _p(out))
+    return out[:nb].copy()
+
+
+def rle_page_decode_i32(page: np.ndarray, n: int) -> np.ndarray:
+    v = np.zeros(n, np.int32)
+    load().orc_rle_page_decode_i32(_p(np.ascontiguousarray(page, np.uint8)), _p(v))
+    return v

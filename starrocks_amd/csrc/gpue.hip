@@ -6062,6 +6062,120 @@ __global__ void k_bshuf_untranspose_i32(const uint8_t* __restrict__ scratch, uin
     }
 }
 
+// ---------------------------------------------------------------------------
+// RLE page decode for int32 (storage ingress):
+// storage/rowset/rle_page.h (4-byte LE num_elements header) over
+// base/bit/rle_encoding.h's Parquet-style RLE/bit-pack hybrid at
+// bit_width = 32 (rle_page.h:82) — every run is byte-aligned:
+//   repeated := varint(count<<1) + 4-byte LE value
+//   literal  := byte(groups<<1|1) + groups*8 LE u32 values
+// Decode is two-phase: a single-wave scan walks the (compact) run headers
+// into a run table — one entry per RUN, not per value — then a grid-stride
+// fill kernel binary-searches the table per output tile. The oracle
+// restates the reference's own encoder byte-for-byte (hand KATs in
+// tests/test_page_decode.py), so GPU<->oracle parity pins the format.
+// ---------------------------------------------------------------------------
+__global__ void k_rle_scan_i32(const uint8_t* __restrict__ page, uint64_t page_bytes,
+                               uint32_t* __restrict__ run_start,
+                               uint64_t* __restrict__ run_info, // bit0: literal; >>1: byte off or value
+                               uint32_t* __restrict__ n_runs_out,
+                               uint32_t* __restrict__ err) {
+    if (blockIdx.x != 0 || threadIdx.x != 0) return;
+    uint32_t n;
+    memcpy(&n, page, 4);
+    uint64_t pos = 4;
+    uint32_t out = 0, nruns = 0;
+    while (out < n) {
+        if (pos >= page_bytes) { *err = 1; break; }
+        // varint indicator
+        uint32_t ind = 0;
+        int shift = 0;
+        for (;;) {
+            uint8_t b = page[pos++];
+            ind |= (uint32_t)(b & 0x7F) << shift;
+            if (!(b & 0x80)) break;
+            shift += 7;
+        }
+        run_start[nruns] = out;
+        if (ind & 1) {
+            uint32_t cnt = (ind >> 1) * 8;
+            run_info[nruns] = (pos << 1) | 1;
+            pos += (uint64_t)cnt * 4;
+            out += cnt; // may pad past n (zero-padded tail group)
+        } else {
+            uint32_t v;
+            memcpy(&v, page + pos, 4);
+            pos += 4;
+            run_info[nruns] = ((uint64_t)v << 1);
+            out += (ind >> 1);
+        }
+        nruns++;
+    }
+    run_start[nruns] = out < n ? n : out;
+    *n_runs_out = nruns;
+}
+
+__global__ void k_rle_fill_i32(const uint8_t* __restrict__ page,
+                               const uint32_t* __restrict__ run_start,
+                               const uint64_t* __restrict__ run_info, uint32_t n_runs,
+                               uint64_t n, int32_t* __restrict__ out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        // binary search: largest run with run_start <= i
+        uint32_t lo = 0, hi = n_runs - 1;
+        while (lo < hi) {
+            uint32_t mid = (lo + hi + 1) >> 1;
+            if ((uint64_t)run_start[mid] <= i) lo = mid;
+            else hi = mid - 1;
+        }
+        uint64_t info = run_info[lo];
+        if (info & 1) { // literal: byte-assemble the LE u32 (unaligned)
+            const uint8_t* p = page + (info >> 1) + (i - run_start[lo]) * 4;
+            out[i] = (int32_t)((uint32_t)p[0] | ((uint32_t)p[1] << 8) |
+                               ((uint32_t)p[2] << 16) | ((uint32_t)p[3] << 24));
+        } else {
+            out[i] = (int32_t)(uint32_t)(info >> 1);
+        }
+    }
+}
+
+extern "C" int gpue_page_decode_rle_i32(gpue_session* s, gpue_dbuf* page,
+                                        uint64_t n_values, gpue_dbuf* out);
+int gpue_page_decode_rle_i32(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
+                             gpue_dbuf* out) {
+    ARG_CHECK(s && page && out && out->bytes >= n_values * 4 && page->bytes >= 4);
+    uint64_t max_runs = page->bytes / 5 + 2; // a run is >= 5 encoded bytes
+    uint32_t* d_start = nullptr;
+    uint64_t* d_info = nullptr;
+    uint32_t* d_meta = nullptr; // [n_runs, err]
+    HIP_CHECK(hipMalloc(&d_start, (max_runs + 1) * 4));
+    HIP_CHECK(hipMalloc(&d_info, max_runs * 8));
+    HIP_CHECK(hipMalloc(&d_meta, 8));
+    HIP_CHECK(hipMemsetAsync(d_meta, 0, 8, s->stream));
+    hipLaunchKernelGGL(k_rle_scan_i32, dim3(1), dim3(64), 0, s->stream,
+                       (const uint8_t*)page->ptr, page->bytes, d_start, d_info, d_meta,
+                       d_meta + 1);
+    uint32_t meta[2] = {0, 0};
+    HIP_CHECK(hipMemcpyAsync(meta, d_meta, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    if (meta[1] || meta[0] == 0) {
+        (void)hipFree(d_start); (void)hipFree(d_info); (void)hipFree(d_meta);
+        if (meta[1]) {
+            snprintf(g_err, sizeof(g_err), "rle page truncated/corrupt");
+            return GPUE_ERR_ARG;
+        }
+        return n_values == 0 ? GPUE_OK : GPUE_ERR_ARG;
+    }
+    hipLaunchKernelGGL(k_rle_fill_i32, dim3(grid_for(n_values)), dim3(BLOCK), 0, s->stream,
+                       (const uint8_t*)page->ptr, d_start, d_info, meta[0], n_values,
+                       (int32_t*)out->ptr);
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_start);
+    (void)hipFree(d_info);
+    (void)hipFree(d_meta);
+    return GPUE_OK;
+}
+
 extern "C" int gpue_page_decode_bshuf_lz4_i32(gpue_session* s, gpue_dbuf* page,
                                               uint32_t n_values, gpue_dbuf* out);
 int gpue_page_decode_bshuf_lz4_i32(gpue_session* s, gpue_dbuf* page, uint32_t n_values,

@@ -179,6 +179,7 @@ def _declare(lib):
         "gpue_sbf_build_i32": (c_i32, [c_vp, c_vp, c_u64, c_i32, c_vp]),
         "gpue_sbf_test_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_i32, c_vp]),
         "gpue_topk_i64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp]),
+        "gpue_page_decode_rle_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp]),
         "gpue_page_decode_bshuf_lz4_i32": (c_i32, [c_vp, c_vp, c_u32, c_vp]),
         "gpue_timer_start": (c_i32, [c_vp]),
         "gpue_timer_stop": (c_i32, [c_vp, ctypes.POINTER(ctypes.c_float)]),
@@ -774,6 +775,11 @@ class Engine:
         _ck(self._lib, self._lib.gpue_partition_2xi32(
             self._h, a._h, b._h, n, num_channels, sp.ctypes.data_as(c_vp), row_indexes._h))
         return sp
+
+    def page_decode_rle_i32(self, page: DBuf, n_values, out: DBuf):
+        """RLE page decode (rle_page.h + rle_encoding.h at bit_width 32)."""
+        _ck(self._lib, self._lib.gpue_page_decode_rle_i32(self._h, page._h,
+                                                          n_values, out._h))
 
     def page_decode_bshuf_lz4_i32(self, page: DBuf, n_values, out: DBuf):
         _ck(self._lib, self._lib.gpue_page_decode_bshuf_lz4_i32(self._h, page._h,
