@@ -6099,10 +6099,12 @@ __global__ void k_rle_scan_i32(const uint8_t* __restrict__ page, uint64_t page_b
         run_start[nruns] = out;
         if (ind & 1) {
             uint32_t cnt = (ind >> 1) * 8;
+            if (pos + (uint64_t)cnt * 4 > page_bytes) { *err = 1; break; }
             run_info[nruns] = (pos << 1) | 1;
             pos += (uint64_t)cnt * 4;
             out += cnt; // may pad past n (zero-padded tail group)
         } else {
+            if (pos + 4 > page_bytes) { *err = 1; break; }
             uint32_t v;
             memcpy(&v, page + pos, 4);
             pos += 4;
