@@ -180,6 +180,8 @@ def _declare(lib):
         "gpue_sbf_test_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_i32, c_vp]),
         "gpue_topk_i64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp]),
         "gpue_page_decode_rle_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp]),
+        "gpue_ubench_bitgather": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_u64, c_i32,
+                                          ctypes.POINTER(ctypes.c_float)]),
         "gpue_page_decode_bshuf_lz4_i32": (c_i32, [c_vp, c_vp, c_u32, c_vp]),
         "gpue_timer_start": (c_i32, [c_vp]),
         "gpue_timer_stop": (c_i32, [c_vp, ctypes.POINTER(ctypes.c_float)]),
@@ -775,6 +777,12 @@ class Engine:
         _ck(self._lib, self._lib.gpue_partition_2xi32(
             self._h, a._h, b._h, n, num_channels, sp.ctypes.data_as(c_vp), row_indexes._h))
         return sp
+
+    def ubench_bitgather(self, idx: DBuf, n, bits: DBuf, nbits_pow2, reps=5) -> float:
+        ms = ctypes.c_float()
+        _ck(self._lib, self._lib.gpue_ubench_bitgather(
+            self._h, idx._h, n, bits._h, nbits_pow2, reps, ctypes.byref(ms)))
+        return ms.value
 
     def page_decode_rle_i32(self, page: DBuf, n_values, out: DBuf):
         """RLE page decode (rle_page.h + rle_encoding.h at bit_width 32)."""
