@@ -558,6 +558,9 @@ void gpue_ingest_destroy(gpue_ingest* g);
  * Two-phase device decode: run-table scan + binary-search parallel fill. */
 int gpue_page_decode_rle_i32(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
                              gpue_dbuf* out);
+/* BOOL variant (bit_width 1, bit-packed literal groups; u8 output) */
+int gpue_page_decode_rle_bool(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
+                              gpue_dbuf* out);
 
 /* ---- event timing on the session stream (bench roofline evidence) ---- */
 int gpue_timer_start(gpue_session* s);

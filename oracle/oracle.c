@@ -1136,6 +1136,113 @@ uint64_t orc_rle_page_encode_i32(const int32_t* values, uint32_t n, uint8_t* out
     return pos;
 }
 
+/* BOOL variant (bit_width = 1, rle_page.h:82): repeated run value padded to
+ * ONE byte (PutAligned(v, Ceil(1,8))); literal groups are bit-packed
+ * LSB-first, one byte per 8 values (BitWriter::PutValue). Same run-breaking
+ * as the int32 encoder. values/decoded are 0/1 uint8 (the reference decodes
+ * bool columns into u8). */
+uint64_t orc_rle_page_encode_bool(const uint8_t* values, uint32_t n, uint8_t* out) {
+    memcpy(out, &n, 4);
+    uint64_t pos = 4;
+    uint8_t buffered[8];
+    int num_buffered = 0;
+    uint32_t literal_count = 0;
+    uint32_t repeat_count = 0;
+    uint8_t current_value = 0;
+    int64_t lit_indicator = -1;
+
+#define FLUSH_LITERAL_B(update)                                                 \
+    do {                                                                        \
+        if (lit_indicator < 0) { lit_indicator = (int64_t)pos; pos += 1; }      \
+        if (num_buffered > 0) {                                                 \
+            uint8_t byte = 0;                                                   \
+            for (int b = 0; b < num_buffered; b++) byte |= (buffered[b] & 1) << b; \
+            out[pos++] = byte;                                                  \
+        }                                                                       \
+        num_buffered = 0;                                                       \
+        if (update) {                                                           \
+            out[lit_indicator] = (uint8_t)(((literal_count / 8) << 1) | 1);     \
+            lit_indicator = -1;                                                 \
+            literal_count = 0;                                                  \
+        }                                                                       \
+    } while (0)
+
+#define FLUSH_REPEATED_B()                                                      \
+    do {                                                                        \
+        pos = rle_put_varint(out, pos, (repeat_count << 1) | 0);                \
+        out[pos++] = current_value;                                             \
+        num_buffered = 0;                                                       \
+        repeat_count = 0;                                                       \
+    } while (0)
+
+    for (uint32_t i = 0; i < n; i++) {
+        uint8_t v = values[i] & 1;
+        if (repeat_count > 0 && v == current_value) {
+            repeat_count++;
+            if (repeat_count > 8) continue;
+        } else {
+            if (repeat_count >= 8) FLUSH_REPEATED_B();
+            repeat_count = 1;
+            current_value = v;
+        }
+        buffered[num_buffered++] = v;
+        if (num_buffered == 8) {
+            if (repeat_count >= 8) {
+                num_buffered = 0;
+                if (literal_count != 0) FLUSH_LITERAL_B(1);
+            } else {
+                literal_count += 8;
+                if (literal_count / 8 + 1 >= (1 << 6)) FLUSH_LITERAL_B(1);
+                else FLUSH_LITERAL_B(0);
+                repeat_count = 0;
+            }
+        }
+    }
+    if (literal_count > 0 || repeat_count > 0 || num_buffered > 0) {
+        int all_repeat = literal_count == 0 &&
+                         (repeat_count == (uint32_t)num_buffered || num_buffered == 0);
+        if (repeat_count > 0 && all_repeat) {
+            FLUSH_REPEATED_B();
+        } else {
+            for (; num_buffered != 0 && num_buffered < 8; num_buffered++)
+                buffered[num_buffered] = 0;
+            literal_count += (uint32_t)num_buffered;
+            FLUSH_LITERAL_B(1);
+            repeat_count = 0;
+        }
+    }
+#undef FLUSH_LITERAL_B
+#undef FLUSH_REPEATED_B
+    return pos;
+}
+
+uint64_t orc_rle_page_decode_bool(const uint8_t* page, uint8_t* values) {
+    uint32_t n;
+    memcpy(&n, page, 4);
+    uint64_t pos = 4;
+    uint32_t done = 0;
+    while (done < n) {
+        uint32_t indicator;
+        pos = rle_get_varint(page, pos, &indicator);
+        if (indicator & 1) {
+            uint32_t groups = indicator >> 1;
+            for (uint32_t g = 0; g < groups; g++) {
+                uint8_t byte = page[pos++];
+                for (int b = 0; b < 8; b++) {
+                    if (done < n) values[done] = (byte >> b) & 1;
+                    done++;
+                }
+            }
+            if (done > n) done = n;
+        } else {
+            uint32_t cnt = indicator >> 1;
+            uint8_t v = page[pos++] & 1;
+            for (uint32_t i = 0; i < cnt && done < n; i++) values[done++] = v;
+        }
+    }
+    return n;
+}
+
 /* RleDecoder<T>::GetBatch (:426-452) at bit_width 32 */
 uint64_t orc_rle_page_decode_i32(const uint8_t* page, int32_t* values) {
     uint32_t n;

@@ -312,3 +312,5 @@ void orc_set_threads(int n);
  * bit_width 32; Parquet-style RLE/bit-pack hybrid, byte-aligned) */
 uint64_t orc_rle_page_encode_i32(const int32_t* values, uint32_t n, uint8_t* out);
 uint64_t orc_rle_page_decode_i32(const uint8_t* page, int32_t* values);
+uint64_t orc_rle_page_encode_bool(const uint8_t* values, uint32_t n, uint8_t* out);
+uint64_t orc_rle_page_decode_bool(const uint8_t* page, uint8_t* values);

@@ -74,6 +74,10 @@ def _decl(lib):
     lib.orc_rle_page_encode_i32.argtypes = [c_vp, u, c_vp]
     lib.orc_rle_page_decode_i32.restype = c_u64
     lib.orc_rle_page_decode_i32.argtypes = [c_vp, c_vp]
+    lib.orc_rle_page_encode_bool.restype = c_u64
+    lib.orc_rle_page_encode_bool.argtypes = [c_vp, u, c_vp]
+    lib.orc_rle_page_decode_bool.restype = c_u64
+    lib.orc_rle_page_decode_bool.argtypes = [c_vp, c_vp]
     lib.orc_xxh3_64_4to8.restype = c_u64
     lib.orc_xxh3_64_4to8.argtypes = [c_vp, c_i32, c_u64]
     lib.orc_xxh3_hash_i32.argtypes = [c_vp, c_u64, c_vp]
@@ -697,4 +701,17 @@ def rle_page_encode_i32(values: np.ndarray) -> np.ndarray:
 def rle_page_decode_i32(page: np.ndarray, n: int) -> np.ndarray:
     v = np.zeros(n, np.int32)
     load().orc_rle_page_decode_i32(_p(np.ascontiguousarray(page, np.uint8)), _p(v))
+    return v
+
+
+def rle_page_encode_bool(values: np.ndarray) -> np.ndarray:
+    a = np.ascontiguousarray(values, np.uint8)
+    out = np.zeros(4 + len(a) + len(a) // 8 + 64, np.uint8)
+    nb = load().orc_rle_page_encode_bool(_p(a), len(a), _p(out))
+    return out[:nb].copy()
+
+
+def rle_page_decode_bool(page: np.ndarray, n: int) -> np.ndarray:
+    v = np.zeros(n, np.uint8)
+    load().orc_rle_page_decode_bool(_p(np.ascontiguousarray(page, np.uint8)), _p(v))
     return v

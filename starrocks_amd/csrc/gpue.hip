@@ -6194,6 +6194,107 @@ __global__ void k_rle_fill_i32(const uint8_t* __restrict__ page,
     }
 }
 
+// BOOL variant (bit_width = 1): repeated value is one byte; literal groups
+// are bit-packed LSB-first, one byte per 8 values. Output u8 per value
+// (the reference decodes bool columns to u8).
+__global__ void k_rle_scan_bool(const uint8_t* __restrict__ page, uint64_t page_bytes,
+                                uint32_t* __restrict__ run_start,
+                                uint64_t* __restrict__ run_info,
+                                uint32_t* __restrict__ n_runs_out,
+                                uint32_t* __restrict__ err) {
+    if (blockIdx.x != 0 || threadIdx.x != 0) return;
+    uint32_t n;
+    memcpy(&n, page, 4);
+    uint64_t pos = 4;
+    uint32_t out = 0, nruns = 0;
+    while (out < n) {
+        if (pos >= page_bytes) { *err = 1; break; }
+        uint32_t ind = 0;
+        int shift = 0;
+        for (;;) {
+            uint8_t b = page[pos++];
+            ind |= (uint32_t)(b & 0x7F) << shift;
+            if (!(b & 0x80)) break;
+            shift += 7;
+        }
+        run_start[nruns] = out;
+        if (ind & 1) {
+            uint32_t groups = ind >> 1;
+            if (pos + groups > page_bytes) { *err = 1; break; }
+            run_info[nruns] = (pos << 1) | 1;
+            pos += groups;
+            out += groups * 8;
+        } else {
+            if (pos + 1 > page_bytes) { *err = 1; break; }
+            run_info[nruns] = ((uint64_t)(page[pos] & 1) << 1);
+            pos += 1;
+            out += (ind >> 1);
+        }
+        nruns++;
+    }
+    run_start[nruns] = out < n ? n : out;
+    *n_runs_out = nruns;
+}
+
+__global__ void k_rle_fill_bool(const uint8_t* __restrict__ page,
+                                const uint32_t* __restrict__ run_start,
+                                const uint64_t* __restrict__ run_info, uint32_t n_runs,
+                                uint64_t n, uint8_t* __restrict__ out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint32_t lo = 0, hi = n_runs - 1;
+        while (lo < hi) {
+            uint32_t mid = (lo + hi + 1) >> 1;
+            if ((uint64_t)run_start[mid] <= i) lo = mid;
+            else hi = mid - 1;
+        }
+        uint64_t info = run_info[lo];
+        if (info & 1) {
+            uint64_t d = i - run_start[lo];
+            out[i] = (page[(info >> 1) + (d >> 3)] >> (d & 7)) & 1;
+        } else {
+            out[i] = (uint8_t)(info >> 1);
+        }
+    }
+}
+
+extern "C" int gpue_page_decode_rle_bool(gpue_session* s, gpue_dbuf* page,
+                                         uint64_t n_values, gpue_dbuf* out);
+int gpue_page_decode_rle_bool(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
+                              gpue_dbuf* out) {
+    ARG_CHECK(s && page && out && out->bytes >= n_values && page->bytes >= 4);
+    uint64_t max_runs = page->bytes + 2; // a bool run is >= 2 encoded bytes
+    uint32_t* d_start = nullptr;
+    uint64_t* d_info = nullptr;
+    uint32_t* d_meta = nullptr;
+    HIP_CHECK(hipMalloc(&d_start, (max_runs + 1) * 4));
+    HIP_CHECK(hipMalloc(&d_info, max_runs * 8));
+    HIP_CHECK(hipMalloc(&d_meta, 8));
+    HIP_CHECK(hipMemsetAsync(d_meta, 0, 8, s->stream));
+    hipLaunchKernelGGL(k_rle_scan_bool, dim3(1), dim3(64), 0, s->stream,
+                       (const uint8_t*)page->ptr, page->bytes, d_start, d_info, d_meta,
+                       d_meta + 1);
+    uint32_t meta[2] = {0, 0};
+    HIP_CHECK(hipMemcpyAsync(meta, d_meta, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    if (meta[1] || meta[0] == 0) {
+        (void)hipFree(d_start); (void)hipFree(d_info); (void)hipFree(d_meta);
+        if (meta[1]) {
+            snprintf(g_err, sizeof(g_err), "rle bool page truncated/corrupt");
+            return GPUE_ERR_ARG;
+        }
+        return n_values == 0 ? GPUE_OK : GPUE_ERR_ARG;
+    }
+    hipLaunchKernelGGL(k_rle_fill_bool, dim3(grid_for(n_values)), dim3(BLOCK), 0, s->stream,
+                       (const uint8_t*)page->ptr, d_start, d_info, meta[0], n_values,
+                       (uint8_t*)out->ptr);
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_start);
+    (void)hipFree(d_info);
+    (void)hipFree(d_meta);
+    return GPUE_OK;
+}
+
 extern "C" int gpue_page_decode_rle_i32(gpue_session* s, gpue_dbuf* page,
                                         uint64_t n_values, gpue_dbuf* out);
 int gpue_page_decode_rle_i32(gpue_session* s, gpue_dbuf* page, uint64_t n_values,

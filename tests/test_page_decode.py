@@ -110,3 +110,41 @@ def test_rle_page_decode_gpu_parity(engine):
         engine.page_decode_rle_i32(pb, 1000, ob)
     pb.free()
     ob.free()
+
+
+def test_rle_bool_page_oracle_roundtrip():
+    """BOOL RLE (bit_width 1, rle_page.h:82): bit-packed literal groups
+    LSB-first, one-byte repeated values. Hand KAT: 100x true -> header +
+    varint(200) + 0x01."""
+    rng = np.random.default_rng(4)
+    pats = [np.ones(100, np.uint8), np.zeros(77, np.uint8),
+            (rng.random(1000) < 0.5).astype(np.uint8),
+            np.repeat((rng.random(60) < 0.5).astype(np.uint8),
+                      rng.integers(1, 50, 60)),
+            np.array([1], np.uint8), np.array([1, 0, 1], np.uint8)]
+    for pat in pats:
+        page = orc.rle_page_encode_bool(pat)
+        assert np.array_equal(orc.rle_page_decode_bool(page, len(pat)), pat)
+    p = orc.rle_page_encode_bool(np.ones(100, np.uint8))
+    assert p.tobytes() == bytes([100, 0, 0, 0, 0xC8, 0x01, 0x01])
+    # 8 literals 1,0,1,1,0,0,1,0 -> indicator 0x03 + LSB-first byte 0x4D
+    p = orc.rle_page_encode_bool(np.array([1, 0, 1, 1, 0, 0, 1, 0], np.uint8))
+    assert p.tobytes() == bytes([8, 0, 0, 0, 0x03, 0x4D])
+
+
+@pytest.mark.gpu
+def test_rle_bool_page_decode_gpu_parity(engine):
+    rng = np.random.default_rng(9)
+    pats = [np.repeat((rng.random(500) < 0.5).astype(np.uint8),
+                      rng.integers(1, 400, 500)),
+            (rng.random(200_000) < 0.3).astype(np.uint8),
+            np.ones(1_000_000, np.uint8)]
+    for pat in pats:
+        page = orc.rle_page_encode_bool(pat)
+        pb = engine.alloc(max(page.nbytes, 4))
+        pb.h2d(page)
+        ob = engine.alloc(max(len(pat), 1))
+        engine.page_decode_rle_bool(pb, len(pat), ob)
+        assert np.array_equal(ob.d2h(np.uint8, len(pat)), pat)
+        pb.free()
+        ob.free()
