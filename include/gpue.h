@@ -160,6 +160,14 @@ int gpue_join_probe_emit_mode_u64(gpue_session* s, gpue_join_table* t, gpue_dbuf
  * is nondeterministic — the emitted match multiset is identical). */
 int gpue_join_build_range_direct_i32(gpue_session* s, gpue_dbuf* keys /*i32, 1-based row 0 sentinel*/,
                                      uint64_t row_count, gpue_join_table** out);
+/* DENSE_RANGE_DIRECT_MAPPING (join_hash_map_method.h:378, .hpp:781-940):
+ * rank/select-compressed direct map — per 32-key group a {start_index,
+ * bitset} pair (2 bits/interval position amortized), first[] sized by the
+ * PRESENT keys; probe = bit test + popcount + one first[] load; chains hold
+ * identical keys (no compare). The selector's choice when the interval
+ * exceeds bucket/L2 but 2b/pos + 4 B/row beats bucket-chained. */
+int gpue_join_build_dense_range_direct_i32(gpue_session* s, gpue_dbuf* keys,
+                                           uint64_t row_count, gpue_join_table** out);
 
 /* ---- JoinHashMapSelector (join_hash_table.cpp:164-344) --------------------
  * The reference's automatic key-constructor + map-method decision, restated

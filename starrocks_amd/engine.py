@@ -86,6 +86,8 @@ def _declare(lib):
         "gpue_scan_filter_i64_lt_sp": (c_i32, [c_vp, c_vp, c_u64, c_i64, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_join_build_payload_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_build_range_direct_i32": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
+        "gpue_join_build_dense_range_direct_i32": (c_i32, [c_vp, c_vp, c_u64,
+                                                           ctypes.POINTER(c_vp)]),
         "gpue_join_build_bucket_chained_u32": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_build_linear_chained_u32": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_build_bucket_chained_u64": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
@@ -395,6 +397,14 @@ class Engine:
         h = c_vp()
         _ck(self._lib, self._lib.gpue_join_build_range_direct_i32(self._h, keys._h, row_count,
                                                                   ctypes.byref(h)))
+        return JoinTable(self, h)
+
+    def join_build_dense_range_direct(self, keys: DBuf, row_count) -> JoinTable:
+        """DENSE_RANGE_DIRECT_MAPPING (rank/select-compressed direct map,
+        join_hash_map_method.h:378)."""
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_join_build_dense_range_direct_i32(
+            self._h, keys._h, row_count, ctypes.byref(h)))
         return JoinTable(self, h)
 
     def join_build_bucket_chained(self, keys: DBuf, row_count) -> JoinTable:

@@ -1358,3 +1358,64 @@ def test_null_aware_left_anti_varchar_parity(engine):
     for b in (b_bytes, b_off, p_bytes, p_off, p_n, op_b, ob_b):
         b.free()
     t.destroy()
+
+
+def test_dense_range_direct_join_parity(engine):
+    """DENSE_RANGE_DIRECT (rank/select 2-bit map, join_hash_map_method.h:378,
+    .hpp:781-940): sparse-wide build keys with duplicates; every probe mode
+    must produce the same multiset as the bucket-chained build of the same
+    rows, and the selector's DENSE decision (10 M rows / 100 M interval in
+    the CPU-threshold scenario) is what routes here via join_build_auto."""
+    rng = np.random.default_rng(77)
+    nbuild, nprobe = 120_000, 400_000
+    # interval ~3M >> nbuild: the dense regime; ~25% duplicate keys
+    base = rng.integers(1, 3_000_000, nbuild - nbuild // 4)
+    keys = np.concatenate([[0], base, rng.choice(base, nbuild // 4)]).astype(np.int32)
+    kb = engine.alloc(keys.nbytes)
+    kb.h2d(keys)
+    td = engine.join_build_dense_range_direct(kb, nbuild)
+    tb = engine.join_build_bucket_chained(kb, nbuild)
+    probe = rng.integers(-10, 3_100_000, nprobe).astype(np.int32)
+    pb = engine.alloc(probe.nbytes)
+    pb.h2d(probe)
+    for mode in (0, 1, 2, 3):
+        cd = engine.join_probe_emit_mode(td, pb, nprobe, mode)
+        cb = engine.join_probe_emit_mode(tb, pb, nprobe, mode)
+        assert cd == cb, mode
+        opd, obd = engine.alloc(max(cd, 1) * 4), engine.alloc(max(cd, 1) * 4)
+        opb, obb = engine.alloc(max(cb, 1) * 4), engine.alloc(max(cb, 1) * 4)
+        engine.join_probe_emit_mode(td, pb, nprobe, mode, opd, obd)
+        engine.join_probe_emit_mode(tb, pb, nprobe, mode, opb, obb)
+        pack = lambda a, b: np.sort(a.astype(np.uint64) << np.uint64(32) | b)
+        assert np.array_equal(
+            pack(opd.d2h(np.uint32, cd), obd.d2h(np.uint32, cd)),
+            pack(opb.d2h(np.uint32, cb), obb.d2h(np.uint32, cb))), mode
+        for x in (opd, obd, opb, obb):
+            x.free()
+    td.destroy()
+    tb.destroy()
+    kb.free()
+    pb.free()
+
+
+def test_join_build_auto_dense_route(engine):
+    """join_build_auto must route a DENSE decision to the dense build and the
+    table must answer probes (CPU-cache thresholds force the DENSE branch)."""
+    rng = np.random.default_rng(78)
+    n = 400_000
+    # interval 2.5M: > bucket 524k, > the 64 KB l2 below, and the dense
+    # inequality interval/4 + rc*4 <= (b+b/10)*4 holds (2.23M <= 2.31M)
+    keys = np.concatenate([[0], rng.integers(1, 2_500_000, n)]).astype(np.int32)
+    kb = engine.alloc(keys.nbytes)
+    kb.h2d(keys)
+    # CPU-threshold scenario forcing the DENSE branch
+    t, method = engine.join_build_auto(kb, n, l2_size=64 << 10, l3_size=32 << 20)
+    assert engine.JM_NAMES[method] == "DENSE_RANGE_DIRECT"
+    probe = keys[1:5001]
+    pb = engine.alloc(probe.nbytes)
+    pb.h2d(probe)
+    cnt = engine.join_probe_emit_mode(t, pb, 5000, 1)  # LEFT_SEMI
+    assert cnt == 5000
+    kb.free()
+    pb.free()
+    t.destroy()
