@@ -1386,10 +1386,18 @@ def test_dense_range_direct_join_parity(engine):
         opb, obb = engine.alloc(max(cb, 1) * 4), engine.alloc(max(cb, 1) * 4)
         engine.join_probe_emit_mode(td, pb, nprobe, mode, opd, obd)
         engine.join_probe_emit_mode(tb, pb, nprobe, mode, opb, obb)
-        pack = lambda a, b: np.sort(a.astype(np.uint64) << np.uint64(32) | b)
-        assert np.array_equal(
-            pack(opd.d2h(np.uint32, cd), obd.d2h(np.uint32, cd)),
-            pack(opb.d2h(np.uint32, cb), obb.d2h(np.uint32, cb))), mode
+        gop, gob = opd.d2h(np.uint32, cd), obd.d2h(np.uint32, cd)
+        eop, eob = opb.d2h(np.uint32, cb), obb.d2h(np.uint32, cb)
+        if mode == 1:
+            # SEMI emits ONE representative build row per matched probe row;
+            # chain order differs between structures, so compare the probe
+            # set + the key-match property (as the reference's own semi
+            # semantics define it)
+            assert np.array_equal(np.sort(gop), np.sort(eop))
+            assert (keys[gob] == probe[gop]).all()
+        else:
+            pack = lambda a, b: np.sort(a.astype(np.uint64) << np.uint64(32) | b)
+            assert np.array_equal(pack(gop, gob), pack(eop, eob)), mode
         for x in (opd, obd, opb, obb):
             x.free()
     td.destroy()
