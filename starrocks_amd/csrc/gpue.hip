@@ -4463,6 +4463,67 @@ k_ub_sum4(const int4* __restrict__ a, const int4* __restrict__ b,
     if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(out, (unsigned long long)sum);
 }
 
+// PF-kernel decomposition legs (DESIGN.md §4d residual): same 120 KB LDS /
+// 1024-thread / 1-block-per-CU geometry as k_q21_star_agg_pf, with phase 2
+// removed. which=2: the 4 NT streams only; which=3: + the per-key LDS
+// prefilter test (maybe-count accumulated to prevent DCE).
+template <bool LDS_TEST>
+__global__ __launch_bounds__(BLOCK_Q21) void
+k_ub_q21_pf_legs(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
+                 const int32_t* __restrict__ od, const int32_t* __restrict__ rv,
+                 uint64_t n, const uint32_t* __restrict__ prefilter, int64_t psmin,
+                 uint64_t psint, unsigned long long* __restrict__ sink) {
+    __shared__ uint32_t pf[PF_WORDS];
+    __shared__ unsigned long long g[NG_Q21];
+    for (uint32_t w = threadIdx.x; w < PF_WORDS; w += blockDim.x)
+        pf[w] = prefilter ? prefilter[w] : 0u;
+    for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x) g[j] = 0;
+    __syncthreads();
+    const uint64_t n4 = n / 4;
+    const int4* __restrict__ pk4 = (const int4*)pk;
+    const int4* __restrict__ sk4 = (const int4*)sk;
+    const int4* __restrict__ od4 = (const int4*)od;
+    const int4* __restrict__ rv4 = (const int4*)rv;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    unsigned long long acc = 0;
+    auto ld4 = [&](const int4* p, uint64_t i) {
+        const uint64_t* q = (const uint64_t*)(p + i);
+        uint64_t lo = __builtin_nontemporal_load(q);
+        uint64_t hi = __builtin_nontemporal_load(q + 1);
+        int4 v;
+        v.x = (int32_t)lo; v.y = (int32_t)(lo >> 32);
+        v.z = (int32_t)hi; v.w = (int32_t)(hi >> 32);
+        return v;
+    };
+    auto quad = [&](int4 p4, int4 s4, int4 o4, int4 r4) {
+        if (LDS_TEST) {
+            #pragma unroll
+            for (int j = 0; j < 4; j++) {
+                uint32_t idx = (uint32_t)((&p4.x)[j] - psmin);
+                bool in = idx < psint;
+                uint32_t fidx = (in ? idx : 0u) & PF_MASK;
+                acc += in & (pf[fidx >> 5] >> (fidx & 31)) & 1u;
+            }
+            acc += (unsigned long long)(uint32_t)(s4.x + o4.x + r4.x);
+        } else {
+            acc += (unsigned long long)(uint32_t)(p4.x + s4.x + o4.x + r4.x);
+        }
+    };
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i + stride < n4; i += 2 * stride) {
+        int4 pa = ld4(pk4, i), sa = ld4(sk4, i), oa = ld4(od4, i), ra = ld4(rv4, i);
+        uint64_t i2 = i + stride;
+        int4 pb_ = ld4(pk4, i2), sb = ld4(sk4, i2), ob = ld4(od4, i2), rb = ld4(rv4, i2);
+        quad(pa, sa, oa, ra);
+        quad(pb_, sb, ob, rb);
+    }
+    for (; i < n4; i += stride) quad(pk4[i], sk4[i], od4[i], rv4[i]);
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        acc += __shfl_down(acc, off, WAVE);
+    if ((threadIdx.x & (WAVE - 1)) == 0 && acc) atomicAdd(sink, acc);
+    if (g[threadIdx.x & (NG_Q21 - 1)] > ~0ull - 1) sink[0] = 0; // keep g live
+}
+
 extern "C" int gpue_ubench_q21(gpue_session* s, int which, gpue_dbuf* pk, gpue_dbuf* sk,
                                gpue_dbuf* od, gpue_dbuf* rv, gpue_dbuf* pbits,
                                int64_t psmin, uint64_t psint, uint64_t n, int grid1024,
@@ -4484,6 +4545,14 @@ int gpue_ubench_q21(gpue_session* s, int which, gpue_dbuf* pk, gpue_dbuf* sk, gp
                 hipLaunchKernelGGL(k_ub_q21_phase1, dim3(grid1024), dim3(BLOCK_Q21), 0,
                                    s->stream, (const int4*)pk->ptr, n4, (const uint32_t*)pbits->ptr,
                                    psmin, psint, d_out);
+            } else if (which == 2 || which == 3) {
+                ARG_CHECK(sk && od && rv);
+                auto kern = which == 3 ? k_ub_q21_pf_legs<true> : k_ub_q21_pf_legs<false>;
+                hipLaunchKernelGGL(kern, dim3(grid1024), dim3(BLOCK_Q21), 0, s->stream,
+                                   (const int32_t*)pk->ptr, (const int32_t*)sk->ptr,
+                                   (const int32_t*)od->ptr, (const int32_t*)rv->ptr, n,
+                                   pbits ? (const uint32_t*)pbits->ptr : nullptr, psmin,
+                                   psint, d_out);
             } else {
                 ARG_CHECK(sk && od && rv);
                 hipLaunchKernelGGL(k_ub_sum4, dim3(grid1024), dim3(BLOCK_Q21), 0, s->stream,

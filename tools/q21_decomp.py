@@ -37,3 +37,36 @@ for grid in (256, 512, 1024):
                                  smin, smax - smin + 1, n, grid, 10, ctypes.byref(ms)))
     print(f"grid {grid}: phase1-only (pk+bitgather): {ms.value:.3f} ms -> {4*n/(ms.value/1e3)/1e9:.0f} GB/s(pk)")
 eng.close()
+
+# r02: PF-kernel geometry legs (120 KB LDS, 1024 threads, 1 block/CU).
+# which=2 streams-only, which=3 + LDS prefilter test. Needs a 64 KB folded
+# prefilter buffer: fold the bitset on host exactly as k_build_prefilter.
+eng2 = Engine(0)
+lib2 = eng2._lib
+lib2.gpue_ubench_q21.restype = c_i32
+lib2.gpue_ubench_q21.argtypes = [c_vp, c_i32] + [c_vp] * 5 + [c_i64, c_u64, c_u64,
+                                 c_i32, c_i32, ctypes.POINTER(ctypes.c_float)]
+cols2 = [eng2.alloc(n * 4) for _ in range(4)]
+eng2.gen_lineorder_q21(42, 0, n, *cols2)
+pf_fold = np.zeros((1 << 19) // 32, np.uint32)
+set_idx = np.flatnonzero(np.kron(bits, np.uint32(1)))  # placeholder; fold below
+# fold: bit i of `bits` -> prefilter bit (i & (2^19-1))
+words = bits
+for w in range(len(words)):
+    v = int(words[w])
+    while v:
+        b = (v & -v).bit_length() - 1
+        i = w * 32 + b
+        f = i & ((1 << 19) - 1)
+        pf_fold[f >> 5] |= np.uint32(1 << (f & 31))
+        v &= v - 1
+pfb = eng2.alloc(pf_fold.nbytes)
+pfb.h2d(pf_fold)
+eng2.sync()
+ms2 = ctypes.c_float()
+for which, name in ((2, "pf-geom streams-only"), (3, "pf-geom streams+LDS-test")):
+    _ck(lib2, lib2.gpue_ubench_q21(eng2._h, which, cols2[0]._h, cols2[1]._h, cols2[2]._h,
+                                   cols2[3]._h, pfb._h, smin, smax - smin + 1, n, 256, 10,
+                                   ctypes.byref(ms2)))
+    print(f"{name}: {ms2.value:.3f} ms -> {16*n/(ms2.value/1e3)/1e9:.0f} GB/s")
+eng2.close()
