@@ -3613,6 +3613,123 @@ k_q21_star_agg_pf(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk
 }
 
 // ---------------------------------------------------------------------------
+// Wave-queue variant of the prefilter kernel (GPUE_Q21_PF=4). The
+// decomposition (tools/q21_decomp.py r02) shows streams + LDS prefilter
+// test run at 1.48 ms — phase 2 alone costs the remaining ~0.8 ms, because
+// with a 10% maybe rate nearly EVERY wave executes every sparse branch body
+// with ~6 active lanes (wave divergence, not memory). Fix: compact maybes
+// into a per-wave LDS queue ({pk,sk,od,rv} int4 entries, ballot+rank push)
+// and drain a FULL 64-lane wave of survivors at a time — phase-2 instruction
+// episodes drop ~8x and every confirm gather issues with all lanes active.
+// 64 KB pf + 56 KB groups + 32 KB queues = 152 KB LDS, 1 block/CU.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(BLOCK_Q21) void
+k_q21_star_agg_pfq(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
+                   const int32_t* __restrict__ od, const int32_t* __restrict__ rv,
+                   uint64_t n, const uint32_t* __restrict__ prefilter, int64_t psmin,
+                   uint64_t psint, const uint16_t* __restrict__ pfirst,
+                   const uint32_t* __restrict__ sbits, int64_t ssmin, uint64_t ssint,
+                   const uint16_t* __restrict__ dfirst, int64_t dmin,
+                   unsigned long long* __restrict__ group_sums) {
+    __shared__ uint32_t pf[PF_WORDS];              // 64 KB folded part filter
+    __shared__ unsigned long long g[NG_Q21];       // 56 KB group sums
+    __shared__ int4 wq[BLOCK_Q21 / WAVE][128];     // 32 KB per-wave maybe queues
+    for (uint32_t w = threadIdx.x; w < PF_WORDS; w += blockDim.x) pf[w] = prefilter[w];
+    for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x) g[j] = 0;
+    __syncthreads();
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x & (WAVE - 1);
+    uint32_t wqn = 0; // wave-uniform queue fill (ballot counts are uniform)
+    const uint64_t n4 = n / 4;
+    const int4* __restrict__ pk4 = (const int4*)pk;
+    const int4* __restrict__ sk4 = (const int4*)sk;
+    const int4* __restrict__ od4 = (const int4*)od;
+    const int4* __restrict__ rv4 = (const int4*)rv;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    auto ld4 = [&](const int4* p, uint64_t i) {
+        const uint64_t* q = (const uint64_t*)(p + i);
+        uint64_t lo = __builtin_nontemporal_load(q);
+        uint64_t hi = __builtin_nontemporal_load(q + 1);
+        int4 v;
+        v.x = (int32_t)lo; v.y = (int32_t)(lo >> 32);
+        v.z = (int32_t)hi; v.w = (int32_t)(hi >> 32);
+        return v;
+    };
+    // drain one full wave of queued maybes: every filter executes with all
+    // lanes carrying real candidates
+    auto drain64 = [&]() {
+        int4 e = wq[wid][wqn - 64 + lane];
+        uint32_t brand1 = pfirst[e.x - 1]; // 0 = fails the exact category filter
+        if (brand1) {
+            uint32_t sidx = (uint32_t)(e.y - ssmin);
+            if (sidx < ssint && ((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) {
+                uint32_t year1 = dfirst[e.z - dmin];
+                atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)],
+                          (unsigned long long)(int64_t)e.w);
+            }
+        }
+        wqn -= 64;
+    };
+    auto quad = [&](int4 p4, int4 s4, int4 o4, int4 r4) {
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint32_t idx = (uint32_t)((&p4.x)[j] - psmin);
+            bool in = idx < psint;
+            uint32_t fidx = (in ? idx : 0u) & PF_MASK;
+            bool maybe = in & (pf[fidx >> 5] >> (fidx & 31)) & 1u;
+            uint64_t m = __ballot(maybe);
+            if (m) {
+                uint32_t rank = __popcll(m & ((1ull << lane) - 1));
+                if (maybe)
+                    wq[wid][wqn + rank] =
+                        make_int4((&p4.x)[j], (&s4.x)[j], (&o4.x)[j], (&r4.x)[j]);
+                wqn += __popcll(m);
+                if (wqn >= 64) drain64();
+            }
+        }
+    };
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i + stride < n4; i += 2 * stride) {
+        int4 pa = ld4(pk4, i), sa = ld4(sk4, i), oa = ld4(od4, i), ra = ld4(rv4, i);
+        uint64_t i2 = i + stride;
+        int4 pb_ = ld4(pk4, i2), sb = ld4(sk4, i2), ob = ld4(od4, i2), rb = ld4(rv4, i2);
+        quad(pa, sa, oa, ra);
+        quad(pb_, sb, ob, rb);
+    }
+    for (; i < n4; i += stride) quad(pk4[i], sk4[i], od4[i], rv4[i]);
+    // drain the partial tail (lanes < wqn active)
+    if (wqn > 0 && lane < (int)wqn) {
+        int4 e = wq[wid][lane];
+        uint32_t brand1 = pfirst[e.x - 1];
+        if (brand1) {
+            uint32_t sidx = (uint32_t)(e.y - ssmin);
+            if (sidx < ssint && ((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) {
+                uint32_t year1 = dfirst[e.z - dmin];
+                atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)],
+                          (unsigned long long)(int64_t)e.w);
+            }
+        }
+    }
+    // scalar row tail (n % 4): exact path, no queue
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (uint64_t r = n4 * 4 + tid; r < n; r += stride) {
+        uint32_t idx = (uint32_t)(pk[r] - psmin);
+        if (idx >= psint) continue;
+        uint32_t fidx = idx & PF_MASK;
+        if (!((pf[fidx >> 5] >> (fidx & 31)) & 1u)) continue;
+        uint32_t brand1 = pfirst[pk[r] - 1];
+        if (!brand1) continue;
+        uint32_t sidx = (uint32_t)(sk[r] - ssmin);
+        if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
+        uint32_t year1 = dfirst[od[r] - dmin];
+        atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)], (unsigned long long)(int64_t)rv[r]);
+    }
+    __syncthreads();
+    for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x)
+        if (g[j] != 0) atomicAdd(&group_sums[j], g[j]);
+}
+
+// ---------------------------------------------------------------------------
 // Two-stream pipelined q21 (GPUE_Q21_PIPE=1): the fused kernel's streaming
 // leg (1.57 ms) and its part-probe gather leg (2.32 ms) measured fully
 // ADDITIVE (profiles/q21_decomp.log) — so split them into an operator pair
@@ -3816,7 +3933,8 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
                                dates->first16, dates->min_key,
                                (unsigned long long*)group_sums->ptr);
         };
-        if (mode == 2 && use_nt) launch(k_q21_star_agg_pf<false, true>, def_grid);
+        if (mode == 4) launch(k_q21_star_agg_pfq, def_grid); // wave-queue variant
+        else if (mode == 2 && use_nt) launch(k_q21_star_agg_pf<false, true>, def_grid);
         else if (mode == 2) launch(k_q21_star_agg_pf<false, false>, def_grid);
         else if (use_nt) launch(k_q21_star_agg_pf<true, true>, def_grid);
         else launch(k_q21_star_agg_pf<true, false>, def_grid);
