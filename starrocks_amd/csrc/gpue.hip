@@ -3670,11 +3670,14 @@ k_q21_star_agg_pfq(const int32_t* __restrict__ pk, const int32_t* __restrict__ s
         }
         wqn -= 64;
     };
-    auto quad = [&](int4 p4, int4 s4, int4 o4, int4 r4) {
+    // IMPORTANT: every loop condition below is WAVE-UNIFORM (based on the
+    // wave's base index, not the lane's) — the queue bookkeeping (wqn) is a
+    // wave-uniform register, so a lane-dependent trip count would desync it
+    auto quad = [&](int4 p4, int4 s4, int4 o4, int4 r4, bool inb) {
         #pragma unroll
         for (int j = 0; j < 4; j++) {
             uint32_t idx = (uint32_t)((&p4.x)[j] - psmin);
-            bool in = idx < psint;
+            bool in = inb & (idx < psint);
             uint32_t fidx = (in ? idx : 0u) & PF_MASK;
             bool maybe = in & (pf[fidx >> 5] >> (fidx & 31)) & 1u;
             uint64_t m = __ballot(maybe);
@@ -3688,15 +3691,24 @@ k_q21_star_agg_pfq(const int32_t* __restrict__ pk, const int32_t* __restrict__ s
             }
         }
     };
-    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    for (; i + stride < n4; i += 2 * stride) {
+    uint64_t base = (uint64_t)blockIdx.x * blockDim.x + (uint64_t)wid * WAVE;
+    uint64_t i = base + lane;
+    // full-wave main loop: both quads of every lane in bounds
+    for (; base + stride + WAVE <= n4; base += 2 * stride, i += 2 * stride) {
         int4 pa = ld4(pk4, i), sa = ld4(sk4, i), oa = ld4(od4, i), ra = ld4(rv4, i);
         uint64_t i2 = i + stride;
         int4 pb_ = ld4(pk4, i2), sb = ld4(sk4, i2), ob = ld4(od4, i2), rb = ld4(rv4, i2);
-        quad(pa, sa, oa, ra);
-        quad(pb_, sb, ob, rb);
+        quad(pa, sa, oa, ra, true);
+        quad(pb_, sb, ob, rb, true);
     }
-    for (; i < n4; i += stride) quad(pk4[i], sk4[i], od4[i], rv4[i]);
+    // wave-uniform predicated remainder (single quads, OOB lanes masked)
+    for (; base < n4; base += stride, i += stride) {
+        bool inb = i < n4;
+        int4 z = make_int4(0, 0, 0, 0);
+        int4 p4 = inb ? pk4[i] : z, s4 = inb ? sk4[i] : z;
+        int4 o4 = inb ? od4[i] : z, r4 = inb ? rv4[i] : z;
+        quad(p4, s4, o4, r4, inb);
+    }
     // drain the partial tail (lanes < wqn active)
     if (wqn > 0 && lane < (int)wqn) {
         int4 e = wq[wid][lane];
