@@ -4120,6 +4120,108 @@ k_q43_star_agg(const int32_t* __restrict__ ck, const int32_t* __restrict__ sk,
         if (g[j] != 0) atomicAdd(&group_sums[j], g[j]);
 }
 
+// ---------------------------------------------------------------------------
+// Wave-queue q43 (GPUE_Q43_PF=4): only the pk stream is read unconditionally
+// (2.4 GB); LDS-prefilter maybes (~10%) queue {pk, row} per wave and a FULL
+// 64-lane drain confirms against the exact bitset, then loads sk/ck/od (and
+// rv/sc for final survivors) as per-row scalars — the deferred-load idea
+// taken to its limit, with phase 2 free of sparse-lane divergence. LDS:
+// 64 KB pf + 6.4 KB groups + 16 KB queues = 87 KB, 1024-thr, 1 block/CU.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(BLOCK_Q21) void
+k_q43_star_agg_pfq(const int32_t* __restrict__ ck, const int32_t* __restrict__ sk,
+                   const int32_t* __restrict__ pk, const int32_t* __restrict__ od,
+                   const int32_t* __restrict__ rv, const int32_t* __restrict__ sc,
+                   uint64_t n, const uint32_t* __restrict__ prefilter,
+                   const uint32_t* __restrict__ cbits, int64_t csmin, uint64_t csint,
+                   const uint32_t* __restrict__ sbits, int64_t ssmin, uint64_t ssint,
+                   const uint16_t* __restrict__ sfirst,
+                   const uint32_t* __restrict__ pbits, int64_t psmin, uint64_t psint,
+                   const uint16_t* __restrict__ pfirst,
+                   const uint16_t* __restrict__ dfirst, int64_t dmin,
+                   unsigned long long* __restrict__ group_sums) {
+    __shared__ uint32_t pf[PF_WORDS];
+    __shared__ unsigned long long g[NG_Q43];
+    __shared__ int2 wq[BLOCK_Q21 / WAVE][128]; // {pk, row32}
+    for (uint32_t w = threadIdx.x; w < PF_WORDS; w += blockDim.x) pf[w] = prefilter[w];
+    for (int j = threadIdx.x; j < NG_Q43; j += blockDim.x) g[j] = 0;
+    __syncthreads();
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x & (WAVE - 1);
+    uint32_t wqn = 0; // wave-uniform (ballot counts); loops below are wave-uniform
+    const uint64_t n4 = n / 4;
+    const int4* __restrict__ pk4 = (const int4*)pk;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    auto process = [&](int2 e) { // full filter chain for one queued candidate
+        uint32_t idx = (uint32_t)(e.x - psmin);
+        if (!((pbits[idx >> 5] >> (idx & 31)) & 1u)) return; // exact part filter
+        uint32_t r = (uint32_t)e.y;
+        int32_t skv = sk[r];
+        uint32_t sidx = (uint32_t)(skv - ssmin);
+        if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) return;
+        uint32_t cidx = (uint32_t)(ck[r] - csmin);
+        if (cidx >= csint || !((cbits[cidx >> 5] >> (cidx & 31)) & 1u)) return;
+        uint32_t dpay = dfirst[od[r] - dmin];
+        if (dpay == 0) return;
+        uint32_t ppay = pfirst[e.x - 1];
+        uint32_t spay = sfirst[skv - 1];
+        atomicAdd(&g[(dpay - 1) * 400 + (spay - 1) * 40 + (ppay - 1)],
+                  (unsigned long long)((int64_t)rv[r] - sc[r]));
+    };
+    auto push = [&](int32_t key, uint32_t row, bool inb) {
+        uint32_t idx = (uint32_t)(key - psmin);
+        bool in = inb & (idx < psint);
+        uint32_t fidx = (in ? idx : 0u) & PF_MASK;
+        bool maybe = in & (pf[fidx >> 5] >> (fidx & 31)) & 1u;
+        uint64_t m = __ballot(maybe);
+        if (m) {
+            uint32_t rank = __popcll(m & ((1ull << lane) - 1));
+            if (maybe) wq[wid][wqn + rank] = make_int2(key, (int32_t)row);
+            wqn += __popcll(m);
+            if (wqn >= 64) {
+                process(wq[wid][wqn - 64 + lane]);
+                wqn -= 64;
+            }
+        }
+    };
+    uint64_t base = (uint64_t)blockIdx.x * blockDim.x + (uint64_t)wid * WAVE;
+    uint64_t i = base + lane;
+    for (; base + stride + WAVE <= n4; base += 2 * stride, i += 2 * stride) {
+        int4 pa = pk4[i];
+        uint64_t i2 = i + stride;
+        int4 pb_ = pk4[i2];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) push((&pa.x)[j], (uint32_t)(i * 4 + j), true);
+        #pragma unroll
+        for (int j = 0; j < 4; j++) push((&pb_.x)[j], (uint32_t)(i2 * 4 + j), true);
+    }
+    for (; base < n4; base += stride, i += stride) {
+        bool inb = i < n4;
+        int4 p4 = inb ? pk4[i] : make_int4(0, 0, 0, 0);
+        #pragma unroll
+        for (int j = 0; j < 4; j++) push((&p4.x)[j], (uint32_t)(i * 4 + j), inb);
+    }
+    if (wqn > 0 && lane < (int)wqn) process(wq[wid][lane]);
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (uint64_t r = n4 * 4 + tid; r < n; r += stride) {
+        uint32_t pidx = (uint32_t)(pk[r] - psmin);
+        if (pidx >= psint || !((pbits[pidx >> 5] >> (pidx & 31)) & 1u)) continue;
+        uint32_t sidx = (uint32_t)(sk[r] - ssmin);
+        if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
+        uint32_t cidx = (uint32_t)(ck[r] - csmin);
+        if (cidx >= csint || !((cbits[cidx >> 5] >> (cidx & 31)) & 1u)) continue;
+        uint32_t dpay = dfirst[od[r] - dmin];
+        if (dpay == 0) continue;
+        uint32_t ppay = pfirst[pk[r] - 1];
+        uint32_t spay = sfirst[sk[r] - 1];
+        atomicAdd(&g[(dpay - 1) * 400 + (spay - 1) * 40 + (ppay - 1)],
+                  (unsigned long long)((int64_t)rv[r] - sc[r]));
+    }
+    __syncthreads();
+    for (int j = threadIdx.x; j < NG_Q43; j += blockDim.x)
+        if (g[j] != 0) atomicAdd(&group_sums[j], g[j]);
+}
+
 extern "C" int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs,
                                        gpue_join_table* supps, gpue_join_table* parts,
                                        gpue_join_table* dates, gpue_dbuf* ck, gpue_dbuf* sk,
@@ -4158,9 +4260,12 @@ int gpue_q43_star_agg_accum_async(gpue_session* s, gpue_join_table* custs,
     ARG_CHECK(custs->bitset && supps->bitset && parts->bitset);
     ARG_CHECK(custs->min_key == 1 && supps->min_key == 1 && parts->min_key == 1);
     const char* pfe = getenv("GPUE_Q43_PF");
-    bool use_pf = (!pfe || atoi(pfe)) && parts->prefilter;
-    int def_grid = use_pf ? 512 : 256; // PF fits 2 blocks/CU
-    auto kern = use_pf ? k_q43_star_agg<true> : k_q43_star_agg<false>;
+    int pfm = pfe ? atoi(pfe) : 1;
+    bool use_pf = pfm && parts->prefilter;
+    bool use_q = pfm == 4 && parts->prefilter; // wave-queue variant
+    int def_grid = use_q ? 256 : (use_pf ? 512 : 256); // queue: 1 block/CU
+    auto kern = use_q ? k_q43_star_agg_pfq
+                      : (use_pf ? k_q43_star_agg<true> : k_q43_star_agg<false>);
     hipLaunchKernelGGL(kern, dim3(env_cap("GPUE_GRID_Q43", def_grid)), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)ck->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)pk->ptr, (const int32_t*)od->ptr,
