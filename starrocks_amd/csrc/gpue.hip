@@ -5841,6 +5841,101 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
         atomicAdd(ngroups, local_claims);
 }
 
+// ---------------------------------------------------------------------------
+// Two-stage wave-queue q3 probe (GPUE_Q3_WQ): the fused kernel's legs run at
+// low lane density (ship survivors 46%, order-bit survivors ~9%), so
+// every wave issues every leg with mostly-idle lanes. Queue A compacts
+// ship-passing rows; its full-wave drain loads lk and probes the order
+// bitset with all 64 lanes carrying real candidates; survivors queue into B,
+// whose full-wave drain loads ext/disc and does the hash-table insert.
+// The 56 MB bitset random-gather itself stays architectural
+// (profiles/r02_bitgather_ubench.json) — this removes the divergent-issue
+// overhead around it. All loop/drain conditions are wave-uniform (the
+// queue counters are wave-uniform registers).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(BLOCK) void
+k_q3_probe_agg_wq(const int64_t* __restrict__ lk, const int64_t* __restrict__ ext,
+                  const int64_t* __restrict__ disc, const int32_t* __restrict__ ship,
+                  uint64_t n, const uint32_t* __restrict__ order_bits, int32_t ship_cutoff,
+                  unsigned long long* __restrict__ slots,
+                  unsigned long long* __restrict__ sums,
+                  unsigned long long* __restrict__ counts, uint64_t cap_mask,
+                  unsigned int* __restrict__ err,
+                  unsigned long long* __restrict__ ngroups) {
+    (void)counts;
+    __shared__ uint32_t qa[BLOCK / WAVE][128]; // rows passing the ship filter
+    __shared__ uint2 qb[BLOCK / WAVE][128];    // {lk32, row} passing the order bitset
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x & (WAVE - 1);
+    uint32_t wqa = 0, wqb = 0; // wave-uniform
+    unsigned long long local_claims = 0;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+
+    auto insert = [&](uint32_t key32, uint32_t row) {
+        unsigned long long k = key32;
+        unsigned long long v = (unsigned long long)(__builtin_nontemporal_load(ext + row) *
+                                                    (100 - __builtin_nontemporal_load(disc + row)));
+        uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
+        uint64_t left = cap_mask + 1;
+        for (;;) {
+            unsigned long long cur = slots[s];
+            if (cur == k) { atomicAdd(&sums[s], v); break; }
+            if (cur == AGG_EMPTY) {
+                unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
+                if (old == AGG_EMPTY || old == k) {
+                    if (old == AGG_EMPTY) local_claims++;
+                    atomicAdd(&sums[s], v);
+                    break;
+                }
+            }
+            s = (s + 1) & cap_mask;
+            if (--left == 0) { atomicOr(err, AGG_ERR_FULL); break; }
+        }
+    };
+    // drain helpers take the count of VALID entries (wave-uniform); lanes
+    // beyond it participate in the ballots with pass=false
+    auto drainB = [&](uint32_t cnt) {
+        bool valid = (uint32_t)lane < cnt;
+        uint2 e = qb[wid][valid ? wqb - cnt + lane : 0];
+        if (valid) insert(e.x, e.y);
+        wqb -= cnt;
+    };
+    auto drainA = [&](uint32_t cnt) {
+        bool valid = (uint32_t)lane < cnt;
+        uint32_t r = qa[wid][valid ? wqa - cnt + lane : 0];
+        uint64_t k = valid ? (uint64_t)__builtin_nontemporal_load(lk + r) : 1;
+        uint64_t o = k - 1;
+        bool pass = valid && ((order_bits[o >> 5] >> (o & 31)) & 1u);
+        uint64_t m = __ballot(pass);
+        wqa -= cnt;
+        if (m) {
+            uint32_t rank = __popcll(m & ((1ull << lane) - 1));
+            if (pass) qb[wid][wqb + rank] = make_uint2((uint32_t)k, r);
+            wqb += __popcll(m);
+            if (wqb >= 64) drainB(64);
+        }
+    };
+    uint64_t base = (uint64_t)blockIdx.x * blockDim.x + (uint64_t)wid * WAVE;
+    uint64_t i = base + lane;
+    for (; base < n; base += stride, i += stride) {
+        bool inb = i < n;
+        bool pass = inb && (__builtin_nontemporal_load(ship + (inb ? i : 0)) > ship_cutoff);
+        uint64_t m = __ballot(pass);
+        if (m) {
+            uint32_t rank = __popcll(m & ((1ull << lane) - 1));
+            if (pass) qa[wid][wqa + rank] = (uint32_t)i;
+            wqa += __popcll(m);
+            if (wqa >= 64) drainA(64);
+        }
+    }
+    if (wqa > 0) drainA(wqa);
+    if (wqb > 0) drainB(wqb);
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        local_claims += __shfl_down(local_claims, off, WAVE);
+    if ((threadIdx.x & (WAVE - 1)) == 0 && local_claims && ngroups)
+        atomicAdd(ngroups, local_claims);
+}
+
 // Decomposition instrument for q3's roofline attribution:
 // legs bitmask accumulates per-leg cost — 1: ship stream, 2: +lk read &
 // order-bits gather, 4: +ext/disc product, 8: +hash-table insert (the full
@@ -6251,11 +6346,19 @@ int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbu
               n_groups);
     int rc = agg_table_reset(at, /*with_counts=*/false);
     if (rc != GPUE_OK) return rc;
-    hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))), dim3(BLOCK), 0, s->stream,
-                       (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
-                       (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
-                       (const uint32_t*)order_bits->ptr, ship_cutoff, at->slots, at->sums,
-                       at->counts, at->cap - 1, s->d_agg_err, at->n_groups);
+    const char* wq = getenv("GPUE_Q3_WQ");
+    if (!wq || atoi(wq)) // wave-queue default; =0 reverts to the fused quad form
+        hipLaunchKernelGGL(k_q3_probe_agg_wq, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))), dim3(BLOCK), 0, s->stream,
+                           (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
+                           (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
+                           (const uint32_t*)order_bits->ptr, ship_cutoff, at->slots,
+                           at->sums, at->counts, at->cap - 1, s->d_agg_err, at->n_groups);
+    else
+        hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))), dim3(BLOCK), 0, s->stream,
+                           (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
+                           (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
+                           (const uint32_t*)order_bits->ptr, ship_cutoff, at->slots,
+                           at->sums, at->counts, at->cap - 1, s->d_agg_err, at->n_groups);
     hipLaunchKernelGGL(k_hash_agg_emit, dim3(grid_for(at->cap)), dim3(BLOCK), 0, s->stream,
                        at->slots, at->sums, at->counts, at->cap, at->cursor, max_out,
                        (uint64_t*)out_keys->ptr, (int64_t*)out_sums->ptr, nullptr);
