@@ -6346,8 +6346,13 @@ int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbu
               n_groups);
     int rc = agg_table_reset(at, /*with_counts=*/false);
     if (rc != GPUE_OK) return rc;
+    // MEASURED NEGATIVE (r02): the two-stage wave-queue probe is 5.76 vs
+    // 5.65 ms — compaction doubles lane density but drops per-lane MLP from
+    // 4 in-flight bitset gathers (int4 quads) to 1 per drain, and the leg is
+    // latency-bound on a 56 MB footprint (profiles/r02_bitgather_ubench.json;
+    // same lesson as round 1's ballot-form rejection). Kept env-gated.
     const char* wq = getenv("GPUE_Q3_WQ");
-    if (!wq || atoi(wq)) // wave-queue default; =0 reverts to the fused quad form
+    if (wq && atoi(wq))
         hipLaunchKernelGGL(k_q3_probe_agg_wq, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))), dim3(BLOCK), 0, s->stream,
                            (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
                            (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
