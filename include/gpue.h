@@ -569,6 +569,13 @@ int gpue_page_decode_rle_i32(gpue_session* s, gpue_dbuf* page, uint64_t n_values
 /* BOOL variant (bit_width 1, bit-packed literal groups; u8 output) */
 int gpue_page_decode_rle_bool(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
                               gpue_dbuf* out);
+/* Frame-of-reference page decode (FOR_ENCODING,
+ * frame_of_reference_coding.{h,cpp}): independent 128-value frames
+ * (LE min + MSB-first bit-packed deltas; formats 0 min-delta / 1 ascending
+ * prefix / 2 raw), decoded one block per frame with an LDS scan for the
+ * ascending format. */
+int gpue_page_decode_for_i32(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
+                             gpue_dbuf* out);
 
 /* ---- event timing on the session stream (bench roofline evidence) ---- */
 int gpue_timer_start(gpue_session* s);

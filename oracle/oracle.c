@@ -1272,6 +1272,155 @@ uint64_t orc_rle_page_decode_i32(const uint8_t* page, int32_t* values) {
 }
 
 /* ====================================================================== */
+/* Frame-of-reference page codec for int32 (FOR_ENCODING,                   */
+/* storage/rowset/frame_of_reference_page.h over                            */
+/* base/bit/frame_of_reference_coding.{h,cpp}). Format (coding.h:76-100):   */
+/*   body:   per 128-value frame: fixed32-LE min, then bit-packed values    */
+/*           MSB-first within bytes (bit_pack, coding.cpp:97-118);          */
+/*           format 0: value-min deltas at bits(max-min);                   */
+/*           format 1 (ascending): value[i]-value[i-1] deltas (delta[0]=0); */
+/*           format 2 (range overflow): raw values at 32 bits               */
+/*   footer: per frame {format u8, bit_width u8}, then frame_value_num u8   */
+/*           (=128), values_num u32-LE                                      */
+/* NOTE an upstream quirk: the reference ENCODER resizes format-2 frame     */
+/* bodies to num*bit_width BYTES (coding.cpp:169 — 8x the packed size),     */
+/* while its own DECODER advances bit_width*128/8 + 4 per frame             */
+/* (coding.cpp:271-277). We restate the decoder-authoritative layout        */
+/* (ceil(num*bw/8) == bw*128/8 for full frames), which the reference        */
+/* decoder reads correctly; format-2 only triggers when a frame's value     */
+/* range exceeds 2^31.                                                      */
+/* ====================================================================== */
+
+static void for_bit_pack_u32(const uint32_t* in, int n, int bw, uint8_t* out) {
+    int bit_index = 0;
+    *out = 0;
+    for (int i = 0; i < n; i++) {
+        for (int k = bw - 1; k >= 0; k--) {
+            if (bit_index > 7) {
+                bit_index = 0;
+                *++out = 0;
+            }
+            *out |= (uint8_t)(((in[i] >> k) & 1u) << (7 - bit_index));
+            bit_index++;
+        }
+    }
+}
+
+static void for_bit_unpack_u32(const uint8_t* in, int n, int bw, uint32_t* out) {
+    int bit_index = 0;
+    for (int i = 0; i < n; i++) {
+        uint32_t v = 0;
+        for (int k = 0; k < bw; k++) {
+            if (bit_index > 7) {
+                in++;
+                bit_index = 0;
+            }
+            v |= (uint32_t)((*in >> (7 - bit_index)) & 1u) << (bw - k - 1);
+            bit_index++;
+        }
+        out[i] = v;
+    }
+}
+
+static inline uint8_t for_bits_u32(uint32_t v) {
+    return v == 0 ? 0 : (uint8_t)(32 - __builtin_clz(v));
+}
+
+uint64_t orc_for_page_encode_i32(const int32_t* values, uint32_t n, uint8_t* out) {
+    const uint32_t* u = (const uint32_t*)values;
+    uint64_t pos = 0;
+    uint32_t nframes = (n + 127) / 128;
+    uint8_t* fmts = (uint8_t*)malloc(nframes ? nframes : 1);
+    uint8_t* bws = (uint8_t*)malloc(nframes ? nframes : 1);
+    for (uint32_t f = 0; f < nframes; f++) {
+        const uint32_t* in = u + (uint64_t)f * 128;
+        int num = (f == nframes - 1) ? (int)(n - f * 128) : 128;
+        uint32_t mn = in[0], mx = in[0];
+        int ascending = 1, keep_original = 0;
+        uint8_t bw = 0;
+        for (int i = 1; i < num; i++) {
+            if (ascending) {
+                if (in[i] < in[i - 1]) ascending = 0;
+                else if ((in[i] >> 1) - (in[i - 1] >> 1) > (0xFFFFFFFFu >> 1)) keep_original = 1;
+                else { uint8_t b = for_bits_u32(in[i] - in[i - 1]); if (b > bw) bw = b; }
+            }
+            if (in[i] < mn) { mn = in[i]; continue; }
+            if (in[i] > mx) mx = in[i];
+        }
+        if (!ascending && (mx >> 1) - (mn >> 1) > (0xFFFFFFFFu >> 1)) keep_original = 1;
+        memcpy(out + pos, &mn, 4);
+        pos += 4;
+        uint32_t tmp[128];
+        if (keep_original) {
+            bw = 32;
+            for_bit_pack_u32(in, num, bw, out + pos);
+            pos += ((uint64_t)num * bw + 7) / 8;
+            fmts[f] = 2;
+        } else if (ascending) {
+            tmp[0] = 0;
+            for (int i = 1; i < num; i++) tmp[i] = in[i] - in[i - 1];
+            for_bit_pack_u32(tmp, num, bw, out + pos);
+            pos += ((uint64_t)num * bw + 7) / 8;
+            fmts[f] = 1;
+        } else {
+            bw = for_bits_u32(mx - mn);
+            for (int i = 0; i < num; i++) tmp[i] = in[i] - mn;
+            for_bit_pack_u32(tmp, num, bw, out + pos);
+            pos += ((uint64_t)num * bw + 7) / 8;
+            fmts[f] = 0;
+        }
+        bws[f] = bw;
+    }
+    for (uint32_t f = 0; f < nframes; f++) {
+        out[pos++] = fmts[f];
+        out[pos++] = bws[f];
+    }
+    out[pos++] = 128; /* frame_value_num */
+    memcpy(out + pos, &n, 4);
+    pos += 4;
+    free(fmts);
+    free(bws);
+    return pos;
+}
+
+uint64_t orc_for_page_decode_i32(const uint8_t* page, uint64_t page_bytes, int32_t* values) {
+    if (page_bytes < 5) return 0;
+    uint8_t frame_value_num = page[page_bytes - 5];
+    uint32_t n;
+    memcpy(&n, page + page_bytes - 4, 4);
+    if (n == 0) return 0;
+    uint32_t nframes = n / frame_value_num + (n % frame_value_num != 0);
+    uint64_t footer = page_bytes - 5 - (uint64_t)nframes * 2;
+    uint64_t off = 0;
+    for (uint32_t f = 0; f < nframes; f++) {
+        uint8_t fmt = page[footer + f * 2];
+        uint8_t bw = page[footer + f * 2 + 1];
+        int num = (f == nframes - 1) ? (int)(n - (uint64_t)f * frame_value_num)
+                                     : frame_value_num;
+        uint32_t mn;
+        memcpy(&mn, page + off, 4);
+        off += 4;
+        uint32_t tmp[256];
+        for_bit_unpack_u32(page + off, num, bw, tmp);
+        uint32_t* o = (uint32_t*)values + (uint64_t)f * frame_value_num;
+        if (fmt == 2) {
+            for (int i = 0; i < num; i++) o[i] = tmp[i];
+        } else if (fmt == 1) {
+            uint32_t acc = mn;
+            for (int i = 0; i < num; i++) { acc += tmp[i]; o[i] = acc; }
+            o[0] = mn; /* delta[0] == 0 */
+        } else {
+            for (int i = 0; i < num; i++) o[i] = mn + tmp[i];
+        }
+        /* decoder-authoritative frame advance (coding.cpp:271-277) uses the
+         * FULL frame size; equal to packed size for full frames */
+        off += (uint64_t)bw * frame_value_num / 8;
+        if (f == nframes - 1) break;
+    }
+    return n;
+}
+
+/* ====================================================================== */
 /* XXH3-64 exchange hash, version 1 (exchange_sink_operator.cpp:604-610:    */
 /* `_exchange_hash_function_version == 1` uses Column::xxh3_hash =          */
 /* HashUtil::xx_hash3_64 = XXH3_64bits_withSeed, chained per key column     */

@@ -314,3 +314,8 @@ uint64_t orc_rle_page_encode_i32(const int32_t* values, uint32_t n, uint8_t* out
 uint64_t orc_rle_page_decode_i32(const uint8_t* page, int32_t* values);
 uint64_t orc_rle_page_encode_bool(const uint8_t* values, uint32_t n, uint8_t* out);
 uint64_t orc_rle_page_decode_bool(const uint8_t* page, uint8_t* values);
+
+/* frame-of-reference page codec for int32 (FOR_ENCODING,
+ * frame_of_reference_coding.{h,cpp}; decoder-authoritative layout) */
+uint64_t orc_for_page_encode_i32(const int32_t* values, uint32_t n, uint8_t* out);
+uint64_t orc_for_page_decode_i32(const uint8_t* page, uint64_t page_bytes, int32_t* values);

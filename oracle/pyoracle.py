@@ -78,6 +78,10 @@ def _decl(lib):
     lib.orc_rle_page_encode_bool.argtypes = [c_vp, u, c_vp]
     lib.orc_rle_page_decode_bool.restype = c_u64
     lib.orc_rle_page_decode_bool.argtypes = [c_vp, c_vp]
+    lib.orc_for_page_encode_i32.restype = c_u64
+    lib.orc_for_page_encode_i32.argtypes = [c_vp, u, c_vp]
+    lib.orc_for_page_decode_i32.restype = c_u64
+    lib.orc_for_page_decode_i32.argtypes = [c_vp, c_u64, c_vp]
     lib.orc_xxh3_64_4to8.restype = c_u64
     lib.orc_xxh3_64_4to8.argtypes = [c_vp, c_i32, c_u64]
     lib.orc_xxh3_hash_i32.argtypes = [c_vp, c_u64, c_vp]
@@ -714,4 +718,19 @@ def rle_page_encode_bool(values: np.ndarray) -> np.ndarray:
 def rle_page_decode_bool(page: np.ndarray, n: int) -> np.ndarray:
     v = np.zeros(n, np.uint8)
     load().orc_rle_page_decode_bool(_p(np.ascontiguousarray(page, np.uint8)), _p(v))
+    return v
+
+
+def for_page_encode_i32(values: np.ndarray) -> np.ndarray:
+    a = np.ascontiguousarray(values, np.int32)
+    out = np.zeros(a.nbytes * 2 + 64, np.uint8)
+    nb = load().orc_for_page_encode_i32(_p(a), len(a), _p(out))
+    return out[:nb].copy()
+
+
+def for_page_decode_i32(page: np.ndarray, n: int) -> np.ndarray:
+    v = np.zeros(n, np.int32)
+    got = load().orc_for_page_decode_i32(_p(np.ascontiguousarray(page, np.uint8)),
+                                         len(page), _p(v))
+    assert got == n
     return v

@@ -6802,6 +6802,125 @@ __global__ void k_rle_fill_i32(const uint8_t* __restrict__ page,
     }
 }
 
+// ---------------------------------------------------------------------------
+// Frame-of-reference page decode for int32 (FOR_ENCODING,
+// frame_of_reference_page.h over frame_of_reference_coding.{h,cpp}; format
+// and the decoder-authoritative frame-advance documented in oracle.c).
+// Frames are independent 128-value units -> one 128-thread block per frame:
+// each thread bit-unpacks its value (MSB-first), format 1 (ascending
+// deltas) does an LDS inclusive scan, formats 0/2 add the frame min / pass
+// through. The footer walk is a single-wave scan kernel like the RLE one.
+// ---------------------------------------------------------------------------
+__global__ void k_for_scan_i32(const uint8_t* __restrict__ page, uint64_t page_bytes,
+                               uint64_t* __restrict__ frame_off, // byte offset of frame body
+                               uint8_t* __restrict__ frame_fmt,
+                               uint8_t* __restrict__ frame_bw,
+                               uint32_t* __restrict__ meta) { // [nframes, err, n, fvn]
+    if (blockIdx.x != 0 || threadIdx.x != 0) return;
+    if (page_bytes < 5) { meta[1] = 1; return; }
+    uint8_t fvn = page[page_bytes - 5];
+    uint32_t n;
+    memcpy(&n, page + page_bytes - 4, 4);
+    meta[2] = n;
+    meta[3] = fvn;
+    if (n == 0 || fvn == 0) { meta[0] = 0; return; }
+    uint32_t nframes = n / fvn + (n % fvn != 0);
+    uint64_t footer = (uint64_t)nframes * 2 + 5;
+    if (footer > page_bytes) { meta[1] = 1; return; }
+    uint64_t fpos = page_bytes - footer;
+    uint64_t off = 0;
+    for (uint32_t f = 0; f < nframes; f++) {
+        uint8_t fmt = page[fpos + f * 2];
+        uint8_t bw = page[fpos + f * 2 + 1];
+        frame_off[f] = off;
+        frame_fmt[f] = fmt;
+        frame_bw[f] = bw;
+        off += 4 + (uint64_t)bw * fvn / 8;
+        if (off > page_bytes) { meta[1] = 1; return; }
+    }
+    meta[0] = nframes;
+}
+
+__global__ void k_for_fill_i32(const uint8_t* __restrict__ page,
+                               const uint64_t* __restrict__ frame_off,
+                               const uint8_t* __restrict__ frame_fmt,
+                               const uint8_t* __restrict__ frame_bw, uint32_t fvn,
+                               uint64_t n, int32_t* __restrict__ out) {
+    __shared__ uint32_t vals[256]; // fvn <= 255 by format (u8)
+    uint32_t f = blockIdx.x;
+    uint32_t num = min((uint64_t)fvn, n - (uint64_t)f * fvn);
+    const uint8_t* body = page + frame_off[f];
+    uint32_t mn;
+    memcpy(&mn, body, 4);
+    const uint8_t* bits = body + 4;
+    uint32_t bw = frame_bw[f];
+    uint32_t t = threadIdx.x;
+    if (t < num) {
+        // bit-unpack value t: MSB-first stream, value bits big-endian
+        uint64_t bit0 = (uint64_t)t * bw;
+        uint32_t v = 0;
+        for (uint32_t k = 0; k < bw; k++) {
+            uint64_t b = bit0 + k;
+            v |= (uint32_t)((bits[b >> 3] >> (7 - (b & 7))) & 1u) << (bw - k - 1);
+        }
+        vals[t] = v;
+    }
+    __syncthreads();
+    uint8_t fmt = frame_fmt[f];
+    uint32_t* o = (uint32_t*)out + (uint64_t)f * fvn;
+    if (fmt == 1) {
+        // ascending: inclusive scan of deltas (delta[0] encoded as 0)
+        for (uint32_t s = 1; s < num; s <<= 1) {
+            uint32_t add = (t < num && t >= s) ? vals[t - s] : 0u;
+            __syncthreads();
+            if (t < num) vals[t] += add;
+            __syncthreads();
+        }
+        if (t < num) o[t] = mn + vals[t];
+    } else if (fmt == 2) {
+        if (t < num) o[t] = vals[t];
+    } else {
+        if (t < num) o[t] = mn + vals[t];
+    }
+}
+
+extern "C" int gpue_page_decode_for_i32(gpue_session* s, gpue_dbuf* page,
+                                        uint64_t n_values, gpue_dbuf* out);
+int gpue_page_decode_for_i32(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
+                             gpue_dbuf* out) {
+    ARG_CHECK(s && page && out && out->bytes >= n_values * 4 && page->bytes >= 5);
+    uint32_t max_frames = (uint32_t)(n_values / 1 + 2); // bounded below by meta check
+    uint64_t* d_off = nullptr;
+    uint8_t *d_fmt = nullptr, *d_bw = nullptr;
+    uint32_t* d_meta = nullptr;
+    HIP_CHECK(hipMalloc(&d_off, (uint64_t)max_frames * 8));
+    HIP_CHECK(hipMalloc(&d_fmt, max_frames));
+    HIP_CHECK(hipMalloc(&d_bw, max_frames));
+    HIP_CHECK(hipMalloc(&d_meta, 16));
+    HIP_CHECK(hipMemsetAsync(d_meta, 0, 16, s->stream));
+    hipLaunchKernelGGL(k_for_scan_i32, dim3(1), dim3(64), 0, s->stream,
+                       (const uint8_t*)page->ptr, page->bytes, d_off, d_fmt, d_bw, d_meta);
+    uint32_t meta[4] = {0, 0, 0, 0};
+    HIP_CHECK(hipMemcpyAsync(meta, d_meta, 16, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    int rc = GPUE_OK;
+    if (meta[1] || meta[2] != n_values || meta[3] == 0 || meta[3] > 255) {
+        snprintf(g_err, sizeof(g_err), "for page: err=%u n=%u (want %llu) fvn=%u",
+                 meta[1], meta[2], (unsigned long long)n_values, meta[3]);
+        rc = GPUE_ERR_ARG;
+    } else if (meta[0] > 0) {
+        hipLaunchKernelGGL(k_for_fill_i32, dim3(meta[0]), dim3(256), 0, s->stream,
+                           (const uint8_t*)page->ptr, d_off, d_fmt, d_bw, meta[3],
+                           n_values, (int32_t*)out->ptr);
+        HIP_CHECK(hipStreamSynchronize(s->stream));
+    }
+    (void)hipFree(d_off);
+    (void)hipFree(d_fmt);
+    (void)hipFree(d_bw);
+    (void)hipFree(d_meta);
+    return rc;
+}
+
 // BOOL variant (bit_width = 1): repeated value is one byte; literal groups
 // are bit-packed LSB-first, one byte per 8 values. Output u8 per value
 // (the reference decodes bool columns to u8).

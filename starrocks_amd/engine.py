@@ -183,6 +183,7 @@ def _declare(lib):
         "gpue_topk_i64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp]),
         "gpue_page_decode_rle_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp]),
         "gpue_page_decode_rle_bool": (c_i32, [c_vp, c_vp, c_u64, c_vp]),
+        "gpue_page_decode_for_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp]),
         "gpue_ubench_bitgather": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_u64, c_i32,
                                           ctypes.POINTER(ctypes.c_float)]),
         "gpue_page_decode_bshuf_lz4_i32": (c_i32, [c_vp, c_vp, c_u32, c_vp]),
@@ -794,6 +795,11 @@ class Engine:
         _ck(self._lib, self._lib.gpue_ubench_bitgather(
             self._h, idx._h, n, bits._h, nbits_pow2, reps, ctypes.byref(ms)))
         return ms.value
+
+    def page_decode_for_i32(self, page: DBuf, n_values, out: DBuf):
+        """Frame-of-reference page decode (FOR_ENCODING)."""
+        _ck(self._lib, self._lib.gpue_page_decode_for_i32(self._h, page._h,
+                                                          n_values, out._h))
 
     def page_decode_rle_bool(self, page: DBuf, n_values, out: DBuf):
         """RLE bool page decode (bit_width 1; u8 output)."""
