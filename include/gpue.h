@@ -576,6 +576,10 @@ int gpue_page_decode_rle_bool(gpue_session* s, gpue_dbuf* page, uint64_t n_value
  * ascending format. */
 int gpue_page_decode_for_i32(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
                              gpue_dbuf* out);
+/* BinaryPlainPage -> BinaryColumn (binary_plain_page.h:28-46: string body +
+ * u32 absolute-offset trailer + count). The dict page's dictionary format. */
+int gpue_page_decode_binary_plain(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
+                                  gpue_dbuf* out_bytes, gpue_dbuf* out_offsets);
 
 /* ---- event timing on the session stream (bench roofline evidence) ---- */
 int gpue_timer_start(gpue_session* s);

@@ -1272,6 +1272,40 @@ uint64_t orc_rle_page_decode_i32(const uint8_t* page, int32_t* values) {
 }
 
 /* ====================================================================== */
+/* BinaryPlainPage codec (PLAIN_ENCODING, binary_plain_page.h:28-46):       */
+/* body = concatenated strings; trailer = one u32-LE ABSOLUTE start offset  */
+/* per string (offsets[0] == 0), then u32-LE num_elems. This is also the    */
+/* page format a dict page's dictionary itself is stored in                 */
+/* (binary_dict_page.cpp).                                                  */
+/* ====================================================================== */
+
+uint64_t orc_binary_plain_encode(const uint8_t* bytes, const uint32_t* offsets,
+                                 uint32_t n, uint8_t* out) {
+    uint32_t body = offsets[n];
+    memcpy(out, bytes, body);
+    uint64_t pos = body;
+    for (uint32_t i = 0; i < n; i++) {
+        memcpy(out + pos, &offsets[i], 4);
+        pos += 4;
+    }
+    memcpy(out + pos, &n, 4);
+    return pos + 4;
+}
+
+/* returns n; fills out_bytes (body) and out_offsets[n+1] */
+uint64_t orc_binary_plain_decode(const uint8_t* page, uint64_t page_bytes,
+                                 uint8_t* out_bytes, uint32_t* out_offsets) {
+    uint32_t n;
+    memcpy(&n, page + page_bytes - 4, 4);
+    uint64_t body = page_bytes - 4 - (uint64_t)n * 4;
+    if (out_bytes) memcpy(out_bytes, page, body);
+    for (uint32_t i = 0; i < n; i++)
+        memcpy(&out_offsets[i], page + body + (uint64_t)i * 4, 4);
+    out_offsets[n] = (uint32_t)body;
+    return n;
+}
+
+/* ====================================================================== */
 /* Frame-of-reference page codec for int32 (FOR_ENCODING,                   */
 /* storage/rowset/frame_of_reference_page.h over                            */
 /* base/bit/frame_of_reference_coding.{h,cpp}). Format (coding.h:76-100):   */

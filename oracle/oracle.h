@@ -319,3 +319,8 @@ uint64_t orc_rle_page_decode_bool(const uint8_t* page, uint8_t* values);
  * frame_of_reference_coding.{h,cpp}; decoder-authoritative layout) */
 uint64_t orc_for_page_encode_i32(const int32_t* values, uint32_t n, uint8_t* out);
 uint64_t orc_for_page_decode_i32(const uint8_t* page, uint64_t page_bytes, int32_t* values);
+/* BinaryPlainPage codec (PLAIN_ENCODING, binary_plain_page.h:28-46) */
+uint64_t orc_binary_plain_encode(const uint8_t* bytes, const uint32_t* offsets,
+                                 uint32_t n, uint8_t* out);
+uint64_t orc_binary_plain_decode(const uint8_t* page, uint64_t page_bytes,
+                                 uint8_t* out_bytes, uint32_t* out_offsets);

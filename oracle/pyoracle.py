@@ -82,6 +82,10 @@ def _decl(lib):
     lib.orc_for_page_encode_i32.argtypes = [c_vp, u, c_vp]
     lib.orc_for_page_decode_i32.restype = c_u64
     lib.orc_for_page_decode_i32.argtypes = [c_vp, c_u64, c_vp]
+    lib.orc_binary_plain_encode.restype = c_u64
+    lib.orc_binary_plain_encode.argtypes = [c_vp, c_vp, u, c_vp]
+    lib.orc_binary_plain_decode.restype = c_u64
+    lib.orc_binary_plain_decode.argtypes = [c_vp, c_u64, c_vp, c_vp]
     lib.orc_xxh3_64_4to8.restype = c_u64
     lib.orc_xxh3_64_4to8.argtypes = [c_vp, c_i32, c_u64]
     lib.orc_xxh3_hash_i32.argtypes = [c_vp, c_u64, c_vp]
@@ -734,3 +738,22 @@ def for_page_decode_i32(page: np.ndarray, n: int) -> np.ndarray:
                                          len(page), _p(v))
     assert got == n
     return v
+
+
+def binary_plain_encode(bytes_: np.ndarray, offsets: np.ndarray) -> np.ndarray:
+    n = len(offsets) - 1
+    out = np.zeros(int(offsets[-1]) + 4 * n + 8, np.uint8)
+    nb = load().orc_binary_plain_encode(_p(np.ascontiguousarray(bytes_, np.uint8)),
+                                        _p(np.ascontiguousarray(offsets, np.uint32)),
+                                        n, _p(out))
+    return out[:nb].copy()
+
+
+def binary_plain_decode(page: np.ndarray, n: int):
+    pg = np.ascontiguousarray(page, np.uint8)
+    body = len(pg) - 4 - 4 * n
+    b = np.zeros(max(body, 1), np.uint8)
+    o = np.zeros(n + 1, np.uint32)
+    got = load().orc_binary_plain_decode(_p(pg), len(pg), _p(b), _p(o))
+    assert got == n
+    return b[:body], o

@@ -6803,6 +6803,56 @@ __global__ void k_rle_fill_i32(const uint8_t* __restrict__ page,
 }
 
 // ---------------------------------------------------------------------------
+// BinaryPlainPage decode (PLAIN_ENCODING, binary_plain_page.h:28-46): body
+// of concatenated strings + u32-LE absolute-offset trailer + u32 count.
+// Decodes straight into the engine's BinaryColumn shape (bytes + u32
+// offsets[n+1]) — the format a dict page's dictionary itself ships in, so
+// this + RLE/bitshuffle code pages + gpue_dict_decode_binary covers the
+// whole dict-encoded varchar ingress chain on device.
+// ---------------------------------------------------------------------------
+__global__ void k_binary_plain_offsets(const uint8_t* __restrict__ page, uint64_t body,
+                                       uint32_t n, uint32_t* __restrict__ out_offsets) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i <= n; i += stride) {
+        if (i == n) {
+            out_offsets[n] = (uint32_t)body;
+        } else {
+            uint32_t v;
+            memcpy(&v, page + body + i * 4, 4);
+            out_offsets[i] = v;
+        }
+    }
+}
+
+extern "C" int gpue_page_decode_binary_plain(gpue_session* s, gpue_dbuf* page,
+                                             uint64_t n_values, gpue_dbuf* out_bytes,
+                                             gpue_dbuf* out_offsets);
+int gpue_page_decode_binary_plain(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
+                                  gpue_dbuf* out_bytes, gpue_dbuf* out_offsets) {
+    ARG_CHECK(s && page && out_bytes && out_offsets && page->bytes >= 4);
+    ARG_CHECK(out_offsets->bytes >= (n_values + 1) * 4);
+    uint32_t n = 0;
+    HIP_CHECK(hipMemcpyAsync(&n, (const uint8_t*)page->ptr + page->bytes - 4, 4,
+                             hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    if (n != n_values || page->bytes < 4 + (uint64_t)n * 4) {
+        snprintf(g_err, sizeof(g_err), "binary_plain page: n=%u want %llu", n,
+                 (unsigned long long)n_values);
+        return GPUE_ERR_ARG;
+    }
+    uint64_t body = page->bytes - 4 - (uint64_t)n * 4;
+    ARG_CHECK(out_bytes->bytes >= body || body == 0);
+    if (body)
+        HIP_CHECK(hipMemcpyAsync(out_bytes->ptr, page->ptr, body,
+                                 hipMemcpyDeviceToDevice, s->stream));
+    hipLaunchKernelGGL(k_binary_plain_offsets, dim3(grid_for(n + 1)), dim3(BLOCK), 0,
+                       s->stream, (const uint8_t*)page->ptr, body, n,
+                       (uint32_t*)out_offsets->ptr);
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    return GPUE_OK;
+}
+
+// ---------------------------------------------------------------------------
 // Frame-of-reference page decode for int32 (FOR_ENCODING,
 // frame_of_reference_page.h over frame_of_reference_coding.{h,cpp}; format
 // and the decoder-authoritative frame-advance documented in oracle.c).
