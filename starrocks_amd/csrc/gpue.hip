@@ -4142,11 +4142,7 @@ k_q43_star_agg_pfq(const int32_t* __restrict__ ck, const int32_t* __restrict__ s
                    unsigned long long* __restrict__ group_sums) {
     __shared__ uint32_t pf[PF_WORDS];
     __shared__ unsigned long long g[NG_Q43];
-    __shared__ int2 wq[BLOCK_Q21 / WAVE][256]; // {pk, row32}; 256-deep so the
-                                               // drain runs 2 entries/lane
-                                               // (2 concurrent filter chains
-                                               // per lane = 2x the MLP of the
-                                               // latency-bound scalar gathers)
+    __shared__ int2 wq[BLOCK_Q21 / WAVE][128]; // {pk, row32}
     for (uint32_t w = threadIdx.x; w < PF_WORDS; w += blockDim.x) pf[w] = prefilter[w];
     for (int j = threadIdx.x; j < NG_Q43; j += blockDim.x) g[j] = 0;
     __syncthreads();
@@ -4172,35 +4168,6 @@ k_q43_star_agg_pfq(const int32_t* __restrict__ ck, const int32_t* __restrict__ s
         atomicAdd(&g[(dpay - 1) * 400 + (spay - 1) * 40 + (ppay - 1)],
                   (unsigned long long)((int64_t)rv[r] - sc[r]));
     };
-    // two candidates per lane with their loads interleaved: the compiler
-    // issues both chains' gathers before either chain's dependent branch,
-    // doubling the in-flight scalar gathers per lane
-    auto process2 = [&](int2 a, int2 b) {
-        uint32_t ia = (uint32_t)(a.x - psmin), ib = (uint32_t)(b.x - psmin);
-        bool pa = (pbits[ia >> 5] >> (ia & 31)) & 1u;
-        bool pb = (pbits[ib >> 5] >> (ib & 31)) & 1u;
-        uint32_t ra = (uint32_t)a.y, rb = (uint32_t)b.y;
-        int32_t ska = pa ? sk[ra] : 0, skb = pb ? sk[rb] : 0;
-        uint32_t sa = (uint32_t)(ska - ssmin), sb_ = (uint32_t)(skb - ssmin);
-        pa = pa && sa < ssint && ((sbits[sa >> 5] >> (sa & 31)) & 1u);
-        pb = pb && sb_ < ssint && ((sbits[sb_ >> 5] >> (sb_ & 31)) & 1u);
-        uint32_t ca = pa ? (uint32_t)(ck[ra] - csmin) : 0;
-        uint32_t cb = pb ? (uint32_t)(ck[rb] - csmin) : 0;
-        pa = pa && ca < csint && ((cbits[ca >> 5] >> (ca & 31)) & 1u);
-        pb = pb && cb < csint && ((cbits[cb >> 5] >> (cb & 31)) & 1u);
-        uint32_t da = pa ? dfirst[od[ra] - dmin] : 0;
-        uint32_t db = pb ? dfirst[od[rb] - dmin] : 0;
-        if (pa && da) {
-            uint32_t ppay = pfirst[a.x - 1], spay = sfirst[ska - 1];
-            atomicAdd(&g[(da - 1) * 400 + (spay - 1) * 40 + (ppay - 1)],
-                      (unsigned long long)((int64_t)rv[ra] - sc[ra]));
-        }
-        if (pb && db) {
-            uint32_t ppay = pfirst[b.x - 1], spay = sfirst[skb - 1];
-            atomicAdd(&g[(db - 1) * 400 + (spay - 1) * 40 + (ppay - 1)],
-                      (unsigned long long)((int64_t)rv[rb] - sc[rb]));
-        }
-    };
     auto push = [&](int32_t key, uint32_t row, bool inb) {
         uint32_t idx = (uint32_t)(key - psmin);
         bool in = inb & (idx < psint);
@@ -4211,11 +4178,9 @@ k_q43_star_agg_pfq(const int32_t* __restrict__ ck, const int32_t* __restrict__ s
             uint32_t rank = __popcll(m & ((1ull << lane) - 1));
             if (maybe) wq[wid][wqn + rank] = make_int2(key, (int32_t)row);
             wqn += __popcll(m);
-            if (wqn >= 128) { // drain 2 entries per lane (interleaved chains)
-                int2 e1 = wq[wid][wqn - 128 + lane];
-                int2 e2 = wq[wid][wqn - 64 + lane];
-                process2(e1, e2);
-                wqn -= 128;
+            if (wqn >= 64) {
+                process(wq[wid][wqn - 64 + lane]);
+                wqn -= 64;
             }
         }
     };
@@ -4235,10 +4200,6 @@ k_q43_star_agg_pfq(const int32_t* __restrict__ ck, const int32_t* __restrict__ s
         int4 p4 = inb ? pk4[i] : make_int4(0, 0, 0, 0);
         #pragma unroll
         for (int j = 0; j < 4; j++) push((&p4.x)[j], (uint32_t)(i * 4 + j), inb);
-    }
-    if (wqn >= 64) {
-        process(wq[wid][wqn - 64 + lane]);
-        wqn -= 64;
     }
     if (wqn > 0 && lane < (int)wqn) process(wq[wid][lane]);
     uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
