@@ -6184,43 +6184,6 @@ __global__ void k_q3_probe_slices_fused(const uint32_t* __restrict__ keys,
         atomicAdd(ngroups, local_claims);
 }
 
-__global__ void k_q3_probe_slice(const uint32_t* __restrict__ keys,
-                                 const long long* __restrict__ vals, uint64_t n,
-                                 const uint32_t* __restrict__ order_bits,
-                                 unsigned long long* __restrict__ slots,
-                                 unsigned long long* __restrict__ sums, uint64_t cap_mask,
-                                 unsigned int* __restrict__ err,
-                                 unsigned long long* __restrict__ ngroups) {
-    unsigned long long local_claims = 0;
-    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
-        unsigned long long k = keys[i];
-        uint64_t o = k - 1;
-        if (!((order_bits[o >> 5] >> (o & 31)) & 1u)) continue;
-        unsigned long long v = (unsigned long long)vals[i];
-        uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
-        uint64_t left = cap_mask + 1;
-        for (;;) {
-            unsigned long long cur = slots[s];
-            if (cur == k) { atomicAdd(&sums[s], v); break; }
-            if (cur == AGG_EMPTY) {
-                unsigned long long old = atomicCAS(&slots[s], AGG_EMPTY, k);
-                if (old == AGG_EMPTY || old == k) {
-                    if (old == AGG_EMPTY) local_claims++;
-                    atomicAdd(&sums[s], v);
-                    break;
-                }
-            }
-            s = (s + 1) & cap_mask;
-            if (--left == 0) { atomicOr(err, AGG_ERR_FULL); break; }
-        }
-    }
-    for (int off = WAVE / 2; off > 0; off >>= 1)
-        local_claims += __shfl_down(local_claims, off, WAVE);
-    if ((threadIdx.x & (WAVE - 1)) == 0 && local_claims && ngroups)
-        atomicAdd(ngroups, local_claims);
-}
-
 extern "C" {
 int gpue_gen_lineitem_q3(gpue_session* s, uint64_t seed, uint64_t row_start, uint64_t n,
                          uint64_t n_orders, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
