@@ -424,6 +424,16 @@ int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs, gpue_join_t
 int gpue_partition_xxh3_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n,
                             uint32_t num_channels, uint64_t* start_points_out,
                             gpue_dbuf* row_indexes_out);
+/* The bucket-shuffle hash path (zlib crc32, seed 0 — the third of the
+ * exchange's three hash functions, exchange_sink_operator.cpp:617-622;
+ * pinned live against python zlib). */
+int gpue_partition_crc_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n,
+                           uint32_t num_channels, uint64_t* start_points_out,
+                           gpue_dbuf* row_indexes_out);
+/* Varchar (BinaryColumn) partition key: FNV over the slice bytes. */
+int gpue_partition_varchar(gpue_session* s, gpue_dbuf* bytes, gpue_dbuf* offsets,
+                           uint64_t n, uint32_t num_channels,
+                           uint64_t* start_points_out, gpue_dbuf* row_indexes_out);
 int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t num_channels,
                        uint64_t* start_points_out, gpue_dbuf* row_indexes_out);
 /* Multi-column partition key: the sink seeds FNV_SEED then CHAINS fnv_hash

@@ -1518,6 +1518,44 @@ void orc_xxh3_hash_i64(const int64_t* col, uint64_t n, uint32_t* hashes) {
         hashes[i] = (uint32_t)orc_xxh3_64_4to8(&col[i], 8, hashes[i]);
 }
 
+/* zlib CRC32 (polynomial 0xEDB88320, standard pre/post inversion) — the
+ * exchange's third hash path: bucket-shuffle partitioning hashes with
+ * HashUtil::zlib_crc_hash seeded 0 (exchange_sink_operator.cpp:617-622,
+ * hash_util.hpp:37-39, column_hash.cpp:53-57). Pinned live against python's
+ * zlib.crc32 in tests. */
+uint32_t orc_zlib_crc32(const void* data, int32_t n, uint32_t seed) {
+    const uint8_t* p = (const uint8_t*)data;
+    uint32_t crc = ~seed;
+    for (int32_t i = 0; i < n; i++) {
+        crc ^= p[i];
+        for (int k = 0; k < 8; k++)
+            crc = (crc >> 1) ^ (0xEDB88320u & (0u - (crc & 1u)));
+    }
+    return ~crc;
+}
+
+void orc_partition_channel_crc_u32(const uint32_t* keys, uint64_t n,
+                                   uint32_t num_channels, uint32_t* channel_ids) {
+#pragma omp parallel for schedule(static)
+    for (uint64_t i = 0; i < n; i++) {
+        uint32_t h = orc_zlib_crc32(&keys[i], 4, 0);
+        channel_ids[i] = (uint32_t)(((uint64_t)h * num_channels) >> 32);
+    }
+}
+
+/* varchar (BinaryColumn) partition key: the default FNV path over the slice
+ * bytes (fnv_hash_column on BinaryColumn, seed FNV_SEED) + ReduceOp */
+void orc_partition_channel_fnv_slice(const uint8_t* bytes, const uint32_t* offsets,
+                                     uint64_t n, uint32_t num_channels,
+                                     uint32_t* channel_ids) {
+#pragma omp parallel for schedule(static)
+    for (uint64_t i = 0; i < n; i++) {
+        uint32_t h = orc_fnv_hash(bytes + offsets[i],
+                                  (int32_t)(offsets[i + 1] - offsets[i]), 0x811C9DC5u);
+        channel_ids[i] = (uint32_t)(((uint64_t)h * num_channels) >> 32);
+    }
+}
+
 /* single-column xxh3 partition: seed init + hash + ReduceOp channel
  * (shuffler.h:71-86 — the Shuffler is hash-version agnostic) */
 void orc_partition_channel_xxh3_u32(const uint32_t* keys, uint64_t n,

@@ -324,3 +324,10 @@ uint64_t orc_binary_plain_encode(const uint8_t* bytes, const uint32_t* offsets,
                                  uint32_t n, uint8_t* out);
 uint64_t orc_binary_plain_decode(const uint8_t* page, uint64_t page_bytes,
                                  uint8_t* out_bytes, uint32_t* out_offsets);
+/* zlib CRC32 + the exchange's crc (bucket-shuffle) and varchar-key paths */
+uint32_t orc_zlib_crc32(const void* data, int32_t n, uint32_t seed);
+void orc_partition_channel_crc_u32(const uint32_t* keys, uint64_t n,
+                                   uint32_t num_channels, uint32_t* channel_ids);
+void orc_partition_channel_fnv_slice(const uint8_t* bytes, const uint32_t* offsets,
+                                     uint64_t n, uint32_t num_channels,
+                                     uint32_t* channel_ids);

@@ -91,6 +91,10 @@ def _decl(lib):
     lib.orc_xxh3_hash_i32.argtypes = [c_vp, c_u64, c_vp]
     lib.orc_xxh3_hash_i64.argtypes = [c_vp, c_u64, c_vp]
     lib.orc_partition_channel_xxh3_u32.argtypes = [c_vp, c_u64, u, c_vp]
+    lib.orc_zlib_crc32.restype = u
+    lib.orc_zlib_crc32.argtypes = [c_vp, c_i32, u]
+    lib.orc_partition_channel_crc_u32.argtypes = [c_vp, c_u64, u, c_vp]
+    lib.orc_partition_channel_fnv_slice.argtypes = [c_vp, c_vp, c_u64, u, c_vp]
     lib.orc_gen_u64.restype = c_u64
     lib.orc_gen_u64.argtypes = [c_u64, c_u64, c_u64]
     lib.orc_gen_dates.argtypes = [c_i32, c_vp, c_vp]
@@ -757,3 +761,24 @@ def binary_plain_decode(page: np.ndarray, n: int):
     got = load().orc_binary_plain_decode(_p(pg), len(pg), _p(b), _p(o))
     assert got == n
     return b[:body], o
+
+
+def zlib_crc32(data: bytes, seed: int = 0) -> int:
+    return load().orc_zlib_crc32(data, len(data), seed)
+
+
+def partition_channels_crc(keys: np.ndarray, num_channels: int) -> np.ndarray:
+    ch = np.empty(len(keys), np.uint32)
+    load().orc_partition_channel_crc_u32(_p(np.ascontiguousarray(keys, np.uint32)),
+                                         len(keys), num_channels, _p(ch))
+    return ch
+
+
+def partition_channels_fnv_slice(bytes_: np.ndarray, offsets: np.ndarray,
+                                 num_channels: int) -> np.ndarray:
+    n = len(offsets) - 1
+    ch = np.empty(n, np.uint32)
+    load().orc_partition_channel_fnv_slice(
+        _p(np.ascontiguousarray(bytes_, np.uint8)),
+        _p(np.ascontiguousarray(offsets, np.uint32)), n, num_channels, _p(ch))
+    return ch

@@ -4330,6 +4330,21 @@ __device__ static inline uint64_t xxh3_64_u64(uint64_t v, uint64_t seed) {
     return xxh3_rrmxmx(input64 ^ bitflip, 8);
 }
 
+// zlib CRC32 (poly 0xEDB88320) — the bucket-shuffle hash path
+// (exchange_sink_operator.cpp:617-622 via HashUtil::zlib_crc_hash, seed 0).
+// Pinned against python zlib through the oracle restatement.
+__device__ static inline uint32_t zlib_crc32_u32(uint32_t key, uint32_t seed) {
+    uint32_t crc = ~seed;
+    #pragma unroll
+    for (int b = 0; b < 4; b++) {
+        crc ^= (key >> (8 * b)) & 0xFFu;
+        #pragma unroll
+        for (int k = 0; k < 8; k++)
+            crc = (crc >> 1) ^ (0xEDB88320u & (0u - (crc & 1u)));
+    }
+    return ~crc;
+}
+
 __device__ static inline uint32_t fnv_u32(uint32_t key, uint32_t seed) {
     uint32_t h = seed;
     #pragma unroll
@@ -4356,8 +4371,9 @@ __global__ void k_partition_hist(const uint32_t* __restrict__ keys, uint64_t n, 
     __syncthreads();
     uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
     for (uint64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
-        uint32_t hash = HV ? (uint32_t)xxh3_64_u32(keys[i], XXH3_SEED_32)
-                           : fnv_u32(keys[i], 0x811C9DC5u);
+        uint32_t hash = HV == 2   ? zlib_crc32_u32(keys[i], 0)
+                        : HV == 1 ? (uint32_t)xxh3_64_u32(keys[i], XXH3_SEED_32)
+                                  : fnv_u32(keys[i], 0x811C9DC5u);
         uint32_t ch = (uint32_t)(((uint64_t)hash * nch) >> 32);
         atomicAdd(&h[ch], 1u);
     }
@@ -4376,8 +4392,9 @@ __global__ void k_partition_emit(const uint32_t* __restrict__ keys, uint64_t n, 
     __syncthreads();
     uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
     for (uint64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
-        uint32_t hash = HV ? (uint32_t)xxh3_64_u32(keys[i], XXH3_SEED_32)
-                           : fnv_u32(keys[i], 0x811C9DC5u);
+        uint32_t hash = HV == 2   ? zlib_crc32_u32(keys[i], 0)
+                        : HV == 1 ? (uint32_t)xxh3_64_u32(keys[i], XXH3_SEED_32)
+                                  : fnv_u32(keys[i], 0x811C9DC5u);
         uint32_t ch = (uint32_t)(((uint64_t)hash * nch) >> 32);
         uint64_t pos = atomicAdd((unsigned long long*)&cursor[ch], 1ull);
         row_indexes[pos] = (uint32_t)i;
@@ -4510,7 +4527,10 @@ static int partition_i32_impl(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint
     uint32_t* d_hist = nullptr;
     uint64_t* d_off = nullptr;
     HIP_CHECK(hipMalloc(&d_hist, (uint64_t)nb * nch * sizeof(uint32_t)));
-    if (hash_version == 1)
+    if (hash_version == 2)
+        hipLaunchKernelGGL(k_partition_hist<2>, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const uint32_t*)keys->ptr, n, tile, nch, d_hist);
+    else if (hash_version == 1)
         hipLaunchKernelGGL(k_partition_hist<1>, dim3(nb), dim3(BLOCK), 0, s->stream,
                            (const uint32_t*)keys->ptr, n, tile, nch, d_hist);
     else
@@ -4534,7 +4554,11 @@ static int partition_i32_impl(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint
     HIP_CHECK(hipMalloc(&d_off, (uint64_t)nb * nch * sizeof(uint64_t)));
     HIP_CHECK(hipMemcpyAsync(d_off, h_off, (uint64_t)nb * nch * sizeof(uint64_t),
                              hipMemcpyHostToDevice, s->stream));
-    if (hash_version == 1)
+    if (hash_version == 2)
+        hipLaunchKernelGGL(k_partition_emit<2>, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const uint32_t*)keys->ptr, n, tile, nch, d_off,
+                           (uint32_t*)row_indexes_out->ptr);
+    else if (hash_version == 1)
         hipLaunchKernelGGL(k_partition_emit<1>, dim3(nb), dim3(BLOCK), 0, s->stream,
                            (const uint32_t*)keys->ptr, n, tile, nch, d_off,
                            (uint32_t*)row_indexes_out->ptr);
@@ -4563,6 +4587,103 @@ extern "C" int gpue_partition_xxh3_i32(gpue_session* s, gpue_dbuf* keys, uint64_
 int gpue_partition_xxh3_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t nch,
                             uint64_t* start_points_out, gpue_dbuf* row_indexes_out) {
     return partition_i32_impl(s, keys, n, nch, start_points_out, row_indexes_out, 1);
+}
+
+// bucket-shuffle hash path (zlib crc32, seed 0) — the third exchange hash
+// (exchange_sink_operator.cpp:617-622)
+extern "C" int gpue_partition_crc_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n,
+                                      uint32_t nch, uint64_t* start_points_out,
+                                      gpue_dbuf* row_indexes_out);
+int gpue_partition_crc_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t nch,
+                           uint64_t* start_points_out, gpue_dbuf* row_indexes_out) {
+    return partition_i32_impl(s, keys, n, nch, start_points_out, row_indexes_out, 2);
+}
+
+// varchar (BinaryColumn) partition key: the exchange hashes the slice bytes
+// (fnv_hash over BinaryColumn, FNV_SEED) — same counting-sort layout
+__global__ void k_partition_hist_vc(const uint8_t* __restrict__ bytes,
+                                    const uint32_t* __restrict__ offsets, uint64_t n,
+                                    uint64_t tile, uint32_t nch,
+                                    uint32_t* __restrict__ block_hist) {
+    __shared__ uint32_t h[MAX_CH];
+    for (uint32_t c = threadIdx.x; c < nch; c += blockDim.x) h[c] = 0;
+    __syncthreads();
+    uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
+    for (uint64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        uint32_t hash = 0x811C9DC5u;
+        for (uint32_t b = offsets[i]; b < offsets[i + 1]; b++)
+            hash = (bytes[b] ^ hash) * 16777619u;
+        uint32_t ch = (uint32_t)(((uint64_t)hash * nch) >> 32);
+        atomicAdd(&h[ch], 1u);
+    }
+    __syncthreads();
+    for (uint32_t c = threadIdx.x; c < nch; c += blockDim.x)
+        block_hist[(uint64_t)blockIdx.x * nch + c] = h[c];
+}
+
+__global__ void k_partition_emit_vc(const uint8_t* __restrict__ bytes,
+                                    const uint32_t* __restrict__ offsets, uint64_t n,
+                                    uint64_t tile, uint32_t nch,
+                                    const uint64_t* __restrict__ block_offsets,
+                                    uint32_t* __restrict__ row_indexes) {
+    __shared__ uint64_t cursor[MAX_CH];
+    for (uint32_t c = threadIdx.x; c < nch; c += blockDim.x)
+        cursor[c] = block_offsets[(uint64_t)blockIdx.x * nch + c];
+    __syncthreads();
+    uint64_t lo = (uint64_t)blockIdx.x * tile, hi = min(lo + tile, n);
+    for (uint64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+        uint32_t hash = 0x811C9DC5u;
+        for (uint32_t b = offsets[i]; b < offsets[i + 1]; b++)
+            hash = (bytes[b] ^ hash) * 16777619u;
+        uint32_t ch = (uint32_t)(((uint64_t)hash * nch) >> 32);
+        uint64_t pos = atomicAdd((unsigned long long*)&cursor[ch], 1ull);
+        row_indexes[pos] = (uint32_t)i;
+    }
+}
+
+extern "C" int gpue_partition_varchar(gpue_session* s, gpue_dbuf* bytes, gpue_dbuf* offsets,
+                                      uint64_t n, uint32_t nch, uint64_t* start_points_out,
+                                      gpue_dbuf* row_indexes_out);
+int gpue_partition_varchar(gpue_session* s, gpue_dbuf* bytes, gpue_dbuf* offsets,
+                           uint64_t n, uint32_t nch, uint64_t* start_points_out,
+                           gpue_dbuf* row_indexes_out) {
+    ARG_CHECK(s && bytes && offsets && start_points_out && row_indexes_out);
+    ARG_CHECK(nch >= 1 && nch <= MAX_CH);
+    ARG_CHECK(offsets->bytes >= (n + 1) * 4 && row_indexes_out->bytes >= n * 4);
+    uint32_t nb = grid_for(n);
+    uint64_t tile = (n + nb - 1) / nb;
+    uint32_t* d_hist = nullptr;
+    uint64_t* d_off = nullptr;
+    HIP_CHECK(hipMalloc(&d_hist, (uint64_t)nb * nch * sizeof(uint32_t)));
+    hipLaunchKernelGGL(k_partition_hist_vc, dim3(nb), dim3(BLOCK), 0, s->stream,
+                       (const uint8_t*)bytes->ptr, (const uint32_t*)offsets->ptr, n, tile,
+                       nch, d_hist);
+    uint32_t* h_hist = (uint32_t*)malloc((uint64_t)nb * nch * sizeof(uint32_t));
+    uint64_t* h_off = (uint64_t*)malloc((uint64_t)nb * nch * sizeof(uint64_t));
+    HIP_CHECK(hipMemcpyAsync(h_hist, d_hist, (uint64_t)nb * nch * sizeof(uint32_t),
+                             hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    uint64_t acc = 0;
+    for (uint32_t c = 0; c < nch; c++) {
+        start_points_out[c] = acc;
+        for (uint32_t b = 0; b < nb; b++) {
+            h_off[(uint64_t)b * nch + c] = acc;
+            acc += h_hist[(uint64_t)b * nch + c];
+        }
+    }
+    start_points_out[nch] = acc;
+    HIP_CHECK(hipMalloc(&d_off, (uint64_t)nb * nch * sizeof(uint64_t)));
+    HIP_CHECK(hipMemcpyAsync(d_off, h_off, (uint64_t)nb * nch * sizeof(uint64_t),
+                             hipMemcpyHostToDevice, s->stream));
+    hipLaunchKernelGGL(k_partition_emit_vc, dim3(nb), dim3(BLOCK), 0, s->stream,
+                       (const uint8_t*)bytes->ptr, (const uint32_t*)offsets->ptr, n, tile,
+                       nch, d_off, (uint32_t*)row_indexes_out->ptr);
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_hist);
+    (void)hipFree(d_off);
+    free(h_hist);
+    free(h_off);
+    return GPUE_OK;
 }
 
 

@@ -125,6 +125,8 @@ def _declare(lib):
         "gpue_q43_star_agg_accum_async": (c_i32, [c_vp] * 5 + [c_vp] * 6 + [c_u64, c_vp]),
         "gpue_q3_probe_accum": (c_i32, [c_vp] + [c_vp] * 4 + [c_u64, c_vp, c_i32, c_vp]),
         "gpue_partition_xxh3_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
+        "gpue_partition_crc_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
+        "gpue_partition_varchar": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
         "gpue_gather_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_gather_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_partition_i64": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
@@ -775,6 +777,23 @@ class Engine:
 
     def ingest_destroy(self, ing):
         self._lib.gpue_ingest_destroy(ing)
+
+    def partition_crc(self, keys: DBuf, n, num_channels, row_indexes: DBuf) -> np.ndarray:
+        """Bucket-shuffle hash path (zlib crc32 seed 0,
+        exchange_sink_operator.cpp:617-622)."""
+        sp = np.zeros(num_channels + 1, dtype=np.uint64)
+        _ck(self._lib, self._lib.gpue_partition_crc_i32(
+            self._h, keys._h, n, num_channels, sp.ctypes.data_as(c_vp), row_indexes._h))
+        return sp
+
+    def partition_varchar(self, bytes_: DBuf, offsets: DBuf, n, num_channels,
+                          row_indexes: DBuf) -> np.ndarray:
+        """Varchar partition key (FNV over the slice bytes)."""
+        sp = np.zeros(num_channels + 1, dtype=np.uint64)
+        _ck(self._lib, self._lib.gpue_partition_varchar(
+            self._h, bytes_._h, offsets._h, n, num_channels,
+            sp.ctypes.data_as(c_vp), row_indexes._h))
+        return sp
 
     def partition_xxh3(self, keys: DBuf, n, num_channels, row_indexes: DBuf) -> np.ndarray:
         """Version-1 exchange hash (xxh3) partition
