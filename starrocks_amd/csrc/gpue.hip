@@ -778,11 +778,16 @@ __global__ void k_filter_emit(const int64_t* __restrict__ in, uint64_t n, int64_
 // sets an error flag and exits (host returns GPUE_ERR_HIP).
 // ---------------------------------------------------------------------------
 static constexpr uint32_t FILT_ITEMS = 8;                     // rows per thread per tile
+                                                              // (ITEMS template below
+                                                              // sweeps 8/16/32)
 static constexpr uint32_t FILT_TILE = BLOCK * FILT_ITEMS;     // 2048 rows
 static constexpr unsigned long long FILT_AGG = 1ull << 62;    // aggregate available
 static constexpr unsigned long long FILT_PREFIX = 2ull << 62; // inclusive prefix available
 static constexpr unsigned long long FILT_CNT_MASK = (1ull << 62) - 1;
 
+template <int ITEMS> // rows per thread per tile: larger tiles -> fewer
+                     // lookback pipeline stages (the per-tile publish/observe
+                     // latency chain dominates at 488 K tiles)
 __global__ void k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, int64_t theta,
                                   int64_t* __restrict__ out,
                                   unsigned long long* __restrict__ tile_desc,
@@ -794,7 +799,8 @@ __global__ void k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, in
     __shared__ uint64_t sh_tile;
     __shared__ unsigned long long sh_excl;
     __shared__ uint32_t wsum[BLOCK / WAVE];
-    const uint64_t n_tiles = (n + FILT_TILE - 1) / FILT_TILE;
+    const uint64_t TILE = (uint64_t)BLOCK * ITEMS;
+    const uint64_t n_tiles = (n + TILE - 1) / TILE;
     int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
     (void)ticket;
     (void)sh_tile;
@@ -805,23 +811,23 @@ __global__ void k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, in
     // tiles). The bounded spin turns any residency miscount into an error
     // return instead of a hang.
     for (uint64_t t = blockIdx.x; t < n_tiles; t += gridDim.x) {
-        uint64_t lo = t * FILT_TILE;
-        uint64_t hi = min(lo + (uint64_t)FILT_TILE, n);
-        uint64_t my = lo + (uint64_t)threadIdx.x * FILT_ITEMS;
-        int64_t v[FILT_ITEMS];
-        uint32_t pm = 0; // predicate bitmask over my 8 rows
-        if (my + FILT_ITEMS <= hi) {
+        uint64_t lo = t * TILE;
+        uint64_t hi = min(lo + TILE, n);
+        uint64_t my = lo + (uint64_t)threadIdx.x * ITEMS;
+        int64_t v[ITEMS];
+        uint32_t pm = 0; // predicate bitmask over my rows (ITEMS <= 32)
+        if (my + ITEMS <= hi) {
             const longlong2* p2 = (const longlong2*)(in + my);
             #pragma unroll
-            for (int j = 0; j < (int)FILT_ITEMS / 2; j++) {
+            for (int j = 0; j < ITEMS / 2; j++) {
                 longlong2 w = p2[j];
                 v[2 * j] = w.x;
                 v[2 * j + 1] = w.y;
             }
             #pragma unroll
-            for (int j = 0; j < (int)FILT_ITEMS; j++) pm |= (v[j] < theta) << j;
+            for (int j = 0; j < ITEMS; j++) pm |= (v[j] < theta) << j;
         } else {
-            for (int j = 0; j < (int)FILT_ITEMS; j++) {
+            for (int j = 0; j < ITEMS; j++) {
                 uint64_t i = my + j;
                 v[j] = (i < hi) ? in[i] : 0;
                 pm |= ((i < hi) && (v[j] < theta)) << j;
@@ -906,7 +912,7 @@ __global__ void k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, in
         __syncthreads();
         uint64_t w = sh_excl + wave_base + my_excl;
         #pragma unroll
-        for (int j = 0; j < (int)FILT_ITEMS; j++)
+        for (int j = 0; j < ITEMS; j++)
             if (pm & (1u << j)) out[w++] = v[j];
         __syncthreads();
     }
@@ -917,26 +923,30 @@ extern "C" int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64
 int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64_t theta,
                                gpue_dbuf* out, uint64_t* out_count) {
     ARG_CHECK(s && in && out && out_count && in->bytes >= n * 8 && n > 0);
-    uint64_t n_tiles = (n + FILT_TILE - 1) / FILT_TILE;
+    int items = env_cap("GPUE_FILT_ITEMS", 16); // tile-size sweep (8/16/32)
+    if (items != 8 && items != 16 && items != 32) items = 16;
+    uint64_t tile = (uint64_t)BLOCK * items;
+    uint64_t n_tiles = (n + tile - 1) / tile;
     unsigned long long* d_desc = nullptr;
     unsigned long long* d_misc = nullptr; // {ticket(unused), total, error}
     HIP_CHECK(hipMalloc(&d_desc, n_tiles * 8));
     HIP_CHECK(hipMalloc(&d_misc, 3 * 8));
     HIP_CHECK(hipMemsetAsync(d_desc, 0, n_tiles * 8, s->stream));
     HIP_CHECK(hipMemsetAsync(d_misc, 0, 3 * 8, s->stream));
+    auto kern = items == 8 ? k_filter_lookback<8>
+               : items == 32 ? k_filter_lookback<32> : k_filter_lookback<16>;
     // fully-resident grid for the static-assignment lookback (see kernel
     // comment); the occupancy API can over-report by one block per CU on
     // SGPR-heavy 256-thread kernels (MI355X_MICROARCH.md) — subtract one
     int blocks_per_cu = 0;
-    HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(&blocks_per_cu,
-                                                           k_filter_lookback, BLOCK, 0));
+    HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(&blocks_per_cu, kern, BLOCK, 0));
     if (blocks_per_cu > 1) blocks_per_cu -= 1;
     if (blocks_per_cu < 1) blocks_per_cu = 1;
     hipDeviceProp_t props;
     HIP_CHECK(hipGetDeviceProperties(&props, s->device));
     uint64_t resident = (uint64_t)props.multiProcessorCount * blocks_per_cu;
     uint32_t nb = (uint32_t)(n_tiles < resident ? n_tiles : resident);
-    hipLaunchKernelGGL(k_filter_lookback, dim3(nb), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(kern, dim3(nb), dim3(BLOCK), 0, s->stream,
                        (const int64_t*)in->ptr, n, theta, (int64_t*)out->ptr, d_desc,
                        d_misc, d_misc + 1, d_misc + 2);
     unsigned long long h_misc[3];
