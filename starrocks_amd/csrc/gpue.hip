@@ -785,10 +785,12 @@ static constexpr unsigned long long FILT_AGG = 1ull << 62;    // aggregate avail
 static constexpr unsigned long long FILT_PREFIX = 2ull << 62; // inclusive prefix available
 static constexpr unsigned long long FILT_CNT_MASK = (1ull << 62) - 1;
 
-template <int ITEMS> // rows per thread per tile: larger tiles -> fewer
-                     // lookback pipeline stages (the per-tile publish/observe
-                     // latency chain dominates at 488 K tiles)
-__global__ void k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, int64_t theta,
+template <int ITEMS, int TPB = BLOCK> // rows/thread and threads/block:
+                     // larger tiles -> fewer lookback pipeline stages (the
+                     // per-tile publish/observe latency chain dominates at
+                     // 488 K tiles)
+__global__ __launch_bounds__(TPB) void
+k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, int64_t theta,
                                   int64_t* __restrict__ out,
                                   unsigned long long* __restrict__ tile_desc,
                                   unsigned long long* __restrict__ ticket,
@@ -798,8 +800,8 @@ __global__ void k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, in
     // order == thread order == row order; ONE block scan per tile.
     __shared__ uint64_t sh_tile;
     __shared__ unsigned long long sh_excl;
-    __shared__ uint32_t wsum[BLOCK / WAVE];
-    const uint64_t TILE = (uint64_t)BLOCK * ITEMS;
+    __shared__ uint32_t wsum[TPB / WAVE];
+    const uint64_t TILE = (uint64_t)TPB * ITEMS;
     const uint64_t n_tiles = (n + TILE - 1) / TILE;
     int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
     (void)ticket;
@@ -848,7 +850,7 @@ __global__ void k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, in
         uint32_t wave_base = 0;
         for (int w = 0; w < wid; w++) wave_base += wsum[w];
         uint32_t tile_count = 0;
-        for (int w = 0; w < BLOCK / WAVE; w++) tile_count += wsum[w];
+        for (int w = 0; w < TPB / WAVE; w++) tile_count += wsum[w];
         (void)wave_total;
         // WAVE-PARALLEL lookback (wave 0): 64 predecessor descriptors per
         // step — a serial walk scans O(grid) aggregates per tile and
@@ -928,7 +930,9 @@ int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64
     // 4x fewer lookback pipeline stages
     int items = env_cap("GPUE_FILT_ITEMS", 32);
     if (items != 8 && items != 16 && items != 32) items = 16;
-    uint64_t tile = (uint64_t)BLOCK * items;
+    int tpb = env_cap("GPUE_FILT_TPB", BLOCK);
+    if (tpb != 256 && tpb != 512 && tpb != 1024) tpb = BLOCK;
+    uint64_t tile = (uint64_t)tpb * items;
     uint64_t n_tiles = (n + tile - 1) / tile;
     unsigned long long* d_desc = nullptr;
     unsigned long long* d_misc = nullptr; // {ticket(unused), total, error}
@@ -936,20 +940,24 @@ int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64
     HIP_CHECK(hipMalloc(&d_misc, 3 * 8));
     HIP_CHECK(hipMemsetAsync(d_desc, 0, n_tiles * 8, s->stream));
     HIP_CHECK(hipMemsetAsync(d_misc, 0, 3 * 8, s->stream));
-    auto kern = items == 8 ? k_filter_lookback<8>
-               : items == 32 ? k_filter_lookback<32> : k_filter_lookback<16>;
+    auto kern = items == 8 ? k_filter_lookback<8, 256>
+               : items == 32 ? k_filter_lookback<32, 256> : k_filter_lookback<16, 256>;
+    if (tpb == 512) kern = items == 32 ? k_filter_lookback<32, 512>
+                                       : k_filter_lookback<16, 512>;
+    else if (tpb == 1024) kern = items == 32 ? k_filter_lookback<32, 1024>
+                                             : k_filter_lookback<16, 1024>;
     // fully-resident grid for the static-assignment lookback (see kernel
     // comment); the occupancy API can over-report by one block per CU on
     // SGPR-heavy 256-thread kernels (MI355X_MICROARCH.md) — subtract one
     int blocks_per_cu = 0;
-    HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(&blocks_per_cu, kern, BLOCK, 0));
+    HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(&blocks_per_cu, kern, tpb, 0));
     if (blocks_per_cu > 1) blocks_per_cu -= 1;
     if (blocks_per_cu < 1) blocks_per_cu = 1;
     hipDeviceProp_t props;
     HIP_CHECK(hipGetDeviceProperties(&props, s->device));
     uint64_t resident = (uint64_t)props.multiProcessorCount * blocks_per_cu;
     uint32_t nb = (uint32_t)(n_tiles < resident ? n_tiles : resident);
-    hipLaunchKernelGGL(kern, dim3(nb), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(kern, dim3(nb), dim3(tpb), 0, s->stream,
                        (const int64_t*)in->ptr, n, theta, (int64_t*)out->ptr, d_desc,
                        d_misc, d_misc + 1, d_misc + 2);
     unsigned long long h_misc[3];
