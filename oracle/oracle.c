@@ -2399,7 +2399,11 @@ static size_t lz4_compress_block(const uint8_t* src, size_t n, uint8_t* dst) {
     static uint32_t table[1 << HASH_LOG];
     memset(table, 0, sizeof(table));
     size_t s = 0, d = 0, anchor = 0;
-    while (n >= MINMATCH + 1 && s + MINMATCH + 5 < n) { /* keep the >=5-literal tail rule */
+    /* LZ4 block end rules: the last 5 bytes are literals AND a match must
+     * START at least 12 bytes before the block end (lz4_Block_format.md
+     * "End of block restrictions") — r01 enforced only the first, producing
+     * streams the upstream decoder rejects (caught by the pyarrow pin). */
+    while (n >= 13 && s + 12 < n) {
         uint32_t seq;
         memcpy(&seq, src + s, 4);
         uint32_t h = (seq * 2654435761u) >> (32 - HASH_LOG);
@@ -2462,6 +2466,14 @@ static size_t lz4_compress_block(const uint8_t* src, size_t n, uint8_t* dst) {
     d += lit;
     return d;
 }
+
+/* extern wrappers over the static block codec so tests can pin the LZ4
+ * BLOCK layer against a published implementation (pyarrow's bundled
+ * lz4_raw codec) — the compression half of the page format's
+ * byte-compatibility (DESIGN.md §4c "one unpinned surface"). */
+uint64_t orc_lz4_compress_block(const uint8_t* src, uint64_t n, uint8_t* dst);
+uint64_t orc_lz4_decompress_block(const uint8_t* src, uint64_t comp_n, uint8_t* dst,
+                                  uint64_t cap);
 
 static size_t lz4_decompress_block(const uint8_t* src, size_t comp_n, uint8_t* dst,
                                    size_t dst_cap) {
@@ -2538,3 +2550,12 @@ uint64_t orc_bshuf_lz4_decode_i32(const uint8_t* in, uint32_t n, int32_t* values
 }
 
 void orc_free(void* p) { free(p); }
+
+uint64_t orc_lz4_compress_block(const uint8_t* src, uint64_t n, uint8_t* dst) {
+    return lz4_compress_block(src, n, dst);
+}
+
+uint64_t orc_lz4_decompress_block(const uint8_t* src, uint64_t comp_n, uint8_t* dst,
+                                  uint64_t cap) {
+    return lz4_decompress_block(src, comp_n, dst, cap);
+}

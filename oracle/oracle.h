@@ -331,3 +331,7 @@ void orc_partition_channel_crc_u32(const uint32_t* keys, uint64_t n,
 void orc_partition_channel_fnv_slice(const uint8_t* bytes, const uint32_t* offsets,
                                      uint64_t n, uint32_t num_channels,
                                      uint32_t* channel_ids);
+/* raw LZ4 block codec (pinned against pyarrow's bundled lz4_raw in tests) */
+uint64_t orc_lz4_compress_block(const uint8_t* src, uint64_t n, uint8_t* dst);
+uint64_t orc_lz4_decompress_block(const uint8_t* src, uint64_t comp_n, uint8_t* dst,
+                                  uint64_t cap);

@@ -271,3 +271,41 @@ def test_dict_varchar_full_chain_gpu(engine):
     t.destroy()
     for x in (dp, db, do, cp, cb, ob, oo, hb, ho):
         x.free()
+
+
+def test_lz4_block_vs_pyarrow():
+    """Pin the LZ4 BLOCK layer of the bitshuffle+LZ4 page against a
+    PUBLISHED implementation (pyarrow's bundled lz4_raw codec) in both
+    directions — closing the compression half of the format's
+    byte-compatibility gap (the bitshuffle bit-transpose half remains a
+    spec restatement; the library is absent offline)."""
+    pa = pytest.importorskip("pyarrow")
+    import ctypes
+    lib = orc.load()
+    lib.orc_lz4_compress_block.restype = np.uint64 and ctypes.c_uint64
+    lib.orc_lz4_compress_block.argtypes = [ctypes.c_void_p, ctypes.c_uint64,
+                                           ctypes.c_void_p]
+    lib.orc_lz4_decompress_block.restype = ctypes.c_uint64
+    lib.orc_lz4_decompress_block.argtypes = [ctypes.c_void_p, ctypes.c_uint64,
+                                             ctypes.c_void_p, ctypes.c_uint64]
+    codec = pa.Codec("lz4_raw")
+    rng = np.random.default_rng(31)
+    cases = [b"", b"a", b"ab" * 500,
+             bytes(rng.integers(0, 4, 5000, dtype=np.uint8)),   # compressible
+             bytes(rng.integers(0, 256, 5000, dtype=np.uint8)), # incompressible
+             bytes(rng.integers(0, 256, 17, dtype=np.uint8)),
+             b"\x00" * 8192]
+    for data in cases:
+        if len(data) == 0:
+            continue
+        # ours -> pyarrow
+        out = np.zeros(len(data) * 2 + 64, np.uint8)
+        nb = lib.orc_lz4_compress_block(data, len(data), out.ctypes.data)
+        assert nb > 0
+        got = bytes(codec.decompress(out[:nb].tobytes(), len(data)))
+        assert got == data, (len(data), "ours->pyarrow")
+        # pyarrow -> ours
+        comp = bytes(codec.compress(data))
+        dec = np.zeros(len(data) + 8, np.uint8)
+        m = lib.orc_lz4_decompress_block(comp, len(comp), dec.ctypes.data, len(dec))
+        assert m == len(data) and dec[:m].tobytes() == data, (len(data), "pyarrow->ours")
