@@ -2393,6 +2393,16 @@ static void bshuf_untranspose_block_i32(const uint8_t* in, uint32_t* out, uint32
     }
 }
 
+/* extern wrappers so tests can cross-check the bit-plane transpose against
+ * an INDEPENDENT numpy restatement of the published bitshuffle algorithm
+ * (two independent restatements agreeing is the strongest offline evidence
+ * available — the library itself is absent). */
+uint64_t orc_bshuf_transpose_i32(const int32_t* in, uint32_t elems, uint8_t* out);
+uint64_t orc_bshuf_transpose_i32(const int32_t* in, uint32_t elems, uint8_t* out) {
+    bshuf_transpose_block_i32((const uint32_t*)in, out, elems);
+    return (uint64_t)(elems / 8) * 32;
+}
+
 /* minimal LZ4 block compressor (format-correct greedy hash matcher) */
 static size_t lz4_compress_block(const uint8_t* src, size_t n, uint8_t* dst) {
     enum { HASH_LOG = 13, MINMATCH = 4 };

@@ -335,3 +335,6 @@ void orc_partition_channel_fnv_slice(const uint8_t* bytes, const uint32_t* offse
 uint64_t orc_lz4_compress_block(const uint8_t* src, uint64_t n, uint8_t* dst);
 uint64_t orc_lz4_decompress_block(const uint8_t* src, uint64_t comp_n, uint8_t* dst,
                                   uint64_t cap);
+/* bit-plane transpose stage (cross-checked vs an independent numpy
+ * restatement of published bitshuffle 0.5.1 in tests) */
+uint64_t orc_bshuf_transpose_i32(const int32_t* in, uint32_t elems, uint8_t* out);
