@@ -590,6 +590,11 @@ int gpue_page_decode_for_i32(gpue_session* s, gpue_dbuf* page, uint64_t n_values
  * u32 absolute-offset trailer + count). The dict page's dictionary format. */
 int gpue_page_decode_binary_plain(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
                                   gpue_dbuf* out_bytes, gpue_dbuf* out_offsets);
+/* BinaryPrefixPage -> BinaryColumn (PREFIX_ENCODING,
+ * binary_prefix_page.{h,cpp}: front coding, restart every 16 entries;
+ * decode parallelizes per restart group). */
+int gpue_page_decode_binary_prefix(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
+                                   gpue_dbuf* out_bytes, gpue_dbuf* out_offsets);
 
 /* ---- event timing on the session stream (bench roofline evidence) ---- */
 int gpue_timer_start(gpue_session* s);

@@ -1272,6 +1272,106 @@ uint64_t orc_rle_page_decode_i32(const uint8_t* page, int32_t* values) {
 }
 
 /* ====================================================================== */
+/* BinaryPrefixPage codec (PREFIX_ENCODING, binary_prefix_page.{h,cpp}):    */
+/* front coding with a restart point every 16 entries.                      */
+/*   Entry   := varint32 shared_len, varint32 unshared_len, unshared bytes  */
+/*              (shared_len forced 0 at restart entries)                    */
+/*   Trailer := u32 num_entries, u8 restart_interval (16),                  */
+/*              u32 restart_offset ^ num_restarts, u32 num_restarts         */
+/* ====================================================================== */
+
+static uint64_t put_varint32(uint8_t* out, uint64_t pos, uint32_t v) {
+    while (v >= 0x80) {
+        out[pos++] = (uint8_t)(v | 0x80);
+        v >>= 7;
+    }
+    out[pos++] = (uint8_t)v;
+    return pos;
+}
+
+static uint64_t get_varint32(const uint8_t* in, uint64_t pos, uint32_t* v) {
+    uint32_t r = 0;
+    int shift = 0;
+    for (;;) {
+        uint8_t b = in[pos++];
+        r |= (uint32_t)(b & 0x7F) << shift;
+        if (!(b & 0x80)) break;
+        shift += 7;
+    }
+    *v = r;
+    return pos;
+}
+
+uint64_t orc_binary_prefix_encode(const uint8_t* bytes, const uint32_t* offsets,
+                                  uint32_t n, uint8_t* out) {
+    uint64_t pos = 0;
+    uint32_t* restarts = (uint32_t*)malloc(((uint64_t)n / 16 + 2) * 4);
+    uint32_t nrestart = 0;
+    uint32_t last_off = 0, last_len = 0;
+    for (uint32_t e = 0; e < n; e++) {
+        const uint8_t* entry = bytes + offsets[e];
+        uint32_t entry_len = offsets[e + 1] - offsets[e];
+        uint32_t share = 0;
+        if (e % 16 == 0) {
+            restarts[nrestart++] = (uint32_t)pos;
+        } else {
+            uint32_t maxs = entry_len < last_len ? entry_len : last_len;
+            share = maxs;
+            for (uint32_t j = 0; j < maxs; j++)
+                if (entry[j] != bytes[last_off + j]) { share = j; break; }
+        }
+        uint32_t non_share = entry_len - share;
+        pos = put_varint32(out, pos, share);
+        pos = put_varint32(out, pos, non_share);
+        memcpy(out + pos, entry + share, non_share);
+        pos += non_share;
+        last_off = offsets[e];
+        last_len = entry_len;
+    }
+    memcpy(out + pos, &n, 4);
+    pos += 4;
+    out[pos++] = 16;
+    for (uint32_t i = 0; i < nrestart; i++) {
+        memcpy(out + pos, &restarts[i], 4);
+        pos += 4;
+    }
+    memcpy(out + pos, &nrestart, 4);
+    free(restarts);
+    return pos + 4;
+}
+
+/* decode the whole page to a BinaryColumn; returns n */
+uint64_t orc_binary_prefix_decode(const uint8_t* page, uint64_t page_bytes,
+                                  uint8_t* out_bytes, uint32_t* out_offsets) {
+    uint32_t nrestart;
+    memcpy(&nrestart, page + page_bytes - 4, 4);
+    uint64_t trailer = page_bytes - 4 - (uint64_t)nrestart * 4 - 1 - 4;
+    uint32_t n;
+    memcpy(&n, page + trailer, 4);
+    uint64_t pos = 0;
+    uint32_t w = 0;
+    out_offsets[0] = 0;
+    uint32_t prev_off = 0, prev_len = 0;
+    for (uint32_t e = 0; e < n; e++) {
+        uint32_t share, non_share;
+        pos = get_varint32(page, pos, &share);
+        pos = get_varint32(page, pos, &non_share);
+        uint32_t start = w;
+        if (out_bytes) {
+            memcpy(out_bytes + w, out_bytes + prev_off, share);
+            memcpy(out_bytes + w + share, page + pos, non_share);
+        }
+        w += share + non_share;
+        pos += non_share;
+        out_offsets[e + 1] = w;
+        prev_off = start;
+        prev_len = share + non_share;
+    }
+    (void)prev_len;
+    return n;
+}
+
+/* ====================================================================== */
 /* BinaryPlainPage codec (PLAIN_ENCODING, binary_plain_page.h:28-46):       */
 /* body = concatenated strings; trailer = one u32-LE ABSOLUTE start offset  */
 /* per string (offsets[0] == 0), then u32-LE num_elems. This is also the    */

@@ -338,3 +338,8 @@ uint64_t orc_lz4_decompress_block(const uint8_t* src, uint64_t comp_n, uint8_t* 
 /* bit-plane transpose stage (cross-checked vs an independent numpy
  * restatement of published bitshuffle 0.5.1 in tests) */
 uint64_t orc_bshuf_transpose_i32(const int32_t* in, uint32_t elems, uint8_t* out);
+/* BinaryPrefixPage codec (PREFIX_ENCODING, binary_prefix_page.{h,cpp}) */
+uint64_t orc_binary_prefix_encode(const uint8_t* bytes, const uint32_t* offsets,
+                                  uint32_t n, uint8_t* out);
+uint64_t orc_binary_prefix_decode(const uint8_t* page, uint64_t page_bytes,
+                                  uint8_t* out_bytes, uint32_t* out_offsets);

@@ -86,6 +86,10 @@ def _decl(lib):
     lib.orc_binary_plain_encode.argtypes = [c_vp, c_vp, u, c_vp]
     lib.orc_binary_plain_decode.restype = c_u64
     lib.orc_binary_plain_decode.argtypes = [c_vp, c_u64, c_vp, c_vp]
+    lib.orc_binary_prefix_encode.restype = c_u64
+    lib.orc_binary_prefix_encode.argtypes = [c_vp, c_vp, u, c_vp]
+    lib.orc_binary_prefix_decode.restype = c_u64
+    lib.orc_binary_prefix_decode.argtypes = [c_vp, c_u64, c_vp, c_vp]
     lib.orc_xxh3_64_4to8.restype = c_u64
     lib.orc_xxh3_64_4to8.argtypes = [c_vp, c_i32, c_u64]
     lib.orc_xxh3_hash_i32.argtypes = [c_vp, c_u64, c_vp]
@@ -782,3 +786,21 @@ def partition_channels_fnv_slice(bytes_: np.ndarray, offsets: np.ndarray,
         _p(np.ascontiguousarray(bytes_, np.uint8)),
         _p(np.ascontiguousarray(offsets, np.uint32)), n, num_channels, _p(ch))
     return ch
+
+
+def binary_prefix_encode(bytes_: np.ndarray, offsets: np.ndarray) -> np.ndarray:
+    n = len(offsets) - 1
+    out = np.zeros(int(offsets[-1]) * 2 + n * 12 + 64, np.uint8)
+    nb = load().orc_binary_prefix_encode(
+        _p(np.ascontiguousarray(bytes_, np.uint8)),
+        _p(np.ascontiguousarray(offsets, np.uint32)), n, _p(out))
+    return out[:nb].copy()
+
+
+def binary_prefix_decode(page: np.ndarray, n: int, total_bytes: int):
+    pg = np.ascontiguousarray(page, np.uint8)
+    b = np.zeros(max(total_bytes, 1), np.uint8)
+    o = np.zeros(n + 1, np.uint32)
+    got = load().orc_binary_prefix_decode(_p(pg), len(pg), _p(b), _p(o))
+    assert got == n
+    return b[:int(o[-1])], o

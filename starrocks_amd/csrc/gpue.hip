@@ -6961,6 +6961,144 @@ int gpue_page_decode_binary_plain(gpue_session* s, gpue_dbuf* page, uint64_t n_v
 }
 
 // ---------------------------------------------------------------------------
+// BinaryPrefixPage decode (PREFIX_ENCODING, binary_prefix_page.{h,cpp}):
+// front coding with a restart point every 16 entries — decode parallelizes
+// per restart GROUP. Phase A (thread per group) parses the varint entry
+// headers into per-entry {share_len, data_off} + lengths; the shared
+// block-scan turns lengths into BinaryColumn offsets; phase B (thread per
+// group) reconstructs bytes sequentially within its group (entry e copies
+// entry e-1's materialized prefix — an intra-thread dependency only, since
+// restarts reset the chain).
+// ---------------------------------------------------------------------------
+__device__ static inline uint64_t bp_varint(const uint8_t* p, uint64_t pos, uint32_t* v) {
+    uint32_t r = 0;
+    int shift = 0;
+    for (;;) {
+        uint8_t b = p[pos++];
+        r |= (uint32_t)(b & 0x7F) << shift;
+        if (!(b & 0x80)) break;
+        shift += 7;
+    }
+    *v = r;
+    return pos;
+}
+
+__global__ void k_bprefix_scan(const uint8_t* __restrict__ page, uint64_t restart_tab,
+                               uint32_t nrestart, uint32_t n,
+                               uint32_t* __restrict__ lens,
+                               uint2* __restrict__ meta) { // {share, data_off}
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; g < nrestart;
+         g += stride) {
+        uint32_t ro;
+        memcpy(&ro, page + restart_tab + g * 4, 4);
+        uint64_t pos = ro;
+        uint32_t e0 = (uint32_t)g * 16;
+        uint32_t e1 = min(e0 + 16, n);
+        for (uint32_t e = e0; e < e1; e++) {
+            uint32_t share, non_share;
+            pos = bp_varint(page, pos, &share);
+            pos = bp_varint(page, pos, &non_share);
+            lens[e] = share + non_share;
+            meta[e] = make_uint2(share, (uint32_t)pos);
+            pos += non_share;
+        }
+    }
+}
+
+__global__ void k_bprefix_offsets(const uint64_t* __restrict__ off64, uint32_t n,
+                                  uint32_t* __restrict__ out_offsets) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t e = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; e <= n; e += stride)
+        out_offsets[e] = (uint32_t)off64[e];
+}
+
+__global__ void k_bprefix_fill(const uint8_t* __restrict__ page, uint32_t nrestart,
+                               uint32_t n, const uint2* __restrict__ meta,
+                               const uint32_t* __restrict__ out_offsets,
+                               uint8_t* __restrict__ out_bytes) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; g < nrestart;
+         g += stride) {
+        uint32_t e0 = (uint32_t)g * 16;
+        uint32_t e1 = min(e0 + 16, n);
+        for (uint32_t e = e0; e < e1; e++) {
+            uint2 m = meta[e];
+            uint32_t w = out_offsets[e];
+            uint32_t prev = e > e0 ? out_offsets[e - 1] : 0;
+            for (uint32_t j = 0; j < m.x; j++) out_bytes[w + j] = out_bytes[prev + j];
+            uint32_t nsh = out_offsets[e + 1] - w - m.x;
+            for (uint32_t j = 0; j < nsh; j++) out_bytes[w + m.x + j] = page[m.y + j];
+        }
+    }
+}
+
+extern "C" int gpue_page_decode_binary_prefix(gpue_session* s, gpue_dbuf* page,
+                                              uint64_t n_values, gpue_dbuf* out_bytes,
+                                              gpue_dbuf* out_offsets);
+int gpue_page_decode_binary_prefix(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
+                                   gpue_dbuf* out_bytes, gpue_dbuf* out_offsets) {
+    ARG_CHECK(s && page && out_bytes && out_offsets && page->bytes >= 13);
+    ARG_CHECK(out_offsets->bytes >= (n_values + 1) * 4);
+    uint32_t nrestart = 0;
+    HIP_CHECK(hipMemcpyAsync(&nrestart, (const uint8_t*)page->ptr + page->bytes - 4, 4,
+                             hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    uint64_t restart_tab = page->bytes - 4 - (uint64_t)nrestart * 4;
+    uint64_t trailer = restart_tab - 1 - 4;
+    uint32_t n = 0;
+    HIP_CHECK(hipMemcpyAsync(&n, (const uint8_t*)page->ptr + trailer, 4,
+                             hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    if (n != n_values || nrestart != (n + 15) / 16) {
+        snprintf(g_err, sizeof(g_err), "binary_prefix page: n=%u restarts=%u want %llu",
+                 n, nrestart, (unsigned long long)n_values);
+        return GPUE_ERR_ARG;
+    }
+    if (n == 0) return GPUE_OK;
+    uint32_t* d_lens = nullptr;
+    uint2* d_meta = nullptr;
+    uint64_t *d_bsums = nullptr, *d_off64 = nullptr;
+    HIP_CHECK(hipMalloc(&d_lens, (uint64_t)n * 4));
+    HIP_CHECK(hipMalloc(&d_meta, (uint64_t)n * 8));
+    uint32_t nb = grid_for(n);
+    uint64_t tile = ((uint64_t)n + nb - 1) / nb;
+    HIP_CHECK(hipMalloc(&d_bsums, (nb + 1) * 8));
+    HIP_CHECK(hipMalloc(&d_off64, ((uint64_t)n + 1) * 8));
+    hipLaunchKernelGGL(k_bprefix_scan, dim3(grid_for(nrestart)), dim3(BLOCK), 0, s->stream,
+                       (const uint8_t*)page->ptr, restart_tab, nrestart, n, d_lens, d_meta);
+    hipLaunchKernelGGL(k_block_sums_u32, dim3(nb), dim3(BLOCK), 0, s->stream, d_lens, n,
+                       tile, d_bsums);
+    hipLaunchKernelGGL(k_scan_small, dim3(1), dim3(1), 0, s->stream, d_bsums, nb);
+    hipLaunchKernelGGL(k_scan_offsets, dim3(nb), dim3(BLOCK), 0, s->stream, d_lens, n, tile,
+                       d_bsums, d_off64);
+    uint64_t total = 0;
+    HIP_CHECK(hipMemcpyAsync(&total, d_bsums + nb, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    int rc = GPUE_OK;
+    if (out_bytes->bytes < total && total > 0) {
+        snprintf(g_err, sizeof(g_err), "binary_prefix: out_bytes %llu < total %llu",
+                 (unsigned long long)out_bytes->bytes, (unsigned long long)total);
+        rc = GPUE_ERR_ARG;
+    } else {
+        // d_off64 holds per-entry EXCLUSIVE offsets; write u32 offsets[ n+1 ]
+        HIP_CHECK(hipMemcpyAsync(d_off64 + n, d_bsums + nb, 8, hipMemcpyDeviceToDevice,
+                                 s->stream));
+        hipLaunchKernelGGL(k_bprefix_offsets, dim3(grid_for(n + 1)), dim3(BLOCK), 0,
+                           s->stream, d_off64, n, (uint32_t*)out_offsets->ptr);
+        hipLaunchKernelGGL(k_bprefix_fill, dim3(grid_for(nrestart)), dim3(BLOCK), 0,
+                           s->stream, (const uint8_t*)page->ptr, nrestart, n, d_meta,
+                           (const uint32_t*)out_offsets->ptr, (uint8_t*)out_bytes->ptr);
+        HIP_CHECK(hipStreamSynchronize(s->stream));
+    }
+    (void)hipFree(d_lens);
+    (void)hipFree(d_meta);
+    (void)hipFree(d_bsums);
+    (void)hipFree(d_off64);
+    return rc;
+}
+
+// ---------------------------------------------------------------------------
 // Frame-of-reference page decode for int32 (FOR_ENCODING,
 // frame_of_reference_page.h over frame_of_reference_coding.{h,cpp}; format
 // and the decoder-authoritative frame-advance documented in oracle.c).

@@ -187,6 +187,7 @@ def _declare(lib):
         "gpue_page_decode_rle_bool": (c_i32, [c_vp, c_vp, c_u64, c_vp]),
         "gpue_page_decode_for_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp]),
         "gpue_page_decode_binary_plain": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_vp]),
+        "gpue_page_decode_binary_prefix": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_vp]),
         "gpue_ubench_bitgather": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_u64, c_i32,
                                           ctypes.POINTER(ctypes.c_float)]),
         "gpue_page_decode_bshuf_lz4_i32": (c_i32, [c_vp, c_vp, c_u32, c_vp]),
@@ -815,6 +816,12 @@ class Engine:
         _ck(self._lib, self._lib.gpue_ubench_bitgather(
             self._h, idx._h, n, bits._h, nbits_pow2, reps, ctypes.byref(ms)))
         return ms.value
+
+    def page_decode_binary_prefix(self, page: DBuf, n_values, out_bytes: DBuf,
+                                  out_offsets: DBuf):
+        """BinaryPrefixPage (front coding, restart every 16) -> BinaryColumn."""
+        _ck(self._lib, self._lib.gpue_page_decode_binary_prefix(
+            self._h, page._h, n_values, out_bytes._h, out_offsets._h))
 
     def page_decode_binary_plain(self, page: DBuf, n_values, out_bytes: DBuf,
                                  out_offsets: DBuf):

@@ -338,3 +338,47 @@ def test_bitshuffle_transpose_vs_independent_numpy():
         nb = lib.orc_bshuf_transpose_i32(vals.ctypes.data, elems, out.ctypes.data)
         assert nb == elems * 4
         assert np.array_equal(out, numpy_bshuf_i32(vals)), elems
+
+
+def test_binary_prefix_page_oracle_roundtrip():
+    """BinaryPrefixPage (front coding, restart every 16 entries;
+    binary_prefix_page.{h,cpp}): sorted dictionary keys compress well and
+    roundtrip exactly, including empty strings and non-multiple-of-16
+    counts."""
+    rng = np.random.default_rng(19)
+    rows = sorted(f"prefix_{i:06d}_{rng.integers(100)}".encode()
+                  for i in range(1000)) + [b"", b"x"]
+    off = np.zeros(len(rows) + 1, np.uint32)
+    np.cumsum([len(r) for r in rows], out=off[1:])
+    bts = np.frombuffer(b"".join(rows), np.uint8).copy()
+    page = orc.binary_prefix_encode(bts, off)
+    assert len(page) < bts.nbytes // 2  # front coding pays on sorted keys
+    db, do = orc.binary_prefix_decode(page, len(rows), bts.nbytes)
+    assert np.array_equal(do, off) and np.array_equal(db, bts)
+
+
+@pytest.mark.gpu
+def test_binary_prefix_page_decode_gpu_parity(engine):
+    rng = np.random.default_rng(20)
+    cases = []
+    rows = sorted(f"key_{i:07d}".encode() + b"s" * int(rng.integers(0, 5))
+                  for i in range(50_000))
+    cases.append(rows)
+    cases.append([bytes(rng.integers(97, 123, int(rng.integers(0, 30)),
+                                     dtype=np.uint8)) for _ in range(997)])
+    for rows in cases:
+        off = np.zeros(len(rows) + 1, np.uint32)
+        np.cumsum([len(r) for r in rows], out=off[1:])
+        bts = np.frombuffer(b"".join(rows), np.uint8).copy()
+        page = orc.binary_prefix_encode(bts, off)
+        pb = engine.alloc(page.nbytes)
+        pb.h2d(page)
+        ob = engine.alloc(max(bts.nbytes, 1))
+        oo = engine.alloc((len(rows) + 1) * 4)
+        engine.page_decode_binary_prefix(pb, len(rows), ob, oo)
+        assert np.array_equal(oo.d2h(np.uint32, len(rows) + 1), off)
+        if bts.nbytes:
+            assert np.array_equal(ob.d2h(np.uint8, bts.nbytes), bts)
+        pb.free()
+        ob.free()
+        oo.free()
