@@ -920,6 +920,121 @@ k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, int64_t theta,
     }
 }
 
+// Re-read emit variant (GPUE_FILT_RR): keep only the 32-bit predicate mask
+// per thread — the emit re-reads matched rows from in[] (at s=0.01 that is
+// ~1% of lines, usually L2-resident) instead of holding ITEMS int64 values
+// in registers. The small register footprint doubles resident blocks per
+// CU, so one block's lookback walk overlaps another block's streaming.
+template <int ITEMS, int TPB>
+__global__ __launch_bounds__(TPB) void
+k_filter_lookback_rr(const int64_t* __restrict__ in, uint64_t n, int64_t theta,
+                     int64_t* __restrict__ out,
+                     unsigned long long* __restrict__ tile_desc,
+                     unsigned long long* __restrict__ ticket,
+                     unsigned long long* __restrict__ total_out,
+                     unsigned long long* __restrict__ error_out) {
+    __shared__ unsigned long long sh_excl;
+    __shared__ uint32_t wsum[TPB / WAVE];
+    const uint64_t TILE = (uint64_t)TPB * ITEMS;
+    const uint64_t n_tiles = (n + TILE - 1) / TILE;
+    int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+    (void)ticket;
+    for (uint64_t t = blockIdx.x; t < n_tiles; t += gridDim.x) {
+        uint64_t lo = t * TILE;
+        uint64_t hi = min(lo + TILE, n);
+        uint64_t my = lo + (uint64_t)threadIdx.x * ITEMS;
+        uint32_t pm = 0;
+        if (my + ITEMS <= hi) {
+            const longlong2* p2 = (const longlong2*)(in + my);
+            #pragma unroll
+            for (int j = 0; j < ITEMS / 2; j++) {
+                longlong2 w = p2[j];
+                pm |= (w.x < theta) << (2 * j);
+                pm |= (w.y < theta) << (2 * j + 1);
+            }
+        } else {
+            for (int j = 0; j < ITEMS; j++) {
+                uint64_t i = my + j;
+                pm |= ((i < hi) && (in[i] < theta)) << j;
+            }
+        }
+        uint32_t c = __popc(pm);
+        uint32_t pre = c;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            uint32_t up = __shfl_up(pre, off, WAVE);
+            if (lane >= off) pre += up;
+        }
+        uint32_t my_excl = pre - c;
+        if (lane == WAVE - 1) wsum[wid] = pre;
+        __syncthreads();
+        uint32_t wave_base = 0;
+        for (int w = 0; w < wid; w++) wave_base += wsum[w];
+        uint32_t tile_count = 0;
+        for (int w = 0; w < TPB / WAVE; w++) tile_count += wsum[w];
+        if (wid == 0) {
+            if (t == 0) {
+                if (lane == 0) {
+                    __hip_atomic_store(&tile_desc[0], FILT_PREFIX | tile_count,
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    sh_excl = 0;
+                }
+            } else {
+                if (lane == 0)
+                    __hip_atomic_store(&tile_desc[t], FILT_AGG | tile_count,
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                unsigned long long running = 0;
+                int64_t base = (int64_t)t - 1;
+                uint64_t spins = 0;
+                for (;;) {
+                    int64_t idx = base - lane;
+                    unsigned long long d =
+                        idx >= 0 ? __hip_atomic_load(&tile_desc[idx], __ATOMIC_RELAXED,
+                                                     __HIP_MEMORY_SCOPE_AGENT)
+                                 : FILT_PREFIX;
+                    unsigned long long flag = d & ~FILT_CNT_MASK;
+                    uint64_t prefix_mask = __ballot(flag == FILT_PREFIX);
+                    uint64_t invalid_mask = __ballot(flag == 0);
+                    int first_prefix = prefix_mask ? (__ffsll((unsigned long long)prefix_mask) - 1) : WAVE;
+                    int first_invalid = invalid_mask ? (__ffsll((unsigned long long)invalid_mask) - 1) : WAVE;
+                    if (first_prefix < first_invalid) {
+                        unsigned long long contrib =
+                            (lane <= first_prefix) ? (d & FILT_CNT_MASK) : 0;
+                        for (int off = WAVE / 2; off > 0; off >>= 1)
+                            contrib += __shfl_down(contrib, off, WAVE);
+                        running += __shfl(contrib, 0, WAVE);
+                        break;
+                    }
+                    if (first_invalid == WAVE) {
+                        unsigned long long contrib = d & FILT_CNT_MASK;
+                        for (int off = WAVE / 2; off > 0; off >>= 1)
+                            contrib += __shfl_down(contrib, off, WAVE);
+                        running += __shfl(contrib, 0, WAVE);
+                        base -= WAVE;
+                        continue;
+                    }
+                    if (++spins > (1ull << 28)) {
+                        if (lane == 0) atomicOr(error_out, 1ull);
+                        running = 0;
+                        break;
+                    }
+                }
+                if (lane == 0) {
+                    __hip_atomic_store(&tile_desc[t], FILT_PREFIX | (running + tile_count),
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    sh_excl = running;
+                }
+            }
+            if (lane == 0 && t == n_tiles - 1) *total_out = sh_excl + tile_count;
+        }
+        __syncthreads();
+        uint64_t w = sh_excl + wave_base + my_excl;
+        #pragma unroll
+        for (int j = 0; j < ITEMS; j++)
+            if (pm & (1u << j)) out[w++] = in[my + j]; // re-read (mostly L2)
+        __syncthreads();
+    }
+}
+
 extern "C" int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n,
                                           int64_t theta, gpue_dbuf* out, uint64_t* out_count);
 int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64_t theta,
@@ -949,6 +1064,14 @@ int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64
                                        : k_filter_lookback<16, 512>;
     else if (tpb == 1024) kern = items == 32 ? k_filter_lookback<32, 1024>
                                              : k_filter_lookback<16, 1024>;
+    const char* rr = getenv("GPUE_FILT_RR");
+    if (rr && atoi(rr)) {
+        kern = items == 32 ? k_filter_lookback_rr<32, 256> : k_filter_lookback_rr<16, 256>;
+        if (tpb == 512) kern = items == 32 ? k_filter_lookback_rr<32, 512>
+                                           : k_filter_lookback_rr<16, 512>;
+        else if (tpb == 1024) kern = items == 32 ? k_filter_lookback_rr<32, 1024>
+                                                 : k_filter_lookback_rr<16, 1024>;
+    }
     // fully-resident grid for the static-assignment lookback (see kernel
     // comment); the occupancy API can over-report by one block per CU on
     // SGPR-heavy 256-thread kernels (MI355X_MICROARCH.md) — subtract one
