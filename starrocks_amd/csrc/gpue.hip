@@ -920,11 +920,11 @@ k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, int64_t theta,
     }
 }
 
-// Re-read emit variant (GPUE_FILT_RR): keep only the 32-bit predicate mask
-// per thread — the emit re-reads matched rows from in[] (at s=0.01 that is
-// ~1% of lines, usually L2-resident) instead of holding ITEMS int64 values
-// in registers. The small register footprint doubles resident blocks per
-// CU, so one block's lookback walk overlaps another block's streaming.
+// Re-read emit variant (GPUE_FILT_RR) — MEASURED WORSE at every selectivity
+// (s=0.01: 2.83 vs 2.14 ms; s=0.5: 12.6 vs 4.45 — the emit re-reads miss L2
+// at an 8 GB footprint; profiles/r02_filter_items_sweep.log). Keeping the
+// 32 values in registers is the right design; this stays env-gated as the
+// recorded negative.
 template <int ITEMS, int TPB>
 __global__ __launch_bounds__(TPB) void
 k_filter_lookback_rr(const int64_t* __restrict__ in, uint64_t n, int64_t theta,
