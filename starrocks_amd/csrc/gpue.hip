@@ -3201,6 +3201,23 @@ __global__ void k_right_mark_rd(const int32_t* __restrict__ probe_keys, uint64_t
     }
 }
 
+// DENSE_RANGE_DIRECT right-mark: rank/select head + chain walk, no compare
+__global__ void k_right_mark_dense(const int32_t* __restrict__ probe_keys, uint64_t n,
+                                   int64_t mn, int64_t mx,
+                                   const uint2* __restrict__ groups,
+                                   const uint32_t* __restrict__ first,
+                                   const uint32_t* __restrict__ next,
+                                   uint32_t* __restrict__ matched_bits) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint32_t b = dense_head(probe_keys, i, mn, mx, groups, first);
+        while (b != 0) {
+            atomicOr(&matched_bits[b >> 5], 1u << (b & 31));
+            b = next[b];
+        }
+    }
+}
+
 __global__ void k_right_mark_keys(const uint32_t* __restrict__ probe_keys, uint64_t n,
                                   int is_linear, uint32_t log_bucket_size,
                                   uint32_t bucket_mask, const uint32_t* __restrict__ first,
@@ -3260,6 +3277,10 @@ int gpue_join_probe_right_i32(gpue_session* s, gpue_join_table* t, gpue_dbuf* pr
         hipLaunchKernelGGL(k_right_mark_rd, dim3(grid_for(n_rows)), dim3(BLOCK), 0, s->stream,
                            (const int32_t*)probe_keys->ptr, n_rows, t->min_key, t->max_key,
                            t->first, t->next, d_bits);
+    } else if (t->kind == gpue_join_table::DENSE_RANGE_DIRECT) {
+        hipLaunchKernelGGL(k_right_mark_dense, dim3(grid_for(n_rows)), dim3(BLOCK), 0,
+                           s->stream, (const int32_t*)probe_keys->ptr, n_rows, t->min_key,
+                           t->max_key, t->dense_groups, t->first, t->next, d_bits);
     } else {
         hipLaunchKernelGGL(k_right_mark_keys, dim3(grid_for(n_rows)), dim3(BLOCK), 0,
                            s->stream, (const uint32_t*)probe_keys->ptr, n_rows,

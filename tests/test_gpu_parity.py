@@ -1427,3 +1427,34 @@ def test_join_build_auto_dense_route(engine):
     kb.free()
     pb.free()
     t.destroy()
+
+
+def test_dense_right_semi_anti_parity(engine):
+    """RIGHT SEMI/ANTI over a DENSE_RANGE_DIRECT table (the probe pass marks
+    matched BUILD rows) must match the bucket-chained build's sets."""
+    rng = np.random.default_rng(81)
+    nbuild, nprobe = 60_000, 200_000
+    keys = np.concatenate([[0], rng.integers(1, 1_500_000, nbuild)]).astype(np.int32)
+    kb = engine.alloc(keys.nbytes)
+    kb.h2d(keys)
+    td = engine.join_build_dense_range_direct(kb, nbuild)
+    tb = engine.join_build_bucket_chained(kb, nbuild)
+    probe = rng.integers(1, 1_600_000, nprobe).astype(np.int32)
+    pb = engine.alloc(probe.nbytes)
+    pb.h2d(probe)
+    for anti in (0, 1):
+        cd = engine.join_probe_right(td, pb, nprobe, anti)
+        cb = engine.join_probe_right(tb, pb, nprobe, anti)
+        assert cd == cb, anti
+        od = engine.alloc(max(cd, 1) * 4)
+        ob2 = engine.alloc(max(cb, 1) * 4)
+        engine.join_probe_right(td, pb, nprobe, anti, od)
+        engine.join_probe_right(tb, pb, nprobe, anti, ob2)
+        assert np.array_equal(np.sort(od.d2h(np.uint32, cd)),
+                              np.sort(ob2.d2h(np.uint32, cb))), anti
+        od.free()
+        ob2.free()
+    td.destroy()
+    tb.destroy()
+    kb.free()
+    pb.free()
