@@ -1035,6 +1035,136 @@ k_filter_lookback_rr(const int64_t* __restrict__ in, uint64_t n, int64_t theta,
     }
 }
 
+// Software-pipelined lookback (GPUE_FILT_PIPE): per iteration the block
+// (1) counts + publishes AGG for the CURRENT tile, (2) ISSUES the next
+// tile's loads, (3) resolves the current prefix (wave-0 walk) and (4)
+// emits — so the ~7.5 us/tile walk+barrier stall hides under the next
+// tile's in-flight loads. Two register tile buffers (v[2][ITEMS]).
+template <int ITEMS, int TPB>
+__global__ __launch_bounds__(TPB) void
+k_filter_lookback_pipe(const int64_t* __restrict__ in, uint64_t n, int64_t theta,
+                       int64_t* __restrict__ out,
+                       unsigned long long* __restrict__ tile_desc,
+                       unsigned long long* __restrict__ ticket,
+                       unsigned long long* __restrict__ total_out,
+                       unsigned long long* __restrict__ error_out) {
+    __shared__ unsigned long long sh_excl;
+    __shared__ uint32_t wsum[TPB / WAVE];
+    const uint64_t TILE = (uint64_t)TPB * ITEMS;
+    const uint64_t n_tiles = (n + TILE - 1) / TILE;
+    int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+    (void)ticket;
+    int64_t v[2][ITEMS];
+    auto load_tile = [&](uint64_t t, int64_t* dst) {
+        uint64_t lo = t * TILE;
+        uint64_t hi = min(lo + TILE, n);
+        uint64_t my = lo + (uint64_t)threadIdx.x * ITEMS;
+        if (my + ITEMS <= hi) {
+            const longlong2* p2 = (const longlong2*)(in + my);
+            #pragma unroll
+            for (int j = 0; j < ITEMS / 2; j++) {
+                longlong2 w = p2[j];
+                dst[2 * j] = w.x;
+                dst[2 * j + 1] = w.y;
+            }
+        } else {
+            for (int j = 0; j < ITEMS; j++) {
+                uint64_t i = my + j;
+                dst[j] = (i < hi) ? in[i] : theta; // >= theta: never matches
+            }
+        }
+    };
+    uint64_t t = blockIdx.x;
+    if (t < n_tiles) load_tile(t, v[0]);
+    int cur = 0;
+    for (; t < n_tiles; t += gridDim.x, cur ^= 1) {
+        uint64_t lo = t * TILE;
+        uint64_t hi = min(lo + TILE, n);
+        uint64_t my = lo + (uint64_t)threadIdx.x * ITEMS;
+        uint32_t pm = 0;
+        #pragma unroll
+        for (int j = 0; j < ITEMS; j++)
+            pm |= ((my + j < hi) && (v[cur][j] < theta)) << j;
+        uint32_t c = __popc(pm);
+        uint32_t pre = c;
+        for (int off = 1; off < WAVE; off <<= 1) {
+            uint32_t up = __shfl_up(pre, off, WAVE);
+            if (lane >= off) pre += up;
+        }
+        uint32_t my_excl = pre - c;
+        if (lane == WAVE - 1) wsum[wid] = pre;
+        __syncthreads();
+        uint32_t wave_base = 0;
+        for (int w = 0; w < wid; w++) wave_base += wsum[w];
+        uint32_t tile_count = 0;
+        for (int w = 0; w < TPB / WAVE; w++) tile_count += wsum[w];
+        // publish AGG immediately, then ISSUE next tile's loads so their
+        // latency overlaps the lookback walk below
+        if (wid == 0 && lane == 0 && t > 0)
+            __hip_atomic_store(&tile_desc[t], FILT_AGG | tile_count, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+        if (t + gridDim.x < n_tiles) load_tile(t + gridDim.x, v[cur ^ 1]);
+        if (wid == 0) {
+            if (t == 0) {
+                if (lane == 0) {
+                    __hip_atomic_store(&tile_desc[0], FILT_PREFIX | tile_count,
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    sh_excl = 0;
+                }
+            } else {
+                unsigned long long running = 0;
+                int64_t base = (int64_t)t - 1;
+                uint64_t spins = 0;
+                for (;;) {
+                    int64_t idx = base - lane;
+                    unsigned long long d =
+                        idx >= 0 ? __hip_atomic_load(&tile_desc[idx], __ATOMIC_RELAXED,
+                                                     __HIP_MEMORY_SCOPE_AGENT)
+                                 : FILT_PREFIX;
+                    unsigned long long flag = d & ~FILT_CNT_MASK;
+                    uint64_t prefix_mask = __ballot(flag == FILT_PREFIX);
+                    uint64_t invalid_mask = __ballot(flag == 0);
+                    int first_prefix = prefix_mask ? (__ffsll((unsigned long long)prefix_mask) - 1) : WAVE;
+                    int first_invalid = invalid_mask ? (__ffsll((unsigned long long)invalid_mask) - 1) : WAVE;
+                    if (first_prefix < first_invalid) {
+                        unsigned long long contrib =
+                            (lane <= first_prefix) ? (d & FILT_CNT_MASK) : 0;
+                        for (int off = WAVE / 2; off > 0; off >>= 1)
+                            contrib += __shfl_down(contrib, off, WAVE);
+                        running += __shfl(contrib, 0, WAVE);
+                        break;
+                    }
+                    if (first_invalid == WAVE) {
+                        unsigned long long contrib = d & FILT_CNT_MASK;
+                        for (int off = WAVE / 2; off > 0; off >>= 1)
+                            contrib += __shfl_down(contrib, off, WAVE);
+                        running += __shfl(contrib, 0, WAVE);
+                        base -= WAVE;
+                        continue;
+                    }
+                    if (++spins > (1ull << 28)) {
+                        if (lane == 0) atomicOr(error_out, 1ull);
+                        running = 0;
+                        break;
+                    }
+                }
+                if (lane == 0) {
+                    __hip_atomic_store(&tile_desc[t], FILT_PREFIX | (running + tile_count),
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    sh_excl = running;
+                }
+            }
+            if (lane == 0 && t == n_tiles - 1) *total_out = sh_excl + tile_count;
+        }
+        __syncthreads();
+        uint64_t w = sh_excl + wave_base + my_excl;
+        #pragma unroll
+        for (int j = 0; j < ITEMS; j++)
+            if (pm & (1u << j)) out[w++] = v[cur][j];
+        __syncthreads();
+    }
+}
+
 extern "C" int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n,
                                           int64_t theta, gpue_dbuf* out, uint64_t* out_count);
 int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64_t theta,
@@ -1064,6 +1194,15 @@ int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64
                                        : k_filter_lookback<16, 512>;
     else if (tpb == 1024) kern = items == 32 ? k_filter_lookback<32, 1024>
                                              : k_filter_lookback<16, 1024>;
+    const char* pipe = getenv("GPUE_FILT_PIPE");
+    if (pipe && atoi(pipe)) {
+        if (tpb == 1024) kern = items == 32 ? k_filter_lookback_pipe<32, 1024>
+                                            : k_filter_lookback_pipe<16, 1024>;
+        else if (tpb == 512) kern = items == 32 ? k_filter_lookback_pipe<32, 512>
+                                                : k_filter_lookback_pipe<16, 512>;
+        else kern = items == 32 ? k_filter_lookback_pipe<32, 256>
+                                : k_filter_lookback_pipe<16, 256>;
+    }
     const char* rr = getenv("GPUE_FILT_RR");
     if (rr && atoi(rr)) {
         kern = items == 32 ? k_filter_lookback_rr<32, 256> : k_filter_lookback_rr<16, 256>;
