@@ -1035,11 +1035,12 @@ k_filter_lookback_rr(const int64_t* __restrict__ in, uint64_t n, int64_t theta,
     }
 }
 
-// Software-pipelined lookback (GPUE_FILT_PIPE): per iteration the block
-// (1) counts + publishes AGG for the CURRENT tile, (2) ISSUES the next
-// tile's loads, (3) resolves the current prefix (wave-0 walk) and (4)
-// emits — so the ~7.5 us/tile walk+barrier stall hides under the next
-// tile's in-flight loads. Two register tile buffers (v[2][ITEMS]).
+// Software-pipelined lookback (GPUE_FILT_PIPE) — MEASURED NEGATIVE
+// (5.37 vs 2.14 ms at s=0.01, every TPB/ITEMS combo): the double tile
+// buffer (v[2][32] = 128 VGPRs + machinery) spills to scratch and the
+// scratch traffic swamps the ~7.5 us/tile walk stall it was meant to hide.
+// Kept env-gated as the recorded negative; the register-resident
+// single-buffer form (32 rows x 1024 threads) remains the winner.
 template <int ITEMS, int TPB>
 __global__ __launch_bounds__(TPB) void
 k_filter_lookback_pipe(const int64_t* __restrict__ in, uint64_t n, int64_t theta,
