@@ -240,3 +240,33 @@ def test_agg_table_grow_gpu(engine):
     for b in (ok_b, os_b, oc_b):
         b.free()
     engine.agg_table_destroy(at)
+
+
+def test_partition_channel_bounds_gpu(engine):
+    """MAX_CH=64 channels supported (the exchange's channel cap); 65 must
+    be a clean argument error."""
+    from starrocks_amd.engine import GpueError
+    n = 100_000
+    keys = engine.alloc(n * 4)
+    engine.gen_u32_mod(keys, 42, 9, 0, n, 0, 0)
+    ri = engine.alloc(n * 4)
+    sp = engine.partition(keys, n, 64, ri)
+    assert sp[-1] == n and len(sp) == 65
+    with pytest.raises(GpueError):
+        engine.partition(keys, n, 65, ri)
+    keys.free()
+    ri.free()
+
+
+def test_ingest_exact_chunk_multiple_gpu(engine):
+    """ingest push where bytes is an exact multiple of the staging chunk."""
+    import numpy as np
+    n = (8 << 20) // 4  # exactly 2 chunks of 4 MB
+    host = np.arange(n, dtype=np.int32)
+    dst = engine.alloc(n * 4)
+    ing = engine.ingest_create(4 << 20)
+    engine.ingest_push(ing, host, dst)
+    engine.ingest_sync(ing)
+    engine.ingest_destroy(ing)
+    assert np.array_equal(dst.d2h(np.int32, n), host)
+    dst.free()
