@@ -49,7 +49,6 @@ lib2.gpue_ubench_q21.argtypes = [c_vp, c_i32] + [c_vp] * 5 + [c_i64, c_u64, c_u6
 cols2 = [eng2.alloc(n * 4) for _ in range(4)]
 eng2.gen_lineorder_q21(42, 0, n, *cols2)
 pf_fold = np.zeros((1 << 19) // 32, np.uint32)
-set_idx = np.flatnonzero(np.kron(bits, np.uint32(1)))  # placeholder; fold below
 # fold: bit i of `bits` -> prefilter bit (i & (2^19-1))
 words = bits
 for w in range(len(words)):
