@@ -152,6 +152,15 @@ int gpue_join_build_linear_chained_u32(gpue_session* s, gpue_dbuf* keys /*u32, 1
  * (join_hash_map_helper.h:46-55), u64 build-key compares along chains. */
 int gpue_join_build_bucket_chained_u64(gpue_session* s, gpue_dbuf* keys /*u64, 1-based*/,
                                        uint64_t row_count, gpue_join_table** out);
+/* 16-byte keys (TYPE_LARGEINT / SERIALIZED_FIXED_SIZE_LARGEINT packing):
+ * the generic JoinKeyHash<T,16> — crc32 over the key bytes, CRC_SEED,
+ * masked by bucket_size-1 (join_hash_map_helper.h:23-30). */
+int gpue_join_build_bucket_chained_u128(gpue_session* s, gpue_dbuf* keys /*16 B, 1-based*/,
+                                        uint64_t row_count, gpue_join_table** out);
+int gpue_join_probe_emit_mode_u128(gpue_session* s, gpue_join_table* t,
+                                   gpue_dbuf* probe_keys, uint64_t n_rows, int mode,
+                                   gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                                   uint64_t* match_count);
 int gpue_join_probe_emit_mode_u64(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
                                   uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
                                   gpue_dbuf* out_build_idx, uint64_t* match_count);

@@ -149,7 +149,7 @@ struct gpue_join_table {
     // method discriminator — the GPU analog of JoinHashMapSelector's choice
     // (reference join_hash_table.cpp:164-344)
     enum Kind { PAYLOAD = 0, RANGE_DIRECT = 1, BUCKET_CHAINED = 2,
-                DENSE_RANGE_DIRECT = 7,
+                DENSE_RANGE_DIRECT = 7, BUCKET_CHAINED128 = 8,
                 LINEAR_CHAINED = 3, VARCHAR = 4, BUCKET_CHAINED64 = 5 } kind = PAYLOAD;
     uint32_t log_bucket_size = 0;
     uint32_t* build_keys = nullptr; // chained methods keep the build keys for the
@@ -158,6 +158,7 @@ struct gpue_join_table {
     uint32_t* key_offsets = nullptr;
     uint8_t* key_nulls = nullptr;   // VARCHAR nullable: is_nulls (1-based), or null
     uint64_t* build_keys64 = nullptr; // BUCKET_CHAINED64: 8-byte build keys
+    ulonglong2* build_keys128 = nullptr; // BUCKET_CHAINED128: 16-byte build keys
 };
 
 int gpue_device_count(int* out) {
@@ -1492,6 +1493,7 @@ void gpue_join_table_destroy(gpue_join_table* t) {
     if (t->key_offsets) (void)hipFree(t->key_offsets);
     if (t->key_nulls) (void)hipFree(t->key_nulls);
     if (t->build_keys64) (void)hipFree(t->build_keys64);
+    if (t->build_keys128) (void)hipFree(t->build_keys128);
     delete t;
 }
 
@@ -3035,6 +3037,158 @@ int gpue_join_probe_emit_varchar_nulls(gpue_session* s, gpue_join_table* t, gpue
                            mode == 6 ? 2 : mode, d_counts,
                            d_offsets, (uint32_t*)out_probe_idx->ptr,
                            (uint32_t*)out_build_idx->ptr);
+        HIP_CHECK(hipStreamSynchronize(s->stream));
+        (void)hipFree(d_offsets);
+    }
+    (void)hipFree(d_counts);
+    (void)hipFree(d_bsums);
+    return GPUE_OK;
+}
+
+// ---------------------------------------------------------------------------
+// 16-byte (LARGEINT / SERIALIZED_FIXED_SIZE_LARGEINT) bucket-chained join —
+// the generic JoinKeyHash<T,16> path: crc_hash_32 over the 16 key bytes,
+// seed CRC_SEED, masked by (bucket_size-1) (join_hash_map_helper.h:23-30).
+// Functionally equal to the varchar path with fixed 16-byte slices (the
+// oracle pin reuses exactly that), minus the per-row offsets indirection.
+// ---------------------------------------------------------------------------
+__global__ void k_build_bucket_chained128(const ulonglong2* __restrict__ keys,
+                                          uint64_t row_count, uint32_t bucket_mask,
+                                          uint32_t* __restrict__ first,
+                                          uint32_t* __restrict__ next) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i <= row_count;
+         i += stride) {
+        ulonglong2 k = keys[i];
+        uint32_t b = crc_hash_32_dev((const uint8_t*)&k, 16, 0x811C9DC5u) & bucket_mask;
+        next[i] = atomicExch(&first[b], (uint32_t)i);
+    }
+}
+
+__global__ void k_probe_count_u128(const ulonglong2* __restrict__ probe_keys, uint64_t n,
+                                   uint32_t bucket_mask, const uint32_t* __restrict__ first,
+                                   const uint32_t* __restrict__ next,
+                                   const ulonglong2* __restrict__ build_keys, int mode,
+                                   uint32_t* __restrict__ row_counts) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        ulonglong2 k = probe_keys[i];
+        uint32_t b = first[crc_hash_32_dev((const uint8_t*)&k, 16, 0x811C9DC5u) & bucket_mask];
+        uint32_t c = 0;
+        while (b != 0) {
+            ulonglong2 bk = build_keys[b];
+            c += (bk.x == k.x) & (bk.y == k.y);
+            b = next[b];
+        }
+        row_counts[i] = join_mode_count(c, mode);
+    }
+}
+
+__global__ void k_probe_emit_u128(const ulonglong2* __restrict__ probe_keys, uint64_t n,
+                                  uint32_t bucket_mask, const uint32_t* __restrict__ first,
+                                  const uint32_t* __restrict__ next,
+                                  const ulonglong2* __restrict__ build_keys, int mode,
+                                  const uint32_t* __restrict__ row_counts,
+                                  const uint64_t* __restrict__ row_offsets,
+                                  uint32_t* __restrict__ out_probe,
+                                  uint32_t* __restrict__ out_build) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        if (row_counts[i] == 0) continue;
+        uint64_t pos = row_offsets[i];
+        ulonglong2 k = probe_keys[i];
+        uint32_t b = first[crc_hash_32_dev((const uint8_t*)&k, 16, 0x811C9DC5u) & bucket_mask];
+        bool any = false;
+        while (b != 0) {
+            ulonglong2 bk = build_keys[b];
+            if ((bk.x == k.x) & (bk.y == k.y)) {
+                any = true;
+                if (mode != 2) {
+                    out_probe[pos] = (uint32_t)i;
+                    out_build[pos] = b;
+                    pos++;
+                }
+                if (mode == 1 || mode == 2) break;
+            }
+            b = next[b];
+        }
+        if (!any && (mode == 2 || mode == 3)) {
+            out_probe[pos] = (uint32_t)i;
+            out_build[pos] = 0;
+        }
+    }
+}
+
+extern "C" {
+int gpue_join_build_bucket_chained_u128(gpue_session* s, gpue_dbuf* keys /*16 B, 1-based*/,
+                                        uint64_t row_count, gpue_join_table** out);
+int gpue_join_probe_emit_mode_u128(gpue_session* s, gpue_join_table* t,
+                                   gpue_dbuf* probe_keys, uint64_t n_rows, int mode,
+                                   gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                                   uint64_t* match_count);
+}
+
+int gpue_join_build_bucket_chained_u128(gpue_session* s, gpue_dbuf* keys,
+                                        uint64_t row_count, gpue_join_table** out) {
+    ARG_CHECK(s && keys && out && row_count > 0 && row_count + 1 < (1ull << 31));
+    ARG_CHECK(keys->bytes >= (row_count + 1) * 16);
+    gpue_join_table* t = new gpue_join_table();
+    t->s = s;
+    t->kind = gpue_join_table::BUCKET_CHAINED128;
+    t->row_count = row_count;
+    t->bucket_size = calc_bucket_size((uint32_t)(row_count + 1));
+    t->log_bucket_size = (uint32_t)__builtin_ctzll(t->bucket_size);
+    HIP_CHECK(hipMalloc(&t->first, t->bucket_size * sizeof(uint32_t)));
+    HIP_CHECK(hipMalloc(&t->next, (row_count + 1) * sizeof(uint32_t)));
+    HIP_CHECK(hipMalloc(&t->build_keys128, (row_count + 1) * sizeof(ulonglong2)));
+    HIP_CHECK(hipMemsetAsync(t->first, 0, t->bucket_size * sizeof(uint32_t), s->stream));
+    HIP_CHECK(hipMemsetAsync(t->next, 0, (row_count + 1) * sizeof(uint32_t), s->stream));
+    HIP_CHECK(hipMemcpyAsync(t->build_keys128, keys->ptr,
+                             (row_count + 1) * sizeof(ulonglong2),
+                             hipMemcpyDeviceToDevice, s->stream));
+    hipLaunchKernelGGL(k_build_bucket_chained128, dim3(grid_for(row_count)), dim3(BLOCK), 0,
+                       s->stream, t->build_keys128, row_count,
+                       (uint32_t)(t->bucket_size - 1), t->first, t->next);
+    HIP_CHECK(hipGetLastError());
+    *out = t;
+    return GPUE_OK;
+}
+
+int gpue_join_probe_emit_mode_u128(gpue_session* s, gpue_join_table* t, gpue_dbuf* probe_keys,
+                                   uint64_t n_rows, int mode, gpue_dbuf* out_probe_idx,
+                                   gpue_dbuf* out_build_idx, uint64_t* match_count) {
+    ARG_CHECK(s && t && probe_keys && match_count);
+    ARG_CHECK(t->kind == gpue_join_table::BUCKET_CHAINED128);
+    ARG_CHECK(mode >= 0 && mode <= 3);
+    ARG_CHECK(probe_keys->bytes >= n_rows * 16);
+    uint32_t nb = grid_for(n_rows);
+    uint64_t tile = (n_rows + nb - 1) / nb;
+    uint32_t* d_counts = nullptr;
+    uint64_t* d_bsums = nullptr;
+    uint64_t* d_offsets = nullptr;
+    HIP_CHECK(hipMalloc(&d_counts, n_rows * 4));
+    HIP_CHECK(hipMalloc(&d_bsums, (nb + 1) * 8));
+    hipLaunchKernelGGL(k_probe_count_u128, dim3(nb), dim3(BLOCK), 0, s->stream,
+                       (const ulonglong2*)probe_keys->ptr, n_rows,
+                       (uint32_t)(t->bucket_size - 1), t->first, t->next, t->build_keys128,
+                       mode, d_counts);
+    hipLaunchKernelGGL(k_block_sums_u32, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts,
+                       n_rows, tile, d_bsums);
+    hipLaunchKernelGGL(k_scan_small, dim3(1), dim3(1), 0, s->stream, d_bsums, nb);
+    uint64_t total = 0;
+    HIP_CHECK(hipMemcpyAsync(&total, d_bsums + nb, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    *match_count = total;
+    if (out_probe_idx && out_build_idx && total > 0) {
+        ARG_CHECK(out_probe_idx->bytes >= total * 4 && out_build_idx->bytes >= total * 4);
+        HIP_CHECK(hipMalloc(&d_offsets, n_rows * 8));
+        hipLaunchKernelGGL(k_scan_offsets, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts,
+                           n_rows, tile, d_bsums, d_offsets);
+        hipLaunchKernelGGL(k_probe_emit_u128, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const ulonglong2*)probe_keys->ptr, n_rows,
+                           (uint32_t)(t->bucket_size - 1), t->first, t->next,
+                           t->build_keys128, mode, d_counts, d_offsets,
+                           (uint32_t*)out_probe_idx->ptr, (uint32_t*)out_build_idx->ptr);
         HIP_CHECK(hipStreamSynchronize(s->stream));
         (void)hipFree(d_offsets);
     }

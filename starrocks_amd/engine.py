@@ -91,6 +91,10 @@ def _declare(lib):
         "gpue_join_build_bucket_chained_u32": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_build_linear_chained_u32": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_build_bucket_chained_u64": (c_i32, [c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
+        "gpue_join_build_bucket_chained_u128": (c_i32, [c_vp, c_vp, c_u64,
+                                                        ctypes.POINTER(c_vp)]),
+        "gpue_join_probe_emit_mode_u128": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp,
+                                                   c_vp, ctypes.POINTER(c_u64)]),
         "gpue_join_probe_emit_mode_u64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp,
                                                   ctypes.POINTER(c_u64)]),
         "gpue_join_table_destroy": (None, [c_vp]),
@@ -417,6 +421,23 @@ class Engine:
         _ck(self._lib, self._lib.gpue_join_build_bucket_chained_u32(
             self._h, keys._h, row_count, ctypes.byref(h)))
         return JoinTable(self, h)
+
+    def join_build_bucket_chained_u128(self, keys: DBuf, row_count) -> JoinTable:
+        """16-byte key (LARGEINT / packed ≤16 B) bucket-chained build
+        (JoinKeyHash<T,16>: crc32 over the key bytes)."""
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_join_build_bucket_chained_u128(
+            self._h, keys._h, row_count, ctypes.byref(h)))
+        return JoinTable(self, h)
+
+    def join_probe_emit_mode_u128(self, table, probe_keys, n_rows, mode,
+                                  out_probe=None, out_build=None) -> int:
+        cnt = c_u64()
+        op = out_probe._h if out_probe else None
+        ob = out_build._h if out_build else None
+        _ck(self._lib, self._lib.gpue_join_probe_emit_mode_u128(
+            self._h, table._h, probe_keys._h, n_rows, mode, op, ob, ctypes.byref(cnt)))
+        return cnt.value
 
     def join_build_bucket_chained_u64(self, keys: DBuf, row_count) -> JoinTable:
         h = c_vp()

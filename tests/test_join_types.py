@@ -2,6 +2,7 @@
 oracle vs brute force on CPU; GPU parity in test_gpu_parity."""
 
 import numpy as np
+import pytest
 
 from oracle import pyoracle as orc
 
@@ -53,3 +54,59 @@ def test_right_semi_anti_vs_brute():
         exp = sorted(j for j in range(1, len(bk))
                      if (bk[j] in probe_set) != bool(anti))
         assert got == exp
+
+
+@pytest.mark.gpu
+def test_u128_key_join_parity(engine):
+    """16-byte key joins (JoinKeyHash<T,16> = crc32 over the key bytes,
+    join_hash_map_helper.h:23-30): the dedicated fixed-16 path must produce
+    the same multisets as the varchar/Slice path over the identical 16-byte
+    slices (which the oracle already pins), for every probe mode."""
+    rng = np.random.default_rng(55)
+    nbuild, nprobe = 20_000, 80_000
+    pool = rng.integers(0, 2**63, (3000, 2), dtype=np.uint64)
+    bkeys = np.zeros((nbuild + 1, 2), np.uint64)
+    bkeys[1:] = pool[rng.integers(0, len(pool), nbuild)]
+    pkeys = pool[rng.integers(0, len(pool), nprobe)].copy()
+    pkeys[::7] = rng.integers(0, 2**63, (len(pkeys[::7]), 2), dtype=np.uint64)
+
+    kb = engine.alloc(bkeys.nbytes)
+    kb.h2d(bkeys)
+    t128 = engine.join_build_bucket_chained_u128(kb, nbuild)
+    pb = engine.alloc(pkeys.nbytes)
+    pb.h2d(pkeys)
+
+    # varchar reference: same bytes as fixed-16 slices
+    boff = (np.arange(nbuild + 2, dtype=np.uint32) * 16)
+    poff = (np.arange(nprobe + 1, dtype=np.uint32) * 16)
+    bb = engine.alloc(bkeys.nbytes)
+    bb.h2d(bkeys.view(np.uint8).reshape(-1))
+    bo = engine.alloc(boff.nbytes)
+    bo.h2d(boff)
+    tvc = engine.join_build_varchar(bb, bo, nbuild)
+    pvb = engine.alloc(pkeys.nbytes)
+    pvb.h2d(pkeys.view(np.uint8).reshape(-1))
+    po = engine.alloc(poff.nbytes)
+    po.h2d(poff)
+
+    for mode in (0, 1, 2, 3):
+        c1 = engine.join_probe_emit_mode_u128(t128, pb, nprobe, mode)
+        c2 = engine.join_probe_emit_varchar_mode(tvc, pvb, po, nprobe, mode)
+        assert c1 == c2, mode
+        o1p, o1b = engine.alloc(max(c1, 1) * 4), engine.alloc(max(c1, 1) * 4)
+        o2p, o2b = engine.alloc(max(c2, 1) * 4), engine.alloc(max(c2, 1) * 4)
+        engine.join_probe_emit_mode_u128(t128, pb, nprobe, mode, o1p, o1b)
+        engine.join_probe_emit_varchar_mode(tvc, pvb, po, nprobe, mode, o2p, o2b)
+        if mode == 1:
+            assert np.array_equal(np.sort(o1p.d2h(np.uint32, c1)),
+                                  np.sort(o2p.d2h(np.uint32, c2)))
+        else:
+            pk_ = lambda a, b: np.sort(a.astype(np.uint64) << np.uint64(32) | b)
+            assert np.array_equal(pk_(o1p.d2h(np.uint32, c1), o1b.d2h(np.uint32, c1)),
+                                  pk_(o2p.d2h(np.uint32, c2), o2b.d2h(np.uint32, c2))), mode
+        for x in (o1p, o1b, o2p, o2b):
+            x.free()
+    t128.destroy()
+    tvc.destroy()
+    for x in (kb, pb, bb, bo, pvb, po):
+        x.free()
