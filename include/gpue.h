@@ -279,6 +279,10 @@ int gpue_join_probe_emit_nulls_i32(gpue_session* s, gpue_join_table* t, gpue_dbu
  * join_hash_map_helper.h:112-136): two int32 key columns -> one 8-byte key. */
 int gpue_pack_keys_2xi32(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b, uint64_t n,
                          gpue_dbuf* out);
+/* two i64 key columns packed into one 16-byte key
+ * (SERIALIZED_FIXED_SIZE_LARGEINT, join_key_constructor.h:40-153) */
+int gpue_pack_keys_2xi64(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b, uint64_t n,
+                         gpue_dbuf* out);
 /* SERIALIZED_VARCHAR / Slice keys (the selector's remaining constructor
  * branch, join_hash_map.cpp:269-281 -> JoinKeyHash<Slice>,
  * join_hash_map_helper.h:57-64: crc_hash_32(bytes,len,CRC_HASH_SEED1) masked

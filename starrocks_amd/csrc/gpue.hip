@@ -3197,6 +3197,30 @@ int gpue_join_probe_emit_mode_u128(gpue_session* s, gpue_join_table* t, gpue_dbu
     return GPUE_OK;
 }
 
+// SERIALIZED_FIXED_SIZE_LARGEINT packing for two 8-byte key columns
+// (join_key_constructor.h:40-153 / serialize_batch_at_interval: keys
+// concatenated little-endian into one 16-byte value)
+__global__ void k_pack_keys_2xi64(const uint64_t* __restrict__ a,
+                                  const uint64_t* __restrict__ b, uint64_t n,
+                                  ulonglong2* __restrict__ out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+        out[i] = make_ulonglong2(a[i], b[i]);
+}
+
+extern "C" int gpue_pack_keys_2xi64(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b,
+                                    uint64_t n, gpue_dbuf* out);
+int gpue_pack_keys_2xi64(gpue_session* s, gpue_dbuf* a, gpue_dbuf* b, uint64_t n,
+                         gpue_dbuf* out) {
+    ARG_CHECK(s && a && b && out);
+    ARG_CHECK(a->bytes >= n * 8 && b->bytes >= n * 8 && out->bytes >= n * 16);
+    hipLaunchKernelGGL(k_pack_keys_2xi64, dim3(grid_for(n)), dim3(BLOCK), 0, s->stream,
+                       (const uint64_t*)a->ptr, (const uint64_t*)b->ptr, n,
+                       (ulonglong2*)out->ptr);
+    HIP_CHECK(hipGetLastError());
+    return GPUE_OK;
+}
+
 // ---------------------------------------------------------------------------
 // SimdBlockFilter — the reference's split-block bloom runtime filter
 // (runtime_filter.h:79-232, runtime_filter.cpp:26-36,114-124; upstream

@@ -108,6 +108,7 @@ def _declare(lib):
         "gpue_join_build_bucket_chained_nulls_u32": (c_i32, [c_vp, c_vp, c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_join_probe_emit_nulls_i32": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_pack_keys_2xi32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
+        "gpue_pack_keys_2xi64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_q1_join_sum": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_u64,
                                      ctypes.POINTER(c_i64), ctypes.POINTER(c_u64)]),
         "gpue_q21_star_agg": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
@@ -535,6 +536,11 @@ class Engine:
             self._h, table._h, pbytes._h, poffsets._h, n_rows, mode, op, ob,
             ctypes.byref(cnt)))
         return cnt.value
+
+    def pack_keys_2xi64(self, a: DBuf, b: DBuf, n, out: DBuf):
+        """Two i64 key columns packed into 16-byte keys
+        (SERIALIZED_FIXED_SIZE_LARGEINT)."""
+        _ck(self._lib, self._lib.gpue_pack_keys_2xi64(self._h, a._h, b._h, n, out._h))
 
     def pack_keys_2xi32(self, a: DBuf, b: DBuf, n, out: DBuf):
         _ck(self._lib, self._lib.gpue_pack_keys_2xi32(self._h, a._h, b._h, n, out._h))
