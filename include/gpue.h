@@ -241,6 +241,12 @@ int gpue_join_select_method(int key_constructor, int lt_class, uint64_t row_coun
 int gpue_join_build_auto_i32(gpue_session* s, gpue_dbuf* keys, uint64_t row_count,
                              int mode, int with_other_conjunct, uint64_t l2_size,
                              uint64_t l3_size, gpue_join_table** out, int* method_out);
+/* 8-byte-key auto build: selector decision with LT_BIGINT reported via
+ * method_out; every physical tier maps onto the u64 bucket-chained table
+ * (the u64 domain has no range-direct/linear specialization here). */
+int gpue_join_build_auto_u64(gpue_session* s, gpue_dbuf* keys, uint64_t row_count,
+                             int mode, int with_other_conjunct, uint64_t l2_size,
+                             uint64_t l3_size, gpue_join_table** out, int* method_out);
 void gpue_join_table_destroy(gpue_join_table* t);
 int gpue_join_table_minmax(gpue_join_table* t, int64_t* min_out, int64_t* max_out);
 /* d2h the first[] array (tests) */

@@ -2190,6 +2190,60 @@ int gpue_join_build_auto_i32(gpue_session* s, gpue_dbuf* keys, uint64_t row_coun
     }
 }
 
+// 8-byte-key auto build: run the selector with LT_BIGINT and report its
+// decision; every physical tier maps onto the u64 bucket-chained table (the
+// u64 key domain has no range-direct/linear specialization in this engine —
+// dense int64 dims in our configs are i32-valued and take the i32 paths).
+__global__ void k_minmax_i64(const int64_t* __restrict__ keys, uint64_t n,
+                             long long* __restrict__ mn, long long* __restrict__ mx) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    long long lmn = 0x7FFFFFFFFFFFFFFFll, lmx = 0x8000000000000000ll;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        long long v = keys[i];
+        lmn = v < lmn ? v : lmn;
+        lmx = v > lmx ? v : lmx;
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        long long a = __shfl_down(lmn, off, WAVE);
+        long long b = __shfl_down(lmx, off, WAVE);
+        lmn = a < lmn ? a : lmn;
+        lmx = b > lmx ? b : lmx;
+    }
+    if ((threadIdx.x & (WAVE - 1)) == 0) {
+        atomicMin(mn, lmn);
+        atomicMax(mx, lmx);
+    }
+}
+
+extern "C" int gpue_join_build_auto_u64(gpue_session* s, gpue_dbuf* keys,
+                                        uint64_t row_count, int mode,
+                                        int with_other_conjunct, uint64_t l2_size,
+                                        uint64_t l3_size, gpue_join_table** out,
+                                        int* method_out);
+int gpue_join_build_auto_u64(gpue_session* s, gpue_dbuf* keys, uint64_t row_count,
+                             int mode, int with_other_conjunct, uint64_t l2_size,
+                             uint64_t l3_size, gpue_join_table** out, int* method_out) {
+    ARG_CHECK(s && keys && out && row_count > 0);
+    ARG_CHECK(keys->bytes >= (row_count + 1) * 8);
+    if (l2_size == 0) l2_size = 4ull << 20;
+    if (l3_size == 0) l3_size = 256ull << 20;
+    long long h_init[2] = {0x7FFFFFFFFFFFFFFFll, (long long)0x8000000000000000ll};
+    long long* d_mm = nullptr;
+    HIP_CHECK(hipMalloc(&d_mm, 16));
+    HIP_CHECK(hipMemcpyAsync(d_mm, h_init, 16, hipMemcpyHostToDevice, s->stream));
+    hipLaunchKernelGGL(k_minmax_i64, dim3(grid_for(row_count)), dim3(BLOCK), 0, s->stream,
+                       (const int64_t*)keys->ptr + 1, row_count, d_mm, d_mm + 1);
+    long long h_mm[2];
+    HIP_CHECK(hipMemcpyAsync(h_mm, d_mm, 16, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_mm);
+    int method = gpue_join_select_method(GPUE_KEYCON_ONE_KEY, GPUE_LT_BIGINT, row_count,
+                                         h_mm[0], h_mm[1], mode, with_other_conjunct, 1, 1,
+                                         l2_size, l3_size);
+    if (method_out) *method_out = method;
+    return gpue_join_build_bucket_chained_u64(s, keys, row_count, out);
+}
+
 // linear-probe lookup to the chain head, then chain walk (lookup_init,
 // join_hash_map_method.hpp:286-368): chains hold identical keys, so the walk
 // needs no further compare

@@ -71,6 +71,9 @@ def _declare(lib):
         "gpue_join_build_auto_i32": (c_i32, [c_vp, c_vp, c_u64, c_i32, c_i32, c_u64,
                                              c_u64, ctypes.POINTER(c_vp),
                                              ctypes.POINTER(c_i32)]),
+        "gpue_join_build_auto_u64": (c_i32, [c_vp, c_vp, c_u64, c_i32, c_i32, c_u64,
+                                             c_u64, ctypes.POINTER(c_vp),
+                                             ctypes.POINTER(c_i32)]),
         "gpue_gen_u32_mod": (c_i32, [c_vp, c_vp, c_u64, c_u64, c_u64, c_u64, c_u32, c_u32]),
         "gpue_gen_i64": (c_i32, [c_vp, c_vp, c_u64, c_u64, c_u64, c_u64]),
         "gpue_gen_lineorder_q1": (c_i32, [c_vp, c_u64, c_u64, c_u64, c_vp, c_vp, c_vp]),
@@ -393,6 +396,17 @@ class Engine:
         h = c_vp()
         m = c_i32()
         _ck(self._lib, self._lib.gpue_join_build_auto_i32(
+            self._h, keys._h, row_count, mode, with_other_conjunct, l2_size, l3_size,
+            ctypes.byref(h), ctypes.byref(m)))
+        return JoinTable(self, h), m.value
+
+    def join_build_auto_u64(self, keys: DBuf, row_count, mode=0, with_other_conjunct=0,
+                            l2_size=0, l3_size=0):
+        """Selector-driven build for 8-byte keys (decision reported; every
+        physical tier maps onto the u64 bucket-chained table)."""
+        h = c_vp()
+        m = c_i32()
+        _ck(self._lib, self._lib.gpue_join_build_auto_u64(
             self._h, keys._h, row_count, mode, with_other_conjunct, l2_size, l3_size,
             ctypes.byref(h), ctypes.byref(m)))
         return JoinTable(self, h), m.value

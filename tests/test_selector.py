@@ -257,3 +257,26 @@ def test_bench_dims_select_range_direct_on_mi355x_defaults():
         m = orc.join_select_method(ONE_KEY, LT_INT, rc, mn, mx, INNER, 0,
                                    l2_size=L2M, l3_size=L3M)
         assert m == RANGE_DIRECT, (rc, orc.JM_NAMES[m])
+
+
+@pytest.mark.gpu
+def test_join_build_auto_u64(engine):
+    """8-byte-key auto build: decision reported (sparse 64-bit interval >=
+    2^32 closes the range-direct gate -> LINEAR decision; physical tier is
+    the u64 bucket table), probes answer correctly."""
+    rng = np.random.default_rng(61)
+    n = 50_000
+    keys = np.concatenate([[0], rng.integers(1, 2**40, n)]).astype(np.uint64)
+    kb = engine.alloc(keys.nbytes)
+    kb.h2d(keys)
+    t, method = engine.join_build_auto_u64(kb, n)
+    assert orc.JM_NAMES[method] == "LINEAR_CHAINED"  # interval >= 2^32
+    probe = np.concatenate([keys[1:1001], rng.integers(1, 2**40, 1000)]).astype(np.uint64)
+    pb = engine.alloc(probe.nbytes)
+    pb.h2d(probe)
+    cnt = engine.join_probe_emit_u64(t, pb, len(probe), mode=1)
+    present = set(keys[1:].tolist())
+    assert cnt == sum(p in present for p in probe.tolist())
+    t.destroy()
+    kb.free()
+    pb.free()
