@@ -142,3 +142,64 @@ def test_partition_counting_sort_stable(keys, nch):
         assert np.array_equal(seg, np.sort(seg))          # stability
         assert all(ch[i] == c for i in seg.tolist())      # correct routing
     assert sp[-1] == len(k)
+
+
+# ---- page-codec properties (round 2): any value sequence must roundtrip ----
+
+rle_vals = st.lists(st.integers(-2**31, 2**31 - 1), min_size=0, max_size=600)
+runny_vals = st.lists(st.tuples(st.integers(-5, 5), st.integers(1, 50)),
+                      min_size=0, max_size=40)
+
+
+@settings(max_examples=60, deadline=None, derandomize=True)
+@given(vals=rle_vals)
+def test_rle_page_roundtrip_property(vals):
+    a = np.array(vals, np.int32)
+    assert np.array_equal(orc.rle_page_decode_i32(orc.rle_page_encode_i32(a), len(a)), a)
+
+
+@settings(max_examples=60, deadline=None, derandomize=True)
+@given(runs=runny_vals)
+def test_rle_page_runs_roundtrip_property(runs):
+    a = np.repeat([v for v, _ in runs], [c for _, c in runs]).astype(np.int32)
+    assert np.array_equal(orc.rle_page_decode_i32(orc.rle_page_encode_i32(a), len(a)), a)
+
+
+@settings(max_examples=60, deadline=None, derandomize=True)
+@given(vals=rle_vals)
+def test_for_page_roundtrip_property(vals):
+    a = np.array(vals, np.int32)
+    if len(a) == 0:
+        return  # FOR pages of zero values are represented by the caller
+    assert np.array_equal(
+        orc.for_page_decode_i32(orc.for_page_encode_i32(a), len(a)), a)
+
+
+@settings(max_examples=60, deadline=None, derandomize=True)
+@given(vals=st.lists(st.integers(-2**31, 2**31 - 1), min_size=1, max_size=400))
+def test_for_page_sorted_roundtrip_property(vals):
+    a = np.sort(np.array(vals, np.int32))  # exercises the ascending format
+    assert np.array_equal(
+        orc.for_page_decode_i32(orc.for_page_encode_i32(a), len(a)), a)
+
+
+@settings(max_examples=40, deadline=None, derandomize=True)
+@given(rows=st.lists(st.binary(min_size=0, max_size=24), min_size=0, max_size=120))
+def test_binary_prefix_roundtrip_property(rows):
+    off = np.zeros(len(rows) + 1, np.uint32)
+    np.cumsum([len(r) for r in rows], out=off[1:])
+    bts = np.frombuffer(b"".join(rows), np.uint8).copy()
+    if len(rows) == 0:
+        return
+    page = orc.binary_prefix_encode(bts, off)
+    db, do = orc.binary_prefix_decode(page, len(rows), max(int(off[-1]), 1))
+    assert np.array_equal(do, off)
+    assert np.array_equal(db, bts)
+
+
+@settings(max_examples=40, deadline=None, derandomize=True)
+@given(bools=st.lists(st.booleans(), min_size=0, max_size=600))
+def test_rle_bool_roundtrip_property(bools):
+    a = np.array(bools, np.uint8)
+    assert np.array_equal(
+        orc.rle_page_decode_bool(orc.rle_page_encode_bool(a), len(a)), a)
