@@ -382,3 +382,36 @@ def test_binary_prefix_page_decode_gpu_parity(engine):
         pb.free()
         ob.free()
         oo.free()
+
+
+@pytest.mark.gpu
+def test_for_page_feeds_q1_join_gpu(engine):
+    """Storage-ingress chain for the metric path: lo_orderdate stored as a
+    frame-of-reference page, decoded on device, then the q1 date-join+SUM
+    runs over the decoded column — results bit-exact vs the oracle pipeline
+    (the same chain test shape as the bitshuffle page in round 1)."""
+    from starrocks_amd import gen
+    n, year = 2_000_000, 1993
+    od, ep, dc = orc.gen_lineorder_q1(42, 0, n)
+    page = orc.for_page_encode_i32(od)
+    pb = engine.alloc(page.nbytes)
+    pb.h2d(page)
+    od_dev = engine.alloc(n * 4)
+    engine.page_decode_for_i32(pb, n, od_dev)
+    ep_dev = engine.alloc(n * 4)
+    ep_dev.h2d(ep)
+    dc_dev = engine.alloc(n * 4)
+    dc_dev.h2d(dc)
+    datekey, dyear = gen.gen_dates()
+    kb = engine.alloc(datekey.nbytes)
+    kb.h2d(datekey.astype(np.int32))
+    pay = np.where(dyear == year, dyear - 1992 + 1, 0).astype(np.uint32)
+    payb = engine.alloc(pay.nbytes)
+    payb.h2d(pay)
+    dates = engine.join_build_payload(kb, payb, len(datekey))
+    s, cnt = engine.q1_join_sum(dates, od_dev, ep_dev, dc_dev, n)
+    es, ecnt = orc.q1_pipeline(42, 0, n, year)
+    assert (s, cnt) == (es, ecnt)
+    dates.destroy()
+    for b in (pb, od_dev, ep_dev, dc_dev, kb, payb):
+        b.free()
