@@ -778,10 +778,8 @@ __global__ void k_filter_emit(const int64_t* __restrict__ in, uint64_t n, int64_
 // algorithmic byte count exactly. Spins are bounded: on timeout the kernel
 // sets an error flag and exits (host returns GPUE_ERR_HIP).
 // ---------------------------------------------------------------------------
-static constexpr uint32_t FILT_ITEMS = 8;                     // rows per thread per tile
-                                                              // (ITEMS template below
-                                                              // sweeps 8/16/32)
-static constexpr uint32_t FILT_TILE = BLOCK * FILT_ITEMS;     // 2048 rows
+// rows-per-thread and threads-per-block are template parameters of the
+// lookback kernels below (r02 sweep: 32 x 1024 default)
 static constexpr unsigned long long FILT_AGG = 1ull << 62;    // aggregate available
 static constexpr unsigned long long FILT_PREFIX = 2ull << 62; // inclusive prefix available
 static constexpr unsigned long long FILT_CNT_MASK = (1ull << 62) - 1;
@@ -4581,8 +4579,6 @@ k_q43_star_agg(const int32_t* __restrict__ ck, const int32_t* __restrict__ sk,
     const int4* __restrict__ sk4 = (const int4*)sk;
     const int4* __restrict__ pk4 = (const int4*)pk;
     const int4* __restrict__ od4 = (const int4*)od;
-    const int4* __restrict__ rv4 = (const int4*)rv;
-    const int4* __restrict__ sc4 = (const int4*)sc;
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     // The part bitset (1/25 selective) gates everything: test it from pk
     // alone, and only LOAD the other five streams for quads with a survivor.
@@ -4859,8 +4855,10 @@ __device__ static inline uint64_t xxh3_64_u32(uint32_t v, uint64_t seed) {
     return xxh3_rrmxmx(input64 ^ bitflip, 4);
 }
 
-// 8-byte input (one i64 key)
-__device__ static inline uint64_t xxh3_64_u64(uint64_t v, uint64_t seed) {
+// 8-byte input (one i64 key) — the i64 exchange-hash hop; the oracle-side
+// twin (orc_xxh3_hash_i64) is vector-pinned, this stays for the i64
+// partition path when a workload needs it
+__device__ [[maybe_unused]] static inline uint64_t xxh3_64_u64(uint64_t v, uint64_t seed) {
     uint32_t s32 = (uint32_t)seed;
     uint32_t sw = __builtin_bswap32(s32);
     seed ^= (uint64_t)sw << 32;
