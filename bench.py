@@ -475,10 +475,12 @@ def main():
             ev_g = [torch.cuda.Event() for _ in range(S)]
             ev_a = [torch.cuda.Event() for _ in range(S)]
 
+            part_scr = eng.alloc(eng.partition_scratch_bytes(rows, world))
+
             def _gather(b):
                 lo, hi = blk[b]
                 nb = hi - lo
-                eng.partition_i64(col_views[b][0], nb, world, ridx)
+                eng.partition_i64_async(col_views[b][0], nb, world, ridx, part_scr)
                 for j, (c, s_) in enumerate(zip(col_views[b], send[b])):
                     if j < 3:
                         eng.gather_u64(c, ridx, nb, s_)
@@ -578,12 +580,17 @@ def main():
             ev_g = [torch.cuda.Event() for _ in range(S)]
             ev_a = [torch.cuda.Event() for _ in range(S)]
 
+            part_scr = eng.alloc(eng.partition_scratch_bytes(rows, world))
+
             def _gather(b):
                 lo, hi = blk[b]
                 nb = hi - lo
                 # partition (fnv->channel + counting sort) + gather: the
-                # exchange sink stage (exchange_sink_operator.cpp:611-660)
-                eng.partition(col_views[b][0], nb, world, ridx)
+                # exchange sink stage (exchange_sink_operator.cpp:611-660).
+                # Async form: splits are static (discovered at setup), so the
+                # timed step never host-syncs here — the comm stream's
+                # all-to-all of block b-1 keeps flowing
+                eng.partition_async(col_views[b][0], nb, world, ridx, part_scr)
                 for c, s_ in zip(col_views[b], send[b]):
                     eng.gather_u32(c, ridx, nb, s_)
 

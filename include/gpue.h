@@ -453,6 +453,17 @@ int gpue_partition_crc_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n,
 int gpue_partition_varchar(gpue_session* s, gpue_dbuf* bytes, gpue_dbuf* offsets,
                            uint64_t n, uint32_t num_channels,
                            uint64_t* start_points_out, gpue_dbuf* row_indexes_out);
+/* Steady-state async partition: hist -> DEVICE-side scan -> emit with no
+ * host round-trip (the sync forms host-scan the histogram to return split
+ * sizes; in the chunked exchange the splits are static per shard, so the
+ * timed step only needs row_indexes). scratch holds the per-(block,channel)
+ * histogram + offsets (>= grid*channels*12 bytes). */
+int gpue_partition_i32_async(gpue_session* s, gpue_dbuf* keys, uint64_t n,
+                             uint32_t num_channels, gpue_dbuf* row_indexes_out,
+                             gpue_dbuf* scratch);
+int gpue_partition_i64_async(gpue_session* s, gpue_dbuf* keys, uint64_t n,
+                             uint32_t num_channels, gpue_dbuf* row_indexes_out,
+                             gpue_dbuf* scratch);
 int gpue_partition_i32(gpue_session* s, gpue_dbuf* keys, uint64_t n, uint32_t num_channels,
                        uint64_t* start_points_out, gpue_dbuf* row_indexes_out);
 /* Multi-column partition key: the sink seeds FNV_SEED then CHAINS fnv_hash

@@ -123,6 +123,8 @@ def _declare(lib):
         "gpue_q21_star_agg_async": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_vp, c_u64, c_vp]),
         "gpue_q21_star_agg_pipe": (c_i32, [c_vp] * 8 + [c_u64, c_vp, c_vp, c_i32]),
         "gpue_partition_i32": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
+        "gpue_partition_i32_async": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
+        "gpue_partition_i64_async": (c_i32, [c_vp, c_vp, c_u64, c_u32, c_vp, c_vp]),
         "gpue_session_stream": (c_vp, [c_vp]),
         "gpue_ingest_create": (c_i32, [c_vp, c_u64, ctypes.POINTER(c_vp)]),
         "gpue_pinned_alloc": (c_i32, [c_vp, c_u64, ctypes.POINTER(c_vp)]),
@@ -819,6 +821,25 @@ class Engine:
 
     def ingest_destroy(self, ing):
         self._lib.gpue_ingest_destroy(ing)
+
+    def partition_scratch_bytes(self, n, num_channels):
+        """Scratch size for the async partition forms (per-(block,channel)
+        hist + offsets)."""
+        nb = min((n + 255) // 256, 4096)  # conservative upper bound on grid
+        return int(nb * num_channels * 12 + 4096)
+
+    def partition_async(self, keys: DBuf, n, num_channels, row_indexes: DBuf,
+                        scratch: DBuf):
+        """Steady-state partition: hist -> device scan -> emit, fully async
+        (no host readback; splits must already be known — they are static
+        per shard)."""
+        _ck(self._lib, self._lib.gpue_partition_i32_async(
+            self._h, keys._h, n, num_channels, row_indexes._h, scratch._h))
+
+    def partition_i64_async(self, keys: DBuf, n, num_channels, row_indexes: DBuf,
+                            scratch: DBuf):
+        _ck(self._lib, self._lib.gpue_partition_i64_async(
+            self._h, keys._h, n, num_channels, row_indexes._h, scratch._h))
 
     def partition_crc(self, keys: DBuf, n, num_channels, row_indexes: DBuf) -> np.ndarray:
         """Bucket-shuffle hash path (zlib crc32 seed 0,
