@@ -1458,3 +1458,30 @@ def test_dense_right_semi_anti_parity(engine):
     tb.destroy()
     kb.free()
     pb.free()
+
+
+def test_partition_async_matches_sync(engine):
+    """The async partition (device-side scan) must produce byte-identical
+    row_indexes to the sync form — the chunked-exchange step relies on the
+    setup-discovered splits describing the async layout exactly."""
+    n, nch = 2_000_000, 8
+    keys = engine.alloc(n * 4)
+    engine.gen_u32_mod(keys, 42, 9, 0, n, 0, 0)
+    ri_sync = engine.alloc(n * 4)
+    sp = engine.partition(keys, n, nch, ri_sync)
+    ri_async = engine.alloc(n * 4)
+    scr = engine.alloc(engine.partition_scratch_bytes(n, nch))
+    engine.partition_async(keys, n, nch, ri_async, scr)
+    engine.sync()
+    assert np.array_equal(ri_sync.d2h(np.uint32, n), ri_async.d2h(np.uint32, n))
+    # i64 form
+    k64 = engine.alloc(n * 8)
+    engine.gen_i64(k64, 42, 3, 0, n)
+    r1 = engine.alloc(n * 4)
+    engine.partition_i64(k64, n, nch, r1)
+    r2 = engine.alloc(n * 4)
+    engine.partition_i64_async(k64, n, nch, r2, scr)
+    engine.sync()
+    assert np.array_equal(r1.d2h(np.uint32, n), r2.d2h(np.uint32, n))
+    for b in (keys, ri_sync, ri_async, scr, k64, r1, r2):
+        b.free()
