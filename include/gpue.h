@@ -626,6 +626,38 @@ int gpue_page_decode_binary_plain(gpue_session* s, gpue_dbuf* page, uint64_t n_v
 int gpue_page_decode_binary_prefix(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
                                    gpue_dbuf* out_bytes, gpue_dbuf* out_offsets);
 
+/* ---- ASOF join (reference LinearChainedAsofJoinHashMap,
+ * join_hash_map_method.h:201-217 + AsofIndex, join_hash_table_descriptor.h:
+ * 59-104 / .cpp:70-134). The selector routes every ASOF join here
+ * (_get_fallback_method, join_hash_table.cpp:257-263). Equi-key lookup plus a
+ * per-key temporal index sorted ascending (LT/LE) or descending (GT/GE),
+ * probed with the reference's branchless lower-bound search; at most one
+ * build match per probe row. GPU layout: open-addressing key slots + one
+ * contiguous (asof_value, row) segment per key — binary search wants
+ * contiguous sorted runs, not the CPU's pointer chains. Tie order among
+ * duplicate (key, asof) pairs is unspecified in the reference (pdqsort is
+ * unstable, comparator reads only asof_value); we pin the deterministic
+ * refinement "smallest build row wins the boundary slot". */
+#define GPUE_ASOF_LT 0 /* probe <  build: match smallest build value >  probe */
+#define GPUE_ASOF_LE 1 /* probe <= build: match smallest build value >= probe */
+#define GPUE_ASOF_GT 2 /* probe >  build: match largest  build value <  probe */
+#define GPUE_ASOF_GE 3 /* probe >= build: match largest  build value <= probe */
+typedef struct gpue_asof_table gpue_asof_table;
+/* keys: (row_count+1) int32 equi keys, asof: (row_count+1) int64 temporal
+ * values; row 0 is the sentinel "no match" row (never matched), as in the
+ * reference's 1-based build rows. opcode: GPUE_ASOF_*. */
+int gpue_asof_build_i32(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* asof,
+                        uint64_t row_count, int opcode, gpue_asof_table** out);
+/* mode: GPUE_JOIN_INNER (matched rows only) or GPUE_JOIN_LEFT_OUTER (miss
+ * emits build index 0), the reference's ASOF_INNER / ASOF_LEFT_OUTER
+ * (join_hash_table.cpp:744). Two-call contract like gpue_join_probe_emit:
+ * null outputs -> count only. Output ordered by probe row. */
+int gpue_asof_probe_emit_i32(gpue_session* s, gpue_asof_table* t, gpue_dbuf* probe_keys,
+                             gpue_dbuf* probe_asof, uint64_t n_rows, int mode,
+                             gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                             uint64_t* match_count);
+int gpue_asof_table_destroy(gpue_asof_table* t);
+
 /* ---- event timing on the session stream (bench roofline evidence) ---- */
 int gpue_timer_start(gpue_session* s);
 int gpue_timer_stop(gpue_session* s, float* ms_out);

@@ -2669,3 +2669,90 @@ uint64_t orc_lz4_decompress_block(const uint8_t* src, uint64_t comp_n, uint8_t* 
                                   uint64_t cap) {
     return lz4_decompress_block(src, comp_n, dst, cap);
 }
+
+/* ------------------------------------------------------------------------
+ * ASOF join — AsofIndex restatement (join_hash_table_descriptor.h:59-104,
+ * .cpp:70-134). Entries {asof_value, row_index} per equi key, sorted by
+ * asof_value ascending for LT/LE, descending for GT/GE (is_descending =
+ * GE||GT, is_strict = LT||GT), probed with the branchless lower-bound
+ * find_asof_match. The reference sorts with pdqsort (unstable) on
+ * asof_value only; we refine ties by row ascending — deterministic, and
+ * identical whenever (key, asof) pairs are distinct per key.
+ * ------------------------------------------------------------------------ */
+typedef struct {
+    int32_t key;
+    int64_t v;
+    uint32_t row;
+} OrcAsofEntry;
+
+static int g_asof_desc; /* qsort comparator direction (single-threaded use) */
+
+static int orc_asof_entry_cmp(const void* pa, const void* pb) {
+    const OrcAsofEntry* a = (const OrcAsofEntry*)pa;
+    const OrcAsofEntry* b = (const OrcAsofEntry*)pb;
+    if (a->key != b->key) return a->key < b->key ? -1 : 1;
+    if (a->v != b->v) {
+        if (g_asof_desc) return a->v > b->v ? -1 : 1;
+        return a->v < b->v ? -1 : 1;
+    }
+    return a->row < b->row ? -1 : (a->row > b->row ? 1 : 0);
+}
+
+/* find_asof_match (:83-108) with _bound_search_iteration (:112-131) inlined:
+ * the unroll-hinted >=8 loop plus tail loop run the same iteration sequence
+ * as this single loop. Returns entries[low].row_index or 0. */
+static uint32_t orc_asof_find(const OrcAsofEntry* e, uint32_t len, int64_t probe,
+                              int opcode) {
+    if (len == 0) return 0;
+    uint32_t size = len, low = 0;
+    while (size > 0) {
+        uint32_t half = size / 2;
+        uint32_t other_half = size - half;
+        uint32_t probe_pos = low + half;
+        uint32_t other_low = low + other_half;
+        int64_t entry = e[probe_pos].v;
+        size = half;
+        int cond;
+        switch (opcode) { /* :112-131 condition per (descending, strict) */
+        case 0: cond = probe >= entry; break;  /* LT: asc, strict */
+        case 1: cond = probe > entry; break;   /* LE: asc, non-strict */
+        case 2: cond = probe <= entry; break;  /* GT: desc, strict */
+        default: cond = probe < entry; break;  /* GE: desc, non-strict */
+        }
+        low = cond ? other_low : low;
+    }
+    return low < len ? e[low].row : 0;
+}
+
+void orc_asof_inner_join(const int32_t* build_keys, const int64_t* build_asof,
+                         uint32_t build_rows, const int32_t* probe_keys,
+                         const int64_t* probe_asof, uint64_t n, int opcode,
+                         uint32_t* out_build) {
+    OrcAsofEntry* e = (OrcAsofEntry*)malloc((size_t)build_rows * sizeof(OrcAsofEntry));
+    for (uint32_t i = 0; i < build_rows; i++) { /* rows are 1-based */
+        e[i].key = build_keys[i + 1];
+        e[i].v = build_asof[i + 1];
+        e[i].row = i + 1;
+    }
+    g_asof_desc = opcode >= 2; /* is_descending = GE||GT (:67) */
+    qsort(e, build_rows, sizeof(OrcAsofEntry), orc_asof_entry_cmp);
+    for (uint64_t i = 0; i < n; i++) {
+        int32_t k = probe_keys[i];
+        /* binary search the sorted-by-key entry array for the key's run */
+        uint32_t lo = 0, hi = build_rows; /* first index with key >= k */
+        while (lo < hi) {
+            uint32_t mid = lo + (hi - lo) / 2;
+            if (e[mid].key < k) lo = mid + 1;
+            else hi = mid;
+        }
+        uint32_t start = lo;
+        hi = build_rows; /* first index with key > k */
+        while (lo < hi) {
+            uint32_t mid = lo + (hi - lo) / 2;
+            if (e[mid].key <= k) lo = mid + 1;
+            else hi = mid;
+        }
+        out_build[i] = orc_asof_find(e + start, lo - start, probe_asof[i], opcode);
+    }
+    free(e);
+}

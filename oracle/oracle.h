@@ -343,3 +343,14 @@ uint64_t orc_binary_prefix_encode(const uint8_t* bytes, const uint32_t* offsets,
                                   uint32_t n, uint8_t* out);
 uint64_t orc_binary_prefix_decode(const uint8_t* page, uint64_t page_bytes,
                                   uint8_t* out_bytes, uint32_t* out_offsets);
+/* ASOF inner/left-outer join restatement (LinearChainedAsofJoinHashMap,
+ * join_hash_map_method.h:201-217 + AsofIndex, join_hash_table_descriptor.h:
+ * 59-104 / .cpp:70-134). opcode 0 LT / 1 LE / 2 GT / 3 GE. Build arrays are
+ * 1-based (row 0 sentinel). out_build[i] = 1-based matched build row for
+ * probe row i, 0 = no match. Tie refinement: duplicate (key, asof) pairs
+ * resolve to the smallest build row (the reference's pdqsort is unstable
+ * there; gpue pins the same refinement). */
+void orc_asof_inner_join(const int32_t* build_keys, const int64_t* build_asof,
+                         uint32_t build_rows, const int32_t* probe_keys,
+                         const int64_t* probe_asof, uint64_t n, int opcode,
+                         uint32_t* out_build);

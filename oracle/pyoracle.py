@@ -90,6 +90,8 @@ def _decl(lib):
     lib.orc_binary_prefix_encode.argtypes = [c_vp, c_vp, u, c_vp]
     lib.orc_binary_prefix_decode.restype = c_u64
     lib.orc_binary_prefix_decode.argtypes = [c_vp, c_u64, c_vp, c_vp]
+    lib.orc_asof_inner_join.restype = None
+    lib.orc_asof_inner_join.argtypes = [c_vp, c_vp, u, c_vp, c_vp, c_u64, c_i32, c_vp]
     lib.orc_xxh3_64_4to8.restype = c_u64
     lib.orc_xxh3_64_4to8.argtypes = [c_vp, c_i32, c_u64]
     lib.orc_xxh3_hash_i32.argtypes = [c_vp, c_u64, c_vp]
@@ -804,3 +806,19 @@ def binary_prefix_decode(page: np.ndarray, n: int, total_bytes: int):
     got = load().orc_binary_prefix_decode(_p(pg), len(pg), _p(b), _p(o))
     assert got == n
     return b[:int(o[-1])], o
+
+
+def asof_inner_join(build_keys, build_asof, probe_keys, probe_asof,
+                    opcode: int) -> np.ndarray:
+    """AsofIndex restatement (join_hash_table_descriptor.cpp:70-134).
+    build_* are 1-based (index 0 = sentinel); returns per-probe-row matched
+    1-based build row (0 = miss). opcode: 0 LT / 1 LE / 2 GT / 3 GE."""
+    bk = np.ascontiguousarray(build_keys, np.int32)
+    ba = np.ascontiguousarray(build_asof, np.int64)
+    pk = np.ascontiguousarray(probe_keys, np.int32)
+    pa = np.ascontiguousarray(probe_asof, np.int64)
+    assert len(bk) == len(ba) and len(pk) == len(pa)
+    out = np.empty(len(pk), np.uint32)
+    load().orc_asof_inner_join(_p(bk), _p(ba), len(bk) - 1, _p(pk), _p(pa),
+                               len(pk), opcode, _p(out))
+    return out

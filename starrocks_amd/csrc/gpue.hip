@@ -8112,3 +8112,470 @@ int gpue_timer_stop(gpue_session* s, float* ms_out) {
     HIP_CHECK(hipEventElapsedTime(ms_out, s->ev_start, s->ev_stop));
     return GPUE_OK;
 }
+
+// ---------------------------------------------------------------------------
+// ASOF join — reference LinearChainedAsofJoinHashMap (join_hash_map_method.h:
+// 201-217): an equi-key hash map whose buckets each carry an AsofIndex
+// (join_hash_table_descriptor.h:59-104) of {asof_value, row_index} entries,
+// sorted by asof_value ascending for LT/LE and descending for GT/GE
+// (is_descending = GE||GT, is_strict = LT||GT, :70-80), probed with the
+// branchless lower-bound find_asof_match (:83-134).
+//
+// MI355X layout: the CPU chains + per-bucket std::vector
+// become ONE open-addressing slot table (u64 per slot: bit63 occupancy flag |
+// u32 key) and ONE contiguous (asof_value[], row[]) arena grouped by slot —
+// the probe's binary search then walks a contiguous sorted run (coalescible,
+// log2(len) dependent loads) instead of pointer chains. Build is four
+// streaming passes (claim+count, scan, scatter, segmented sort); the
+// segmented sort runs one block per distinct key: LDS bitonic up to 4096
+// entries (48 KB of the 160 KB LDS), single-block global bitonic over a
+// pow2-padded segment beyond that (segments get pow2 capacity in the arena
+// so the network needs no virtual-padding cases).
+//
+// Tie pinning: the reference sorts with pdqsort (unstable) comparing ONLY
+// asof_value (join_hash_table_descriptor.cpp:70-80), so which duplicate
+// (key, asof) row lands at the match boundary is implementation-defined
+// there. We sort by (asof_value, row) lexicographic (row ascending in both
+// directions) — deterministic, and identical to the reference whenever
+// (key, asof) pairs are distinct. The oracle pins the same refinement.
+// ---------------------------------------------------------------------------
+struct gpue_asof_table {
+    gpue_session* s = nullptr;
+    int opcode = 0;          // GPUE_ASOF_*
+    uint64_t row_count = 0;  // build rows (1-based, row 0 sentinel)
+    uint64_t n_slots = 0;    // pow2 open-addressing size
+    uint32_t log_slots = 0;
+    uint64_t* slots = nullptr;   // (1ull<<63)|key when occupied, 0 empty
+    uint2* meta = nullptr;       // per slot {segment start, length} in the arena
+    int64_t* asof_vals = nullptr;  // arena: padded-pow2 segments, sorted
+    uint32_t* asof_rows = nullptr; // matching 1-based build rows
+};
+
+#define ASOF_OCCUPIED (1ull << 63)
+
+// claim-or-find a slot for key; returns slot index. Winners append the slot
+// to the occupied list (drives the per-segment sort grid).
+__device__ static inline uint32_t asof_slot_insert(uint32_t k, uint32_t log_slots,
+                                                   uint32_t mask, uint64_t* slots,
+                                                   uint32_t* occ, uint32_t* occ_cursor) {
+    uint64_t want = ASOF_OCCUPIED | (uint64_t)k;
+    uint32_t p = join_hash_u32(k, log_slots);
+    for (;;) {
+        uint64_t prev = atomicCAS((unsigned long long*)&slots[p], 0ull,
+                                  (unsigned long long)want);
+        if (prev == 0ull) {
+            occ[atomicAdd(occ_cursor, 1u)] = p;
+            return p;
+        }
+        if (prev == want) return p;
+        p = (p + 1) & mask;
+    }
+}
+
+// lookup-only probe; returns slot or 0xFFFFFFFF on miss
+__device__ static inline uint32_t asof_slot_find(uint32_t k, uint32_t log_slots,
+                                                 uint32_t mask,
+                                                 const uint64_t* __restrict__ slots) {
+    uint64_t want = ASOF_OCCUPIED | (uint64_t)k;
+    uint32_t p = join_hash_u32(k, log_slots);
+    for (;;) {
+        uint64_t v = slots[p];
+        if (v == want) return p;
+        if (v == 0ull) return 0xFFFFFFFFu;
+        p = (p + 1) & mask;
+    }
+}
+
+__global__ void k_asof_insert(const int32_t* __restrict__ keys, uint64_t row_count,
+                              uint32_t log_slots, uint32_t mask, uint64_t* slots,
+                              uint32_t* counts, uint32_t* occ, uint32_t* occ_cursor) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i <= row_count;
+         i += stride) {
+        uint32_t p = asof_slot_insert((uint32_t)keys[i], log_slots, mask, slots, occ,
+                                      occ_cursor);
+        atomicAdd(&counts[p], 1u);
+    }
+}
+
+// per-slot arena capacity = next pow2 of the entry count: pow2 segments let
+// the >LDS sort run a plain bitonic network with sentinel padding in place
+__global__ void k_asof_caps(const uint32_t* __restrict__ counts, uint64_t n_slots,
+                            uint32_t* __restrict__ caps) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_slots;
+         i += stride) {
+        uint32_t c = counts[i];
+        caps[i] = c <= 1 ? c : (2u << (31 - __clz(c - 1)));
+    }
+}
+
+__global__ void k_asof_pack_meta(const uint64_t* __restrict__ cap_offsets,
+                                 const uint32_t* __restrict__ counts, uint64_t n_slots,
+                                 uint2* __restrict__ meta) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_slots;
+         i += stride)
+        meta[i] = make_uint2((uint32_t)cap_offsets[i], counts[i]);
+}
+
+__global__ void k_asof_scatter(const int32_t* __restrict__ keys,
+                               const int64_t* __restrict__ asof, uint64_t row_count,
+                               uint32_t log_slots, uint32_t mask,
+                               const uint64_t* __restrict__ slots,
+                               const uint2* __restrict__ meta, uint32_t* __restrict__ cursors,
+                               int64_t* __restrict__ vals, uint32_t* __restrict__ rows) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i <= row_count;
+         i += stride) {
+        uint32_t p = asof_slot_find((uint32_t)keys[i], log_slots, mask, slots);
+        uint32_t pos = meta[p].x + atomicAdd(&cursors[p], 1u);
+        vals[pos] = asof[i];
+        rows[pos] = (uint32_t)i;
+    }
+}
+
+// composite order: by asof_value in the opcode's direction, row ascending on
+// ties (the pinned refinement; see the section comment)
+__device__ static inline bool asof_before(int64_t av, uint32_t ar, int64_t bv, uint32_t br,
+                                          bool descending) {
+    if (av != bv) return descending ? (av > bv) : (av < bv);
+    return ar < br;
+}
+
+// one block per occupied slot. len <= 4096: LDS bitonic (48 KB static).
+// Larger: sentinel-pad the segment to its pow2 capacity in place, then the
+// same bitonic network block-strided over global memory (__syncthreads is
+// the stage barrier — valid because exactly one block owns the segment).
+// Sentinels (+inf,row ~0u for ascending / -inf,~0u for descending) order
+// strictly after every real entry, so positions [0,len) hold exactly the
+// real entries post-sort and the probe never reads the pad.
+#define ASOF_LDS_CAP 4096
+__global__ void k_asof_sort_segments(const uint32_t* __restrict__ occ, uint32_t n_occ,
+                                     const uint2* __restrict__ meta, int descending,
+                                     int64_t* __restrict__ vals,
+                                     uint32_t* __restrict__ rows) {
+    __shared__ int64_t s_val[ASOF_LDS_CAP];
+    __shared__ uint32_t s_row[ASOF_LDS_CAP];
+    for (uint32_t seg = blockIdx.x; seg < n_occ; seg += gridDim.x) {
+        uint2 m = meta[occ[seg]];
+        uint32_t base = m.x, len = m.y;
+        if (len <= 1) continue;
+        uint32_t cap = 2u << (31 - __clz(len - 1)); // pow2 >= len (len >= 2)
+        const int64_t pad_val = descending ? INT64_MIN : INT64_MAX;
+        if (len <= ASOF_LDS_CAP) {
+            for (uint32_t i = threadIdx.x; i < cap; i += blockDim.x) {
+                s_val[i] = i < len ? vals[base + i] : pad_val;
+                s_row[i] = i < len ? rows[base + i] : 0xFFFFFFFFu;
+            }
+            __syncthreads();
+            for (uint32_t k = 2; k <= cap; k <<= 1) {
+                for (uint32_t j = k >> 1; j > 0; j >>= 1) {
+                    for (uint32_t i = threadIdx.x; i < cap; i += blockDim.x) {
+                        uint32_t ij = i ^ j;
+                        if (ij > i) {
+                            bool up = (i & k) == 0;
+                            bool sw = asof_before(s_val[ij], s_row[ij], s_val[i], s_row[i],
+                                                  descending);
+                            if (sw == up) {
+                                int64_t tv = s_val[i]; s_val[i] = s_val[ij]; s_val[ij] = tv;
+                                uint32_t tr = s_row[i]; s_row[i] = s_row[ij]; s_row[ij] = tr;
+                            }
+                        }
+                    }
+                    __syncthreads();
+                }
+            }
+            for (uint32_t i = threadIdx.x; i < len; i += blockDim.x) {
+                vals[base + i] = s_val[i];
+                rows[base + i] = s_row[i];
+            }
+            __syncthreads();
+        } else {
+            for (uint32_t i = len + threadIdx.x; i < cap; i += blockDim.x) {
+                vals[base + i] = pad_val;
+                rows[base + i] = 0xFFFFFFFFu;
+            }
+            __syncthreads();
+            for (uint32_t k = 2; k <= cap; k <<= 1) {
+                for (uint32_t j = k >> 1; j > 0; j >>= 1) {
+                    for (uint32_t i = threadIdx.x; i < cap; i += blockDim.x) {
+                        uint32_t ij = i ^ j;
+                        if (ij > i) {
+                            bool up = (i & k) == 0;
+                            bool sw = asof_before(vals[base + ij], rows[base + ij],
+                                                  vals[base + i], rows[base + i],
+                                                  descending);
+                            if (sw == up) {
+                                int64_t tv = vals[base + i];
+                                vals[base + i] = vals[base + ij];
+                                vals[base + ij] = tv;
+                                uint32_t tr = rows[base + i];
+                                rows[base + i] = rows[base + ij];
+                                rows[base + ij] = tr;
+                            }
+                        }
+                    }
+                    __syncthreads();
+                }
+            }
+        }
+    }
+}
+
+// find_asof_match restated (join_hash_table_descriptor.cpp:83-134): branchless
+// lower bound over the sorted segment; the CPU's two while-loops (an
+// unroll-hinted >=8 loop then the tail) are one loop here — the #pragma GCC
+// unroll is a scheduling hint, the iteration sequence is identical.
+// OP: 0 LT, 1 LE, 2 GT, 3 GE. "low = other_low" condition per variant
+// (:112-131): LT probe>=entry / LE probe>entry / GT probe<=entry /
+// GE probe<entry. Returns the 1-based build row, 0 on no match (:105-108).
+template <int OP>
+__device__ static inline uint32_t asof_find(const int64_t* __restrict__ vals,
+                                            const uint32_t* __restrict__ rows,
+                                            uint32_t base, uint32_t len, int64_t probe) {
+    if (len == 0) return 0;
+    uint32_t size = len, low = 0;
+    while (size > 0) {
+        uint32_t half = size >> 1;
+        uint32_t probe_pos = low + half;
+        uint32_t other_low = low + (size - half);
+        int64_t entry = vals[base + probe_pos];
+        size = half;
+        bool cond;
+        if (OP == 0) cond = probe >= entry;
+        else if (OP == 1) cond = probe > entry;
+        else if (OP == 2) cond = probe <= entry;
+        else cond = probe < entry;
+        low = cond ? other_low : low;
+    }
+    return low < len ? rows[base + low] : 0;
+}
+
+template <int OP>
+__global__ void k_asof_probe_count(const int32_t* __restrict__ pkeys,
+                                   const int64_t* __restrict__ pasof, uint64_t n,
+                                   uint32_t log_slots, uint32_t mask,
+                                   const uint64_t* __restrict__ slots,
+                                   const uint2* __restrict__ meta,
+                                   const int64_t* __restrict__ vals,
+                                   const uint32_t* __restrict__ rows, int mode,
+                                   uint32_t* __restrict__ row_counts) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        uint32_t p = asof_slot_find((uint32_t)pkeys[i], log_slots, mask, slots);
+        uint32_t match = 0;
+        if (p != 0xFFFFFFFFu) {
+            uint2 m = meta[p];
+            match = asof_find<OP>(vals, rows, m.x, m.y, pasof[i]);
+        }
+        row_counts[i] = match ? 1u : (mode == GPUE_JOIN_LEFT_OUTER ? 1u : 0u);
+    }
+}
+
+template <int OP>
+__global__ void k_asof_probe_emit(const int32_t* __restrict__ pkeys,
+                                  const int64_t* __restrict__ pasof, uint64_t n,
+                                  uint32_t log_slots, uint32_t mask,
+                                  const uint64_t* __restrict__ slots,
+                                  const uint2* __restrict__ meta,
+                                  const int64_t* __restrict__ vals,
+                                  const uint32_t* __restrict__ rows, int mode,
+                                  const uint32_t* __restrict__ row_counts,
+                                  const uint64_t* __restrict__ row_offsets,
+                                  uint32_t* __restrict__ out_probe,
+                                  uint32_t* __restrict__ out_build) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+        if (row_counts[i] == 0) continue;
+        uint32_t p = asof_slot_find((uint32_t)pkeys[i], log_slots, mask, slots);
+        uint32_t match = 0;
+        if (p != 0xFFFFFFFFu) {
+            uint2 m = meta[p];
+            match = asof_find<OP>(vals, rows, m.x, m.y, pasof[i]);
+        }
+        uint64_t pos = row_offsets[i];
+        out_probe[pos] = (uint32_t)i;
+        out_build[pos] = match; // 0 = LEFT_OUTER miss, as the reference's row 0
+    }
+}
+
+int gpue_asof_build_i32(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* asof,
+                        uint64_t row_count, int opcode, gpue_asof_table** out) {
+    ARG_CHECK(s && keys && asof && out && row_count > 0 && row_count + 1 < (1ull << 31));
+    ARG_CHECK(opcode >= GPUE_ASOF_LT && opcode <= GPUE_ASOF_GE);
+    ARG_CHECK(keys->bytes >= (row_count + 1) * 4 && asof->bytes >= (row_count + 1) * 8);
+    gpue_asof_table* t = new gpue_asof_table();
+    t->s = s;
+    t->opcode = opcode;
+    t->row_count = row_count;
+    uint64_t want = 2 * (row_count + 1);
+    t->n_slots = 64;
+    while (t->n_slots < want) t->n_slots <<= 1;
+    t->log_slots = (uint32_t)__builtin_ctzll(t->n_slots);
+    uint32_t mask = (uint32_t)(t->n_slots - 1);
+    uint32_t* d_counts = nullptr;
+    uint32_t* d_caps = nullptr;
+    uint32_t* d_occ = nullptr;
+    uint32_t* d_misc = nullptr; // occ cursor + scatter cursors
+    uint64_t* d_bsums = nullptr;
+    uint64_t* d_offsets = nullptr;
+    HIP_CHECK(hipMalloc(&t->slots, t->n_slots * 8));
+    HIP_CHECK(hipMalloc(&t->meta, t->n_slots * 8));
+    HIP_CHECK(hipMalloc(&d_counts, t->n_slots * 4));
+    HIP_CHECK(hipMalloc(&d_caps, t->n_slots * 4));
+    HIP_CHECK(hipMalloc(&d_occ, row_count * 4));
+    HIP_CHECK(hipMalloc(&d_misc, (t->n_slots + 1) * 4));
+    HIP_CHECK(hipMemsetAsync(t->slots, 0, t->n_slots * 8, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_counts, 0, t->n_slots * 4, s->stream));
+    HIP_CHECK(hipMemsetAsync(d_misc, 0, (t->n_slots + 1) * 4, s->stream));
+    uint32_t nb_rows = grid_for(row_count);
+    hipLaunchKernelGGL(k_asof_insert, dim3(nb_rows), dim3(BLOCK), 0, s->stream,
+                       (const int32_t*)keys->ptr, row_count, t->log_slots, mask, t->slots,
+                       d_counts, d_occ, d_misc);
+    // pow2 arena capacities, block-scanned to segment starts
+    uint32_t nb_slots = grid_for(t->n_slots);
+    uint64_t tile = (t->n_slots + nb_slots - 1) / nb_slots;
+    HIP_CHECK(hipMalloc(&d_bsums, (nb_slots + 1) * 8));
+    HIP_CHECK(hipMalloc(&d_offsets, t->n_slots * 8));
+    hipLaunchKernelGGL(k_asof_caps, dim3(nb_slots), dim3(BLOCK), 0, s->stream, d_counts,
+                       t->n_slots, d_caps);
+    hipLaunchKernelGGL(k_block_sums_u32, dim3(nb_slots), dim3(BLOCK), 0, s->stream, d_caps,
+                       t->n_slots, tile, d_bsums);
+    hipLaunchKernelGGL(k_scan_small, dim3(1), dim3(1), 0, s->stream, d_bsums, nb_slots);
+    hipLaunchKernelGGL(k_scan_offsets, dim3(nb_slots), dim3(BLOCK), 0, s->stream, d_caps,
+                       t->n_slots, tile, d_bsums, d_offsets);
+    hipLaunchKernelGGL(k_asof_pack_meta, dim3(nb_slots), dim3(BLOCK), 0, s->stream,
+                       d_offsets, d_counts, t->n_slots, t->meta);
+    uint64_t arena = 0; // padded total <= 2*row_count
+    uint32_t n_occ = 0;
+    HIP_CHECK(hipMemcpyAsync(&arena, d_bsums + nb_slots, 8, hipMemcpyDeviceToHost,
+                             s->stream));
+    HIP_CHECK(hipMemcpyAsync(&n_occ, d_misc, 4, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    HIP_CHECK(hipMalloc(&t->asof_vals, (arena ? arena : 1) * 8));
+    HIP_CHECK(hipMalloc(&t->asof_rows, (arena ? arena : 1) * 4));
+    hipLaunchKernelGGL(k_asof_scatter, dim3(nb_rows), dim3(BLOCK), 0, s->stream,
+                       (const int32_t*)keys->ptr, (const int64_t*)asof->ptr, row_count,
+                       t->log_slots, mask, t->slots, t->meta, d_misc + 1, t->asof_vals,
+                       t->asof_rows);
+    int descending = opcode >= GPUE_ASOF_GT; // is_descending = GE||GT (:67)
+    uint32_t nb_sort = n_occ < MAX_GRID ? (n_occ ? n_occ : 1) : (uint32_t)MAX_GRID;
+    hipLaunchKernelGGL(k_asof_sort_segments, dim3(nb_sort), dim3(BLOCK), 0, s->stream,
+                       d_occ, n_occ, t->meta, descending, t->asof_vals, t->asof_rows);
+    HIP_CHECK(hipGetLastError());
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(d_counts);
+    (void)hipFree(d_caps);
+    (void)hipFree(d_occ);
+    (void)hipFree(d_misc);
+    (void)hipFree(d_bsums);
+    (void)hipFree(d_offsets);
+    *out = t;
+    return GPUE_OK;
+}
+
+int gpue_asof_probe_emit_i32(gpue_session* s, gpue_asof_table* t, gpue_dbuf* probe_keys,
+                             gpue_dbuf* probe_asof, uint64_t n_rows, int mode,
+                             gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                             uint64_t* match_count) {
+    ARG_CHECK(s && t && probe_keys && probe_asof && match_count);
+    ARG_CHECK(mode == GPUE_JOIN_INNER || mode == GPUE_JOIN_LEFT_OUTER);
+    ARG_CHECK(probe_keys->bytes >= n_rows * 4 && probe_asof->bytes >= n_rows * 8);
+    uint32_t nb = grid_for(n_rows);
+    uint64_t tile = (n_rows + nb - 1) / nb;
+    uint32_t mask = (uint32_t)(t->n_slots - 1);
+    uint32_t* d_counts = nullptr;
+    uint64_t* d_bsums = nullptr;
+    uint64_t* d_offsets = nullptr;
+    HIP_CHECK(hipMalloc(&d_counts, (n_rows ? n_rows : 1) * 4));
+    HIP_CHECK(hipMalloc(&d_bsums, (nb + 1) * 8));
+    switch (t->opcode) {
+    case GPUE_ASOF_LT:
+        hipLaunchKernelGGL(k_asof_probe_count<0>, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr,
+                           n_rows, t->log_slots, mask, t->slots, t->meta, t->asof_vals,
+                           t->asof_rows, mode, d_counts);
+        break;
+    case GPUE_ASOF_LE:
+        hipLaunchKernelGGL(k_asof_probe_count<1>, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr,
+                           n_rows, t->log_slots, mask, t->slots, t->meta, t->asof_vals,
+                           t->asof_rows, mode, d_counts);
+        break;
+    case GPUE_ASOF_GT:
+        hipLaunchKernelGGL(k_asof_probe_count<2>, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr,
+                           n_rows, t->log_slots, mask, t->slots, t->meta, t->asof_vals,
+                           t->asof_rows, mode, d_counts);
+        break;
+    default:
+        hipLaunchKernelGGL(k_asof_probe_count<3>, dim3(nb), dim3(BLOCK), 0, s->stream,
+                           (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr,
+                           n_rows, t->log_slots, mask, t->slots, t->meta, t->asof_vals,
+                           t->asof_rows, mode, d_counts);
+        break;
+    }
+    hipLaunchKernelGGL(k_block_sums_u32, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts,
+                       n_rows, tile, d_bsums);
+    hipLaunchKernelGGL(k_scan_small, dim3(1), dim3(1), 0, s->stream, d_bsums, nb);
+    uint64_t total = 0;
+    HIP_CHECK(hipMemcpyAsync(&total, d_bsums + nb, 8, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    *match_count = total;
+    if (out_probe_idx && out_build_idx && total > 0) {
+        ARG_CHECK(out_probe_idx->bytes >= total * 4 && out_build_idx->bytes >= total * 4);
+        HIP_CHECK(hipMalloc(&d_offsets, n_rows * 8));
+        hipLaunchKernelGGL(k_scan_offsets, dim3(nb), dim3(BLOCK), 0, s->stream, d_counts,
+                           n_rows, tile, d_bsums, d_offsets);
+        switch (t->opcode) {
+        case GPUE_ASOF_LT:
+            hipLaunchKernelGGL(k_asof_probe_emit<0>, dim3(nb), dim3(BLOCK), 0, s->stream,
+                               (const int32_t*)probe_keys->ptr,
+                               (const int64_t*)probe_asof->ptr, n_rows, t->log_slots, mask,
+                               t->slots, t->meta, t->asof_vals, t->asof_rows, mode,
+                               d_counts, d_offsets, (uint32_t*)out_probe_idx->ptr,
+                               (uint32_t*)out_build_idx->ptr);
+            break;
+        case GPUE_ASOF_LE:
+            hipLaunchKernelGGL(k_asof_probe_emit<1>, dim3(nb), dim3(BLOCK), 0, s->stream,
+                               (const int32_t*)probe_keys->ptr,
+                               (const int64_t*)probe_asof->ptr, n_rows, t->log_slots, mask,
+                               t->slots, t->meta, t->asof_vals, t->asof_rows, mode,
+                               d_counts, d_offsets, (uint32_t*)out_probe_idx->ptr,
+                               (uint32_t*)out_build_idx->ptr);
+            break;
+        case GPUE_ASOF_GT:
+            hipLaunchKernelGGL(k_asof_probe_emit<2>, dim3(nb), dim3(BLOCK), 0, s->stream,
+                               (const int32_t*)probe_keys->ptr,
+                               (const int64_t*)probe_asof->ptr, n_rows, t->log_slots, mask,
+                               t->slots, t->meta, t->asof_vals, t->asof_rows, mode,
+                               d_counts, d_offsets, (uint32_t*)out_probe_idx->ptr,
+                               (uint32_t*)out_build_idx->ptr);
+            break;
+        default:
+            hipLaunchKernelGGL(k_asof_probe_emit<3>, dim3(nb), dim3(BLOCK), 0, s->stream,
+                               (const int32_t*)probe_keys->ptr,
+                               (const int64_t*)probe_asof->ptr, n_rows, t->log_slots, mask,
+                               t->slots, t->meta, t->asof_vals, t->asof_rows, mode,
+                               d_counts, d_offsets, (uint32_t*)out_probe_idx->ptr,
+                               (uint32_t*)out_build_idx->ptr);
+            break;
+        }
+        HIP_CHECK(hipStreamSynchronize(s->stream));
+        (void)hipFree(d_offsets);
+    }
+    (void)hipFree(d_counts);
+    (void)hipFree(d_bsums);
+    return GPUE_OK;
+}
+
+int gpue_asof_table_destroy(gpue_asof_table* t) {
+    if (!t) return GPUE_OK;
+    (void)hipFree(t->slots);
+    (void)hipFree(t->meta);
+    (void)hipFree(t->asof_vals);
+    (void)hipFree(t->asof_rows);
+    delete t;
+    return GPUE_OK;
+}

@@ -201,6 +201,10 @@ def _declare(lib):
         "gpue_ubench_bitgather": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_u64, c_i32,
                                           ctypes.POINTER(ctypes.c_float)]),
         "gpue_page_decode_bshuf_lz4_i32": (c_i32, [c_vp, c_vp, c_u32, c_vp]),
+        "gpue_asof_build_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp]),
+        "gpue_asof_probe_emit_i32": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_i32,
+                                             c_vp, c_vp, ctypes.POINTER(c_u64)]),
+        "gpue_asof_table_destroy": (c_i32, [c_vp]),
         "gpue_timer_start": (c_i32, [c_vp]),
         "gpue_timer_stop": (c_i32, [c_vp, ctypes.POINTER(ctypes.c_float)]),
     }
@@ -262,6 +266,20 @@ class JoinTable:
     def destroy(self):
         if self._h:
             self._lib.gpue_join_table_destroy(self._h)
+            self._h = None
+
+
+class AsofTable:
+    """ASOF join table (LinearChainedAsofJoinHashMap, join_hash_map_method.h:
+    201-217): equi-key slots + per-key temporal index sorted per opcode."""
+
+    def __init__(self, eng: "Engine", handle):
+        self._lib = eng._lib
+        self._h = handle
+
+    def destroy(self):
+        if self._h:
+            self._lib.gpue_asof_table_destroy(self._h)
             self._h = None
 
 
@@ -493,6 +511,33 @@ class Engine:
         ob = out_build._h if out_build else None
         _ck(self._lib, self._lib.gpue_join_probe_emit_mode_i32(
             self._h, table._h, probe_keys._h, n_rows, mode, op, ob, ctypes.byref(cnt)))
+        return cnt.value
+
+    # --- ASOF join (join_hash_map_method.h:201-217 + AsofIndex,
+    # join_hash_table_descriptor.h:59-104 / .cpp:70-134) ---
+    ASOF_NAMES = {0: "LT", 1: "LE", 2: "GT", 3: "GE"}
+
+    def asof_build(self, keys: DBuf, asof: DBuf, row_count, opcode) -> AsofTable:
+        """keys: (row_count+1) int32 equi keys, asof: (row_count+1) int64
+        temporal values, both 1-based (row 0 sentinel). opcode: 0 LT / 1 LE /
+        2 GT / 3 GE (the probe-vs-build temporal condition)."""
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_asof_build_i32(
+            self._h, keys._h, asof._h, row_count, opcode, ctypes.byref(h)))
+        return AsofTable(self, h)
+
+    def asof_probe_emit(self, table: AsofTable, probe_keys: DBuf, probe_asof: DBuf,
+                        n_rows, mode=0, out_probe=None, out_build=None) -> int:
+        """Two-call contract like join_probe_emit: without outputs, returns the
+        match count; with outputs, fills (probe row, 1-based build row) pairs
+        ordered by probe row. mode 0 = ASOF inner, 3 = ASOF left outer (miss
+        emits build row 0)."""
+        cnt = c_u64()
+        op = out_probe._h if out_probe else None
+        ob = out_build._h if out_build else None
+        _ck(self._lib, self._lib.gpue_asof_probe_emit_i32(
+            self._h, table._h, probe_keys._h, probe_asof._h, n_rows, mode, op, ob,
+            ctypes.byref(cnt)))
         return cnt.value
 
     def join_build_bucket_chained_nulls(self, keys: DBuf, nulls: DBuf, row_count):
