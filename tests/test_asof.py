@@ -159,3 +159,22 @@ def test_gpu_ties_and_small_segments(engine):
         for b in (op, ob, pkb, pab, kb, ab):
             b.free()
         t.destroy()
+
+
+def test_oracle_randomized_sweep():
+    """Wider randomized sweep: many seeds x opcodes x shapes against the
+    brute force, including single-key builds, wide key spaces (mostly
+    misses), and negative/positive value mixes."""
+    shapes = [
+        (30, 80, 1, 1, 10),      # one key only
+        (50, 120, 1, 200, 40),   # sparse keys, mostly misses
+        (120, 150, 1, 6, 5),     # dense keys, heavy duplicates
+    ]
+    for opcode in OPCODES:
+        for si, (nb, np_, klo, khi, span) in enumerate(shapes):
+            for seed in range(3):
+                bk, ba, pk, pa = _case(7000 + opcode * 100 + si * 10 + seed,
+                                       nb, np_, klo, khi, span)
+                got = orc.asof_inner_join(bk, ba, pk, pa, opcode)
+                want = brute_force(bk, ba, pk, pa, opcode)
+                assert np.array_equal(got, want), (OPCODES[opcode], si, seed)
