@@ -656,6 +656,19 @@ int gpue_asof_probe_emit_i32(gpue_session* s, gpue_asof_table* t, gpue_dbuf* pro
                              gpue_dbuf* probe_asof, uint64_t n_rows, int mode,
                              gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
                              uint64_t* match_count);
+/* Nullable variants: is_nulls is a (row_count+1) u8 mask — the caller ORs the
+ * equi-key and temporal null masks, and flagged build rows are skipped
+ * exactly as the reference's is_null_row does
+ * (join_hash_table_descriptor.h:447-456); null probe rows never match
+ * (INNER drops them, LEFT_OUTER emits build row 0). */
+int gpue_asof_build_nulls_i32(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* asof,
+                              gpue_dbuf* is_nulls, uint64_t row_count, int opcode,
+                              gpue_asof_table** out);
+int gpue_asof_probe_emit_nulls_i32(gpue_session* s, gpue_asof_table* t,
+                                   gpue_dbuf* probe_keys, gpue_dbuf* probe_asof,
+                                   gpue_dbuf* probe_nulls, uint64_t n_rows, int mode,
+                                   gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                                   uint64_t* match_count);
 int gpue_asof_table_destroy(gpue_asof_table* t);
 
 /* ---- event timing on the session stream (bench roofline evidence) ---- */

@@ -2724,29 +2724,41 @@ static uint32_t orc_asof_find(const OrcAsofEntry* e, uint32_t len, int64_t probe
     return low < len ? e[low].row : 0;
 }
 
-void orc_asof_inner_join(const int32_t* build_keys, const int64_t* build_asof,
-                         uint32_t build_rows, const int32_t* probe_keys,
-                         const int64_t* probe_asof, uint64_t n, int opcode,
-                         uint32_t* out_build) {
+/* nulls variant: build rows flagged in build_nulls (1-based; the equi-key
+ * and temporal masks ORed by the caller) are skipped per is_null_row
+ * (join_hash_table_descriptor.h:447-456); null probe rows never match. */
+void orc_asof_inner_join_nulls(const int32_t* build_keys, const int64_t* build_asof,
+                               const uint8_t* build_nulls, uint32_t build_rows,
+                               const int32_t* probe_keys, const int64_t* probe_asof,
+                               const uint8_t* probe_nulls, uint64_t n, int opcode,
+                               uint32_t* out_build) {
     OrcAsofEntry* e = (OrcAsofEntry*)malloc((size_t)build_rows * sizeof(OrcAsofEntry));
+    uint32_t m = 0;
     for (uint32_t i = 0; i < build_rows; i++) { /* rows are 1-based */
-        e[i].key = build_keys[i + 1];
-        e[i].v = build_asof[i + 1];
-        e[i].row = i + 1;
+        if (build_nulls && build_nulls[i + 1]) continue;
+        e[m].key = build_keys[i + 1];
+        e[m].v = build_asof[i + 1];
+        e[m].row = i + 1;
+        m++;
     }
+    uint32_t n_live = m;
     g_asof_desc = opcode >= 2; /* is_descending = GE||GT (:67) */
-    qsort(e, build_rows, sizeof(OrcAsofEntry), orc_asof_entry_cmp);
+    qsort(e, n_live, sizeof(OrcAsofEntry), orc_asof_entry_cmp);
     for (uint64_t i = 0; i < n; i++) {
+        if (probe_nulls && probe_nulls[i]) {
+            out_build[i] = 0;
+            continue;
+        }
         int32_t k = probe_keys[i];
         /* binary search the sorted-by-key entry array for the key's run */
-        uint32_t lo = 0, hi = build_rows; /* first index with key >= k */
+        uint32_t lo = 0, hi = n_live; /* first index with key >= k */
         while (lo < hi) {
             uint32_t mid = lo + (hi - lo) / 2;
             if (e[mid].key < k) lo = mid + 1;
             else hi = mid;
         }
         uint32_t start = lo;
-        hi = build_rows; /* first index with key > k */
+        hi = n_live; /* first index with key > k */
         while (lo < hi) {
             uint32_t mid = lo + (hi - lo) / 2;
             if (e[mid].key <= k) lo = mid + 1;
@@ -2755,4 +2767,12 @@ void orc_asof_inner_join(const int32_t* build_keys, const int64_t* build_asof,
         out_build[i] = orc_asof_find(e + start, lo - start, probe_asof[i], opcode);
     }
     free(e);
+}
+
+void orc_asof_inner_join(const int32_t* build_keys, const int64_t* build_asof,
+                         uint32_t build_rows, const int32_t* probe_keys,
+                         const int64_t* probe_asof, uint64_t n, int opcode,
+                         uint32_t* out_build) {
+    orc_asof_inner_join_nulls(build_keys, build_asof, NULL, build_rows, probe_keys,
+                              probe_asof, NULL, n, opcode, out_build);
 }

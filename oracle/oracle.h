@@ -354,3 +354,11 @@ void orc_asof_inner_join(const int32_t* build_keys, const int64_t* build_asof,
                          uint32_t build_rows, const int32_t* probe_keys,
                          const int64_t* probe_asof, uint64_t n, int opcode,
                          uint32_t* out_build);
+/* nulls variant: 1-based build_nulls (equi+temporal masks ORed) skips build
+ * rows (join_hash_table_descriptor.h:447-456); null probe rows never match.
+ * Either mask may be NULL. */
+void orc_asof_inner_join_nulls(const int32_t* build_keys, const int64_t* build_asof,
+                               const uint8_t* build_nulls, uint32_t build_rows,
+                               const int32_t* probe_keys, const int64_t* probe_asof,
+                               const uint8_t* probe_nulls, uint64_t n, int opcode,
+                               uint32_t* out_build);

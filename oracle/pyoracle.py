@@ -92,6 +92,9 @@ def _decl(lib):
     lib.orc_binary_prefix_decode.argtypes = [c_vp, c_u64, c_vp, c_vp]
     lib.orc_asof_inner_join.restype = None
     lib.orc_asof_inner_join.argtypes = [c_vp, c_vp, u, c_vp, c_vp, c_u64, c_i32, c_vp]
+    lib.orc_asof_inner_join_nulls.restype = None
+    lib.orc_asof_inner_join_nulls.argtypes = [c_vp, c_vp, c_vp, u, c_vp, c_vp, c_vp,
+                                              c_u64, c_i32, c_vp]
     lib.orc_xxh3_64_4to8.restype = c_u64
     lib.orc_xxh3_64_4to8.argtypes = [c_vp, c_i32, c_u64]
     lib.orc_xxh3_hash_i32.argtypes = [c_vp, c_u64, c_vp]
@@ -821,4 +824,21 @@ def asof_inner_join(build_keys, build_asof, probe_keys, probe_asof,
     out = np.empty(len(pk), np.uint32)
     load().orc_asof_inner_join(_p(bk), _p(ba), len(bk) - 1, _p(pk), _p(pa),
                                len(pk), opcode, _p(out))
+    return out
+
+
+def asof_inner_join_nulls(build_keys, build_asof, build_nulls, probe_keys,
+                          probe_asof, probe_nulls, opcode: int) -> np.ndarray:
+    """Nulls variant (join_hash_table_descriptor.h:447-456): flagged build
+    rows skipped, null probe rows unmatched. Either mask may be None."""
+    bk = np.ascontiguousarray(build_keys, np.int32)
+    ba = np.ascontiguousarray(build_asof, np.int64)
+    pk = np.ascontiguousarray(probe_keys, np.int32)
+    pa = np.ascontiguousarray(probe_asof, np.int64)
+    bn = None if build_nulls is None else np.ascontiguousarray(build_nulls, np.uint8)
+    pn = None if probe_nulls is None else np.ascontiguousarray(probe_nulls, np.uint8)
+    out = np.empty(len(pk), np.uint32)
+    load().orc_asof_inner_join_nulls(
+        _p(bk), _p(ba), None if bn is None else _p(bn), len(bk) - 1,
+        _p(pk), _p(pa), None if pn is None else _p(pn), len(pk), opcode, _p(out))
     return out

@@ -8186,12 +8186,17 @@ __device__ static inline uint32_t asof_slot_find(uint32_t k, uint32_t log_slots,
     }
 }
 
-__global__ void k_asof_insert(const int32_t* __restrict__ keys, uint64_t row_count,
+// nulls (1-based, may be null): rows flagged null — equi key OR temporal —
+// are skipped, the reference's is_null_row (join_hash_table_descriptor.h:
+// 447-456; the caller ORs the two null masks into one)
+__global__ void k_asof_insert(const int32_t* __restrict__ keys,
+                              const uint8_t* __restrict__ nulls, uint64_t row_count,
                               uint32_t log_slots, uint32_t mask, uint64_t* slots,
                               uint32_t* counts, uint32_t* occ, uint32_t* occ_cursor) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i <= row_count;
          i += stride) {
+        if (nulls && nulls[i]) continue;
         uint32_t p = asof_slot_insert((uint32_t)keys[i], log_slots, mask, slots, occ,
                                       occ_cursor);
         atomicAdd(&counts[p], 1u);
@@ -8220,7 +8225,8 @@ __global__ void k_asof_pack_meta(const uint64_t* __restrict__ cap_offsets,
 }
 
 __global__ void k_asof_scatter(const int32_t* __restrict__ keys,
-                               const int64_t* __restrict__ asof, uint64_t row_count,
+                               const int64_t* __restrict__ asof,
+                               const uint8_t* __restrict__ nulls, uint64_t row_count,
                                uint32_t log_slots, uint32_t mask,
                                const uint64_t* __restrict__ slots,
                                const uint2* __restrict__ meta, uint32_t* __restrict__ cursors,
@@ -8228,6 +8234,7 @@ __global__ void k_asof_scatter(const int32_t* __restrict__ keys,
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = 1 + (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i <= row_count;
          i += stride) {
+        if (nulls && nulls[i]) continue;
         uint32_t p = asof_slot_find((uint32_t)keys[i], log_slots, mask, slots);
         uint32_t pos = meta[p].x + atomicAdd(&cursors[p], 1u);
         vals[pos] = asof[i];
@@ -8354,7 +8361,8 @@ __device__ static inline uint32_t asof_find(const int64_t* __restrict__ vals,
 
 template <int OP>
 __global__ void k_asof_probe_count(const int32_t* __restrict__ pkeys,
-                                   const int64_t* __restrict__ pasof, uint64_t n,
+                                   const int64_t* __restrict__ pasof,
+                                   const uint8_t* __restrict__ pnulls, uint64_t n,
                                    uint32_t log_slots, uint32_t mask,
                                    const uint64_t* __restrict__ slots,
                                    const uint2* __restrict__ meta,
@@ -8363,11 +8371,13 @@ __global__ void k_asof_probe_count(const int32_t* __restrict__ pkeys,
                                    uint32_t* __restrict__ row_counts) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
-        uint32_t p = asof_slot_find((uint32_t)pkeys[i], log_slots, mask, slots);
         uint32_t match = 0;
-        if (p != 0xFFFFFFFFu) {
-            uint2 m = meta[p];
-            match = asof_find<OP>(vals, rows, m.x, m.y, pasof[i]);
+        if (!(pnulls && pnulls[i])) { // null probe rows never match
+            uint32_t p = asof_slot_find((uint32_t)pkeys[i], log_slots, mask, slots);
+            if (p != 0xFFFFFFFFu) {
+                uint2 m = meta[p];
+                match = asof_find<OP>(vals, rows, m.x, m.y, pasof[i]);
+            }
         }
         row_counts[i] = match ? 1u : (mode == GPUE_JOIN_LEFT_OUTER ? 1u : 0u);
     }
@@ -8375,7 +8385,8 @@ __global__ void k_asof_probe_count(const int32_t* __restrict__ pkeys,
 
 template <int OP>
 __global__ void k_asof_probe_emit(const int32_t* __restrict__ pkeys,
-                                  const int64_t* __restrict__ pasof, uint64_t n,
+                                  const int64_t* __restrict__ pasof,
+                                  const uint8_t* __restrict__ pnulls, uint64_t n,
                                   uint32_t log_slots, uint32_t mask,
                                   const uint64_t* __restrict__ slots,
                                   const uint2* __restrict__ meta,
@@ -8388,11 +8399,13 @@ __global__ void k_asof_probe_emit(const int32_t* __restrict__ pkeys,
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
     for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
         if (row_counts[i] == 0) continue;
-        uint32_t p = asof_slot_find((uint32_t)pkeys[i], log_slots, mask, slots);
         uint32_t match = 0;
-        if (p != 0xFFFFFFFFu) {
-            uint2 m = meta[p];
-            match = asof_find<OP>(vals, rows, m.x, m.y, pasof[i]);
+        if (!(pnulls && pnulls[i])) {
+            uint32_t p = asof_slot_find((uint32_t)pkeys[i], log_slots, mask, slots);
+            if (p != 0xFFFFFFFFu) {
+                uint2 m = meta[p];
+                match = asof_find<OP>(vals, rows, m.x, m.y, pasof[i]);
+            }
         }
         uint64_t pos = row_offsets[i];
         out_probe[pos] = (uint32_t)i;
@@ -8400,8 +8413,9 @@ __global__ void k_asof_probe_emit(const int32_t* __restrict__ pkeys,
     }
 }
 
-int gpue_asof_build_i32(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* asof,
-                        uint64_t row_count, int opcode, gpue_asof_table** out) {
+static int asof_build_impl(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* asof,
+                           const uint8_t* d_nulls, uint64_t row_count, int opcode,
+                           gpue_asof_table** out) {
     ARG_CHECK(s && keys && asof && out && row_count > 0 && row_count + 1 < (1ull << 31));
     ARG_CHECK(opcode >= GPUE_ASOF_LT && opcode <= GPUE_ASOF_GE);
     ARG_CHECK(keys->bytes >= (row_count + 1) * 4 && asof->bytes >= (row_count + 1) * 8);
@@ -8431,8 +8445,8 @@ int gpue_asof_build_i32(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* asof,
     HIP_CHECK(hipMemsetAsync(d_misc, 0, (t->n_slots + 1) * 4, s->stream));
     uint32_t nb_rows = grid_for(row_count);
     hipLaunchKernelGGL(k_asof_insert, dim3(nb_rows), dim3(BLOCK), 0, s->stream,
-                       (const int32_t*)keys->ptr, row_count, t->log_slots, mask, t->slots,
-                       d_counts, d_occ, d_misc);
+                       (const int32_t*)keys->ptr, d_nulls, row_count, t->log_slots, mask,
+                       t->slots, d_counts, d_occ, d_misc);
     // pow2 arena capacities, block-scanned to segment starts
     uint32_t nb_slots = grid_for(t->n_slots);
     uint64_t tile = (t->n_slots + nb_slots - 1) / nb_slots;
@@ -8456,9 +8470,9 @@ int gpue_asof_build_i32(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* asof,
     HIP_CHECK(hipMalloc(&t->asof_vals, (arena ? arena : 1) * 8));
     HIP_CHECK(hipMalloc(&t->asof_rows, (arena ? arena : 1) * 4));
     hipLaunchKernelGGL(k_asof_scatter, dim3(nb_rows), dim3(BLOCK), 0, s->stream,
-                       (const int32_t*)keys->ptr, (const int64_t*)asof->ptr, row_count,
-                       t->log_slots, mask, t->slots, t->meta, d_misc + 1, t->asof_vals,
-                       t->asof_rows);
+                       (const int32_t*)keys->ptr, (const int64_t*)asof->ptr, d_nulls,
+                       row_count, t->log_slots, mask, t->slots, t->meta, d_misc + 1,
+                       t->asof_vals, t->asof_rows);
     int descending = opcode >= GPUE_ASOF_GT; // is_descending = GE||GT (:67)
     uint32_t nb_sort = n_occ < MAX_GRID ? (n_occ ? n_occ : 1) : (uint32_t)MAX_GRID;
     hipLaunchKernelGGL(k_asof_sort_segments, dim3(nb_sort), dim3(BLOCK), 0, s->stream,
@@ -8475,10 +8489,23 @@ int gpue_asof_build_i32(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* asof,
     return GPUE_OK;
 }
 
-int gpue_asof_probe_emit_i32(gpue_session* s, gpue_asof_table* t, gpue_dbuf* probe_keys,
-                             gpue_dbuf* probe_asof, uint64_t n_rows, int mode,
-                             gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
-                             uint64_t* match_count) {
+int gpue_asof_build_i32(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* asof,
+                        uint64_t row_count, int opcode, gpue_asof_table** out) {
+    return asof_build_impl(s, keys, asof, nullptr, row_count, opcode, out);
+}
+
+int gpue_asof_build_nulls_i32(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* asof,
+                              gpue_dbuf* is_nulls, uint64_t row_count, int opcode,
+                              gpue_asof_table** out) {
+    ARG_CHECK(s && is_nulls && is_nulls->bytes >= row_count + 1);
+    return asof_build_impl(s, keys, asof, (const uint8_t*)is_nulls->ptr, row_count,
+                           opcode, out);
+}
+
+static int asof_probe_impl(gpue_session* s, gpue_asof_table* t, gpue_dbuf* probe_keys,
+                           gpue_dbuf* probe_asof, const uint8_t* d_pnulls, uint64_t n_rows,
+                           int mode, gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                           uint64_t* match_count) {
     ARG_CHECK(s && t && probe_keys && probe_asof && match_count);
     ARG_CHECK(mode == GPUE_JOIN_INNER || mode == GPUE_JOIN_LEFT_OUTER);
     ARG_CHECK(probe_keys->bytes >= n_rows * 4 && probe_asof->bytes >= n_rows * 8);
@@ -8493,25 +8520,25 @@ int gpue_asof_probe_emit_i32(gpue_session* s, gpue_asof_table* t, gpue_dbuf* pro
     switch (t->opcode) {
     case GPUE_ASOF_LT:
         hipLaunchKernelGGL(k_asof_probe_count<0>, dim3(nb), dim3(BLOCK), 0, s->stream,
-                           (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr,
+                           (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr, d_pnulls,
                            n_rows, t->log_slots, mask, t->slots, t->meta, t->asof_vals,
                            t->asof_rows, mode, d_counts);
         break;
     case GPUE_ASOF_LE:
         hipLaunchKernelGGL(k_asof_probe_count<1>, dim3(nb), dim3(BLOCK), 0, s->stream,
-                           (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr,
+                           (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr, d_pnulls,
                            n_rows, t->log_slots, mask, t->slots, t->meta, t->asof_vals,
                            t->asof_rows, mode, d_counts);
         break;
     case GPUE_ASOF_GT:
         hipLaunchKernelGGL(k_asof_probe_count<2>, dim3(nb), dim3(BLOCK), 0, s->stream,
-                           (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr,
+                           (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr, d_pnulls,
                            n_rows, t->log_slots, mask, t->slots, t->meta, t->asof_vals,
                            t->asof_rows, mode, d_counts);
         break;
     default:
         hipLaunchKernelGGL(k_asof_probe_count<3>, dim3(nb), dim3(BLOCK), 0, s->stream,
-                           (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr,
+                           (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr, d_pnulls,
                            n_rows, t->log_slots, mask, t->slots, t->meta, t->asof_vals,
                            t->asof_rows, mode, d_counts);
         break;
@@ -8532,7 +8559,7 @@ int gpue_asof_probe_emit_i32(gpue_session* s, gpue_asof_table* t, gpue_dbuf* pro
         case GPUE_ASOF_LT:
             hipLaunchKernelGGL(k_asof_probe_emit<0>, dim3(nb), dim3(BLOCK), 0, s->stream,
                                (const int32_t*)probe_keys->ptr,
-                               (const int64_t*)probe_asof->ptr, n_rows, t->log_slots, mask,
+                               (const int64_t*)probe_asof->ptr, d_pnulls, n_rows, t->log_slots, mask,
                                t->slots, t->meta, t->asof_vals, t->asof_rows, mode,
                                d_counts, d_offsets, (uint32_t*)out_probe_idx->ptr,
                                (uint32_t*)out_build_idx->ptr);
@@ -8540,7 +8567,7 @@ int gpue_asof_probe_emit_i32(gpue_session* s, gpue_asof_table* t, gpue_dbuf* pro
         case GPUE_ASOF_LE:
             hipLaunchKernelGGL(k_asof_probe_emit<1>, dim3(nb), dim3(BLOCK), 0, s->stream,
                                (const int32_t*)probe_keys->ptr,
-                               (const int64_t*)probe_asof->ptr, n_rows, t->log_slots, mask,
+                               (const int64_t*)probe_asof->ptr, d_pnulls, n_rows, t->log_slots, mask,
                                t->slots, t->meta, t->asof_vals, t->asof_rows, mode,
                                d_counts, d_offsets, (uint32_t*)out_probe_idx->ptr,
                                (uint32_t*)out_build_idx->ptr);
@@ -8548,7 +8575,7 @@ int gpue_asof_probe_emit_i32(gpue_session* s, gpue_asof_table* t, gpue_dbuf* pro
         case GPUE_ASOF_GT:
             hipLaunchKernelGGL(k_asof_probe_emit<2>, dim3(nb), dim3(BLOCK), 0, s->stream,
                                (const int32_t*)probe_keys->ptr,
-                               (const int64_t*)probe_asof->ptr, n_rows, t->log_slots, mask,
+                               (const int64_t*)probe_asof->ptr, d_pnulls, n_rows, t->log_slots, mask,
                                t->slots, t->meta, t->asof_vals, t->asof_rows, mode,
                                d_counts, d_offsets, (uint32_t*)out_probe_idx->ptr,
                                (uint32_t*)out_build_idx->ptr);
@@ -8556,7 +8583,7 @@ int gpue_asof_probe_emit_i32(gpue_session* s, gpue_asof_table* t, gpue_dbuf* pro
         default:
             hipLaunchKernelGGL(k_asof_probe_emit<3>, dim3(nb), dim3(BLOCK), 0, s->stream,
                                (const int32_t*)probe_keys->ptr,
-                               (const int64_t*)probe_asof->ptr, n_rows, t->log_slots, mask,
+                               (const int64_t*)probe_asof->ptr, d_pnulls, n_rows, t->log_slots, mask,
                                t->slots, t->meta, t->asof_vals, t->asof_rows, mode,
                                d_counts, d_offsets, (uint32_t*)out_probe_idx->ptr,
                                (uint32_t*)out_build_idx->ptr);
@@ -8568,6 +8595,25 @@ int gpue_asof_probe_emit_i32(gpue_session* s, gpue_asof_table* t, gpue_dbuf* pro
     (void)hipFree(d_counts);
     (void)hipFree(d_bsums);
     return GPUE_OK;
+}
+
+int gpue_asof_probe_emit_i32(gpue_session* s, gpue_asof_table* t, gpue_dbuf* probe_keys,
+                             gpue_dbuf* probe_asof, uint64_t n_rows, int mode,
+                             gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                             uint64_t* match_count) {
+    return asof_probe_impl(s, t, probe_keys, probe_asof, nullptr, n_rows, mode,
+                           out_probe_idx, out_build_idx, match_count);
+}
+
+int gpue_asof_probe_emit_nulls_i32(gpue_session* s, gpue_asof_table* t,
+                                   gpue_dbuf* probe_keys, gpue_dbuf* probe_asof,
+                                   gpue_dbuf* probe_nulls, uint64_t n_rows, int mode,
+                                   gpue_dbuf* out_probe_idx, gpue_dbuf* out_build_idx,
+                                   uint64_t* match_count) {
+    ARG_CHECK(s && probe_nulls && probe_nulls->bytes >= n_rows);
+    return asof_probe_impl(s, t, probe_keys, probe_asof,
+                           (const uint8_t*)probe_nulls->ptr, n_rows, mode, out_probe_idx,
+                           out_build_idx, match_count);
 }
 
 int gpue_asof_table_destroy(gpue_asof_table* t) {

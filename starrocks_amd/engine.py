@@ -202,6 +202,11 @@ def _declare(lib):
                                           ctypes.POINTER(ctypes.c_float)]),
         "gpue_page_decode_bshuf_lz4_i32": (c_i32, [c_vp, c_vp, c_u32, c_vp]),
         "gpue_asof_build_i32": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp]),
+        "gpue_asof_build_nulls_i32": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_i32,
+                                              c_vp]),
+        "gpue_asof_probe_emit_nulls_i32": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_vp, c_u64,
+                                                   c_i32, c_vp, c_vp,
+                                                   ctypes.POINTER(c_u64)]),
         "gpue_asof_probe_emit_i32": (c_i32, [c_vp, c_vp, c_vp, c_vp, c_u64, c_i32,
                                              c_vp, c_vp, ctypes.POINTER(c_u64)]),
         "gpue_asof_table_destroy": (c_i32, [c_vp]),
@@ -538,6 +543,27 @@ class Engine:
         _ck(self._lib, self._lib.gpue_asof_probe_emit_i32(
             self._h, table._h, probe_keys._h, probe_asof._h, n_rows, mode, op, ob,
             ctypes.byref(cnt)))
+        return cnt.value
+
+    def asof_build_nulls(self, keys: DBuf, asof: DBuf, nulls: DBuf, row_count,
+                         opcode) -> AsofTable:
+        """Nullable build: nulls is the (row_count+1) u8 OR of the equi-key
+        and temporal null masks; flagged rows are skipped
+        (join_hash_table_descriptor.h:447-456)."""
+        h = c_vp()
+        _ck(self._lib, self._lib.gpue_asof_build_nulls_i32(
+            self._h, keys._h, asof._h, nulls._h, row_count, opcode, ctypes.byref(h)))
+        return AsofTable(self, h)
+
+    def asof_probe_emit_nulls(self, table: AsofTable, probe_keys: DBuf,
+                              probe_asof: DBuf, probe_nulls: DBuf, n_rows, mode=0,
+                              out_probe=None, out_build=None) -> int:
+        cnt = c_u64()
+        op = out_probe._h if out_probe else None
+        ob = out_build._h if out_build else None
+        _ck(self._lib, self._lib.gpue_asof_probe_emit_nulls_i32(
+            self._h, table._h, probe_keys._h, probe_asof._h, probe_nulls._h, n_rows,
+            mode, op, ob, ctypes.byref(cnt)))
         return cnt.value
 
     def join_build_bucket_chained_nulls(self, keys: DBuf, nulls: DBuf, row_count):

@@ -178,3 +178,59 @@ def test_oracle_randomized_sweep():
                 got = orc.asof_inner_join(bk, ba, pk, pa, opcode)
                 want = brute_force(bk, ba, pk, pa, opcode)
                 assert np.array_equal(got, want), (OPCODES[opcode], si, seed)
+
+
+def test_oracle_nulls_vs_bruteforce():
+    """Nulls variant: flagged build rows skipped, null probe rows unmatched
+    (join_hash_table_descriptor.h:447-456) — checked by masking inside the
+    brute force's input instead."""
+    rng = np.random.default_rng(99)
+    for opcode in OPCODES:
+        bk, ba, pk, pa = _case(900 + opcode, 80, 250, 1, 10, 30)
+        bn = np.concatenate([[0], rng.integers(0, 2, len(bk) - 1)]).astype(np.uint8)
+        pn = rng.integers(0, 2, len(pk)).astype(np.uint8)
+        got = orc.asof_inner_join_nulls(bk, ba, bn, pk, pa, pn, opcode)
+        # brute force: null build rows get a key no probe carries; null
+        # probe rows zeroed after the fact
+        bk_m = bk.copy()
+        bk_m[bn != 0] = -(10**9)
+        want = brute_force(bk_m, ba, pk, pa, opcode)
+        want[pn != 0] = 0
+        assert np.array_equal(got, want), OPCODES[opcode]
+        # None masks == the plain entry point
+        assert np.array_equal(orc.asof_inner_join_nulls(bk, ba, None, pk, pa, None, opcode),
+                              orc.asof_inner_join(bk, ba, pk, pa, opcode))
+
+
+@pytest.mark.gpu
+def test_gpu_nulls_parity(engine):
+    rng = np.random.default_rng(321)
+    for opcode in (0, 3):  # one ascending, one descending
+        n_build, n_probe = 100_000, 1_000_000
+        bk = np.concatenate([[0], rng.integers(1, 20_001, n_build)]).astype(np.int32)
+        ba = np.concatenate([[0], rng.integers(-10**9, 10**9, n_build)]).astype(np.int64)
+        bn = np.concatenate([[0], (rng.random(n_build) < 0.3)]).astype(np.uint8)
+        pk = rng.integers(1, 20_001, n_probe).astype(np.int32)
+        pa = rng.integers(-10**9, 10**9, n_probe).astype(np.int64)
+        pn = (rng.random(n_probe) < 0.2).astype(np.uint8)
+        want = orc.asof_inner_join_nulls(bk, ba, bn, pk, pa, pn, opcode)
+
+        kb = engine.alloc(bk.nbytes); kb.h2d(bk)
+        ab = engine.alloc(ba.nbytes); ab.h2d(ba)
+        nb = engine.alloc(bn.nbytes); nb.h2d(bn)
+        t = engine.asof_build_nulls(kb, ab, nb, n_build, opcode)
+        pkb = engine.alloc(pk.nbytes); pkb.h2d(pk)
+        pab = engine.alloc(pa.nbytes); pab.h2d(pa)
+        pnb = engine.alloc(pn.nbytes); pnb.h2d(pn)
+        cnt = engine.asof_probe_emit_nulls(t, pkb, pab, pnb, n_probe, 3)
+        assert cnt == n_probe
+        op = engine.alloc(cnt * 4)
+        ob = engine.alloc(cnt * 4)
+        engine.asof_probe_emit_nulls(t, pkb, pab, pnb, n_probe, 3, op, ob)
+        assert np.array_equal(ob.d2h(np.uint32, cnt), want), OPCODES[opcode]
+        # INNER count = non-null matched rows
+        cnt0 = engine.asof_probe_emit_nulls(t, pkb, pab, pnb, n_probe, 0)
+        assert cnt0 == int(np.count_nonzero(want))
+        for b in (op, ob, pkb, pab, pnb, kb, ab, nb):
+            b.free()
+        t.destroy()

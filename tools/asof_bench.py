@@ -11,8 +11,11 @@ Run: gpurun -- 'python tools/asof_bench.py > gpurun_out/asof_bench.json'
 """
 
 import json
+import os
+import sys
 import time
 
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 import numpy as np
 
 from starrocks_amd.engine import Engine
