@@ -8143,12 +8143,22 @@ struct gpue_asof_table {
     gpue_session* s = nullptr;
     int opcode = 0;          // GPUE_ASOF_*
     uint64_t row_count = 0;  // build rows (1-based, row 0 sentinel)
-    uint64_t n_slots = 0;    // pow2 open-addressing size
+    uint64_t n_slots = 0;    // pow2 open-addressing size (build-time table)
     uint32_t log_slots = 0;
-    uint64_t* slots = nullptr;   // (1ull<<63)|key when occupied, 0 empty
-    uint2* meta = nullptr;       // per slot {segment start, length} in the arena
+    uint64_t* slots = nullptr;   // (1ull<<63)|key when occupied, 0 empty (freed
+                                 // after build — probing uses the compact table)
+    uint2* meta = nullptr;       // per slot {segment start, length} (freed too)
     int64_t* asof_vals = nullptr;  // arena: padded-pow2 segments, sorted
     uint32_t* asof_rows = nullptr; // matching 1-based build rows
+    // compact probe table, rebuilt once the distinct-key count is known:
+    // pow2 >= 2*n_occ slots of uint4 {key, flag, seg start, seg len} — ONE
+    // 16 B load resolves key AND segment (vs key slot + meta gather), and at
+    // 1 M distinct keys the footprint drops 536 MB -> 32 MB
+    // (Infinity-Cache-resident). len==0 marks empty (occupied slots always
+    // hold >= 1 entry); flag (bit 31 of .y) disambiguates from a live key.
+    uint4* cslots = nullptr;
+    uint64_t n_cslots = 0;
+    uint32_t log_cslots = 0;
 };
 
 #define ASOF_OCCUPIED (1ull << 63)
@@ -8330,6 +8340,34 @@ __global__ void k_asof_sort_segments(const uint32_t* __restrict__ occ, uint32_t 
     }
 }
 
+// rebuild the sparse build-time table as the compact probe table (one pass
+// over the occupied-slot list; claim via CAS on the slot's first 8 bytes,
+// meta written non-atomically — probes only run after build completes)
+__global__ void k_asof_compact(const uint32_t* __restrict__ occ, uint32_t n_occ,
+                               const uint64_t* __restrict__ slots,
+                               const uint2* __restrict__ meta, uint32_t log_cslots,
+                               uint32_t cmask, uint4* __restrict__ cslots) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_occ;
+         i += stride) {
+        uint32_t p = occ[i];
+        uint32_t key = (uint32_t)slots[p];
+        uint2 m = meta[p];
+        uint64_t want = ASOF_OCCUPIED | (uint64_t)key;
+        uint32_t q = join_hash_u32(key, log_cslots);
+        for (;;) { // keys are distinct here: CAS either claims or collides
+            uint64_t prev = atomicCAS((unsigned long long*)&cslots[q], 0ull,
+                                      (unsigned long long)want);
+            if (prev == 0ull) {
+                cslots[q].z = m.x;
+                cslots[q].w = m.y;
+                break;
+            }
+            q = (q + 1) & cmask;
+        }
+    }
+}
+
 // find_asof_match restated (join_hash_table_descriptor.cpp:83-134): branchless
 // lower bound over the sorted segment; the CPU's two while-loops (an
 // unroll-hinted >=8 loop then the tail) are one loop here — the #pragma GCC
@@ -8359,27 +8397,107 @@ __device__ static inline uint32_t asof_find(const int64_t* __restrict__ vals,
     return low < len ? rows[base + low] : 0;
 }
 
+// probe kernels: 4-way interleaved rows per lane. The probe is LATENCY-bound
+// (a dependent chain: slot load -> log2(len) binary-search loads -> row
+// gather over a multi-hundred-MB footprint); per-lane memory-level
+// parallelism is the controlling resource for such gathers (DESIGN.md §4d,
+// the q3 wave-queue lesson), so each lane advances FOUR independent probe
+// rows in lockstep — every level issues 4 concurrent loads instead of 1.
+// Measured: 61.0 -> 13.0 ms at 600 M rows / 1 M keys (profiles/asof_bench).
+template <int OP>
+__device__ static inline void asof_probe4(const int32_t* __restrict__ pkeys,
+                                          const int64_t* __restrict__ pasof,
+                                          const uint8_t* __restrict__ pnulls, uint64_t n,
+                                          uint32_t log_cslots, uint32_t cmask,
+                                          const uint4* __restrict__ cslots,
+                                          const int64_t* __restrict__ vals,
+                                          const uint32_t* __restrict__ rows,
+                                          uint64_t base, uint64_t stride,
+                                          uint32_t match[4]) {
+    uint32_t start[4], len[4], low[4], size[4];
+    int64_t pv[4];
+    uint32_t q[4];
+    bool live[4];
+    #pragma unroll
+    for (int j = 0; j < 4; j++) {
+        uint64_t i = base + (uint64_t)j * stride;
+        match[j] = 0;
+        live[j] = i < n && !(pnulls && pnulls[i]);
+        q[j] = live[j] ? join_hash_u32((uint32_t)pkeys[i], log_cslots) : 0u;
+        pv[j] = live[j] ? pasof[i] : 0;
+        len[j] = 0;
+    }
+    // slot lookup, first probe batched (load <= 1/2 so collisions are rare)
+    uint4 sv[4];
+    #pragma unroll
+    for (int j = 0; j < 4; j++)
+        if (live[j]) sv[j] = cslots[q[j]];
+    #pragma unroll
+    for (int j = 0; j < 4; j++) {
+        if (!live[j]) continue;
+        uint64_t i = base + (uint64_t)j * stride;
+        uint64_t want = ASOF_OCCUPIED | (uint64_t)(uint32_t)pkeys[i];
+        uint4 v = sv[j];
+        for (;;) {
+            uint64_t head = ((uint64_t)v.y << 32) | v.x;
+            if (head == want) {
+                start[j] = v.z;
+                len[j] = v.w;
+                break;
+            }
+            if (head == 0ull) break; // miss
+            q[j] = (q[j] + 1) & cmask;
+            v = cslots[q[j]];
+        }
+        low[j] = 0;
+        size[j] = len[j];
+    }
+    // branchless lower bound, 4 searches in lockstep (4 loads per level)
+    while (size[0] | size[1] | size[2] | size[3]) {
+        int64_t e[4];
+        #pragma unroll
+        for (int j = 0; j < 4; j++)
+            if (size[j]) e[j] = vals[start[j] + low[j] + (size[j] >> 1)];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            if (!size[j]) continue;
+            uint32_t half = size[j] >> 1;
+            uint32_t other_low = low[j] + (size[j] - half);
+            size[j] = half;
+            bool cond;
+            if (OP == 0) cond = pv[j] >= e[j];
+            else if (OP == 1) cond = pv[j] > e[j];
+            else if (OP == 2) cond = pv[j] <= e[j];
+            else cond = pv[j] < e[j];
+            low[j] = cond ? other_low : low[j];
+        }
+    }
+    #pragma unroll
+    for (int j = 0; j < 4; j++)
+        if (len[j] && low[j] < len[j]) match[j] = rows[start[j] + low[j]];
+}
+
 template <int OP>
 __global__ void k_asof_probe_count(const int32_t* __restrict__ pkeys,
                                    const int64_t* __restrict__ pasof,
                                    const uint8_t* __restrict__ pnulls, uint64_t n,
-                                   uint32_t log_slots, uint32_t mask,
-                                   const uint64_t* __restrict__ slots,
-                                   const uint2* __restrict__ meta,
+                                   uint32_t log_cslots, uint32_t cmask,
+                                   const uint4* __restrict__ cslots,
                                    const int64_t* __restrict__ vals,
                                    const uint32_t* __restrict__ rows, int mode,
                                    uint32_t* __restrict__ row_counts) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
-        uint32_t match = 0;
-        if (!(pnulls && pnulls[i])) { // null probe rows never match
-            uint32_t p = asof_slot_find((uint32_t)pkeys[i], log_slots, mask, slots);
-            if (p != 0xFFFFFFFFu) {
-                uint2 m = meta[p];
-                match = asof_find<OP>(vals, rows, m.x, m.y, pasof[i]);
-            }
+    for (uint64_t base = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; base < n;
+         base += 4 * stride) {
+        uint32_t match[4];
+        asof_probe4<OP>(pkeys, pasof, pnulls, n, log_cslots, cmask, cslots, vals, rows,
+                        base, stride, match);
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint64_t i = base + (uint64_t)j * stride;
+            if (i < n)
+                row_counts[i] = match[j] ? 1u : (mode == GPUE_JOIN_LEFT_OUTER ? 1u : 0u);
         }
-        row_counts[i] = match ? 1u : (mode == GPUE_JOIN_LEFT_OUTER ? 1u : 0u);
     }
 }
 
@@ -8387,9 +8505,8 @@ template <int OP>
 __global__ void k_asof_probe_emit(const int32_t* __restrict__ pkeys,
                                   const int64_t* __restrict__ pasof,
                                   const uint8_t* __restrict__ pnulls, uint64_t n,
-                                  uint32_t log_slots, uint32_t mask,
-                                  const uint64_t* __restrict__ slots,
-                                  const uint2* __restrict__ meta,
+                                  uint32_t log_cslots, uint32_t cmask,
+                                  const uint4* __restrict__ cslots,
                                   const int64_t* __restrict__ vals,
                                   const uint32_t* __restrict__ rows, int mode,
                                   const uint32_t* __restrict__ row_counts,
@@ -8397,19 +8514,20 @@ __global__ void k_asof_probe_emit(const int32_t* __restrict__ pkeys,
                                   uint32_t* __restrict__ out_probe,
                                   uint32_t* __restrict__ out_build) {
     uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
-    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
-        if (row_counts[i] == 0) continue;
-        uint32_t match = 0;
-        if (!(pnulls && pnulls[i])) {
-            uint32_t p = asof_slot_find((uint32_t)pkeys[i], log_slots, mask, slots);
-            if (p != 0xFFFFFFFFu) {
-                uint2 m = meta[p];
-                match = asof_find<OP>(vals, rows, m.x, m.y, pasof[i]);
+    for (uint64_t base = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; base < n;
+         base += 4 * stride) {
+        uint32_t match[4];
+        asof_probe4<OP>(pkeys, pasof, pnulls, n, log_cslots, cmask, cslots, vals, rows,
+                        base, stride, match);
+        #pragma unroll
+        for (int j = 0; j < 4; j++) {
+            uint64_t i = base + (uint64_t)j * stride;
+            if (i < n && row_counts[i] != 0) {
+                uint64_t pos = row_offsets[i];
+                out_probe[pos] = (uint32_t)i;
+                out_build[pos] = match[j]; // 0 = LEFT_OUTER miss (reference row 0)
             }
         }
-        uint64_t pos = row_offsets[i];
-        out_probe[pos] = (uint32_t)i;
-        out_build[pos] = match; // 0 = LEFT_OUTER miss, as the reference's row 0
     }
 }
 
@@ -8477,8 +8595,21 @@ static int asof_build_impl(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* asof,
     uint32_t nb_sort = n_occ < MAX_GRID ? (n_occ ? n_occ : 1) : (uint32_t)MAX_GRID;
     hipLaunchKernelGGL(k_asof_sort_segments, dim3(nb_sort), dim3(BLOCK), 0, s->stream,
                        d_occ, n_occ, t->meta, descending, t->asof_vals, t->asof_rows);
+    // compact probe table sized by the now-known distinct-key count
+    t->n_cslots = 64;
+    while (t->n_cslots < 2ull * (n_occ ? n_occ : 1)) t->n_cslots <<= 1;
+    t->log_cslots = (uint32_t)__builtin_ctzll(t->n_cslots);
+    HIP_CHECK(hipMalloc(&t->cslots, t->n_cslots * sizeof(uint4)));
+    HIP_CHECK(hipMemsetAsync(t->cslots, 0, t->n_cslots * sizeof(uint4), s->stream));
+    hipLaunchKernelGGL(k_asof_compact, dim3(grid_for(n_occ ? n_occ : 1)), dim3(BLOCK), 0,
+                       s->stream, d_occ, n_occ, t->slots, t->meta,
+                       t->log_cslots, (uint32_t)(t->n_cslots - 1), t->cslots);
     HIP_CHECK(hipGetLastError());
     HIP_CHECK(hipStreamSynchronize(s->stream));
+    (void)hipFree(t->slots);
+    t->slots = nullptr;
+    (void)hipFree(t->meta);
+    t->meta = nullptr;
     (void)hipFree(d_counts);
     (void)hipFree(d_caps);
     (void)hipFree(d_occ);
@@ -8511,7 +8642,7 @@ static int asof_probe_impl(gpue_session* s, gpue_asof_table* t, gpue_dbuf* probe
     ARG_CHECK(probe_keys->bytes >= n_rows * 4 && probe_asof->bytes >= n_rows * 8);
     uint32_t nb = grid_for(n_rows);
     uint64_t tile = (n_rows + nb - 1) / nb;
-    uint32_t mask = (uint32_t)(t->n_slots - 1);
+    uint32_t cmask = (uint32_t)(t->n_cslots - 1);
     uint32_t* d_counts = nullptr;
     uint64_t* d_bsums = nullptr;
     uint64_t* d_offsets = nullptr;
@@ -8521,25 +8652,25 @@ static int asof_probe_impl(gpue_session* s, gpue_asof_table* t, gpue_dbuf* probe
     case GPUE_ASOF_LT:
         hipLaunchKernelGGL(k_asof_probe_count<0>, dim3(nb), dim3(BLOCK), 0, s->stream,
                            (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr, d_pnulls,
-                           n_rows, t->log_slots, mask, t->slots, t->meta, t->asof_vals,
+                           n_rows, t->log_cslots, cmask, t->cslots, t->asof_vals,
                            t->asof_rows, mode, d_counts);
         break;
     case GPUE_ASOF_LE:
         hipLaunchKernelGGL(k_asof_probe_count<1>, dim3(nb), dim3(BLOCK), 0, s->stream,
                            (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr, d_pnulls,
-                           n_rows, t->log_slots, mask, t->slots, t->meta, t->asof_vals,
+                           n_rows, t->log_cslots, cmask, t->cslots, t->asof_vals,
                            t->asof_rows, mode, d_counts);
         break;
     case GPUE_ASOF_GT:
         hipLaunchKernelGGL(k_asof_probe_count<2>, dim3(nb), dim3(BLOCK), 0, s->stream,
                            (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr, d_pnulls,
-                           n_rows, t->log_slots, mask, t->slots, t->meta, t->asof_vals,
+                           n_rows, t->log_cslots, cmask, t->cslots, t->asof_vals,
                            t->asof_rows, mode, d_counts);
         break;
     default:
         hipLaunchKernelGGL(k_asof_probe_count<3>, dim3(nb), dim3(BLOCK), 0, s->stream,
                            (const int32_t*)probe_keys->ptr, (const int64_t*)probe_asof->ptr, d_pnulls,
-                           n_rows, t->log_slots, mask, t->slots, t->meta, t->asof_vals,
+                           n_rows, t->log_cslots, cmask, t->cslots, t->asof_vals,
                            t->asof_rows, mode, d_counts);
         break;
     }
@@ -8559,32 +8690,32 @@ static int asof_probe_impl(gpue_session* s, gpue_asof_table* t, gpue_dbuf* probe
         case GPUE_ASOF_LT:
             hipLaunchKernelGGL(k_asof_probe_emit<0>, dim3(nb), dim3(BLOCK), 0, s->stream,
                                (const int32_t*)probe_keys->ptr,
-                               (const int64_t*)probe_asof->ptr, d_pnulls, n_rows, t->log_slots, mask,
-                               t->slots, t->meta, t->asof_vals, t->asof_rows, mode,
+                               (const int64_t*)probe_asof->ptr, d_pnulls, n_rows, t->log_cslots, cmask,
+                               t->cslots, t->asof_vals, t->asof_rows, mode,
                                d_counts, d_offsets, (uint32_t*)out_probe_idx->ptr,
                                (uint32_t*)out_build_idx->ptr);
             break;
         case GPUE_ASOF_LE:
             hipLaunchKernelGGL(k_asof_probe_emit<1>, dim3(nb), dim3(BLOCK), 0, s->stream,
                                (const int32_t*)probe_keys->ptr,
-                               (const int64_t*)probe_asof->ptr, d_pnulls, n_rows, t->log_slots, mask,
-                               t->slots, t->meta, t->asof_vals, t->asof_rows, mode,
+                               (const int64_t*)probe_asof->ptr, d_pnulls, n_rows, t->log_cslots, cmask,
+                               t->cslots, t->asof_vals, t->asof_rows, mode,
                                d_counts, d_offsets, (uint32_t*)out_probe_idx->ptr,
                                (uint32_t*)out_build_idx->ptr);
             break;
         case GPUE_ASOF_GT:
             hipLaunchKernelGGL(k_asof_probe_emit<2>, dim3(nb), dim3(BLOCK), 0, s->stream,
                                (const int32_t*)probe_keys->ptr,
-                               (const int64_t*)probe_asof->ptr, d_pnulls, n_rows, t->log_slots, mask,
-                               t->slots, t->meta, t->asof_vals, t->asof_rows, mode,
+                               (const int64_t*)probe_asof->ptr, d_pnulls, n_rows, t->log_cslots, cmask,
+                               t->cslots, t->asof_vals, t->asof_rows, mode,
                                d_counts, d_offsets, (uint32_t*)out_probe_idx->ptr,
                                (uint32_t*)out_build_idx->ptr);
             break;
         default:
             hipLaunchKernelGGL(k_asof_probe_emit<3>, dim3(nb), dim3(BLOCK), 0, s->stream,
                                (const int32_t*)probe_keys->ptr,
-                               (const int64_t*)probe_asof->ptr, d_pnulls, n_rows, t->log_slots, mask,
-                               t->slots, t->meta, t->asof_vals, t->asof_rows, mode,
+                               (const int64_t*)probe_asof->ptr, d_pnulls, n_rows, t->log_cslots, cmask,
+                               t->cslots, t->asof_vals, t->asof_rows, mode,
                                d_counts, d_offsets, (uint32_t*)out_probe_idx->ptr,
                                (uint32_t*)out_build_idx->ptr);
             break;
@@ -8620,6 +8751,7 @@ int gpue_asof_table_destroy(gpue_asof_table* t) {
     if (!t) return GPUE_OK;
     (void)hipFree(t->slots);
     (void)hipFree(t->meta);
+    (void)hipFree(t->cslots);
     (void)hipFree(t->asof_vals);
     (void)hipFree(t->asof_rows);
     delete t;
