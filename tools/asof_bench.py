@@ -1,11 +1,13 @@
 """ASOF join perf characterization (DESIGN.md §4c-r2 evidence).
 
-Times gpue_asof_build_i32 (claim/scan/scatter/segmented-bitonic) and the
-probe at lineorder scale: the probe streams 12 B/row (i32 key + i64 value)
-and then does one slot gather + one meta gather + ~log2(entries-per-key)
-dependent 8 B gathers per row, so its envelope sits between the pure-stream
-roofline and the divergent-gather request-rate wall (cf. the q3 bit-gather
-ubench). Prints one JSON line.
+Times gpue_asof_build_i32 (claim/scan/scatter/segmented-bitonic/compact) and
+the probe at lineorder scale: the probe streams 12 B/row (i32 key + i64
+value), then one compact-slot load + ~log2(entries-per-key) search loads +
+one row gather — ~6 divergent requests/row. Measured 10.0 Grows/s at 600 M
+rows / 1 M keys = 60 G random requests/s, the divergent-gather request
+envelope (profiles/r02_asof_kernel_stats.txt; the bit-gather ubench ceiling
+is 57.4 Gprobe/s). A compact-table fold and 4-way per-lane interleaving both
+measured flat — the wall is request rate, not latency. Prints one JSON line.
 
 Run: gpurun -- 'python tools/asof_bench.py > gpurun_out/asof_bench.json'
 """
