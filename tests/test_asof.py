@@ -234,3 +234,57 @@ def test_gpu_nulls_parity(engine):
         for b in (op, ob, pkb, pab, pnb, kb, ab, nb):
             b.free()
         t.destroy()
+
+
+@pytest.mark.gpu
+def test_gpu_asof_agg_chain(engine):
+    """Composability: ASOF join -> payload gather by build row -> generic CAS
+    hash aggregate, vs the same chain through the oracle. The 'latest value
+    as of each event, summed per key' shape ASOF joins exist for."""
+    rng = np.random.default_rng(777)
+    n_build, n_probe = 50_000, 500_000
+    bk = np.concatenate([[0], rng.integers(1, 2_001, n_build)]).astype(np.int32)
+    ba = np.concatenate([[0], rng.integers(0, 10**9, n_build)]).astype(np.int64)
+    bval = np.concatenate([[0], rng.integers(1, 1000, n_build)]).astype(np.uint32)
+    pk = rng.integers(1, 2_001, n_probe).astype(np.int32)
+    pa = rng.integers(0, 10**9, n_probe).astype(np.int64)
+
+    # oracle chain: per-probe matched build row -> gather bval -> sum by key
+    want_match = orc.asof_inner_join(bk, ba, pk, pa, 3)  # GE: latest <= probe
+    hit = want_match != 0
+    expect = np.zeros(2001, np.int64)
+    np.add.at(expect, pk[hit], bval[want_match[hit]].astype(np.int64))
+
+    # GPU chain through the generic operators
+    kb = engine.alloc(bk.nbytes); kb.h2d(bk)
+    ab = engine.alloc(ba.nbytes); ab.h2d(ba)
+    t = engine.asof_build(kb, ab, n_build, 3)
+    pkb = engine.alloc(pk.nbytes); pkb.h2d(pk)
+    pab = engine.alloc(pa.nbytes); pab.h2d(pa)
+    cnt = engine.asof_probe_emit(t, pkb, pab, n_probe, 0)
+    assert cnt == int(np.count_nonzero(hit))
+    op = engine.alloc(max(cnt, 1) * 4)
+    ob = engine.alloc(max(cnt, 1) * 4)
+    engine.asof_probe_emit(t, pkb, pab, n_probe, 0, op, ob)
+    # gather: group key by probe row, value by matched build row
+    bval_d = engine.alloc(bval.nbytes); bval_d.h2d(bval)
+    gkey = engine.alloc(max(cnt, 1) * 4)
+    gval = engine.alloc(max(cnt, 1) * 4)
+    engine.gather_u32(pkb, op, cnt, gkey)
+    engine.gather_u32(bval_d, ob, cnt, gval)
+    # widen to the agg's u64 key / i64 value layout
+    keys64 = engine.alloc(max(cnt, 1) * 8)
+    keys64.h2d(gkey.d2h(np.uint32, cnt).astype(np.uint64))
+    vals64 = engine.alloc(max(cnt, 1) * 8)
+    vals64.h2d(gval.d2h(np.uint32, cnt).astype(np.int64))
+    ok_b = engine.alloc(4096 * 8)
+    os_b = engine.alloc(4096 * 8)
+    ng = engine.hash_agg_sum_u64(keys64, vals64, cnt, ok_b, os_b, max_out=4096)
+    got = np.zeros(2001, np.int64)
+    gk = ok_b.d2h(np.uint64, ng)
+    gs = os_b.d2h(np.int64, ng)
+    got[gk.astype(np.int64)] = gs
+    assert np.array_equal(got, expect)
+    for b in (kb, ab, pkb, pab, op, ob, bval_d, gkey, gval, keys64, vals64, ok_b, os_b):
+        b.free()
+    t.destroy()
