@@ -288,3 +288,31 @@ def test_gpu_asof_agg_chain(engine):
     for b in (kb, ab, pkb, pab, op, ob, bval_d, gkey, gval, keys64, vals64, ok_b, os_b):
         b.free()
     t.destroy()
+
+
+@pytest.mark.gpu
+def test_gpu_edge_all_null_build_and_empty_probe(engine):
+    """Degenerate shapes: every build row null (no groups exist at all) and a
+    zero-row probe must both work, not crash or hang."""
+    n_build = 1_000
+    bk = np.arange(n_build + 1, dtype=np.int32)
+    ba = np.arange(n_build + 1, dtype=np.int64)
+    bn = np.ones(n_build + 1, np.uint8)
+    kb = engine.alloc(bk.nbytes); kb.h2d(bk)
+    ab = engine.alloc(ba.nbytes); ab.h2d(ba)
+    nb = engine.alloc(bn.nbytes); nb.h2d(bn)
+    t = engine.asof_build_nulls(kb, ab, nb, n_build, 0)
+    pk = np.arange(1, 101, dtype=np.int32)
+    pa = np.zeros(100, np.int64)
+    pkb = engine.alloc(pk.nbytes); pkb.h2d(pk)
+    pab = engine.alloc(pa.nbytes); pab.h2d(pa)
+    assert engine.asof_probe_emit(t, pkb, pab, 100, 0) == 0   # all miss
+    assert engine.asof_probe_emit(t, pkb, pab, 100, 3) == 100  # outer: all rows
+    assert engine.asof_probe_emit(t, pkb, pab, 0, 0) == 0      # empty probe
+    ob = engine.alloc(100 * 4)
+    op = engine.alloc(100 * 4)
+    engine.asof_probe_emit(t, pkb, pab, 100, 3, op, ob)
+    assert not np.any(ob.d2h(np.uint32, 100))
+    for b in (kb, ab, nb, pkb, pab, op, ob):
+        b.free()
+    t.destroy()
