@@ -919,6 +919,141 @@ k_filter_lookback(const int64_t* __restrict__ in, uint64_t n, int64_t theta,
     }
 }
 
+// Wave-coalesced layout variant (GPUE_FILT_WL): the baseline kernel gives
+// each THREAD 32 consecutive rows, so every 16 B load instruction touches 64
+// DISTINCT 256 B-strided cache lines — 4x the TA line-request work per byte
+// of a lane-coalesced stream (the coalesced 3-stream ubench runs 5658 GB/s
+// where the filter runs 3784). Here each WAVE owns a contiguous 2048-row
+// chunk and lanes advance through it in lockstep 1 KB steps (64 lanes x
+// 16 B contiguous per instruction); emit order is reconstructed per step
+// from two ballots (even/odd element ranks), with the running wave offset
+// chained in a wave-uniform register — same registers (v[32]), same
+// descriptor protocol and grid sizing as the baseline.
+template <int TPB>
+__global__ __launch_bounds__(TPB) void
+k_filter_lookback_wl(const int64_t* __restrict__ in, uint64_t n, int64_t theta,
+                     int64_t* __restrict__ out,
+                     unsigned long long* __restrict__ tile_desc,
+                     unsigned long long* __restrict__ ticket,
+                     unsigned long long* __restrict__ total_out,
+                     unsigned long long* __restrict__ error_out) {
+    __shared__ unsigned long long sh_excl;
+    __shared__ uint32_t wsum[TPB / WAVE];
+    constexpr int STEPS = 16;             // 16 steps x 2 values = 32 rows/thread
+    constexpr uint64_t CHUNK = (uint64_t)STEPS * 2 * WAVE; // 2048 rows per wave
+    const uint64_t TILE = (uint64_t)(TPB / WAVE) * CHUNK;
+    const uint64_t n_tiles = (n + TILE - 1) / TILE;
+    int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+    (void)ticket;
+    for (uint64_t t = blockIdx.x; t < n_tiles; t += gridDim.x) {
+        uint64_t lo = t * TILE;
+        uint64_t hi = min(lo + TILE, n);
+        uint64_t lo_w = lo + (uint64_t)wid * CHUNK;
+        int64_t v[2 * STEPS];
+        uint32_t wave_total = 0;
+        if (lo_w + CHUNK <= hi) { // full chunk: unguarded 1 KB coalesced steps
+            const longlong2* p2 = (const longlong2*)(in + lo_w);
+            #pragma unroll
+            for (int j = 0; j < STEPS; j++) {
+                longlong2 w = p2[j * WAVE + lane];
+                v[2 * j] = w.x;
+                v[2 * j + 1] = w.y;
+                wave_total += __popcll(__ballot(w.x < theta));
+                wave_total += __popcll(__ballot(w.y < theta));
+            }
+        } else {
+            #pragma unroll
+            for (int j = 0; j < STEPS; j++) {
+                uint64_t r0 = lo_w + (uint64_t)j * 2 * WAVE + 2 * lane;
+                // OOB elements get v == theta: the LT predicate is false
+                v[2 * j] = r0 < hi ? in[r0] : theta;
+                v[2 * j + 1] = r0 + 1 < hi ? in[r0 + 1] : theta;
+                wave_total += __popcll(__ballot(v[2 * j] < theta));
+                wave_total += __popcll(__ballot(v[2 * j + 1] < theta));
+            }
+        }
+        if (lane == 0) wsum[wid] = wave_total;
+        __syncthreads();
+        uint32_t wave_base = 0;
+        for (int w = 0; w < wid; w++) wave_base += wsum[w];
+        uint32_t tile_count = 0;
+        for (int w = 0; w < TPB / WAVE; w++) tile_count += wsum[w];
+        // lookback — identical protocol to the baseline kernel
+        if (wid == 0) {
+            if (t == 0) {
+                if (lane == 0) {
+                    __hip_atomic_store(&tile_desc[0], FILT_PREFIX | tile_count,
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    sh_excl = 0;
+                }
+            } else {
+                if (lane == 0)
+                    __hip_atomic_store(&tile_desc[t], FILT_AGG | tile_count,
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                unsigned long long running = 0;
+                int64_t base = (int64_t)t - 1;
+                uint64_t spins = 0;
+                for (;;) {
+                    int64_t idx = base - lane;
+                    unsigned long long d =
+                        idx >= 0 ? __hip_atomic_load(&tile_desc[idx], __ATOMIC_RELAXED,
+                                                     __HIP_MEMORY_SCOPE_AGENT)
+                                 : FILT_PREFIX;
+                    unsigned long long flag = d & ~FILT_CNT_MASK;
+                    uint64_t prefix_mask = __ballot(flag == FILT_PREFIX);
+                    uint64_t invalid_mask = __ballot(flag == 0);
+                    int first_prefix = prefix_mask ? (__ffsll((unsigned long long)prefix_mask) - 1) : WAVE;
+                    int first_invalid = invalid_mask ? (__ffsll((unsigned long long)invalid_mask) - 1) : WAVE;
+                    if (first_prefix < first_invalid) {
+                        unsigned long long contrib =
+                            (lane <= first_prefix) ? (d & FILT_CNT_MASK) : 0;
+                        for (int off = WAVE / 2; off > 0; off >>= 1)
+                            contrib += __shfl_down(contrib, off, WAVE);
+                        running += __shfl(contrib, 0, WAVE);
+                        break;
+                    }
+                    if (first_invalid == WAVE) {
+                        unsigned long long contrib = d & FILT_CNT_MASK;
+                        for (int off = WAVE / 2; off > 0; off >>= 1)
+                            contrib += __shfl_down(contrib, off, WAVE);
+                        running += __shfl(contrib, 0, WAVE);
+                        base -= WAVE;
+                        continue;
+                    }
+                    if (++spins > (1ull << 28)) {
+                        if (lane == 0) atomicOr(error_out, 1ull);
+                        running = 0;
+                        break;
+                    }
+                }
+                if (lane == 0) {
+                    __hip_atomic_store(&tile_desc[t], FILT_PREFIX | (running + tile_count),
+                                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    sh_excl = running;
+                }
+            }
+            if (lane == 0 && t == n_tiles - 1) *total_out = sh_excl + tile_count;
+        }
+        __syncthreads();
+        // in-order emit: per step, two ballots rank the 128 contiguous rows
+        // (order = (lane, element)); the wave's running offset is uniform
+        uint64_t running = sh_excl + wave_base;
+        uint64_t lml = ((uint64_t)1 << lane) - 1;
+        #pragma unroll
+        for (int j = 0; j < STEPS; j++) {
+            bool p0 = v[2 * j] < theta;
+            bool p1 = v[2 * j + 1] < theta;
+            uint64_t m0 = __ballot(p0);
+            uint64_t m1 = __ballot(p1);
+            uint32_t below = __popcll(m0 & lml) + __popcll(m1 & lml);
+            if (p0) out[running + below] = v[2 * j];
+            if (p1) out[running + below + (p0 ? 1u : 0u)] = v[2 * j + 1];
+            running += __popcll(m0) + __popcll(m1);
+        }
+        __syncthreads();
+    }
+}
+
 // Re-read emit variant (GPUE_FILT_RR) — MEASURED WORSE at every selectivity
 // (s=0.01: 2.83 vs 2.14 ms; s=0.5: 12.6 vs 4.45 — the emit re-reads miss L2
 // at an 8 GB footprint; profiles/r02_filter_items_sweep.log). Keeping the
@@ -1175,6 +1310,8 @@ int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64
     // 4x fewer lookback pipeline stages
     int items = env_cap("GPUE_FILT_ITEMS", 32);
     if (items != 8 && items != 16 && items != 32) items = 16;
+    const char* wl_env = getenv("GPUE_FILT_WL");
+    if (wl_env && atoi(wl_env)) items = 32; // the wl kernel's tile is TPB*32
     // TPB sweep (same log): 1024-thread blocks (32 K-row tiles) win at every
     // selectivity — s=0.01 2.14 ms (3784 GB/s), and single-pass now beats
     // the two-pass form even at s=0.5 (4.45 vs 4.81 ms)
@@ -1202,6 +1339,11 @@ int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64
                                                 : k_filter_lookback_pipe<16, 512>;
         else kern = items == 32 ? k_filter_lookback_pipe<32, 256>
                                 : k_filter_lookback_pipe<16, 256>;
+    }
+    const char* wl = getenv("GPUE_FILT_WL");
+    if (wl && atoi(wl)) {
+        kern = tpb == 1024 ? k_filter_lookback_wl<1024>
+               : tpb == 512 ? k_filter_lookback_wl<512> : k_filter_lookback_wl<256>;
     }
     const char* rr = getenv("GPUE_FILT_RR");
     if (rr && atoi(rr)) {
