@@ -1310,8 +1310,13 @@ int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64
     // 4x fewer lookback pipeline stages
     int items = env_cap("GPUE_FILT_ITEMS", 32);
     if (items != 8 && items != 16 && items != 32) items = 16;
+    // wave-coalesced layout is the measured default (r02: s=0.01 2.13 ->
+    // 1.73 ms, s=0.5 4.45 -> 2.64 ms — profiles/r02_filter_wl.log); an
+    // explicit GPUE_FILT_WL=0 or a non-default ITEMS sweep selects the
+    // per-thread-contiguous baseline
     const char* wl_env = getenv("GPUE_FILT_WL");
-    if (wl_env && atoi(wl_env)) items = 32; // the wl kernel's tile is TPB*32
+    bool use_wl = wl_env ? atoi(wl_env) != 0 : items == 32;
+    if (use_wl) items = 32; // the wl kernel's tile is TPB*32
     // TPB sweep (same log): 1024-thread blocks (32 K-row tiles) win at every
     // selectivity — s=0.01 2.14 ms (3784 GB/s), and single-pass now beats
     // the two-pass form even at s=0.5 (4.45 vs 4.81 ms)
@@ -1340,8 +1345,7 @@ int gpue_scan_filter_i64_lt_sp(gpue_session* s, gpue_dbuf* in, uint64_t n, int64
         else kern = items == 32 ? k_filter_lookback_pipe<32, 256>
                                 : k_filter_lookback_pipe<16, 256>;
     }
-    const char* wl = getenv("GPUE_FILT_WL");
-    if (wl && atoi(wl)) {
+    if (use_wl) {
         kern = tpb == 1024 ? k_filter_lookback_wl<1024>
                : tpb == 512 ? k_filter_lookback_wl<512> : k_filter_lookback_wl<256>;
     }
