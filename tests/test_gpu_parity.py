@@ -1498,3 +1498,29 @@ def test_partition_async_matches_sync(engine):
         assert len(np.unique(rows)) == len(rows)
     for b in (keys, ri_sync, ri_async, scr, k64, r1, r2):
         b.free()
+
+
+@pytest.mark.gpu
+def test_filter_singlepass_layout_ab(engine):
+    """Both single-pass layouts — the wave-coalesced default and the
+    per-thread-contiguous baseline (GPUE_FILT_WL=0) — must be bit-exact
+    including order on the same input."""
+    import os
+    n = 7_345_677  # non-multiple of every tile size
+    inp = engine.alloc(n * 8)
+    engine.gen_i64(inp, SEED, 9, 0, n)
+    data = inp.d2h(np.int64, n)
+    theta = int(np.quantile(data.astype(np.float64), 0.13))
+    out = engine.alloc(n * 8)
+    cnt = engine.scan_filter_i64_lt_sp(inp, n, theta, out)
+    got_wl = out.d2h(np.int64, cnt)
+    os.environ["GPUE_FILT_WL"] = "0"
+    try:
+        cnt2 = engine.scan_filter_i64_lt_sp(inp, n, theta, out)
+        got_base = out.d2h(np.int64, cnt2)
+    finally:
+        del os.environ["GPUE_FILT_WL"]
+    assert cnt == cnt2
+    assert np.array_equal(got_wl, got_base)
+    inp.free()
+    out.free()
