@@ -5814,6 +5814,8 @@ int gpue_agg_table_create(gpue_session* s, uint64_t capacity, gpue_agg_table** o
 void gpue_agg_table_destroy(gpue_agg_table* t);
 }
 
+static int agg_table_reset(gpue_agg_table* t, bool with_counts = true);
+
 int gpue_agg_table_create(gpue_session* s, uint64_t capacity, gpue_agg_table** out) {
     ARG_CHECK(s && out && capacity >= 16);
     uint64_t cap = 16;
@@ -5825,7 +5827,11 @@ int gpue_agg_table_create(gpue_session* s, uint64_t capacity, gpue_agg_table** o
     HIP_CHECK(hipMalloc(&t->counts, cap * 8));
     HIP_CHECK(hipMalloc(&t->cursor, 8));
     HIP_CHECK(hipMalloc(&t->n_groups, 8));
-    HIP_CHECK(hipMemsetAsync(t->n_groups, 0, 8, s->stream));
+    // a fresh table must be READY: pushing into uninitialized slots would
+    // walk garbage chains (a degenerate O(n*cap) crawl) and a garbage slot
+    // aliasing a real key would silently corrupt that group's sum
+    int rc = agg_table_reset(t, true);
+    if (rc != GPUE_OK) return rc;
     *out = t;
     return GPUE_OK;
 }
@@ -5842,7 +5848,7 @@ void gpue_agg_table_destroy(gpue_agg_table* t) {
     delete t;
 }
 
-static int agg_table_reset(gpue_agg_table* t, bool with_counts = true) {
+static int agg_table_reset(gpue_agg_table* t, bool with_counts) {
     HIP_CHECK(hipMemsetAsync(t->slots, 0xFF, t->cap * 8, t->s->stream));
     HIP_CHECK(hipMemsetAsync(t->sums, 0, t->cap * 8, t->s->stream));
     if (with_counts) HIP_CHECK(hipMemsetAsync(t->counts, 0, t->cap * 8, t->s->stream));
