@@ -336,7 +336,15 @@ typedef struct gpue_agg_table gpue_agg_table; /* defined with the agg section be
  * pre-aggregated partials, merge_batch semantics aggregate.h:158-168;
  * update_only=1 aggregates only rows whose group exists and marks the rest
  * in miss_mask — the selective pre-agg form); probe_hits =
- * build_hash_map_with_selection's hit count; emit = convert_hash_map_to_chunk. */
+ * build_hash_map_with_selection's hit count; emit = convert_hash_map_to_chunk.
+ * update_only is a FLAGS word: bit 0 = update-only (above); bit 1 =
+ * GPUE_AGG_SUM_ONLY — skip the per-row group-count atomic for SUM-only
+ * aggregations (the insert leg is atomic-rate bound,
+ * profiles/r02_agg_card_ubench.json; a query with no COUNT/AVG function
+ * needs no counts and the fused kernels already omit them). Under
+ * SUM_ONLY, emit's out_counts for groups touched only by such pushes
+ * read 0. */
+#define GPUE_AGG_SUM_ONLY 2
 int gpue_hash_agg_push_u64(gpue_session* s, gpue_agg_table* at, gpue_dbuf* keys,
                            gpue_dbuf* vals, gpue_dbuf* cnts, uint64_t n, int update_only,
                            gpue_dbuf* miss_mask, uint64_t* hits_out);

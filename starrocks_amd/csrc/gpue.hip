@@ -6366,14 +6366,14 @@ __global__ void k_hash_agg_push(const uint64_t* __restrict__ keys,
             if (cur == k) {
                 if (vals) {
                     atomicAdd(&sums[slot], v);
-                    atomicAdd(&counts[slot], c);
+                    if (!(update_only & 2)) atomicAdd(&counts[slot], c);
                 }
                 local_hits++;
                 if (miss_mask) miss_mask[i] = 0;
                 break;
             }
             if (cur == AGG_EMPTY) {
-                if (update_only) { // new group: leave for pass-through
+                if (update_only & 1) { // new group: leave for pass-through
                     if (miss_mask) miss_mask[i] = 1;
                     break;
                 }
@@ -6381,7 +6381,7 @@ __global__ void k_hash_agg_push(const uint64_t* __restrict__ keys,
                 if (old == AGG_EMPTY || old == k) {
                     if (vals) {
                         atomicAdd(&sums[slot], v);
-                        atomicAdd(&counts[slot], c);
+                        if (!(update_only & 2)) atomicAdd(&counts[slot], c);
                     }
                     if (old == k) local_hits++;
                     else local_claims++;
@@ -6426,6 +6426,7 @@ static int hash_agg_push_impl(gpue_session* s, gpue_agg_table* at, gpue_dbuf* ke
                               gpue_dbuf* vals, gpue_dbuf* cnts, uint64_t n, int update_only,
                               gpue_dbuf* miss_mask, uint64_t* hits_out) {
     ARG_CHECK(s && at && keys && keys->bytes >= n * 8);
+    ARG_CHECK(update_only >= 0 && update_only <= 3);
     ARG_CHECK(!vals || vals->bytes >= n * 8);
     ARG_CHECK(!cnts || cnts->bytes >= n * 8);
     ARG_CHECK(!miss_mask || miss_mask->bytes >= n);

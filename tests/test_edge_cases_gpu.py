@@ -270,3 +270,36 @@ def test_ingest_exact_chunk_multiple_gpu(engine):
     engine.ingest_destroy(ing)
     assert np.array_equal(dst.d2h(np.int32, n), host)
     dst.free()
+
+
+def test_hash_agg_sum_only_flag(engine):
+    """GPUE_AGG_SUM_ONLY (flags bit 1): sums identical to the default path,
+    the per-row count atomic skipped (counts emit as 0)."""
+    rng = np.random.default_rng(31)
+    n, ngroups = 2_000_000, 50_000
+    keys_h = rng.integers(0, ngroups, n).astype(np.uint64)
+    vals_h = rng.integers(-100, 100, n).astype(np.int64)
+    keys = engine.alloc(n * 8); keys.h2d(keys_h)
+    vals = engine.alloc(n * 8); vals.h2d(vals_h)
+    ok = engine.alloc(2 * ngroups * 8)
+    os_ = engine.alloc(2 * ngroups * 8)
+    oc = engine.alloc(2 * ngroups * 8)
+
+    def run(flags):
+        at = engine.agg_table_create(2 * ngroups)
+        engine.hash_agg_push(at, keys, vals, n, update_only=flags)
+        got = engine.hash_agg_emit(at, ok, os_, 2 * ngroups, out_counts=oc)
+        engine.agg_table_destroy(at)
+        gk = ok.d2h(np.uint64, got)
+        gs = os_.d2h(np.int64, got)
+        gc = oc.d2h(np.int64, got)
+        order = np.argsort(gk)
+        return gk[order], gs[order], gc[order]
+
+    k0, s0, c0 = run(0)
+    k2, s2, c2 = run(2)
+    assert np.array_equal(k0, k2)
+    assert np.array_equal(s0, s2)
+    assert np.all(c0 >= 1) and not np.any(c2)
+    for b in (keys, vals, ok, os_, oc):
+        b.free()
