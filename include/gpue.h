@@ -624,6 +624,10 @@ int gpue_page_decode_rle_bool(gpue_session* s, gpue_dbuf* page, uint64_t n_value
  * ascending format. */
 int gpue_page_decode_for_i32(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
                              gpue_dbuf* out);
+/* PlainPage numeric decode (plain_page.h:51,83-102,148-158): u32 LE count
+ * + raw LE values; the fallback numeric encoding (encoding_info.cpp). */
+int gpue_page_decode_plain_i32(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
+                               gpue_dbuf* out);
 /* BinaryPlainPage -> BinaryColumn (binary_plain_page.h:28-46: string body +
  * u32 absolute-offset trailer + count). The dict page's dictionary format. */
 int gpue_page_decode_binary_plain(gpue_session* s, gpue_dbuf* page, uint64_t n_values,

@@ -2776,3 +2776,27 @@ void orc_asof_inner_join(const int32_t* build_keys, const int64_t* build_asof,
     orc_asof_inner_join_nulls(build_keys, build_asof, NULL, build_rows, probe_keys,
                               probe_asof, NULL, n, opcode, out_build);
 }
+
+/* PlainPage numeric codec (plain_page.h:51,83-102,148-158): u32 LE count
+ * header + raw LE fixed-size values. The fallback encoding every numeric
+ * type can take (encoding_info.cpp). */
+uint64_t orc_plain_page_encode_i32(const int32_t* values, uint32_t n, uint8_t* out) {
+    out[0] = (uint8_t)n;
+    out[1] = (uint8_t)(n >> 8);
+    out[2] = (uint8_t)(n >> 16);
+    out[3] = (uint8_t)(n >> 24);
+    memcpy(out + 4, values, (size_t)n * 4);
+    return 4 + (uint64_t)n * 4;
+}
+
+/* returns the element count, or UINT64_MAX on a malformed page (size
+ * mismatch check, plain_page.h:148-161) */
+uint64_t orc_plain_page_decode_i32(const uint8_t* page, uint64_t page_bytes,
+                                   int32_t* values) {
+    if (page_bytes < 4) return UINT64_MAX;
+    uint32_t n = (uint32_t)page[0] | ((uint32_t)page[1] << 8) |
+                 ((uint32_t)page[2] << 16) | ((uint32_t)page[3] << 24);
+    if (page_bytes != 4 + (uint64_t)n * 4) return UINT64_MAX;
+    memcpy(values, page + 4, (size_t)n * 4);
+    return n;
+}

@@ -362,3 +362,7 @@ void orc_asof_inner_join_nulls(const int32_t* build_keys, const int64_t* build_a
                                const int32_t* probe_keys, const int64_t* probe_asof,
                                const uint8_t* probe_nulls, uint64_t n, int opcode,
                                uint32_t* out_build);
+/* PlainPage numeric codec (plain_page.h:51,83-102,148-158) */
+uint64_t orc_plain_page_encode_i32(const int32_t* values, uint32_t n, uint8_t* out);
+uint64_t orc_plain_page_decode_i32(const uint8_t* page, uint64_t page_bytes,
+                                   int32_t* values);

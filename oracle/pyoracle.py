@@ -90,6 +90,10 @@ def _decl(lib):
     lib.orc_binary_prefix_encode.argtypes = [c_vp, c_vp, u, c_vp]
     lib.orc_binary_prefix_decode.restype = c_u64
     lib.orc_binary_prefix_decode.argtypes = [c_vp, c_u64, c_vp, c_vp]
+    lib.orc_plain_page_encode_i32.restype = c_u64
+    lib.orc_plain_page_encode_i32.argtypes = [c_vp, u, c_vp]
+    lib.orc_plain_page_decode_i32.restype = c_u64
+    lib.orc_plain_page_decode_i32.argtypes = [c_vp, c_u64, c_vp]
     lib.orc_asof_inner_join.restype = None
     lib.orc_asof_inner_join.argtypes = [c_vp, c_vp, u, c_vp, c_vp, c_u64, c_i32, c_vp]
     lib.orc_asof_inner_join_nulls.restype = None
@@ -842,3 +846,18 @@ def asof_inner_join_nulls(build_keys, build_asof, build_nulls, probe_keys,
         _p(bk), _p(ba), None if bn is None else _p(bn), len(bk) - 1,
         _p(pk), _p(pa), None if pn is None else _p(pn), len(pk), opcode, _p(out))
     return out
+
+
+def plain_page_encode_i32(values) -> np.ndarray:
+    v = np.ascontiguousarray(values, np.int32)
+    out = np.zeros(4 + v.nbytes, np.uint8)
+    nb = load().orc_plain_page_encode_i32(_p(v), len(v), _p(out))
+    return out[:nb].copy()
+
+
+def plain_page_decode_i32(page, n: int) -> np.ndarray:
+    pg = np.ascontiguousarray(page, np.uint8)
+    out = np.zeros(max(n, 1), np.int32)
+    got = load().orc_plain_page_decode_i32(_p(pg), len(pg), _p(out))
+    assert got == n, got
+    return out[:n]

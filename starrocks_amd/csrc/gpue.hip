@@ -8249,6 +8249,40 @@ int gpue_topk_i64(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* vals, uint64_t n,
 }
 
 // ---------------------------------------------------------------------------
+// PlainPage numeric decode (plain_page.h:51,83-102,148-158): u32 LE count
+// header + raw LE values — the fallback encoding every numeric type can
+// take (encoding_info.cpp). Decode is a validated device copy.
+// ---------------------------------------------------------------------------
+__global__ void k_plain_copy_i32(const int32_t* __restrict__ body, uint64_t n,
+                                 int32_t* __restrict__ out) {
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+        out[i] = body[i];
+}
+
+int gpue_page_decode_plain_i32(gpue_session* s, gpue_dbuf* page, uint64_t n_values,
+                               gpue_dbuf* out) {
+    ARG_CHECK(s && page && out && page->bytes >= 4 && out->bytes >= n_values * 4);
+    uint32_t count = 0;
+    HIP_CHECK(hipMemcpyAsync(&count, page->ptr, 4, hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    if (count != n_values || page->bytes < 4 + (uint64_t)count * 4) {
+        snprintf(g_err, sizeof(g_err),
+                 "plain page malformed: count %u, expected %llu, page %llu B", count,
+                 (unsigned long long)n_values, (unsigned long long)page->bytes);
+        return GPUE_ERR_ARG;
+    }
+    // header is 4 B so the body is not 16 B aligned in the page buffer —
+    // plain i32 copy (HBM-bound either way)
+    hipLaunchKernelGGL(k_plain_copy_i32, dim3(grid_stream(n_values)), dim3(BLOCK), 0,
+                       s->stream, (const int32_t*)((const uint8_t*)page->ptr + 4),
+                       n_values, (int32_t*)out->ptr);
+    HIP_CHECK(hipGetLastError());
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    return GPUE_OK;
+}
+
+// ---------------------------------------------------------------------------
 // stream event timing (bench roofline evidence — HIP events on the session
 // stream, the stream every kernel above launches on)
 // ---------------------------------------------------------------------------

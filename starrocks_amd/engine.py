@@ -194,6 +194,7 @@ def _declare(lib):
         "gpue_sbf_test_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_i32, c_vp]),
         "gpue_topk_i64": (c_i32, [c_vp, c_vp, c_vp, c_u64, c_i32, c_vp, c_vp]),
         "gpue_page_decode_rle_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp]),
+        "gpue_page_decode_plain_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp]),
         "gpue_page_decode_rle_bool": (c_i32, [c_vp, c_vp, c_u64, c_vp]),
         "gpue_page_decode_for_i32": (c_i32, [c_vp, c_vp, c_u64, c_vp]),
         "gpue_page_decode_binary_plain": (c_i32, [c_vp, c_vp, c_u64, c_vp, c_vp]),
@@ -971,6 +972,11 @@ class Engine:
         """RLE bool page decode (bit_width 1; u8 output)."""
         _ck(self._lib, self._lib.gpue_page_decode_rle_bool(self._h, page._h,
                                                            n_values, out._h))
+
+    def page_decode_plain_i32(self, page: DBuf, n_values, out: DBuf):
+        """PlainPage numeric decode (plain_page.h:51,83-102,148-158)."""
+        _ck(self._lib, self._lib.gpue_page_decode_plain_i32(
+            self._h, page._h, n_values, out._h))
 
     def page_decode_rle_i32(self, page: DBuf, n_values, out: DBuf):
         """RLE page decode (rle_page.h + rle_encoding.h at bit_width 32)."""
