@@ -143,7 +143,8 @@ def test_gpu_ties_and_small_segments(engine):
     # tie-heavy small case across all opcodes: duplicate (key, asof) pairs
     # must resolve to the smallest build row on both sides
     for opcode in OPCODES:
-        bk, ba, pk, pa = _case(500 + opcode, 5_000, 100_000, 1, 40, 3, tie_heavy=True)
+        # negative keys included: build/probe must agree through the u32 cast
+        bk, ba, pk, pa = _case(500 + opcode, 5_000, 100_000, -15, 40, 3, tie_heavy=True)
         want = orc.asof_inner_join(bk, ba, pk, pa, opcode)
         kb = engine.alloc(bk.nbytes); kb.h2d(bk)
         ab = engine.alloc(ba.nbytes); ab.h2d(ba)
@@ -169,6 +170,7 @@ def test_oracle_randomized_sweep():
         (30, 80, 1, 1, 10),      # one key only
         (50, 120, 1, 200, 40),   # sparse keys, mostly misses
         (120, 150, 1, 6, 5),     # dense keys, heavy duplicates
+        (80, 150, -20, 20, 30),  # NEGATIVE keys (the u32-cast slot path)
     ]
     for opcode in OPCODES:
         for si, (nb, np_, klo, khi, span) in enumerate(shapes):
