@@ -141,6 +141,10 @@ struct gpue_join_table {
                                  // halves the random-gather footprint (L2 per XCD is 4 MiB)
     uint32_t* prefilter = nullptr; // 2^19-bit (64 KB) fold of `bitset` for
                                    // LDS-resident prefiltering (k_q21_star_agg_pf)
+    uint32_t* prefilter2w = nullptr; // WIDE split fold (2 x 2^19 bits,
+                                     // 128 KB): q43's 6.4 KB group array
+                                     // leaves LDS room for full-size folds
+                                     // (fpr ~1% vs the single fold's 6.1%)
     uint32_t* prefilter2 = nullptr; // split two-probe fold (2 x 2^18 bits):
                                     // words [0,8192) = i & MASK18,
                                     // [8192,16384) = (i*2654435761)>>14 —
@@ -1474,6 +1478,25 @@ __global__ void k_set_bounds(const uint32_t* __restrict__ first, uint64_t interv
 // split two-probe fold: two independent 2^18-bit arrays in one 64 KB
 // allocation — bloom math at 56 K passing keys: per-array density 0.214,
 // double-probe false-positive 0.193^2 = 3.7% vs the single 2^19 fold's 6.1%
+// wide split fold: two 2^19-bit arrays (128 KB total) for kernels whose LDS
+// budget allows full-size folds — per-array density 0.107, double-probe
+// false-positive 0.101^2 = 1.0%
+__global__ void k_build_prefilter_split_w(const uint32_t* __restrict__ bitset,
+                                          uint64_t set_interval,
+                                          uint32_t* __restrict__ pf2) {
+    const uint32_t M19 = (1u << 19) - 1;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; i < set_interval;
+         i += stride) {
+        if ((bitset[i >> 5] >> (i & 31)) & 1u) {
+            uint32_t a = (uint32_t)i & M19;
+            uint32_t b = ((uint32_t)i * 2654435761u) >> 13; // top 19 bits
+            atomicOr(&pf2[a >> 5], 1u << (a & 31));
+            atomicOr(&pf2[(1u << 14) + (b >> 5)], 1u << (b & 31));
+        }
+    }
+}
+
 __global__ void k_build_prefilter_split(const uint32_t* __restrict__ bitset,
                                         uint64_t set_interval,
                                         uint32_t* __restrict__ pf2) {
@@ -1611,6 +1634,10 @@ int gpue_join_build_payload_i32(gpue_session* s, gpue_dbuf* keys, gpue_dbuf* pay
     HIP_CHECK(hipMemsetAsync(t->prefilter2, 0, (1u << 19) / 8, s->stream));
     hipLaunchKernelGGL(k_build_prefilter_split, dim3(grid_for(set_interval)), dim3(BLOCK),
                        0, s->stream, t->bitset, set_interval, t->prefilter2);
+    HIP_CHECK(hipMalloc(&t->prefilter2w, (1u << 20) / 8));
+    HIP_CHECK(hipMemsetAsync(t->prefilter2w, 0, (1u << 20) / 8, s->stream));
+    hipLaunchKernelGGL(k_build_prefilter_split_w, dim3(grid_for(set_interval)), dim3(BLOCK),
+                       0, s->stream, t->bitset, set_interval, t->prefilter2w);
     uint32_t* d_ovf = nullptr;
     HIP_CHECK(hipMalloc(&d_ovf, sizeof(uint32_t)));
     HIP_CHECK(hipMemsetAsync(d_ovf, 0, sizeof(uint32_t), s->stream));
@@ -1659,6 +1686,7 @@ void gpue_join_table_destroy(gpue_join_table* t) {
     if (t->first16) (void)hipFree(t->first16);
     if (t->prefilter) (void)hipFree(t->prefilter);
     if (t->prefilter2) (void)hipFree(t->prefilter2);
+    if (t->prefilter2w) (void)hipFree(t->prefilter2w);
     if (t->dense_groups) (void)hipFree(t->dense_groups);
     if (t->build_keys) (void)hipFree(t->build_keys);
     if (t->key_bytes) (void)hipFree(t->key_bytes);
@@ -5080,6 +5108,111 @@ k_q43_star_agg_pfq(const int32_t* __restrict__ ck, const int32_t* __restrict__ s
         if (g[j] != 0) atomicAdd(&group_sums[j], g[j]);
 }
 
+// ---------------------------------------------------------------------------
+// Wide split two-probe q43 (GPUE_Q43_PF=6): two full 2^19-bit folds (128 KB
+// LDS — affordable because q43's group array is only 6.4 KB) cut the
+// false-maybe rate 6.1% -> ~1.0%, nearly halving drain episodes. Same
+// wave-queue structure; `prefilter` here is the table's prefilter2w.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(BLOCK_Q21) void
+k_q43_star_agg_pfq2(const int32_t* __restrict__ ck, const int32_t* __restrict__ sk,
+                   const int32_t* __restrict__ pk, const int32_t* __restrict__ od,
+                   const int32_t* __restrict__ rv, const int32_t* __restrict__ sc,
+                   uint64_t n, const uint32_t* __restrict__ prefilter,
+                   const uint32_t* __restrict__ cbits, int64_t csmin, uint64_t csint,
+                   const uint32_t* __restrict__ sbits, int64_t ssmin, uint64_t ssint,
+                   const uint16_t* __restrict__ sfirst,
+                   const uint32_t* __restrict__ pbits, int64_t psmin, uint64_t psint,
+                   const uint16_t* __restrict__ pfirst,
+                   const uint16_t* __restrict__ dfirst, int64_t dmin,
+                   unsigned long long* __restrict__ group_sums) {
+    __shared__ uint32_t pfa[PF_WORDS]; // 64 KB fold A (i & M19)
+    __shared__ uint32_t pfb[PF_WORDS]; // 64 KB fold B (hash19(i))
+    __shared__ unsigned long long g[NG_Q43];
+    __shared__ int2 wq[BLOCK_Q21 / WAVE][128]; // {pk, row32}
+    for (uint32_t w = threadIdx.x; w < PF_WORDS; w += blockDim.x) {
+        pfa[w] = prefilter[w];
+        pfb[w] = prefilter[PF_WORDS + w];
+    }
+    for (int j = threadIdx.x; j < NG_Q43; j += blockDim.x) g[j] = 0;
+    __syncthreads();
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x & (WAVE - 1);
+    uint32_t wqn = 0; // wave-uniform (ballot counts); loops below are wave-uniform
+    const uint64_t n4 = n / 4;
+    const int4* __restrict__ pk4 = (const int4*)pk;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    auto process = [&](int2 e) { // full filter chain for one queued candidate
+        uint32_t idx = (uint32_t)(e.x - psmin);
+        if (!((pbits[idx >> 5] >> (idx & 31)) & 1u)) return; // exact part filter
+        uint32_t r = (uint32_t)e.y;
+        int32_t skv = sk[r];
+        uint32_t sidx = (uint32_t)(skv - ssmin);
+        if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) return;
+        uint32_t cidx = (uint32_t)(ck[r] - csmin);
+        if (cidx >= csint || !((cbits[cidx >> 5] >> (cidx & 31)) & 1u)) return;
+        uint32_t dpay = dfirst[od[r] - dmin];
+        if (dpay == 0) return;
+        uint32_t ppay = pfirst[e.x - 1];
+        uint32_t spay = sfirst[skv - 1];
+        atomicAdd(&g[(dpay - 1) * 400 + (spay - 1) * 40 + (ppay - 1)],
+                  (unsigned long long)((int64_t)rv[r] - sc[r]));
+    };
+    auto push = [&](int32_t key, uint32_t row, bool inb) {
+        uint32_t idx = (uint32_t)(key - psmin);
+        bool in = inb & (idx < psint);
+        uint32_t ia = (in ? idx : 0u) & PF_MASK;
+        uint32_t ib = ((in ? idx : 0u) * 2654435761u) >> 13;
+        bool maybe = in & (pfa[ia >> 5] >> (ia & 31)) & (pfb[ib >> 5] >> (ib & 31)) & 1u;
+        uint64_t m = __ballot(maybe);
+        if (m) {
+            uint32_t rank = __popcll(m & ((1ull << lane) - 1));
+            if (maybe) wq[wid][wqn + rank] = make_int2(key, (int32_t)row);
+            wqn += __popcll(m);
+            if (wqn >= 64) {
+                process(wq[wid][wqn - 64 + lane]);
+                wqn -= 64;
+            }
+        }
+    };
+    uint64_t base = (uint64_t)blockIdx.x * blockDim.x + (uint64_t)wid * WAVE;
+    uint64_t i = base + lane;
+    for (; base + stride + WAVE <= n4; base += 2 * stride, i += 2 * stride) {
+        int4 pa = pk4[i];
+        uint64_t i2 = i + stride;
+        int4 pb_ = pk4[i2];
+        #pragma unroll
+        for (int j = 0; j < 4; j++) push((&pa.x)[j], (uint32_t)(i * 4 + j), true);
+        #pragma unroll
+        for (int j = 0; j < 4; j++) push((&pb_.x)[j], (uint32_t)(i2 * 4 + j), true);
+    }
+    for (; base < n4; base += stride, i += stride) {
+        bool inb = i < n4;
+        int4 p4 = inb ? pk4[i] : make_int4(0, 0, 0, 0);
+        #pragma unroll
+        for (int j = 0; j < 4; j++) push((&p4.x)[j], (uint32_t)(i * 4 + j), inb);
+    }
+    if (wqn > 0 && lane < (int)wqn) process(wq[wid][lane]);
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (uint64_t r = n4 * 4 + tid; r < n; r += stride) {
+        uint32_t pidx = (uint32_t)(pk[r] - psmin);
+        if (pidx >= psint || !((pbits[pidx >> 5] >> (pidx & 31)) & 1u)) continue;
+        uint32_t sidx = (uint32_t)(sk[r] - ssmin);
+        if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
+        uint32_t cidx = (uint32_t)(ck[r] - csmin);
+        if (cidx >= csint || !((cbits[cidx >> 5] >> (cidx & 31)) & 1u)) continue;
+        uint32_t dpay = dfirst[od[r] - dmin];
+        if (dpay == 0) continue;
+        uint32_t ppay = pfirst[pk[r] - 1];
+        uint32_t spay = sfirst[sk[r] - 1];
+        atomicAdd(&g[(dpay - 1) * 400 + (spay - 1) * 40 + (ppay - 1)],
+                  (unsigned long long)((int64_t)rv[r] - sc[r]));
+    }
+    __syncthreads();
+    for (int j = threadIdx.x; j < NG_Q43; j += blockDim.x)
+        if (g[j] != 0) atomicAdd(&group_sums[j], g[j]);
+}
+
 extern "C" int gpue_q43_star_agg_async(gpue_session* s, gpue_join_table* custs,
                                        gpue_join_table* supps, gpue_join_table* parts,
                                        gpue_join_table* dates, gpue_dbuf* ck, gpue_dbuf* sk,
@@ -5118,17 +5251,21 @@ int gpue_q43_star_agg_accum_async(gpue_session* s, gpue_join_table* custs,
     ARG_CHECK(custs->bitset && supps->bitset && parts->bitset);
     ARG_CHECK(custs->min_key == 1 && supps->min_key == 1 && parts->min_key == 1);
     const char* pfe = getenv("GPUE_Q43_PF");
-    int pfm = pfe ? atoi(pfe) : 4; // wave-queue default: 1.15 vs 1.86 ms (r02)
+    int pfm = pfe ? atoi(pfe) : 4; // wave-queue default: 1.15 vs 1.86 ms (r02);
+                                   // mode 6 = wide split-fold experiment
     bool use_pf = pfm && parts->prefilter;
-    bool use_q = pfm == 4 && parts->prefilter; // wave-queue variant
+    bool use_q = (pfm == 4 || pfm == 6) && parts->prefilter; // wave-queue forms
     int def_grid = use_q ? 256 : (use_pf ? 512 : 256); // queue: 1 block/CU
-    auto kern = use_q ? k_q43_star_agg_pfq
-                      : (use_pf ? k_q43_star_agg<true> : k_q43_star_agg<false>);
+    auto kern = pfm == 6 && parts->prefilter2w
+                    ? k_q43_star_agg_pfq2
+                    : (use_q ? k_q43_star_agg_pfq
+                             : (use_pf ? k_q43_star_agg<true> : k_q43_star_agg<false>));
     hipLaunchKernelGGL(kern, dim3(env_cap("GPUE_GRID_Q43", def_grid)), dim3(BLOCK_Q21), 0, s->stream,
                        (const int32_t*)ck->ptr, (const int32_t*)sk->ptr,
                        (const int32_t*)pk->ptr, (const int32_t*)od->ptr,
                        (const int32_t*)rv->ptr, (const int32_t*)sc->ptr, n,
-                       parts->prefilter,
+                       pfm == 6 && parts->prefilter2w ? parts->prefilter2w
+                                                      : parts->prefilter,
                        custs->bitset, custs->set_min,
                        (uint64_t)(custs->set_max - custs->set_min + 1),
                        supps->bitset, supps->set_min,
