@@ -5251,8 +5251,9 @@ int gpue_q43_star_agg_accum_async(gpue_session* s, gpue_join_table* custs,
     ARG_CHECK(custs->bitset && supps->bitset && parts->bitset);
     ARG_CHECK(custs->min_key == 1 && supps->min_key == 1 && parts->min_key == 1);
     const char* pfe = getenv("GPUE_Q43_PF");
-    int pfm = pfe ? atoi(pfe) : 4; // wave-queue default: 1.15 vs 1.86 ms (r02);
-                                   // mode 6 = wide split-fold experiment
+    int pfm = pfe ? atoi(pfe) : 6; // wide split-fold default: 0.922 vs
+                                   // 1.150 ms interleaved A/B (r02 late);
+                                   // mode 4 = single-fold wave-queue
     bool use_pf = pfm && parts->prefilter;
     bool use_q = (pfm == 4 || pfm == 6) && parts->prefilter; // wave-queue forms
     int def_grid = use_q ? 256 : (use_pf ? 512 : 256); // queue: 1 block/CU
