@@ -4614,6 +4614,110 @@ k_q21_star_agg_pfq2(const int32_t* __restrict__ pk, const int32_t* __restrict__ 
 
 
 // ---------------------------------------------------------------------------
+// Deferred-load wave-queue q21 (GPUE_Q21_PF=7, experiment): the q43 form
+// applied to q21 — stream ONLY pk (2.4 GB instead of 9.6 GB), queue {pk,row}
+// maybes through the split two-probe prefilter, and let drains load sk (4%
+// of rows), then od/rv (~0.8%) on demand. Line-fetch estimate: 2.4 GB pk +
+// ~1.15 GB sk lines + ~0.6 GB od/rv lines = ~4.1 GB — if drain issue-rate
+// holds like q43's, this beats the full-stream 1.57 ms.
+// LDS: 2x32 KB folds + 56 KB groups + 16 KB {pk,row} queues = 136 KB.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(BLOCK_Q21) void
+k_q21_star_agg_pfq3(const int32_t* __restrict__ pk, const int32_t* __restrict__ sk,
+                    const int32_t* __restrict__ od, const int32_t* __restrict__ rv,
+                    uint64_t n, const uint32_t* __restrict__ prefilter, int64_t psmin,
+                    uint64_t psint, const uint16_t* __restrict__ pfirst,
+                    const uint32_t* __restrict__ sbits, int64_t ssmin, uint64_t ssint,
+                    const uint16_t* __restrict__ dfirst, int64_t dmin,
+                    unsigned long long* __restrict__ group_sums) {
+    __shared__ uint32_t pfa[1 << 13];
+    __shared__ uint32_t pfb[1 << 13];
+    __shared__ unsigned long long g[NG_Q21];
+    __shared__ int2 wq[BLOCK_Q21 / WAVE][128]; // {pk, row32}
+    for (uint32_t w = threadIdx.x; w < (1u << 13); w += blockDim.x) {
+        pfa[w] = prefilter[w];
+        pfb[w] = prefilter[(1u << 13) + w];
+    }
+    for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x) g[j] = 0;
+    __syncthreads();
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x & (WAVE - 1);
+    uint32_t wqn = 0; // wave-uniform (ballot counts); loops are wave-uniform
+    const uint64_t n4 = n / 4;
+    const int4* __restrict__ pk4 = (const int4*)pk;
+    uint64_t stride = (uint64_t)gridDim.x * blockDim.x;
+    auto process = [&](int2 e) { // confirm cascade with on-demand loads
+        uint32_t brand1 = pfirst[e.x - 1]; // exact category filter (0 = fail)
+        if (!brand1) return;
+        uint32_t r = (uint32_t)e.y;
+        uint32_t sidx = (uint32_t)(sk[r] - ssmin);
+        if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) return;
+        uint32_t year1 = dfirst[od[r] - dmin];
+        atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)],
+                  (unsigned long long)(int64_t)rv[r]);
+    };
+    auto push = [&](int32_t key, uint32_t row, bool inb) {
+        uint32_t idx = (uint32_t)(key - psmin);
+        bool in = inb & (idx < psint);
+        uint32_t ia = (in ? idx : 0u) & ((1u << 18) - 1);
+        uint32_t ib = ((in ? idx : 0u) * 2654435761u) >> 14;
+        bool maybe = in & (pfa[ia >> 5] >> (ia & 31)) & (pfb[ib >> 5] >> (ib & 31)) & 1u;
+        uint64_t m = __ballot(maybe);
+        if (m) {
+            uint32_t rank = __popcll(m & ((1ull << lane) - 1));
+            if (maybe) wq[wid][wqn + rank] = make_int2(key, (int32_t)row);
+            wqn += __popcll(m);
+            if (wqn >= 64) {
+                process(wq[wid][wqn - 64 + lane]);
+                wqn -= 64;
+            }
+        }
+    };
+    auto ld4nt = [&](const int4* p, uint64_t i) {
+        const uint64_t* q = (const uint64_t*)(p + i);
+        uint64_t lo = __builtin_nontemporal_load(q);
+        uint64_t hi = __builtin_nontemporal_load(q + 1);
+        int4 v;
+        v.x = (int32_t)lo; v.y = (int32_t)(lo >> 32);
+        v.z = (int32_t)hi; v.w = (int32_t)(hi >> 32);
+        return v;
+    };
+    uint64_t base = (uint64_t)blockIdx.x * blockDim.x + (uint64_t)wid * WAVE;
+    uint64_t i = base + lane;
+    for (; base + stride + WAVE <= n4; base += 2 * stride, i += 2 * stride) {
+        int4 pa = ld4nt(pk4, i);
+        uint64_t i2 = i + stride;
+        int4 pb_ = ld4nt(pk4, i2);
+        #pragma unroll
+        for (int j = 0; j < 4; j++) push((&pa.x)[j], (uint32_t)(i * 4 + j), true);
+        #pragma unroll
+        for (int j = 0; j < 4; j++) push((&pb_.x)[j], (uint32_t)(i2 * 4 + j), true);
+    }
+    for (; base < n4; base += stride, i += stride) {
+        bool inb = i < n4;
+        int4 p4 = inb ? pk4[i] : make_int4(0, 0, 0, 0);
+        #pragma unroll
+        for (int j = 0; j < 4; j++) push((&p4.x)[j], (uint32_t)(i * 4 + j), inb);
+    }
+    if (wqn > 0 && lane < (int)wqn) process(wq[wid][lane]);
+    // scalar row tail (n % 4): exact path
+    uint64_t tid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (uint64_t r = n4 * 4 + tid; r < n; r += stride) {
+        uint32_t idx = (uint32_t)(pk[r] - psmin);
+        if (idx >= psint) continue;
+        uint32_t brand1 = pfirst[pk[r] - 1];
+        if (!brand1) continue;
+        uint32_t sidx = (uint32_t)(sk[r] - ssmin);
+        if (sidx >= ssint || !((sbits[sidx >> 5] >> (sidx & 31)) & 1u)) continue;
+        uint32_t year1 = dfirst[od[r] - dmin];
+        atomicAdd(&g[(year1 - 1) * 1000 + (brand1 - 1)], (unsigned long long)(int64_t)rv[r]);
+    }
+    __syncthreads();
+    for (int j = threadIdx.x; j < NG_Q21; j += blockDim.x)
+        if (g[j] != 0) atomicAdd(&group_sums[j], g[j]);
+}
+
+// ---------------------------------------------------------------------------
 // Two-stream pipelined q21 (GPUE_Q21_PIPE=1): the fused kernel's streaming
 // leg (1.57 ms) and its part-probe gather leg (2.32 ms) measured fully
 // ADDITIVE (profiles/q21_decomp.log) — so split them into an operator pair
@@ -4821,7 +4925,19 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
                                dates->first16, dates->min_key,
                                (unsigned long long*)group_sums->ptr);
         };
-        if (mode == 6) { // split two-probe prefilter experiment
+        if (mode == 7) { // deferred-load (q43-form) experiment
+            hipLaunchKernelGGL(k_q21_star_agg_pfq3, dim3(env_cap("GPUE_GRID_PF", def_grid)),
+                               dim3(tpb), 0, s->stream, (const int32_t*)pk->ptr,
+                               (const int32_t*)sk->ptr, (const int32_t*)od->ptr,
+                               (const int32_t*)rv->ptr, n, parts->prefilter2,
+                               parts->set_min,
+                               (uint64_t)(parts->set_max - parts->set_min + 1),
+                               parts->first16, supps->bitset, supps->set_min,
+                               (uint64_t)(supps->set_max - supps->set_min + 1),
+                               dates->first16, dates->min_key,
+                               (unsigned long long*)group_sums->ptr);
+        }
+        else if (mode == 6) { // split two-probe prefilter experiment
             hipLaunchKernelGGL(k_q21_star_agg_pfq2, dim3(env_cap("GPUE_GRID_PF", def_grid)),
                                dim3(tpb), 0, s->stream, (const int32_t*)pk->ptr,
                                (const int32_t*)sk->ptr, (const int32_t*)od->ptr,
