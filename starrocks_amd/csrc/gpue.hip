@@ -4899,11 +4899,12 @@ int gpue_q21_star_agg_async(gpue_session* s, gpue_join_table* parts, gpue_join_t
     HIP_CHECK(hipMemsetAsync(group_sums->ptr, 0, NG_Q21 * sizeof(int64_t), s->stream));
     if (q21_pf() && parts->prefilter) {
         const char* e = getenv("GPUE_Q21_PF");
-        int mode = e ? atoi(e) : 6; // split two-probe prefilter default:
-                                    // 1.58-1.60 vs 1.64-1.70 ms interleaved
-                                    // A/B at 200 steps (r02 late); mode 4 =
-                                    // the single-fold wave-queue (1.67 vs
-                                    // 2.25 ms over mode 1 earlier in r02)
+        int mode = e ? atoi(e) : 7; // deferred-load default (the q43 form:
+                                    // stream pk only, drain sk/od/rv on
+                                    // demand): 1.106-1.133 vs mode 6's
+                                    // 1.559-1.560 ms interleaved (r02 late).
+                                    // 6 = full-stream + split fold; 4 =
+                                    // single-fold wave-queue; 1 = prefilter
         // NT streams measured 2.267 vs 2.575 ms (frac 0.54 vs 0.47,
         // gpurun_out/q21_var_sweep.log r02) — default on; the global-group
         // 2-block variant (mode 2) measured 3.2 ms and stays for evidence
