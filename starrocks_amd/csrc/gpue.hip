@@ -7112,6 +7112,9 @@ __global__ void k_q3_order_bits_bl(const int32_t* __restrict__ ocust,
 // makes half those gathers wasted traffic, and the conditional ext/disc
 // loads serialize. COUNT is not in Q3's select list — one CAS-claim + one
 // atomicAdd per passing row.
+template <bool CT> // CT: CACHED ext/disc gathers — at ~9% survivor density
+                   // adjacent survivors share 64 B lines, and the NT hint
+                   // forces a re-fetch per survivor (the q21 mode-7 lesson)
 __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
                                const int64_t* __restrict__ ext,
                                const int64_t* __restrict__ disc,
@@ -7134,8 +7137,9 @@ __global__ void k_q3_probe_agg(const int64_t* __restrict__ lk,
         unsigned long long k = (unsigned long long)__builtin_nontemporal_load(lk + i);
         uint64_t o = k - 1;
         if (!((order_bits[o >> 5] >> (o & 31)) & 1u)) continue;
-        unsigned long long v = (unsigned long long)(__builtin_nontemporal_load(ext + i) *
-                                                    (100 - __builtin_nontemporal_load(disc + i)));
+        unsigned long long v = (unsigned long long)(
+            (CT ? ext[i] : __builtin_nontemporal_load(ext + i)) *
+            (100 - (CT ? disc[i] : __builtin_nontemporal_load(disc + i))));
         uint64_t s = ((k * 11400714819323198485ull) >> 32) & cap_mask;
         uint64_t left = cap_mask + 1;
         for (;;) {
@@ -7621,6 +7625,17 @@ extern "C" int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ex
                                    gpue_dbuf* order_bits, int32_t ship_cutoff,
                                    gpue_agg_table* at, gpue_dbuf* out_keys,
                                    gpue_dbuf* out_sums, uint64_t max_out, uint64_t* n_groups);
+
+// GPUE_Q3_CT=1: cached ext/disc gathers in the fused q3 kernel (A/B vs the
+// NT default, which keeps L2 free for the 56 MB order bitset)
+static bool q3_ct() {
+    static int v = -1;
+    if (v < 0) {
+        const char* e = getenv("GPUE_Q3_CT");
+        v = (e && atoi(e)) ? 1 : 0;
+    }
+    return v == 1;
+}
 int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf* disc,
                         gpue_dbuf* ship, uint64_t n, gpue_dbuf* order_bits,
                         int32_t ship_cutoff, gpue_agg_table* at, gpue_dbuf* out_keys,
@@ -7642,7 +7657,7 @@ int gpue_q3_probe_agg_t(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbu
                            (const uint32_t*)order_bits->ptr, ship_cutoff, at->slots,
                            at->sums, at->counts, at->cap - 1, s->d_agg_err, at->n_groups);
     else
-        hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))), dim3(BLOCK), 0, s->stream,
+        hipLaunchKernelGGL(q3_ct() ? k_q3_probe_agg<true> : k_q3_probe_agg<false>, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))), dim3(BLOCK), 0, s->stream,
                            (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
                            (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
                            (const uint32_t*)order_bits->ptr, ship_cutoff, at->slots,
@@ -7676,7 +7691,7 @@ int gpue_q3_probe_accum(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbu
                         gpue_dbuf* ship, uint64_t n, gpue_dbuf* order_bits,
                         int32_t ship_cutoff, gpue_agg_table* at) {
     ARG_CHECK(s && lk && ext && disc && ship && order_bits && at);
-    hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(q3_ct() ? k_q3_probe_agg<true> : k_q3_probe_agg<false>, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))), dim3(BLOCK), 0, s->stream,
                        (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
                        (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
                        (const uint32_t*)order_bits->ptr, ship_cutoff, at->slots, at->sums,
@@ -7845,7 +7860,7 @@ int gpue_q3_probe_agg(gpue_session* s, gpue_dbuf* lk, gpue_dbuf* ext, gpue_dbuf*
     HIP_CHECK(hipMemsetAsync(d_sums, 0, cap * 8, s->stream));
     HIP_CHECK(hipMemsetAsync(d_counts, 0, cap * 8, s->stream));
     HIP_CHECK(hipMemsetAsync(d_cursor, 0, 8, s->stream));
-    hipLaunchKernelGGL(k_q3_probe_agg, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))), dim3(BLOCK), 0, s->stream,
+    hipLaunchKernelGGL(q3_ct() ? k_q3_probe_agg<true> : k_q3_probe_agg<false>, dim3(grid_capped(n, env_cap("GPUE_GRID_Q3", MAX_GRID))), dim3(BLOCK), 0, s->stream,
                        (const int64_t*)lk->ptr, (const int64_t*)ext->ptr,
                        (const int64_t*)disc->ptr, (const int32_t*)ship->ptr, n,
                        (const uint32_t*)order_bits->ptr, ship_cutoff, d_slots, d_sums,
